@@ -1,0 +1,200 @@
+"""ModelRunner: device-side execution of scheduled batches.
+
+Decode steps are captured into hipGraphs per batch-size bucket (torch's CUDA
+graph API is hipGraph on ROCm).  The capture includes the whole transformer
+forward — custom CDNA4 kernels, hipBLASLt GEMMs and (at TP>1) the RCCL
+all-reduces — so steady-state decode replays with one graph launch instead
+of ~1k kernel launches (MI355X boundary cost ≈1.2-1.9 us each, see
+/opt/skills/guides/MI355X_MICROARCH.md 'boundary').
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional
+
+import torch
+
+from hyperspot import ops
+from hyperspot.models import build_model
+from hyperspot.models.llama import ForwardMeta
+
+from .config import EngineConfig
+from .kv_cache import (BlockManager, allocate_kv_caches,
+                       compute_num_gpu_blocks)
+from .request import Request
+from .scheduler import ScheduledBatch
+
+log = logging.getLogger(__name__)
+
+
+class ModelRunner:
+    def __init__(self, config: EngineConfig, device: Optional[str] = None):
+        self.config = config
+        self.spec = config.spec()
+        self.device = torch.device(
+            device if device is not None
+            else ("cuda" if torch.cuda.is_available() else "cpu"))
+        self.dtype = torch.bfloat16 if self.device.type == "cuda" \
+            else torch.float32
+        torch.manual_seed(config.seed)
+        self.model = build_model(self.spec, dtype=self.dtype).to(self.device)
+        self.model.eval()
+        self.num_blocks = compute_num_gpu_blocks(self.spec, config, self.device)
+        kv_heads_local = self.spec.num_kv_heads // max(config.tp_size, 1)
+        self.kv_caches = allocate_kv_caches(
+            self.spec.num_layers, self.num_blocks, kv_heads_local,
+            config.block_size, self.spec.head_dim, self.dtype, self.device)
+        self.block_manager = BlockManager(self.num_blocks, config.block_size)
+        self.max_blocks_per_seq = (config.max_model_len + config.block_size - 1) \
+            // config.block_size
+        self._gen = torch.Generator().manual_seed(config.seed)
+        self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
+        self._graph_io: Dict[int, dict] = {}
+        self._graph_pool = None
+
+    # ---------------- input preparation ----------------
+
+    def _prefill_inputs(self, batch: ScheduledBatch):
+        ids: List[int] = []
+        pos: List[int] = []
+        slots: List[int] = []
+        starts = [0]
+        logit_rows = []
+        for req in batch.requests:
+            n = req.num_prompt_tokens
+            ids.extend(req.prompt_token_ids)
+            pos.extend(range(n))
+            bm = self.block_manager
+            slots.extend(bm.slot_of(req.request_id, p) for p in range(n))
+            starts.append(starts[-1] + n)
+            logit_rows.append(starts[-1] - 1)
+        d = self.device
+        meta = ForwardMeta(
+            mode="prefill",
+            positions=torch.tensor(pos, dtype=torch.long, device=d),
+            slot_mapping=torch.tensor(slots, dtype=torch.long, device=d),
+            seq_start=torch.tensor(starts, dtype=torch.int32, device=d),
+            max_seqlen=max(r.num_prompt_tokens for r in batch.requests),
+            logits_indices=torch.tensor(logit_rows, dtype=torch.long, device=d),
+        )
+        input_ids = torch.tensor(ids, dtype=torch.long, device=d)
+        return input_ids, meta
+
+    def _decode_inputs(self, batch: ScheduledBatch):
+        bm = self.block_manager
+        ids, pos, slots, seq_lens = [], [], [], []
+        tables = []
+        for req in batch.requests:
+            ids.append(req.last_token_id)
+            p = bm.seq_tokens[req.request_id]       # next position index
+            pos.append(p)
+            slots.append(bm.append_slot(req.request_id))
+            seq_lens.append(p + 1)
+            tables.append(bm.table(req.request_id))
+        max_t = max(len(t) for t in tables)
+        bt = torch.zeros(len(tables), max_t, dtype=torch.int32)
+        for i, t in enumerate(tables):
+            bt[i, :len(t)] = torch.tensor(t, dtype=torch.int32)
+        d = self.device
+        meta = ForwardMeta(
+            mode="decode",
+            positions=torch.tensor(pos, dtype=torch.long, device=d),
+            slot_mapping=torch.tensor(slots, dtype=torch.long, device=d),
+            block_tables=bt.to(d),
+            seq_lens=torch.tensor(seq_lens, dtype=torch.int32, device=d),
+        )
+        input_ids = torch.tensor(ids, dtype=torch.long, device=d)
+        return input_ids, meta
+
+    # ---------------- execution ----------------
+
+    @torch.inference_mode()
+    def execute(self, batch: ScheduledBatch) -> torch.Tensor:
+        """Run one step; returns sampled token ids [num_seqs] (cpu)."""
+        if batch.mode == "prefill":
+            input_ids, meta = self._prefill_inputs(batch)
+            logits = self.model(input_ids, meta, self.kv_caches)
+        else:
+            input_ids, meta = self._decode_inputs(batch)
+            logits = self._decode_forward(input_ids, meta)
+        return self._sample(logits, batch.requests)
+
+    def _decode_forward(self, input_ids, meta):
+        B = input_ids.shape[0]
+        bucket = self._bucket_for(B)
+        if bucket is None or self.device.type != "cuda" \
+                or self.config.enforce_eager:
+            return self.model(input_ids, meta, self.kv_caches)
+        if bucket not in self._graphs:
+            self._capture(bucket)
+        io = self._graph_io[bucket]
+        io["input_ids"][:B] = input_ids
+        io["input_ids"][B:] = 0
+        io["positions"][:B] = meta.positions
+        io["positions"][B:] = 0
+        io["slot_mapping"][:B] = meta.slot_mapping
+        io["slot_mapping"][B:] = -1          # padded rows write nothing
+        io["block_tables"].zero_()
+        io["block_tables"][:B, :meta.block_tables.shape[1]] = meta.block_tables
+        io["seq_lens"][:B] = meta.seq_lens
+        io["seq_lens"][B:] = 1
+        self._graphs[bucket].replay()
+        return io["logits"][:B]
+
+    def _bucket_for(self, b: int) -> Optional[int]:
+        for s in self.config.graph_batch_sizes:
+            if s >= b:
+                return s
+        return None
+
+    def _capture(self, bucket: int) -> None:
+        d = self.device
+        log.info("capturing decode hipGraph for batch=%d", bucket)
+        io = {
+            "input_ids": torch.zeros(bucket, dtype=torch.long, device=d),
+            "positions": torch.zeros(bucket, dtype=torch.long, device=d),
+            "slot_mapping": torch.full((bucket,), -1, dtype=torch.long, device=d),
+            "block_tables": torch.zeros(bucket, self.max_blocks_per_seq,
+                                        dtype=torch.int32, device=d),
+            "seq_lens": torch.ones(bucket, dtype=torch.int32, device=d),
+        }
+        meta = ForwardMeta(mode="decode", positions=io["positions"],
+                           slot_mapping=io["slot_mapping"],
+                           block_tables=io["block_tables"],
+                           seq_lens=io["seq_lens"])
+        # warmup outside capture (allocator, hipBLASLt heuristics, RCCL)
+        for _ in range(2):
+            self.model(io["input_ids"], meta, self.kv_caches)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g, pool=self._graph_pool):
+            io["logits"] = self.model(io["input_ids"], meta, self.kv_caches)
+        if self._graph_pool is None:
+            self._graph_pool = g.pool()
+        self._graphs[bucket] = g
+        self._graph_io[bucket] = io
+
+    def capture_all_graphs(self) -> None:
+        if self.device.type != "cuda" or self.config.enforce_eager:
+            return
+        for b in self.config.graph_batch_sizes:
+            if b <= self.config.max_num_seqs:
+                self._capture(b)
+
+    # ---------------- sampling ----------------
+
+    def _sample(self, logits: torch.Tensor, requests: List[Request]) -> torch.Tensor:
+        B = logits.shape[0]
+        temps = torch.tensor([r.sampling.temperature for r in requests],
+                             dtype=torch.float32)
+        if bool((temps == 0).all()):
+            return ops.greedy_sample(logits).cpu()
+        top_p = torch.tensor([r.sampling.top_p for r in requests],
+                             dtype=torch.float32)
+        top_k = torch.tensor([r.sampling.top_k for r in requests],
+                             dtype=torch.int32)
+        uniform = torch.rand(B, generator=self._gen)
+        d = logits.device
+        return ops.sample(logits, temps.to(d), top_p.to(d), top_k.to(d),
+                          uniform.to(d)).cpu()

@@ -1,0 +1,121 @@
+"""Continuous-batching scheduler.
+
+Each step is EITHER a prefill batch (new sequences, whole prompts) or a
+decode batch (one token for every running sequence).  Keeping the decode
+step homogeneous is what makes it hipGraph-capturable per batch bucket
+(SURVEY.md §7 hard-part 2: graphs want static shapes — bucketed batches).
+
+Admission / fairness hooks (per-tenant quotas) are applied by the
+serverless-runtime scheduler layer above (hyperspot.serving.admission);
+this class is the device-level scheduler.
+"""
+
+from __future__ import annotations
+
+from collections import deque
+from dataclasses import dataclass, field
+from typing import Deque, List, Optional
+
+from .config import EngineConfig
+from .kv_cache import BlockManager
+from .request import Request, RequestState
+
+
+@dataclass
+class ScheduledBatch:
+    mode: str                       # "prefill" | "decode"
+    requests: List[Request] = field(default_factory=list)
+    # prefill: per-request number of tokens being processed (== prompt len)
+    num_tokens: int = 0
+
+    @property
+    def empty(self) -> bool:
+        return not self.requests
+
+
+class Scheduler:
+    def __init__(self, config: EngineConfig, block_manager: BlockManager):
+        self.config = config
+        self.bm = block_manager
+        self.waiting: Deque[Request] = deque()
+        self.running: List[Request] = []
+        # keep a couple of free blocks as headroom before admitting prefills
+        self.watermark = max(1, int(0.01 * block_manager.num_blocks))
+
+    def add(self, req: Request) -> None:
+        self.waiting.append(req)
+
+    def abort(self, request_id: str) -> None:
+        for q in (self.waiting, self.running):
+            for r in list(q):
+                if r.request_id == request_id:
+                    q.remove(r)
+                    if r.state == RequestState.RUNNING:
+                        self.bm.free(r.request_id)
+                    r.state = RequestState.FINISHED
+
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    def schedule(self) -> ScheduledBatch:
+        # Admit prefills first (throughput: keep the decode batch full).
+        batch = self._schedule_prefill()
+        if not batch.empty:
+            return batch
+        return self._schedule_decode()
+
+    def _schedule_prefill(self) -> ScheduledBatch:
+        batch = ScheduledBatch(mode="prefill")
+        budget = self.config.max_num_batched_tokens
+        while self.waiting:
+            req = self.waiting[0]
+            n = req.num_prompt_tokens
+            if n > self.config.max_model_len:
+                self.waiting.popleft()
+                req.state = RequestState.FINISHED
+                continue
+            if len(self.running) + len(batch.requests) >= self.config.max_num_seqs:
+                break
+            if n > budget:
+                break
+            if not self.bm.can_allocate(n, self.watermark):
+                break
+            self.waiting.popleft()
+            self.bm.allocate(req.request_id, n)
+            req.state = RequestState.RUNNING
+            batch.requests.append(req)
+            budget -= n
+            batch.num_tokens += n
+        if batch.requests:
+            self.running.extend(batch.requests)
+        return batch
+
+    def _schedule_decode(self) -> ScheduledBatch:
+        batch = ScheduledBatch(mode="decode")
+        # Ensure every running sequence can take one more token; preempt
+        # (recompute) the youngest sequences if the pool is out of pages.
+        while self.running:
+            # total fresh-block demand of this decode step vs the free list
+            need = sum(1 for req in self.running
+                       if self.bm.seq_tokens[req.request_id]
+                       % self.bm.block_size == 0)
+            if need <= self.bm.num_free:
+                break
+            victim = self.running.pop()          # youngest (appended last)
+            self.bm.free(victim.request_id)
+            victim.state = RequestState.PREEMPTED
+            # recompute path: prompt + generated so far becomes the new prompt
+            victim.prompt_token_ids = victim.prompt_token_ids + victim.output_token_ids
+            victim.output_token_ids = []
+            self.waiting.appendleft(victim)
+        batch.requests = list(self.running)
+        batch.num_tokens = len(batch.requests)
+        return batch
+
+    def finish_step(self) -> List[Request]:
+        """Remove finished requests from running, free their pages."""
+        done = [r for r in self.running if r.finished]
+        for r in done:
+            self.bm.free(r.request_id)
+        self.running = [r for r in self.running if not r.finished]
+        return done

@@ -1,0 +1,154 @@
+"""Op dispatch layer.
+
+On GPU (ROCm) every hot op runs a hand-written CDNA4 HIP kernel from the
+in-tree extension ``hyperspot._C`` (csrc/, built for gfx950 only).  On CPU the
+pure-PyTorch references in :mod:`hyperspot.ops.torch_ref` run instead (CI has
+no GPU).  There is no CUDA path, no Triton, no multi-backend dispatch — a CUDA
+device without the extension is a hard error, never a silent eager fallback.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from . import torch_ref
+
+_C = None
+_C_ERR: Exception | None = None
+try:  # built by `python csrc/setup.py build_ext --inplace` (gfx950)
+    from hyperspot import _C as _C  # type: ignore
+except Exception as e:  # pragma: no cover - exercised only on GPU boxes
+    _C_ERR = e
+
+
+def have_native() -> bool:
+    return _C is not None
+
+
+def _native():
+    if _C is None:
+        raise RuntimeError(
+            "hyperspot._C (gfx950 HIP extension) is not built but a GPU tensor "
+            "reached the op layer. Build it in-tree with "
+            "`python csrc/setup.py build_ext --inplace` "
+            f"(import error: {_C_ERR!r})")
+    return _C
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Tensor:
+    if x.is_cuda:
+        out = torch.empty_like(x)
+        _native().rmsnorm(out, x, weight, eps)
+        return out
+    return torch_ref.rmsnorm(x, weight, eps)
+
+
+def fused_add_rmsnorm(x: torch.Tensor, residual: torch.Tensor,
+                      weight: torch.Tensor, eps: float):
+    """residual += x (in place); x_out = rmsnorm(residual) (in place on x)."""
+    if x.is_cuda:
+        _native().fused_add_rmsnorm(x, residual, weight, eps)
+        return x, residual
+    out, res = torch_ref.fused_add_rmsnorm(x, residual, weight, eps)
+    return out, res
+
+
+def rope_kv_append(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                   positions: torch.Tensor, cos_sin: torch.Tensor,
+                   slot_mapping: torch.Tensor,
+                   k_cache: torch.Tensor, v_cache: torch.Tensor) -> None:
+    """Fused: NeoX RoPE on q,k (in place) + scatter k,v into the paged cache.
+
+    q [T,H,D], k/v [T,KV,D]; cos_sin [max_pos, D] fp32 (cos half | sin half);
+    slot_mapping [T] int64 (-1 = skip append, e.g. sliding-window drop).
+    """
+    if q.is_cuda:
+        _native().rope_kv_append(q, k, v, positions, cos_sin, slot_mapping,
+                                 k_cache, v_cache)
+        return
+    d = q.shape[-1]
+    cs = cos_sin[positions.long()]
+    cos, sin = cs[:, : d // 2], cs[:, d // 2:]
+    q.copy_(torch_ref._rotate_neox(q, cos, sin))
+    k.copy_(torch_ref._rotate_neox(k, cos, sin))
+    mask = slot_mapping >= 0
+    if mask.any():
+        torch_ref.kv_cache_append(k[mask], v[mask], k_cache, v_cache,
+                                  slot_mapping[mask])
+
+
+def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
+                      v_cache: torch.Tensor, block_tables: torch.Tensor,
+                      seq_lens: torch.Tensor, scale: float) -> torch.Tensor:
+    if q.is_cuda:
+        out = torch.empty_like(q)
+        _native().paged_attn_decode(out, q, k_cache, v_cache, block_tables,
+                                    seq_lens, scale)
+        return out
+    return torch_ref.paged_attn_decode(q, k_cache, v_cache, block_tables,
+                                       seq_lens, scale)
+
+
+def prefill_attn(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                 seq_start: torch.Tensor, max_seqlen: int,
+                 scale: float) -> torch.Tensor:
+    if q.is_cuda:
+        out = torch.empty_like(q)
+        _native().prefill_attn(out, q, k, v, seq_start, max_seqlen, scale)
+        return out
+    return torch_ref.prefill_attn(q, k, v, seq_start, scale)
+
+
+def silu_mul(gate_up: torch.Tensor) -> torch.Tensor:
+    if gate_up.is_cuda:
+        i = gate_up.shape[-1] // 2
+        out = torch.empty(*gate_up.shape[:-1], i, dtype=gate_up.dtype,
+                          device=gate_up.device)
+        _native().silu_mul(out, gate_up)
+        return out
+    return torch_ref.silu_mul(gate_up)
+
+
+def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
+    if logits.is_cuda:
+        out = torch.empty(logits.shape[0], dtype=torch.long,
+                          device=logits.device)
+        _native().greedy_sample(out, logits)
+        return out
+    return logits.float().argmax(-1)
+
+
+def sample(logits: torch.Tensor, temperature: torch.Tensor,
+           top_p: torch.Tensor, top_k: torch.Tensor,
+           uniform: torch.Tensor) -> torch.Tensor:
+    """Batch sampling. Greedy rows (temperature==0) take the argmax kernel;
+    stochastic rows go through filtering + an inverse-CDF draw."""
+    if not logits.is_cuda:
+        return torch_ref.sample(logits, temperature, top_p, top_k, uniform)
+    if bool((temperature == 0).all()):
+        return greedy_sample(logits)
+    # Stochastic path: torch does the (sort-based) top-k/top-p filtering,
+    # the HIP kernel does temperature softmax + inverse-CDF in one pass.
+    lf = logits.float()
+    B, V = lf.shape
+    t = temperature.clamp_min(1e-6)[:, None]
+    row = lf / t
+    if bool((top_k > 0).any()):
+        k = top_k.clamp(0, V)
+        kth = torch.where(
+            k > 0,
+            row.topk(int(k.max().clamp(min=1)), dim=-1).values.gather(
+                1, (k.clamp(min=1) - 1)[:, None]).squeeze(1),
+            torch.full_like(row[:, 0], float("-inf")))
+        row = torch.where(row < kth[:, None], float("-inf"), row)
+    if bool((top_p < 1.0).any()):
+        sp, idx = row.softmax(-1).sort(descending=True)
+        cum = sp.cumsum(-1)
+        drop = (cum - sp) >= top_p[:, None]
+        row = row.masked_fill(drop.gather(1, idx.argsort(-1)), float("-inf"))
+    out = torch.empty(B, dtype=torch.long, device=logits.device)
+    _native().inv_cdf_sample(out, row, uniform)
+    greedy = temperature == 0
+    if bool(greedy.any()):
+        out = torch.where(greedy, greedy_sample(logits), out)
+    return out
