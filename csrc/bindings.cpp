@@ -1,0 +1,146 @@
+// Python bindings for the gfx950 kernel set (hyperspot._C).
+//
+// Host-side glue only — every kernel lives in a .hip translation unit and
+// is reached through a plain launcher so this file needs no device code.
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_bf16.h>
+
+#include <stdexcept>
+
+using bf16 = __hip_bfloat16;
+
+void launch_rmsnorm(bf16*, const bf16*, const bf16*, float, long, int,
+                    hipStream_t);
+void launch_fused_add_rmsnorm(bf16*, bf16*, const bf16*, float, long, int,
+                              hipStream_t);
+void launch_rope_kv_append(bf16*, bf16*, const bf16*, const long*,
+                           const float*, const long*, bf16*, bf16*, long,
+                           int, int, int, int, hipStream_t);
+void launch_paged_attn(bf16*, const bf16*, const bf16*, const bf16*,
+                       const int*, const int*, const int*, long, int, int,
+                       int, int, int, float, hipStream_t);
+void launch_silu_mul(bf16*, const bf16*, long, int, hipStream_t);
+void launch_greedy_sample(long*, const bf16*, long, int, hipStream_t);
+void launch_inv_cdf_sample(long*, const float*, const float*, long, int,
+                           hipStream_t);
+
+namespace {
+
+hipStream_t stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check(const torch::Tensor& t, torch::ScalarType dt, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == dt, name, " has wrong dtype");
+}
+
+bf16* bf(torch::Tensor& t) { return reinterpret_cast<bf16*>(t.data_ptr()); }
+const bf16* cbf(const torch::Tensor& t) {
+  return reinterpret_cast<const bf16*>(t.data_ptr());
+}
+
+void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+             double eps) {
+  check(out, torch::kBFloat16, "out");
+  check(x, torch::kBFloat16, "x");
+  check(w, torch::kBFloat16, "w");
+  const int hidden = (int)x.size(-1);
+  TORCH_CHECK(hidden % 8 == 0 && hidden <= 16384, "hidden size");
+  launch_rmsnorm(bf(out), cbf(x), cbf(w), (float)eps, x.numel() / hidden,
+                 hidden, stream());
+}
+
+void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
+                       torch::Tensor w, double eps) {
+  check(x, torch::kBFloat16, "x");
+  check(residual, torch::kBFloat16, "residual");
+  const int hidden = (int)x.size(-1);
+  TORCH_CHECK(hidden % 8 == 0 && hidden <= 16384, "hidden size");
+  launch_fused_add_rmsnorm(bf(x), bf(residual), cbf(w), (float)eps,
+                           x.numel() / hidden, hidden, stream());
+}
+
+void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                    torch::Tensor positions, torch::Tensor cos_sin,
+                    torch::Tensor slot_mapping, torch::Tensor k_cache,
+                    torch::Tensor v_cache) {
+  check(q, torch::kBFloat16, "q");
+  check(k, torch::kBFloat16, "k");
+  check(v, torch::kBFloat16, "v");
+  check(cos_sin, torch::kFloat32, "cos_sin");
+  check(positions, torch::kInt64, "positions");
+  check(slot_mapping, torch::kInt64, "slot_mapping");
+  const long T = q.size(0);
+  const int H = (int)q.size(1), KV = (int)k.size(1), D = (int)q.size(2);
+  const int BS = (int)k_cache.size(2);
+  launch_rope_kv_append(bf(q), bf(k), cbf(v),
+                        positions.data_ptr<long>(),
+                        cos_sin.data_ptr<float>(),
+                        slot_mapping.data_ptr<long>(), bf(k_cache),
+                        bf(v_cache), T, H, KV, D, BS, stream());
+}
+
+void paged_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
+                torch::Tensor v_cache, torch::Tensor block_tables,
+                torch::Tensor ctx_lens,
+                c10::optional<torch::Tensor> row_seq, double scale) {
+  check(out, torch::kBFloat16, "out");
+  check(q, torch::kBFloat16, "q");
+  check(block_tables, torch::kInt32, "block_tables");
+  check(ctx_lens, torch::kInt32, "ctx_lens");
+  const long R = q.size(0);
+  const int H = (int)q.size(1), D = (int)q.size(2);
+  const int KV = (int)k_cache.size(1), BS = (int)k_cache.size(2);
+  TORCH_CHECK(H % KV == 0, "GQA group");
+  const int* rs = nullptr;
+  if (row_seq.has_value()) {
+    check(*row_seq, torch::kInt32, "row_seq");
+    rs = row_seq->data_ptr<int>();
+  }
+  launch_paged_attn(bf(out), cbf(q), cbf(k_cache), cbf(v_cache),
+                    block_tables.data_ptr<int>(), ctx_lens.data_ptr<int>(),
+                    rs, R, KV, H / KV, D, (int)block_tables.size(1), BS,
+                    (float)scale, stream());
+}
+
+void silu_mul(torch::Tensor out, torch::Tensor x) {
+  check(out, torch::kBFloat16, "out");
+  check(x, torch::kBFloat16, "x");
+  const int inter = (int)out.size(-1);
+  TORCH_CHECK(x.size(-1) == 2 * inter && inter % 8 == 0, "shape");
+  launch_silu_mul(bf(out), cbf(x), out.numel() / inter, inter, stream());
+}
+
+void greedy_sample(torch::Tensor out, torch::Tensor logits) {
+  check(out, torch::kInt64, "out");
+  check(logits, torch::kBFloat16, "logits");
+  launch_greedy_sample(out.data_ptr<long>(), cbf(logits), logits.size(0),
+                       (int)logits.size(1), stream());
+}
+
+void inv_cdf_sample(torch::Tensor out, torch::Tensor logits,
+                    torch::Tensor uniform) {
+  check(out, torch::kInt64, "out");
+  check(logits, torch::kFloat32, "logits");
+  check(uniform, torch::kFloat32, "uniform");
+  launch_inv_cdf_sample(out.data_ptr<long>(), logits.data_ptr<float>(),
+                        uniform.data_ptr<float>(), logits.size(0),
+                        (int)logits.size(1), stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "HyperSpot-AMD gfx950 (CDNA4) kernels";
+  m.def("rmsnorm", &rmsnorm);
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
+  m.def("rope_kv_append", &rope_kv_append);
+  m.def("paged_attn", &paged_attn);
+  m.def("silu_mul", &silu_mul);
+  m.def("greedy_sample", &greedy_sample);
+  m.def("inv_cdf_sample", &inv_cdf_sample);
+}

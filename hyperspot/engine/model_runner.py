@@ -61,7 +61,9 @@ class ModelRunner:
         slots: List[int] = []
         starts = [0]
         logit_rows = []
-        for req in batch.requests:
+        row_seq: List[int] = []
+        tables = []
+        for si, req in enumerate(batch.requests):
             n = req.num_prompt_tokens
             ids.extend(req.prompt_token_ids)
             pos.extend(range(n))
@@ -69,13 +71,23 @@ class ModelRunner:
             slots.extend(bm.slot_of(req.request_id, p) for p in range(n))
             starts.append(starts[-1] + n)
             logit_rows.append(starts[-1] - 1)
+            row_seq.extend([si] * n)
+            tables.append(bm.table(req.request_id))
+        max_t = max(len(t) for t in tables)
+        bt = torch.zeros(len(tables), max_t, dtype=torch.int32)
+        for i, t in enumerate(tables):
+            bt[i, :len(t)] = torch.tensor(t, dtype=torch.int32)
         d = self.device
+        positions = torch.tensor(pos, dtype=torch.long, device=d)
         meta = ForwardMeta(
             mode="prefill",
-            positions=torch.tensor(pos, dtype=torch.long, device=d),
+            positions=positions,
             slot_mapping=torch.tensor(slots, dtype=torch.long, device=d),
             seq_start=torch.tensor(starts, dtype=torch.int32, device=d),
             max_seqlen=max(r.num_prompt_tokens for r in batch.requests),
+            row_seq=torch.tensor(row_seq, dtype=torch.int32, device=d),
+            ctx_lens=(positions + 1).to(torch.int32),
+            block_tables=bt.to(d),
             logits_indices=torch.tensor(logit_rows, dtype=torch.long, device=d),
         )
         input_ids = torch.tensor(ids, dtype=torch.long, device=d)

@@ -38,7 +38,9 @@ class ForwardMeta:
     # prefill
     seq_start: Optional[torch.Tensor] = None   # [num_seqs+1] int32
     max_seqlen: int = 0
-    # decode
+    row_seq: Optional[torch.Tensor] = None     # [T] int32 row -> seq index
+    ctx_lens: Optional[torch.Tensor] = None    # [T] int32 = position + 1
+    # decode (block_tables also used by GPU prefill, indexed via row_seq)
     block_tables: Optional[torch.Tensor] = None  # [B, max_blocks] int32
     seq_lens: Optional[torch.Tensor] = None      # [B] int32
     # logits are computed only for these token rows (last token per seq)
@@ -65,12 +67,7 @@ class Attention(nn.Module):
         k_cache, v_cache = kv_cache[0], kv_cache[1]
         ops.rope_kv_append(q, k, v, meta.positions, cos_sin,
                            meta.slot_mapping, k_cache, v_cache)
-        if meta.mode == "prefill":
-            o = ops.prefill_attn(q, k, v, meta.seq_start, meta.max_seqlen,
-                                 self.scale)
-        else:
-            o = ops.paged_attn_decode(q, k_cache, v_cache, meta.block_tables,
-                                      meta.seq_lens, self.scale)
+        o = ops.attention(q, k, v, k_cache, v_cache, meta, self.scale)
         return self.o_proj(o.view(T, -1))
 
 

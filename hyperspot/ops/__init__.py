@@ -82,21 +82,32 @@ def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
                       seq_lens: torch.Tensor, scale: float) -> torch.Tensor:
     if q.is_cuda:
         out = torch.empty_like(q)
-        _native().paged_attn_decode(out, q, k_cache, v_cache, block_tables,
-                                    seq_lens, scale)
+        _native().paged_attn(out, q, k_cache, v_cache, block_tables,
+                             seq_lens, None, scale)
         return out
     return torch_ref.paged_attn_decode(q, k_cache, v_cache, block_tables,
                                        seq_lens, scale)
 
 
-def prefill_attn(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
-                 seq_start: torch.Tensor, max_seqlen: int,
-                 scale: float) -> torch.Tensor:
+def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+              k_cache: torch.Tensor, v_cache: torch.Tensor, meta,
+              scale: float) -> torch.Tensor:
+    """Attention for one layer, prefill or decode, reading the paged cache.
+
+    The ONE gfx950 kernel serves both modes: decode rows attend over
+    ctx=seq_len cache slots; prefill rows attend over slots 0..pos (their
+    own k/v were appended by the fused RoPE kernel just before), which
+    realises causal varlen attention without a separate kernel.
+    """
+    if meta.mode == "decode":
+        return paged_attn_decode(q, k_cache, v_cache, meta.block_tables,
+                                 meta.seq_lens, scale)
     if q.is_cuda:
         out = torch.empty_like(q)
-        _native().prefill_attn(out, q, k, v, seq_start, max_seqlen, scale)
+        _native().paged_attn(out, q, k_cache, v_cache, meta.block_tables,
+                             meta.ctx_lens, meta.row_seq, scale)
         return out
-    return torch_ref.prefill_attn(q, k, v, seq_start, scale)
+    return torch_ref.prefill_attn(q, k, v, meta.seq_start, scale)
 
 
 def silu_mul(gate_up: torch.Tensor) -> torch.Tensor:

@@ -1,0 +1,72 @@
+"""GPU engine tests: end-to-end decode on the HIP kernel path, hipGraph
+capture, and native-extension enforcement."""
+
+import pytest
+import torch
+
+import hyperspot.ops as ops
+from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+
+pytestmark = pytest.mark.gpu
+
+
+def _cfg(**kw):
+    base = dict(model="tiny-llama", max_num_seqs=8,
+                max_num_batched_tokens=512, max_model_len=256,
+                num_gpu_blocks=128, enforce_eager=True)
+    base.update(kw)
+    return EngineConfig(**base)
+
+
+def test_native_extension_loaded():
+    assert ops.have_native(), "GPU run requires the in-tree gfx950 extension"
+
+
+def test_generate_deterministic_on_gpu():
+    eng = LLMEngine(_cfg())
+    sp = SamplingParams(temperature=0.0, max_tokens=12)
+    a = eng.generate([[1, 2, 3, 4, 5]], sp)[0]
+    b = eng.generate([[1, 2, 3, 4, 5]], sp)[0]
+    assert a == b
+    assert len(a) == 12
+
+
+def test_decode_consistent_with_prefill_recompute_gpu():
+    """Greedy continuation via incremental paged decode must match feeding
+    the grown prompt through prefill again (same kernels, same dtype)."""
+    eng = LLMEngine(_cfg())
+    sp1 = SamplingParams(temperature=0.0, max_tokens=6)
+    prompt = [3, 7, 11, 13, 17]
+    out = eng.generate([prompt], sp1)[0]
+    # recompute: prompt + first k outputs, ask for 1 token
+    for k in range(0, 5):
+        nxt = eng.generate([prompt + out[:k]],
+                           SamplingParams(temperature=0.0, max_tokens=1))[0]
+        assert nxt[0] == out[k], (k, nxt, out)
+
+
+def test_hipgraph_decode_matches_eager():
+    sp = SamplingParams(temperature=0.0, max_tokens=10)
+    prompts = [[1, 2, 3, 4], [9, 8, 7], [5, 5, 5, 5, 5]]
+    eager = LLMEngine(_cfg(enforce_eager=True)).generate(prompts, sp)
+    graphed_engine = LLMEngine(_cfg(enforce_eager=False,
+                                    graph_batch_sizes=(1, 2, 4, 8)))
+    graphed = graphed_engine.generate(prompts, sp)
+    assert eager == graphed
+    assert len(graphed_engine.runner._graphs) > 0, "graphs were not used"
+
+
+def test_batched_gpu_greedy_equals_single():
+    eng = LLMEngine(_cfg())
+    sp = SamplingParams(temperature=0.0, max_tokens=5)
+    p1, p2 = [2, 4, 6, 8], [1, 3, 5, 7, 9, 11]
+    both = eng.generate([p1, p2], sp)
+    assert both[0] == eng.generate([p1], sp)[0]
+    assert both[1] == eng.generate([p2], sp)[0]
+
+
+def test_mixtral_tiny_runs_on_gpu():
+    eng = LLMEngine(_cfg(model="tiny-moe"))
+    out = eng.generate([[1, 2, 3]], SamplingParams(temperature=0.0,
+                                                   max_tokens=4))[0]
+    assert len(out) == 4

@@ -1,0 +1,186 @@
+"""GPU numerics tests: every gfx950 HIP kernel vs the plain-PyTorch fp32
+reference of the same op (hyperspot.ops.torch_ref).  bf16 I/O tolerances."""
+
+import pytest
+import torch
+
+import hyperspot.ops as ops
+from hyperspot.ops import torch_ref as R
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+def setup_module():
+    assert ops.have_native(), \
+        "hyperspot._C must be built on GPU boxes (python csrc/build.py)"
+
+
+def _randn_bf16(*shape, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    return torch.randn(*shape, generator=g).bfloat16().to(DEV)
+
+
+@pytest.mark.parametrize("hidden", [2048, 4096, 8192])
+def test_rmsnorm(hidden):
+    x = _randn_bf16(33, hidden)
+    w = _randn_bf16(hidden, seed=1)
+    out = ops.rmsnorm(x, w, 1e-5)
+    ref = R.rmsnorm(x.float().cpu(), w.float().cpu(), 1e-5)
+    assert torch.allclose(out.float().cpu(), ref, atol=3e-2, rtol=3e-2)
+
+
+def test_fused_add_rmsnorm():
+    x = _randn_bf16(17, 4096)
+    res = _randn_bf16(17, 4096, seed=2)
+    w = _randn_bf16(4096, seed=3)
+    x_ref, res_ref = x.float().cpu(), res.float().cpu()
+    out, new_res = ops.fused_add_rmsnorm(x, res, w, 1e-5)
+    ref_out, ref_res = R.fused_add_rmsnorm(x_ref, res_ref,
+                                           w.float().cpu(), 1e-5)
+    assert torch.allclose(new_res.float().cpu(), ref_res, atol=3e-2, rtol=3e-2)
+    assert torch.allclose(out.float().cpu(), ref_out, atol=4e-2, rtol=4e-2)
+
+
+def test_rope_kv_append():
+    T, H, KV, D, BS, NB = 9, 8, 2, 128, 16, 8
+    q = _randn_bf16(T, H, D)
+    k = _randn_bf16(T, KV, D, seed=1)
+    v = _randn_bf16(T, KV, D, seed=2)
+    pos = torch.tensor([0, 1, 2, 5, 9, 100, 101, 4, 7], device=DEV)
+    slots = torch.tensor([0, 1, 2, 17, 18, 33, 40, -1, 55], device=DEV)
+    cos_sin = R.rope_cos_sin(torch.arange(256), D, 500000.0).to(DEV)
+    k_cache = torch.zeros(NB, KV, BS, D, dtype=torch.bfloat16, device=DEV)
+    v_cache = torch.zeros_like(k_cache)
+    q_ref, k_ref = R.apply_rope(q.float().cpu(), k.float().cpu(),
+                                pos.cpu(), 500000.0)
+    ops.rope_kv_append(q, k, v, pos, cos_sin, slots, k_cache, v_cache)
+    assert torch.allclose(q.float().cpu(), q_ref, atol=3e-2, rtol=3e-2)
+    assert torch.allclose(k.float().cpu(), k_ref, atol=3e-2, rtol=3e-2)
+    # cache contents (skip slot -1 row)
+    kc_ref = torch.zeros(NB, KV, BS, D)
+    vc_ref = torch.zeros(NB, KV, BS, D)
+    mask = slots.cpu() >= 0
+    R.kv_cache_append(k.float().cpu()[mask], v.float().cpu()[mask],
+                      kc_ref, vc_ref, slots.cpu()[mask])
+    assert torch.allclose(k_cache.float().cpu(), kc_ref, atol=3e-2, rtol=3e-2)
+    assert torch.allclose(v_cache.float().cpu(), vc_ref, atol=3e-2, rtol=3e-2)
+    # slot -1 must not be written anywhere: caches zero outside ref slots
+    untouched = torch.ones(NB * BS, dtype=torch.bool)
+    untouched[slots.cpu()[mask]] = False
+    kc = k_cache.float().cpu().permute(0, 2, 1, 3).reshape(NB * BS, KV, D)
+    assert (kc[untouched] == 0).all()
+
+
+@pytest.mark.parametrize("group,heads_kv", [(4, 8), (8, 1), (1, 2)])
+def test_paged_attn_decode(group, heads_kv):
+    torch.manual_seed(0)
+    B, D, BS, NB = 5, 128, 16, 64
+    H = group * heads_kv
+    maxb = 8
+    q = _randn_bf16(B, H, D)
+    k_cache = _randn_bf16(NB, heads_kv, BS, D, seed=1)
+    v_cache = _randn_bf16(NB, heads_kv, BS, D, seed=2)
+    bt = torch.randperm(NB, device=DEV, dtype=torch.int32)[: B * maxb] \
+        .reshape(B, maxb).contiguous()
+    seq_lens = torch.tensor([1, 16, 17, 100, 128], dtype=torch.int32,
+                            device=DEV)
+    out = ops.paged_attn_decode(q, k_cache, v_cache, bt, seq_lens,
+                                D ** -0.5)
+    ref = R.paged_attn_decode(q.float().cpu(), k_cache.float().cpu(),
+                              v_cache.float().cpu(), bt.cpu(),
+                              seq_lens.cpu(), D ** -0.5)
+    assert torch.allclose(out.float().cpu(), ref, atol=2e-2, rtol=2e-2)
+
+
+def test_paged_attn_prefill_matches_varlen_ref():
+    """Prefill through the cache == varlen causal attention."""
+    torch.manual_seed(0)
+    H, KV, D, BS = 4, 2, 128, 16
+    lens = [33, 7, 64]
+    T = sum(lens)
+    NB = 32
+    q = _randn_bf16(T, H, D)
+    k = _randn_bf16(T, KV, D, seed=1)
+    v = _randn_bf16(T, KV, D, seed=2)
+    k_cache = torch.zeros(NB, KV, BS, D, dtype=torch.bfloat16, device=DEV)
+    v_cache = torch.zeros_like(k_cache)
+    # identity-ish paging: seq i gets consecutive blocks
+    tables, slots, row_seq, ctx = [], [], [], []
+    next_blk = 0
+    for si, n in enumerate(lens):
+        nb = (n + BS - 1) // BS
+        blocks = list(range(next_blk, next_blk + nb))
+        next_blk += nb
+        tables.append(blocks + [0] * (8 - nb))
+        for p in range(n):
+            slots.append(blocks[p // BS] * BS + p % BS)
+            row_seq.append(si)
+            ctx.append(p + 1)
+    slots_t = torch.tensor(slots, device=DEV)
+    R.kv_cache_append(k.float(), v.float(),
+                      kc := torch.zeros(NB, KV, BS, D),
+                      vc := torch.zeros(NB, KV, BS, D), slots_t.cpu())
+    k_cache.copy_(kc.bfloat16())
+    v_cache.copy_(vc.bfloat16())
+    bt = torch.tensor(tables, dtype=torch.int32, device=DEV)
+    out = torch.empty_like(q)
+    from hyperspot import _C
+    _C.paged_attn(out, q, k_cache, v_cache, bt,
+                  torch.tensor(ctx, dtype=torch.int32, device=DEV),
+                  torch.tensor(row_seq, dtype=torch.int32, device=DEV),
+                  D ** -0.5)
+    ss = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32)
+    ref = R.prefill_attn(q.float().cpu(), k.float().cpu(), v.float().cpu(),
+                         ss, D ** -0.5)
+    assert torch.allclose(out.float().cpu(), ref, atol=2e-2, rtol=2e-2)
+
+
+def test_silu_mul():
+    x = _randn_bf16(65, 2 * 1024)
+    out = ops.silu_mul(x)
+    ref = R.silu_mul(x.float().cpu())
+    assert torch.allclose(out.float().cpu(), ref, atol=2e-2, rtol=2e-2)
+
+
+def test_greedy_sample():
+    torch.manual_seed(0)
+    logits = _randn_bf16(64, 128256)
+    out = ops.greedy_sample(logits)
+    # ties are possible in bf16: check the VALUE is the max
+    mx = logits.float().max(-1).values
+    picked = logits.float().gather(1, out[:, None].to(DEV)).squeeze(1)
+    assert torch.equal(picked.cpu(), mx.cpu())
+
+
+def test_inv_cdf_sample_matches_reference():
+    torch.manual_seed(0)
+    B, V = 16, 128256
+    logits = torch.randn(B, V, device=DEV)
+    u = torch.rand(B, device=DEV)
+    from hyperspot import _C
+    out = torch.empty(B, dtype=torch.long, device=DEV)
+    _C.inv_cdf_sample(out, logits, u)
+    ref = R.sample(logits.cpu(), torch.ones(B), torch.ones(B),
+                   torch.zeros(B, dtype=torch.int32), u.cpu())
+    # fp-association differences can shift the crossing by a hair when u
+    # lands exactly on a boundary; require >= 15/16 exact matches and
+    # adjacent-index otherwise
+    same = (out.cpu() == ref).sum().item()
+    assert same >= B - 1, (out.cpu(), ref)
+
+
+def test_sample_dispatch_topk_topp_gpu():
+    torch.manual_seed(0)
+    B, V = 8, 1000
+    logits = _randn_bf16(B, V).float()
+    topk = logits.topk(5, dim=-1).indices.cpu()
+    for trial in range(5):
+        t = ops.sample(logits, torch.ones(B, device=DEV),
+                       torch.ones(B, device=DEV),
+                       torch.full((B,), 5, dtype=torch.int32, device=DEV),
+                       torch.rand(B, device=DEV))
+        for b in range(B):
+            assert t[b].cpu() in topk[b]
