@@ -1,0 +1,96 @@
+// Native HTTP/1.1 server for the api-gateway module.
+//
+// The reference host plane serves HTTP with axum/hyper on tokio
+// (modules/system/api-gateway); this rebuild is a compact native server:
+// accept loop + connection threads, keep-alive, Content-Length bodies,
+// chunked streaming responses for SSE (the llm-gateway stream contract,
+// reference modules/llm-gateway/docs/DESIGN.md:289-311).
+#pragma once
+
+#include <atomic>
+#include <functional>
+#include <map>
+#include <memory>
+#include <string>
+#include <thread>
+#include <vector>
+
+#include "../util/json.h"
+
+namespace hs {
+
+struct HttpRequest {
+  std::string method;
+  std::string target;           // raw path?query
+  std::string path;
+  std::map<std::string, std::string> query;
+  std::map<std::string, std::string> headers;   // lower-case keys
+  std::map<std::string, std::string> path_params;
+  std::string body;
+  std::string request_id;
+  std::string peer;
+  Json extensions = Json::object();   // SecurityContext etc.
+
+  std::string header(const std::string& k, const std::string& dflt = "") const {
+    auto it = headers.find(k);
+    return it == headers.end() ? dflt : it->second;
+  }
+};
+
+// Streaming-capable response writer bound to one connection.
+class ResponseWriter {
+ public:
+  explicit ResponseWriter(int fd) : fd_(fd) {}
+
+  // one-shot response
+  void respond(int status, const std::string& content_type,
+               const std::string& body,
+               const std::vector<std::pair<std::string, std::string>>&
+                   extra_headers = {});
+  // streaming (SSE): chunked transfer encoding
+  void begin_stream(int status, const std::string& content_type,
+                    const std::vector<std::pair<std::string, std::string>>&
+                        extra_headers = {});
+  bool write_chunk(const std::string& data);   // false if client went away
+  void end_stream();
+
+  bool started() const { return started_; }
+  bool keep_alive = true;
+
+ private:
+  bool send_all(const char* p, size_t n);
+  int fd_;
+  bool started_ = false;
+  bool streaming_ = false;
+};
+
+using HttpHandler = std::function<void(HttpRequest&, ResponseWriter&)>;
+
+class HttpServer {
+ public:
+  HttpServer(std::string bind_addr, HttpHandler handler,
+             size_t body_limit = 16 * 1024 * 1024);
+  ~HttpServer();
+
+  // binds + starts accept thread; returns false on bind failure
+  bool start();
+  void stop();
+  int port() const { return port_; }   // resolved port (0 -> ephemeral)
+
+ private:
+  void accept_loop();
+  void handle_conn(int fd, std::string peer);
+
+  std::string bind_addr_;
+  HttpHandler handler_;
+  size_t body_limit_;
+  int listen_fd_ = -1;
+  int port_ = 0;
+  std::atomic<bool> running_{false};
+  std::thread accept_thread_;
+  std::atomic<int> live_conns_{0};
+};
+
+std::string url_decode(const std::string& s);
+
+}  // namespace hs

@@ -38,6 +38,8 @@ class ModelRunner:
         self.dtype = torch.bfloat16 if self.device.type == "cuda" \
             else torch.float32
         torch.manual_seed(config.seed)
+        from hyperspot.parallel.layers import set_init_device
+        set_init_device(self.device)
         self.model = build_model(self.spec, dtype=self.dtype).to(self.device)
         self.model.eval()
         self.num_blocks = compute_num_gpu_blocks(self.spec, config, self.device)

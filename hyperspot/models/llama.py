@@ -123,20 +123,18 @@ class LlamaForCausalLM(nn.Module):
         super().__init__()
         self.spec = spec
         self.dtype = dtype
-        g = torch.Generator().manual_seed(1234)
-        emb = torch.empty(spec.vocab_size, spec.hidden_size,
-                          dtype=torch.float32)
-        emb.normal_(0.0, 0.02, generator=g)
+        from hyperspot.parallel.layers import _init_weight
         # Embedding + LM head replicated (vocab GEMM needs no collective;
         # 288 GB HBM3E/GPU makes the duplicated 2 GB irrelevant at 70B).
-        self.embed = nn.Parameter(emb.to(dtype), requires_grad=False)
+        self.embed = nn.Parameter(
+            _init_weight(spec.vocab_size, spec.hidden_size, dtype, 900001),
+            requires_grad=False)
         if spec.tie_embeddings:
             self.lm_head = self.embed
         else:
-            lm = torch.empty(spec.vocab_size, spec.hidden_size,
-                             dtype=torch.float32)
-            lm.normal_(0.0, 0.02, generator=g)
-            self.lm_head = nn.Parameter(lm.to(dtype), requires_grad=False)
+            self.lm_head = nn.Parameter(
+                _init_weight(spec.vocab_size, spec.hidden_size, dtype,
+                             900002), requires_grad=False)
         self.layers = nn.ModuleList([
             DecoderLayer(spec, i, dtype, mlp_cls=self.mlp_cls)
             for i in range(spec.num_layers)])

@@ -119,7 +119,7 @@ def test_paged_attn_prefill_matches_varlen_ref():
             row_seq.append(si)
             ctx.append(p + 1)
     slots_t = torch.tensor(slots, device=DEV)
-    R.kv_cache_append(k.float(), v.float(),
+    R.kv_cache_append(k.float().cpu(), v.float().cpu(),
                       kc := torch.zeros(NB, KV, BS, D),
                       vc := torch.zeros(NB, KV, BS, D), slots_t.cpu())
     k_cache.copy_(kc.bfloat16())
