@@ -1,0 +1,198 @@
+// ModKit — the module runtime of the host plane.
+//
+// Native C++ re-creation of the reference's Rust modkit (libs/modkit):
+//   - Module capability contracts      (reference src/contracts.rs:37-135)
+//   - registry + topo-sorted phases    (src/registry.rs:310, runtime/host_runtime.rs:717)
+//   - ClientHub type-erased DI         (src/client_hub.rs:123)
+//   - OperationSpec/OpenAPI registry   (src/api/operation_builder.rs:196, openapi_registry.rs:83)
+//   - RFC-9457 Problem                 (libs/modkit-errors/src/problem.rs:41)
+//   - SecurityContext / AccessScope    (libs/modkit-security/src/context.rs:23)
+#pragma once
+
+#include <functional>
+#include <map>
+#include <memory>
+#include <optional>
+#include <string>
+#include <vector>
+
+#include "../http/http.h"
+#include "../util/json.h"
+
+namespace hs {
+
+// ---------------------------------------------------------------- problems
+struct Problem {
+  int status = 500;
+  std::string title = "Internal Server Error";
+  std::string type = "about:blank";
+  std::string detail;
+  std::string code;      // stable machine-readable error code
+
+  Json to_json(const std::string& instance = "") const {
+    Json j = Json::object();
+    j["type"] = type;
+    j["title"] = title;
+    j["status"] = status;
+    if (!detail.empty()) j["detail"] = detail;
+    if (!code.empty()) j["code"] = code;
+    if (!instance.empty()) j["instance"] = instance;
+    return j;
+  }
+  static Problem not_found(const std::string& detail = "") {
+    return {404, "Not Found", "about:blank", detail, ""};
+  }
+  static Problem bad_request(const std::string& detail = "") {
+    return {400, "Bad Request", "about:blank", detail, ""};
+  }
+  static Problem unauthorized(const std::string& detail = "") {
+    return {401, "Unauthorized", "about:blank", detail, ""};
+  }
+  static Problem forbidden(const std::string& detail = "") {
+    return {403, "Forbidden", "about:blank", detail, ""};
+  }
+};
+
+void respond_problem(ResponseWriter& w, const Problem& p,
+                     const std::string& instance = "");
+
+// ---------------------------------------------------- security (PDP/PEP)
+// DEFAULT_TENANT_ID must match the reference's modkit-security constant
+// (config/static-tenants.yaml:36 comments pin this UUID).
+inline const char* kDefaultTenantId = "00000000-df51-5b42-9538-d2b56b7ee953";
+inline const char* kDefaultSubjectId = "00000000-7f5e-5e2f-a26c-6c08aa2eaa4a";
+
+struct SecurityContext {
+  std::string subject_id;
+  std::string subject_type = "user";   // user | service | anonymous
+  std::string tenant_id;
+  std::vector<std::string> scopes;
+
+  static SecurityContext anonymous() {
+    return {"", "anonymous", "", {}};
+  }
+  static SecurityContext default_ctx() {   // auth-disabled mode
+    return {kDefaultSubjectId, "user", kDefaultTenantId, {}};
+  }
+  Json to_json() const;
+  static SecurityContext from_json(const Json& j);
+};
+
+// ------------------------------------------------------------- operations
+struct RateLimitCfg {
+  double rps = 50;
+  double burst = 100;
+  int in_flight = 64;
+};
+
+struct OperationSpec {
+  std::string method;        // GET/POST/...
+  std::string path;          // /module/v1/things/{id}
+  std::string operation_id;
+  std::string summary;
+  std::vector<std::string> tags;
+  bool authenticated = false;  // one of authenticated/is_public must be set
+  bool is_public = false;
+  std::optional<RateLimitCfg> rate_limit;
+  std::vector<std::string> allowed_content_types;   // 415 enforcement
+  Json request_schema;        // JSON schema (null = none)
+  std::map<int, std::string> responses;             // status -> description
+  Json response_schema;       // schema of the 200 response
+  std::vector<std::string> license_features;
+  bool sse = false;
+};
+
+struct Route {
+  OperationSpec spec;
+  HttpHandler handler;
+  std::vector<std::string> segments;   // pre-split path
+};
+
+// Typed route collector passed to modules during the `rest` phase —
+// the OperationBuilder equivalent: register() refuses specs that don't
+// declare an auth stance (typestate in the reference; runtime check here).
+class RestRegistry {
+ public:
+  void register_op(OperationSpec spec, HttpHandler handler);
+  const std::vector<Route>& routes() const { return routes_; }
+  // returns nullptr if no match; fills params
+  const Route* match(const std::string& method, const std::string& path,
+                     std::map<std::string, std::string>& params,
+                     bool* path_exists = nullptr) const;
+  Json build_openapi(const std::string& title, const std::string& version,
+                     const std::string& description) const;
+  void add_schema(const std::string& name, Json schema) {
+    schemas_[name] = std::move(schema);
+  }
+
+ private:
+  std::vector<Route> routes_;
+  std::map<std::string, Json> schemas_;
+};
+
+// -------------------------------------------------------------- ClientHub
+// Type-erased inter-module DI: modules publish SDK client objects under an
+// interface name; consumers resolve them without direct dependencies.
+class ClientHub {
+ public:
+  template <typename T>
+  void register_client(const std::string& iface, std::shared_ptr<T> impl) {
+    clients_[iface] = std::static_pointer_cast<void>(impl);
+  }
+  template <typename T>
+  std::shared_ptr<T> get(const std::string& iface) const {
+    auto it = clients_.find(iface);
+    if (it == clients_.end()) return nullptr;
+    return std::static_pointer_cast<T>(it->second);
+  }
+  bool has(const std::string& iface) const { return clients_.count(iface); }
+
+ private:
+  std::map<std::string, std::shared_ptr<void>> clients_;
+};
+
+// ---------------------------------------------------------------- modules
+class ModuleCtx {
+ public:
+  Json config;          // modules.<name>.config
+  Json full_config;     // whole AppConfig (read-only use)
+  ClientHub* hub = nullptr;
+  std::string home_dir;
+  bool* cancel = nullptr;   // set true on shutdown
+};
+
+class Module {
+ public:
+  virtual ~Module() = default;
+  virtual std::string name() const = 0;
+  virtual std::vector<std::string> deps() const { return {}; }
+  // capabilities
+  virtual bool is_rest_host() const { return false; }
+  virtual bool is_stateful() const { return false; }
+  // phases (reference host_runtime.rs: pre_init/db/init/post_init/rest/
+  // start/stop; grpc+oop handled by the owning modules directly)
+  virtual void init(ModuleCtx& ctx) {}
+  virtual void post_init(ModuleCtx& ctx) {}
+  virtual void register_rest(ModuleCtx& ctx, RestRegistry& rest) {}
+  virtual void start(ModuleCtx& ctx) {}
+  virtual void stop(ModuleCtx& ctx) {}
+};
+
+class ModuleRegistry {
+ public:
+  void add(std::shared_ptr<Module> m) { modules_.push_back(std::move(m)); }
+  // topo-sort by deps; throws on cycle/missing dep
+  std::vector<std::shared_ptr<Module>> sorted() const;
+  const std::vector<std::shared_ptr<Module>>& all() const { return modules_; }
+
+ private:
+  std::vector<std::shared_ptr<Module>> modules_;
+};
+
+// --------------------------------------------------------------- config
+// Layered config: struct defaults -> YAML file -> APP__SECTION__KEY env ->
+// CLI overrides (reference libs/modkit/src/bootstrap/config/mod.rs:270-283).
+Json load_app_config(const std::string& yaml_path,
+                     const std::map<std::string, std::string>& cli_overrides);
+
+}  // namespace hs

@@ -1,0 +1,518 @@
+#include "system_modules.h"
+
+#include <sys/sysinfo.h>
+#include <unistd.h>
+
+#include <cstring>
+#include <fstream>
+
+#include "../util/log.h"
+
+namespace hs {
+
+static SecurityContext sec_of(HttpRequest& req) {
+  return SecurityContext::from_json(req.extensions.at("security"));
+}
+
+// ------------------------------------------------------- tenant-resolver
+
+namespace {
+
+class StaticTenantResolver : public TenantResolverClient {
+ public:
+  explicit StaticTenantResolver(const Json& cfg) {
+    // config shape of static-tr-plugin (config/static-tenants.yaml:30-47):
+    // {vendor, priority, tenants: [{id, name, status, type, parent_id?}]}
+    const Json& ts = cfg.at("tenants");
+    if (ts.is_array()) {
+      for (const auto& t : ts.arr()) {
+        Tenant tn{t.at("id").as_string(), t.at("name").as_string(),
+                  t.at("status").as_string("active"),
+                  t.at("type").as_string("tenant"),
+                  t.at("parent_id").as_string()};
+        tenants_[tn.id] = tn;
+      }
+    }
+    if (tenants_.empty()) {
+      Tenant root{kDefaultTenantId, "Default Tenant", "active", "root", ""};
+      tenants_[root.id] = root;
+    }
+  }
+  std::optional<Tenant> get_tenant(const std::string& id) override {
+    auto it = tenants_.find(id);
+    if (it == tenants_.end()) return std::nullopt;
+    return it->second;
+  }
+  std::vector<Tenant> get_tenants() override {
+    std::vector<Tenant> out;
+    for (auto& [_, t] : tenants_) out.push_back(t);
+    return out;
+  }
+  std::vector<Tenant> get_ancestors(const std::string& id) override {
+    std::vector<Tenant> out;
+    auto cur = get_tenant(id);
+    while (cur && !cur->parent_id.empty()) {
+      cur = get_tenant(cur->parent_id);
+      if (cur) out.push_back(*cur);
+    }
+    return out;
+  }
+  std::vector<Tenant> get_descendants(const std::string& id) override {
+    std::vector<Tenant> out;
+    for (auto& [_, t] : tenants_) {
+      for (auto a = get_tenant(t.id); a;) {
+        if (a->parent_id == id) { out.push_back(t); break; }
+        a = a->parent_id.empty() ? std::nullopt : get_tenant(a->parent_id);
+      }
+    }
+    return out;
+  }
+  bool is_ancestor(const std::string& a, const std::string& b) override {
+    for (auto& t : get_ancestors(b))
+      if (t.id == a) return true;
+    return false;
+  }
+
+ private:
+  std::map<std::string, Tenant> tenants_;
+};
+
+}  // namespace
+
+void TenantResolverModule::init(ModuleCtx& ctx) {
+  ctx.hub->register_client<TenantResolverClient>(
+      "tenant-resolver",
+      std::make_shared<StaticTenantResolver>(ctx.config));
+}
+
+// -------------------------------------------------------- authn-resolver
+
+namespace {
+
+class StaticAuthnResolver : public AuthnResolverClient {
+ public:
+  explicit StaticAuthnResolver(const Json& cfg) {
+    // static-authn-plugin shape (plugins/static-authn-plugin/src/config.rs):
+    // {tokens: [{token, subject_id, subject_tenant_id, subject_type?,
+    //            scopes?: []}]}
+    const Json& toks = cfg.at("tokens");
+    if (toks.is_array()) {
+      for (const auto& t : toks.arr()) {
+        SecurityContext c;
+        c.subject_id = t.at("subject_id").as_string();
+        c.tenant_id = t.at("subject_tenant_id").as_string(kDefaultTenantId);
+        c.subject_type = t.at("subject_type").as_string("user");
+        if (t.at("scopes").is_array())
+          for (auto& s : t.at("scopes").arr())
+            c.scopes.push_back(s.as_string());
+        map_[t.at("token").as_string()] = c;
+      }
+    }
+  }
+  std::optional<SecurityContext> authenticate(
+      const std::string& bearer) override {
+    auto it = map_.find(bearer);
+    if (it == map_.end()) return std::nullopt;
+    return it->second;
+  }
+
+ private:
+  std::map<std::string, SecurityContext> map_;
+};
+
+}  // namespace
+
+void AuthnResolverModule::init(ModuleCtx& ctx) {
+  ctx.hub->register_client<AuthnResolverClient>(
+      "authn-resolver", std::make_shared<StaticAuthnResolver>(ctx.config));
+}
+
+// -------------------------------------------------------- authz-resolver
+
+namespace {
+
+// Static PDP: allow actions inside the subject's own tenant (or its
+// descendants), deny cross-tenant.  Returned tenant_scope is the AccessScope
+// the PEP (handlers) must filter rows by; empty scope = deny-all.
+class StaticAuthzResolver : public AuthzResolverClient {
+ public:
+  explicit StaticAuthzResolver(ClientHub* hub) : hub_(hub) {}
+  EvaluationResponse evaluate(const EvaluationRequest& r) override {
+    EvaluationResponse resp;
+    if (r.subject.subject_type == "anonymous") {
+      resp.deny_reason = "anonymous subject";
+      return resp;
+    }
+    const std::string& own = r.subject.tenant_id;
+    if (r.tenant_id.empty() || r.tenant_id == own) {
+      resp.allow = true;
+      resp.tenant_scope = {own};
+      return resp;
+    }
+    auto tr = hub_->get<TenantResolverClient>("tenant-resolver");
+    if (tr && tr->is_ancestor(own, r.tenant_id)) {
+      resp.allow = true;
+      resp.tenant_scope = {r.tenant_id};
+      return resp;
+    }
+    resp.deny_reason = "cross-tenant access denied";
+    return resp;
+  }
+
+ private:
+  ClientHub* hub_;
+};
+
+}  // namespace
+
+void AuthzResolverModule::init(ModuleCtx& ctx) {
+  ctx.hub->register_client<AuthzResolverClient>(
+      "authz-resolver", std::make_shared<StaticAuthzResolver>(ctx.hub));
+}
+
+// -------------------------------------------------------- types-registry
+
+void TypesRegistryModule::init(ModuleCtx& ctx) {}
+
+void TypesRegistryModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
+  // REST surface per reference: /types-registry/v1/entities{,/{gts_id}}
+  OperationSpec list;
+  list.method = "GET";
+  list.path = "/types-registry/v1/entities";
+  list.operation_id = "types_registry_list";
+  list.summary = "List GTS entities (wildcard filter via ?filter=)";
+  list.authenticated = true;
+  list.tags = {"types-registry"};
+  rest.register_op(list, [this](HttpRequest& rq, ResponseWriter& w) {
+    std::string filter = rq.query.count("filter") ? rq.query["filter"] : "";
+    Json items = Json::array();
+    std::lock_guard<std::mutex> lk(mu_);
+    for (auto& [id, e] : entities_) {
+      if (!filter.empty()) {
+        // '*' suffix wildcard
+        std::string pre = filter;
+        bool wild = !pre.empty() && pre.back() == '*';
+        if (wild) pre.pop_back();
+        if (wild ? id.rfind(pre, 0) != 0 : id != filter) continue;
+      }
+      items.push_back(e);
+    }
+    Json out = Json::object();
+    out["items"] = items;
+    w.respond(200, "application/json", out.dump());
+  });
+
+  OperationSpec reg;
+  reg.method = "POST";
+  reg.path = "/types-registry/v1/entities";
+  reg.operation_id = "types_registry_register";
+  reg.summary = "Batch-register GTS types/instances";
+  reg.authenticated = true;
+  reg.allowed_content_types = {"application/json"};
+  reg.tags = {"types-registry"};
+  rest.register_op(reg, [this](HttpRequest& rq, ResponseWriter& w) {
+    Json body;
+    try { body = Json::parse(rq.body); }
+    catch (...) { throw Problem::bad_request("invalid JSON body"); }
+    const Json& ents = body.at("entities");
+    if (!ents.is_array())
+      throw Problem::bad_request("'entities' array required");
+    int n = 0;
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      for (const auto& e : ents.arr()) {
+        std::string id = e.at("gts_id").as_string();
+        if (id.empty()) throw Problem::bad_request("entity without gts_id");
+        entities_[id] = e;
+        ++n;
+      }
+    }
+    Json out = Json::object();
+    out["registered"] = n;
+    w.respond(201, "application/json", out.dump());
+  });
+
+  OperationSpec get;
+  get.method = "GET";
+  get.path = "/types-registry/v1/entities/{gts_id}";
+  get.operation_id = "types_registry_get";
+  get.authenticated = true;
+  get.tags = {"types-registry"};
+  rest.register_op(get, [this](HttpRequest& rq, ResponseWriter& w) {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto it = entities_.find(rq.path_params["gts_id"]);
+    if (it == entities_.end()) throw Problem::not_found("no such entity");
+    w.respond(200, "application/json", it->second.dump());
+  });
+}
+
+// -------------------------------------------------------- nodes-registry
+
+void NodesRegistryModule::init(ModuleCtx& ctx) {}
+
+Json NodesRegistryModule::node_info() const {
+  Json n = Json::object();
+  char host[256] = {0};
+  gethostname(host, sizeof host - 1);
+  n["hostname"] = host;
+  n["os"] = "linux";
+  n["num_cpus"] = (long)sysconf(_SC_NPROCESSORS_ONLN);
+  struct sysinfo si{};
+  if (sysinfo(&si) == 0) {
+    n["mem_total_bytes"] = (double)si.totalram * si.mem_unit;
+    n["mem_free_bytes"] = (double)si.freeram * si.mem_unit;
+  }
+  // GPU inventory: AMD KFD topology (the reference uses NVML enumeration,
+  // libs/modkit-node-info/src/gpu_collector_linux.rs:22; MI355X-native is
+  // the KFD sysfs tree)
+  Json gpus = Json::array();
+  for (int i = 0; i < 64; ++i) {
+    std::string base = "/sys/class/kfd/kfd/topology/nodes/" +
+                       std::to_string(i);
+    std::ifstream f(base + "/properties");
+    if (!f) break;
+    std::string key;
+    long simd = 0, uid = 0;
+    Json g = Json::object();
+    std::string line;
+    while (std::getline(f, line)) {
+      auto sp = line.find(' ');
+      if (sp == std::string::npos) continue;
+      std::string k = line.substr(0, sp);
+      long v = atol(line.c_str() + sp + 1);
+      if (k == "simd_count") simd = v;
+      if (k == "unique_id") uid = v;
+      if (k == "gfx_target_version") g["gfx_target_version"] = v;
+    }
+    if (simd > 0) {   // CPU nodes have simd_count 0
+      g["index"] = (long)gpus.size();
+      g["simd_count"] = simd;
+      g["unique_id"] = uid;
+      gpus.push_back(g);
+    }
+  }
+  n["gpus"] = gpus;
+  return n;
+}
+
+void NodesRegistryModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
+  OperationSpec nodes;
+  nodes.method = "GET";
+  nodes.path = "/nodes-registry/v1/nodes";
+  nodes.operation_id = "nodes_list";
+  nodes.summary = "Node inventory";
+  nodes.authenticated = true;
+  nodes.tags = {"nodes-registry"};
+  rest.register_op(nodes, [this](HttpRequest& rq, ResponseWriter& w) {
+    Json out = Json::object();
+    Json items = Json::array();
+    Json self = node_info();
+    self["id"] = "local";
+    items.push_back(self);
+    out["items"] = items;
+    w.respond(200, "application/json", out.dump());
+  });
+  OperationSpec sysinfo_op;
+  sysinfo_op.method = "GET";
+  sysinfo_op.path = "/nodes-registry/v1/nodes/{id}/sysinfo";
+  sysinfo_op.operation_id = "nodes_sysinfo";
+  sysinfo_op.authenticated = true;
+  sysinfo_op.tags = {"nodes-registry"};
+  rest.register_op(sysinfo_op, [this](HttpRequest& rq, ResponseWriter& w) {
+    if (rq.path_params["id"] != "local")
+      throw Problem::not_found("unknown node");
+    w.respond(200, "application/json", node_info().dump());
+  });
+}
+
+// -------------------------------------------------------- model-registry
+
+namespace {
+
+class LocalModelRegistry : public ModelRegistryClient {
+ public:
+  LocalModelRegistry(const Json& cfg, const std::string& serving_model) {
+    // models from config: modules.model-registry.config.models:
+    // [{canonical_id, name, architecture, context_window, managed...}]
+    const Json& ms = cfg.at("models");
+    if (ms.is_array())
+      for (auto& m : ms.arr())
+        models_[m.at("canonical_id").as_string()] = m;
+    if (models_.empty()) {
+      std::vector<std::string> ids = {"local::llama3-8b",
+                                      "local::llama3-70b",
+                                      "local::mixtral-8x7b"};
+      // whatever the llm-gateway actually serves is always registered
+      if (!serving_model.empty())
+        ids.push_back("local::" + serving_model);
+      for (const std::string& cid : ids) {
+        if (models_.count(cid)) continue;
+        Json m = Json::object();
+        m["canonical_id"] = cid;
+        m["provider_slug"] = "local";
+        m["provider_model_id"] = cid.substr(cid.find("::") + 2);
+        m["managed"] = true;
+        m["architecture"] =
+            cid.find("mixtral") != std::string::npos ? "mistral" : "llama";
+        m["format"] = "safetensors";
+        m["lifecycle_status"] = "production";
+        Json caps = Json::object();
+        caps["text_input"] = true;
+        caps["text_output"] = true;
+        caps["streaming"] = true;
+        m["capabilities"] = caps;
+        m["context_window"] = 8192;
+        models_[cid] = m;
+      }
+    }
+  }
+  std::optional<Json> get_tenant_model(const std::string& tenant,
+                                       const std::string& canonical)
+      override {
+    // canonical id = {provider_slug}::{provider_model_id}, split on FIRST
+    // "::" (reference PRD.md:11); bare model ids default to "local::"
+    std::string cid = canonical.find("::") == std::string::npos
+        ? "local::" + canonical : canonical;
+    auto it = models_.find(cid);
+    if (it == models_.end()) return std::nullopt;
+    return it->second;
+  }
+  std::vector<Json> list_tenant_models(const std::string&) override {
+    std::vector<Json> out;
+    for (auto& [_, m] : models_) out.push_back(m);
+    return out;
+  }
+
+ private:
+  std::map<std::string, Json> models_;
+};
+
+std::shared_ptr<LocalModelRegistry> g_registry;
+
+}  // namespace
+
+void ModelRegistryModule::init(ModuleCtx& ctx) {
+  g_registry = std::make_shared<LocalModelRegistry>(
+      ctx.config,
+      ctx.full_config.path("modules.llm-gateway.config.model")
+          .as_string(""));
+  ctx.hub->register_client<ModelRegistryClient>("model-registry",
+                                                g_registry);
+}
+
+void ModelRegistryModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
+  OperationSpec list;
+  list.method = "GET";
+  list.path = "/model-registry/v1/models";
+  list.operation_id = "models_list";
+  list.summary = "List tenant-visible models";
+  list.authenticated = true;
+  list.tags = {"model-registry"};
+  rest.register_op(list, [](HttpRequest& rq, ResponseWriter& w) {
+    Json items = Json::array();
+    for (auto& m : g_registry->list_tenant_models(
+             sec_of(rq).tenant_id))
+      items.push_back(m);
+    Json out = Json::object();
+    out["items"] = items;
+    w.respond(200, "application/json", out.dump());
+  });
+  OperationSpec get;
+  get.method = "GET";
+  get.path = "/model-registry/v1/models/{canonical_id}";
+  get.operation_id = "models_get";
+  get.authenticated = true;
+  get.tags = {"model-registry"};
+  rest.register_op(get, [](HttpRequest& rq, ResponseWriter& w) {
+    auto m = g_registry->get_tenant_model(sec_of(rq).tenant_id,
+                                          rq.path_params["canonical_id"]);
+    if (!m) throw Problem::not_found("model_not_found");
+    w.respond(200, "application/json", m->dump());
+  });
+}
+
+// ------------------------------------------------------------- credstore
+
+namespace {
+
+class InMemoryCredStore : public CredStoreClient {
+ public:
+  std::optional<std::string> get(const std::string& tenant,
+                                 const std::string& ref) override {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto it = store_.find(tenant + "\x1f" + ref);
+    if (it == store_.end()) return std::nullopt;
+    return it->second;
+  }
+  void put(const std::string& tenant, const std::string& ref,
+           const std::string& value) override {
+    std::lock_guard<std::mutex> lk(mu_);
+    store_[tenant + "\x1f" + ref] = value;
+  }
+  bool del(const std::string& tenant, const std::string& ref) override {
+    std::lock_guard<std::mutex> lk(mu_);
+    return store_.erase(tenant + "\x1f" + ref) > 0;
+  }
+
+ private:
+  std::mutex mu_;
+  std::map<std::string, std::string> store_;
+};
+
+std::shared_ptr<InMemoryCredStore> g_creds;
+
+}  // namespace
+
+void CredStoreModule::init(ModuleCtx& ctx) {
+  g_creds = std::make_shared<InMemoryCredStore>();
+  ctx.hub->register_client<CredStoreClient>("credstore", g_creds);
+}
+
+void CredStoreModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
+  // REST per reference DESIGN.md:283-289; tenant always from SecurityCtx;
+  // inaccessible secrets are ALWAYS 404 (anti-enumeration, DESIGN.md:336).
+  OperationSpec put;
+  put.method = "PUT";
+  put.path = "/credstore/v1/secrets/{ref}";
+  put.operation_id = "credstore_put";
+  put.authenticated = true;
+  put.allowed_content_types = {"application/json"};
+  put.tags = {"credstore"};
+  rest.register_op(put, [](HttpRequest& rq, ResponseWriter& w) {
+    Json body;
+    try { body = Json::parse(rq.body); }
+    catch (...) { throw Problem::bad_request("invalid JSON"); }
+    if (!body.at("value").is_string())
+      throw Problem::bad_request("'value' string required");
+    g_creds->put(sec_of(rq).tenant_id, rq.path_params["ref"],
+                 body.at("value").as_string());
+    w.respond(204, "application/json", "");
+  });
+  OperationSpec get;
+  get.method = "GET";
+  get.path = "/credstore/v1/secrets/{ref}";
+  get.operation_id = "credstore_get";
+  get.authenticated = true;
+  get.tags = {"credstore"};
+  rest.register_op(get, [](HttpRequest& rq, ResponseWriter& w) {
+    auto v = g_creds->get(sec_of(rq).tenant_id, rq.path_params["ref"]);
+    if (!v) throw Problem::not_found();   // never reveal existence
+    Json out = Json::object();
+    out["ref"] = rq.path_params["ref"];
+    out["value"] = *v;
+    w.respond(200, "application/json", out.dump());
+  });
+  OperationSpec del;
+  del.method = "DELETE";
+  del.path = "/credstore/v1/secrets/{ref}";
+  del.operation_id = "credstore_delete";
+  del.authenticated = true;
+  del.tags = {"credstore"};
+  rest.register_op(del, [](HttpRequest& rq, ResponseWriter& w) {
+    if (!g_creds->del(sec_of(rq).tenant_id, rq.path_params["ref"]))
+      throw Problem::not_found();
+    w.respond(204, "application/json", "");
+  });
+}
+
+}  // namespace hs

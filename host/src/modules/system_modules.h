@@ -1,0 +1,129 @@
+// System modules (control plane): authn/authz/tenant resolvers with static
+// plugins, types-registry (GTS store), nodes-registry, model-registry,
+// credstore, file-storage.
+//
+// Reference crates: modules/system/{authn-resolver,authz-resolver,
+// tenant-resolver,types-registry,nodes-registry}, spec-only
+// modules/{model-registry,credstore,file-storage}.
+#pragma once
+
+#include <mutex>
+
+#include "../modkit/modkit.h"
+#include "api_gateway.h"
+
+namespace hs {
+
+// ---- tenant-resolver (static-tr-plugin: YAML tenant tree,
+//      config/static-tenants.yaml shape) ----
+struct Tenant {
+  std::string id, name, status, type, parent_id;
+};
+
+struct TenantResolverClient {
+  virtual ~TenantResolverClient() = default;
+  virtual std::optional<Tenant> get_tenant(const std::string& id) = 0;
+  virtual std::vector<Tenant> get_tenants() = 0;
+  virtual std::vector<Tenant> get_ancestors(const std::string& id) = 0;
+  virtual std::vector<Tenant> get_descendants(const std::string& id) = 0;
+  virtual bool is_ancestor(const std::string& a, const std::string& b) = 0;
+};
+
+class TenantResolverModule : public Module {
+ public:
+  std::string name() const override { return "tenant-resolver"; }
+  void init(ModuleCtx& ctx) override;
+};
+
+// ---- authn-resolver (static-authn-plugin: token -> identity map) ----
+class AuthnResolverModule : public Module {
+ public:
+  std::string name() const override { return "authn-resolver"; }
+  void init(ModuleCtx& ctx) override;
+};
+
+// ---- authz-resolver (PDP; static plugin = allow within own tenant) ----
+struct EvaluationRequest {
+  SecurityContext subject;
+  std::string action;       // read | write | admin
+  std::string resource;     // e.g. "llm:chat"
+  std::string tenant_id;
+};
+struct EvaluationResponse {
+  bool allow = false;
+  std::string deny_reason;
+  std::vector<std::string> tenant_scope;   // row-level AccessScope
+};
+struct AuthzResolverClient {
+  virtual ~AuthzResolverClient() = default;
+  virtual EvaluationResponse evaluate(const EvaluationRequest& r) = 0;
+};
+
+class AuthzResolverModule : public Module {
+ public:
+  std::string name() const override { return "authz-resolver"; }
+  std::vector<std::string> deps() const override {
+    return {"tenant-resolver"};
+  }
+  void init(ModuleCtx& ctx) override;
+};
+
+// ---- types-registry (GTS entity store; plugin discovery backbone) ----
+class TypesRegistryModule : public Module {
+ public:
+  std::string name() const override { return "types-registry"; }
+  void init(ModuleCtx& ctx) override;
+  void register_rest(ModuleCtx& ctx, RestRegistry& rest) override;
+
+ private:
+  std::mutex mu_;
+  std::map<std::string, Json> entities_;   // gts_id -> entity
+};
+
+// ---- nodes-registry (node inventory incl. GPU info) ----
+class NodesRegistryModule : public Module {
+ public:
+  std::string name() const override { return "nodes-registry"; }
+  void init(ModuleCtx& ctx) override;
+  void register_rest(ModuleCtx& ctx, RestRegistry& rest) override;
+
+ private:
+  Json node_info() const;
+};
+
+// ---- model-registry (canonical {provider}::{model} resolution,
+//      tenant approvals — reference modules/model-registry/docs/PRD.md) ----
+struct ModelRegistryClient {
+  virtual ~ModelRegistryClient() = default;
+  // canonical id "local::llama3-8b"; empty optional = not found/approved
+  virtual std::optional<Json> get_tenant_model(const std::string& tenant,
+                                               const std::string& canonical)
+      = 0;
+  virtual std::vector<Json> list_tenant_models(const std::string& tenant) = 0;
+};
+
+class ModelRegistryModule : public Module {
+ public:
+  std::string name() const override { return "model-registry"; }
+  void init(ModuleCtx& ctx) override;
+  void register_rest(ModuleCtx& ctx, RestRegistry& rest) override;
+};
+
+// ---- credstore (tenant-scoped secrets; always-404 on inaccessible) ----
+struct CredStoreClient {
+  virtual ~CredStoreClient() = default;
+  virtual std::optional<std::string> get(const std::string& tenant,
+                                         const std::string& ref) = 0;
+  virtual void put(const std::string& tenant, const std::string& ref,
+                   const std::string& value) = 0;
+  virtual bool del(const std::string& tenant, const std::string& ref) = 0;
+};
+
+class CredStoreModule : public Module {
+ public:
+  std::string name() const override { return "credstore"; }
+  void init(ModuleCtx& ctx) override;
+  void register_rest(ModuleCtx& ctx, RestRegistry& rest) override;
+};
+
+}  // namespace hs

@@ -1,0 +1,81 @@
+"""Byte-level tokenizer + chat templating.
+
+There is no network in this environment, so no real checkpoint/tokenizer
+files exist; serving uses a deterministic byte-level scheme mapped into the
+model's vocab (a production deployment would load the checkpoint's
+tokenizer via `transformers`, which is installed).  Round-trips exactly for
+any UTF-8 text.
+"""
+
+from __future__ import annotations
+
+from typing import List
+
+BOS = 1
+EOS = 2
+OFFSET = 4          # byte b -> token b + OFFSET
+
+
+class ByteTokenizer:
+    def __init__(self, vocab_size: int):
+        assert vocab_size >= 256 + OFFSET
+        self.vocab_size = vocab_size
+        self.eos_token_id = EOS
+
+    def encode(self, text: str, add_bos: bool = False) -> List[int]:
+        ids = [b + OFFSET for b in text.encode("utf-8")]
+        return ([BOS] if add_bos else []) + ids
+
+    def decode(self, ids: List[int]) -> str:
+        data = bytes(i - OFFSET for i in ids if OFFSET <= i < 256 + OFFSET)
+        return data.decode("utf-8", errors="replace")
+
+
+class StreamDetokenizer:
+    """Incremental detokenizer that only emits complete UTF-8 sequences."""
+
+    def __init__(self, tok: ByteTokenizer):
+        self.tok = tok
+        self.buf = b""
+
+    def push(self, token_id: int) -> str:
+        if not (OFFSET <= token_id < 256 + OFFSET):
+            return ""
+        self.buf += bytes([token_id - OFFSET])
+        try:
+            out = self.buf.decode("utf-8")
+            self.buf = b""
+            return out
+        except UnicodeDecodeError as e:
+            if e.start > 0:
+                out = self.buf[:e.start].decode("utf-8", errors="replace")
+                self.buf = self.buf[e.start:]
+                return out
+            if len(self.buf) >= 4:     # invalid sequence, flush
+                out = self.buf.decode("utf-8", errors="replace")
+                self.buf = b""
+                return out
+            return ""
+
+    def flush(self) -> str:
+        out = self.buf.decode("utf-8", errors="replace")
+        self.buf = b""
+        return out
+
+
+def render_chat(messages: list) -> str:
+    """Minimal deterministic chat template over the GTS message shape
+    (role + content[] parts — reference message.v1 schema: content is
+    ALWAYS an array of parts; a bare string is tolerated for robustness)."""
+    out = []
+    for m in messages:
+        role = m.get("role", "user")
+        content = m.get("content", [])
+        if isinstance(content, str):
+            text = content
+        else:
+            text = "".join(p.get("text", "") for p in content
+                           if isinstance(p, dict) and p.get("type", "text") == "text")
+        out.append(f"<|{role}|>\n{text}\n")
+    out.append("<|assistant|>\n")
+    return "".join(out)

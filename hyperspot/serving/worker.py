@@ -1,0 +1,224 @@
+"""Per-GPU engine worker process.
+
+Spawned by the C++ host plane's llm-gateway module (the reference's OoP
+module pattern — SURVEY.md §3.5: config by argv/env, liveness by "info"
+round-trips).  Speaks newline-delimited JSON over a Unix socket:
+
+  -> {"type":"chat","id":r,"model":m,"messages":[...],"params":{...}}
+  <- {"event":"delta","id":r,"text":t,"token_id":n}
+  <- {"event":"done","id":r,"finish_reason":f,"usage":{...}}
+  -> {"type":"abort","id":r}
+  -> {"type":"info"}           <- {"event":"info","ready":true,...}
+
+One stepping thread drives the continuous-batching engine; connection
+threads only enqueue requests and drain per-request token queues, so the
+GPU loop never blocks on a slow client.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import logging
+import os
+import queue
+import socket
+import threading
+import time
+
+log = logging.getLogger("hyperspot.worker")
+
+
+class WorkerState:
+    def __init__(self, engine, tokenizer):
+        self.engine = engine
+        self.tokenizer = tokenizer
+        self.lock = threading.Lock()          # guards engine scheduler
+        self.new_work = threading.Condition(self.lock)
+        self.streams = {}                     # rid -> queue.Queue
+        self.started = {}                     # rid -> prompt_len
+        self.stop = False
+
+    def submit(self, rid, prompt_ids, sampling):
+        q = queue.Queue()
+        with self.new_work:
+            self.streams[rid] = q
+            self.started[rid] = len(prompt_ids)
+            self.engine.add_request(prompt_ids, sampling, request_id=rid)
+            self.new_work.notify()
+        return q
+
+    def abort(self, rid):
+        with self.new_work:
+            self.engine.abort_request(rid)
+            self.streams.pop(rid, None)
+
+    def step_loop(self):
+        while not self.stop:
+            with self.new_work:
+                while not self.engine.has_work() and not self.stop:
+                    self.new_work.wait(timeout=0.5)
+                if self.stop:
+                    return
+                outputs = self.engine.step()
+            for out in outputs:
+                q = self.streams.get(out.request_id)
+                if q is not None:
+                    q.put(out)
+                    if out.finished:
+                        self.streams.pop(out.request_id, None)
+
+
+def handle_conn(conn: socket.socket, state: WorkerState, model_name: str):
+    f = conn.makefile("rwb")
+
+    def send(obj):
+        try:
+            f.write((json.dumps(obj) + "\n").encode())
+            f.flush()
+            return True
+        except (BrokenPipeError, OSError):
+            return False
+
+    try:
+        for raw in f:
+            try:
+                msg = json.loads(raw)
+            except json.JSONDecodeError:
+                send({"event": "error", "message": "bad json"})
+                continue
+            t = msg.get("type")
+            if t == "info":
+                send({"event": "info", "ready": True, "model": model_name,
+                      "num_running": state.engine.num_running,
+                      "num_waiting": state.engine.num_waiting,
+                      "kv_blocks_free": state.engine.runner.block_manager.num_free,
+                      "kv_blocks_total": state.engine.runner.num_blocks})
+            elif t == "abort":
+                state.abort(msg.get("id", ""))
+            elif t == "chat":
+                _run_chat(msg, state, send)
+            else:
+                send({"event": "error", "message": f"unknown type {t!r}"})
+    except (ConnectionResetError, BrokenPipeError, OSError):
+        pass
+    finally:
+        try:
+            conn.close()
+        except OSError:
+            pass
+
+
+def _run_chat(msg, state: WorkerState, send):
+    from hyperspot.engine import SamplingParams
+    from .tokenizer import StreamDetokenizer, render_chat
+
+    rid = msg.get("id") or f"r{time.monotonic_ns()}"
+    params = msg.get("params") or {}
+    prompt_ids = msg.get("prompt_ids")
+    if prompt_ids is None:
+        text = render_chat(msg.get("messages") or [])
+        prompt_ids = state.tokenizer.encode(text, add_bos=True)
+    max_model_len = state.engine.config.max_model_len
+    budget = max_model_len - len(prompt_ids) - 1
+    if budget <= 0:
+        send({"event": "error", "id": rid,
+              "message": "prompt exceeds context window"})
+        return
+    sampling = SamplingParams(
+        temperature=float(params.get("temperature", 0.7)),
+        top_p=float(params.get("top_p", 1.0)),
+        top_k=int(params.get("top_k", 0)),
+        max_tokens=min(int(params.get("max_tokens", 256)), budget),
+        ignore_eos=bool(params.get("ignore_eos", False)),
+    )
+    q = state.submit(rid, prompt_ids, sampling)
+    detok = StreamDetokenizer(state.tokenizer)
+    n_out = 0
+    while True:
+        try:
+            out = q.get(timeout=600)
+        except queue.Empty:
+            state.abort(rid)
+            send({"event": "error", "id": rid, "message": "engine stall"})
+            return
+        n_out += 1
+        text = detok.push(out.token_id)
+        if text and not out.finished:
+            if not send({"event": "delta", "id": rid, "text": text,
+                         "token_id": out.token_id}):
+                state.abort(rid)
+                return
+        if out.finished:
+            text += detok.flush()
+            if text:
+                send({"event": "delta", "id": rid, "text": text,
+                      "token_id": out.token_id})
+            send({"event": "done", "id": rid,
+                  "finish_reason": out.finish_reason.value
+                  if out.finish_reason else "stop",
+                  "usage": {"input_tokens": len(prompt_ids),
+                            "output_tokens": n_out}})
+            return
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--uds", required=True)
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--max-num-seqs", type=int, default=256)
+    ap.add_argument("--max-model-len", type=int, default=8192)
+    ap.add_argument("--num-gpu-blocks", type=int, default=0)
+    ap.add_argument("--eager", action="store_true")
+    ap.add_argument("--device", default=None)
+    ap.add_argument("--tp", type=int, default=1)
+    args = ap.parse_args()
+
+    logging.basicConfig(level=logging.INFO,
+                        format="%(asctime)s %(levelname)s %(name)s %(message)s")
+    import torch
+    from hyperspot.engine import EngineConfig, LLMEngine
+    from hyperspot.serving.tokenizer import ByteTokenizer
+
+    on_gpu = torch.cuda.is_available() and args.device != "cpu"
+    cfg = EngineConfig(
+        model=args.model, max_num_seqs=args.max_num_seqs,
+        max_model_len=args.max_model_len,
+        num_gpu_blocks=args.num_gpu_blocks or None,
+        enforce_eager=args.eager or not on_gpu, tp_size=args.tp)
+    if args.tp > 1:
+        from hyperspot.parallel.state import initialize_model_parallel
+        initialize_model_parallel(tp_size=args.tp)
+    log.info("loading engine model=%s device=%s", args.model, args.device)
+    eng = LLMEngine(cfg, device=args.device,
+                    eos_token_id=ByteTokenizer(cfg.spec().vocab_size).eos_token_id)
+    if not cfg.enforce_eager:
+        eng.capture_graphs()
+    tok = ByteTokenizer(cfg.spec().vocab_size)
+    state = WorkerState(eng, tok)
+    stepper = threading.Thread(target=state.step_loop, daemon=True)
+    stepper.start()
+
+    try:
+        os.unlink(args.uds)
+    except FileNotFoundError:
+        pass
+    srv = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    srv.bind(args.uds)
+    srv.listen(64)
+    log.info("worker ready on %s", args.uds)
+    try:
+        while True:
+            conn, _ = srv.accept()
+            threading.Thread(target=handle_conn,
+                             args=(conn, state, args.model),
+                             daemon=True).start()
+    except KeyboardInterrupt:
+        pass
+    finally:
+        state.stop = True
+        srv.close()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,394 @@
+"""End-to-end tests of the C++ host plane (hyperspot-server) with the
+Python engine worker on CPU — the reference's testing/e2e pytest suite
+re-created for the rebuild (mock-free: the real binary, the real worker).
+
+Covers BASELINE config 1 (/health /healthz /docs on config/no-db.yaml
+posture) plus the llm-gateway chat contract (sync + SSE + [DONE]) and the
+multi-tenant auth path.
+"""
+
+import json
+import os
+import socket
+import subprocess
+import time
+import urllib.error
+import urllib.request
+from pathlib import Path
+
+import pytest
+
+ROOT = Path(__file__).resolve().parent.parent
+BIN = ROOT / "host" / "build" / "hyperspot-server"
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def _build_binary():
+    if not BIN.exists():
+        r = subprocess.run(["make", "-j8"], cwd=ROOT / "host",
+                           capture_output=True, text=True)
+        assert r.returncode == 0, r.stdout + r.stderr
+    return BIN
+
+
+def _http(method, url, body=None, token=None, timeout=30):
+    req = urllib.request.Request(url, method=method)
+    if token:
+        req.add_header("authorization", f"Bearer {token}")
+    data = None
+    if body is not None:
+        data = json.dumps(body).encode()
+        req.add_header("content-type", "application/json")
+    try:
+        with urllib.request.urlopen(req, data=data, timeout=timeout) as r:
+            return r.status, r.read().decode()
+    except urllib.error.HTTPError as e:
+        return e.code, e.read().decode()
+
+
+class ServerProc:
+    def __init__(self, config_path, port, extra_env=None):
+        _build_binary()
+        env = dict(os.environ, **(extra_env or {}))
+        env["APP__MODULES__API_DASH_GATEWAY"] = ""  # unused; kept simple
+        self.port = port
+        self.proc = subprocess.Popen(
+            [str(BIN), "run", "--config", str(config_path),
+             "--set", "modules.api-gateway.config.bind_addr",
+             f'"127.0.0.1:{port}"'],
+            cwd=ROOT, env=env, stdout=subprocess.PIPE,
+            stderr=subprocess.PIPE)
+
+    def wait_ready(self, timeout=60):
+        t0 = time.time()
+        while time.time() - t0 < timeout:
+            if self.proc.poll() is not None:
+                out = self.proc.stderr.read().decode()
+                raise RuntimeError(f"server died: {out[-2000:]}")
+            try:
+                st, _ = _http("GET", f"http://127.0.0.1:{self.port}/healthz",
+                              timeout=2)
+                if st == 200:
+                    return
+            except Exception:
+                pass
+            time.sleep(0.2)
+        raise TimeoutError("server did not come up")
+
+    def wait_worker(self, timeout=120, token=None):
+        t0 = time.time()
+        url = f"http://127.0.0.1:{self.port}/llm-gateway/v1/status"
+        while time.time() - t0 < timeout:
+            st, body = _http("GET", url, token=token)
+            if st == 200 and json.loads(body).get("worker_ready"):
+                return
+            time.sleep(0.5)
+        raise TimeoutError("worker not ready")
+
+    def stop(self):
+        self.proc.terminate()
+        try:
+            self.proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            self.proc.kill()
+            self.proc.wait()
+
+
+@pytest.fixture(scope="module")
+def server():
+    # dedicated socket path to avoid clashes with other runs
+    import tempfile
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-e2e-")
+    port = _free_port()
+    cfg = f"""
+server:
+  home_dir: "/tmp/hs-e2e"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      enable_docs: true
+      cors_enabled: true
+      auth_disabled: true
+  llm-gateway:
+    config:
+      model: "tiny-llama"
+      worker_socket: "{sock}"
+      auto_start_worker: true
+      worker:
+        device: "cpu"
+        eager: true
+        max_num_seqs: 8
+        num_gpu_blocks: 256
+"""
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(cfg)
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        srv.wait_worker()
+        yield srv
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
+
+
+BASE = "http://127.0.0.1:{}"
+
+
+def test_healthz(server):
+    st, body = _http("GET", BASE.format(server.port) + "/healthz")
+    assert st == 200 and body == "ok"
+
+
+def test_health_json(server):
+    st, body = _http("GET", BASE.format(server.port) + "/health")
+    j = json.loads(body)
+    assert st == 200 and j["status"] == "ok" and "uptime_seconds" in j
+
+
+def test_docs_and_openapi(server):
+    st, body = _http("GET", BASE.format(server.port) + "/docs")
+    assert st == 200 and "<html" in body.lower()
+    st, body = _http("GET", BASE.format(server.port) + "/openapi.json")
+    j = json.loads(body)
+    assert st == 200
+    assert j["openapi"].startswith("3.1")
+    assert "/llm-gateway/v1/chat/completions" in j["paths"]
+    assert "/v1/chat/completions" in j["paths"]
+    assert "Problem" in j["components"]["schemas"]
+
+
+def test_not_found_is_problem_json(server):
+    st, body = _http("GET", BASE.format(server.port) + "/nope")
+    j = json.loads(body)
+    assert st == 404 and j["status"] == 404 and j["title"] == "Not Found"
+
+
+def test_validation_error(server):
+    st, body = _http("POST", BASE.format(server.port) + "/v1/chat/completions",
+                     body={"model": "llama3-8b"})
+    j = json.loads(body)
+    assert st == 400 and j["code"] == "validation_error"
+
+
+def test_model_not_found(server):
+    st, body = _http("POST", BASE.format(server.port) + "/v1/chat/completions",
+                     body={"model": "nope-13b",
+                           "messages": [{"role": "user", "content":
+                                         [{"type": "text", "text": "hi"}]}]})
+    j = json.loads(body)
+    assert st == 404 and j["code"] == "model_not_found"
+
+
+def test_chat_completion_sync(server):
+    st, body = _http("POST", BASE.format(server.port) + "/v1/chat/completions",
+                     body={"model": "tiny-llama",
+                           "messages": [{"role": "user", "content":
+                                         [{"type": "text", "text": "hello"}]}],
+                           "max_tokens": 16, "temperature": 0.0})
+    assert st == 200, body
+    j = json.loads(body)
+    assert j["model_used"] == "local::tiny-llama"
+    assert j["usage"]["output_tokens"] >= 1
+    assert isinstance(j["content"], list) and j["content"][0]["type"] == "text"
+
+
+def test_chat_completion_sse_stream(server):
+    url = BASE.format(server.port) + "/llm-gateway/v1/chat/completions"
+    req = urllib.request.Request(url, method="POST")
+    req.add_header("content-type", "application/json")
+    data = json.dumps({"model": "tiny-llama", "stream": True,
+                       "messages": [{"role": "user", "content":
+                                     [{"type": "text", "text": "hi"}]}],
+                       "max_tokens": 8, "temperature": 0.0}).encode()
+    with urllib.request.urlopen(req, data=data, timeout=60) as r:
+        assert r.status == 200
+        assert r.headers["content-type"].startswith("text/event-stream")
+        raw = r.read().decode()
+    events = [l[6:] for l in raw.splitlines() if l.startswith("data: ")]
+    assert events[-1] == "[DONE]"
+    first = json.loads(events[0])
+    assert first["delta"].get("role") == "assistant"
+    final = json.loads(events[-2])
+    assert final.get("finish_reason") in ("stop", "length")
+    assert "usage" in final
+
+
+def test_model_registry_routes(server):
+    st, body = _http("GET", BASE.format(server.port) +
+                     "/model-registry/v1/models")
+    j = json.loads(body)
+    assert st == 200
+    ids = [m["canonical_id"] for m in j["items"]]
+    assert "local::llama3-8b" in ids
+    st, body = _http("GET", BASE.format(server.port) +
+                     "/model-registry/v1/models/local::llama3-8b")
+    assert st == 200
+
+
+def test_types_registry_roundtrip(server):
+    base = BASE.format(server.port) + "/types-registry/v1/entities"
+    ent = {"gts_id": "gts.x.core.test.v1~acme.thing.v1~",
+           "kind": "instance", "payload": {"a": 1}}
+    st, body = _http("POST", base, body={"entities": [ent]})
+    assert st == 201, body
+    st, body = _http("GET", base + "?filter=gts.x.core.test.*")
+    j = json.loads(body)
+    assert st == 200 and len(j["items"]) == 1
+    st, body = _http("GET", base + "/" + ent["gts_id"])
+    assert st == 200 and json.loads(body)["gts_id"] == ent["gts_id"]
+
+
+def test_nodes_registry(server):
+    st, body = _http("GET", BASE.format(server.port) +
+                     "/nodes-registry/v1/nodes")
+    j = json.loads(body)
+    assert st == 200 and j["items"][0]["id"] == "local"
+    assert "gpus" in j["items"][0]
+
+
+def test_credstore_tenant_scoping(server):
+    base = BASE.format(server.port) + "/credstore/v1/secrets/apikey"
+    st, _ = _http("PUT", base, body={"value": "s3cr3t"})
+    assert st == 204
+    st, body = _http("GET", base)
+    assert st == 200 and json.loads(body)["value"] == "s3cr3t"
+    st, _ = _http("DELETE", base)
+    assert st == 204
+    st, _ = _http("GET", base)
+    assert st == 404
+
+
+def test_unsupported_media_type(server):
+    url = BASE.format(server.port) + "/v1/chat/completions"
+    req = urllib.request.Request(url, method="POST", data=b"x=1")
+    req.add_header("content-type", "application/x-www-form-urlencoded")
+    try:
+        with urllib.request.urlopen(req, timeout=10) as r:
+            st = r.status
+            body = r.read().decode()
+    except urllib.error.HTTPError as e:
+        st, body = e.code, e.read().decode()
+    assert st == 415 and json.loads(body)["status"] == 415
+
+
+@pytest.fixture(scope="module")
+def mt_server():
+    """Multi-tenant server: static tenants + static bearer tokens."""
+    import tempfile
+    port = _free_port()
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-mt-")
+    cfg = f"""
+server:
+  home_dir: "/tmp/hs-e2e-mt"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: false
+  tenant-resolver:
+    config:
+      tenants:
+        - id: "00000000-df51-5b42-9538-d2b56b7ee953"
+          name: "Root"
+          status: "active"
+          type: "root"
+        - id: "11111111-1111-1111-1111-111111111111"
+          name: "Acme"
+          status: "active"
+          type: "tenant"
+          parent_id: "00000000-df51-5b42-9538-d2b56b7ee953"
+  authn-resolver:
+    config:
+      tokens:
+        - token: "root-token"
+          subject_id: "root-user"
+          subject_tenant_id: "00000000-df51-5b42-9538-d2b56b7ee953"
+        - token: "acme-token"
+          subject_id: "acme-user"
+          subject_tenant_id: "11111111-1111-1111-1111-111111111111"
+  llm-gateway:
+    config:
+      model: "tiny-llama"
+      worker_socket: "{sock}"
+      auto_start_worker: false
+"""
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(cfg)
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        yield srv
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
+
+
+def test_auth_required_without_token(mt_server):
+    st, body = _http("GET", BASE.format(mt_server.port) +
+                     "/model-registry/v1/models")
+    j = json.loads(body)
+    assert st == 401 and j["status"] == 401
+
+
+def test_auth_invalid_token(mt_server):
+    st, _ = _http("GET", BASE.format(mt_server.port) +
+                  "/model-registry/v1/models", token="wrong")
+    assert st == 401
+
+
+def test_auth_valid_token(mt_server):
+    st, body = _http("GET", BASE.format(mt_server.port) +
+                     "/model-registry/v1/models", token="acme-token")
+    assert st == 200 and "items" in json.loads(body)
+
+
+def test_public_routes_skip_auth(mt_server):
+    st, _ = _http("GET", BASE.format(mt_server.port) + "/healthz")
+    assert st == 200
+
+
+def test_credstore_isolated_between_tenants(mt_server):
+    base = BASE.format(mt_server.port) + "/credstore/v1/secrets/shared-name"
+    st, _ = _http("PUT", base, body={"value": "acme-secret"},
+                  token="acme-token")
+    assert st == 204
+    # other tenant sees 404 (anti-enumeration), not 403
+    st, _ = _http("GET", base, token="root-token")
+    assert st == 404
+    st, body = _http("GET", base, token="acme-token")
+    assert st == 200 and json.loads(body)["value"] == "acme-secret"
+
+
+def test_cli_check_and_introspection():
+    _build_binary()
+    r = subprocess.run([str(BIN), "check", "--config",
+                        str(ROOT / "config" / "no-db.yaml")],
+                       capture_output=True, text=True, cwd=ROOT)
+    assert r.returncode == 0 and "config OK" in r.stdout
+    r = subprocess.run([str(BIN), "--list-modules"], capture_output=True,
+                       text=True, cwd=ROOT)
+    assert "llm-gateway" in r.stdout and "api-gateway" in r.stdout
+    r = subprocess.run([str(BIN), "--print-config", "--config",
+                        str(ROOT / "config" / "no-db.yaml")],
+                       capture_output=True, text=True, cwd=ROOT)
+    cfg = json.loads(r.stdout)
+    assert cfg["modules"]["api-gateway"]["config"]["auth_disabled"] is True
+    r = subprocess.run([str(BIN), "--dump-modules-config-json", "--config",
+                        str(ROOT / "config" / "no-db.yaml")],
+                       capture_output=True, text=True, cwd=ROOT)
+    assert "llm-gateway" in json.loads(r.stdout)
