@@ -28,7 +28,7 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
     const bf16* __restrict__ k_cache, const bf16* __restrict__ v_cache,
     const int* __restrict__ block_tables, const int* __restrict__ ctx_lens,
     const int* __restrict__ row_seq, int max_blocks, int num_kv_heads,
-    float scale) {
+    float scale, long q_stride) {
   constexpr int SLICE = D / 4;        // K elems per lane (4 lanes/token)
   constexpr int VPL = D / 64;         // V elems per lane (lane owns a slice)
   const int row = blockIdx.x;
@@ -49,7 +49,7 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
   for (int i = threadIdx.x; i < GROUP * D; i += 256) {
     const int h = i / D, d = i % D;
     q_lds[h][d] = bf2f(*(const unsigned short*)(
-        q + ((long)row * H + kvh * GROUP + h) * D + d)) * scale;
+        q + (long)row * q_stride + (kvh * GROUP + h) * D + d)) * scale;
   }
   __syncthreads();
 
@@ -165,7 +165,7 @@ void launch_paged_attn(bf16* out, const bf16* q, const bf16* k_cache,
                        const int* ctx_lens, const int* row_seq,
                        long num_rows, int num_kv_heads, int group, int D,
                        int max_blocks, int block_size, float scale,
-                       hipStream_t stream) {
+                       long q_stride, hipStream_t stream) {
   if (block_size != ATTN_BS)
     throw std::runtime_error("paged_attn: block_size must be 16");
   dim3 grid((unsigned)num_rows, (unsigned)num_kv_heads);
@@ -173,7 +173,7 @@ void launch_paged_attn(bf16* out, const bf16* q, const bf16* k_cache,
     if (group == G && D == DD) {                                            \
       paged_attn_kernel<G, DD><<<grid, 256, 0, stream>>>(                   \
           out, q, k_cache, v_cache, block_tables, ctx_lens, row_seq,        \
-          max_blocks, num_kv_heads, scale);                                 \
+          max_blocks, num_kv_heads, scale, q_stride);                       \
       return;                                                               \
     }
   CASE(1, 128) CASE(2, 128) CASE(4, 128) CASE(8, 128)

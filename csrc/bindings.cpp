@@ -17,10 +17,11 @@ void launch_fused_add_rmsnorm(bf16*, bf16*, const bf16*, float, long, int,
                               hipStream_t);
 void launch_rope_kv_append(bf16*, bf16*, const bf16*, const long*,
                            const float*, const long*, bf16*, bf16*, long,
-                           int, int, int, int, hipStream_t);
+                           int, int, int, int, long, long, long,
+                           hipStream_t);
 void launch_paged_attn(bf16*, const bf16*, const bf16*, const bf16*,
                        const int*, const int*, const int*, long, int, int,
-                       int, int, int, float, hipStream_t);
+                       int, int, int, float, long, hipStream_t);
 void launch_silu_mul(bf16*, const bf16*, long, int, hipStream_t);
 void launch_greedy_sample(long*, const bf16*, long, int, hipStream_t);
 void launch_inv_cdf_sample(long*, const float*, const float*, long, int,
@@ -64,13 +65,23 @@ void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
                            x.numel() / hidden, hidden, stream());
 }
 
+// q/k/v may be strided views into one fused qkv buffer: require only the
+// head dim contiguous (stride(2)==1) and head stride == D.
+void check_qkv_view(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.dim() == 3 && t.stride(2) == 1 &&
+                  t.stride(1) == t.size(2),
+              name, " must be [T, H, D] with contiguous heads");
+}
+
 void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                     torch::Tensor positions, torch::Tensor cos_sin,
                     torch::Tensor slot_mapping, torch::Tensor k_cache,
                     torch::Tensor v_cache) {
-  check(q, torch::kBFloat16, "q");
-  check(k, torch::kBFloat16, "k");
-  check(v, torch::kBFloat16, "v");
+  check_qkv_view(q, "q");
+  check_qkv_view(k, "k");
+  check_qkv_view(v, "v");
   check(cos_sin, torch::kFloat32, "cos_sin");
   check(positions, torch::kInt64, "positions");
   check(slot_mapping, torch::kInt64, "slot_mapping");
@@ -81,7 +92,8 @@ void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                         positions.data_ptr<long>(),
                         cos_sin.data_ptr<float>(),
                         slot_mapping.data_ptr<long>(), bf(k_cache),
-                        bf(v_cache), T, H, KV, D, BS, stream());
+                        bf(v_cache), T, H, KV, D, BS, q.stride(0),
+                        k.stride(0), v.stride(0), stream());
 }
 
 void paged_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
@@ -89,7 +101,7 @@ void paged_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                 torch::Tensor ctx_lens,
                 c10::optional<torch::Tensor> row_seq, double scale) {
   check(out, torch::kBFloat16, "out");
-  check(q, torch::kBFloat16, "q");
+  check_qkv_view(q, "q");
   check(block_tables, torch::kInt32, "block_tables");
   check(ctx_lens, torch::kInt32, "ctx_lens");
   const long R = q.size(0);
@@ -104,7 +116,7 @@ void paged_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
   launch_paged_attn(bf(out), cbf(q), cbf(k_cache), cbf(v_cache),
                     block_tables.data_ptr<int>(), ctx_lens.data_ptr<int>(),
                     rs, R, KV, H / KV, D, (int)block_tables.size(1), BS,
-                    (float)scale, stream());
+                    (float)scale, q.stride(0), stream());
 }
 
 void silu_mul(torch::Tensor out, torch::Tensor x) {

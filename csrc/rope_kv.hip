@@ -10,12 +10,15 @@
 // cos_sin: [max_pos, D] float (cos | sin halves), positions: [T] i64
 // slot_mapping: [T] i64 (slot = block*BS + off; <0 = no append)
 // k_cache/v_cache: [NB, KV, BS, D] bf16
+// q/k/v may be strided views into one fused qkv GEMM output buffer
+// (token stride qs/ks/vs in elements; head dim contiguous).
 template <int D>
 __global__ __launch_bounds__(256) void rope_kv_append_kernel(
     bf16* __restrict__ q, bf16* __restrict__ k, const bf16* __restrict__ v,
     const long* __restrict__ positions, const float* __restrict__ cos_sin,
     const long* __restrict__ slot_mapping, bf16* __restrict__ k_cache,
-    bf16* __restrict__ v_cache, int H, int KV, int block_size) {
+    bf16* __restrict__ v_cache, int H, int KV, int block_size,
+    long qs_stride, long ks_stride, long vs_stride) {
   constexpr int HALF = D / 2;
   __shared__ float cs[D];
   const long t = blockIdx.x;
@@ -34,7 +37,7 @@ __global__ __launch_bounds__(256) void rope_kv_append_kernel(
   const int total_q = H * PAIR2;
   for (int u = threadIdx.x; u < total_q; u += 256) {
     const int h = u / PAIR2, j = (u % PAIR2) * 2;
-    bf16* base = q + (t * H + h) * D;
+    bf16* base = q + t * qs_stride + h * D;
     ushort2 lo = *reinterpret_cast<ushort2*>(base + j);
     ushort2 hi = *reinterpret_cast<ushort2*>(base + HALF + j);
     float c0 = cs[j], c1 = cs[j + 1];
@@ -49,7 +52,7 @@ __global__ __launch_bounds__(256) void rope_kv_append_kernel(
   const int total_k = KV * PAIR2;
   for (int u = threadIdx.x; u < total_k; u += 256) {
     const int h = u / PAIR2, j = (u % PAIR2) * 2;
-    bf16* base = k + (t * KV + h) * D;
+    bf16* base = k + t * ks_stride + h * D;
     ushort2 lo = *reinterpret_cast<ushort2*>(base + j);
     ushort2 hi = *reinterpret_cast<ushort2*>(base + HALF + j);
     float c0 = cs[j], c1 = cs[j + 1];
@@ -71,7 +74,7 @@ __global__ __launch_bounds__(256) void rope_kv_append_kernel(
     const int total_v = KV * D / 8;
     for (int u = threadIdx.x; u < total_v; u += 256) {
       const int h = u / (D / 8), j = (u % (D / 8)) * 8;
-      const bf16* src = v + (t * KV + h) * D + j;
+      const bf16* src = v + t * vs_stride + h * D + j;
       bf16* vc = v_cache + (((blk * KV + h) * block_size + off) * D) + j;
       *reinterpret_cast<uint4*>(vc) = *reinterpret_cast<const uint4*>(src);
     }
@@ -82,16 +85,17 @@ void launch_rope_kv_append(bf16* q, bf16* k, const bf16* v,
                            const long* positions, const float* cos_sin,
                            const long* slot_mapping, bf16* k_cache,
                            bf16* v_cache, long T, int H, int KV, int D,
-                           int block_size, hipStream_t stream) {
+                           int block_size, long qs, long ks, long vs,
+                           hipStream_t stream) {
   dim3 grid((unsigned)T);
   if (D == 128)
     rope_kv_append_kernel<128><<<grid, 256, 0, stream>>>(
         q, k, v, positions, cos_sin, slot_mapping, k_cache, v_cache,
-        H, KV, block_size);
+        H, KV, block_size, qs, ks, vs);
   else if (D == 64)
     rope_kv_append_kernel<64><<<grid, 256, 0, stream>>>(
         q, k, v, positions, cos_sin, slot_mapping, k_cache, v_cache,
-        H, KV, block_size);
+        H, KV, block_size, qs, ks, vs);
   else
     throw std::runtime_error("rope_kv_append: unsupported head_dim");
 }

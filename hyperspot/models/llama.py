@@ -104,7 +104,9 @@ class DecoderLayer(nn.Module):
                 meta: ForwardMeta, kv_cache: torch.Tensor,
                 cos_sin: torch.Tensor):
         if residual is None:
-            residual = x.clone()
+            # x is the fresh embedding-gather output: safe to alias as the
+            # residual (fused_add_rmsnorm mutates it in place from layer 1 on)
+            residual = x
             h = ops.rmsnorm(x, self.input_norm_w, self.eps)
         else:
             h, residual = ops.fused_add_rmsnorm(x, residual,

@@ -81,7 +81,7 @@ def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
                       v_cache: torch.Tensor, block_tables: torch.Tensor,
                       seq_lens: torch.Tensor, scale: float) -> torch.Tensor:
     if q.is_cuda:
-        out = torch.empty_like(q)
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _native().paged_attn(out, q, k_cache, v_cache, block_tables,
                              seq_lens, None, scale)
         return out
@@ -103,7 +103,7 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         return paged_attn_decode(q, k_cache, v_cache, meta.block_tables,
                                  meta.seq_lens, scale)
     if q.is_cuda:
-        out = torch.empty_like(q)
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _native().paged_attn(out, q, k_cache, v_cache, meta.block_tables,
                              meta.ctx_lens, meta.row_seq, scale)
         return out

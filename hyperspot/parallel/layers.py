@@ -98,13 +98,16 @@ class QKVParallelLinear(nn.Module):
             vf[r * ks:(r + 1) * ks]], 0), requires_grad=False)
 
     def forward(self, x: torch.Tensor):
+        """Returns strided [T, n, D] views into ONE fused qkv buffer — no
+        .contiguous() copies; the HIP kernels take the token stride."""
         qkv = x @ self.weight.t()
         qs, ks = self.nh * self.head_dim, self.nkv * self.head_dim
-        q, k, v = qkv.split([qs, ks, ks], dim=-1)
         T = x.shape[0]
-        return (q.view(T, self.nh, self.head_dim).contiguous(),
-                k.view(T, self.nkv, self.head_dim).contiguous(),
-                v.view(T, self.nkv, self.head_dim).contiguous())
+        d = self.head_dim
+        q = qkv[:, :qs].view(T, self.nh, d)
+        k = qkv[:, qs:qs + ks].view(T, self.nkv, d)
+        v = qkv[:, qs + ks:].view(T, self.nkv, d)
+        return q, k, v
 
 
 class RowParallelLinear(nn.Module):
