@@ -22,6 +22,9 @@ void launch_rope_kv_append(bf16*, bf16*, const bf16*, const long*,
 void launch_paged_attn(bf16*, const bf16*, const bf16*, const bf16*,
                        const int*, const int*, const int*, long, int, int,
                        int, int, int, float, long, hipStream_t);
+void launch_attn_prefill_mfma(bf16*, const bf16*, const bf16*,
+                              const bf16*, const int*, int, int, int, int,
+                              int, float, long, long, long, hipStream_t);
 void launch_silu_mul(bf16*, const bf16*, long, int, hipStream_t);
 void launch_greedy_sample(long*, const bf16*, long, int, hipStream_t);
 void launch_inv_cdf_sample(long*, const float*, const float*, long, int,
@@ -119,6 +122,22 @@ void paged_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                     (float)scale, q.stride(0), stream());
 }
 
+void attn_prefill_mfma(torch::Tensor out, torch::Tensor q, torch::Tensor k,
+                       torch::Tensor v, torch::Tensor seq_start,
+                       long max_seqlen, double scale) {
+  check(out, torch::kBFloat16, "out");
+  check_qkv_view(q, "q");
+  check_qkv_view(k, "k");
+  check_qkv_view(v, "v");
+  check(seq_start, torch::kInt32, "seq_start");
+  const int H = (int)q.size(1), D = (int)q.size(2);
+  const int KV = (int)k.size(1);
+  launch_attn_prefill_mfma(
+      bf(out), cbf(q), cbf(k), cbf(v), seq_start.data_ptr<int>(),
+      (int)seq_start.numel() - 1, (int)max_seqlen, H, KV, D, (float)scale,
+      q.stride(0), k.stride(0), v.stride(0), stream());
+}
+
 void silu_mul(torch::Tensor out, torch::Tensor x) {
   check(out, torch::kBFloat16, "out");
   check(x, torch::kBFloat16, "x");
@@ -152,6 +171,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
   m.def("rope_kv_append", &rope_kv_append);
   m.def("paged_attn", &paged_attn);
+  m.def("attn_prefill_mfma", &attn_prefill_mfma);
   m.def("silu_mul", &silu_mul);
   m.def("greedy_sample", &greedy_sample);
   m.def("inv_cdf_sample", &inv_cdf_sample);

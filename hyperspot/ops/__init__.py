@@ -104,8 +104,15 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                                  meta.seq_lens, scale)
     if q.is_cuda:
         out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
-        _native().paged_attn(out, q, k_cache, v_cache, meta.block_tables,
-                             meta.ctx_lens, meta.row_seq, scale)
+        if q.shape[-1] == 128:
+            # MFMA flash prefill (K/V straight from the qkv projection)
+            _native().attn_prefill_mfma(out, q, k, v, meta.seq_start,
+                                        meta.max_seqlen, scale)
+        else:
+            # fallback: per-row attention over the freshly appended cache
+            _native().paged_attn(out, q, k_cache, v_cache,
+                                 meta.block_tables, meta.ctx_lens,
+                                 meta.row_seq, scale)
         return out
     return torch_ref.prefill_attn(q, k, v, meta.seq_start, scale)
 

@@ -138,6 +138,47 @@ def test_paged_attn_prefill_matches_varlen_ref():
     assert torch.allclose(out.float().cpu(), ref, atol=2e-2, rtol=2e-2)
 
 
+@pytest.mark.parametrize("lens,H,KV", [
+    ([5, 300, 64], 8, 2),     # partial tiles + >256-row tile crossing
+    ([513], 4, 4),            # MHA, 3 q tiles
+    ([33, 16], 8, 1),         # group=8
+])
+def test_attn_prefill_mfma_vs_ref(lens, H, KV):
+    torch.manual_seed(1)
+    D = 128
+    T = sum(lens)
+    q = _randn_bf16(T, H, D, seed=11)
+    k = _randn_bf16(T, KV, D, seed=12)
+    v = _randn_bf16(T, KV, D, seed=13)
+    ss = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32, device=DEV)
+    out = torch.empty_like(q)
+    from hyperspot import _C
+    _C.attn_prefill_mfma(out, q, k, v, ss, max(lens), D ** -0.5)
+    ref = R.prefill_attn(q.float().cpu(), k.float().cpu(), v.float().cpu(),
+                         ss.cpu(), D ** -0.5)
+    assert torch.allclose(out.float().cpu(), ref, atol=2.5e-2, rtol=2.5e-2)
+
+
+def test_attn_prefill_mfma_strided_views():
+    """Exercise the fused-qkv strided-view path end to end."""
+    torch.manual_seed(2)
+    H, KV, D = 4, 2, 128
+    lens = [40, 17]
+    T = sum(lens)
+    qkv = _randn_bf16(T, (H + 2 * KV) * D, seed=21)
+    q = qkv[:, :H * D].view(T, H, D)
+    k = qkv[:, H * D:(H + KV) * D].view(T, KV, D)
+    v = qkv[:, (H + KV) * D:].view(T, KV, D)
+    ss = torch.tensor([0, 40, 57], dtype=torch.int32, device=DEV)
+    out = torch.empty(T, H, D, dtype=torch.bfloat16, device=DEV)
+    from hyperspot import _C
+    _C.attn_prefill_mfma(out, q, k, v, ss, 40, D ** -0.5)
+    ref = R.prefill_attn(q.float().cpu(), k.float().cpu(), v.float().cpu(),
+                         ss.cpu(), D ** -0.5)
+    assert torch.allclose(out.float().cpu(), ref, atol=2.5e-2, rtol=2.5e-2)
+
+
 def test_silu_mul():
     x = _randn_bf16(65, 2 * 1024)
     out = ops.silu_mul(x)
