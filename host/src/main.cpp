@@ -15,6 +15,7 @@
 #include "modkit/modkit.h"
 #include "modules/api_gateway.h"
 #include "modules/llm_gateway.h"
+#include "modules/oagw.h"
 #include "modules/system_modules.h"
 #include "util/log.h"
 
@@ -45,6 +46,10 @@ int run_server(const Json& cfg, bool check_only) {
   registry.add(std::make_shared<ModelRegistryModule>());
   registry.add(std::make_shared<CredStoreModule>());
   registry.add(std::make_shared<ServerlessRuntimeModule>());
+  registry.add(std::make_shared<OagwModule>());
+  registry.add(std::make_shared<FileStorageModule>());
+  auto orch = std::make_shared<ModuleOrchestratorModule>();
+  registry.add(orch);
   registry.add(std::make_shared<LlmGatewayModule>());
 
   ClientHub hub;
@@ -64,6 +69,11 @@ int run_server(const Json& cfg, bool check_only) {
   // phases: init -> post_init -> rest -> start (reference
   // host_runtime.rs:717 run_phases_internal; db/grpc/oop phases live
   // inside the owning modules here)
+  {
+    std::vector<std::string> names;
+    for (auto& m : sorted) names.push_back(m->name());
+    orch->set_modules(names);
+  }
   try {
     for (auto& [m, ctx] : mods) m->init(ctx);
     for (auto& [m, ctx] : mods) m->post_init(ctx);
@@ -96,7 +106,8 @@ int list_modules(const Json& cfg) {
   for (const char* m : {"api-gateway", "authn-resolver", "tenant-resolver",
                         "authz-resolver", "types-registry",
                         "nodes-registry", "model-registry", "credstore",
-                        "serverless-runtime", "llm-gateway"})
+                        "serverless-runtime", "oagw", "file-storage",
+                        "module-orchestrator", "llm-gateway"})
     std::cout << m << "\n";
   return 0;
 }

@@ -74,14 +74,28 @@ const Route* RestRegistry::match(const std::string& method,
   if (path_exists) *path_exists = false;
   const Route* best = nullptr;
   for (const auto& r : routes_) {
-    if (r.segments.size() != segs.size()) continue;
+    const bool wildcard = !r.segments.empty() &&
+        r.segments.back().rfind("{*", 0) == 0;
+    if (wildcard ? segs.size() < r.segments.size() - 1
+                 : r.segments.size() != segs.size())
+      continue;
     std::map<std::string, std::string> p;
     bool ok = true;
-    for (size_t i = 0; i < segs.size(); ++i) {
+    const size_t fixed = r.segments.size() - (wildcard ? 1 : 0);
+    for (size_t i = 0; i < fixed && ok; ++i) {
       const std::string& rs = r.segments[i];
       if (rs.size() >= 2 && rs.front() == '{' && rs.back() == '}')
-        p[rs.substr(1, rs.size() - 2)] = segs[i];
-      else if (rs != segs[i]) { ok = false; break; }
+        p[rs.substr(1, rs.size() - 2)] = i < segs.size() ? segs[i] : "";
+      else if (i >= segs.size() || rs != segs[i]) ok = false;
+    }
+    if (ok && wildcard) {
+      std::string restpath;
+      for (size_t i = fixed; i < segs.size(); ++i) {
+        if (!restpath.empty()) restpath += '/';
+        restpath += segs[i];
+      }
+      const std::string& w = r.segments.back();
+      p[w.substr(2, w.size() - 3)] = restpath;
     }
     if (!ok) continue;
     if (path_exists) *path_exists = true;

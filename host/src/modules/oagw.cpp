@@ -1,0 +1,254 @@
+#include "oagw.h"
+
+#include "../http/client.h"
+#include "../util/log.h"
+#include "system_modules.h"
+
+namespace hs {
+
+namespace {
+
+SecurityContext sec_of(HttpRequest& req) {
+  return SecurityContext::from_json(req.extensions.at("security"));
+}
+
+// validate the Upstream model shape (oagw-sdk/src/models.rs:272-284)
+void validate_upstream(const Json& u) {
+  if (!u.at("alias").is_string() || u.at("alias").as_string().empty())
+    throw Problem::bad_request("'alias' required");
+  const Json& eps = u.path("server.endpoints");
+  if (!eps.is_array() || eps.size() == 0)
+    throw Problem::bad_request("'server.endpoints' required");
+  for (const auto& e : eps.arr()) {
+    const std::string scheme = e.at("scheme").as_string("http");
+    if (scheme != "http" && scheme != "https")
+      throw Problem::bad_request("unsupported scheme '" + scheme +
+                                 "' (http/https only)");
+    if (!e.at("host").is_string())
+      throw Problem::bad_request("endpoint host required");
+  }
+}
+
+}  // namespace
+
+void OagwModule::init(ModuleCtx& ctx) {
+  hub_ = ctx.hub;
+  // config-seeded upstreams (the reference's InMemoryCredentialResolver
+  // pattern: config `upstreams:` list for static deployments)
+  const Json& ups = ctx.config.at("upstreams");
+  if (ups.is_array()) {
+    for (auto u : ups.arr()) {
+      std::string id = "up-" + std::to_string(next_id_++);
+      u["id"] = id;
+      if (!u.contains("tenant_id")) u["tenant_id"] = kDefaultTenantId;
+      if (!u.contains("enabled")) u["enabled"] = true;
+      upstreams_[id] = u;
+    }
+  }
+}
+
+Json* OagwModule::find_upstream(const std::string& tenant,
+                                const std::string& alias) {
+  for (auto& [id, u] : upstreams_) {
+    if (u.at("alias").as_string() == alias &&
+        u.at("enabled").as_bool(true) &&
+        (u.at("tenant_id").as_string() == tenant ||
+         u.at("tenant_id").as_string() == kDefaultTenantId))
+      return &u;
+  }
+  return nullptr;
+}
+
+void OagwModule::proxy(HttpRequest& req, ResponseWriter& w) {
+  SecurityContext sec = sec_of(req);
+  std::string alias = req.path_params["alias"];
+  std::string suffix = req.path_params["path"];
+  Json up;
+  {
+    std::lock_guard<std::mutex> lk(mu_);
+    Json* u = find_upstream(sec.tenant_id, alias);
+    if (!u)
+      throw Problem{404, "Not Found", "about:blank",
+                    "no upstream with alias '" + alias + "'",
+                    "upstream_not_found"};
+    up = *u;
+    // per-upstream rate limit (token-bucket default,
+    // oagw-sdk RateLimitConfig)
+    const Json& rl = up.at("rate_limit");
+    if (rl.is_object()) {
+      auto& b = limiters_[up.at("id").as_string()];
+      if (!b)
+        b = std::make_unique<TokenBucket>(rl.at("sustained").as_number(50),
+                                          rl.at("burst").as_number(100));
+      if (!b->try_acquire())
+        throw Problem{429, "Too Many Requests", "about:blank",
+                      "upstream rate limit", "rate_limited"};
+    }
+  }
+  const Json& ep = up.path("server.endpoints").arr()[0];
+  const std::string host = ep.at("host").as_string();
+  const int port = (int)ep.at("port").as_int(80);
+
+  // header policy: drop hop-by-hop + authorization (never forwarded),
+  // pass the rest (reference src/infra/proxy/headers.rs allow-list idea)
+  std::map<std::string, std::string> fwd;
+  for (auto& [k, v] : req.headers) {
+    if (k == "host" || k == "connection" || k == "authorization" ||
+        k == "content-length" || k == "transfer-encoding")
+      continue;
+    fwd[k] = v;
+  }
+  // auth plugin: apikey header injection, value inline or from credstore
+  const Json& auth = up.at("auth");
+  if (auth.is_object() &&
+      auth.at("plugin_type").as_string() == "apikey") {
+    std::string header = auth.path("config.header").as_string("x-api-key");
+    std::string value = auth.path("config.value").as_string("");
+    std::string ref = auth.path("config.credential_ref").as_string("");
+    if (!ref.empty()) {
+      auto cs = hub_->get<CredStoreClient>("credstore");
+      auto v = cs ? cs->get(sec.tenant_id, ref) : std::nullopt;
+      if (!v)
+        throw Problem{502, "Bad Gateway", "about:blank",
+                      "credential_ref not resolvable", "provider_error"};
+      value = *v;
+    }
+    if (!value.empty()) fwd[header] = value;
+  }
+
+  std::string target = "/" + suffix;
+  if (req.target.find('?') != std::string::npos)
+    target += req.target.substr(req.target.find('?'));
+
+  // forward (buffered; the llm engine path does not go through OAGW)
+  auto resp = http_request(host, port, req.method, target, fwd, req.body,
+                           10000);
+  if (!resp)
+    throw Problem{502, "Bad Gateway", "about:blank",
+                  "upstream connect failed", "provider_error"};
+  std::string ct = resp->headers.count("content-type")
+      ? resp->headers["content-type"] : "application/octet-stream";
+  std::vector<std::pair<std::string, std::string>> hdrs;
+  for (auto& [k, v] : resp->headers) {
+    if (k == "content-type" || k == "content-length" ||
+        k == "connection" || k == "transfer-encoding")
+      continue;
+    hdrs.push_back({k, v});
+  }
+  hdrs.push_back({"x-oagw-upstream", alias});
+  w.respond(resp->status, ct, resp->body, hdrs);
+}
+
+void OagwModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
+  auto crud = [this, &rest](const std::string& kind,
+                            std::map<std::string, Json>* store,
+                            bool validate) {
+    OperationSpec list;
+    list.method = "GET";
+    list.path = "/oagw/v1/" + kind;
+    list.operation_id = "oagw_" + kind + "_list";
+    list.authenticated = true;
+    list.tags = {"oagw"};
+    rest.register_op(list, [this, store](HttpRequest& rq,
+                                         ResponseWriter& w) {
+      // ListQuery top/skip (oagw-sdk models.rs:293-303)
+      long top = rq.query.count("top") ? atol(rq.query["top"].c_str()) : 50;
+      long skip = rq.query.count("skip") ? atol(rq.query["skip"].c_str()) : 0;
+      SecurityContext sec = sec_of(rq);
+      Json items = Json::array();
+      std::lock_guard<std::mutex> lk(mu_);
+      long i = 0;
+      for (auto& [id, u] : *store) {
+        if (u.at("tenant_id").as_string() != sec.tenant_id &&
+            u.at("tenant_id").as_string() != kDefaultTenantId)
+          continue;
+        if (i++ < skip) continue;
+        if ((long)items.size() >= top) break;
+        items.push_back(u);
+      }
+      Json out = Json::object();
+      out["items"] = items;
+      w.respond(200, "application/json", out.dump());
+    });
+
+    OperationSpec create;
+    create.method = "POST";
+    create.path = "/oagw/v1/" + kind;
+    create.operation_id = "oagw_" + kind + "_create";
+    create.authenticated = true;
+    create.allowed_content_types = {"application/json"};
+    create.tags = {"oagw"};
+    rest.register_op(create, [this, store, validate, kind](
+                                 HttpRequest& rq, ResponseWriter& w) {
+      Json body;
+      try { body = Json::parse(rq.body); }
+      catch (...) { throw Problem::bad_request("invalid JSON"); }
+      if (validate) validate_upstream(body);
+      SecurityContext sec = sec_of(rq);
+      std::lock_guard<std::mutex> lk(mu_);
+      std::string id = kind.substr(0, 2) + "-" + std::to_string(next_id_++);
+      body["id"] = id;
+      body["tenant_id"] = sec.tenant_id;
+      if (!body.contains("enabled")) body["enabled"] = true;
+      (*store)[id] = body;
+      w.respond(201, "application/json", body.dump());
+    });
+
+    OperationSpec get;
+    get.method = "GET";
+    get.path = "/oagw/v1/" + kind + "/{id}";
+    get.operation_id = "oagw_" + kind + "_get";
+    get.authenticated = true;
+    get.tags = {"oagw"};
+    rest.register_op(get, [this, store](HttpRequest& rq,
+                                        ResponseWriter& w) {
+      SecurityContext sec = sec_of(rq);
+      std::lock_guard<std::mutex> lk(mu_);
+      auto it = store->find(rq.path_params["id"]);
+      if (it == store->end() ||
+          (it->second.at("tenant_id").as_string() != sec.tenant_id &&
+           it->second.at("tenant_id").as_string() != kDefaultTenantId))
+        throw Problem::not_found();
+      w.respond(200, "application/json", it->second.dump());
+    });
+
+    OperationSpec del;
+    del.method = "DELETE";
+    del.path = "/oagw/v1/" + kind + "/{id}";
+    del.operation_id = "oagw_" + kind + "_delete";
+    del.authenticated = true;
+    del.tags = {"oagw"};
+    rest.register_op(del, [this, store](HttpRequest& rq,
+                                        ResponseWriter& w) {
+      SecurityContext sec = sec_of(rq);
+      std::lock_guard<std::mutex> lk(mu_);
+      auto it = store->find(rq.path_params["id"]);
+      if (it == store->end() ||
+          it->second.at("tenant_id").as_string() != sec.tenant_id)
+        throw Problem::not_found();
+      store->erase(it);
+      w.respond(204, "application/json", "");
+    });
+  };
+  crud("upstreams", &upstreams_, true);
+  crud("routes", &routes_, false);
+
+  OperationSpec px;
+  px.method = "POST";
+  px.path = "/oagw/v1/proxy/{alias}/{*path}";
+  px.operation_id = "oagw_proxy_post";
+  px.summary = "Proxy a request through a configured upstream";
+  px.authenticated = true;
+  px.tags = {"oagw"};
+  rest.register_op(px, [this](HttpRequest& rq, ResponseWriter& w) {
+    proxy(rq, w);
+  });
+  OperationSpec pxg = px;
+  pxg.method = "GET";
+  pxg.operation_id = "oagw_proxy_get";
+  rest.register_op(pxg, [this](HttpRequest& rq, ResponseWriter& w) {
+    proxy(rq, w);
+  });
+}
+
+}  // namespace hs

@@ -1,6 +1,7 @@
 #include "system_modules.h"
 
 #include <sys/sysinfo.h>
+#include <sys/stat.h>
 #include <unistd.h>
 
 #include <cstring>
@@ -512,6 +513,136 @@ void CredStoreModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
     if (!g_creds->del(sec_of(rq).tenant_id, rq.path_params["ref"]))
       throw Problem::not_found();
     w.respond(204, "application/json", "");
+  });
+}
+
+// ----------------------------------------------------------- file-storage
+
+namespace {
+
+std::string sanitize_rel_path(const std::string& p) {
+  if (p.empty() || p.find("..") != std::string::npos || p.front() == '/')
+    throw Problem::bad_request("invalid file name");
+  return p;
+}
+
+void mkdirs(const std::string& path) {
+  std::string cur;
+  for (size_t i = 0; i < path.size(); ++i) {
+    if (path[i] == '/' && !cur.empty()) mkdir(cur.c_str(), 0755);
+    cur += path[i];
+  }
+  mkdir(cur.c_str(), 0755);
+}
+
+class LocalFileStorage : public FileStorageClient {
+ public:
+  explicit LocalFileStorage(std::string root) : root_(std::move(root)) {}
+  std::string root_for(const std::string& tenant) override {
+    std::string dir = root_ + "/" + tenant;
+    mkdirs(dir);
+    return dir;
+  }
+
+ private:
+  std::string root_;
+};
+
+std::shared_ptr<LocalFileStorage> g_files;
+
+}  // namespace
+
+void FileStorageModule::init(ModuleCtx& ctx) {
+  std::string home = ctx.home_dir;
+  if (!home.empty() && home[0] == '~') {
+    const char* h = getenv("HOME");
+    home = std::string(h ? h : "/tmp") + home.substr(1);
+  }
+  root_ = ctx.config.at("root").as_string(home + "/file-storage");
+  mkdirs(root_);
+  g_files = std::make_shared<LocalFileStorage>(root_);
+  ctx.hub->register_client<FileStorageClient>("file-storage", g_files);
+}
+
+void FileStorageModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
+  OperationSpec put;
+  put.method = "PUT";
+  put.path = "/file-storage/v1/files/{*name}";
+  put.operation_id = "files_put";
+  put.summary = "Store a file (checkpoints, media)";
+  put.authenticated = true;
+  put.tags = {"file-storage"};
+  rest.register_op(put, [](HttpRequest& rq, ResponseWriter& w) {
+    std::string rel = sanitize_rel_path(rq.path_params["name"]);
+    std::string dir = g_files->root_for(sec_of(rq).tenant_id);
+    std::string full = dir + "/" + rel;
+    auto slash = full.rfind('/');
+    mkdirs(full.substr(0, slash));
+    std::ofstream f(full, std::ios::binary | std::ios::trunc);
+    if (!f) throw Problem{500, "Internal Server Error", "about:blank",
+                          "cannot write file", ""};
+    f.write(rq.body.data(), (long)rq.body.size());
+    Json meta = Json::object();
+    meta["name"] = rel;
+    meta["size"] = rq.body.size();
+    meta["mime"] = rq.header("content-type", "application/octet-stream");
+    w.respond(201, "application/json", meta.dump());
+  });
+
+  OperationSpec get;
+  get.method = "GET";
+  get.path = "/file-storage/v1/files/{*name}";
+  get.operation_id = "files_get";
+  get.authenticated = true;
+  get.tags = {"file-storage"};
+  rest.register_op(get, [](HttpRequest& rq, ResponseWriter& w) {
+    std::string rel = sanitize_rel_path(rq.path_params["name"]);
+    std::string full =
+        g_files->root_for(sec_of(rq).tenant_id) + "/" + rel;
+    std::ifstream f(full, std::ios::binary);
+    if (!f) throw Problem::not_found();
+    std::stringstream ss;
+    ss << f.rdbuf();
+    w.respond(200, "application/octet-stream", ss.str());
+  });
+
+  OperationSpec del;
+  del.method = "DELETE";
+  del.path = "/file-storage/v1/files/{*name}";
+  del.operation_id = "files_delete";
+  del.authenticated = true;
+  del.tags = {"file-storage"};
+  rest.register_op(del, [](HttpRequest& rq, ResponseWriter& w) {
+    std::string rel = sanitize_rel_path(rq.path_params["name"]);
+    std::string full =
+        g_files->root_for(sec_of(rq).tenant_id) + "/" + rel;
+    if (unlink(full.c_str()) != 0) throw Problem::not_found();
+    w.respond(204, "application/json", "");
+  });
+}
+
+// ---------------------------------------------------- module-orchestrator
+
+void ModuleOrchestratorModule::register_rest(ModuleCtx& ctx,
+                                             RestRegistry& rest) {
+  OperationSpec list;
+  list.method = "GET";
+  list.path = "/module-orchestrator/v1/modules";
+  list.operation_id = "orchestrator_modules";
+  list.summary = "Registered module instances";
+  list.authenticated = true;
+  list.tags = {"module-orchestrator"};
+  rest.register_op(list, [this](HttpRequest& rq, ResponseWriter& w) {
+    Json items = Json::array();
+    for (auto& n : module_names_) {
+      Json m = Json::object();
+      m["name"] = n;
+      m["status"] = "running";
+      items.push_back(m);
+    }
+    Json out = Json::object();
+    out["items"] = items;
+    w.respond(200, "application/json", out.dump());
   });
 }
 

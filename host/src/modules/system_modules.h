@@ -126,4 +126,37 @@ class CredStoreModule : public Module {
   void register_rest(ModuleCtx& ctx, RestRegistry& rest) override;
 };
 
+// ---- file-storage (binary store; serves model checkpoints for registry
+//      hot-swap — reference modules/file-storage/docs/PRD.md) ----
+struct FileStorageClient {
+  virtual ~FileStorageClient() = default;
+  virtual std::string root_for(const std::string& tenant) = 0;
+};
+
+class FileStorageModule : public Module {
+ public:
+  std::string name() const override { return "file-storage"; }
+  void init(ModuleCtx& ctx) override;
+  void register_rest(ModuleCtx& ctx, RestRegistry& rest) override;
+
+ private:
+  std::string root_;
+};
+
+// ---- module-orchestrator (module/worker instance visibility;
+//      reference modules/system/module-orchestrator DirectoryService) ----
+class ModuleOrchestratorModule : public Module {
+ public:
+  std::string name() const override { return "module-orchestrator"; }
+  void register_rest(ModuleCtx& ctx, RestRegistry& rest) override;
+  void post_init(ModuleCtx& ctx) override { registry_size_ = 0; }
+  void set_modules(std::vector<std::string> names) {
+    module_names_ = std::move(names);
+  }
+
+ private:
+  std::vector<std::string> module_names_;
+  size_t registry_size_ = 0;
+};
+
 }  // namespace hs

@@ -1,0 +1,166 @@
+"""e2e: OAGW control+data plane and file-storage (the reference's
+oagw/tests/proxy_integration.rs + e2e suites, rebuilt in pytest against a
+live mock upstream)."""
+
+import json
+import threading
+import http.server
+from pathlib import Path
+
+import pytest
+
+from tests.test_host_e2e import (ServerProc, _free_port, _http, BASE)
+
+
+class MockUpstream(http.server.BaseHTTPRequestHandler):
+    """Echo server recording received headers."""
+
+    def _respond(self):
+        length = int(self.headers.get("content-length", 0))
+        body = self.rfile.read(length) if length else b""
+        out = json.dumps({
+            "path": self.path,
+            "method": self.command,
+            "api_key": self.headers.get("x-api-key", ""),
+            "authorization": self.headers.get("authorization", ""),
+            "echo": body.decode(errors="replace"),
+        }).encode()
+        self.send_response(200)
+        self.send_header("content-type", "application/json")
+        self.send_header("content-length", str(len(out)))
+        self.end_headers()
+        self.wfile.write(out)
+
+    do_GET = _respond
+    do_POST = _respond
+
+    def log_message(self, *a):
+        pass
+
+
+@pytest.fixture(scope="module")
+def upstream():
+    srv = http.server.ThreadingHTTPServer(("127.0.0.1", 0), MockUpstream)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    yield srv.server_address[1]
+    srv.shutdown()
+
+
+@pytest.fixture(scope="module")
+def server(tmp_path_factory):
+    import tempfile
+    port = _free_port()
+    home = tmp_path_factory.mktemp("hs-oagw-home")
+    cfg = f"""
+server:
+  home_dir: "{home}"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+  llm-gateway:
+    config:
+      auto_start_worker: false
+"""
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(cfg)
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        yield srv
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
+
+
+def test_upstream_crud_and_proxy(server, upstream):
+    base = BASE.format(server.port)
+    up = {"alias": "prov", "server": {"endpoints": [
+        {"scheme": "http", "host": "127.0.0.1", "port": upstream}]},
+        "auth": {"plugin_type": "apikey",
+                 "config": {"header": "x-api-key", "value": "k-123"}}}
+    st, body = _http("POST", base + "/oagw/v1/upstreams", body=up)
+    assert st == 201, body
+    uid = json.loads(body)["id"]
+
+    st, body = _http("GET", base + "/oagw/v1/upstreams")
+    assert st == 200 and len(json.loads(body)["items"]) == 1
+
+    # data plane: inject api key, never forward client authorization
+    st, body = _http("POST", base + "/oagw/v1/proxy/prov/v1/chat?x=1",
+                     body={"q": 2}, token="client-secret-token")
+    assert st == 200, body
+    j = json.loads(body)
+    assert j["path"] == "/v1/chat?x=1"
+    assert j["api_key"] == "k-123"
+    assert j["authorization"] == ""          # stripped (credential hygiene)
+    assert json.loads(j["echo"]) == {"q": 2}
+
+    st, _ = _http("DELETE", base + f"/oagw/v1/upstreams/{uid}")
+    assert st == 204
+    st, body = _http("POST", base + "/oagw/v1/proxy/prov/v1/chat", body={})
+    assert st == 404 and json.loads(body)["code"] == "upstream_not_found"
+
+
+def test_upstream_validation(server):
+    base = BASE.format(server.port)
+    st, body = _http("POST", base + "/oagw/v1/upstreams", body={"alias": ""})
+    assert st == 400
+    st, body = _http("POST", base + "/oagw/v1/upstreams",
+                     body={"alias": "x", "server": {"endpoints": [
+                         {"scheme": "ftp", "host": "h"}]}})
+    assert st == 400 and "scheme" in json.loads(body)["detail"]
+
+
+def test_credential_ref_from_credstore(server, upstream):
+    base = BASE.format(server.port)
+    st, _ = _http("PUT", base + "/credstore/v1/secrets/prov-key",
+                  body={"value": "from-credstore"})
+    assert st == 204
+    up = {"alias": "prov2", "server": {"endpoints": [
+        {"scheme": "http", "host": "127.0.0.1", "port": upstream}]},
+        "auth": {"plugin_type": "apikey",
+                 "config": {"header": "x-api-key",
+                            "credential_ref": "prov-key"}}}
+    st, _ = _http("POST", base + "/oagw/v1/upstreams", body=up)
+    assert st == 201
+    st, body = _http("GET", base + "/oagw/v1/proxy/prov2/ping")
+    assert st == 200 and json.loads(body)["api_key"] == "from-credstore"
+
+
+def test_file_storage_roundtrip(server):
+    base = BASE.format(server.port)
+    data = {"value": "x" * 1000}
+    st, body = _http("PUT", base +
+                     "/file-storage/v1/files/ckpt/llama/model.bin",
+                     body=data)
+    assert st == 201, body
+    meta = json.loads(body)
+    assert meta["name"] == "ckpt/llama/model.bin" and meta["size"] > 900
+    st, body = _http("GET", base +
+                     "/file-storage/v1/files/ckpt/llama/model.bin")
+    assert st == 200 and json.loads(body) == data
+    st, _ = _http("DELETE", base +
+                  "/file-storage/v1/files/ckpt/llama/model.bin")
+    assert st == 204
+    st, _ = _http("GET", base + "/file-storage/v1/files/ckpt/llama/model.bin")
+    assert st == 404
+
+
+def test_file_storage_path_traversal_blocked(server):
+    base = BASE.format(server.port)
+    st, _ = _http("PUT", base + "/file-storage/v1/files/../evil",
+                  body={"v": 1})
+    assert st == 400
+
+
+def test_module_orchestrator_lists_modules(server):
+    st, body = _http("GET", BASE.format(server.port) +
+                     "/module-orchestrator/v1/modules")
+    names = [m["name"] for m in json.loads(body)["items"]]
+    assert "llm-gateway" in names and "oagw" in names
