@@ -8,24 +8,41 @@ not a parallelism one.  Layout per layer:
 
 so one (block, kv_head) is a contiguous block_size*head_dim*2B tile — 4 KB at
 block 16 / head_dim 128 — the unit the decode kernel streams.
+
+The per-sequence state lives in flat numpy arrays indexed by a row id
+(tables matrix, token counts) so the decode-step input preparation is a few
+vectorized ops instead of per-sequence Python loops — at batch 256 the
+Python path was costing more than the GPU step itself.
 """
 
 from __future__ import annotations
 
 from typing import Dict, List
 
+import numpy as np
 import torch
 
 
 class BlockManager:
-    """Free-list allocator over the paged pool + per-sequence block tables."""
+    """Free-list allocator over the paged pool + per-sequence block tables.
 
-    def __init__(self, num_blocks: int, block_size: int):
+    Sequences occupy a row in a [capacity, max_blocks] int32 table matrix;
+    `rows_state()` exposes the numpy views the ModelRunner batches over.
+    """
+
+    def __init__(self, num_blocks: int, block_size: int,
+                 capacity: int = 1024, max_blocks_per_seq: int = 2048):
         self.num_blocks = num_blocks
         self.block_size = block_size
+        self.capacity = capacity
+        self.max_blocks_per_seq = max_blocks_per_seq
         self.free_blocks: List[int] = list(range(num_blocks - 1, -1, -1))
-        self.tables: Dict[str, List[int]] = {}
-        self.seq_tokens: Dict[str, int] = {}
+        self.row_of: Dict[str, int] = {}
+        self._free_rows: List[int] = list(range(capacity - 1, -1, -1))
+        self.tables_np = np.zeros((capacity, max_blocks_per_seq),
+                                  dtype=np.int32)
+        self.ntables_np = np.zeros(capacity, dtype=np.int32)
+        self.tokens_np = np.zeros(capacity, dtype=np.int64)
 
     @property
     def num_free(self) -> int:
@@ -41,38 +58,76 @@ class BlockManager:
         need = self.blocks_needed(num_tokens)
         if need > self.num_free:
             raise RuntimeError("KV pool exhausted")
+        if not self._free_rows:
+            raise RuntimeError("sequence table capacity exhausted")
+        row = self._free_rows.pop()
+        self.row_of[seq_id] = row
         blocks = [self.free_blocks.pop() for _ in range(need)]
-        self.tables[seq_id] = blocks
-        self.seq_tokens[seq_id] = num_tokens
+        self.tables_np[row, :need] = blocks
+        self.ntables_np[row] = need
+        self.tokens_np[row] = num_tokens
         return blocks
 
+    def tokens_of(self, seq_id: str) -> int:
+        return int(self.tokens_np[self.row_of[seq_id]])
+
     def can_append(self, seq_id: str) -> bool:
-        n = self.seq_tokens[seq_id]
-        return n % self.block_size != 0 or self.num_free > 0
+        row = self.row_of[seq_id]
+        return int(self.tokens_np[row]) % self.block_size != 0 \
+            or self.num_free > 0
 
     def append_slot(self, seq_id: str) -> int:
         """Reserve the slot for one more token; returns its flat slot id."""
-        n = self.seq_tokens[seq_id]
-        table = self.tables[seq_id]
+        row = self.row_of[seq_id]
+        n = int(self.tokens_np[row])
         if n % self.block_size == 0:
             if not self.free_blocks:
                 raise RuntimeError("KV pool exhausted")
-            table.append(self.free_blocks.pop())
-        self.seq_tokens[seq_id] = n + 1
-        return table[n // self.block_size] * self.block_size + n % self.block_size
+            nt = int(self.ntables_np[row])
+            self.tables_np[row, nt] = self.free_blocks.pop()
+            self.ntables_np[row] = nt + 1
+        self.tokens_np[row] = n + 1
+        blk = int(self.tables_np[row, n // self.block_size])
+        return blk * self.block_size + n % self.block_size
 
     def slot_of(self, seq_id: str, pos: int) -> int:
-        table = self.tables[seq_id]
-        return table[pos // self.block_size] * self.block_size + pos % self.block_size
+        row = self.row_of[seq_id]
+        return int(self.tables_np[row, pos // self.block_size]) \
+            * self.block_size + pos % self.block_size
 
     def free(self, seq_id: str) -> None:
-        blocks = self.tables.pop(seq_id, None)
-        if blocks:
-            self.free_blocks.extend(reversed(blocks))
-        self.seq_tokens.pop(seq_id, None)
+        row = self.row_of.pop(seq_id, None)
+        if row is None:
+            return
+        nt = int(self.ntables_np[row])
+        self.free_blocks.extend(int(b)
+                                for b in self.tables_np[row, :nt][::-1])
+        self.ntables_np[row] = 0
+        self.tokens_np[row] = 0
+        self._free_rows.append(row)
 
     def table(self, seq_id: str) -> List[int]:
-        return self.tables[seq_id]
+        row = self.row_of[seq_id]
+        return [int(b) for b in
+                self.tables_np[row, : int(self.ntables_np[row])]]
+
+    # ---- vectorized decode-step state (ModelRunner fast path) ----
+
+    def append_slots_batch(self, rows: np.ndarray) -> np.ndarray:
+        """Reserve one slot for every row; returns flat slot ids [B]."""
+        n = self.tokens_np[rows]
+        boundary = np.nonzero(n % self.block_size == 0)[0]
+        for i in boundary:          # rare: one new block per 16 steps/seq
+            row = int(rows[i])
+            if not self.free_blocks:
+                raise RuntimeError("KV pool exhausted")
+            nt = int(self.ntables_np[row])
+            self.tables_np[row, nt] = self.free_blocks.pop()
+            self.ntables_np[row] = nt + 1
+        blk = self.tables_np[rows, n // self.block_size]
+        slots = blk.astype(np.int64) * self.block_size + n % self.block_size
+        self.tokens_np[rows] = n + 1
+        return slots
 
 
 def allocate_kv_caches(num_layers: int, num_blocks: int, kv_heads: int,

@@ -47,10 +47,22 @@ class ModelRunner:
         self.kv_caches = allocate_kv_caches(
             self.spec.num_layers, self.num_blocks, kv_heads_local,
             config.block_size, self.spec.head_dim, self.dtype, self.device)
-        self.block_manager = BlockManager(self.num_blocks, config.block_size)
         self.max_blocks_per_seq = (config.max_model_len + config.block_size - 1) \
             // config.block_size
+        self.block_manager = BlockManager(
+            self.num_blocks, config.block_size,
+            capacity=max(config.max_num_seqs * 2, 64),
+            max_blocks_per_seq=self.max_blocks_per_seq)
         self._gen = torch.Generator().manual_seed(config.seed)
+        # pinned staging for the decode hot path (vectorized input prep)
+        pin = self.device.type == "cuda"
+        cap = self.block_manager.capacity
+        self._st_ids = torch.empty(cap, dtype=torch.long, pin_memory=pin)
+        self._st_pos = torch.empty(cap, dtype=torch.long, pin_memory=pin)
+        self._st_slots = torch.empty(cap, dtype=torch.long, pin_memory=pin)
+        self._st_lens = torch.empty(cap, dtype=torch.int32, pin_memory=pin)
+        self._st_bt_flat = torch.empty(cap * self.max_blocks_per_seq,
+                                       dtype=torch.int32, pin_memory=pin)
         self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
         self._graph_io: Dict[int, dict] = {}
         self._graph_pool = None
@@ -63,14 +75,19 @@ class ModelRunner:
         slots: List[int] = []
         starts = [0]
         logit_rows = []
+        import numpy as np
         row_seq: List[int] = []
         tables = []
+        bm = self.block_manager
+        bs = bm.block_size
         for si, req in enumerate(batch.requests):
             n = req.num_prompt_tokens
             ids.extend(req.prompt_token_ids)
             pos.extend(range(n))
-            bm = self.block_manager
-            slots.extend(bm.slot_of(req.request_id, p) for p in range(n))
+            row = bm.row_of[req.request_id]
+            p = np.arange(n)
+            s = bm.tables_np[row, p // bs].astype(np.int64) * bs + p % bs
+            slots.extend(s.tolist())
             starts.append(starts[-1] + n)
             logit_rows.append(starts[-1] - 1)
             row_seq.extend([si] * n)
@@ -96,29 +113,36 @@ class ModelRunner:
         return input_ids, meta
 
     def _decode_inputs(self, batch: ScheduledBatch):
+        """Vectorized decode-step prep: numpy state + pinned staging (the
+        per-sequence Python loop cost more than the GPU step at batch 256)."""
+        import numpy as np
         bm = self.block_manager
-        ids, pos, slots, seq_lens = [], [], [], []
-        tables = []
-        for req in batch.requests:
-            ids.append(req.last_token_id)
-            p = bm.seq_tokens[req.request_id]       # next position index
-            pos.append(p)
-            slots.append(bm.append_slot(req.request_id))
-            seq_lens.append(p + 1)
-            tables.append(bm.table(req.request_id))
-        max_t = max(len(t) for t in tables)
-        bt = torch.zeros(len(tables), max_t, dtype=torch.int32)
-        for i, t in enumerate(tables):
-            bt[i, :len(t)] = torch.tensor(t, dtype=torch.int32)
+        reqs = batch.requests
+        B = len(reqs)
+        rows = np.fromiter((bm.row_of[r.request_id] for r in reqs),
+                           dtype=np.int64, count=B)
+        ids = np.fromiter((r.last_token_id for r in reqs),
+                          dtype=np.int64, count=B)
+        pos = bm.tokens_np[rows].copy()             # next position index
+        slots = bm.append_slots_batch(rows)
+        max_nt = int(bm.ntables_np[rows].max())
+        bt = bm.tables_np[rows[:, None], np.arange(max_nt)[None, :]]
+        self._st_ids.numpy()[:B] = ids
+        self._st_pos.numpy()[:B] = pos
+        self._st_slots.numpy()[:B] = slots
+        self._st_lens.numpy()[:B] = pos + 1
+        bt_stage = self._st_bt_flat[:B * max_nt].view(B, max_nt)
+        bt_stage.numpy()[:] = bt
         d = self.device
+        nb = d.type == "cuda"
         meta = ForwardMeta(
             mode="decode",
-            positions=torch.tensor(pos, dtype=torch.long, device=d),
-            slot_mapping=torch.tensor(slots, dtype=torch.long, device=d),
-            block_tables=bt.to(d),
-            seq_lens=torch.tensor(seq_lens, dtype=torch.int32, device=d),
+            positions=self._st_pos[:B].to(d, non_blocking=nb),
+            slot_mapping=self._st_slots[:B].to(d, non_blocking=nb),
+            block_tables=bt_stage.to(d, non_blocking=nb),
+            seq_lens=self._st_lens[:B].to(d, non_blocking=nb),
         )
-        input_ids = torch.tensor(ids, dtype=torch.long, device=d)
+        input_ids = self._st_ids[:B].to(d, non_blocking=nb)
         return input_ids, meta
 
     # ---------------- execution ----------------

@@ -97,7 +97,7 @@ class Scheduler:
         while self.running:
             # total fresh-block demand of this decode step vs the free list
             need = sum(1 for req in self.running
-                       if self.bm.seq_tokens[req.request_id]
+                       if self.bm.tokens_of(req.request_id)
                        % self.bm.block_size == 0)
             if need <= self.bm.num_free:
                 break
