@@ -1,21 +1,23 @@
-// Unified paged attention for gfx950 — serves BOTH the decode step (one
-// query token per sequence) and prefill (every prompt row attends over the
-// cache slots the fused RoPE+append kernel just wrote, so no separate
-// varlen kernel is needed; GEMMs dominate prefill and the K/V re-reads hit
-// the 256 MiB Infinity Cache).
+// Unified paged attention for gfx950 — serves the decode step (one query
+// token per sequence) and short/odd-shape prefill fallback (the MFMA flash
+// kernel covers head_dim-128 prefill).
 //
 // Geometry (memory-bound design, wave64-native):
 //   grid = (num_query_rows, num_kv_heads); block = 256 threads = 4 waves.
 //   A workgroup owns one (row, kv_head): it processes the whole GQA group
 //   (GROUP query heads) so K/V stream from HBM exactly once per group.
-//   Waves split the KV blocks round-robin (flash-decode style) and keep
-//   per-head online-softmax state (m, l, acc); a final LDS combine merges
-//   the four waves.
-//   K tile read: 4 lanes per token x 32B-slices -> 1 KiB coalesced per
-//   16-token page; V read: lane owns a D/64-element slice of every row.
+//   Waves split the KV pages round-robin (flash-decode style) with
+//   per-head online-softmax state; a final LDS combine merges the waves.
 //
-// BLOCK_SIZE is fixed at 16 tokens (one KV page = 16 x 128 x 2B = 4 KiB per
-// head — the pool unit sized for 288 GB HBM3E).
+//   K page read: 4 lanes per token x 64B slices -> 4 KiB coalesced/page.
+//   V page read: lane owns a D/64-element column slice of all 16 rows.
+//   v2 (profiles/r01: v1 was LDS-issue-bound at 155us/layer, 14x off
+//   roofline): the lane's q slice lives in REGISTERS as raw bf16 pairs
+//   (GROUP<=4) or is read as float4 from LDS (GROUP=8); score/softmax
+//   LDS traffic is float4; V rows are preloaded per page.
+//
+// BLOCK_SIZE is fixed at 16 tokens (one KV page = 16 x 128 x 2B = 4 KiB
+// per head — the pool unit sized for 288 GB HBM3E).
 #include <float.h>
 
 #include "common.h"
@@ -31,6 +33,7 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
     float scale, long q_stride) {
   constexpr int SLICE = D / 4;        // K elems per lane (4 lanes/token)
   constexpr int VPL = D / 64;         // V elems per lane (lane owns a slice)
+  constexpr bool QREG = (GROUP <= 4); // q slice in registers vs LDS
   const int row = blockIdx.x;
   const int kvh = blockIdx.y;
   const int H = num_kv_heads * GROUP;
@@ -39,25 +42,41 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
   const int* bt = block_tables + (long)table * max_blocks;
   const int nblocks = (ctx + ATTN_BS - 1) / ATTN_BS;
 
-  __shared__ float q_lds[GROUP][D];
-  __shared__ float sc[4][GROUP][ATTN_BS];
-  __shared__ float comb_o[4][GROUP][D];
+  __shared__ __attribute__((aligned(16))) float q_lds[GROUP][D];
+  __shared__ __attribute__((aligned(16))) float sc[4][GROUP][ATTN_BS];
+  __shared__ __attribute__((aligned(16))) float comb_o[4][GROUP][D];
   __shared__ float comb_m[4][GROUP];
   __shared__ float comb_l[4][GROUP];
-
-  // stage the query group, pre-scaled
-  for (int i = threadIdx.x; i < GROUP * D; i += 256) {
-    const int h = i / D, d = i % D;
-    q_lds[h][d] = bf2f(*(const unsigned short*)(
-        q + (long)row * q_stride + (kvh * GROUP + h) * D + d)) * scale;
-  }
-  __syncthreads();
 
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int tok = lane >> 2;          // 0..15
   const int sl = lane & 3;            // K slice index
   const int d0 = sl * SLICE;
+
+  // ---- stage the query group ----
+  unsigned q_pk[QREG ? GROUP : 1][QREG ? SLICE / 2 : 1];  // raw bf16 pairs
+  if constexpr (QREG) {
+    const bf16* qp = q + (long)row * q_stride + (long)kvh * GROUP * D + d0;
+    #pragma unroll
+    for (int h = 0; h < GROUP; ++h) {
+      #pragma unroll
+      for (int i = 0; i < SLICE / 8; ++i) {
+        uint4 r4 = *reinterpret_cast<const uint4*>(qp + h * D + i * 8);
+        q_pk[h][i * 4 + 0] = r4.x;
+        q_pk[h][i * 4 + 1] = r4.y;
+        q_pk[h][i * 4 + 2] = r4.z;
+        q_pk[h][i * 4 + 3] = r4.w;
+      }
+    }
+  } else {
+    for (int i = threadIdx.x; i < GROUP * D; i += 256) {
+      const int h = i / D, d = i % D;
+      q_lds[h][d] = bf2f(*(const unsigned short*)(
+          q + (long)row * q_stride + (kvh * GROUP + h) * D + d));
+    }
+    __syncthreads();
+  }
 
   float m[GROUP], l[GROUP], acc[GROUP][VPL];
   #pragma unroll
@@ -71,35 +90,61 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
     const long blk = bt[b];
     const int nb = min(ATTN_BS, ctx - b * ATTN_BS);
     // ---- K dot: lane covers SLICE elems of its token's key ----
-    float kf[SLICE];
+    unsigned k_pk[SLICE / 2];
     {
       const bf16* kp = k_cache +
           (((blk * num_kv_heads + kvh) * ATTN_BS + tok) * D + d0);
       #pragma unroll
       for (int i = 0; i < SLICE / 8; ++i) {
-        bf16x8 kv8 = load_bf16x8(kp + i * 8);
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) kf[i * 8 + j] = bf16x8_get(kv8, j);
+        uint4 r4 = *reinterpret_cast<const uint4*>(kp + i * 8);
+        k_pk[i * 4 + 0] = r4.x;
+        k_pk[i * 4 + 1] = r4.y;
+        k_pk[i * 4 + 2] = r4.z;
+        k_pk[i * 4 + 3] = r4.w;
       }
     }
     #pragma unroll
     for (int h = 0; h < GROUP; ++h) {
       float p = 0.f;
-      #pragma unroll
-      for (int i = 0; i < SLICE; ++i) p += q_lds[h][d0 + i] * kf[i];
+      if constexpr (QREG) {
+        #pragma unroll
+        for (int i = 0; i < SLICE / 2; ++i) {
+          const unsigned qa = q_pk[h][i], ka = k_pk[i];
+          p += bf2f((unsigned short)(qa & 0xffff)) *
+               bf2f((unsigned short)(ka & 0xffff));
+          p += bf2f((unsigned short)(qa >> 16)) *
+               bf2f((unsigned short)(ka >> 16));
+        }
+      } else {
+        #pragma unroll
+        for (int i = 0; i < SLICE / 4; ++i) {
+          const float4 qv = *reinterpret_cast<const float4*>(
+              &q_lds[h][d0 + i * 4]);
+          p += qv.x * bf2f((unsigned short)(k_pk[i * 2] & 0xffff));
+          p += qv.y * bf2f((unsigned short)(k_pk[i * 2] >> 16));
+          p += qv.z * bf2f((unsigned short)(k_pk[i * 2 + 1] & 0xffff));
+          p += qv.w * bf2f((unsigned short)(k_pk[i * 2 + 1] >> 16));
+        }
+      }
       p += __shfl_xor(p, 1);
       p += __shfl_xor(p, 2);
       if (sl == 0)
-        sc[wid][h][tok] = (tok < nb) ? p : -FLT_MAX;
+        sc[wid][h][tok] = (tok < nb) ? p * scale : -FLT_MAX;
     }
-    // wave-private LDS area; DS ops of one wave are in program order,
-    // so no barrier is needed before re-reading sc.
+    // wave-private LDS area; DS ops of one wave are in program order.
+    // ---- online softmax update + exponentiate in LDS ----
+    float mn_h[GROUP];
     #pragma unroll
     for (int h = 0; h < GROUP; ++h) {
       float bm = -FLT_MAX;
       #pragma unroll
-      for (int t = 0; t < ATTN_BS; ++t) bm = fmaxf(bm, sc[wid][h][t]);
+      for (int t4 = 0; t4 < ATTN_BS / 4; ++t4) {
+        const float4 s4 = *reinterpret_cast<const float4*>(
+            &sc[wid][h][t4 * 4]);
+        bm = fmaxf(fmaxf(bm, fmaxf(s4.x, s4.y)), fmaxf(s4.z, s4.w));
+      }
       const float mn = fmaxf(m[h], bm);
+      mn_h[h] = mn;
       const float alpha = (m[h] == -FLT_MAX) ? 0.f : __expf(m[h] - mn);
       #pragma unroll
       for (int i = 0; i < VPL; ++i) acc[h][i] *= alpha;
@@ -108,30 +153,42 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
       if (sl == 0)  // one lane per token exponentiates it
         sc[wid][h][tok] = (tok < nb) ? __expf(sc[wid][h][tok] - mn) : 0.f;
     }
-    // ---- V accumulate: lane owns elements [lane*VPL, lane*VPL+VPL) ----
+    // ---- V: preload the lane's column slice of all 16 rows ----
     const bf16* vbase = v_cache +
         ((blk * num_kv_heads + kvh) * ATTN_BS) * D + lane * VPL;
-    for (int t = 0; t < nb; ++t) {
-      float vf[VPL];
-      if constexpr (VPL == 2) {
-        ushort2 vv = *reinterpret_cast<const ushort2*>(vbase + (long)t * D);
-        vf[0] = bf2f(vv.x); vf[1] = bf2f(vv.y);
+    float vf[ATTN_BS][VPL];
+    #pragma unroll
+    for (int t = 0; t < ATTN_BS; ++t) {
+      if (t < nb) {
+        if constexpr (VPL == 2) {
+          ushort2 vv = *reinterpret_cast<const ushort2*>(vbase + (long)t * D);
+          vf[t][0] = bf2f(vv.x); vf[t][1] = bf2f(vv.y);
+        } else {
+          vf[t][0] = bf2f(*(const unsigned short*)(vbase + (long)t * D));
+        }
       } else {
-        vf[0] = bf2f(*(const unsigned short*)(vbase + (long)t * D));
-      }
-      #pragma unroll
-      for (int h = 0; h < GROUP; ++h) {
-        const float p = sc[wid][h][t];
         #pragma unroll
-        for (int i = 0; i < VPL; ++i) acc[h][i] += p * vf[i];
+        for (int i = 0; i < VPL; ++i) vf[t][i] = 0.f;
       }
     }
     #pragma unroll
     for (int h = 0; h < GROUP; ++h) {
       float s = 0.f;
       #pragma unroll
-      for (int t = 0; t < ATTN_BS; ++t) s += sc[wid][h][t];
+      for (int t4 = 0; t4 < ATTN_BS / 4; ++t4) {
+        const float4 p4 = *reinterpret_cast<const float4*>(
+            &sc[wid][h][t4 * 4]);
+        #pragma unroll
+        for (int i = 0; i < VPL; ++i) {
+          acc[h][i] += p4.x * vf[t4 * 4 + 0][i];
+          acc[h][i] += p4.y * vf[t4 * 4 + 1][i];
+          acc[h][i] += p4.z * vf[t4 * 4 + 2][i];
+          acc[h][i] += p4.w * vf[t4 * 4 + 3][i];
+        }
+        s += p4.x + p4.y + p4.z + p4.w;
+      }
       l[h] += s;
+      (void)mn_h;
     }
   }
 

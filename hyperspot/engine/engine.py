@@ -76,6 +76,20 @@ class LLMEngine:
     def capture_graphs(self) -> None:
         self.runner.capture_all_graphs()
 
+    def embed(self, prompts):
+        """Mean-pooled embeddings; caller must not be mid-step."""
+        return self.runner.embed(prompts)
+
+    def swap_weights(self, checkpoint_path: str) -> float:
+        """Live weight hot-swap (same architecture); KV pool and captured
+        decode graphs survive (in-place parameter copy)."""
+        return self.runner.swap_weights(checkpoint_path)
+
+    def save_checkpoint(self, path: str) -> None:
+        from .checkpoint import save_checkpoint
+        save_checkpoint(self.runner.model, path,
+                        {"model": self.config.model})
+
     def generate(self, prompts: List[List[int]],
                  sampling: Optional[SamplingParams] = None):
         """Synchronous batch generation helper (tests / offline)."""

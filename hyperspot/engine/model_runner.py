@@ -220,6 +220,43 @@ class ModelRunner:
             if b <= self.config.max_num_seqs:
                 self._capture(b)
 
+    # ---------------- embeddings ----------------
+
+    @torch.inference_mode()
+    def embed(self, prompts: List[List[int]]) -> torch.Tensor:
+        """Mean-pooled final hidden state per prompt -> [N, hidden] fp32.
+
+        Runs outside the scheduler on temporary KV pages (freed at the end);
+        serves the llm-gateway /embeddings contract."""
+        from .request import Request
+        from .config import SamplingParams
+        bm = self.block_manager
+        fake = []
+        try:
+            for i, p in enumerate(prompts):
+                rid = f"__embed_{i}"
+                bm.allocate(rid, len(p))
+                fake.append(Request(request_id=rid, prompt_token_ids=list(p),
+                                    sampling=SamplingParams()))
+            batch = ScheduledBatch(mode="prefill", requests=fake,
+                                   num_tokens=sum(len(p) for p in prompts))
+            input_ids, meta = self._prefill_inputs(batch)
+            meta.return_hidden = True
+            hidden = self.model(input_ids, meta, self.kv_caches).float()
+            out = torch.empty(len(prompts), hidden.shape[-1],
+                              dtype=torch.float32, device=hidden.device)
+            ss = meta.seq_start
+            for i in range(len(prompts)):
+                out[i] = hidden[int(ss[i]):int(ss[i + 1])].mean(0)
+        finally:
+            for i in range(len(prompts)):
+                bm.free(f"__embed_{i}")
+        return out.cpu()
+
+    def swap_weights(self, checkpoint_path: str) -> float:
+        from .checkpoint import load_checkpoint_into
+        return load_checkpoint_into(self.model, checkpoint_path)
+
     # ---------------- sampling ----------------
 
     def _sample(self, logits: torch.Tensor, requests: List[Request]) -> torch.Tensor:

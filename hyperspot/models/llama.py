@@ -45,6 +45,8 @@ class ForwardMeta:
     seq_lens: Optional[torch.Tensor] = None      # [B] int32
     # logits are computed only for these token rows (last token per seq)
     logits_indices: Optional[torch.Tensor] = None
+    # embeddings mode: return final hidden states instead of logits
+    return_hidden: bool = False
 
 
 class Attention(nn.Module):
@@ -155,6 +157,8 @@ class LlamaForCausalLM(nn.Module):
             x, residual = layer(x, residual, meta, kv_caches[i], self.cos_sin)
         x, _ = ops.fused_add_rmsnorm(x, residual, self.final_norm_w,
                                      self.spec.rms_eps)
+        if meta.return_hidden:
+            return x
         if meta.logits_indices is not None:
             x = x[meta.logits_indices]
         return x @ self.lm_head.t()

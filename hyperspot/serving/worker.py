@@ -98,6 +98,24 @@ def handle_conn(conn: socket.socket, state: WorkerState, model_name: str):
                 state.abort(msg.get("id", ""))
             elif t == "chat":
                 _run_chat(msg, state, send)
+            elif t == "embeddings":
+                _run_embeddings(msg, state, send)
+            elif t == "swap":
+                path = msg.get("checkpoint", "")
+                try:
+                    with state.lock:
+                        secs = state.engine.swap_weights(path)
+                    send({"event": "swapped", "seconds": secs,
+                          "checkpoint": path})
+                except Exception as e:
+                    send({"event": "error", "message": f"swap failed: {e}"})
+            elif t == "save_checkpoint":
+                try:
+                    with state.lock:
+                        state.engine.save_checkpoint(msg.get("path", ""))
+                    send({"event": "saved", "path": msg.get("path", "")})
+                except Exception as e:
+                    send({"event": "error", "message": f"save failed: {e}"})
             else:
                 send({"event": "error", "message": f"unknown type {t!r}"})
     except (ConnectionResetError, BrokenPipeError, OSError):
@@ -160,6 +178,32 @@ def _run_chat(msg, state: WorkerState, send):
                   "usage": {"input_tokens": len(prompt_ids),
                             "output_tokens": n_out}})
             return
+
+
+def _run_embeddings(msg, state: WorkerState, send):
+    inputs = msg.get("input")
+    if isinstance(inputs, str):
+        inputs = [inputs]
+    if not isinstance(inputs, list) or not inputs:
+        send({"event": "error", "message": "input must be string or array"})
+        return
+    prompts = [state.tokenizer.encode(str(t), add_bos=True) for t in inputs]
+    try:
+        with state.lock:
+            vecs = state.engine.embed(prompts)
+    except Exception as e:
+        send({"event": "error", "message": f"embed failed: {e}"})
+        return
+    dims = msg.get("dimensions")
+    data = []
+    for i, v in enumerate(vecs):
+        emb = v.tolist()
+        if isinstance(dims, int) and 0 < dims < len(emb):
+            emb = emb[:dims]
+        data.append(emb)
+    send({"event": "embeddings", "data": data,
+          "usage": {"input_tokens": sum(len(p) for p in prompts),
+                    "output_tokens": 0}})
 
 
 def main():
