@@ -217,6 +217,219 @@ __global__ __launch_bounds__(256) void paged_attn_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// v3 (D=128, GROUP<=4): latency-oriented redesign.  v2's q-in-register
+// layout (4 lanes/token, 32-elem q slice) cost 244 VGPRs -> occupancy 2
+// waves/SIMD, and its V read was 16x 4-byte loads/page; with two dependent
+// HBM round trips per page the kernel ran 5x off the KV-stream roofline
+// (334us/layer at bs=256 ctx=512, profiles/r01).  v3:
+//   - K phase: 8 lanes/token (16-elem slice) -> q_pk is GROUP*8 dwords
+//     (32 VGPRs at GROUP=4); dot via v_dot2_f32_bf16 (guide: packed bf16
+//     dot at f32 accumulate, 6x fewer VALU ops than scalar cvt+fma).
+//   - V phase: lane owns an 8-elem dim slice (lane&15) of 4 consecutive
+//     tokens (lane>>4) -> 4x dwordx4 loads/page; PV accumulate in packed
+//     f32 FMAs (SLP-packed float2 math), P stays f32 (no bf16 round).
+//   - All K+V loads for the page are issued together at the top of the
+//     loop so one HBM round trip covers both phases.
+//   - Softmax is shuffle-only (score all-reduce over slice lanes, max and
+//     denominator over token lanes); LDS is used once per page to move
+//     P from K-lane-layout to V-lane-layout (wave-private, DS ops of one
+//     wave are in program order).
+// ---------------------------------------------------------------------------
+
+DEV float dot2_bf16(unsigned a, unsigned b, float c) {
+  asm("v_dot2_f32_bf16 %0, %1, %2, %0" : "+v"(c) : "v"(a), "v"(b));
+  return c;
+}
+
+template <int GROUP>
+__global__ __launch_bounds__(256) void paged_attn_v3_kernel(
+    bf16* __restrict__ out, const bf16* __restrict__ q,
+    const bf16* __restrict__ k_cache, const bf16* __restrict__ v_cache,
+    const int* __restrict__ block_tables, const int* __restrict__ ctx_lens,
+    const int* __restrict__ row_seq, int max_blocks, int num_kv_heads,
+    float scale, long q_stride) {
+  constexpr int D = 128;
+  const int row = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int H = num_kv_heads * GROUP;
+  const int ctx = ctx_lens[row];
+  const int table = row_seq ? row_seq[row] : row;
+  const int* bt = block_tables + (long)table * max_blocks;
+  const int nblocks = (ctx + ATTN_BS - 1) / ATTN_BS;
+
+  __shared__ __attribute__((aligned(16))) float sc[4][GROUP][ATTN_BS];
+  __shared__ __attribute__((aligned(16))) float comb_o[4][GROUP][D];
+  __shared__ float comb_m[4][GROUP];
+  __shared__ float comb_l[4][GROUP];
+
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  // K-phase role: token t_loc = lane>>3 (8 per pass, 2 passes/page),
+  // slice sl = lane&7 -> elems [sl*16, sl*16+16).
+  const int t_loc = lane >> 3;
+  const int sl = lane & 7;
+  // V-phase role: dim slice ds = lane&15 -> elems [ds*8, ds*8+8),
+  // token group g = lane>>4 -> tokens [4g, 4g+4).
+  const int ds = lane & 15;
+  const int g = lane >> 4;
+
+  // ---- query group: raw packed-bf16 slice, GROUP*8 dwords ----
+  unsigned q_pk[GROUP][8];
+  {
+    const bf16* qp = q + (long)row * q_stride + (long)kvh * GROUP * D
+                     + sl * 16;
+    #pragma unroll
+    for (int h = 0; h < GROUP; ++h) {
+      uint4 a = *reinterpret_cast<const uint4*>(qp + h * D);
+      uint4 b = *reinterpret_cast<const uint4*>(qp + h * D + 8);
+      q_pk[h][0] = a.x; q_pk[h][1] = a.y; q_pk[h][2] = a.z; q_pk[h][3] = a.w;
+      q_pk[h][4] = b.x; q_pk[h][5] = b.y; q_pk[h][6] = b.z; q_pk[h][7] = b.w;
+    }
+  }
+
+  float m[GROUP], l[GROUP];
+  float2 acc[GROUP][4];
+  #pragma unroll
+  for (int h = 0; h < GROUP; ++h) {
+    m[h] = -FLT_MAX; l[h] = 0.f;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) acc[h][j] = make_float2(0.f, 0.f);
+  }
+
+  for (int b = wid; b < nblocks; b += 4) {
+    const long blk = bt[b];
+    const int nb = min(ATTN_BS, ctx - b * ATTN_BS);
+    const bf16* pbase = k_cache + ((blk * num_kv_heads + kvh) * ATTN_BS) * D;
+    const bf16* vbase = v_cache + ((blk * num_kv_heads + kvh) * ATTN_BS) * D;
+    // ---- issue every load for this page up front (one round trip) ----
+    uint4 kr[2][2], vr[4];
+    #pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      const bf16* kp = pbase + (p * 8 + t_loc) * D + sl * 16;
+      kr[p][0] = *reinterpret_cast<const uint4*>(kp);
+      kr[p][1] = *reinterpret_cast<const uint4*>(kp + 8);
+    }
+    #pragma unroll
+    for (int i = 0; i < 4; ++i)
+      vr[i] = *reinterpret_cast<const uint4*>(vbase + (4 * g + i) * D
+                                              + ds * 8);
+    // ---- scores: dot2 over the lane's 16-elem slice, then all-reduce
+    // over the 8 slice lanes (bits 0..2) ----
+    float s[GROUP][2];
+    #pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      const unsigned* kw = reinterpret_cast<const unsigned*>(&kr[p][0]);
+      #pragma unroll
+      for (int h = 0; h < GROUP; ++h) {
+        float d = 0.f;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) d = dot2_bf16(q_pk[h][j], kw[j], d);
+        d += __shfl_xor(d, 1);
+        d += __shfl_xor(d, 2);
+        d += __shfl_xor(d, 4);
+        const int tok = p * 8 + t_loc;
+        s[h][p] = (tok < nb) ? d * scale : -FLT_MAX;
+      }
+    }
+    // ---- online softmax, shuffle-only (token index lives in bits 3..5) --
+    #pragma unroll
+    for (int h = 0; h < GROUP; ++h) {
+      float bm = fmaxf(s[h][0], s[h][1]);
+      bm = fmaxf(bm, __shfl_xor(bm, 8));
+      bm = fmaxf(bm, __shfl_xor(bm, 16));
+      bm = fmaxf(bm, __shfl_xor(bm, 32));
+      const float mn = fmaxf(m[h], bm);
+      const float alpha = (m[h] == -FLT_MAX) ? 0.f : __expf(m[h] - mn);
+      const float p0 = __expf(s[h][0] - mn);  // masked lanes underflow to 0
+      const float p1 = __expf(s[h][1] - mn);
+      float ps = p0 + p1;
+      ps += __shfl_xor(ps, 8);
+      ps += __shfl_xor(ps, 16);
+      ps += __shfl_xor(ps, 32);
+      l[h] = l[h] * alpha + ps;
+      m[h] = mn;
+      #pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        acc[h][j].x *= alpha;
+        acc[h][j].y *= alpha;
+      }
+      if (sl == 0) {
+        sc[wid][h][t_loc] = p0;
+        sc[wid][h][8 + t_loc] = p1;
+      }
+    }
+    // ---- PV: lane accumulates its dim slice over its 4 tokens ----
+    float4 p4[GROUP];
+    #pragma unroll
+    for (int h = 0; h < GROUP; ++h)
+      p4[h] = *reinterpret_cast<const float4*>(&sc[wid][h][4 * g]);
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const unsigned* vw = reinterpret_cast<const unsigned*>(&vr[i]);
+      float2 vf[4];
+      #pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        vf[j].x = bf2f((unsigned short)(vw[j] & 0xffff));
+        vf[j].y = bf2f((unsigned short)(vw[j] >> 16));
+      }
+      #pragma unroll
+      for (int h = 0; h < GROUP; ++h) {
+        const float p = (&p4[h].x)[i];
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          acc[h][j].x += p * vf[j].x;
+          acc[h][j].y += p * vf[j].y;
+        }
+      }
+    }
+  }
+
+  // ---- reduce the 4 token groups (bits 4..5); m,l are wave-uniform ----
+  #pragma unroll
+  for (int h = 0; h < GROUP; ++h) {
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      acc[h][j].x += __shfl_xor(acc[h][j].x, 16);
+      acc[h][j].x += __shfl_xor(acc[h][j].x, 32);
+      acc[h][j].y += __shfl_xor(acc[h][j].y, 16);
+      acc[h][j].y += __shfl_xor(acc[h][j].y, 32);
+    }
+  }
+  // ---- combine the 4 waves via LDS ----
+  if (lane < 16) {
+    #pragma unroll
+    for (int h = 0; h < GROUP; ++h)
+      #pragma unroll
+      for (int j = 0; j < 4; ++j)
+        *reinterpret_cast<float2*>(&comb_o[wid][h][ds * 8 + j * 2]) =
+            acc[h][j];
+  }
+  if (lane == 0) {
+    #pragma unroll
+    for (int h = 0; h < GROUP; ++h) {
+      comb_m[wid][h] = m[h];
+      comb_l[wid][h] = l[h];
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < GROUP * D; i += 256) {
+    const int h = i / D, d = i % D;
+    float M = fmaxf(fmaxf(comb_m[0][h], comb_m[1][h]),
+                    fmaxf(comb_m[2][h], comb_m[3][h]));
+    float o = 0.f, L = 0.f;
+    #pragma unroll
+    for (int w = 0; w < 4; ++w) {
+      const float f = (comb_m[w][h] == -FLT_MAX) ? 0.f
+                                                 : __expf(comb_m[w][h] - M);
+      o += f * comb_o[w][h][d];
+      L += f * comb_l[w][h];
+    }
+    *(unsigned short*)(out + ((long)row * H + kvh * GROUP + h) * D + d) =
+        f2bf(o / L);
+  }
+}
+
 void launch_paged_attn(bf16* out, const bf16* q, const bf16* k_cache,
                        const bf16* v_cache, const int* block_tables,
                        const int* ctx_lens, const int* row_seq,
@@ -226,6 +439,15 @@ void launch_paged_attn(bf16* out, const bf16* q, const bf16* k_cache,
   if (block_size != ATTN_BS)
     throw std::runtime_error("paged_attn: block_size must be 16");
   dim3 grid((unsigned)num_rows, (unsigned)num_kv_heads);
+  #define CASE3(G)                                                          \
+    if (group == G && D == 128) {                                           \
+      paged_attn_v3_kernel<G><<<grid, 256, 0, stream>>>(                    \
+          out, q, k_cache, v_cache, block_tables, ctx_lens, row_seq,        \
+          max_blocks, num_kv_heads, scale, q_stride);                       \
+      return;                                                               \
+    }
+  CASE3(1) CASE3(2) CASE3(4)
+  #undef CASE3
   #define CASE(G, DD)                                                       \
     if (group == G && D == DD) {                                            \
       paged_attn_kernel<G, DD><<<grid, 256, 0, stream>>>(                   \
@@ -233,7 +455,7 @@ void launch_paged_attn(bf16* out, const bf16* q, const bf16* k_cache,
           max_blocks, num_kv_heads, scale, q_stride);                       \
       return;                                                               \
     }
-  CASE(1, 128) CASE(2, 128) CASE(4, 128) CASE(8, 128)
+  CASE(8, 128)
   CASE(1, 64) CASE(2, 64) CASE(4, 64) CASE(8, 64)
   #undef CASE
   throw std::runtime_error("paged_attn: unsupported (group, head_dim)");
