@@ -27,6 +27,37 @@ from .scheduler import ScheduledBatch
 
 log = logging.getLogger(__name__)
 
+_TUNABLEOP_LOADED = False
+
+
+def _load_tunableop_results() -> None:
+    """Load committed hipBLASLt algo selections (profiles/tunableop_gfx950.csv).
+
+    The default hipBLASLt heuristic picks ~1.5-3x-off-roofline kernels for
+    the skinny decode GEMMs (M=64..256); offline tuning (tools/tune_gemms.py
+    on an MI355X) selects per-shape algorithms which TunableOp then replays.
+    No tuning happens at serve time — results are read-only.
+    """
+    global _TUNABLEOP_LOADED
+    if _TUNABLEOP_LOADED:
+        return
+    _TUNABLEOP_LOADED = True
+    import os
+    path = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.dirname(os.path.abspath(__file__)))),
+        "profiles", "tunableop_gfx950.csv")
+    if not os.path.exists(path):
+        return
+    try:
+        t = torch.cuda.tunable
+        t.enable(True)
+        t.tuning_enable(False)   # replay only; never tune in the hot path
+        t.record_untuned_enable(False)
+        t.read_file(path)
+        log.info("TunableOp: loaded %s", path)
+    except Exception as e:  # pragma: no cover
+        log.warning("TunableOp load failed: %s", e)
+
 
 class ModelRunner:
     def __init__(self, config: EngineConfig, device: Optional[str] = None):
@@ -38,6 +69,8 @@ class ModelRunner:
         self.dtype = torch.bfloat16 if self.device.type == "cuda" \
             else torch.float32
         torch.manual_seed(config.seed)
+        if self.device.type == "cuda":
+            _load_tunableop_results()
         from hyperspot.parallel.layers import set_init_device
         set_init_device(self.device)
         self.model = build_model(self.spec, dtype=self.dtype).to(self.device)
