@@ -39,7 +39,7 @@ SHAPES_70B_TP8 = dict(hidden=8192, inter=28672 // 8, q=8192 // 8,
 
 
 def gemm_shapes(include_70b=False):
-    ms = [64, 128, 256, 512, 2048, 8192]
+    ms = [64, 128, 256, 384, 512, 768, 1024, 2048, 8192]
     specs = (SHAPES_8B, SHAPES_70B_TP8) if include_70b else (SHAPES_8B,)
     for spec in specs:
         h, it = spec["hidden"], spec["inter"]
@@ -49,7 +49,7 @@ def gemm_shapes(include_70b=False):
             yield m, spec["q"], h       # o proj
             yield m, h, 2 * it          # gate_up
             yield m, it, h              # down
-        for m in [64, 256, 512]:
+        for m in [64, 256, 512, 1024]:
             yield m, h, spec["vocab"]   # logits head
 
 
