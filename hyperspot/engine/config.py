@@ -110,9 +110,11 @@ class EngineConfig:
     tp_size: int = 1
     ep_size: int = 1
     dtype: str = "bfloat16"
+    quant: Optional[str] = None          # None (bf16) | "fp8" (e4m3fn weights)
     seed: int = 0
     # decode hipGraph capture batch buckets (padded up to nearest)
-    graph_batch_sizes: tuple = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256)
+    graph_batch_sizes: tuple = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128,
+                                192, 256, 384, 512, 768, 1024)
 
     def spec(self) -> ModelSpec:
         return get_model_spec(self.model)

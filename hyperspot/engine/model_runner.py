@@ -71,8 +71,9 @@ class ModelRunner:
         torch.manual_seed(config.seed)
         if self.device.type == "cuda":
             _load_tunableop_results()
-        from hyperspot.parallel.layers import set_init_device
+        from hyperspot.parallel.layers import set_init_device, set_quant_mode
         set_init_device(self.device)
+        set_quant_mode(config.quant)
         self.model = build_model(self.spec, dtype=self.dtype).to(self.device)
         self.model.eval()
         self.num_blocks = compute_num_gpu_blocks(self.spec, config, self.device)

@@ -18,6 +18,9 @@ from .state import (get_tp_rank, get_tp_size,
 
 
 _INIT_DEVICE = "cpu"
+_QUANT_MODE = None  # None (bf16) | "fp8"
+
+FP8_MAX = 448.0  # OCP e4m3fn max-normal; gfx950 MFMA/hipBLASLt native fp8
 
 
 def set_init_device(device) -> None:
@@ -25,6 +28,61 @@ def set_init_device(device) -> None:
     the 8B/70B shapes; no checkpoints exist in this offline environment)."""
     global _INIT_DEVICE
     _INIT_DEVICE = device
+
+
+def set_quant_mode(mode) -> None:
+    """Weight format for subsequently-constructed linear layers.
+
+    "fp8": per-output-channel float8_e4m3fn weights + fp32 scales; GEMMs run
+    torch._scaled_mm (hipBLASLt fp8 MFMA on gfx950) with per-token dynamic
+    activation scales.  BASELINE config 5 (fp8 checkpoints, halved weight
+    stream).  None: plain bf16.
+    """
+    global _QUANT_MODE
+    assert mode in (None, "fp8"), mode
+    _QUANT_MODE = mode
+
+
+def quantize_weight_fp8(w: torch.Tensor):
+    """[N,K] → (float8_e4m3fn [N,K], fp32 scale [N]) per-output-channel."""
+    wf = w.float()
+    s = (wf.abs().amax(dim=1) / FP8_MAX).clamp_min(1e-8)
+    q = (wf / s[:, None]).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
+    return q, s
+
+
+def quant_fp8_rowwise(x: torch.Tensor):
+    """Per-token dynamic activation quant: [T,K] → (fp8 [T,K], fp32 [T])."""
+    xf = x.float()
+    s = (xf.abs().amax(dim=-1) / FP8_MAX).clamp_min(1e-8)
+    q = (xf / s[:, None]).clamp(-FP8_MAX, FP8_MAX).to(torch.float8_e4m3fn)
+    return q, s
+
+
+class _LinearCompute(nn.Module):
+    """Shared weight-storage / GEMM path for the TP linear layers."""
+
+    def _finalize_weight(self, w: torch.Tensor) -> None:
+        self.quant = _QUANT_MODE
+        if self.quant == "fp8":
+            q, s = quantize_weight_fp8(w)
+            self.weight = nn.Parameter(q, requires_grad=False)
+            self.weight_scale = nn.Parameter(s, requires_grad=False)
+        else:
+            self.weight = nn.Parameter(w, requires_grad=False)
+
+    def _mm(self, x: torch.Tensor) -> torch.Tensor:
+        if self.quant == "fp8":
+            xq, xs = quant_fp8_rowwise(x)
+            if x.is_cuda:
+                return torch._scaled_mm(
+                    xq, self.weight.t(), scale_a=xs[:, None].contiguous(),
+                    scale_b=self.weight_scale[None, :].contiguous(),
+                    out_dtype=x.dtype)
+            # CPU test path: identical numerics model, emulated in fp32
+            wf = self.weight.float() * self.weight_scale[:, None]
+            return ((xq.float() * xs[:, None]) @ wf.t()).to(x.dtype)
+        return x @ self.weight.t()
 
 
 def _init_weight(out_f: int, in_f: int, dtype: torch.dtype, seed_tag: int):
@@ -38,7 +96,7 @@ def _init_weight(out_f: int, in_f: int, dtype: torch.dtype, seed_tag: int):
     return w.to(dtype)
 
 
-class ColumnParallelLinear(nn.Module):
+class ColumnParallelLinear(_LinearCompute):
     def __init__(self, in_features: int, out_features: int,
                  dtype: torch.dtype, seed_tag: int = 0):
         super().__init__()
@@ -47,10 +105,10 @@ class ColumnParallelLinear(nn.Module):
         self.in_features, self.out_features = in_features, out_features
         shard = out_features // tp
         full = _init_weight(out_features, in_features, dtype, seed_tag)
-        self.weight = nn.Parameter(full[r * shard:(r + 1) * shard], requires_grad=False)
+        self._finalize_weight(full[r * shard:(r + 1) * shard].contiguous())
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return x @ self.weight.t()
+        return self._mm(x)
 
     def load_full_weight(self, w: torch.Tensor) -> None:
         tp, r = get_tp_size(), get_tp_rank()
@@ -58,7 +116,7 @@ class ColumnParallelLinear(nn.Module):
         self.weight.data.copy_(w[r * shard:(r + 1) * shard])
 
 
-class MergedColumnParallelLinear(nn.Module):
+class MergedColumnParallelLinear(_LinearCompute):
     """Several column-parallel projections fused into one GEMM
     (gate_proj | up_proj).  Each sub-projection is sharded independently so
     the shard layout matches per-projection splits downstream."""
@@ -74,13 +132,13 @@ class MergedColumnParallelLinear(nn.Module):
             full = _init_weight(o, in_features, dtype, seed_tag + i)
             sh = o // tp
             parts.append(full[r * sh:(r + 1) * sh])
-        self.weight = nn.Parameter(torch.cat(parts, 0), requires_grad=False)
+        self._finalize_weight(torch.cat(parts, 0))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return x @ self.weight.t()
+        return self._mm(x)
 
 
-class QKVParallelLinear(nn.Module):
+class QKVParallelLinear(_LinearCompute):
     def __init__(self, hidden: int, head_dim: int, num_heads: int,
                  num_kv_heads: int, dtype: torch.dtype, seed_tag: int = 0):
         super().__init__()
@@ -93,14 +151,14 @@ class QKVParallelLinear(nn.Module):
         kf = _init_weight(num_kv_heads * head_dim, hidden, dtype, seed_tag + 1)
         vf = _init_weight(num_kv_heads * head_dim, hidden, dtype, seed_tag + 2)
         qs, ks = self.nh * head_dim, self.nkv * head_dim
-        self.weight = nn.Parameter(torch.cat([
+        self._finalize_weight(torch.cat([
             qf[r * qs:(r + 1) * qs], kf[r * ks:(r + 1) * ks],
-            vf[r * ks:(r + 1) * ks]], 0), requires_grad=False)
+            vf[r * ks:(r + 1) * ks]], 0))
 
     def forward(self, x: torch.Tensor):
         """Returns strided [T, n, D] views into ONE fused qkv buffer — no
         .contiguous() copies; the HIP kernels take the token stride."""
-        qkv = x @ self.weight.t()
+        qkv = self._mm(x)
         qs, ks = self.nh * self.head_dim, self.nkv * self.head_dim
         T = x.shape[0]
         d = self.head_dim
@@ -110,7 +168,7 @@ class QKVParallelLinear(nn.Module):
         return q, k, v
 
 
-class RowParallelLinear(nn.Module):
+class RowParallelLinear(_LinearCompute):
     def __init__(self, in_features: int, out_features: int,
                  dtype: torch.dtype, seed_tag: int = 0):
         super().__init__()
@@ -119,9 +177,8 @@ class RowParallelLinear(nn.Module):
         self.in_features, self.out_features = in_features, out_features
         shard = in_features // tp
         full = _init_weight(out_features, in_features, dtype, seed_tag)
-        self.weight = nn.Parameter(full[:, r * shard:(r + 1) * shard].contiguous(),
-                                   requires_grad=False)
+        self._finalize_weight(full[:, r * shard:(r + 1) * shard].contiguous())
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        out = x @ self.weight.t()
+        out = self._mm(x)
         return tensor_model_parallel_all_reduce(out)

@@ -1,0 +1,110 @@
+"""fp8 (e4m3fn) weight-quant path — BASELINE config 5.
+
+CPU tests emulate _scaled_mm numerics in fp32 (same quant model); the GPU
+test (marked) runs the real hipBLASLt fp8 MFMA path and the engine e2e.
+Reference contract: modules/model-registry PRD `format` field and
+BASELINE.json config 5 (fp8 checkpoints, hot-swap).
+"""
+
+import pytest
+import torch
+
+from hyperspot.parallel import layers as L
+
+
+@pytest.fixture(autouse=True)
+def _reset_quant():
+    yield
+    L.set_quant_mode(None)
+
+
+def test_weight_quant_roundtrip():
+    w = torch.randn(64, 128) * 0.02
+    q, s = L.quantize_weight_fp8(w)
+    assert q.dtype == torch.float8_e4m3fn and s.shape == (64,)
+    back = q.float() * s[:, None]
+    rel = (back - w.float()).abs().max() / w.abs().max()
+    assert rel < 0.05, rel.item()
+
+
+def test_act_quant_rowwise():
+    x = torch.randn(16, 128)
+    q, s = L.quant_fp8_rowwise(x)
+    back = q.float() * s[:, None]
+    assert (back - x).abs().max() < 0.1 * x.abs().max()
+
+
+def test_fp8_linear_matches_bf16_cpu():
+    torch.manual_seed(0)
+    L.set_quant_mode(None)
+    lin = L.ColumnParallelLinear(256, 128, torch.float32, seed_tag=7)
+    L.set_quant_mode("fp8")
+    lin8 = L.ColumnParallelLinear(256, 128, torch.float32, seed_tag=7)
+    assert lin8.quant == "fp8" and lin8.weight.dtype == torch.float8_e4m3fn
+    x = torch.randn(8, 256)
+    y, y8 = lin(x), lin8(x)
+    rel = (y - y8).float().norm() / y.float().norm()
+    assert rel < 0.06, rel.item()   # K=256: fp8 w+act quant noise ~4%
+
+
+def test_fp8_engine_generates_cpu():
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    cfg = EngineConfig(model="tiny-llama", max_num_seqs=2,
+                       max_num_batched_tokens=256, max_model_len=128,
+                       num_gpu_blocks=64, enforce_eager=True, quant="fp8")
+    eng = LLMEngine(cfg, device="cpu")
+    out = eng.generate([[1, 2, 3, 4]],
+                       SamplingParams(temperature=0.0, max_tokens=6))[0]
+    assert len(out) == 6
+
+
+def test_fp8_checkpoint_roundtrip(tmp_path):
+    """fp8 state dict saves/loads; hot-swap copies into resident params."""
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    from hyperspot.engine.checkpoint import (load_checkpoint_into,
+                                             save_checkpoint)
+    cfg = EngineConfig(model="tiny-llama", max_num_seqs=2,
+                       max_num_batched_tokens=256, max_model_len=128,
+                       num_gpu_blocks=64, enforce_eager=True, quant="fp8",
+                       seed=3)
+    eng = LLMEngine(cfg, device="cpu")
+    path = str(tmp_path / "ck.safetensors")
+    save_checkpoint(eng.runner.model, path, meta={"format": "fp8"})
+    before = eng.generate([[5, 6, 7]],
+                          SamplingParams(temperature=0.0, max_tokens=4))[0]
+    # scribble on a weight, swap back, behavior must be restored
+    lin = eng.runner.model.layers[0].mlp.down
+    lin.weight_scale.data.mul_(3.0)
+    load_checkpoint_into(eng.runner.model, path)
+    after = eng.generate([[5, 6, 7]],
+                         SamplingParams(temperature=0.0, max_tokens=4))[0]
+    assert before == after
+
+
+@pytest.mark.gpu
+def test_fp8_scaled_mm_matches_bf16_gpu():
+    dev = "cuda:0"
+    torch.manual_seed(0)
+    L.set_quant_mode(None)
+    L.set_init_device(dev)
+    lin = L.ColumnParallelLinear(4096, 1024, torch.bfloat16, seed_tag=9)
+    L.set_quant_mode("fp8")
+    lin8 = L.ColumnParallelLinear(4096, 1024, torch.bfloat16, seed_tag=9)
+    L.set_init_device("cpu")
+    x = (torch.randn(64, 4096, device=dev) * 0.5).to(torch.bfloat16)
+    y = lin.forward(x)
+    y8 = lin8.forward(x)
+    rel = (y - y8).float().norm() / y.float().norm()
+    assert rel < 0.03, rel.item()
+
+
+@pytest.mark.gpu
+def test_fp8_engine_generates_gpu():
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    cfg = EngineConfig(model="llama3-8b", max_num_seqs=4,
+                       max_num_batched_tokens=2048, max_model_len=512,
+                       num_gpu_blocks=256, enforce_eager=True, quant="fp8")
+    eng = LLMEngine(cfg, device="cuda:0")
+    out = eng.generate([[1, 2, 3, 4, 5, 6, 7, 8]],
+                       SamplingParams(temperature=0.0, max_tokens=8))[0]
+    assert len(out) == 8
