@@ -93,6 +93,35 @@ def bench_gemm(iters):
                                  1)}), flush=True)
 
 
+def bench_fp8_gemm(iters):
+    from hyperspot.parallel.layers import quant_fp8_rowwise, quantize_weight_fp8
+    dev = "cuda:0"
+    for (m, k, n, tag) in [(1024, 4096, 6144, "qkv"),
+                           (1024, 4096, 28672, "gate_up"),
+                           (1024, 14336, 4096, "down"),
+                           (256, 4096, 28672, "gate_up_bs256")]:
+        x = torch.randn(m, k, dtype=torch.bfloat16, device=dev)
+        w = torch.randn(n, k, dtype=torch.bfloat16, device=dev) * 0.02
+        wq, ws = quantize_weight_fp8(w)
+        ws_row = ws[None, :].contiguous()
+        us_bf16 = timed(lambda: torch.nn.functional.linear(x, w), iters)
+
+        def fp8_full():
+            xq, xs = quant_fp8_rowwise(x)
+            return torch._scaled_mm(xq, wq.t(), scale_a=xs[:, None],
+                                    scale_b=ws_row, out_dtype=torch.bfloat16)
+        us_fp8 = timed(fp8_full, iters)
+        xq, xs = quant_fp8_rowwise(x)
+        xs_col = xs[:, None].contiguous()
+        us_fp8_mm = timed(lambda: torch._scaled_mm(
+            xq, wq.t(), scale_a=xs_col, scale_b=ws_row,
+            out_dtype=torch.bfloat16), iters)
+        print(json.dumps({
+            "kernel": f"fp8_{tag}", "shape": f"{m}x{k}x{n}",
+            "us_bf16": round(us_bf16, 1), "us_fp8_withquant": round(us_fp8, 1),
+            "us_fp8_mm_only": round(us_fp8_mm, 1)}), flush=True)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--which", default="attn")
@@ -103,6 +132,8 @@ def main():
         bench_attn(args.iters)
     if args.which in ("gemm", "all"):
         bench_gemm(args.iters)
+    if args.which in ("fp8", "all"):
+        bench_fp8_gemm(args.iters)
 
 
 if __name__ == "__main__":

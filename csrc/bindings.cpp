@@ -26,6 +26,13 @@ void launch_attn_prefill_mfma(bf16*, const bf16*, const bf16*,
                               const bf16*, const int*, int, int, int, int,
                               int, float, long, long, long, hipStream_t);
 void launch_silu_mul(bf16*, const bf16*, long, int, hipStream_t);
+void launch_quant_fp8(unsigned char*, float*, const bf16*, long, int, long,
+                      hipStream_t);
+void launch_fused_add_rmsnorm_fp8(unsigned char*, float*, const bf16*,
+                                  bf16*, const bf16*, float, long, int,
+                                  hipStream_t);
+void launch_silu_mul_fp8(unsigned char*, float*, const bf16*, long, int,
+                         hipStream_t);
 void launch_greedy_sample(long*, const bf16*, long, int, hipStream_t);
 void launch_inv_cdf_sample(long*, const float*, const float*, long, int,
                            hipStream_t);
@@ -165,9 +172,53 @@ void inv_cdf_sample(torch::Tensor out, torch::Tensor logits,
 
 }  // namespace
 
+
+unsigned char* u8(torch::Tensor& t) {
+  return reinterpret_cast<unsigned char*>(t.data_ptr());
+}
+
+void quant_fp8(torch::Tensor out, torch::Tensor scales, torch::Tensor x) {
+  check(out, torch::kFloat8_e4m3fn, "out");
+  check(scales, torch::kFloat, "scales");
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16, "x");
+  const int hidden = (int)x.size(-1);
+  TORCH_CHECK(x.stride(-1) == 1, "x inner dim must be contiguous");
+  const long rows = x.numel() / hidden;
+  const long stride = x.dim() > 1 ? x.stride(0) : hidden;
+  launch_quant_fp8(u8(out), scales.data_ptr<float>(), cbf(x), rows, hidden,
+                   stride, stream());
+}
+
+void fused_add_rmsnorm_fp8(torch::Tensor out, torch::Tensor scales,
+                           torch::Tensor x, torch::Tensor residual,
+                           torch::Tensor w, double eps) {
+  check(out, torch::kFloat8_e4m3fn, "out");
+  check(scales, torch::kFloat, "scales");
+  check(x, torch::kBFloat16, "x");
+  check(residual, torch::kBFloat16, "residual");
+  check(w, torch::kBFloat16, "w");
+  const int hidden = (int)x.size(-1);
+  launch_fused_add_rmsnorm_fp8(u8(out), scales.data_ptr<float>(), cbf(x),
+                               bf(residual), cbf(w), (float)eps,
+                               x.numel() / hidden, hidden, stream());
+}
+
+void silu_mul_fp8(torch::Tensor out, torch::Tensor scales, torch::Tensor x) {
+  check(out, torch::kFloat8_e4m3fn, "out");
+  check(scales, torch::kFloat, "scales");
+  check(x, torch::kBFloat16, "x");
+  const int inter = (int)out.size(-1);
+  TORCH_CHECK(x.size(-1) == 2 * inter, "x must be [rows, 2*inter]");
+  launch_silu_mul_fp8(u8(out), scales.data_ptr<float>(), cbf(x),
+                      out.numel() / inter, inter, stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "HyperSpot-AMD gfx950 (CDNA4) kernels";
   m.def("rmsnorm", &rmsnorm);
+  m.def("quant_fp8", &quant_fp8);
+  m.def("fused_add_rmsnorm_fp8", &fused_add_rmsnorm_fp8);
+  m.def("silu_mul_fp8", &silu_mul_fp8);
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
   m.def("rope_kv_append", &rope_kv_append);
   m.def("paged_attn", &paged_attn);

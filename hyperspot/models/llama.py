@@ -62,15 +62,18 @@ class Attention(nn.Module):
             spec.num_heads * spec.head_dim, spec.hidden_size, dtype,
             seed_tag=layer_idx * 10 + 4)
 
-    def forward(self, x: torch.Tensor, meta: ForwardMeta,
+    def forward(self, x, meta: ForwardMeta,
                 kv_cache: torch.Tensor, cos_sin: torch.Tensor) -> torch.Tensor:
-        T = x.shape[0]
         q, k, v = self.qkv(x)
+        T = q.shape[0]
         k_cache, v_cache = kv_cache[0], kv_cache[1]
         ops.rope_kv_append(q, k, v, meta.positions, cos_sin,
                            meta.slot_mapping, k_cache, v_cache)
         o = ops.attention(q, k, v, k_cache, v_cache, meta, self.scale)
-        return self.o_proj(o.view(T, -1))
+        o = o.view(T, -1)
+        if self.o_proj.quant == "fp8" and o.is_cuda:
+            o = ops.quant_fp8(o)   # fused amax+scale+pack (csrc/quant.hip)
+        return self.o_proj(o)
 
 
 class LlamaMLP(nn.Module):
@@ -85,8 +88,10 @@ class LlamaMLP(nn.Module):
         self.tp = get_tp_size()
         self.inter_shard = spec.intermediate_size // self.tp
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x) -> torch.Tensor:
         gu = self.gate_up(x)
+        if self.down.quant == "fp8" and gu.is_cuda:
+            return self.down(ops.silu_mul_q(gu))
         return self.down(ops.silu_mul(gu))
 
 
@@ -105,17 +110,29 @@ class DecoderLayer(nn.Module):
     def forward(self, x: torch.Tensor, residual: Optional[torch.Tensor],
                 meta: ForwardMeta, kv_cache: torch.Tensor,
                 cos_sin: torch.Tensor):
+        # fp8 GPU path: the norm kernels emit per-token-scaled fp8 directly
+        # (csrc/quant.hip), so GEMM inputs never take an extra quant pass.
+        fp8 = self.attn.qkv.quant == "fp8" and x.is_cuda
         if residual is None:
             # x is the fresh embedding-gather output: safe to alias as the
             # residual (fused_add_rmsnorm mutates it in place from layer 1 on)
             residual = x
             h = ops.rmsnorm(x, self.input_norm_w, self.eps)
+            if fp8:
+                h = ops.quant_fp8(h)
+        elif fp8:
+            h, residual = ops.fused_add_rmsnorm_q(x, residual,
+                                                  self.input_norm_w, self.eps)
         else:
             h, residual = ops.fused_add_rmsnorm(x, residual,
                                                 self.input_norm_w, self.eps)
         h = self.attn(h, meta, kv_cache, cos_sin)
-        h, residual = ops.fused_add_rmsnorm(h, residual, self.post_norm_w,
-                                            self.eps)
+        if fp8:
+            h, residual = ops.fused_add_rmsnorm_q(h, residual,
+                                                  self.post_norm_w, self.eps)
+        else:
+            h, residual = ops.fused_add_rmsnorm(h, residual, self.post_norm_w,
+                                                self.eps)
         h = self.mlp(h)
         return h, residual
 

@@ -9,6 +9,8 @@ device without the extension is a hard error, never a silent eager fallback.
 
 from __future__ import annotations
 
+from typing import NamedTuple
+
 import torch
 
 from . import torch_ref
@@ -125,6 +127,51 @@ def silu_mul(gate_up: torch.Tensor) -> torch.Tensor:
         _native().silu_mul(out, gate_up)
         return out
     return torch_ref.silu_mul(gate_up)
+
+
+class QTensor(NamedTuple):
+    """Per-token-scaled fp8 activation: data e4m3fn [T,K], scale fp32 [T].
+
+    Produced on GPU by the fused quant kernels (csrc/quant.hip) so fp8
+    GEMM inputs never take an extra elementwise pass; consumed by the TP
+    linear layers via torch._scaled_mm (hipBLASLt fp8 MFMA)."""
+
+    data: torch.Tensor
+    scale: torch.Tensor
+
+
+def quant_fp8(x: torch.Tensor) -> QTensor:
+    """[T,K] bf16 -> QTensor (GPU; row-per-workgroup amax+scale+pack)."""
+    assert x.is_cuda
+    T, K = x.shape
+    out = torch.empty(T, K, dtype=torch.float8_e4m3fn, device=x.device)
+    scales = torch.empty(T, dtype=torch.float32, device=x.device)
+    _native().quant_fp8(out, scales, x)
+    return QTensor(out, scales)
+
+
+def fused_add_rmsnorm_q(x: torch.Tensor, residual: torch.Tensor,
+                        weight: torch.Tensor, eps: float):
+    """residual += x (in place); returns (QTensor of rmsnorm(residual)*w,
+    residual).  GPU-only fused producer for the fp8 GEMM path."""
+    assert x.is_cuda
+    T, K = x.shape
+    out = torch.empty(T, K, dtype=torch.float8_e4m3fn, device=x.device)
+    scales = torch.empty(T, dtype=torch.float32, device=x.device)
+    _native().fused_add_rmsnorm_fp8(out, scales, x, residual, weight, eps)
+    return QTensor(out, scales), residual
+
+
+def silu_mul_q(gate_up: torch.Tensor) -> QTensor:
+    """silu(g)*u -> fp8 QTensor (GPU fused producer)."""
+    assert gate_up.is_cuda
+    i = gate_up.shape[-1] // 2
+    out = torch.empty(*gate_up.shape[:-1], i, dtype=torch.float8_e4m3fn,
+                      device=gate_up.device)
+    scales = torch.empty(gate_up.shape[0], dtype=torch.float32,
+                         device=gate_up.device)
+    _native().silu_mul_fp8(out, scales, gate_up)
+    return QTensor(out, scales)
 
 
 def greedy_sample(logits: torch.Tensor) -> torch.Tensor:

@@ -71,13 +71,20 @@ class _LinearCompute(nn.Module):
         else:
             self.weight = nn.Parameter(w, requires_grad=False)
 
-    def _mm(self, x: torch.Tensor) -> torch.Tensor:
+    def _mm(self, x) -> torch.Tensor:
+        from hyperspot.ops import QTensor
+        if isinstance(x, QTensor):
+            # pre-quantized by a fused producer kernel (csrc/quant.hip)
+            return torch._scaled_mm(
+                x.data, self.weight.t(), scale_a=x.scale.unsqueeze(1),
+                scale_b=self.weight_scale.unsqueeze(0),
+                out_dtype=torch.bfloat16)
         if self.quant == "fp8":
             xq, xs = quant_fp8_rowwise(x)
             if x.is_cuda:
                 return torch._scaled_mm(
                     xq, self.weight.t(), scale_a=xs[:, None].contiguous(),
-                    scale_b=self.weight_scale[None, :].contiguous(),
+                    scale_b=self.weight_scale.unsqueeze(0),
                     out_dtype=x.dtype)
             # CPU test path: identical numerics model, emulated in fp32
             wf = self.weight.float() * self.weight_scale[:, None]
@@ -160,7 +167,7 @@ class QKVParallelLinear(_LinearCompute):
         .contiguous() copies; the HIP kernels take the token stride."""
         qkv = self._mm(x)
         qs, ks = self.nh * self.head_dim, self.nkv * self.head_dim
-        T = x.shape[0]
+        T = qkv.shape[0]
         d = self.head_dim
         q = qkv[:, :qs].view(T, self.nh, d)
         k = qkv[:, qs:qs + ks].view(T, self.nkv, d)
