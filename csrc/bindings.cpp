@@ -33,6 +33,12 @@ void launch_fused_add_rmsnorm_fp8(unsigned char*, float*, const bf16*,
                                   hipStream_t);
 void launch_silu_mul_fp8(unsigned char*, float*, const bf16*, long, int,
                          hipStream_t);
+void launch_moe_align(int*, int*, int*, const int*, int, int, int,
+                      hipStream_t);
+void launch_moe_gemm(bf16*, const bf16*, const bf16*, const int*,
+                     const int*, int, int, int, int, hipStream_t);
+void launch_moe_combine(bf16*, const bf16*, const float*, const int*, long,
+                        int, int, hipStream_t);
 void launch_greedy_sample(long*, const bf16*, long, int, hipStream_t);
 void launch_inv_cdf_sample(long*, const float*, const float*, long, int,
                            hipStream_t);
@@ -213,12 +219,55 @@ void silu_mul_fp8(torch::Tensor out, torch::Tensor scales, torch::Tensor x) {
                       out.numel() / inter, inter, stream());
 }
 
+
+void moe_align(torch::Tensor sorted_ids, torch::Tensor tile_expert,
+               torch::Tensor inv_pos, torch::Tensor flat_ids,
+               long num_experts) {
+  check(sorted_ids, torch::kInt, "sorted_ids");
+  check(tile_expert, torch::kInt, "tile_expert");
+  check(inv_pos, torch::kInt, "inv_pos");
+  check(flat_ids, torch::kInt, "flat_ids");
+  launch_moe_align(sorted_ids.data_ptr<int>(), tile_expert.data_ptr<int>(),
+                   inv_pos.data_ptr<int>(), flat_ids.data_ptr<int>(),
+                   (int)flat_ids.numel(), (int)num_experts,
+                   (int)tile_expert.numel(), stream());
+}
+
+void moe_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+              torch::Tensor sorted_ids, torch::Tensor tile_expert,
+              long gather_div) {
+  check(out, torch::kBFloat16, "out");
+  check(x, torch::kBFloat16, "x");
+  check(w, torch::kBFloat16, "w");
+  check(sorted_ids, torch::kInt, "sorted_ids");
+  check(tile_expert, torch::kInt, "tile_expert");
+  TORCH_CHECK(w.dim() == 3, "w must be [E, N, K]");
+  launch_moe_gemm(bf(out), cbf(x), cbf(w), sorted_ids.data_ptr<int>(),
+                  tile_expert.data_ptr<int>(), (int)tile_expert.numel(),
+                  (int)w.size(1), (int)w.size(2), (int)gather_div,
+                  stream());
+}
+
+void moe_combine(torch::Tensor out, torch::Tensor y, torch::Tensor wts,
+                 torch::Tensor inv_pos, long topk) {
+  check(out, torch::kBFloat16, "out");
+  check(y, torch::kBFloat16, "y");
+  check(wts, torch::kFloat, "wts");
+  check(inv_pos, torch::kInt, "inv_pos");
+  launch_moe_combine(bf(out), cbf(y), wts.data_ptr<float>(),
+                     inv_pos.data_ptr<int>(), out.size(0), (int)topk,
+                     (int)out.size(-1), stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "HyperSpot-AMD gfx950 (CDNA4) kernels";
   m.def("rmsnorm", &rmsnorm);
   m.def("quant_fp8", &quant_fp8);
   m.def("fused_add_rmsnorm_fp8", &fused_add_rmsnorm_fp8);
   m.def("silu_mul_fp8", &silu_mul_fp8);
+  m.def("moe_align", &moe_align);
+  m.def("moe_gemm", &moe_gemm);
+  m.def("moe_combine", &moe_combine);
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
   m.def("rope_kv_append", &rope_kv_append);
   m.def("paged_attn", &paged_attn);

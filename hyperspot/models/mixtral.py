@@ -65,6 +65,10 @@ class MixtralMoE(nn.Module):
         return w / w.sum(-1, keepdim=True), ids
 
     def _local_moe(self, x, weights, ids):
+        if x.is_cuda and ops.have_native():
+            # grouped MFMA kernels, capture-safe (csrc/moe.hip)
+            local = (ids - self.local_expert_start).to(torch.int32)
+            return ops.moe_ffn(x, self.w13, self.w2, weights, local)
         out = torch.zeros_like(x)
         flat_ids = ids.reshape(-1)
         flat_w = weights.reshape(-1)
@@ -106,12 +110,18 @@ class MixtralMoE(nn.Module):
         dist.all_to_all_single(recv_x, send_x, rc, sc, group=group)
         dist.all_to_all_single(recv_eids, send_eids, rc, sc, group=group)
         # local grouped FFN
-        y = torch.empty_like(recv_x)
         local = recv_eids - self.local_expert_start
-        for e in range(self.experts_per_rank):
-            m = local == e
-            if bool(m.any()):
-                y[m] = self._expert_ffn(recv_x[m], e)
+        if recv_x.is_cuda and ops.have_native():
+            ones = torch.ones(recv_x.shape[0], 1, device=x.device,
+                              dtype=torch.float32)
+            y = ops.moe_ffn(recv_x, self.w13, self.w2, ones,
+                            local.view(-1, 1).to(torch.int32))
+        else:
+            y = torch.empty_like(recv_x)
+            for e in range(self.experts_per_rank):
+                m = local == e
+                if bool(m.any()):
+                    y[m] = self._expert_ffn(recv_x[m], e)
         # combine back
         back = torch.empty(sum(sc), H, dtype=x.dtype, device=x.device)
         dist.all_to_all_single(back, y, sc, rc, group=group)

@@ -174,6 +174,43 @@ def silu_mul_q(gate_up: torch.Tensor) -> QTensor:
     return QTensor(out, scales)
 
 
+MOE_BM = 32  # sorted-pair tile granularity of the MoE kernels (csrc/moe.hip)
+
+
+def moe_ffn(x: torch.Tensor, w13: torch.Tensor, w2: torch.Tensor,
+            weights: torch.Tensor, ids: torch.Tensor) -> torch.Tensor:
+    """Grouped expert SwiGLU FFN (GPU, hipGraph-capture-safe).
+
+    x [T,H]; w13 [E_local, 2I, H]; w2 [E_local, H, I]; weights [T,K] fp32/bf16
+    routing weights; ids [T,K] LOCAL expert indices.  Token->expert sorting,
+    the two grouped MFMA GEMMs and the weighted combine all run on-device
+    with grids that depend only on (T, K, E) — no host sync, no dynamic
+    shapes (MoE dispatch/combine + grouped GEMM).
+    """
+    T, H = x.shape
+    K = ids.shape[1]
+    E = w13.shape[0]
+    TK = T * K
+    ntiles = (TK + E * (MOE_BM - 1) + MOE_BM - 1) // MOE_BM
+    P = ntiles * MOE_BM
+    dev = x.device
+    flat = ids.reshape(-1).to(torch.int32)
+    sorted_ids = torch.empty(P, dtype=torch.int32, device=dev)
+    tile_expert = torch.empty(ntiles, dtype=torch.int32, device=dev)
+    inv_pos = torch.empty(TK, dtype=torch.int32, device=dev)
+    n = _native()
+    n.moe_align(sorted_ids, tile_expert, inv_pos, flat, E)
+    I2 = w13.shape[1]
+    h1 = torch.empty(P, I2, dtype=torch.bfloat16, device=dev)
+    n.moe_gemm(h1, x, w13, sorted_ids, tile_expert, K)   # gather pair//K
+    a = silu_mul(h1)
+    y = torch.empty(P, H, dtype=torch.bfloat16, device=dev)
+    n.moe_gemm(y, a, w2, sorted_ids, tile_expert, 0)     # identity rows
+    out = torch.empty(T, H, dtype=torch.bfloat16, device=dev)
+    n.moe_combine(out, y, weights.float().contiguous(), inv_pos, K)
+    return out
+
+
 def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
     if logits.is_cuda:
         out = torch.empty(logits.shape[0], dtype=torch.long,

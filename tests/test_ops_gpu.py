@@ -225,3 +225,33 @@ def test_sample_dispatch_topk_topp_gpu():
                        torch.rand(B, device=DEV))
         for b in range(B):
             assert t[b].cpu() in topk[b]
+
+
+@pytest.mark.gpu
+def test_moe_grouped_ffn_matches_ref():
+    """Grouped MFMA expert GEMMs + combine vs the per-expert loop."""
+    import hyperspot.ops as ops
+    torch.manual_seed(0)
+    dev = "cuda:0"
+    T, H, I, E, K = 96, 256, 256, 4, 2
+    x = (torch.randn(T, H, device=dev) * 0.3).to(torch.bfloat16)
+    w13 = (torch.randn(E, 2 * I, H, device=dev) * 0.05).to(torch.bfloat16)
+    w2 = (torch.randn(E, H, I, device=dev) * 0.05).to(torch.bfloat16)
+    logits = torch.randn(T, E, device=dev)
+    w, ids = logits.softmax(-1).topk(K, dim=-1)
+    w = (w / w.sum(-1, keepdim=True)).float()
+
+    out = ops.moe_ffn(x, w13, w2, w, ids.to(torch.int32))
+
+    ref = torch.zeros(T, H, device=dev, dtype=torch.float32)
+    xf = x.float()
+    for t in range(T):
+        for k in range(K):
+            e = int(ids[t, k])
+            gu = xf[t] @ w13[e].float().t()
+            g, u = gu[:I], gu[I:]
+            y = (g * torch.sigmoid(g) * u) @ w2[e].float().t()
+            ref[t] += w[t, k] * y
+    err = (out.float() - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err < 0.02 * scale + 0.02, (err, scale)
