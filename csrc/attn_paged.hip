@@ -242,17 +242,24 @@ DEV float dot2_bf16(unsigned a, unsigned b, float c) {
   return c;
 }
 
+// SPLIT > 1 (flash-decode): grid.z WGs share one (row, kvh), each owning
+// pages [4*z+wid :: 4*SPLIT]; per-WG partials (m, l, o) land in `partial`
+// [rows, kvh, SPLIT, GROUP, D+2] fp32 and a second kernel merges them.
+// Used when rows*kvh alone cannot fill the 256 CUs (small batch, or the
+// kvh=1 GQA shard of Llama-70B at TP=8).
 template <int GROUP>
 __global__ __launch_bounds__(256) void paged_attn_v3_kernel(
     bf16* __restrict__ out, const bf16* __restrict__ q,
     const bf16* __restrict__ k_cache, const bf16* __restrict__ v_cache,
     const int* __restrict__ block_tables, const int* __restrict__ ctx_lens,
     const int* __restrict__ row_seq, int max_blocks, int num_kv_heads,
-    float scale, long q_stride) {
+    float scale, long q_stride, int h_total, int g_stride, int h_off,
+    float* __restrict__ partial) {
   constexpr int D = 128;
   const int row = blockIdx.x;
   const int kvh = blockIdx.y;
-  const int H = num_kv_heads * GROUP;
+  const int split = gridDim.z;
+  const int H = h_total;
   const int ctx = ctx_lens[row];
   const int table = row_seq ? row_seq[row] : row;
   const int* bt = block_tables + (long)table * max_blocks;
@@ -277,8 +284,8 @@ __global__ __launch_bounds__(256) void paged_attn_v3_kernel(
   // ---- query group: raw packed-bf16 slice, GROUP*8 dwords ----
   unsigned q_pk[GROUP][8];
   {
-    const bf16* qp = q + (long)row * q_stride + (long)kvh * GROUP * D
-                     + sl * 16;
+    const bf16* qp = q + (long)row * q_stride
+                     + ((long)kvh * g_stride + h_off) * D + sl * 16;
     #pragma unroll
     for (int h = 0; h < GROUP; ++h) {
       uint4 a = *reinterpret_cast<const uint4*>(qp + h * D);
@@ -297,7 +304,7 @@ __global__ __launch_bounds__(256) void paged_attn_v3_kernel(
     for (int j = 0; j < 4; ++j) acc[h][j] = make_float2(0.f, 0.f);
   }
 
-  for (int b = wid; b < nblocks; b += 4) {
+  for (int b = 4 * blockIdx.z + wid; b < nblocks; b += 4 * split) {
     const long blk = bt[b];
     const int nb = min(ATTN_BS, ctx - b * ATTN_BS);
     const bf16* pbase = k_cache + ((blk * num_kv_heads + kvh) * ATTN_BS) * D;
@@ -425,7 +432,46 @@ __global__ __launch_bounds__(256) void paged_attn_v3_kernel(
       o += f * comb_o[w][h][d];
       L += f * comb_l[w][h];
     }
-    *(unsigned short*)(out + ((long)row * H + kvh * GROUP + h) * D + d) =
+    if (split == 1) {
+      *(unsigned short*)(out +
+          ((long)row * H + kvh * g_stride + h + h_off) * D + d) =
+          f2bf(o / L);
+    } else {
+      // partial layout: [row][kvh][z][h][D+2] fp32 (o | m | l)
+      float* pr = partial +
+          ((((long)row * gridDim.y + kvh) * split + blockIdx.z) * GROUP + h)
+          * (D + 2);
+      pr[d] = o;
+      if (d == 0) { pr[D] = M; pr[D + 1] = L; }
+    }
+  }
+}
+
+// merge the z-split partials: grid (rows, kvh); 256 threads cover GROUP*D
+template <int GROUP>
+__global__ __launch_bounds__(256) void paged_attn_v3_merge_kernel(
+    bf16* __restrict__ out, const float* __restrict__ partial, int split,
+    int h_total, int g_stride, int h_off) {
+  constexpr int D = 128;
+  const int row = blockIdx.x;
+  const int kvh = blockIdx.y;
+  for (int i = threadIdx.x; i < GROUP * D; i += 256) {
+    const int h = i / D, d = i % D;
+    const float* base = partial +
+        (((long)row * gridDim.y + kvh) * split * GROUP + h) * (D + 2);
+    float M = -FLT_MAX;
+    for (int z = 0; z < split; ++z)
+      M = fmaxf(M, base[(long)z * GROUP * (D + 2) + D]);
+    float o = 0.f, L = 0.f;
+    for (int z = 0; z < split; ++z) {
+      const float* pz = base + (long)z * GROUP * (D + 2);
+      const float mz = pz[D];
+      const float f = (mz == -FLT_MAX) ? 0.f : __expf(mz - M);
+      o += f * pz[d];
+      L += f * pz[D + 1];
+    }
+    *(unsigned short*)(out +
+        ((long)row * h_total + kvh * g_stride + h + h_off) * D + d) =
         f2bf(o / L);
   }
 }
@@ -435,19 +481,41 @@ void launch_paged_attn(bf16* out, const bf16* q, const bf16* k_cache,
                        const int* ctx_lens, const int* row_seq,
                        long num_rows, int num_kv_heads, int group, int D,
                        int max_blocks, int block_size, float scale,
-                       long q_stride, hipStream_t stream) {
+                       long q_stride, float* split_ws, int split,
+                       hipStream_t stream) {
   if (block_size != ATTN_BS)
     throw std::runtime_error("paged_attn: block_size must be 16");
-  dim3 grid((unsigned)num_rows, (unsigned)num_kv_heads);
-  #define CASE3(G)                                                          \
-    if (group == G && D == 128) {                                           \
-      paged_attn_v3_kernel<G><<<grid, 256, 0, stream>>>(                    \
-          out, q, k_cache, v_cache, block_tables, ctx_lens, row_seq,        \
-          max_blocks, num_kv_heads, scale, q_stride);                       \
-      return;                                                               \
+  // v3 (D=128): GROUP<=4 direct; GROUP=8 as two GROUP=4 half-calls over
+  // the same KV stream (the kvh=1 70B-TP8 shard: 2x a small KV re-read
+  // buys the v3 structure + occupancy).  split>1 = flash-decode page
+  // split with an fp32 partial workspace + merge kernel.
+  if (D == 128 && (group <= 4 || group == 8)) {
+    const int calls = (group == 8) ? 2 : 1;
+    const int g = (group == 8) ? 4 : group;
+    const int h_total = num_kv_heads * group;
+    for (int c = 0; c < calls; ++c) {
+      const int h_off = c * 4;
+      dim3 grid((unsigned)num_rows, (unsigned)num_kv_heads,
+                (unsigned)split);
+      dim3 mgrid((unsigned)num_rows, (unsigned)num_kv_heads);
+      #define CASE3(G)                                                      \
+        if (g == G) {                                                       \
+          paged_attn_v3_kernel<G><<<grid, 256, 0, stream>>>(                \
+              out, q, k_cache, v_cache, block_tables, ctx_lens, row_seq,    \
+              max_blocks, num_kv_heads, scale, q_stride, h_total, group,    \
+              h_off, split_ws);                                             \
+          if (split > 1)                                                    \
+            paged_attn_v3_merge_kernel<G><<<mgrid, 256, 0, stream>>>(       \
+                out, split_ws, split, h_total, group, h_off);               \
+        }
+      CASE3(1) CASE3(2) CASE3(4)
+      #undef CASE3
     }
-  CASE3(1) CASE3(2) CASE3(4)
-  #undef CASE3
+    return;
+  }
+  if (split != 1)
+    throw std::runtime_error("paged_attn: split needs the v3 path");
+  dim3 grid((unsigned)num_rows, (unsigned)num_kv_heads);
   #define CASE(G, DD)                                                       \
     if (group == G && D == DD) {                                            \
       paged_attn_kernel<G, DD><<<grid, 256, 0, stream>>>(                   \
@@ -455,7 +523,6 @@ void launch_paged_attn(bf16* out, const bf16* q, const bf16* k_cache,
           max_blocks, num_kv_heads, scale, q_stride);                       \
       return;                                                               \
     }
-  CASE(8, 128)
   CASE(1, 64) CASE(2, 64) CASE(4, 64) CASE(8, 64)
   #undef CASE
   throw std::runtime_error("paged_attn: unsupported (group, head_dim)");

@@ -21,7 +21,8 @@ void launch_rope_kv_append(bf16*, bf16*, const bf16*, const long*,
                            hipStream_t);
 void launch_paged_attn(bf16*, const bf16*, const bf16*, const bf16*,
                        const int*, const int*, const int*, long, int, int,
-                       int, int, int, float, long, hipStream_t);
+                       int, int, int, float, long, float*, int,
+                       hipStream_t);
 void launch_attn_prefill_mfma(bf16*, const bf16*, const bf16*,
                               const bf16*, const int*, int, int, int, int,
                               int, float, long, long, long, hipStream_t);
@@ -115,7 +116,8 @@ void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
 void paged_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                 torch::Tensor v_cache, torch::Tensor block_tables,
                 torch::Tensor ctx_lens,
-                c10::optional<torch::Tensor> row_seq, double scale) {
+                c10::optional<torch::Tensor> row_seq, double scale,
+                c10::optional<torch::Tensor> split_ws, long split) {
   check(out, torch::kBFloat16, "out");
   check_qkv_view(q, "q");
   check(block_tables, torch::kInt32, "block_tables");
@@ -129,10 +131,18 @@ void paged_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
     check(*row_seq, torch::kInt32, "row_seq");
     rs = row_seq->data_ptr<int>();
   }
+  float* ws = nullptr;
+  if (split > 1) {
+    TORCH_CHECK(split_ws.has_value(), "split>1 needs a workspace");
+    check(*split_ws, torch::kFloat, "split_ws");
+    TORCH_CHECK(split_ws->numel() >= R * KV * split * (H / KV) * (D + 2),
+                "split_ws too small");
+    ws = split_ws->data_ptr<float>();
+  }
   launch_paged_attn(bf(out), cbf(q), cbf(k_cache), cbf(v_cache),
                     block_tables.data_ptr<int>(), ctx_lens.data_ptr<int>(),
                     rs, R, KV, H / KV, D, (int)block_tables.size(1), BS,
-                    (float)scale, q.stride(0), stream());
+                    (float)scale, q.stride(0), ws, (int)split, stream());
 }
 
 void attn_prefill_mfma(torch::Tensor out, torch::Tensor q, torch::Tensor k,

@@ -79,13 +79,37 @@ def rope_kv_append(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                                   slot_mapping[mask])
 
 
+import os
+
+_ATTN_SPLIT_TARGET = int(os.environ.get("HYPERSPOT_ATTN_SPLIT_TARGET", "256"))
+
+
+def _attn_split(rows: int, kvh: int) -> int:
+    """Flash-decode page-split factor: bring the workgroup count up to
+    ~_ATTN_SPLIT_TARGET when rows*kvh alone cannot fill the 256 CUs (small
+    batch, or the kvh=1 GQA shard of 70B at TP=8).  Measured (MI355X, profiles/
+    r01_decode_v3.md): splitting at >=256 WGs already loses — the fp32
+    partial round trip + merge launch cost more than the occupancy buys —
+    so only genuinely tiny grids (rows*kvh < 256, e.g. batch<32 at TP=8)
+    split.  Static in (rows, kvh) => capture-safe."""
+    s = max(1, min(16, _ATTN_SPLIT_TARGET // max(1, rows * kvh)))
+    return 1 if s <= 1 else s
+
+
 def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
                       v_cache: torch.Tensor, block_tables: torch.Tensor,
                       seq_lens: torch.Tensor, scale: float) -> torch.Tensor:
     if q.is_cuda:
         out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+        kvh = k_cache.shape[1]
+        split = _attn_split(q.shape[0], kvh) if q.shape[-1] == 128 else 1
+        ws = None
+        if split > 1:
+            ws = torch.empty(q.shape[0] * kvh * split
+                             * (q.shape[1] // kvh) * 130,
+                             dtype=torch.float32, device=q.device)
         _native().paged_attn(out, q, k_cache, v_cache, block_tables,
-                             seq_lens, None, scale)
+                             seq_lens, None, scale, ws, split)
         return out
     return torch_ref.paged_attn_decode(q, k_cache, v_cache, block_tables,
                                        seq_lens, scale)
@@ -114,7 +138,7 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
             # fallback: per-row attention over the freshly appended cache
             _native().paged_attn(out, q, k_cache, v_cache,
                                  meta.block_tables, meta.ctx_lens,
-                                 meta.row_seq, scale)
+                                 meta.row_seq, scale, None, 1)
         return out
     return torch_ref.prefill_attn(q, k, v, meta.seq_start, scale)
 

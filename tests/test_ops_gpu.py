@@ -130,7 +130,7 @@ def test_paged_attn_prefill_matches_varlen_ref():
     _C.paged_attn(out, q, k_cache, v_cache, bt,
                   torch.tensor(ctx, dtype=torch.int32, device=DEV),
                   torch.tensor(row_seq, dtype=torch.int32, device=DEV),
-                  D ** -0.5)
+                  D ** -0.5, None, 1)
     ss = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
                       dtype=torch.int32)
     ref = R.prefill_attn(q.float().cpu(), k.float().cpu(), v.float().cpu(),
