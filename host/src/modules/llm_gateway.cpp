@@ -7,12 +7,19 @@
 #include <sys/wait.h>
 #include <unistd.h>
 
+#include <chrono>
 #include <cstring>
 
 #include "../util/log.h"
 #include "system_modules.h"
 
 namespace hs {
+
+static double now_s() {
+  return std::chrono::duration<double>(
+             std::chrono::system_clock::now().time_since_epoch())
+      .count();
+}
 
 // ------------------------------------------------------ serverless-runtime
 
@@ -58,6 +65,25 @@ class StaticAdmission : public AdmissionClient {
   int max_concurrent_;
   double rps_, burst_;
 };
+
+// base64 for embedding_response encoding_format=base64 (little-endian f32,
+// the OpenAI-compatible convention the schema references)
+std::string b64_encode(const unsigned char* data, size_t n) {
+  static const char* tbl =
+      "ABCDEFGHIJKLMNOPQRSTUVWXYZabcdefghijklmnopqrstuvwxyz0123456789+/";
+  std::string out;
+  out.reserve((n + 2) / 3 * 4);
+  for (size_t i = 0; i < n; i += 3) {
+    unsigned v = data[i] << 16;
+    if (i + 1 < n) v |= data[i + 1] << 8;
+    if (i + 2 < n) v |= data[i + 2];
+    out.push_back(tbl[(v >> 18) & 63]);
+    out.push_back(tbl[(v >> 12) & 63]);
+    out.push_back(i + 1 < n ? tbl[(v >> 6) & 63] : '=');
+    out.push_back(i + 2 < n ? tbl[v & 63] : '=');
+  }
+  return out;
+}
 
 }  // namespace
 
@@ -126,6 +152,8 @@ void LlmGatewayModule::init(ModuleCtx& ctx) {
   auto_start_ = ctx.config.at("auto_start_worker").as_bool(true);
   python_ = ctx.config.at("python").as_string("python3");
   worker_cfg_ = ctx.config.at("worker");
+  budget_tokens_ =
+      (uint64_t)ctx.config.path("usage.budget_tokens_per_tenant").as_int(0);
 }
 
 bool LlmGatewayModule::worker_ready() {
@@ -171,6 +199,10 @@ void LlmGatewayModule::spawn_worker() {
         args.push_back("--tp");
         args.push_back(std::to_string(worker_cfg_.at("tp").as_int(1)));
       }
+      if (worker_cfg_.contains("quant")) {
+        args.push_back("--quant");
+        args.push_back(worker_cfg_.at("quant").as_string());
+      }
     }
     std::vector<char*> argv;
     for (auto& a : args) argv.push_back(const_cast<char*>(a.c_str()));
@@ -188,9 +220,17 @@ void LlmGatewayModule::start(ModuleCtx& ctx) {
     unlink(socket_path_.c_str());
     spawn_worker();
   }
+  stopping_ = false;
+  job_thread_ = std::thread([this] { job_loop(); });
 }
 
 void LlmGatewayModule::stop(ModuleCtx& ctx) {
+  {
+    std::lock_guard<std::mutex> lk(jobs_mu_);
+    stopping_ = true;
+  }
+  jobs_cv_.notify_all();
+  if (job_thread_.joinable()) job_thread_.join();
   if (worker_pid_ > 0) {
     kill(worker_pid_, SIGTERM);
     int st = 0;
@@ -206,6 +246,97 @@ void LlmGatewayModule::stop(ModuleCtx& ctx) {
       waitpid(worker_pid_, &st, 0);
     }
   }
+}
+
+// ------------------------------------------------- shared pipeline pieces
+
+Json LlmGatewayModule::resolve_model(const SecurityContext& sec,
+                                     const std::string& model) {
+  auto reg = hub_->get<ModelRegistryClient>("model-registry");
+  auto resolved = reg ? reg->get_tenant_model(sec.tenant_id, model)
+                      : std::nullopt;
+  if (!resolved)
+    throw Problem{404, "Not Found", "about:blank",
+                  "model '" + model + "' not found or not approved",
+                  "model_not_found"};
+  return *resolved;
+}
+
+void LlmGatewayModule::check_budget(const std::string& tenant) {
+  if (!budget_tokens_) return;
+  std::lock_guard<std::mutex> lk(usage_mu_);
+  auto it = usage_.find(tenant);
+  if (it != usage_.end() &&
+      it->second.input_tokens + it->second.output_tokens >= budget_tokens_)
+    throw Problem{429, "Too Many Requests", "about:blank",
+                  "tenant token budget exhausted", "budget_exceeded"};
+}
+
+void LlmGatewayModule::record_usage(const std::string& tenant,
+                                    const Json& usage) {
+  const uint64_t in = (uint64_t)usage.at("input_tokens").as_int(0);
+  const uint64_t out = (uint64_t)usage.at("output_tokens").as_int(0);
+  m_input_tokens_ += in;
+  m_output_tokens_ += out;
+  std::lock_guard<std::mutex> lk(usage_mu_);
+  auto& u = usage_[tenant];
+  u.input_tokens += in;
+  u.output_tokens += out;
+  u.requests += 1;
+}
+
+Json LlmGatewayModule::run_chat_blocking(const Json& body,
+                                         const Json& resolved,
+                                         const std::string& rid) {
+  EngineConn conn(socket_path_);
+  if (!conn.ok())
+    throw Problem{503, "Service Unavailable", "about:blank",
+                  "engine connection failed", "provider_error"};
+  Json wreq = Json::object();
+  wreq["type"] = "chat";
+  wreq["id"] = rid;
+  wreq["model"] = resolved.at("provider_model_id").as_string();
+  wreq["messages"] = body.at("messages");
+  Json params = body.at("params");
+  if (params.is_null()) params = Json::object();
+  for (const char* f : {"temperature", "top_p", "top_k", "max_tokens",
+                        "seed"})
+    if (body.contains(f)) params[f] = body.at(f);
+  wreq["params"] = params;
+  if (!conn.send_json(wreq))
+    throw Problem{502, "Bad Gateway", "about:blank", "engine write failed",
+                  "provider_error"};
+  std::string text;
+  Json usage;
+  std::string finish = "stop";
+  while (true) {
+    auto msg = conn.read_json();
+    if (!msg)
+      throw Problem{504, "Gateway Timeout", "about:blank",
+                    "engine timed out", "provider_timeout"};
+    const std::string ev = msg->at("event").as_string();
+    if (ev == "delta") text += msg->at("text").as_string();
+    else if (ev == "done") {
+      usage = msg->at("usage");
+      finish = msg->at("finish_reason").as_string("stop");
+      break;
+    } else if (ev == "error") {
+      throw Problem{502, "Bad Gateway", "about:blank",
+                    msg->at("message").as_string(), "provider_error"};
+    }
+  }
+  Json part = Json::object();
+  part["type"] = "text";
+  part["text"] = text;
+  Json content = Json::array();
+  content.push_back(part);
+  Json resp = Json::object();
+  resp["content"] = content;
+  resp["usage"] = usage;
+  resp["model_used"] = resolved.at("canonical_id").as_string();
+  resp["fallback_used"] = false;
+  resp["finish_reason"] = finish;
+  return resp;
 }
 
 // Build a stream_chunk.v1-shaped SSE event
@@ -224,6 +355,7 @@ static std::string sse_chunk(const std::string& id, const std::string& model,
 }
 
 void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
+  m_requests_++;
   Json body;
   try { body = Json::parse(req.body); }
   catch (...) {
@@ -242,14 +374,18 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   SecurityContext sec =
       SecurityContext::from_json(req.extensions.at("security"));
 
+  // async=true: the engine is a sync backend, so the gateway simulates a
+  // job (model-registry PRD.md:204-214 sync/async abstraction)
+  if (body.at("async").as_bool(false)) {
+    auto job = submit_job(sec, body);
+    w.respond(202, "application/json", job_json(*job).dump(),
+              {{"x-request-id", req.request_id}});
+    return;
+  }
+
   // model resolution via model-registry (DESIGN.md:317-346)
-  auto reg = hub_->get<ModelRegistryClient>("model-registry");
-  auto resolved = reg ? reg->get_tenant_model(sec.tenant_id, model)
-                      : std::nullopt;
-  if (!resolved)
-    throw Problem{404, "Not Found", "about:blank",
-                  "model '" + model + "' not found or not approved",
-                  "model_not_found"};
+  Json resolved = resolve_model(sec, model);
+  check_budget(sec.tenant_id);
 
   // per-tenant admission (serverless-runtime quota machinery)
   auto adm = hub_->get<AdmissionClient>("serverless-runtime");
@@ -268,6 +404,17 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
                   "inference engine is not ready", "provider_error"};
 
   const std::string rid = "chat-" + std::to_string(req_ctr_.fetch_add(1));
+
+  if (!stream) {
+    Json resp = run_chat_blocking(body, resolved, rid);
+    record_usage(sec.tenant_id, resp.at("usage"));
+    resp.erase("finish_reason");
+    w.respond(200, "application/json", resp.dump(),
+              {{"x-request-id", req.request_id}});
+    return;
+  }
+
+  m_streams_++;
   EngineConn conn(socket_path_);
   if (!conn.ok())
     throw Problem{503, "Service Unavailable", "about:blank",
@@ -275,12 +422,10 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   Json wreq = Json::object();
   wreq["type"] = "chat";
   wreq["id"] = rid;
-  wreq["model"] = resolved->at("provider_model_id").as_string();
+  wreq["model"] = resolved.at("provider_model_id").as_string();
   wreq["messages"] = body.at("messages");
   Json params = body.at("params");
   if (params.is_null()) params = Json::object();
-  // OpenAI-style top-level sampling fields accepted additively (the v1
-  // schema has none — Appendix B note)
   for (const char* f : {"temperature", "top_p", "top_k", "max_tokens",
                         "seed"})
     if (body.contains(f)) params[f] = body.at(f);
@@ -289,48 +434,13 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
     throw Problem{502, "Bad Gateway", "about:blank", "engine write failed",
                   "provider_error"};
 
-  if (!stream) {
-    std::string text;
-    Json usage;
-    std::string finish = "stop";
-    while (true) {
-      auto msg = conn.read_json();
-      if (!msg)
-        throw Problem{504, "Gateway Timeout", "about:blank",
-                      "engine timed out", "provider_timeout"};
-      const std::string ev = msg->at("event").as_string();
-      if (ev == "delta") text += msg->at("text").as_string();
-      else if (ev == "done") {
-        usage = msg->at("usage");
-        finish = msg->at("finish_reason").as_string("stop");
-        break;
-      } else if (ev == "error") {
-        throw Problem{502, "Bad Gateway", "about:blank",
-                      msg->at("message").as_string(), "provider_error"};
-      }
-    }
-    // response.v1: content[] parts + usage + model_used (required)
-    Json part = Json::object();
-    part["type"] = "text";
-    part["text"] = text;
-    Json content = Json::array();
-    content.push_back(part);
-    Json resp = Json::object();
-    resp["content"] = content;
-    resp["usage"] = usage;
-    resp["model_used"] = resolved->at("canonical_id").as_string();
-    resp["fallback_used"] = false;
-    (void)finish;
-    w.respond(200, "application/json", resp.dump(),
-              {{"x-request-id", req.request_id}});
-    return;
-  }
-
   // SSE stream per DESIGN.md:289-311: role chunk, delta chunks, final
   // finish_reason+usage chunk, then data: [DONE]
+  const auto t0 = std::chrono::steady_clock::now();
+  bool first_token = false;
   w.begin_stream(200, "text/event-stream",
                  {{"x-request-id", req.request_id}});
-  const std::string canonical = resolved->at("canonical_id").as_string();
+  const std::string canonical = resolved.at("canonical_id").as_string();
   Json role_delta = Json::object();
   role_delta["role"] = "assistant";
   w.write_chunk(sse_chunk(rid, canonical, role_delta));
@@ -343,6 +453,13 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
     }
     const std::string ev = msg->at("event").as_string();
     if (ev == "delta") {
+      if (!first_token) {
+        first_token = true;
+        m_ttft_us_sum_ += (uint64_t)
+            std::chrono::duration_cast<std::chrono::microseconds>(
+                std::chrono::steady_clock::now() - t0).count();
+        m_ttft_count_++;
+      }
       Json d = Json::object();
       d["content"] = msg->at("text").as_string();
       if (!w.write_chunk(sse_chunk(rid, canonical, d))) {
@@ -354,11 +471,13 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
         break;
       }
     } else if (ev == "done") {
+      record_usage(sec.tenant_id, msg->at("usage"));
       w.write_chunk(sse_chunk(rid, canonical, Json::object(),
                               msg->at("finish_reason").as_string("stop"),
                               msg->at("usage")));
       break;
     } else if (ev == "error") {
+      m_errors_++;
       w.write_chunk("data: {\"error\":\"provider_error\"}\n\n");
       break;
     }
@@ -366,6 +485,165 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   if (!client_gone) w.write_chunk("data: [DONE]\n\n");
   w.end_stream();
 }
+
+// -------------------------------------------------------------- embeddings
+
+void LlmGatewayModule::embeddings_handler(HttpRequest& req,
+                                          ResponseWriter& w) {
+  m_requests_++;
+  Json body;
+  try { body = Json::parse(req.body); }
+  catch (...) {
+    throw Problem{400, "Bad Request", "about:blank", "invalid JSON body",
+                  "validation_error"};
+  }
+  // embedding_request.v1: model + input (string | array of strings)
+  const std::string model = body.at("model").as_string();
+  const Json& input = body.at("input");
+  if (model.empty() || (!input.is_string() && !input.is_array()))
+    throw Problem{400, "Bad Request", "about:blank",
+                  "'model' and 'input' (string or array) are required",
+                  "validation_error"};
+  const std::string fmt =
+      body.at("encoding_format").as_string("float");
+  if (fmt != "float" && fmt != "base64")
+    throw Problem{400, "Bad Request", "about:blank",
+                  "encoding_format must be float|base64",
+                  "validation_error"};
+
+  SecurityContext sec =
+      SecurityContext::from_json(req.extensions.at("security"));
+  Json resolved = resolve_model(sec, model);
+  check_budget(sec.tenant_id);
+  if (!worker_ready())
+    throw Problem{503, "Service Unavailable", "about:blank",
+                  "inference engine is not ready", "provider_error"};
+
+  EngineConn conn(socket_path_);
+  if (!conn.ok())
+    throw Problem{503, "Service Unavailable", "about:blank",
+                  "engine connection failed", "provider_error"};
+  Json wreq = Json::object();
+  wreq["type"] = "embeddings";
+  wreq["input"] = input;
+  if (body.contains("dimensions")) wreq["dimensions"] = body.at("dimensions");
+  if (!conn.send_json(wreq))
+    throw Problem{502, "Bad Gateway", "about:blank", "engine write failed",
+                  "provider_error"};
+  auto msg = conn.read_json();
+  if (!msg)
+    throw Problem{504, "Gateway Timeout", "about:blank", "engine timed out",
+                  "provider_timeout"};
+  if (msg->at("event").as_string() == "error")
+    throw Problem{502, "Bad Gateway", "about:blank",
+                  msg->at("message").as_string(), "provider_error"};
+
+  record_usage(sec.tenant_id, msg->at("usage"));
+  Json data = Json::array();
+  const Json& vecs = msg->at("data");
+  for (size_t i = 0; i < vecs.size(); ++i) {
+    Json item = Json::object();
+    item["index"] = (int64_t)i;
+    if (fmt == "base64") {
+      const Json& v = vecs.at(i);
+      std::vector<float> f(v.size());
+      for (size_t j = 0; j < v.size(); ++j)
+        f[j] = (float)v.at(j).as_number();
+      item["embedding"] = b64_encode(
+          reinterpret_cast<const unsigned char*>(f.data()),
+          f.size() * sizeof(float));
+    } else {
+      item["embedding"] = vecs.at(i);
+    }
+    data.push_back(item);
+  }
+  Json resp = Json::object();
+  resp["data"] = data;
+  resp["model"] = resolved.at("canonical_id").as_string();
+  resp["usage"] = msg->at("usage");
+  w.respond(200, "application/json", resp.dump(),
+            {{"x-request-id", req.request_id}});
+}
+
+// ------------------------------------------------------------------- jobs
+
+Json LlmGatewayModule::job_json(const Job& j) const {
+  Json o = Json::object();
+  o["id"] = j.id;
+  o["status"] = j.status;
+  o["created_at"] = j.created_at;
+  if (j.finished_at > 0) o["finished_at"] = j.finished_at;
+  if (!j.batch_id.empty()) o["batch_id"] = j.batch_id;
+  if (j.status == "succeeded") o["result"] = j.result;
+  if (j.status == "failed") o["error"] = j.error;
+  return o;
+}
+
+std::shared_ptr<LlmGatewayModule::Job> LlmGatewayModule::submit_job(
+    const SecurityContext& sec, Json body, const std::string& batch_id) {
+  auto job = std::make_shared<Job>();
+  job->id = "job-" + std::to_string(req_ctr_.fetch_add(1));
+  job->tenant = sec.tenant_id;
+  job->batch_id = batch_id;
+  job->created_at = now_s();
+  body.erase("async");
+  body.erase("stream");
+  job->request = std::move(body);
+  {
+    std::lock_guard<std::mutex> lk(jobs_mu_);
+    jobs_[job->id] = job;
+    job_queue_.push_back(job->id);
+  }
+  jobs_cv_.notify_one();
+  return job;
+}
+
+void LlmGatewayModule::job_loop() {
+  while (true) {
+    std::shared_ptr<Job> job;
+    {
+      std::unique_lock<std::mutex> lk(jobs_mu_);
+      jobs_cv_.wait(lk, [&] { return stopping_ || !job_queue_.empty(); });
+      if (stopping_) return;
+      const std::string id = job_queue_.front();
+      job_queue_.pop_front();
+      auto it = jobs_.find(id);
+      if (it == jobs_.end()) continue;
+      job = it->second;
+      if (job->status != "queued") continue;   // cancelled while queued
+      job->status = "running";
+    }
+    Json result;
+    std::string err;
+    try {
+      SecurityContext sec;
+      sec.tenant_id = job->tenant;
+      Json resolved =
+          resolve_model(sec, job->request.at("model").as_string());
+      check_budget(job->tenant);
+      result = run_chat_blocking(job->request, resolved, job->id);
+    } catch (const Problem& p) {
+      err = p.code.empty() ? "provider_error" : p.code;
+    } catch (...) {
+      err = "provider_error";
+    }
+    {
+      std::lock_guard<std::mutex> lk(jobs_mu_);
+      if (err.empty()) {
+        job->status = "succeeded";
+        result.erase("finish_reason");
+        job->result = result;
+      } else {
+        job->status = "failed";
+        job->error = err;
+      }
+      job->finished_at = now_s();
+    }
+    if (err.empty()) record_usage(job->tenant, job->result.at("usage"));
+  }
+}
+
+// ------------------------------------------------------------------ REST
 
 void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
   Json req_schema = Json::object();
@@ -379,6 +657,7 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
     props["messages"] = msgs;
     Json b = Json::object(); b["type"] = "boolean";
     props["stream"] = b;
+    props["async"] = b;
     Json n = Json::object(); n["type"] = "number";
     props["temperature"] = n;
     props["top_p"] = n;
@@ -403,14 +682,291 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
     op.path = p;
     op.operation_id = std::string("chat_completions") +
         (p[1] == 'v' ? "_alias" : "");
-    op.summary = "Chat completion (sync or SSE stream)";
+    op.summary = "Chat completion (sync, SSE stream, or async job)";
     op.authenticated = true;
     op.allowed_content_types = {"application/json"};
     op.tags = {"llm-gateway"};
     op.request_schema = req_schema;
     op.responses[200] = "completion response or SSE stream";
+    op.responses[202] = "async job accepted";
     op.sse = true;
     rest.register_op(op, handler);
+  }
+
+  for (const char* p : {"/llm-gateway/v1/embeddings", "/v1/embeddings"}) {
+    OperationSpec op;
+    op.method = "POST";
+    op.path = p;
+    op.operation_id = std::string("embeddings") +
+        (p[1] == 'v' ? "_alias" : "");
+    op.summary = "Embeddings (embedding_request.v1)";
+    op.authenticated = true;
+    op.allowed_content_types = {"application/json"};
+    op.tags = {"llm-gateway"};
+    op.responses[200] = "embedding_response.v1";
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      embeddings_handler(rq, w);
+    });
+  }
+
+  // ---- jobs (DESIGN.md:262-270 async surface) ----
+  {
+    OperationSpec op;
+    op.method = "POST";
+    op.path = "/llm-gateway/v1/jobs";
+    op.operation_id = "create_job";
+    op.summary = "Submit an async chat job";
+    op.authenticated = true;
+    op.allowed_content_types = {"application/json"};
+    op.tags = {"llm-gateway"};
+    op.request_schema = req_schema;
+    op.responses[202] = "job.v1";
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      m_requests_++;
+      Json body;
+      try { body = Json::parse(rq.body); }
+      catch (...) { throw Problem::bad_request("invalid JSON body"); }
+      if (body.at("model").as_string().empty() ||
+          !body.at("messages").is_array())
+        throw Problem{400, "Bad Request", "about:blank",
+                      "'model' and 'messages' are required",
+                      "validation_error"};
+      auto sec = SecurityContext::from_json(rq.extensions.at("security"));
+      auto job = submit_job(sec, body);
+      w.respond(202, "application/json", job_json(*job).dump());
+    });
+  }
+  {
+    OperationSpec op;
+    op.method = "GET";
+    op.path = "/llm-gateway/v1/jobs";
+    op.operation_id = "list_jobs";
+    op.summary = "List this tenant's jobs";
+    op.authenticated = true;
+    op.tags = {"llm-gateway"};
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      auto sec = SecurityContext::from_json(rq.extensions.at("security"));
+      Json items = Json::array();
+      std::lock_guard<std::mutex> lk(jobs_mu_);
+      for (auto& [id, j] : jobs_)
+        if (j->tenant == sec.tenant_id) items.push_back(job_json(*j));
+      Json out = Json::object();
+      out["items"] = items;
+      w.respond(200, "application/json", out.dump());
+    });
+  }
+  {
+    OperationSpec op;
+    op.method = "GET";
+    op.path = "/llm-gateway/v1/jobs/{id}";
+    op.operation_id = "get_job";
+    op.summary = "Job status/result";
+    op.authenticated = true;
+    op.tags = {"llm-gateway"};
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      auto sec = SecurityContext::from_json(rq.extensions.at("security"));
+      std::lock_guard<std::mutex> lk(jobs_mu_);
+      auto it = jobs_.find(rq.path_params.at("id"));
+      if (it == jobs_.end() || it->second->tenant != sec.tenant_id)
+        throw Problem{404, "Not Found", "about:blank", "no such job",
+                      "job_not_found"};
+      w.respond(200, "application/json", job_json(*it->second).dump());
+    });
+  }
+  {
+    OperationSpec op;
+    op.method = "DELETE";
+    op.path = "/llm-gateway/v1/jobs/{id}";
+    op.operation_id = "cancel_job";
+    op.summary = "Cancel a queued job";
+    op.authenticated = true;
+    op.tags = {"llm-gateway"};
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      auto sec = SecurityContext::from_json(rq.extensions.at("security"));
+      std::lock_guard<std::mutex> lk(jobs_mu_);
+      auto it = jobs_.find(rq.path_params.at("id"));
+      if (it == jobs_.end() || it->second->tenant != sec.tenant_id)
+        throw Problem{404, "Not Found", "about:blank", "no such job",
+                      "job_not_found"};
+      if (it->second->status == "queued") {
+        it->second->status = "cancelled";
+        it->second->finished_at = now_s();
+      }
+      w.respond(200, "application/json", job_json(*it->second).dump());
+    });
+  }
+
+  // ---- batches ----
+  {
+    OperationSpec op;
+    op.method = "POST";
+    op.path = "/llm-gateway/v1/batches";
+    op.operation_id = "create_batch";
+    op.summary = "Submit a batch of chat requests";
+    op.authenticated = true;
+    op.allowed_content_types = {"application/json"};
+    op.tags = {"llm-gateway"};
+    op.responses[202] = "batch.v1";
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      m_requests_++;
+      Json body;
+      try { body = Json::parse(rq.body); }
+      catch (...) { throw Problem::bad_request("invalid JSON body"); }
+      const Json& reqs = body.at("requests");
+      if (!reqs.is_array() || reqs.size() == 0)
+        throw Problem{400, "Bad Request", "about:blank",
+                      "'requests' (non-empty array) is required",
+                      "validation_error"};
+      auto sec = SecurityContext::from_json(rq.extensions.at("security"));
+      Batch b;
+      b.id = "batch-" + std::to_string(req_ctr_.fetch_add(1));
+      b.tenant = sec.tenant_id;
+      b.created_at = now_s();
+      for (size_t i = 0; i < reqs.size(); ++i)
+        b.job_ids.push_back(submit_job(sec, reqs.at(i), b.id)->id);
+      Json out = Json::object();
+      out["id"] = b.id;
+      out["status"] = "queued";
+      out["num_requests"] = (int64_t)b.job_ids.size();
+      out["created_at"] = b.created_at;
+      {
+        std::lock_guard<std::mutex> lk(jobs_mu_);
+        batches_[b.id] = std::move(b);
+      }
+      w.respond(202, "application/json", out.dump());
+    });
+  }
+  {
+    OperationSpec op;
+    op.method = "GET";
+    op.path = "/llm-gateway/v1/batches/{id}";
+    op.operation_id = "get_batch";
+    op.summary = "Batch status + per-request results";
+    op.authenticated = true;
+    op.tags = {"llm-gateway"};
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      auto sec = SecurityContext::from_json(rq.extensions.at("security"));
+      std::lock_guard<std::mutex> lk(jobs_mu_);
+      auto it = batches_.find(rq.path_params.at("id"));
+      if (it == batches_.end() || it->second.tenant != sec.tenant_id)
+        throw Problem{404, "Not Found", "about:blank", "no such batch",
+                      "job_not_found"};
+      Json items = Json::array();
+      size_t done = 0, failed = 0;
+      for (auto& id : it->second.job_ids) {
+        auto jit = jobs_.find(id);
+        if (jit == jobs_.end()) continue;
+        items.push_back(job_json(*jit->second));
+        const auto& st = jit->second->status;
+        if (st == "succeeded") done++;
+        else if (st == "failed" || st == "cancelled") failed++;
+      }
+      Json out = Json::object();
+      out["id"] = it->second.id;
+      out["created_at"] = it->second.created_at;
+      out["num_requests"] = (int64_t)it->second.job_ids.size();
+      out["status"] =
+          (done + failed == it->second.job_ids.size())
+              ? (failed ? "completed_with_errors" : "completed")
+              : "in_progress";
+      out["jobs"] = items;
+      w.respond(200, "application/json", out.dump());
+    });
+  }
+
+  // ---- usage (PRD.md:224-232 usage tracking) ----
+  {
+    OperationSpec op;
+    op.method = "GET";
+    op.path = "/llm-gateway/v1/usage";
+    op.operation_id = "get_usage";
+    op.summary = "This tenant's accumulated usage";
+    op.authenticated = true;
+    op.tags = {"llm-gateway"};
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      auto sec = SecurityContext::from_json(rq.extensions.at("security"));
+      Json out = Json::object();
+      std::lock_guard<std::mutex> lk(usage_mu_);
+      auto& u = usage_[sec.tenant_id];
+      out["tenant_id"] = sec.tenant_id;
+      out["input_tokens"] = (int64_t)u.input_tokens;
+      out["output_tokens"] = (int64_t)u.output_tokens;
+      out["requests"] = (int64_t)u.requests;
+      if (budget_tokens_) {
+        out["budget_tokens"] = (int64_t)budget_tokens_;
+        out["budget_remaining"] = (int64_t)std::max<int64_t>(
+            0, (int64_t)budget_tokens_
+                   - (int64_t)(u.input_tokens + u.output_tokens));
+      }
+      w.respond(200, "application/json", out.dump());
+    });
+  }
+
+  // ---- Prometheus metrics (the surface the reference
+  // lacks; serving KPIs + engine KV occupancy) ----
+  {
+    OperationSpec op;
+    op.method = "GET";
+    op.path = "/metrics";
+    op.operation_id = "metrics";
+    op.summary = "Prometheus metrics";
+    op.is_public = true;
+    op.tags = {"observability"};
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      std::string t;
+      auto add = [&](const char* name, const char* help, const char* type,
+                     double v) {
+        t += "# HELP " + std::string(name) + " " + help + "\n";
+        t += "# TYPE " + std::string(name) + " " + type + "\n";
+        t += std::string(name) + " " + std::to_string(v) + "\n";
+      };
+      add("hyperspot_requests_total", "Gateway requests", "counter",
+          (double)m_requests_.load());
+      add("hyperspot_sse_streams_total", "SSE streams started", "counter",
+          (double)m_streams_.load());
+      add("hyperspot_engine_errors_total", "Engine-side errors", "counter",
+          (double)m_errors_.load());
+      add("hyperspot_input_tokens_total", "Prompt tokens", "counter",
+          (double)m_input_tokens_.load());
+      add("hyperspot_output_tokens_total", "Generated tokens", "counter",
+          (double)m_output_tokens_.load());
+      if (m_ttft_count_)
+        add("hyperspot_ttft_seconds_avg", "Mean stream TTFT", "gauge",
+            (double)m_ttft_us_sum_ / 1e6 / (double)m_ttft_count_);
+      if (worker_ready()) {
+        EngineConn c(socket_path_);
+        Json q = Json::object();
+        q["type"] = "info";
+        if (c.ok() && c.send_json(q)) {
+          if (auto r = c.read_json(3000)) {
+            add("hyperspot_engine_running", "Sequences decoding", "gauge",
+                (double)r->at("num_running").as_int(0));
+            add("hyperspot_engine_waiting", "Sequences queued", "gauge",
+                (double)r->at("num_waiting").as_int(0));
+            const double total = (double)r->at("kv_blocks_total").as_int(0);
+            const double freeb = (double)r->at("kv_blocks_free").as_int(0);
+            add("hyperspot_kv_blocks_total", "KV pool pages", "gauge",
+                total);
+            add("hyperspot_kv_blocks_free", "Free KV pages", "gauge",
+                freeb);
+            if (total > 0)
+              add("hyperspot_kv_occupancy", "KV pool occupancy 0..1",
+                  "gauge", (total - freeb) / total);
+          }
+        }
+      }
+      // per-tenant usage
+      {
+        std::lock_guard<std::mutex> lk(usage_mu_);
+        t += "# HELP hyperspot_tenant_tokens_total Tokens per tenant\n";
+        t += "# TYPE hyperspot_tenant_tokens_total counter\n";
+        for (auto& [tenant, u] : usage_)
+          t += "hyperspot_tenant_tokens_total{tenant=\"" + tenant +
+               "\"} " +
+               std::to_string(u.input_tokens + u.output_tokens) + "\n";
+      }
+      w.respond(200, "text/plain; version=0.0.4", t);
+    });
   }
 
   OperationSpec status;

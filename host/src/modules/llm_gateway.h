@@ -6,9 +6,16 @@
 // engine (hyperspot.serving.worker, one process per GPU) over a Unix
 // socket — the reference's OoP module pattern with the
 // provider call replaced by the engine (SURVEY.md §3.3 note).
+//
+// Surface (DESIGN.md:262-270): POST /chat/completions (sync + SSE),
+// POST /embeddings, POST/GET/DELETE /jobs{,/{id}}, POST/GET
+// /batches{,/{id}}; plus usage tracking (PRD.md:224-232) and a
+// Prometheus /metrics endpoint (SURVEY.md §5.5 gap the build must fill).
 #pragma once
 
 #include <atomic>
+#include <condition_variable>
+#include <deque>
 #include <mutex>
 #include <thread>
 
@@ -63,8 +70,42 @@ class LlmGatewayModule : public Module {
   bool worker_ready();
 
  private:
+  // async job state machine (DESIGN.md job/batch schemas; statuses
+  // queued|running|succeeded|failed|cancelled)
+  struct Job {
+    std::string id, tenant, batch_id;
+    std::string status = "queued";
+    Json request;
+    Json result;
+    std::string error;       // problem code on failure
+    double created_at = 0, finished_at = 0;
+  };
+  struct Batch {
+    std::string id, tenant;
+    std::vector<std::string> job_ids;
+    double created_at = 0;
+  };
+  struct TenantUsage {
+    uint64_t input_tokens = 0, output_tokens = 0, requests = 0;
+  };
+
   void chat_handler(HttpRequest& req, ResponseWriter& w);
+  void embeddings_handler(HttpRequest& req, ResponseWriter& w);
   void spawn_worker();
+
+  // shared pipeline pieces
+  Json resolve_model(const SecurityContext& sec, const std::string& model);
+  void check_budget(const std::string& tenant);
+  void record_usage(const std::string& tenant, const Json& usage);
+  // blocking non-stream chat against the engine; throws Problem
+  Json run_chat_blocking(const Json& body, const Json& resolved,
+                         const std::string& rid);
+
+  // jobs
+  std::shared_ptr<Job> submit_job(const SecurityContext& sec, Json body,
+                                  const std::string& batch_id = "");
+  Json job_json(const Job& j) const;
+  void job_loop();
 
   ClientHub* hub_ = nullptr;
   std::string model_ = "llama3-8b";
@@ -75,6 +116,25 @@ class LlmGatewayModule : public Module {
   pid_t worker_pid_ = -1;
   std::atomic<bool> ready_{false};
   std::atomic<uint64_t> req_ctr_{0};
+
+  // jobs/batches (in-memory, tenant-scoped)
+  std::mutex jobs_mu_;
+  std::condition_variable jobs_cv_;
+  std::deque<std::string> job_queue_;
+  std::map<std::string, std::shared_ptr<Job>> jobs_;
+  std::map<std::string, Batch> batches_;
+  std::thread job_thread_;
+  std::atomic<bool> stopping_{false};
+
+  // usage tracker + budget (tokens per tenant; 0 = unlimited)
+  std::mutex usage_mu_;
+  std::map<std::string, TenantUsage> usage_;
+  uint64_t budget_tokens_ = 0;
+
+  // metrics
+  std::atomic<uint64_t> m_requests_{0}, m_streams_{0}, m_errors_{0};
+  std::atomic<uint64_t> m_input_tokens_{0}, m_output_tokens_{0};
+  std::atomic<uint64_t> m_ttft_us_sum_{0}, m_ttft_count_{0};
 };
 
 }  // namespace hs

@@ -84,6 +84,14 @@ class Json {
     auto it = obj().find(k);
     return it == obj().end() ? null_json : it->second;
   }
+  const Json& at(size_t i) const {
+    static const Json null_json;
+    if (!is_array() || i >= arr().size()) return null_json;
+    return arr()[i];
+  }
+  void erase(const std::string& k) {
+    if (is_object()) obj().erase(k);
+  }
   // dotted-path lookup ("modules.llm-gateway.config")
   const Json& path(const std::string& dotted) const;
 

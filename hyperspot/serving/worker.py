@@ -216,6 +216,7 @@ def main():
     ap.add_argument("--eager", action="store_true")
     ap.add_argument("--device", default=None)
     ap.add_argument("--tp", type=int, default=1)
+    ap.add_argument("--quant", default=None, choices=["fp8"])
     args = ap.parse_args()
 
     logging.basicConfig(level=logging.INFO,
@@ -229,7 +230,8 @@ def main():
         model=args.model, max_num_seqs=args.max_num_seqs,
         max_model_len=args.max_model_len,
         num_gpu_blocks=args.num_gpu_blocks or None,
-        enforce_eager=args.eager or not on_gpu, tp_size=args.tp)
+        enforce_eager=args.eager or not on_gpu, tp_size=args.tp,
+        quant=args.quant)
     if args.tp > 1:
         from hyperspot.parallel.state import initialize_model_parallel
         initialize_model_parallel(tp_size=args.tp)

@@ -392,3 +392,98 @@ def test_cli_check_and_introspection():
                         str(ROOT / "config" / "no-db.yaml")],
                        capture_output=True, text=True, cwd=ROOT)
     assert "llm-gateway" in json.loads(r.stdout)
+
+
+def test_embeddings_endpoint(server):
+    st, body = _http("POST",
+                     BASE.format(server.port) + "/llm-gateway/v1/embeddings",
+                     {"model": "tiny-llama", "input": ["hello", "world"]})
+    assert st == 200, body
+    d = json.loads(body)
+    assert len(d["data"]) == 2
+    assert d["data"][0]["index"] == 0
+    assert isinstance(d["data"][0]["embedding"], list)
+    assert d["usage"]["input_tokens"] > 0
+    assert d["model"].endswith("tiny-llama")
+    # base64 variant
+    st, body = _http("POST",
+                     BASE.format(server.port) + "/v1/embeddings",
+                     {"model": "tiny-llama", "input": "hi",
+                      "encoding_format": "base64"})
+    assert st == 200, body
+    d = json.loads(body)
+    assert isinstance(d["data"][0]["embedding"], str)
+    import base64
+    raw = base64.b64decode(d["data"][0]["embedding"])
+    assert len(raw) % 4 == 0 and len(raw) > 0
+
+
+def test_async_job_lifecycle(server):
+    url = BASE.format(server.port)
+    st, body = _http("POST", url + "/llm-gateway/v1/jobs",
+                     {"model": "tiny-llama",
+                      "messages": [{"role": "user", "content":
+                                    [{"type": "text", "text": "hi"}]}],
+                      "max_tokens": 4})
+    assert st == 202, body
+    jid = json.loads(body)["id"]
+    for _ in range(200):
+        st, body = _http("GET", url + f"/llm-gateway/v1/jobs/{jid}")
+        assert st == 200, body
+        j = json.loads(body)
+        if j["status"] in ("succeeded", "failed"):
+            break
+        time.sleep(0.2)
+    assert j["status"] == "succeeded", j
+    assert j["result"]["usage"]["output_tokens"] > 0
+    assert j["result"]["content"][0]["type"] == "text"
+    # listing contains it; unknown id is a job_not_found problem
+    st, body = _http("GET", url + "/llm-gateway/v1/jobs")
+    assert any(x["id"] == jid for x in json.loads(body)["items"])
+    st, body = _http("GET", url + "/llm-gateway/v1/jobs/nope")
+    assert st == 404 and json.loads(body)["code"] == "job_not_found"
+
+
+def test_async_chat_flag_makes_job(server):
+    url = BASE.format(server.port)
+    st, body = _http("POST", url + "/v1/chat/completions",
+                     {"model": "tiny-llama", "async": True,
+                      "messages": [{"role": "user", "content":
+                                    [{"type": "text", "text": "x"}]}],
+                      "max_tokens": 2})
+    assert st == 202, body
+    assert json.loads(body)["status"] in ("queued", "running")
+
+
+def test_batches(server):
+    url = BASE.format(server.port)
+    reqs = [{"model": "tiny-llama",
+             "messages": [{"role": "user", "content":
+                           [{"type": "text", "text": f"n{i}"}]}],
+             "max_tokens": 2} for i in range(3)]
+    st, body = _http("POST", url + "/llm-gateway/v1/batches",
+                     {"requests": reqs})
+    assert st == 202, body
+    bid = json.loads(body)["id"]
+    for _ in range(300):
+        st, body = _http("GET", url + f"/llm-gateway/v1/batches/{bid}")
+        b = json.loads(body)
+        if b["status"].startswith("completed"):
+            break
+        time.sleep(0.2)
+    assert b["status"] == "completed", b
+    assert b["num_requests"] == 3
+    assert all(j["status"] == "succeeded" for j in b["jobs"])
+
+
+def test_usage_and_metrics(server):
+    url = BASE.format(server.port)
+    st, body = _http("GET", url + "/llm-gateway/v1/usage")
+    assert st == 200, body
+    u = json.loads(body)
+    assert u["input_tokens"] > 0 and u["requests"] > 0
+    st, body = _http("GET", url + "/metrics")
+    assert st == 200
+    assert "hyperspot_requests_total" in body
+    assert "hyperspot_kv_blocks_total" in body
+    assert "hyperspot_tenant_tokens_total{tenant=" in body
