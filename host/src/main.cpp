@@ -15,6 +15,7 @@
 #include "modkit/modkit.h"
 #include "modules/api_gateway.h"
 #include "modules/llm_gateway.h"
+#include "modules/user_settings.h"
 #include "modules/oagw.h"
 #include "modules/system_modules.h"
 #include "util/log.h"
@@ -51,6 +52,7 @@ int run_server(const Json& cfg, bool check_only) {
   auto orch = std::make_shared<ModuleOrchestratorModule>();
   registry.add(orch);
   registry.add(std::make_shared<LlmGatewayModule>());
+  registry.add(std::make_shared<UserSettingsModule>());
 
   ClientHub hub;
   bool cancel_flag = false;
@@ -107,7 +109,8 @@ int list_modules(const Json& cfg) {
                         "authz-resolver", "types-registry",
                         "nodes-registry", "model-registry", "credstore",
                         "serverless-runtime", "oagw", "file-storage",
-                        "module-orchestrator", "llm-gateway"})
+                        "module-orchestrator", "llm-gateway",
+                        "simple-user-settings"})
     std::cout << m << "\n";
   return 0;
 }

@@ -326,6 +326,9 @@ modules:
       model: "tiny-llama"
       worker_socket: "{sock}"
       auto_start_worker: false
+  simple-user-settings:
+    database:
+      file: "{tempfile.mktemp(suffix='.db', prefix='hs-sus-')}"
 """
     cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
     cfg_path.write_text(cfg)
@@ -487,3 +490,76 @@ def test_usage_and_metrics(server):
     assert "hyperspot_requests_total" in body
     assert "hyperspot_kv_blocks_total" in body
     assert "hyperspot_tenant_tokens_total{tenant=" in body
+
+
+def test_settings_crud_and_tenant_isolation(mt_server):
+    url = BASE.format(mt_server.port)
+    # acme writes a setting
+    st, body = _http("PUT", url + "/simple-user-settings/v1/settings/theme",
+                     {"value": {"mode": "dark"}}, token="acme-token")
+    assert st == 200, body
+    st, body = _http("GET", url + "/simple-user-settings/v1/settings/theme",
+                     token="acme-token")
+    assert st == 200 and json.loads(body)["value"] == {"mode": "dark"}
+    # root (different tenant + subject) cannot see it — secure scope
+    st, body = _http("GET", url + "/simple-user-settings/v1/settings/theme",
+                     token="root-token")
+    assert st == 404, body
+    # update in place
+    st, _ = _http("PUT", url + "/simple-user-settings/v1/settings/theme",
+                  {"value": "light"}, token="acme-token")
+    st, body = _http("GET", url + "/simple-user-settings/v1/settings/theme",
+                     token="acme-token")
+    assert json.loads(body)["value"] == "light"
+    # delete
+    st, _ = _http("DELETE", url + "/simple-user-settings/v1/settings/theme",
+                  token="acme-token")
+    assert st == 204
+    st, _ = _http("GET", url + "/simple-user-settings/v1/settings/theme",
+                  token="acme-token")
+    assert st == 404
+
+
+def test_settings_cursor_pagination_and_filter(mt_server):
+    url = BASE.format(mt_server.port)
+    for i in range(7):
+        st, _ = _http("PUT",
+                      url + f"/simple-user-settings/v1/settings/pg{i:02d}",
+                      {"value": i}, token="acme-token")
+        assert st == 200
+    seen = []
+    cursor = ""
+    for _ in range(10):
+        q = "?$top=3&$filter=contains(key,'pg')"
+        if cursor:
+            q += f"&cursor={cursor}"
+        st, body = _http("GET",
+                         url + "/simple-user-settings/v1/settings" + q,
+                         token="acme-token")
+        assert st == 200, body
+        d = json.loads(body)
+        seen += [it["key"] for it in d["items"]]
+        cursor = d["page_info"].get("next_cursor", "")
+        if not cursor:
+            break
+    assert seen == [f"pg{i:02d}" for i in range(7)], seen
+    # filter eq (urlencoded)
+    from urllib.parse import quote
+    st, body = _http(
+        "GET",
+        url + "/simple-user-settings/v1/settings?$filter="
+        + quote("key eq 'pg03'"), token="acme-token")
+    d = json.loads(body)
+    assert [it["key"] for it in d["items"]] == ["pg03"]
+    # disallowed field
+    st, body = _http(
+        "GET",
+        url + "/simple-user-settings/v1/settings?$filter="
+        + quote("value eq 'x'"), token="acme-token")
+    assert st == 400
+
+
+def test_settings_pdp_deny_anonymous(mt_server):
+    url = BASE.format(mt_server.port)
+    st, body = _http("GET", url + "/simple-user-settings/v1/settings")
+    assert st == 401  # no token at the gateway
