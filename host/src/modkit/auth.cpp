@@ -1,0 +1,178 @@
+#include "auth.h"
+
+#include <openssl/bio.h>
+#include <openssl/evp.h>
+#include <openssl/hmac.h>
+#include <openssl/pem.h>
+
+#include <ctime>
+#include <vector>
+
+namespace hs {
+
+namespace {
+
+int b64url_val(char c) {
+  if (c >= 'A' && c <= 'Z') return c - 'A';
+  if (c >= 'a' && c <= 'z') return c - 'a' + 26;
+  if (c >= '0' && c <= '9') return c - '0' + 52;
+  if (c == '-' || c == '+') return 62;
+  if (c == '_' || c == '/') return 63;
+  return -1;
+}
+
+std::optional<std::string> b64url_decode(const std::string& s) {
+  std::string out;
+  unsigned acc = 0;
+  int bits = 0;
+  for (char c : s) {
+    if (c == '=') break;
+    int v = b64url_val(c);
+    if (v < 0) return std::nullopt;
+    acc = (acc << 6) | (unsigned)v;
+    bits += 6;
+    if (bits >= 8) {
+      bits -= 8;
+      out.push_back((char)((acc >> bits) & 0xff));
+    }
+  }
+  return out;
+}
+
+std::string b64url_encode(const unsigned char* d, size_t n) {
+  static const char* tbl =
+      "ABCDEFGHIJKLMNOPQRSTUVWXYZabcdefghijklmnopqrstuvwxyz0123456789-_";
+  std::string out;
+  for (size_t i = 0; i < n; i += 3) {
+    unsigned v = d[i] << 16;
+    if (i + 1 < n) v |= d[i + 1] << 8;
+    if (i + 2 < n) v |= d[i + 2];
+    out.push_back(tbl[(v >> 18) & 63]);
+    out.push_back(tbl[(v >> 12) & 63]);
+    if (i + 1 < n) out.push_back(tbl[(v >> 6) & 63]);
+    if (i + 2 < n) out.push_back(tbl[v & 63]);
+  }
+  return out;
+}
+
+bool hmac_sha256(const std::string& key, const std::string& msg,
+                 unsigned char out[32]) {
+  unsigned int len = 32;
+  return HMAC(EVP_sha256(), key.data(), (int)key.size(),
+              (const unsigned char*)msg.data(), msg.size(), out,
+              &len) != nullptr;
+}
+
+bool rs256_verify(const std::string& pem, const std::string& msg,
+                  const std::string& sig) {
+  BIO* bio = BIO_new_mem_buf(pem.data(), (int)pem.size());
+  if (!bio) return false;
+  EVP_PKEY* pkey = PEM_read_bio_PUBKEY(bio, nullptr, nullptr, nullptr);
+  BIO_free(bio);
+  if (!pkey) return false;
+  EVP_MD_CTX* ctx = EVP_MD_CTX_new();
+  bool ok = false;
+  if (ctx &&
+      EVP_DigestVerifyInit(ctx, nullptr, EVP_sha256(), nullptr, pkey) == 1 &&
+      EVP_DigestVerify(ctx, (const unsigned char*)sig.data(), sig.size(),
+                       (const unsigned char*)msg.data(), msg.size()) == 1)
+    ok = true;
+  if (ctx) EVP_MD_CTX_free(ctx);
+  EVP_PKEY_free(pkey);
+  return ok;
+}
+
+bool const_eq(const std::string& a, const std::string& b) {
+  if (a.size() != b.size()) return false;
+  unsigned char d = 0;
+  for (size_t i = 0; i < a.size(); ++i)
+    d |= (unsigned char)(a[i] ^ b[i]);
+  return d == 0;
+}
+
+}  // namespace
+
+std::optional<SecurityContext> JwtValidator::validate(
+    const std::string& token, std::string* err) const {
+  auto fail = [&](const char* m) {
+    if (err) *err = m;
+    return std::nullopt;
+  };
+  size_t d1 = token.find('.');
+  size_t d2 = token.rfind('.');
+  if (d1 == std::string::npos || d2 == d1) return fail("malformed token");
+  const std::string signed_part = token.substr(0, d2);
+  auto hdr_raw = b64url_decode(token.substr(0, d1));
+  auto pl_raw = b64url_decode(token.substr(d1 + 1, d2 - d1 - 1));
+  auto sig_raw = b64url_decode(token.substr(d2 + 1));
+  if (!hdr_raw || !pl_raw || !sig_raw) return fail("bad base64url");
+  Json hdr, pl;
+  try {
+    hdr = Json::parse(*hdr_raw);
+    pl = Json::parse(*pl_raw);
+  } catch (...) {
+    return fail("bad JSON");
+  }
+  const std::string alg = hdr.at("alg").as_string();
+  if (alg == "HS256") {
+    if (hs256_secret.empty()) return fail("HS256 not configured");
+    unsigned char mac[32];
+    if (!hmac_sha256(hs256_secret, signed_part, mac))
+      return fail("hmac failed");
+    if (!const_eq(std::string((char*)mac, 32), *sig_raw))
+      return fail("bad signature");
+  } else if (alg == "RS256") {
+    if (rs256_public_pem.empty()) return fail("RS256 not configured");
+    if (!rs256_verify(rs256_public_pem, signed_part, *sig_raw))
+      return fail("bad signature");
+  } else {
+    return fail("unsupported alg");   // incl. alg=none — always rejected
+  }
+  const long now = (long)time(nullptr);
+  if (pl.contains("exp") && now > pl.at("exp").as_int(0) + leeway_s)
+    return fail("token expired");
+  if (pl.contains("nbf") && now + leeway_s < pl.at("nbf").as_int(0))
+    return fail("token not yet valid");
+  if (!issuer.empty() && pl.at("iss").as_string() != issuer)
+    return fail("wrong issuer");
+  if (!audience.empty()) {
+    const Json& aud = pl.at("aud");
+    bool ok = aud.as_string() == audience;
+    if (!ok && aud.is_array())
+      for (auto& a : aud.arr())
+        if (a.as_string() == audience) ok = true;
+    if (!ok) return fail("wrong audience");
+  }
+  SecurityContext c;
+  c.subject_id = pl.at("sub").as_string();
+  if (c.subject_id.empty()) return fail("missing sub");
+  c.tenant_id = pl.at(tenant_claim).as_string(kDefaultTenantId);
+  c.subject_type = pl.at("typ").as_string("user");
+  // scopes: RFC8693-style space-separated "scope" or array "scp"
+  const std::string sc = pl.at("scope").as_string();
+  size_t p = 0;
+  while (p < sc.size()) {
+    size_t q = sc.find(' ', p);
+    if (q == std::string::npos) q = sc.size();
+    if (q > p) c.scopes.push_back(sc.substr(p, q - p));
+    p = q + 1;
+  }
+  if (pl.at("scp").is_array())
+    for (auto& x : pl.at("scp").arr()) c.scopes.push_back(x.as_string());
+  return c;
+}
+
+std::string JwtValidator::sign_hs256(const Json& claims) const {
+  const std::string hdr = R"({"alg":"HS256","typ":"JWT"})";
+  std::string part =
+      b64url_encode((const unsigned char*)hdr.data(), hdr.size()) + "." +
+      [&] {
+        std::string p = claims.dump();
+        return b64url_encode((const unsigned char*)p.data(), p.size());
+      }();
+  unsigned char mac[32];
+  hmac_sha256(hs256_secret, part, mac);
+  return part + "." + b64url_encode(mac, 32);
+}
+
+}  // namespace hs

@@ -1,5 +1,7 @@
 #include "system_modules.h"
 
+#include "../modkit/auth.h"
+
 #include <sys/sysinfo.h>
 #include <sys/stat.h>
 #include <unistd.h>
@@ -90,9 +92,21 @@ void TenantResolverModule::init(ModuleCtx& ctx) {
 
 namespace {
 
+// static tokens + optional JWT validation (modkit-auth equivalent;
+// config.jwt: {hs256_secret | rs256_public_pem, issuer?, audience?,
+// tenant_claim?, leeway_s?})
 class StaticAuthnResolver : public AuthnResolverClient {
  public:
   explicit StaticAuthnResolver(const Json& cfg) {
+    const Json& jw = cfg.at("jwt");
+    if (jw.is_object()) {
+      jwt_.hs256_secret = jw.at("hs256_secret").as_string("");
+      jwt_.rs256_public_pem = jw.at("rs256_public_pem").as_string("");
+      jwt_.issuer = jw.at("issuer").as_string("");
+      jwt_.audience = jw.at("audience").as_string("");
+      jwt_.tenant_claim = jw.at("tenant_claim").as_string("tid");
+      jwt_.leeway_s = (int)jw.at("leeway_s").as_int(30);
+    }
     // static-authn-plugin shape (plugins/static-authn-plugin/src/config.rs):
     // {tokens: [{token, subject_id, subject_tenant_id, subject_type?,
     //            scopes?: []}]}
@@ -113,12 +127,14 @@ class StaticAuthnResolver : public AuthnResolverClient {
   std::optional<SecurityContext> authenticate(
       const std::string& bearer) override {
     auto it = map_.find(bearer);
-    if (it == map_.end()) return std::nullopt;
-    return it->second;
+    if (it != map_.end()) return it->second;
+    if (jwt_.configured()) return jwt_.validate(bearer);
+    return std::nullopt;
   }
 
  private:
   std::map<std::string, SecurityContext> map_;
+  JwtValidator jwt_;
 };
 
 }  // namespace

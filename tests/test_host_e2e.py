@@ -314,6 +314,9 @@ modules:
           parent_id: "00000000-df51-5b42-9538-d2b56b7ee953"
   authn-resolver:
     config:
+      jwt:
+        hs256_secret: "e2e-test-secret"
+        issuer: "hyperspot-e2e"
       tokens:
         - token: "root-token"
           subject_id: "root-user"
@@ -563,3 +566,44 @@ def test_settings_pdp_deny_anonymous(mt_server):
     url = BASE.format(mt_server.port)
     st, body = _http("GET", url + "/simple-user-settings/v1/settings")
     assert st == 401  # no token at the gateway
+
+
+def _mint_jwt(claims, secret="e2e-test-secret"):
+    import base64
+    import hashlib
+    import hmac as hm
+
+    def b64u(b):
+        return base64.urlsafe_b64encode(b).rstrip(b"=").decode()
+    part = (b64u(json.dumps({"alg": "HS256", "typ": "JWT"}).encode())
+            + "." + b64u(json.dumps(claims).encode()))
+    sig = hm.new(secret.encode(), part.encode(), hashlib.sha256).digest()
+    return part + "." + b64u(sig)
+
+
+def test_jwt_hs256_auth(mt_server):
+    url = BASE.format(mt_server.port)
+    now = int(time.time())
+    tok = _mint_jwt({"sub": "jwt-user", "iss": "hyperspot-e2e",
+                     "tid": "11111111-1111-1111-1111-111111111111",
+                     "exp": now + 600, "scope": "chat settings"})
+    st, body = _http("GET", url + "/simple-user-settings/v1/settings",
+                     token=tok)
+    assert st == 200, body
+    # expired token rejected
+    tok = _mint_jwt({"sub": "jwt-user", "iss": "hyperspot-e2e",
+                     "exp": now - 600})
+    st, body = _http("GET", url + "/simple-user-settings/v1/settings",
+                     token=tok)
+    assert st == 401, body
+    # wrong signature rejected
+    tok = _mint_jwt({"sub": "x", "iss": "hyperspot-e2e", "exp": now + 600},
+                    secret="wrong")
+    st, _ = _http("GET", url + "/simple-user-settings/v1/settings",
+                  token=tok)
+    assert st == 401
+    # wrong issuer rejected
+    tok = _mint_jwt({"sub": "x", "iss": "evil", "exp": now + 600})
+    st, _ = _http("GET", url + "/simple-user-settings/v1/settings",
+                  token=tok)
+    assert st == 401
