@@ -16,6 +16,7 @@
 #include "modules/api_gateway.h"
 #include "modules/llm_gateway.h"
 #include "modules/user_settings.h"
+#include "modules/file_parser.h"
 #include "modules/oagw.h"
 #include "modules/system_modules.h"
 #include "util/log.h"
@@ -53,6 +54,7 @@ int run_server(const Json& cfg, bool check_only) {
   registry.add(orch);
   registry.add(std::make_shared<LlmGatewayModule>());
   registry.add(std::make_shared<UserSettingsModule>());
+  registry.add(std::make_shared<FileParserModule>());
 
   ClientHub hub;
   bool cancel_flag = false;
@@ -110,7 +112,7 @@ int list_modules(const Json& cfg) {
                         "nodes-registry", "model-registry", "credstore",
                         "serverless-runtime", "oagw", "file-storage",
                         "module-orchestrator", "llm-gateway",
-                        "simple-user-settings"})
+                        "simple-user-settings", "file-parser"})
     std::cout << m << "\n";
   return 0;
 }

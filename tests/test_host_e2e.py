@@ -607,3 +607,43 @@ def test_jwt_hs256_auth(mt_server):
     st, _ = _http("GET", url + "/simple-user-settings/v1/settings",
                   token=tok)
     assert st == 401
+
+
+def test_file_parser(server):
+    url = BASE.format(server.port)
+    st, body = _http("GET", url + "/file-parser/v1/info")
+    assert st == 200 and "html" in json.loads(body)["backends"]
+    # raw upload html -> markdown
+    req = urllib.request.Request(
+        url + "/file-parser/v1/upload/markdown?filename=t.html",
+        method="POST",
+        data=b"<h1>Title</h1><p>Hello <b>world</b></p><ul><li>a</li>"
+             b"<li>b</li></ul>")
+    with urllib.request.urlopen(req, timeout=10) as r:
+        d = json.loads(r.read())
+    assert "# Title" in d["content"] and "- a" in d["content"], d
+    assert "**world**" in d["content"]
+    # parse-local on a temp txt file
+    import tempfile
+    f = tempfile.NamedTemporaryFile(suffix=".txt", delete=False)
+    f.write(b"local file contents")
+    f.close()
+    st, body = _http("POST", url + "/file-parser/v1/parse-local",
+                     {"path": f.name})
+    assert st == 200 and json.loads(body)["content"] == "local file contents"
+    # traversal rejected
+    st, _ = _http("POST", url + "/file-parser/v1/parse-local",
+                  {"path": "/etc/../etc/passwd"})
+    assert st == 403
+    # multipart upload csv -> markdown table
+    boundary = "XbOuNdArYx"
+    mp = (f"--{boundary}\r\ncontent-disposition: form-data; "
+          f'name="file"; filename="d.csv"\r\n\r\n'
+          f"a,b\r\n1,2\r\n--{boundary}--\r\n").encode()
+    req = urllib.request.Request(
+        url + "/file-parser/v1/upload/markdown", method="POST", data=mp)
+    req.add_header("content-type",
+                   f"multipart/form-data; boundary={boundary}")
+    with urllib.request.urlopen(req, timeout=10) as r:
+        d = json.loads(r.read())
+    assert "| a | b |" in d["content"], d
