@@ -2,9 +2,13 @@
 
 #include "../modkit/auth.h"
 
-#include <sys/sysinfo.h>
+#include <signal.h>
 #include <sys/stat.h>
+#include <sys/sysinfo.h>
+#include <sys/wait.h>
 #include <unistd.h>
+
+#include <chrono>
 
 #include <cstring>
 #include <fstream>
@@ -639,6 +643,106 @@ void FileStorageModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
 
 // ---------------------------------------------------- module-orchestrator
 
+static double mono_s() {
+  return std::chrono::duration<double>(
+             std::chrono::steady_clock::now().time_since_epoch()).count();
+}
+
+void ModuleOrchestratorModule::init(ModuleCtx& ctx) {
+  // collect modules declared runtime.type: oop (A.6 envelope)
+  const Json& mods = ctx.full_config.at("modules");
+  if (mods.is_object()) {
+    for (auto& [mname, mcfg] : mods.obj()) {
+      if (mcfg.path("runtime.type").as_string() != "oop") continue;
+      const Json& ex = mcfg.path("runtime.execution");
+      OopSpec sp;
+      sp.name = mname;
+      sp.exe = ex.at("executable_path").as_string();
+      if (ex.at("args").is_array())
+        for (auto& a : ex.at("args").arr())
+          sp.args.push_back(a.as_string());
+      if (ex.at("environment").is_object())
+        for (auto& [k, v] : ex.at("environment").obj())
+          sp.env[k] = v.as_string();
+      sp.config = mcfg.at("config");
+      oop_specs_.push_back(std::move(sp));
+    }
+  }
+  const std::string bind = ctx.full_config
+      .path("modules.api-gateway.config.bind_addr")
+      .as_string("127.0.0.1:8087");
+  directory_endpoint_ = "http://" + bind + "/module-orchestrator/v1";
+}
+
+void ModuleOrchestratorModule::start(ModuleCtx& ctx) {
+  for (auto& sp : oop_specs_) {
+    int pfd[2];
+    if (pipe(pfd) != 0) continue;
+    pid_t pid = fork();
+    if (pid == 0) {
+      setpgid(0, 0);
+      dup2(pfd[1], 1);
+      dup2(pfd[1], 2);
+      close(pfd[0]);
+      close(pfd[1]);
+      // reference host_runtime.rs:56-59 env contract
+      setenv("MODKIT_MODULE_CONFIG", sp.config.dump().c_str(), 1);
+      setenv("MODKIT_DIRECTORY_ENDPOINT", directory_endpoint_.c_str(), 1);
+      setenv("MODKIT_MODULE_NAME", sp.name.c_str(), 1);
+      for (auto& [k, v] : sp.env) setenv(k.c_str(), v.c_str(), 1);
+      std::vector<std::string> args = {sp.exe};
+      for (auto& a : sp.args) args.push_back(a);
+      std::vector<char*> argv;
+      for (auto& a : args) argv.push_back(const_cast<char*>(a.c_str()));
+      argv.push_back(nullptr);
+      execvp(argv[0], argv.data());
+      _exit(127);
+    }
+    close(pfd[1]);
+    Child c;
+    c.name = sp.name;
+    c.pid = pid;
+    c.out_fd = pfd[0];
+    // log forwarder: child stdout/stderr lines -> host log
+    c.fwd = std::thread([fd = pfd[0], mname = sp.name] {
+      std::string buf;
+      char tmp[4096];
+      while (true) {
+        ssize_t r = read(fd, tmp, sizeof tmp);
+        if (r <= 0) break;
+        buf.append(tmp, (size_t)r);
+        size_t nl;
+        while ((nl = buf.find('\n')) != std::string::npos) {
+          LOG_INFO(("oop:" + mname).c_str(), "%s",
+                   buf.substr(0, nl).c_str());
+          buf.erase(0, nl + 1);
+        }
+      }
+      close(fd);
+    });
+    children_.push_back(std::move(c));
+    LOG_INFO("module-orchestrator", "spawned oop module %s pid=%d",
+             sp.name.c_str(), pid);
+  }
+}
+
+void ModuleOrchestratorModule::stop(ModuleCtx& ctx) {
+  for (auto& c : children_) {
+    if (c.pid > 0) {
+      kill(c.pid, SIGTERM);
+      int st = 0;
+      for (int i = 0; i < 30 && waitpid(c.pid, &st, WNOHANG) == 0; ++i)
+        usleep(100000);
+      if (waitpid(c.pid, &st, WNOHANG) == 0) {
+        kill(c.pid, SIGKILL);
+        waitpid(c.pid, &st, 0);
+      }
+    }
+    if (c.fwd.joinable()) c.fwd.join();
+  }
+  children_.clear();
+}
+
 void ModuleOrchestratorModule::register_rest(ModuleCtx& ctx,
                                              RestRegistry& rest) {
   OperationSpec list;
@@ -656,10 +760,118 @@ void ModuleOrchestratorModule::register_rest(ModuleCtx& ctx,
       m["status"] = "running";
       items.push_back(m);
     }
+    for (auto& c : children_) {
+      Json m = Json::object();
+      m["name"] = c.name;
+      int st = 0;
+      m["status"] = (c.pid > 0 && waitpid(c.pid, &st, WNOHANG) == 0)
+                        ? "running" : "exited";
+      m["runtime"] = "oop";
+      m["pid"] = (long)c.pid;
+      items.push_back(m);
+    }
     Json out = Json::object();
     out["items"] = items;
     w.respond(200, "application/json", out.dump());
   });
+
+  // DirectoryService semantics over REST (proto/directory/v1:
+  // RegisterInstance / Heartbeat / Resolve / List)
+  {
+    OperationSpec op;
+    op.method = "POST";
+    op.path = "/module-orchestrator/v1/instances/register";
+    op.operation_id = "register_instance";
+    op.summary = "Register an OoP module instance";
+    op.authenticated = true;
+    op.allowed_content_types = {"application/json"};
+    op.tags = {"module-orchestrator"};
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      Json body;
+      try { body = Json::parse(rq.body); }
+      catch (...) { throw Problem::bad_request("invalid JSON body"); }
+      const std::string name = body.at("name").as_string();
+      if (name.empty()) throw Problem::bad_request("'name' is required");
+      std::lock_guard<std::mutex> lk(inst_mu_);
+      Instance in;
+      in.id = name + "-" + std::to_string(++inst_ctr_);
+      in.name = name;
+      in.endpoint = body.at("endpoint").as_string();
+      in.meta = body.at("meta");
+      in.last_heartbeat = mono_s();
+      instances_[in.id] = in;
+      Json out = Json::object();
+      out["id"] = in.id;
+      w.respond(200, "application/json", out.dump());
+    });
+  }
+  {
+    OperationSpec op;
+    op.method = "POST";
+    op.path = "/module-orchestrator/v1/instances/{id}/heartbeat";
+    op.operation_id = "instance_heartbeat";
+    op.summary = "Instance liveness heartbeat";
+    op.authenticated = true;
+    op.tags = {"module-orchestrator"};
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      std::lock_guard<std::mutex> lk(inst_mu_);
+      auto it = instances_.find(rq.path_params.at("id"));
+      if (it == instances_.end())
+        throw Problem::not_found("no such instance");
+      it->second.last_heartbeat = mono_s();
+      w.respond(204, "application/json", "");
+    });
+  }
+  {
+    OperationSpec op;
+    op.method = "GET";
+    op.path = "/module-orchestrator/v1/instances";
+    op.operation_id = "list_instances";
+    op.summary = "Registered instances with liveness";
+    op.authenticated = true;
+    op.tags = {"module-orchestrator"};
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      Json items = Json::array();
+      const double now = mono_s();
+      std::lock_guard<std::mutex> lk(inst_mu_);
+      for (auto& [id, in] : instances_) {
+        Json m = Json::object();
+        m["id"] = id;
+        m["name"] = in.name;
+        m["endpoint"] = in.endpoint;
+        m["alive"] = (now - in.last_heartbeat) < 15.0;
+        if (!in.meta.is_null()) m["meta"] = in.meta;
+        items.push_back(m);
+      }
+      Json out = Json::object();
+      out["items"] = items;
+      w.respond(200, "application/json", out.dump());
+    });
+  }
+  {
+    OperationSpec op;
+    op.method = "GET";
+    op.path = "/module-orchestrator/v1/instances/resolve/{name}";
+    op.operation_id = "resolve_instance";
+    op.summary = "Resolve a live instance endpoint by module name";
+    op.authenticated = true;
+    op.tags = {"module-orchestrator"};
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      const std::string name = rq.path_params.at("name");
+      const double now = mono_s();
+      std::lock_guard<std::mutex> lk(inst_mu_);
+      const Instance* best = nullptr;
+      for (auto& [id, in] : instances_)
+        if (in.name == name && (now - in.last_heartbeat) < 15.0 &&
+            (!best || in.last_heartbeat > best->last_heartbeat))
+          best = &in;
+      if (!best) throw Problem::not_found("no live instance of " + name);
+      Json out = Json::object();
+      out["id"] = best->id;
+      out["endpoint"] = best->endpoint;
+      w.respond(200, "application/json", out.dump());
+    });
+  }
 }
 
 }  // namespace hs

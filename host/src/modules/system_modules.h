@@ -8,6 +8,7 @@
 #pragma once
 
 #include <mutex>
+#include <thread>
 
 #include "../modkit/modkit.h"
 #include "api_gateway.h"
@@ -145,18 +146,51 @@ class FileStorageModule : public Module {
 
 // ---- module-orchestrator (module/worker instance visibility;
 //      reference modules/system/module-orchestrator DirectoryService) ----
+// module-orchestrator: module listing + the OoP (out-of-process) module
+// runtime — spawn children declared as modules.<name>.runtime.type: oop
+// (reference A.6 envelope: execution.{executable_path,args,environment}),
+// pass config by env (MODKIT_MODULE_CONFIG) and the directory endpoint
+// (MODKIT_DIRECTORY_ENDPOINT), forward child stdout into the host log
+// (backends/log_forwarder.rs), track register/heartbeat liveness
+// (proto/directory/v1 semantics over REST).
 class ModuleOrchestratorModule : public Module {
  public:
   std::string name() const override { return "module-orchestrator"; }
+  bool is_stateful() const override { return true; }
+  void init(ModuleCtx& ctx) override;
+  void start(ModuleCtx& ctx) override;
+  void stop(ModuleCtx& ctx) override;
   void register_rest(ModuleCtx& ctx, RestRegistry& rest) override;
-  void post_init(ModuleCtx& ctx) override { registry_size_ = 0; }
   void set_modules(std::vector<std::string> names) {
     module_names_ = std::move(names);
   }
 
  private:
+  struct OopSpec {
+    std::string name, exe;
+    std::vector<std::string> args;
+    std::map<std::string, std::string> env;
+    Json config;
+  };
+  struct Child {
+    std::string name;
+    pid_t pid = -1;
+    int out_fd = -1;
+    std::thread fwd;
+  };
+  struct Instance {
+    std::string id, name, endpoint;
+    double last_heartbeat = 0;
+    Json meta;
+  };
+
   std::vector<std::string> module_names_;
-  size_t registry_size_ = 0;
+  std::vector<OopSpec> oop_specs_;
+  std::vector<Child> children_;
+  std::string directory_endpoint_;
+  std::mutex inst_mu_;
+  std::map<std::string, Instance> instances_;
+  uint64_t inst_ctr_ = 0;
 };
 
 }  // namespace hs

@@ -647,3 +647,94 @@ def test_file_parser(server):
     with urllib.request.urlopen(req, timeout=10) as r:
         d = json.loads(r.read())
     assert "| a | b |" in d["content"], d
+
+
+@pytest.fixture(scope="module")
+def oop_server(tmp_path_factory):
+    """Server with one OoP child module (A.6 runtime envelope)."""
+    import tempfile
+    port = _free_port()
+    tmp = tmp_path_factory.mktemp("oop")
+    child = tmp / "child.py"
+    child.write_text("""
+import json, os, time, urllib.request
+cfg = json.loads(os.environ["MODKIT_MODULE_CONFIG"])
+base = os.environ["MODKIT_DIRECTORY_ENDPOINT"]
+name = os.environ["MODKIT_MODULE_NAME"]
+print("child starting with marker", cfg["marker"], flush=True)
+req = urllib.request.Request(base + "/instances/register", method="POST",
+    data=json.dumps({"name": name, "endpoint": "uds:///tmp/x.sock",
+                     "meta": {"marker": cfg["marker"]}}).encode(),
+    headers={"content-type": "application/json"})
+iid = json.loads(urllib.request.urlopen(req, timeout=5).read())["id"]
+while True:
+    urllib.request.urlopen(urllib.request.Request(
+        base + f"/instances/{iid}/heartbeat", method="POST", data=b""),
+        timeout=5)
+    time.sleep(0.5)
+""")
+    cfg = f"""
+server:
+  home_dir: "/tmp/hs-e2e-oop"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+  llm-gateway:
+    config:
+      auto_start_worker: false
+  demo-oop:
+    config:
+      marker: "xyzzy42"
+    runtime:
+      type: oop
+      execution:
+        executable_path: "python3"
+        args: ["{child}"]
+"""
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(cfg)
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        yield srv
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
+
+
+def test_oop_spawn_register_heartbeat(oop_server):
+    url = BASE.format(oop_server.port)
+    # child registers within a few seconds and stays alive via heartbeats
+    inst = None
+    for _ in range(50):
+        st, body = _http("GET", url + "/module-orchestrator/v1/instances")
+        items = json.loads(body)["items"]
+        live = [i for i in items if i["name"] == "demo-oop" and i["alive"]]
+        if live:
+            inst = live[0]
+            break
+        time.sleep(0.3)
+    assert inst, "oop child never registered"
+    assert inst["meta"]["marker"] == "xyzzy42"  # MODKIT_MODULE_CONFIG flowed
+    # resolve by name
+    st, body = _http("GET",
+                     url + "/module-orchestrator/v1/instances/resolve/demo-oop")
+    assert st == 200 and json.loads(body)["endpoint"].startswith("uds://")
+    # child pid listed as running in the module list
+    st, body = _http("GET", url + "/module-orchestrator/v1/modules")
+    mods = json.loads(body)["items"]
+    oop = [m for m in mods if m.get("runtime") == "oop"][0]
+    assert oop["status"] == "running" and oop["pid"] > 0
+    oop_server._oop_pid = oop["pid"]
+
+
+def test_oop_child_terminated_on_stop(oop_server):
+    # relies on ordering: runs after the register test in the same module
+    pid = getattr(oop_server, "_oop_pid", None)
+    assert pid
+    os.kill(pid, 0)  # alive now; fixture teardown must reap it
