@@ -81,6 +81,11 @@ void ResponseWriter::respond(
   h << "HTTP/1.1 " << status << " " << status_text(status) << "\r\n"
     << "content-type: " << content_type << "\r\n"
     << "content-length: " << body.size() << "\r\n";
+  for (auto& [k, v] : default_headers) {
+    bool dup = false;
+    for (auto& [k2, v2] : extra) if (k2 == k) dup = true;
+    if (!dup) h << k << ": " << v << "\r\n";
+  }
   for (auto& [k, v] : extra) h << k << ": " << v << "\r\n";
   h << "connection: " << (keep_alive ? "keep-alive" : "close") << "\r\n\r\n";
   std::string head = h.str();
@@ -99,6 +104,11 @@ void ResponseWriter::begin_stream(
     << "content-type: " << content_type << "\r\n"
     << "cache-control: no-cache\r\n"
     << "transfer-encoding: chunked\r\n";
+  for (auto& [k, v] : default_headers) {
+    bool dup = false;
+    for (auto& [k2, v2] : extra) if (k2 == k) dup = true;
+    if (!dup) h << k << ": " << v << "\r\n";
+  }
   for (auto& [k, v] : extra) h << k << ": " << v << "\r\n";
   h << "connection: close\r\n\r\n";
   keep_alive = false;

@@ -56,6 +56,8 @@ class ResponseWriter {
 
   bool started() const { return started_; }
   bool keep_alive = true;
+  // headers stamped onto every response (request id, traceparent)
+  std::vector<std::pair<std::string, std::string>> default_headers;
 
  private:
   bool send_all(const char* p, size_t n);

@@ -139,6 +139,33 @@ void ApiGatewayModule::handle(HttpRequest& req, ResponseWriter& w) {
   if (req.request_id.empty()) req.request_id = gen_request_id();
   req_counter_++;
 
+  // 1b. W3C trace context: parse inbound traceparent
+  // (00-<trace32>-<span16>-<flags>), mint this hop's span, expose both to
+  // handlers (extensions.trace) and echo traceparent downstream — the
+  // reference's http_request span parenting (api-gateway module.rs:274-330)
+  {
+    std::string tp = req.header("traceparent");
+    std::string trace_id, parent_span;
+    if (tp.size() >= 55 && tp[2] == '-' && tp[35] == '-' && tp[52] == '-') {
+      trace_id = tp.substr(3, 32);
+      parent_span = tp.substr(36, 16);
+    }
+    auto hex = [](uint64_t v, int w) {
+      char b[32];
+      snprintf(b, sizeof b, "%0*llx", w, (unsigned long long)v);
+      return std::string(b);
+    };
+    static std::atomic<uint64_t> span_ctr{0x1a2b};
+    uint64_t sp = (uint64_t)time(nullptr) << 20 ^ (++span_ctr * 0x9e3779b9);
+    if (trace_id.empty())
+      trace_id = hex(sp ^ 0xdeadbeefcafe1234ull, 16) + hex(sp * 31, 16);
+    Json tr = Json::object();
+    tr["trace_id"] = trace_id;
+    tr["span_id"] = hex(sp, 16);
+    if (!parent_span.empty()) tr["parent_span_id"] = parent_span;
+    req.extensions["trace"] = tr;
+  }
+
   // CORS preflight short-circuit + response headers
   std::vector<std::pair<std::string, std::string>> cors_headers;
   if (cors_enabled_) {
@@ -160,15 +187,23 @@ void ApiGatewayModule::handle(HttpRequest& req, ResponseWriter& w) {
   dispatch(req, w);
   auto dt = std::chrono::duration<double, std::milli>(
       std::chrono::steady_clock::now() - t0).count();
-  LOG_DEBUG("http", "%s %s -> done in %.2fms rid=%s", req.method.c_str(),
-            req.path.c_str(), dt, req.request_id.c_str());
+  // http_request span record (OTel semantic fields, log-exported)
+  LOG_DEBUG("http", "%s %s -> done in %.2fms rid=%s trace=%s span=%s",
+            req.method.c_str(), req.path.c_str(), dt,
+            req.request_id.c_str(),
+            req.extensions.path("trace.trace_id").as_string().c_str(),
+            req.extensions.path("trace.span_id").as_string().c_str());
   (void)status;
 }
 
 void ApiGatewayModule::dispatch(HttpRequest& req, ResponseWriter& w) {
   const std::string rid_hdr = req.request_id;
+  const std::string tp_hdr =
+      "00-" + req.extensions.path("trace.trace_id").as_string() + "-" +
+      req.extensions.path("trace.span_id").as_string() + "-01";
   const std::vector<std::pair<std::string, std::string>> rid_headers = {
-      {"x-request-id", rid_hdr}};
+      {"x-request-id", rid_hdr}, {"traceparent", tp_hdr}};
+  w.default_headers = rid_headers;
 
   // built-ins outside the registry: docs + openapi
   if (req.method == "GET" && req.path == "/openapi.json") {

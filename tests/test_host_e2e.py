@@ -738,3 +738,15 @@ def test_oop_child_terminated_on_stop(oop_server):
     pid = getattr(oop_server, "_oop_pid", None)
     assert pid
     os.kill(pid, 0)  # alive now; fixture teardown must reap it
+
+
+def test_traceparent_propagation(server):
+    url = BASE.format(server.port) + "/healthz"
+    req = urllib.request.Request(url)
+    req.add_header("traceparent",
+                   "00-0123456789abcdef0123456789abcdef-00f067aa0ba902b7-01")
+    with urllib.request.urlopen(req, timeout=5) as r:
+        tp = r.headers.get("traceparent", "")
+    assert tp.startswith("00-0123456789abcdef0123456789abcdef-"), tp
+    # a fresh span id was minted for this hop
+    assert tp.split("-")[2] != "00f067aa0ba902b7"
