@@ -6,18 +6,23 @@
 // counts live in device memory and never reach the host.
 //
 //   moe_align:   top-k pair ids -> per-expert sorted order, padded to
-//                BM=32-row tiles; emits sorted_ids (pair idx | -1 pad),
+//                MOE_BM-row tiles; emits sorted_ids (pair idx | -1 pad),
 //                tile_expert (expert per 32-row tile | -1 unused) and
 //                inv_pos (pair -> sorted position) in ONE workgroup.
 //   moe_gemm:    out[p, n] = x[row(p)] @ W[expert(tile(p))]^T over the
-//                sorted space; 32x128 tile per workgroup, 4 waves, LDS
-//                XOR-swizzled A/B, v_mfma_f32_32x32x16_bf16 (fragment
-//                conventions validated by csrc/attn_prefill_mfma.hip).
+//                sorted space; 128x128 tile per workgroup (8 waves, each
+//                32x64 via 2x v_mfma_f32_32x32x16_bf16 per k-chunk), LDS
+//                XOR-swizzled A/B.  BM=128 is a WEIGHT-TRAFFIC choice:
+//                with 32-row tiles each expert's [N,K] matrix was
+//                re-streamed once per m-tile (~9x at batch 1024 -> the
+//                measured 139 ms Mixtral step); 128-row tiles cut the
+//                re-read to ceil(tokens_e/128) at the price of padding
+//                compute, and padding is cheap where HBM is the bound.
 //   moe_combine: out[t] = sum_k w[t,k] * y[inv_pos[t*K+k]]  (gather, no
 //                atomics).
 #include "common.h"
 
-#define MOE_BM 32   // sorted rows per expert tile (the align granularity)
+#define MOE_BM 128  // sorted rows per expert tile (the align granularity)
 
 // ---------------------------------------------------------------- align
 __global__ __launch_bounds__(256) void moe_align_kernel(
@@ -76,7 +81,7 @@ DEV int swz(int row, int byte_in_row) {
 // gather_div > 0: A row p comes from x[sorted_ids[p] / gather_div]
 // gather_div == 0: A row p is x[p] (identity; padded rows are junk and the
 // combine step drops them).
-__global__ __launch_bounds__(256) void moe_gemm_kernel(
+__global__ __launch_bounds__(512) void moe_gemm_kernel(
     bf16* __restrict__ out, const bf16* __restrict__ x,
     const bf16* __restrict__ w, const int* __restrict__ sorted_ids,
     const int* __restrict__ tile_expert, int N, int K, int gather_div) {
@@ -85,7 +90,10 @@ __global__ __launch_bounds__(256) void moe_gemm_kernel(
   const int m0 = blockIdx.x * MOE_BM;
   const int n0 = blockIdx.y * 128;
 
-  __shared__ __attribute__((aligned(16))) unsigned char As[MOE_BM * 128];
+  // single-buffered tiles: a 2-deep LDS ring (64 KiB) was measured 10%
+  // SLOWER here — it halves WG residency (guide: explicit dbuf at HIP
+  // source is not a lever; occupancy is)
+  __shared__ __attribute__((aligned(16))) unsigned char As[128 * 128];
   __shared__ __attribute__((aligned(16))) unsigned char Bs[128 * 128];
 
   const int tid = threadIdx.x;
@@ -93,9 +101,13 @@ __global__ __launch_bounds__(256) void moe_gemm_kernel(
   const int lane = tid & 63;
   const int lcol = lane & 31;
   const int lhalf = lane >> 5;
+  const int wave_m = wid & 3;   // 4 row-blocks of 32
+  const int wave_n = wid >> 2;  // 2 col-blocks of 64
 
-  // A loader role: one 16-B chunk per thread (32 rows x 8 chunks)
-  const int ar = tid >> 3, ac = tid & 7;
+  // loader role: thread covers 32 B of one row (4 threads per 128-row tile
+  // side, 64-elem K slice = 128 B per row)
+  const int ar = tid >> 2;        // 0..127
+  const int ac32 = (tid & 3) * 32;  // byte offset of this thread's 32 B
   int a_src = -1;
   if (gather_div > 0) {
     const int pair = sorted_ids[m0 + ar];
@@ -104,46 +116,51 @@ __global__ __launch_bounds__(256) void moe_gemm_kernel(
     a_src = m0 + ar;
   }
 
-  f32x16 acc;
+  f32x16 acc[2];
   #pragma unroll
-  for (int i = 0; i < 16; ++i) acc[i] = 0.f;
+  for (int nb = 0; nb < 2; ++nb)
+    #pragma unroll
+    for (int i = 0; i < 16; ++i) acc[nb][i] = 0.f;
 
+  const bf16* ap0 = a_src >= 0 ? x + (long)a_src * K + ac32 / 2 : nullptr;
+  const bf16* bp0 = w + ((long)e * N + n0 + ar) * K + ac32 / 2;
   for (int k0 = 0; k0 < K; k0 += 64) {
-    // ---- stage A (32x64) and B (128x64) ----
-    {
-      uint4 av = uint4{0, 0, 0, 0};
-      if (a_src >= 0)
-        av = *reinterpret_cast<const uint4*>(x + (long)a_src * K + k0
-                                             + ac * 8);
-      *reinterpret_cast<uint4*>(&As[swz(ar, ac * 16)]) = av;
-      #pragma unroll
-      for (int c = 0; c < 4; ++c) {
-        const int lin = tid * 4 + c;
-        const int br = lin >> 3, bc = lin & 7;
-        uint4 bv = *reinterpret_cast<const uint4*>(
-            w + ((long)e * N + n0 + br) * K + k0 + bc * 8);
-        *reinterpret_cast<uint4*>(&Bs[swz(br, bc * 16)]) = bv;
-      }
+    uint4 av0 = uint4{0, 0, 0, 0}, av1 = uint4{0, 0, 0, 0};
+    if (ap0) {
+      av0 = *reinterpret_cast<const uint4*>(ap0 + k0);
+      av1 = *reinterpret_cast<const uint4*>(ap0 + k0 + 8);
     }
+    uint4 bv0 = *reinterpret_cast<const uint4*>(bp0 + k0);
+    uint4 bv1 = *reinterpret_cast<const uint4*>(bp0 + k0 + 8);
+    *reinterpret_cast<uint4*>(&As[swz(ar, ac32)]) = av0;
+    *reinterpret_cast<uint4*>(&As[swz(ar, ac32 + 16)]) = av1;
+    *reinterpret_cast<uint4*>(&Bs[swz(ar, ac32)]) = bv0;
+    *reinterpret_cast<uint4*>(&Bs[swz(ar, ac32 + 16)]) = bv1;
     __syncthreads();
-    // ---- 4 MFMAs: C(32x32) += A(32x16) * B(16x32) per k-chunk ----
     #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
       short8 af = *reinterpret_cast<const short8*>(
-          &As[swz(lcol, kk * 32 + lhalf * 16)]);
-      short8 bf = *reinterpret_cast<const short8*>(
-          &Bs[swz(wid * 32 + lcol, kk * 32 + lhalf * 16)]);
-      acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, acc, 0, 0, 0);
+          &As[swz(wave_m * 32 + lcol, kk * 32 + lhalf * 16)]);
+      #pragma unroll
+      for (int nb = 0; nb < 2; ++nb) {
+        short8 bf = *reinterpret_cast<const short8*>(
+            &Bs[swz(wave_n * 64 + nb * 32 + lcol,
+                    kk * 32 + lhalf * 16)]);
+        acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, acc[nb],
+                                                          0, 0, 0);
+      }
     }
     __syncthreads();
   }
   // ---- epilogue: C row = (reg&3) + 8*(reg>>2) + 4*lhalf, col = lcol ----
   #pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    const int m = (r & 3) + 8 * (r >> 2) + 4 * lhalf;
-    *(unsigned short*)(out + (long)(m0 + m) * N + n0 + wid * 32 + lcol) =
-        f2bf(acc[r]);
-  }
+  for (int nb = 0; nb < 2; ++nb)
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int m = (r & 3) + 8 * (r >> 2) + 4 * lhalf;
+      *(unsigned short*)(out + (long)(m0 + wave_m * 32 + m) * N + n0 +
+                         wave_n * 64 + nb * 32 + lcol) = f2bf(acc[nb][r]);
+    }
 }
 
 // ------------------------------------------------------------- combine
@@ -188,7 +205,7 @@ void launch_moe_gemm(bf16* out, const bf16* x, const bf16* w,
   if (N % 128 || K % 64)
     throw std::runtime_error("moe_gemm: N%128 or K%64 != 0");
   dim3 grid((unsigned)ntiles_max, (unsigned)(N / 128));
-  moe_gemm_kernel<<<grid, 256, 0, stream>>>(out, x, w, sorted_ids,
+  moe_gemm_kernel<<<grid, 512, 0, stream>>>(out, x, w, sorted_ids,
                                             tile_expert, N, K, gather_div);
 }
 
