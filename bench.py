@@ -39,7 +39,7 @@ def main():
     ap.add_argument("--model", default="llama3-8b")
     ap.add_argument("--tp", type=int, default=1,
                     help=">1: one tensor-parallel engine over all ranks")
-    ap.add_argument("--batch", type=int, default=1024)
+    ap.add_argument("--batch", type=int, default=2048)
     ap.add_argument("--prompt-len", type=int, default=512)
     ap.add_argument("--eager", action="store_true")
     ap.add_argument("--quant", default=None, choices=[None, "fp8"],
