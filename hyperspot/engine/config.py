@@ -114,7 +114,7 @@ class EngineConfig:
     seed: int = 0
     # decode hipGraph capture batch buckets (padded up to nearest)
     graph_batch_sizes: tuple = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128,
-                                192, 256, 384, 512, 768, 1024)
+                                192, 256, 384, 512, 768, 1024, 1536, 2048)
 
     def spec(self) -> ModelSpec:
         return get_model_spec(self.model)

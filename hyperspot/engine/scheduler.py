@@ -77,7 +77,13 @@ class Scheduler:
             if len(self.running) + len(batch.requests) >= self.config.max_num_seqs:
                 break
             if n > budget:
-                break
+                # a prompt longer than the per-step token budget would
+                # starve under whole-prompt admission: let it run ALONE
+                # (the MFMA prefill kernel handles any seqlen)
+                if batch.empty and n <= self.config.max_model_len:
+                    budget = n
+                else:
+                    break
             if not self.bm.can_allocate(n, self.watermark):
                 break
             self.waiting.popleft()

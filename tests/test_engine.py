@@ -146,3 +146,17 @@ def test_moe_engine_runs():
     out2 = eng.generate([[1, 2, 3, 4]], SamplingParams(temperature=0.0,
                                                        max_tokens=5))[0]
     assert out == out2
+
+
+def test_long_prompt_exceeding_step_budget_is_not_starved():
+    """A prompt longer than max_num_batched_tokens must still be served
+    (scheduled alone), not wait forever."""
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    cfg = EngineConfig(model="tiny-llama", max_num_seqs=4,
+                       max_num_batched_tokens=64, max_model_len=512,
+                       num_gpu_blocks=64, enforce_eager=True)
+    eng = LLMEngine(cfg, device="cpu")
+    long_prompt = list(range(1, 200))          # 199 tokens > 64 budget
+    out = eng.generate([long_prompt, [1, 2, 3]],
+                       SamplingParams(temperature=0.0, max_tokens=4))
+    assert len(out[0]) == 4 and len(out[1]) == 4
