@@ -45,6 +45,9 @@ def main():
     ap.add_argument("--quant", default=None, choices=[None, "fp8"],
                     help="fp8: e4m3fn weights + per-token act scales "
                          "(separate config line, NOT the bf16 headline)")
+    ap.add_argument("--kv-dtype", default="bfloat16",
+                    choices=["bfloat16", "fp8"],
+                    help="paged KV cache dtype (fp8 halves attention HBM)")
     ap.add_argument("--device", default=None)
     args = ap.parse_args()
 
@@ -71,7 +74,8 @@ def main():
         model=args.model, max_num_seqs=args.batch,
         max_num_batched_tokens=max(8192, args.prompt_len),
         max_model_len=max_len, enforce_eager=args.eager or not on_gpu,
-        tp_size=tp, quant=args.quant, seed=1234 + (0 if tp > 1 else rank))
+        tp_size=tp, quant=args.quant, kv_dtype=args.kv_dtype,
+        seed=1234 + (0 if tp > 1 else rank))
     t0 = time.monotonic()
     eng = LLMEngine(cfg, device=args.device)
     log(rank, f"engine init {time.monotonic() - t0:.1f}s "
@@ -152,7 +156,9 @@ def main():
             "higher_is_better": True,
             "scaling": "strong" if tp > 1 else "weak",
             "vs_baseline": None,
-            "dtype": args.quant or "bf16" if on_gpu else "fp32(cpu-dev-run)",
+            "dtype": (((args.quant or "bf16")
+                       + ("+fp8kv" if args.kv_dtype == "fp8" else ""))
+                      if on_gpu else "fp32(cpu-dev-run)"),
             "data": "synthetic",
             "ttft_ms_p50": round(ttft_p50, 2),
             "config": {

@@ -242,12 +242,22 @@ DEV float dot2_bf16(unsigned a, unsigned b, float c) {
   return c;
 }
 
+// fp8-KV support: unpack 4 e4m3fn bytes -> 2 packed-bf16 dwords (for the
+// dot2 K path) or 2 float2 (for the PV path).  gfx950-native converts.
+typedef __attribute__((ext_vector_type(2))) float float2v;
+
+DEV unsigned pk_bf16(float a, float b) {
+  unsigned r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(a), "v"(b));
+  return r;
+}
+
 // SPLIT > 1 (flash-decode): grid.z WGs share one (row, kvh), each owning
 // pages [4*z+wid :: 4*SPLIT]; per-WG partials (m, l, o) land in `partial`
 // [rows, kvh, SPLIT, GROUP, D+2] fp32 and a second kernel merges them.
 // Used when rows*kvh alone cannot fill the 256 CUs (small batch, or the
 // kvh=1 GQA shard of Llama-70B at TP=8).
-template <int GROUP>
+template <int GROUP, bool KV8>
 __global__ __launch_bounds__(256) void paged_attn_v3_kernel(
     bf16* __restrict__ out, const bf16* __restrict__ q,
     const bf16* __restrict__ k_cache, const bf16* __restrict__ v_cache,
@@ -307,26 +317,57 @@ __global__ __launch_bounds__(256) void paged_attn_v3_kernel(
   for (int b = 4 * blockIdx.z + wid; b < nblocks; b += 4 * split) {
     const long blk = bt[b];
     const int nb = min(ATTN_BS, ctx - b * ATTN_BS);
-    const bf16* pbase = k_cache + ((blk * num_kv_heads + kvh) * ATTN_BS) * D;
-    const bf16* vbase = v_cache + ((blk * num_kv_heads + kvh) * ATTN_BS) * D;
+    const long page_off = ((blk * num_kv_heads + kvh) * ATTN_BS) * D;
+    const bf16* pbase = k_cache + page_off;
+    const bf16* vbase = v_cache + page_off;
+    const unsigned char* pbase8 =
+        reinterpret_cast<const unsigned char*>(k_cache) + page_off;
+    const unsigned char* vbase8 =
+        reinterpret_cast<const unsigned char*>(v_cache) + page_off;
     // ---- issue every load for this page up front (one round trip) ----
-    uint4 kr[2][2], vr[4];
+    uint4 kr[2][KV8 ? 1 : 2];
+    uint4 vr4[KV8 ? 1 : 4];
+    uint2 vr2[KV8 ? 4 : 1];
     #pragma unroll
     for (int p = 0; p < 2; ++p) {
-      const bf16* kp = pbase + (p * 8 + t_loc) * D + sl * 16;
-      kr[p][0] = *reinterpret_cast<const uint4*>(kp);
-      kr[p][1] = *reinterpret_cast<const uint4*>(kp + 8);
+      if constexpr (KV8) {
+        kr[p][0] = *reinterpret_cast<const uint4*>(
+            pbase8 + (p * 8 + t_loc) * D + sl * 16);
+      } else {
+        const bf16* kp = pbase + (p * 8 + t_loc) * D + sl * 16;
+        kr[p][0] = *reinterpret_cast<const uint4*>(kp);
+        kr[p][1] = *reinterpret_cast<const uint4*>(kp + 8);
+      }
     }
     #pragma unroll
-    for (int i = 0; i < 4; ++i)
-      vr[i] = *reinterpret_cast<const uint4*>(vbase + (4 * g + i) * D
-                                              + ds * 8);
+    for (int i = 0; i < 4; ++i) {
+      if constexpr (KV8)
+        vr2[i] = *reinterpret_cast<const uint2*>(vbase8 + (4 * g + i) * D
+                                                 + ds * 8);
+      else
+        vr4[i] = *reinterpret_cast<const uint4*>(vbase + (4 * g + i) * D
+                                                 + ds * 8);
+    }
     // ---- scores: dot2 over the lane's 16-elem slice, then all-reduce
     // over the 8 slice lanes (bits 0..2) ----
     float s[GROUP][2];
     #pragma unroll
     for (int p = 0; p < 2; ++p) {
-      const unsigned* kw = reinterpret_cast<const unsigned*>(&kr[p][0]);
+      unsigned kw[8];
+      if constexpr (KV8) {
+        const unsigned* kq = reinterpret_cast<const unsigned*>(&kr[p][0]);
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          float2v lo = __builtin_amdgcn_cvt_pk_f32_fp8(kq[i], false);
+          float2v hi = __builtin_amdgcn_cvt_pk_f32_fp8(kq[i], true);
+          kw[2 * i] = pk_bf16(lo.x, lo.y);
+          kw[2 * i + 1] = pk_bf16(hi.x, hi.y);
+        }
+      } else {
+        #pragma unroll
+        for (int j = 0; j < 8; ++j)
+          kw[j] = reinterpret_cast<const unsigned*>(&kr[p][0])[j];
+      }
       #pragma unroll
       for (int h = 0; h < GROUP; ++h) {
         float d = 0.f;
@@ -373,12 +414,23 @@ __global__ __launch_bounds__(256) void paged_attn_v3_kernel(
       p4[h] = *reinterpret_cast<const float4*>(&sc[wid][h][4 * g]);
     #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      const unsigned* vw = reinterpret_cast<const unsigned*>(&vr[i]);
       float2 vf[4];
-      #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        vf[j].x = bf2f((unsigned short)(vw[j] & 0xffff));
-        vf[j].y = bf2f((unsigned short)(vw[j] >> 16));
+      if constexpr (KV8) {
+        const unsigned* vw = reinterpret_cast<const unsigned*>(&vr2[i]);
+        #pragma unroll
+        for (int jj = 0; jj < 2; ++jj) {
+          float2v lo = __builtin_amdgcn_cvt_pk_f32_fp8(vw[jj], false);
+          float2v hi = __builtin_amdgcn_cvt_pk_f32_fp8(vw[jj], true);
+          vf[2 * jj].x = lo.x; vf[2 * jj].y = lo.y;
+          vf[2 * jj + 1].x = hi.x; vf[2 * jj + 1].y = hi.y;
+        }
+      } else {
+        const unsigned* vw = reinterpret_cast<const unsigned*>(&vr4[i]);
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          vf[j].x = bf2f((unsigned short)(vw[j] & 0xffff));
+          vf[j].y = bf2f((unsigned short)(vw[j] >> 16));
+        }
       }
       #pragma unroll
       for (int h = 0; h < GROUP; ++h) {
@@ -482,7 +534,7 @@ void launch_paged_attn(bf16* out, const bf16* q, const bf16* k_cache,
                        long num_rows, int num_kv_heads, int group, int D,
                        int max_blocks, int block_size, float scale,
                        long q_stride, float* split_ws, int split,
-                       hipStream_t stream) {
+                       bool kv_fp8, hipStream_t stream) {
   if (block_size != ATTN_BS)
     throw std::runtime_error("paged_attn: block_size must be 16");
   // v3 (D=128): GROUP<=4 direct; GROUP=8 as two GROUP=4 half-calls over
@@ -500,10 +552,16 @@ void launch_paged_attn(bf16* out, const bf16* q, const bf16* k_cache,
       dim3 mgrid((unsigned)num_rows, (unsigned)num_kv_heads);
       #define CASE3(G)                                                      \
         if (g == G) {                                                       \
-          paged_attn_v3_kernel<G><<<grid, 256, 0, stream>>>(                \
-              out, q, k_cache, v_cache, block_tables, ctx_lens, row_seq,    \
-              max_blocks, num_kv_heads, scale, q_stride, h_total, group,    \
-              h_off, split_ws);                                             \
+          if (kv_fp8)                                                       \
+            paged_attn_v3_kernel<G, true><<<grid, 256, 0, stream>>>(        \
+                out, q, k_cache, v_cache, block_tables, ctx_lens, row_seq,  \
+                max_blocks, num_kv_heads, scale, q_stride, h_total, group,  \
+                h_off, split_ws);                                           \
+          else                                                              \
+            paged_attn_v3_kernel<G, false><<<grid, 256, 0, stream>>>(       \
+                out, q, k_cache, v_cache, block_tables, ctx_lens, row_seq,  \
+                max_blocks, num_kv_heads, scale, q_stride, h_total, group,  \
+                h_off, split_ws);                                           \
           if (split > 1)                                                    \
             paged_attn_v3_merge_kernel<G><<<mgrid, 256, 0, stream>>>(       \
                 out, split_ws, split, h_total, group, h_off);               \
@@ -515,6 +573,8 @@ void launch_paged_attn(bf16* out, const bf16* q, const bf16* k_cache,
   }
   if (split != 1)
     throw std::runtime_error("paged_attn: split needs the v3 path");
+  if (kv_fp8)
+    throw std::runtime_error("paged_attn: fp8 KV needs head_dim 128");
   dim3 grid((unsigned)num_rows, (unsigned)num_kv_heads);
   #define CASE(G, DD)                                                       \
     if (group == G && D == DD) {                                            \

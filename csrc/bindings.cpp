@@ -17,11 +17,11 @@ void launch_fused_add_rmsnorm(bf16*, bf16*, const bf16*, float, long, int,
                               hipStream_t);
 void launch_rope_kv_append(bf16*, bf16*, const bf16*, const long*,
                            const float*, const long*, bf16*, bf16*, long,
-                           int, int, int, int, long, long, long,
+                           int, int, int, int, long, long, long, bool,
                            hipStream_t);
 void launch_paged_attn(bf16*, const bf16*, const bf16*, const bf16*,
                        const int*, const int*, const int*, long, int, int,
-                       int, int, int, float, long, float*, int,
+                       int, int, int, float, long, float*, int, bool,
                        hipStream_t);
 void launch_attn_prefill_mfma(bf16*, const bf16*, const bf16*,
                               const bf16*, const int*, int, int, int, int,
@@ -105,12 +105,14 @@ void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   const long T = q.size(0);
   const int H = (int)q.size(1), KV = (int)k.size(1), D = (int)q.size(2);
   const int BS = (int)k_cache.size(2);
+  const bool kv_fp8 =
+      k_cache.scalar_type() == torch::kFloat8_e4m3fn;
   launch_rope_kv_append(bf(q), bf(k), cbf(v),
                         positions.data_ptr<long>(),
                         cos_sin.data_ptr<float>(),
                         slot_mapping.data_ptr<long>(), bf(k_cache),
                         bf(v_cache), T, H, KV, D, BS, q.stride(0),
-                        k.stride(0), v.stride(0), stream());
+                        k.stride(0), v.stride(0), kv_fp8, stream());
 }
 
 void paged_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
@@ -139,10 +141,15 @@ void paged_attn(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                 "split_ws too small");
     ws = split_ws->data_ptr<float>();
   }
+  const bool kv_fp8 =
+      k_cache.scalar_type() == torch::kFloat8_e4m3fn;
+  TORCH_CHECK(v_cache.scalar_type() == k_cache.scalar_type(),
+              "k/v cache dtype mismatch");
   launch_paged_attn(bf(out), cbf(q), cbf(k_cache), cbf(v_cache),
                     block_tables.data_ptr<int>(), ctx_lens.data_ptr<int>(),
                     rs, R, KV, H / KV, D, (int)block_tables.size(1), BS,
-                    (float)scale, q.stride(0), ws, (int)split, stream());
+                    (float)scale, q.stride(0), ws, (int)split, kv_fp8,
+                    stream());
 }
 
 void attn_prefill_mfma(torch::Tensor out, torch::Tensor q, torch::Tensor k,

@@ -12,7 +12,10 @@
 // k_cache/v_cache: [NB, KV, BS, D] bf16
 // q/k/v may be strided views into one fused qkv GEMM output buffer
 // (token stride qs/ks/vs in elements; head dim contiguous).
-template <int D>
+// KV8: the paged cache stores OCP e4m3fn bytes (fp8 KV mode — halves the
+// decode-attention HBM stream; scale fixed at 1.0, k/v magnitudes sit in
+// the e4m3 normal range).
+template <int D, bool KV8>
 __global__ __launch_bounds__(256) void rope_kv_append_kernel(
     bf16* __restrict__ q, bf16* __restrict__ k, const bf16* __restrict__ v,
     const long* __restrict__ positions, const float* __restrict__ cos_sin,
@@ -59,24 +62,53 @@ __global__ __launch_bounds__(256) void rope_kv_append_kernel(
     float s0 = cs[HALF + j], s1 = cs[HALF + j + 1];
     float x0 = bf2f(lo.x), x1 = bf2f(lo.y);
     float y0 = bf2f(hi.x), y1 = bf2f(hi.y);
-    lo.x = f2bf(x0 * c0 - y0 * s0); lo.y = f2bf(x1 * c1 - y1 * s1);
-    hi.x = f2bf(y0 * c0 + x0 * s0); hi.y = f2bf(y1 * c1 + x1 * s1);
+    const float r0 = x0 * c0 - y0 * s0, r1 = x1 * c1 - y1 * s1;
+    const float r2 = y0 * c0 + x0 * s0, r3 = y1 * c1 + x1 * s1;
+    lo.x = f2bf(r0); lo.y = f2bf(r1);
+    hi.x = f2bf(r2); hi.y = f2bf(r3);
     *reinterpret_cast<ushort2*>(base + j) = lo;
     *reinterpret_cast<ushort2*>(base + HALF + j) = hi;
     if (slot >= 0) {
-      bf16* kc = k_cache + (((blk * KV + h) * block_size + off) * D);
-      *reinterpret_cast<ushort2*>(kc + j) = lo;
-      *reinterpret_cast<ushort2*>(kc + HALF + j) = hi;
+      if constexpr (KV8) {
+        unsigned char* kc = reinterpret_cast<unsigned char*>(k_cache) +
+            (((blk * KV + h) * block_size + off) * D);
+        unsigned plo = __builtin_amdgcn_cvt_pk_fp8_f32(r0, r1, 0u, false);
+        unsigned phi = __builtin_amdgcn_cvt_pk_fp8_f32(r2, r3, 0u, false);
+        *reinterpret_cast<unsigned short*>(kc + j) = (unsigned short)plo;
+        *reinterpret_cast<unsigned short*>(kc + HALF + j) =
+            (unsigned short)phi;
+      } else {
+        bf16* kc = k_cache + (((blk * KV + h) * block_size + off) * D);
+        *reinterpret_cast<ushort2*>(kc + j) = lo;
+        *reinterpret_cast<ushort2*>(kc + HALF + j) = hi;
+      }
     }
   }
-  // v append: straight copy, 8-elem vectors
+  // v append: straight copy (bf16) or pack-to-fp8, 8-elem vectors
   if (slot >= 0) {
     const int total_v = KV * D / 8;
     for (int u = threadIdx.x; u < total_v; u += 256) {
       const int h = u / (D / 8), j = (u % (D / 8)) * 8;
       const bf16* src = v + t * vs_stride + h * D + j;
-      bf16* vc = v_cache + (((blk * KV + h) * block_size + off) * D) + j;
-      *reinterpret_cast<uint4*>(vc) = *reinterpret_cast<const uint4*>(src);
+      if constexpr (KV8) {
+        bf16x8 vv = load_bf16x8(src);
+        unsigned w0 = 0, w1 = 0;
+        w0 = __builtin_amdgcn_cvt_pk_fp8_f32(bf16x8_get(vv, 0),
+                                             bf16x8_get(vv, 1), w0, false);
+        w0 = __builtin_amdgcn_cvt_pk_fp8_f32(bf16x8_get(vv, 2),
+                                             bf16x8_get(vv, 3), w0, true);
+        w1 = __builtin_amdgcn_cvt_pk_fp8_f32(bf16x8_get(vv, 4),
+                                             bf16x8_get(vv, 5), w1, false);
+        w1 = __builtin_amdgcn_cvt_pk_fp8_f32(bf16x8_get(vv, 6),
+                                             bf16x8_get(vv, 7), w1, true);
+        uint2 pk{w0, w1};
+        *reinterpret_cast<uint2*>(reinterpret_cast<unsigned char*>(v_cache)
+            + (((blk * KV + h) * block_size + off) * D) + j) = pk;
+      } else {
+        bf16* vc = v_cache + (((blk * KV + h) * block_size + off) * D) + j;
+        *reinterpret_cast<uint4*>(vc) =
+            *reinterpret_cast<const uint4*>(src);
+      }
     }
   }
 }
@@ -86,16 +118,20 @@ void launch_rope_kv_append(bf16* q, bf16* k, const bf16* v,
                            const long* slot_mapping, bf16* k_cache,
                            bf16* v_cache, long T, int H, int KV, int D,
                            int block_size, long qs, long ks, long vs,
-                           hipStream_t stream) {
+                           bool kv_fp8, hipStream_t stream) {
   dim3 grid((unsigned)T);
-  if (D == 128)
-    rope_kv_append_kernel<128><<<grid, 256, 0, stream>>>(
+  if (D == 128 && kv_fp8)
+    rope_kv_append_kernel<128, true><<<grid, 256, 0, stream>>>(
         q, k, v, positions, cos_sin, slot_mapping, k_cache, v_cache,
         H, KV, block_size, qs, ks, vs);
-  else if (D == 64)
-    rope_kv_append_kernel<64><<<grid, 256, 0, stream>>>(
+  else if (D == 128)
+    rope_kv_append_kernel<128, false><<<grid, 256, 0, stream>>>(
+        q, k, v, positions, cos_sin, slot_mapping, k_cache, v_cache,
+        H, KV, block_size, qs, ks, vs);
+  else if (D == 64 && !kv_fp8)
+    rope_kv_append_kernel<64, false><<<grid, 256, 0, stream>>>(
         q, k, v, positions, cos_sin, slot_mapping, k_cache, v_cache,
         H, KV, block_size, qs, ks, vs);
   else
-    throw std::runtime_error("rope_kv_append: unsupported head_dim");
+    throw std::runtime_error("rope_kv_append: unsupported head_dim/dtype");
 }

@@ -111,6 +111,7 @@ class EngineConfig:
     ep_size: int = 1
     dtype: str = "bfloat16"
     quant: Optional[str] = None          # None (bf16) | "fp8" (e4m3fn weights)
+    kv_dtype: str = "bfloat16"           # "bfloat16" | "fp8" (e4m3fn cache)
     seed: int = 0
     # decode hipGraph capture batch buckets (padded up to nearest)
     graph_batch_sizes: tuple = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128,

@@ -149,6 +149,7 @@ def compute_num_gpu_blocks(spec, config, device, safety_frac: float = None) -> i
     free, total = torch.cuda.mem_get_info(device)
     budget = int(total * frac) - (total - free)
     kv_heads_local = spec.num_kv_heads // max(config.tp_size, 1)
+    elem_bytes = 1 if getattr(config, "kv_dtype", "bfloat16") == "fp8" else 2
     per_block = (2 * spec.num_layers * kv_heads_local * config.block_size *
-                 spec.head_dim * torch.finfo(torch.bfloat16).bits // 8)
+                 spec.head_dim * elem_bytes)
     return max(budget // per_block, 16)

@@ -78,9 +78,11 @@ class ModelRunner:
         self.model.eval()
         self.num_blocks = compute_num_gpu_blocks(self.spec, config, self.device)
         kv_heads_local = self.spec.num_kv_heads // max(config.tp_size, 1)
+        assert config.kv_dtype in ("bfloat16", "fp8"), config.kv_dtype
+        kv_dt = torch.float8_e4m3fn if config.kv_dtype == "fp8"             else self.dtype
         self.kv_caches = allocate_kv_caches(
             self.spec.num_layers, self.num_blocks, kv_heads_local,
-            config.block_size, self.spec.head_dim, self.dtype, self.device)
+            config.block_size, self.spec.head_dim, kv_dt, self.device)
         self.max_blocks_per_seq = (config.max_model_len + config.block_size - 1) \
             // config.block_size
         self.block_manager = BlockManager(

@@ -217,6 +217,8 @@ def main():
     ap.add_argument("--device", default=None)
     ap.add_argument("--tp", type=int, default=1)
     ap.add_argument("--quant", default=None, choices=["fp8"])
+    ap.add_argument("--kv-dtype", default="bfloat16",
+                    choices=["bfloat16", "fp8"])
     args = ap.parse_args()
 
     logging.basicConfig(level=logging.INFO,
@@ -231,7 +233,7 @@ def main():
         max_model_len=args.max_model_len,
         num_gpu_blocks=args.num_gpu_blocks or None,
         enforce_eager=args.eager or not on_gpu, tp_size=args.tp,
-        quant=args.quant)
+        quant=args.quant, kv_dtype=args.kv_dtype)
     if args.tp > 1:
         from hyperspot.parallel.state import initialize_model_parallel
         initialize_model_parallel(tp_size=args.tp)
