@@ -3,6 +3,8 @@
 #include <arpa/inet.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
+#include <openssl/sha.h>
+#include <poll.h>
 #include <sys/socket.h>
 #include <unistd.h>
 
@@ -298,6 +300,129 @@ void HttpServer::handle_conn(int fd, std::string peer) {
                 "\"status\":404}");
     if (!w.keep_alive) return;
   }
+}
+
+
+
+// ------------------------------------------------------------- WebSocket
+
+bool WsConn::send_text(const std::string& p) {
+  std::string hdr;
+  hdr.push_back((char)0x81);           // FIN + text
+  if (p.size() < 126) {
+    hdr.push_back((char)p.size());
+  } else if (p.size() < 65536) {
+    hdr.push_back(126);
+    hdr.push_back((char)(p.size() >> 8));
+    hdr.push_back((char)(p.size() & 0xff));
+  } else {
+    hdr.push_back(127);
+    for (int i = 7; i >= 0; --i)
+      hdr.push_back((char)((p.size() >> (8 * i)) & 0xff));
+  }
+  std::string f = hdr + p;
+  const char* d = f.data();
+  size_t n = f.size();
+  while (n) {
+    ssize_t w = send(fd_, d, n, MSG_NOSIGNAL);
+    if (w <= 0) return false;
+    d += w;
+    n -= (size_t)w;
+  }
+  return true;
+}
+
+void WsConn::send_close() {
+  const char f[2] = {(char)0x88, 0};
+  send(fd_, f, 2, MSG_NOSIGNAL);
+}
+
+std::optional<std::string> WsConn::recv_text(int timeout_ms) {
+  while (true) {
+    // try to parse one frame from buf_
+    if (buf_.size() >= 2) {
+      const unsigned char b0 = buf_[0], b1 = buf_[1];
+      const int opcode = b0 & 0x0f;
+      const bool masked = b1 & 0x80;
+      size_t len = b1 & 0x7f;
+      size_t off = 2;
+      if (len == 126) {
+        if (buf_.size() >= 4) {
+          len = ((unsigned char)buf_[2] << 8) | (unsigned char)buf_[3];
+          off = 4;
+        } else len = SIZE_MAX;
+      } else if (len == 127) {
+        if (buf_.size() >= 10) {
+          len = 0;
+          for (int i = 0; i < 8; ++i)
+            len = (len << 8) | (unsigned char)buf_[2 + i];
+          off = 10;
+        } else len = SIZE_MAX;
+      }
+      if (len != SIZE_MAX) {
+        const size_t need = off + (masked ? 4 : 0) + len;
+        if (buf_.size() >= need) {
+          std::string payload = buf_.substr(off + (masked ? 4 : 0), len);
+          if (masked) {
+            const unsigned char* mk =
+                (const unsigned char*)buf_.data() + off;
+            for (size_t i = 0; i < payload.size(); ++i)
+              payload[i] = (char)(payload[i] ^ mk[i & 3]);
+          }
+          buf_.erase(0, need);
+          if (opcode == 0x8) return std::nullopt;          // close
+          if (opcode == 0x9) {                             // ping -> pong
+            std::string pong;
+            pong.push_back((char)0x8a);
+            pong.push_back((char)payload.size());
+            pong += payload;
+            send(fd_, pong.data(), pong.size(), MSG_NOSIGNAL);
+            continue;
+          }
+          if (opcode == 0x1 || opcode == 0x2) return payload;
+          continue;                                        // pong etc.
+        }
+      }
+    }
+    struct pollfd pf{fd_, POLLIN, 0};
+    if (poll(&pf, 1, timeout_ms) <= 0) return std::nullopt;
+    char tmp[8192];
+    ssize_t r = recv(fd_, tmp, sizeof tmp, 0);
+    if (r <= 0) return std::nullopt;
+    buf_.append(tmp, (size_t)r);
+  }
+}
+
+std::optional<WsConn> websocket_upgrade(const HttpRequest& req,
+                                        ResponseWriter& w) {
+  if (req.header("upgrade").find("websocket") == std::string::npos)
+    return std::nullopt;
+  const std::string key = req.header("sec-websocket-key");
+  if (key.empty()) return std::nullopt;
+  const std::string magic = key + "258EAFA5-E914-47DA-95CA-C5AB0DC85B11";
+  unsigned char sha[20];
+  SHA1((const unsigned char*)magic.data(), magic.size(), sha);
+  static const char* tbl =
+      "ABCDEFGHIJKLMNOPQRSTUVWXYZabcdefghijklmnopqrstuvwxyz0123456789+/";
+  std::string acc;
+  for (int i = 0; i < 20; i += 3) {
+    unsigned v = sha[i] << 16;
+    if (i + 1 < 20) v |= sha[i + 1] << 8;
+    if (i + 2 < 20) v |= sha[i + 2];
+    acc.push_back(tbl[(v >> 18) & 63]);
+    acc.push_back(tbl[(v >> 12) & 63]);
+    acc.push_back(i + 1 < 20 ? tbl[(v >> 6) & 63] : '=');
+    acc.push_back(i + 2 < 20 ? tbl[v & 63] : '=');
+  }
+  std::string resp =
+      "HTTP/1.1 101 Switching Protocols\r\n"
+      "upgrade: websocket\r\nconnection: Upgrade\r\n"
+      "sec-websocket-accept: " + acc + "\r\n\r\n";
+  ssize_t sr = send(w.raw_fd(), resp.data(), resp.size(), MSG_NOSIGNAL);
+  if (sr != (ssize_t)resp.size()) return std::nullopt;
+  w.mark_started();
+  w.keep_alive = false;
+  return WsConn(w.raw_fd());
 }
 
 }  // namespace hs

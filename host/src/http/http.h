@@ -9,6 +9,7 @@
 
 #include <atomic>
 #include <functional>
+#include <optional>
 #include <map>
 #include <memory>
 #include <string>
@@ -58,6 +59,8 @@ class ResponseWriter {
   bool keep_alive = true;
   // headers stamped onto every response (request id, traceparent)
   std::vector<std::pair<std::string, std::string>> default_headers;
+  int raw_fd() const { return fd_; }
+  void mark_started() { started_ = true; }   // used by websocket upgrade
 
  private:
   bool send_all(const char* p, size_t n);
@@ -65,6 +68,29 @@ class ResponseWriter {
   bool started_ = false;
   bool streaming_ = false;
 };
+
+// ------------------------------------------------------------- WebSocket
+// Minimal RFC-6455 server side for the llm-gateway /realtime contract:
+// handshake (Sec-WebSocket-Accept via SHA-1+base64), text frames, close.
+class WsConn {
+ public:
+  explicit WsConn(int fd) : fd_(fd) {}
+  bool send_text(const std::string& payload);
+  // one text message (handles masking + fragmentation-free frames);
+  // nullopt on close/error/timeout
+  std::optional<std::string> recv_text(int timeout_ms = 120000);
+  void send_close();
+
+ private:
+  int fd_;
+  std::string buf_;
+};
+
+// Performs the 101 handshake if the request asks for it; returns the
+// socket wrapped for frame IO (the HTTP connection loop must stop reusing
+// the socket: keep_alive is cleared and the response marked started).
+std::optional<WsConn> websocket_upgrade(const HttpRequest& req,
+                                        ResponseWriter& w);
 
 using HttpHandler = std::function<void(HttpRequest&, ResponseWriter&)>;
 

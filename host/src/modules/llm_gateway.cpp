@@ -969,6 +969,109 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
     });
   }
 
+  // ---- WS /realtime (DESIGN.md:262-270): JSON messages over RFC-6455.
+  // client: {"type":"input_text","text":...,"max_tokens"?:N}
+  // server: {"type":"delta","text":...} ... {"type":"done","usage":{...}}
+  for (const char* p : {"/llm-gateway/v1/realtime", "/v1/realtime"}) {
+    OperationSpec op;
+    op.method = "GET";
+    op.path = p;
+    op.operation_id = std::string("realtime") + (p[1] == 'v' ? "_alias" : "");
+    op.summary = "Realtime WebSocket session";
+    op.authenticated = true;
+    op.tags = {"llm-gateway"};
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      auto sec = SecurityContext::from_json(rq.extensions.at("security"));
+      auto ws = websocket_upgrade(rq, w);
+      if (!ws)
+        throw Problem{426, "Upgrade Required", "about:blank",
+                      "websocket upgrade required", "validation_error"};
+      while (true) {
+        auto msg = ws->recv_text();
+        if (!msg) break;
+        Json m;
+        try { m = Json::parse(*msg); }
+        catch (...) {
+          ws->send_text("{\"type\":\"error\",\"code\":"
+                        "\"validation_error\"}");
+          continue;
+        }
+        const std::string type = m.at("type").as_string();
+        if (type == "session.close") break;
+        if (type != "input_text") {
+          ws->send_text("{\"type\":\"error\",\"code\":"
+                        "\"validation_error\"}");
+          continue;
+        }
+        if (!worker_ready()) {
+          ws->send_text("{\"type\":\"error\",\"code\":"
+                        "\"provider_error\"}");
+          continue;
+        }
+        EngineConn conn(socket_path_);
+        const std::string rid =
+            "rt-" + std::to_string(req_ctr_.fetch_add(1));
+        Json wreq = Json::object();
+        wreq["type"] = "chat";
+        wreq["id"] = rid;
+        wreq["model"] = model_;
+        Json part = Json::object();
+        part["type"] = "text";
+        part["text"] = m.at("text").as_string();
+        Json content = Json::array();
+        content.push_back(part);
+        Json um = Json::object();
+        um["role"] = "user";
+        um["content"] = content;
+        Json msgs = Json::array();
+        msgs.push_back(um);
+        wreq["messages"] = msgs;
+        Json params = Json::object();
+        if (m.contains("max_tokens")) params["max_tokens"] = m.at("max_tokens");
+        if (m.contains("temperature"))
+          params["temperature"] = m.at("temperature");
+        wreq["params"] = params;
+        if (!conn.ok() || !conn.send_json(wreq)) {
+          ws->send_text("{\"type\":\"error\",\"code\":"
+                        "\"provider_error\"}");
+          continue;
+        }
+        bool gone = false;
+        while (true) {
+          auto ev = conn.read_json();
+          if (!ev) break;
+          const std::string e = ev->at("event").as_string();
+          if (e == "delta") {
+            Json d = Json::object();
+            d["type"] = "delta";
+            d["text"] = ev->at("text");
+            if (!ws->send_text(d.dump())) { gone = true; break; }
+          } else if (e == "done") {
+            record_usage(sec.tenant_id, ev->at("usage"));
+            Json d = Json::object();
+            d["type"] = "done";
+            d["usage"] = ev->at("usage");
+            d["finish_reason"] = ev->at("finish_reason");
+            ws->send_text(d.dump());
+            break;
+          } else if (e == "error") {
+            ws->send_text("{\"type\":\"error\",\"code\":"
+                          "\"provider_error\"}");
+            break;
+          }
+        }
+        if (gone) {
+          Json ab = Json::object();
+          ab["type"] = "abort";
+          ab["id"] = rid;
+          conn.send_json(ab);
+          break;
+        }
+      }
+      ws->send_close();
+    });
+  }
+
   OperationSpec status;
   status.method = "GET";
   status.path = "/llm-gateway/v1/status";

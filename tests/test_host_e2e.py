@@ -750,3 +750,88 @@ def test_traceparent_propagation(server):
     assert tp.startswith("00-0123456789abcdef0123456789abcdef-"), tp
     # a fresh span id was minted for this hop
     assert tp.split("-")[2] != "00f067aa0ba902b7"
+
+
+def _ws_handshake_and_chat(host, port, path, text):
+    """Minimal RFC-6455 client: handshake + one text roundtrip."""
+    import base64
+    import hashlib
+    import struct
+    s = socket.create_connection((host, port), timeout=30)
+    key = base64.b64encode(os.urandom(16)).decode()
+    req = (f"GET {path} HTTP/1.1\r\nhost: {host}\r\n"
+           f"upgrade: websocket\r\nconnection: Upgrade\r\n"
+           f"sec-websocket-key: {key}\r\nsec-websocket-version: 13\r\n\r\n")
+    s.sendall(req.encode())
+    hdr = b""
+    while b"\r\n\r\n" not in hdr:
+        hdr += s.recv(4096)
+    head = hdr.split(b"\r\n\r\n")[0].decode()
+    assert "101" in head.split("\r\n")[0], head
+    want = base64.b64encode(hashlib.sha1(
+        (key + "258EAFA5-E914-47DA-95CA-C5AB0DC85B11").encode()
+    ).digest()).decode()
+    assert want in head, head
+
+    def send_text(payload):
+        data = payload.encode()
+        mask = os.urandom(4)
+        frame = bytearray([0x81])
+        if len(data) < 126:
+            frame.append(0x80 | len(data))
+        else:
+            frame.append(0x80 | 126)
+            frame += struct.pack(">H", len(data))
+        frame += mask
+        frame += bytes(b ^ mask[i % 4] for i, b in enumerate(data))
+        s.sendall(frame)
+
+    buf = bytearray(hdr.split(b"\r\n\r\n", 1)[1])
+
+    def recv_text():
+        nonlocal buf
+        while True:
+            if len(buf) >= 2:
+                ln = buf[1] & 0x7F
+                off = 2
+                if ln == 126 and len(buf) >= 4:
+                    ln = struct.unpack(">H", bytes(buf[2:4]))[0]
+                    off = 4
+                if len(buf) >= off + ln:
+                    op = buf[0] & 0x0F
+                    payload = bytes(buf[off:off + ln])
+                    del buf[:off + ln]
+                    if op == 8:
+                        return None
+                    if op == 1:
+                        return payload.decode()
+                    continue
+            chunk = s.recv(65536)
+            if not chunk:
+                return None
+            buf += chunk
+
+    send_text(json.dumps({"type": "input_text", "text": text,
+                          "max_tokens": 5}))
+    deltas, done = [], None
+    for _ in range(200):
+        m = recv_text()
+        if m is None:
+            break
+        d = json.loads(m)
+        if d["type"] == "delta":
+            deltas.append(d["text"])
+        elif d["type"] == "done":
+            done = d
+            break
+        elif d["type"] == "error":
+            raise AssertionError(d)
+    s.close()
+    return deltas, done
+
+
+def test_realtime_websocket(server):
+    deltas, done = _ws_handshake_and_chat("127.0.0.1", server.port,
+                                          "/llm-gateway/v1/realtime", "hi")
+    assert done is not None and done["usage"]["output_tokens"] > 0
+    assert len(deltas) >= 1
