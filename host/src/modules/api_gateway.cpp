@@ -50,6 +50,9 @@ void ApiGatewayModule::init(ModuleCtx& ctx) {
       (int)c.path("defaults.rate_limit.in_flight").as_int(64);
   body_limit_ =
       (size_t)c.path("defaults.body_limit_bytes").as_int(16 * 1024 * 1024);
+  if (c.path("license.features").is_array())
+    for (auto& f : c.path("license.features").arr())
+      licensed_features_.push_back(f.as_string());
 }
 
 void ApiGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
@@ -247,6 +250,22 @@ void ApiGatewayModule::dispatch(HttpRequest& req, ResponseWriter& w) {
                           "content-type '" + base + "' not allowed", ""},
                       req.path);
       return;
+    }
+  }
+
+  // license validation (reference src/middleware/license_validation.rs:
+  // route specs declare required features; the gateway config grants them)
+  if (!route->spec.license_features.empty()) {
+    for (auto& need : route->spec.license_features) {
+      bool have = false;
+      for (auto& f : licensed_features_)
+        if (f == need) { have = true; break; }
+      if (!have) {
+        respond_problem(w, {403, "Forbidden", "about:blank",
+                            "license feature '" + need + "' not granted",
+                            "license_required"}, req.path);
+        return;
+      }
     }
   }
 

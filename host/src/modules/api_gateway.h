@@ -67,6 +67,7 @@ class ApiGatewayModule : public Module {
   bool cors_enabled_ = false;
   bool auth_disabled_ = false;
   Json cors_cfg_;
+  std::vector<std::string> licensed_features_;
   std::string openapi_title_ = "API Documentation";
   std::string openapi_version_ = "0.1.0";
   std::string openapi_desc_;

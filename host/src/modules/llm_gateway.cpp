@@ -154,6 +154,7 @@ void LlmGatewayModule::init(ModuleCtx& ctx) {
   worker_cfg_ = ctx.config.at("worker");
   budget_tokens_ =
       (uint64_t)ctx.config.path("usage.budget_tokens_per_tenant").as_int(0);
+  license_feature_ = ctx.config.at("require_license_feature").as_string("");
 }
 
 bool LlmGatewayModule::worker_ready() {
@@ -690,6 +691,8 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
     op.responses[200] = "completion response or SSE stream";
     op.responses[202] = "async job accepted";
     op.sse = true;
+    if (!license_feature_.empty())
+      op.license_features = {license_feature_};
     rest.register_op(op, handler);
   }
 

@@ -130,6 +130,7 @@ class LlmGatewayModule : public Module {
   std::mutex usage_mu_;
   std::map<std::string, TenantUsage> usage_;
   uint64_t budget_tokens_ = 0;
+  std::string license_feature_;
 
   // metrics
   std::atomic<uint64_t> m_requests_{0}, m_streams_{0}, m_errors_{0};

@@ -835,3 +835,49 @@ def test_realtime_websocket(server):
                                           "/llm-gateway/v1/realtime", "hi")
     assert done is not None and done["usage"]["output_tokens"] > 0
     assert len(deltas) >= 1
+
+
+@pytest.fixture(scope="module")
+def licensed_server():
+    """Gateway grants no license features; llm-gateway chat requires one."""
+    import tempfile
+    port = _free_port()
+    cfg = f"""
+server:
+  home_dir: "/tmp/hs-e2e-lic"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+      license:
+        features: ["basic"]
+  llm-gateway:
+    config:
+      auto_start_worker: false
+      require_license_feature: "inference-pro"
+"""
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(cfg)
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        yield srv
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
+
+
+def test_license_validation_blocks_route(licensed_server):
+    url = BASE.format(licensed_server.port)
+    st, body = _http("POST", url + "/v1/chat/completions",
+                     {"model": "x", "messages": [{"role": "user",
+                                                  "content": []}]})
+    assert st == 403, body
+    assert json.loads(body)["code"] == "license_required"
+    # unlicensed routes unaffected
+    st, _ = _http("GET", url + "/healthz")
+    assert st == 200
