@@ -217,11 +217,32 @@ void LlmGatewayModule::spawn_worker() {
 }
 
 void LlmGatewayModule::start(ModuleCtx& ctx) {
+  stopping_ = false;
   if (auto_start_) {
     unlink(socket_path_.c_str());
     spawn_worker();
+    // failure detection + elastic recovery: reap a dead
+    // engine worker and respawn it; the gateway 503s only while the
+    // replacement loads
+    watchdog_ = std::thread([this] {
+      while (!stopping_) {
+        for (int i = 0; i < 20 && !stopping_; ++i) usleep(100000);
+        if (stopping_) break;
+        if (worker_pid_ > 0) {
+          int st = 0;
+          if (waitpid(worker_pid_, &st, WNOHANG) == worker_pid_) {
+            LOG_ERROR("llm-gateway",
+                      "engine worker pid=%d died (status=%d); respawning",
+                      worker_pid_, st);
+            ready_ = false;
+            m_worker_restarts_++;
+            unlink(socket_path_.c_str());
+            spawn_worker();
+          }
+        }
+      }
+    });
   }
-  stopping_ = false;
   job_thread_ = std::thread([this] { job_loop(); });
 }
 
@@ -232,6 +253,7 @@ void LlmGatewayModule::stop(ModuleCtx& ctx) {
   }
   jobs_cv_.notify_all();
   if (job_thread_.joinable()) job_thread_.join();
+  if (watchdog_.joinable()) watchdog_.join();
   if (worker_pid_ > 0) {
     kill(worker_pid_, SIGTERM);
     int st = 0;
@@ -929,6 +951,8 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
           (double)m_streams_.load());
       add("hyperspot_engine_errors_total", "Engine-side errors", "counter",
           (double)m_errors_.load());
+      add("hyperspot_worker_restarts_total", "Engine worker respawns",
+          "counter", (double)m_worker_restarts_.load());
       add("hyperspot_input_tokens_total", "Prompt tokens", "counter",
           (double)m_input_tokens_.load());
       add("hyperspot_output_tokens_total", "Generated tokens", "counter",

@@ -124,7 +124,9 @@ class LlmGatewayModule : public Module {
   std::map<std::string, std::shared_ptr<Job>> jobs_;
   std::map<std::string, Batch> batches_;
   std::thread job_thread_;
+  std::thread watchdog_;
   std::atomic<bool> stopping_{false};
+  std::atomic<uint64_t> m_worker_restarts_{0};
 
   // usage tracker + budget (tokens per tenant; 0 = unlimited)
   std::mutex usage_mu_;

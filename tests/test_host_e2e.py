@@ -881,3 +881,40 @@ def test_license_validation_blocks_route(licensed_server):
     # unlicensed routes unaffected
     st, _ = _http("GET", url + "/healthz")
     assert st == 200
+
+
+def test_worker_crash_recovery(server):
+    """Failure detection + elastic recovery: kill the engine worker; the
+    watchdog respawns it and chat serves again."""
+    url = BASE.format(server.port)
+    st, body = _http("GET", url + "/llm-gateway/v1/status")
+    eng_pid = None
+    # find the worker via the process table (child of the server)
+    out = subprocess.run(["pgrep", "-P", str(server.proc.pid)],
+                         capture_output=True, text=True).stdout.split()
+    for pid in out:
+        with open(f"/proc/{pid}/cmdline") as f:
+            if "hyperspot.serving.worker" in f.read():
+                eng_pid = int(pid)
+    assert eng_pid, "no worker child found"
+    os.kill(eng_pid, 9)
+    # watchdog notices within ~2s and respawns; wait for ready again
+    deadline = time.time() + 60
+    ok = False
+    while time.time() < deadline:
+        try:
+            st, body = _http("POST", url + "/v1/chat/completions",
+                             {"model": "tiny-llama",
+                              "messages": [{"role": "user", "content":
+                                            [{"type": "text",
+                                              "text": "back?"}]}],
+                              "max_tokens": 2}, timeout=10)
+            if st == 200:
+                ok = True
+                break
+        except Exception:
+            pass
+        time.sleep(1.0)
+    assert ok, "worker did not recover"
+    st, body = _http("GET", url + "/metrics")
+    assert "hyperspot_worker_restarts_total 1" in body
