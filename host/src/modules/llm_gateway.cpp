@@ -996,6 +996,87 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
     });
   }
 
+  // ---- checkpoint save / live hot-swap (BASELINE config 5:
+  // model-registry + file-storage checkpoints; the engine swaps weights
+  // in place so the KV pool and captured hipGraphs survive) ----
+  {
+    OperationSpec op;
+    op.method = "POST";
+    op.path = "/llm-gateway/v1/checkpoints/save";
+    op.operation_id = "save_checkpoint";
+    op.summary = "Persist current engine weights to file-storage";
+    op.authenticated = true;
+    op.allowed_content_types = {"application/json"};
+    op.tags = {"llm-gateway"};
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      Json body;
+      try { body = Json::parse(rq.body); }
+      catch (...) { throw Problem::bad_request("invalid JSON body"); }
+      auto sec = SecurityContext::from_json(rq.extensions.at("security"));
+      std::string name = body.at("name").as_string("checkpoint");
+      auto fs = hub_->get<FileStorageClient>("file-storage");
+      if (!fs)
+        throw Problem{503, "Service Unavailable", "about:blank",
+                      "file-storage unavailable", "provider_error"};
+      const std::string path =
+          fs->root_for(sec.tenant_id) + "/" + name + ".safetensors";
+      if (!worker_ready())
+        throw Problem{503, "Service Unavailable", "about:blank",
+                      "engine not ready", "provider_error"};
+      EngineConn conn(socket_path_);
+      Json wr = Json::object();
+      wr["type"] = "save_checkpoint";
+      wr["path"] = path;
+      if (!conn.ok() || !conn.send_json(wr))
+        throw Problem{502, "Bad Gateway", "about:blank",
+                      "engine write failed", "provider_error"};
+      auto r = conn.read_json(600000);
+      if (!r || r->at("event").as_string() != "saved")
+        throw Problem{502, "Bad Gateway", "about:blank",
+                      r ? r->at("message").as_string() : "save timed out",
+                      "provider_error"};
+      Json out = Json::object();
+      out["path"] = path;
+      w.respond(200, "application/json", out.dump());
+    });
+  }
+  {
+    OperationSpec op;
+    op.method = "POST";
+    op.path = "/llm-gateway/v1/checkpoints/swap";
+    op.operation_id = "swap_checkpoint";
+    op.summary = "Live weight hot-swap (KV pool + graphs survive)";
+    op.authenticated = true;
+    op.allowed_content_types = {"application/json"};
+    op.tags = {"llm-gateway"};
+    rest.register_op(op, [this](HttpRequest& rq, ResponseWriter& w) {
+      Json body;
+      try { body = Json::parse(rq.body); }
+      catch (...) { throw Problem::bad_request("invalid JSON body"); }
+      const std::string path = body.at("path").as_string();
+      if (path.empty()) throw Problem::bad_request("'path' is required");
+      if (!worker_ready())
+        throw Problem{503, "Service Unavailable", "about:blank",
+                      "engine not ready", "provider_error"};
+      EngineConn conn(socket_path_);
+      Json wr = Json::object();
+      wr["type"] = "swap";
+      wr["checkpoint"] = path;
+      if (!conn.ok() || !conn.send_json(wr))
+        throw Problem{502, "Bad Gateway", "about:blank",
+                      "engine write failed", "provider_error"};
+      auto r = conn.read_json(600000);
+      if (!r || r->at("event").as_string() != "swapped")
+        throw Problem{502, "Bad Gateway", "about:blank",
+                      r ? r->at("message").as_string() : "swap timed out",
+                      "provider_error"};
+      Json out = Json::object();
+      out["checkpoint"] = path;
+      out["seconds"] = r->at("seconds");
+      w.respond(200, "application/json", out.dump());
+    });
+  }
+
   // ---- WS /realtime (DESIGN.md:262-270): JSON messages over RFC-6455.
   // client: {"type":"input_text","text":...,"max_tokens"?:N}
   // server: {"type":"delta","text":...} ... {"type":"done","usage":{...}}

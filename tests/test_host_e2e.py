@@ -918,3 +918,29 @@ def test_worker_crash_recovery(server):
     assert ok, "worker did not recover"
     st, body = _http("GET", url + "/metrics")
     assert "hyperspot_worker_restarts_total 1" in body
+
+
+def test_checkpoint_save_and_hot_swap(server):
+    """BASELINE config 5 flow over REST: save current weights into
+    file-storage, then live-swap them back in; serving continues."""
+    url = BASE.format(server.port)
+    st, body = _http("POST", url + "/llm-gateway/v1/checkpoints/save",
+                     {"name": "ck-e2e"}, timeout=120)
+    assert st == 200, body
+    path = json.loads(body)["path"]
+    assert path.endswith("ck-e2e.safetensors") and os.path.exists(path)
+    st, body = _http("POST", url + "/llm-gateway/v1/checkpoints/swap",
+                     {"path": path}, timeout=120)
+    assert st == 200, body
+    assert json.loads(body)["seconds"] >= 0
+    # engine still serves after the swap
+    st, body = _http("POST", url + "/v1/chat/completions",
+                     {"model": "tiny-llama",
+                      "messages": [{"role": "user", "content":
+                                    [{"type": "text", "text": "post-swap"}]}],
+                      "max_tokens": 3})
+    assert st == 200, body
+    # swapping a missing checkpoint is a clean 502, not a crash
+    st, body = _http("POST", url + "/llm-gateway/v1/checkpoints/swap",
+                     {"path": "/nonexistent.safetensors"}, timeout=60)
+    assert st == 502, body
