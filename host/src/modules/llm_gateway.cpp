@@ -155,30 +155,70 @@ void LlmGatewayModule::init(ModuleCtx& ctx) {
   budget_tokens_ =
       (uint64_t)ctx.config.path("usage.budget_tokens_per_tenant").as_int(0);
   license_feature_ = ctx.config.at("require_license_feature").as_string("");
+  // worker fleet: explicit device list, or count (devices 0..count-1)
+  std::vector<int> devices;
+  if (worker_cfg_.at("devices").is_array())
+    for (auto& d : worker_cfg_.at("devices").arr())
+      devices.push_back((int)d.as_int(0));
+  else
+    for (int i = 0; i < (int)worker_cfg_.at("count").as_int(1); ++i)
+      devices.push_back(i);
+  if (devices.empty()) devices.push_back(0);
+  for (size_t i = 0; i < devices.size(); ++i) {
+    auto w = std::make_unique<Worker>();
+    w->index = (int)i;
+    w->device = devices[i];
+    w->socket = devices.size() == 1
+                    ? socket_path_
+                    : socket_path_ + "." + std::to_string(i);
+    workers_.push_back(std::move(w));
+  }
 }
 
-bool LlmGatewayModule::worker_ready() {
-  if (ready_) return true;
-  EngineConn c(socket_path_);
+bool LlmGatewayModule::probe_worker(Worker& wk) {
+  if (wk.ready) return true;
+  EngineConn c(wk.socket);
   if (!c.ok()) return false;
   Json q = Json::object();
   q["type"] = "info";
   if (!c.send_json(q)) return false;
   auto r = c.read_json(3000);
   if (r && r->at("ready").as_bool()) {
-    ready_ = true;
+    wk.ready = true;
     return true;
   }
   return false;
 }
 
-void LlmGatewayModule::spawn_worker() {
+bool LlmGatewayModule::worker_ready() {
+  for (auto& w : workers_)
+    if (probe_worker(*w)) return true;
+  return false;
+}
+
+LlmGatewayModule::Worker* LlmGatewayModule::pick_worker() {
+  Worker* best = nullptr;
+  for (auto& w : workers_) {
+    if (!probe_worker(*w)) continue;
+    if (!best || w->in_flight.load() < best->in_flight.load())
+      best = w.get();
+  }
+  if (best) best->in_flight++;
+  return best;
+}
+
+void LlmGatewayModule::spawn_one(Worker& wk) {
   pid_t pid = fork();
   if (pid == 0) {
     setpgid(0, 0);
+    if (workers_.size() > 1) {
+      const std::string dev = std::to_string(wk.device);
+      setenv("HIP_VISIBLE_DEVICES", dev.c_str(), 1);
+      setenv("CUDA_VISIBLE_DEVICES", dev.c_str(), 1);
+    }
     std::vector<std::string> args = {
         python_, "-m", "hyperspot.serving.worker",
-        "--uds", socket_path_, "--model", model_};
+        "--uds", wk.socket, "--model", model_};
     if (worker_cfg_.is_object()) {
       if (worker_cfg_.contains("max_num_seqs")) {
         args.push_back("--max-num-seqs");
@@ -204,6 +244,10 @@ void LlmGatewayModule::spawn_worker() {
         args.push_back("--quant");
         args.push_back(worker_cfg_.at("quant").as_string());
       }
+      if (worker_cfg_.contains("kv_dtype")) {
+        args.push_back("--kv-dtype");
+        args.push_back(worker_cfg_.at("kv_dtype").as_string());
+      }
     }
     std::vector<char*> argv;
     for (auto& a : args) argv.push_back(const_cast<char*>(a.c_str()));
@@ -211,33 +255,41 @@ void LlmGatewayModule::spawn_worker() {
     execvp(argv[0], argv.data());
     _exit(127);
   }
-  worker_pid_ = pid;
-  LOG_INFO("llm-gateway", "spawned engine worker pid=%d model=%s sock=%s",
-           pid, model_.c_str(), socket_path_.c_str());
+  wk.pid = pid;
+  LOG_INFO("llm-gateway",
+           "spawned engine worker %d pid=%d device=%d sock=%s", wk.index,
+           pid, wk.device, wk.socket.c_str());
+}
+
+void LlmGatewayModule::spawn_worker() {
+  for (auto& w : workers_) {
+    unlink(w->socket.c_str());
+    spawn_one(*w);
+  }
 }
 
 void LlmGatewayModule::start(ModuleCtx& ctx) {
   stopping_ = false;
   if (auto_start_) {
-    unlink(socket_path_.c_str());
     spawn_worker();
-    // failure detection + elastic recovery: reap a dead
-    // engine worker and respawn it; the gateway 503s only while the
-    // replacement loads
+    // failure detection + elastic recovery: reap dead
+    // engine workers and respawn them; requests route to the remaining
+    // live workers meanwhile
     watchdog_ = std::thread([this] {
       while (!stopping_) {
         for (int i = 0; i < 20 && !stopping_; ++i) usleep(100000);
         if (stopping_) break;
-        if (worker_pid_ > 0) {
+        for (auto& w : workers_) {
+          if (w->pid <= 0) continue;
           int st = 0;
-          if (waitpid(worker_pid_, &st, WNOHANG) == worker_pid_) {
+          if (waitpid(w->pid, &st, WNOHANG) == w->pid) {
             LOG_ERROR("llm-gateway",
-                      "engine worker pid=%d died (status=%d); respawning",
-                      worker_pid_, st);
-            ready_ = false;
+                      "engine worker %d pid=%d died (status=%d); respawn",
+                      w->index, w->pid, st);
+            w->ready = false;
             m_worker_restarts_++;
-            unlink(socket_path_.c_str());
-            spawn_worker();
+            unlink(w->socket.c_str());
+            spawn_one(*w);
           }
         }
       }
@@ -254,19 +306,20 @@ void LlmGatewayModule::stop(ModuleCtx& ctx) {
   jobs_cv_.notify_all();
   if (job_thread_.joinable()) job_thread_.join();
   if (watchdog_.joinable()) watchdog_.join();
-  if (worker_pid_ > 0) {
-    kill(worker_pid_, SIGTERM);
+  for (auto& w : workers_) {
+    if (w->pid <= 0) continue;
+    kill(w->pid, SIGTERM);
     int st = 0;
     for (int i = 0; i < 50; ++i) {
-      if (waitpid(worker_pid_, &st, WNOHANG) == worker_pid_) {
-        worker_pid_ = -1;
+      if (waitpid(w->pid, &st, WNOHANG) == w->pid) {
+        w->pid = -1;
         break;
       }
       usleep(100000);
     }
-    if (worker_pid_ > 0) {
-      kill(worker_pid_, SIGKILL);
-      waitpid(worker_pid_, &st, 0);
+    if (w->pid > 0) {
+      kill(w->pid, SIGKILL);
+      waitpid(w->pid, &st, 0);
     }
   }
 }
@@ -311,7 +364,11 @@ void LlmGatewayModule::record_usage(const std::string& tenant,
 Json LlmGatewayModule::run_chat_blocking(const Json& body,
                                          const Json& resolved,
                                          const std::string& rid) {
-  EngineConn conn(socket_path_);
+  Lease lease{pick_worker()};
+  if (!lease.w)
+    throw Problem{503, "Service Unavailable", "about:blank",
+                  "no engine worker ready", "provider_error"};
+  EngineConn conn(lease.w->socket);
   if (!conn.ok())
     throw Problem{503, "Service Unavailable", "about:blank",
                   "engine connection failed", "provider_error"};
@@ -438,7 +495,11 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   }
 
   m_streams_++;
-  EngineConn conn(socket_path_);
+  Lease lease{pick_worker()};
+  if (!lease.w)
+    throw Problem{503, "Service Unavailable", "about:blank",
+                  "no engine worker ready", "provider_error"};
+  EngineConn conn(lease.w->socket);
   if (!conn.ok())
     throw Problem{503, "Service Unavailable", "about:blank",
                   "engine connection failed", "provider_error"};
@@ -542,7 +603,11 @@ void LlmGatewayModule::embeddings_handler(HttpRequest& req,
     throw Problem{503, "Service Unavailable", "about:blank",
                   "inference engine is not ready", "provider_error"};
 
-  EngineConn conn(socket_path_);
+  Lease lease{pick_worker()};
+  if (!lease.w)
+    throw Problem{503, "Service Unavailable", "about:blank",
+                  "no engine worker ready", "provider_error"};
+  EngineConn conn(lease.w->socket);
   if (!conn.ok())
     throw Problem{503, "Service Unavailable", "about:blank",
                   "engine connection failed", "provider_error"};
@@ -960,26 +1025,37 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
       if (m_ttft_count_)
         add("hyperspot_ttft_seconds_avg", "Mean stream TTFT", "gauge",
             (double)m_ttft_us_sum_ / 1e6 / (double)m_ttft_count_);
-      if (worker_ready()) {
-        EngineConn c(socket_path_);
-        Json q = Json::object();
-        q["type"] = "info";
-        if (c.ok() && c.send_json(q)) {
-          if (auto r = c.read_json(3000)) {
-            add("hyperspot_engine_running", "Sequences decoding", "gauge",
-                (double)r->at("num_running").as_int(0));
-            add("hyperspot_engine_waiting", "Sequences queued", "gauge",
-                (double)r->at("num_waiting").as_int(0));
-            const double total = (double)r->at("kv_blocks_total").as_int(0);
-            const double freeb = (double)r->at("kv_blocks_free").as_int(0);
-            add("hyperspot_kv_blocks_total", "KV pool pages", "gauge",
-                total);
-            add("hyperspot_kv_blocks_free", "Free KV pages", "gauge",
-                freeb);
-            if (total > 0)
-              add("hyperspot_kv_occupancy", "KV pool occupancy 0..1",
-                  "gauge", (total - freeb) / total);
+      {
+        double running = 0, waiting = 0, total = 0, freeb = 0;
+        int live = 0;
+        for (auto& wk : workers_) {
+          if (!probe_worker(*wk)) continue;
+          EngineConn c(wk->socket);
+          Json q = Json::object();
+          q["type"] = "info";
+          if (c.ok() && c.send_json(q)) {
+            if (auto r = c.read_json(3000)) {
+              live++;
+              running += (double)r->at("num_running").as_int(0);
+              waiting += (double)r->at("num_waiting").as_int(0);
+              total += (double)r->at("kv_blocks_total").as_int(0);
+              freeb += (double)r->at("kv_blocks_free").as_int(0);
+            }
           }
+        }
+        add("hyperspot_workers_live", "Ready engine workers", "gauge",
+            (double)live);
+        if (live) {
+          add("hyperspot_engine_running", "Sequences decoding", "gauge",
+              running);
+          add("hyperspot_engine_waiting", "Sequences queued", "gauge",
+              waiting);
+          add("hyperspot_kv_blocks_total", "KV pool pages", "gauge",
+              total);
+          add("hyperspot_kv_blocks_free", "Free KV pages", "gauge", freeb);
+          if (total > 0)
+            add("hyperspot_kv_occupancy", "KV pool occupancy 0..1",
+                "gauge", (total - freeb) / total);
         }
       }
       // per-tenant usage
@@ -1023,7 +1099,7 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
       if (!worker_ready())
         throw Problem{503, "Service Unavailable", "about:blank",
                       "engine not ready", "provider_error"};
-      EngineConn conn(socket_path_);
+      EngineConn conn(workers_[0]->socket);
       Json wr = Json::object();
       wr["type"] = "save_checkpoint";
       wr["path"] = path;
@@ -1058,21 +1134,30 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
       if (!worker_ready())
         throw Problem{503, "Service Unavailable", "about:blank",
                       "engine not ready", "provider_error"};
-      EngineConn conn(socket_path_);
-      Json wr = Json::object();
-      wr["type"] = "swap";
-      wr["checkpoint"] = path;
-      if (!conn.ok() || !conn.send_json(wr))
-        throw Problem{502, "Bad Gateway", "about:blank",
-                      "engine write failed", "provider_error"};
-      auto r = conn.read_json(600000);
-      if (!r || r->at("event").as_string() != "swapped")
-        throw Problem{502, "Bad Gateway", "about:blank",
-                      r ? r->at("message").as_string() : "swap timed out",
-                      "provider_error"};
+      double secs = 0;
+      int swapped = 0;
+      for (auto& wk : workers_) {
+        if (!probe_worker(*wk)) continue;
+        EngineConn conn(wk->socket);
+        Json wr = Json::object();
+        wr["type"] = "swap";
+        wr["checkpoint"] = path;
+        if (!conn.ok() || !conn.send_json(wr))
+          throw Problem{502, "Bad Gateway", "about:blank",
+                        "engine write failed", "provider_error"};
+        auto r = conn.read_json(600000);
+        if (!r || r->at("event").as_string() != "swapped")
+          throw Problem{502, "Bad Gateway", "about:blank",
+                        r ? r->at("message").as_string()
+                          : "swap timed out",
+                        "provider_error"};
+        secs = std::max(secs, r->at("seconds").as_number());
+        swapped++;
+      }
       Json out = Json::object();
       out["checkpoint"] = path;
-      out["seconds"] = r->at("seconds");
+      out["seconds"] = secs;
+      out["workers_swapped"] = (long)swapped;
       w.respond(200, "application/json", out.dump());
     });
   }
@@ -1111,12 +1196,13 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
                         "\"validation_error\"}");
           continue;
         }
-        if (!worker_ready()) {
+        Lease lease{pick_worker()};
+        if (!lease.w) {
           ws->send_text("{\"type\":\"error\",\"code\":"
                         "\"provider_error\"}");
           continue;
         }
-        EngineConn conn(socket_path_);
+        EngineConn conn(lease.w->socket);
         const std::string rid =
             "rt-" + std::to_string(req_ctr_.fetch_add(1));
         Json wreq = Json::object();
@@ -1191,15 +1277,25 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
     Json out = Json::object();
     out["model"] = model_;
     out["worker_ready"] = worker_ready();
-    if (ready_) {
-      EngineConn c(socket_path_);
-      Json q = Json::object();
-      q["type"] = "info";
-      if (c.ok() && c.send_json(q)) {
-        auto r = c.read_json(3000);
-        if (r) out["engine"] = *r;
+    Json ws_json = Json::array();
+    for (auto& wk : workers_) {
+      Json wj = Json::object();
+      wj["index"] = (long)wk->index;
+      wj["device"] = (long)wk->device;
+      wj["ready"] = (bool)wk->ready;
+      wj["in_flight"] = (long)wk->in_flight.load();
+      if (probe_worker(*wk)) {
+        EngineConn c(wk->socket);
+        Json q = Json::object();
+        q["type"] = "info";
+        if (c.ok() && c.send_json(q))
+          if (auto r = c.read_json(3000)) wj["engine"] = *r;
       }
+      ws_json.push_back(wj);
     }
+    out["workers"] = ws_json;
+    if (ws_json.size() > 0 && ws_json.at(0).contains("engine"))
+      out["engine"] = ws_json.at(0).at("engine");
     w.respond(200, "application/json", out.dump());
   });
 }

@@ -67,7 +67,7 @@ class LlmGatewayModule : public Module {
   void stop(ModuleCtx& ctx) override;
 
   const std::string& socket_path() const { return socket_path_; }
-  bool worker_ready();
+  bool worker_ready();                      // any worker ready
 
  private:
   // async job state machine (DESIGN.md job/batch schemas; statuses
@@ -113,8 +113,26 @@ class LlmGatewayModule : public Module {
   bool auto_start_ = true;
   std::string python_ = "python3";
   Json worker_cfg_;
-  pid_t worker_pid_ = -1;
-  std::atomic<bool> ready_{false};
+
+  // data-parallel worker fleet: one engine process per GPU
+  // (worker.count / worker.devices config); requests go to the
+  // least-loaded ready worker
+  struct Worker {
+    int index = 0;
+    int device = 0;
+    std::string socket;
+    pid_t pid = -1;
+    std::atomic<bool> ready{false};
+    std::atomic<int> in_flight{0};
+  };
+  struct Lease {
+    Worker* w = nullptr;
+    ~Lease() { if (w) w->in_flight--; }
+  };
+  std::vector<std::unique_ptr<Worker>> workers_;
+  Worker* pick_worker();                    // least-loaded ready worker
+  bool probe_worker(Worker& wk);
+  void spawn_one(Worker& wk);
   std::atomic<uint64_t> req_ctr_{0};
 
   // jobs/batches (in-memory, tenant-scoped)

@@ -944,3 +944,72 @@ def test_checkpoint_save_and_hot_swap(server):
     st, body = _http("POST", url + "/llm-gateway/v1/checkpoints/swap",
                      {"path": "/nonexistent.safetensors"}, timeout=60)
     assert st == 502, body
+
+
+@pytest.fixture(scope="module")
+def dp_server():
+    """Two CPU engine workers behind one gateway (data-parallel fleet)."""
+    import tempfile
+    port = _free_port()
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-dp-")
+    cfg = f"""
+server:
+  home_dir: "/tmp/hs-e2e-dp"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+  llm-gateway:
+    config:
+      model: "tiny-llama"
+      worker_socket: "{sock}"
+      auto_start_worker: true
+      worker:
+        count: 2
+        device: "cpu"
+        eager: true
+        max_num_seqs: 8
+        num_gpu_blocks: 256
+"""
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(cfg)
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        srv.wait_worker()
+        yield srv
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
+
+
+def test_dp_worker_fleet(dp_server):
+    url = BASE.format(dp_server.port)
+    # both workers come up and are listed
+    deadline = time.time() + 120
+    while time.time() < deadline:
+        st, body = _http("GET", url + "/llm-gateway/v1/status")
+        ws = json.loads(body)["workers"]
+        if len(ws) == 2 and all(w["ready"] for w in ws):
+            break
+        time.sleep(0.5)
+    assert len(ws) == 2 and all(w["ready"] for w in ws), ws
+    # requests spread across the fleet and all succeed
+    import concurrent.futures as cf
+    def one(i):
+        st, body = _http("POST", url + "/v1/chat/completions",
+                         {"model": "tiny-llama",
+                          "messages": [{"role": "user", "content":
+                                        [{"type": "text",
+                                          "text": f"q{i}"}]}],
+                          "max_tokens": 3}, timeout=60)
+        return st
+    with cf.ThreadPoolExecutor(8) as ex:
+        results = list(ex.map(one, range(12)))
+    assert all(r == 200 for r in results), results
+    st, body = _http("GET", url + "/metrics")
+    assert "hyperspot_workers_live 2" in body
