@@ -64,7 +64,8 @@ std::optional<ClientResponse> http_request(
     const std::string& target,
     const std::map<std::string, std::string>& headers,
     const std::string& body, int connect_timeout_ms,
-    const std::function<bool(const char*, size_t)>& on_chunk) {
+    const std::function<bool(const char*, size_t)>& on_chunk,
+    const std::function<void(const ClientResponse&)>& on_headers) {
   int fd = connect_to(host, port, connect_timeout_ms);
   if (fd < 0) return std::nullopt;
   std::ostringstream req;
@@ -120,6 +121,7 @@ std::optional<ClientResponse> http_request(
       pos = e + 2;
     }
   }
+  if (on_headers) on_headers(resp);
   std::string rest = buf.substr(hdr_end + 4);
   const bool chunked =
       resp.headers.count("transfer-encoding") &&

@@ -17,11 +17,14 @@ struct ClientResponse {
 };
 
 // on_chunk: called per body chunk in streaming mode; return false to abort.
+// on_headers: called once after the status line + headers are parsed (lets
+// a proxy begin its own stream before the body arrives — SSE pass-through).
 std::optional<ClientResponse> http_request(
     const std::string& host, int port, const std::string& method,
     const std::string& target,
     const std::map<std::string, std::string>& headers,
     const std::string& body, int connect_timeout_ms = 10000,
-    const std::function<bool(const char*, size_t)>& on_chunk = nullptr);
+    const std::function<bool(const char*, size_t)>& on_chunk = nullptr,
+    const std::function<void(const ClientResponse&)>& on_headers = nullptr);
 
 }  // namespace hs

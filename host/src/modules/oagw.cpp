@@ -120,7 +120,40 @@ void OagwModule::proxy(HttpRequest& req, ResponseWriter& w) {
   if (req.target.find('?') != std::string::npos)
     target += req.target.substr(req.target.find('?'));
 
-  // forward (buffered; the llm engine path does not go through OAGW)
+  // SSE pass-through (reference oagw: no total timeout so SSE can
+  // stream, src/infra/proxy/service.rs:43-49): when the client asks for
+  // an event stream, chunks are forwarded as they arrive
+  if (req.header("accept").find("text/event-stream") != std::string::npos) {
+    bool started = false;
+    int upstatus = 200;
+    std::string upct = "text/event-stream";
+    auto r = http_request(
+        host, port, req.method, target, fwd, req.body, 10000,
+        [&](const char* p, size_t n) {
+          if (!started) {
+            w.begin_stream(upstatus, upct, {{"x-oagw-upstream", alias}});
+            started = true;
+          }
+          return w.write_chunk(std::string(p, n));
+        },
+        [&](const ClientResponse& hr) {
+          upstatus = hr.status;
+          auto it = hr.headers.find("content-type");
+          if (it != hr.headers.end()) upct = it->second;
+        });
+    if (started) {
+      w.end_stream();
+      return;
+    }
+    if (!r)
+      throw Problem{502, "Bad Gateway", "about:blank",
+                    "upstream connect failed", "provider_error"};
+    // empty-bodied response: fall through with the buffered result
+    w.respond(r->status, upct, "", {{"x-oagw-upstream", alias}});
+    return;
+  }
+
+  // forward (buffered)
   auto resp = http_request(host, port, req.method, target, fwd, req.body,
                            10000);
   if (!resp)

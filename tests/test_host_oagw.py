@@ -16,6 +16,17 @@ class MockUpstream(http.server.BaseHTTPRequestHandler):
     """Echo server recording received headers."""
 
     def _respond(self):
+        if self.path.startswith("/events"):
+            # SSE endpoint: 3 events, trickled
+            import time
+            self.send_response(200)
+            self.send_header("content-type", "text/event-stream")
+            self.end_headers()
+            for i in range(3):
+                self.wfile.write(f"data: ev{i}\n\n".encode())
+                self.wfile.flush()
+                time.sleep(0.05)
+            return
         length = int(self.headers.get("content-length", 0))
         body = self.rfile.read(length) if length else b""
         out = json.dumps({
@@ -164,3 +175,23 @@ def test_module_orchestrator_lists_modules(server):
                      "/module-orchestrator/v1/modules")
     names = [m["name"] for m in json.loads(body)["items"]]
     assert "llm-gateway" in names and "oagw" in names
+
+
+def test_proxy_sse_passthrough(server, upstream):
+    """SSE streams through the OAGW data plane chunk-by-chunk (reference
+    service.rs: no total timeout so SSE can stream)."""
+    import urllib.request
+    base = BASE.format(server.port)
+    up = {"alias": "ssesvc", "server": {"endpoints": [
+        {"scheme": "http", "host": "127.0.0.1", "port": upstream}]},
+        "protocol": "gts.x.core.net.protocol.v1~x.core.http.rest.v1",
+        "enabled": True}
+    st, body = _http("POST", base + "/oagw/v1/upstreams", body=up)
+    assert st == 201, body
+    req = urllib.request.Request(base + "/oagw/v1/proxy/ssesvc/events")
+    req.add_header("accept", "text/event-stream")
+    with urllib.request.urlopen(req, timeout=30) as r:
+        assert r.headers.get("content-type", "").startswith(
+            "text/event-stream")
+        data = r.read().decode()
+    assert "data: ev0" in data and "data: ev2" in data
