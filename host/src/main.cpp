@@ -17,6 +17,7 @@
 #include "modules/llm_gateway.h"
 #include "modules/user_settings.h"
 #include "modules/file_parser.h"
+#include "modules/users_info.h"
 #include "modules/oagw.h"
 #include "modules/system_modules.h"
 #include "util/log.h"
@@ -55,6 +56,7 @@ int run_server(const Json& cfg, bool check_only) {
   registry.add(std::make_shared<LlmGatewayModule>());
   registry.add(std::make_shared<UserSettingsModule>());
   registry.add(std::make_shared<FileParserModule>());
+  registry.add(std::make_shared<UsersInfoModule>());
 
   ClientHub hub;
   bool cancel_flag = false;
@@ -112,7 +114,8 @@ int list_modules(const Json& cfg) {
                         "nodes-registry", "model-registry", "credstore",
                         "serverless-runtime", "oagw", "file-storage",
                         "module-orchestrator", "llm-gateway",
-                        "simple-user-settings", "file-parser"})
+                        "simple-user-settings", "file-parser",
+                        "users-info"})
     std::cout << m << "\n";
   return 0;
 }

@@ -332,6 +332,9 @@ modules:
   simple-user-settings:
     database:
       file: "{tempfile.mktemp(suffix='.db', prefix='hs-sus-')}"
+  users-info:
+    database:
+      file: "{tempfile.mktemp(suffix='.db', prefix='hs-ui-')}"
 """
     cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
     cfg_path.write_text(cfg)
@@ -1013,3 +1016,87 @@ def test_dp_worker_fleet(dp_server):
     assert all(r == 200 for r in results), results
     st, body = _http("GET", url + "/metrics")
     assert "hyperspot_workers_live 2" in body
+
+
+def test_users_info_crud_pagination_isolation(mt_server):
+    """The users-info blueprint: CRUD + unique-email conflict + cursor
+    pagination + tenant isolation (the reference example's test matrix)."""
+    url = BASE.format(mt_server.port)
+    st, body = _http("POST", url + "/users-info/v1/cities",
+                     {"name": "Utrecht"}, token="acme-token")
+    assert st == 201, body
+    city = json.loads(body)["id"]
+    ids = []
+    for i in range(5):
+        st, body = _http("POST", url + "/users-info/v1/users",
+                         {"email": f"u{i}@acme.io", "display_name": f"U{i}",
+                          "city_id": city}, token="acme-token")
+        assert st == 201, body
+        ids.append(json.loads(body)["id"])
+    # unique email per tenant
+    st, body = _http("POST", url + "/users-info/v1/users",
+                     {"email": "u0@acme.io"}, token="acme-token")
+    assert st == 409
+    # same email under ANOTHER tenant is fine (tenant-scoped index)
+    st, _ = _http("POST", url + "/users-info/v1/users",
+                  {"email": "u0@acme.io"}, token="root-token")
+    assert st == 201
+    # pagination walks all 5 in email order
+    seen, cursor = [], ""
+    while True:
+        q = "?$top=2"
+        if cursor:
+            q += f"&cursor={cursor}"
+        st, body = _http("GET", url + "/users-info/v1/users" + q,
+                         token="acme-token")
+        d = json.loads(body)
+        seen += [u["email"] for u in d["items"]]
+        cursor = d["page_info"].get("next_cursor", "")
+        if not cursor:
+            break
+    assert seen == [f"u{i}@acme.io" for i in range(5)], seen
+    # root tenant doesn't see acme's display names
+    st, body = _http("GET", url + "/users-info/v1/users",
+                     token="root-token")
+    emails = [u["email"] for u in json.loads(body)["items"]]
+    assert emails == ["u0@acme.io"]
+    # filter
+    from urllib.parse import quote
+    st, body = _http("GET", url + "/users-info/v1/users?$filter="
+                     + quote("email eq 'u3@acme.io'"), token="acme-token")
+    assert [u["email"] for u in json.loads(body)["items"]] == ["u3@acme.io"]
+    # get/delete + sdk-visible count via another module is implicit
+    st, _ = _http("DELETE", url + f"/users-info/v1/users/{ids[0]}",
+                  token="acme-token")
+    assert st == 204
+    st, _ = _http("GET", url + f"/users-info/v1/users/{ids[0]}",
+                  token="acme-token")
+    assert st == 404
+
+
+def test_users_info_sse_events(mt_server):
+    """SSE event stream delivers user.created while subscribed."""
+    import queue
+    url = BASE.format(mt_server.port)
+    q = queue.Queue()
+
+    def listen():
+        req = urllib.request.Request(url + "/users-info/v1/users/events")
+        req.add_header("authorization", "Bearer acme-token")
+        with urllib.request.urlopen(req, timeout=30) as r:
+            for raw in r:
+                line = raw.decode().strip()
+                if line.startswith("data: "):
+                    q.put(json.loads(line[6:]))
+                    return
+
+    import threading
+    t = threading.Thread(target=listen, daemon=True)
+    t.start()
+    time.sleep(0.5)
+    st, _ = _http("POST", url + "/users-info/v1/users",
+                  {"email": "sse@acme.io"}, token="acme-token")
+    assert st == 201
+    ev = q.get(timeout=20)
+    assert ev["type"] == "user.created"
+    assert ev["user"]["email"] == "sse@acme.io"
