@@ -67,6 +67,9 @@ class LLMEngine:
         tokens = self.runner.execute(batch)
         outputs: List[StepOutput] = []
         for req, tok in zip(batch.requests, tokens.tolist()):
+            if batch.mode == "prefill" and \
+                    req.num_computed_tokens < req.num_prompt_tokens:
+                continue     # mid-prompt chunk: no token is sampled yet
             req.append_output(int(tok), self.eos_token_id)
             outputs.append(StepOutput(req.request_id, int(tok), req.finished,
                                       req.finish_reason))

@@ -68,6 +68,26 @@ class BlockManager:
         self.tokens_np[row] = num_tokens
         return blocks
 
+    def can_extend(self, seq_id: str, num_tokens: int,
+                   watermark: int = 0) -> bool:
+        row = self.row_of[seq_id]
+        need = self.blocks_needed(int(self.tokens_np[row]) + num_tokens) \
+            - int(self.ntables_np[row])
+        return need <= self.num_free - watermark
+
+    def extend(self, seq_id: str, num_tokens: int) -> None:
+        """Grow an existing sequence by num_tokens (chunked prefill)."""
+        row = self.row_of[seq_id]
+        total = int(self.tokens_np[row]) + num_tokens
+        need = self.blocks_needed(total) - int(self.ntables_np[row])
+        if need > self.num_free:
+            raise RuntimeError("KV pool exhausted")
+        for _ in range(need):
+            nt = int(self.ntables_np[row])
+            self.tables_np[row, nt] = self.free_blocks.pop()
+            self.ntables_np[row] = nt + 1
+        self.tokens_np[row] = total
+
     def tokens_of(self, seq_id: str) -> int:
         return int(self.tokens_np[self.row_of[seq_id]])
 

@@ -116,12 +116,16 @@ class ModelRunner:
         tables = []
         bm = self.block_manager
         bs = bm.block_size
+        fresh = True
         for si, req in enumerate(batch.requests):
-            n = req.num_prompt_tokens
-            ids.extend(req.prompt_token_ids)
-            pos.extend(range(n))
+            cs = req.chunk_start
+            n = req.chunk_len or req.num_prompt_tokens
+            if cs > 0:
+                fresh = False      # continuation chunk attends over cache
+            ids.extend(req.prompt_token_ids[cs:cs + n])
+            pos.extend(range(cs, cs + n))
             row = bm.row_of[req.request_id]
-            p = np.arange(n)
+            p = np.arange(cs, cs + n)
             s = bm.tables_np[row, p // bs].astype(np.int64) * bs + p % bs
             slots.extend(s.tolist())
             starts.append(starts[-1] + n)
@@ -139,7 +143,9 @@ class ModelRunner:
             positions=positions,
             slot_mapping=torch.tensor(slots, dtype=torch.long, device=d),
             seq_start=torch.tensor(starts, dtype=torch.int32, device=d),
-            max_seqlen=max(r.num_prompt_tokens for r in batch.requests),
+            max_seqlen=max((r.chunk_len or r.num_prompt_tokens)
+                           for r in batch.requests),
+            fresh_prefill=fresh,
             row_seq=torch.tensor(row_seq, dtype=torch.int32, device=d),
             ctx_lens=(positions + 1).to(torch.int32),
             block_tables=bt.to(d),

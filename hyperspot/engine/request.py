@@ -34,6 +34,11 @@ class Request:
     arrival_time: float = field(default_factory=time.monotonic)
     output_token_ids: List[int] = field(default_factory=list)
     num_generated: int = 0      # survives preemption/recompute
+    # chunked prefill progress (tokens whose KV is computed) + the chunk
+    # scheduled for the CURRENT step
+    num_computed_tokens: int = 0
+    chunk_start: int = 0
+    chunk_len: int = 0
     state: RequestState = RequestState.WAITING
     finish_reason: Optional[FinishReason] = None
     first_token_time: Optional[float] = None

@@ -65,35 +65,47 @@ class Scheduler:
         return self._schedule_decode()
 
     def _schedule_prefill(self) -> ScheduledBatch:
+        """Chunked prefill: a prompt longer than the per-step token budget
+        is computed over several steps (head-of-line chunks); its KV grows
+        via BlockManager.extend and it only joins the decode set once the
+        last chunk is in.  Whole short prompts pack into one step as
+        before."""
         batch = ScheduledBatch(mode="prefill")
         budget = self.config.max_num_batched_tokens
-        while self.waiting:
+        completed = []
+        while self.waiting and budget > 0:
             req = self.waiting[0]
-            n = req.num_prompt_tokens
-            if n > self.config.max_model_len:
+            total = req.num_prompt_tokens
+            if total > self.config.max_model_len:
                 self.waiting.popleft()
                 req.state = RequestState.FINISHED
                 continue
-            if len(self.running) + len(batch.requests) >= self.config.max_num_seqs:
+            done = req.num_computed_tokens
+            if done == 0 and len(self.running) + len(batch.requests)                     >= self.config.max_num_seqs:
                 break
-            if n > budget:
-                # a prompt longer than the per-step token budget would
-                # starve under whole-prompt admission: let it run ALONE
-                # (the MFMA prefill kernel handles any seqlen)
-                if batch.empty and n <= self.config.max_model_len:
-                    budget = n
-                else:
+            chunk = min(total - done, budget)
+            if chunk <= 0:
+                break
+            if done == 0:
+                if not self.bm.can_allocate(chunk, self.watermark):
                     break
-            if not self.bm.can_allocate(n, self.watermark):
-                break
-            self.waiting.popleft()
-            self.bm.allocate(req.request_id, n)
-            req.state = RequestState.RUNNING
+                self.bm.allocate(req.request_id, chunk)
+            else:
+                if not self.bm.can_extend(req.request_id, chunk,
+                                          self.watermark):
+                    break
+                self.bm.extend(req.request_id, chunk)
+            req.chunk_start, req.chunk_len = done, chunk
+            req.num_computed_tokens = done + chunk
             batch.requests.append(req)
-            budget -= n
-            batch.num_tokens += n
-        if batch.requests:
-            self.running.extend(batch.requests)
+            budget -= chunk
+            batch.num_tokens += chunk
+            if req.num_computed_tokens == total:
+                self.waiting.popleft()
+                req.state = RequestState.RUNNING
+                completed.append(req)
+            # else: head-of-line request keeps its place; budget is 0 now
+        self.running.extend(completed)
         return batch
 
     def _schedule_decode(self) -> ScheduledBatch:
@@ -113,6 +125,7 @@ class Scheduler:
             # recompute path: prompt + generated so far becomes the new prompt
             victim.prompt_token_ids = victim.prompt_token_ids + victim.output_token_ids
             victim.output_token_ids = []
+            victim.num_computed_tokens = 0
             self.waiting.appendleft(victim)
         batch.requests = list(self.running)
         batch.num_tokens = len(batch.requests)

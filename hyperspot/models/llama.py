@@ -47,6 +47,11 @@ class ForwardMeta:
     logits_indices: Optional[torch.Tensor] = None
     # embeddings mode: return final hidden states instead of logits
     return_hidden: bool = False
+    # prefill: True when every row starts at position 0 (chunk == whole
+    # prompt) -> the self-contained MFMA flash kernel applies; False for
+    # chunked-prefill continuation batches, which must attend over the
+    # paged cache (per-row ctx_lens)
+    fresh_prefill: bool = True
 
 
 class Attention(nn.Module):

@@ -130,17 +130,22 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                                  meta.seq_lens, scale)
     if q.is_cuda:
         out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
-        if q.shape[-1] == 128:
+        if q.shape[-1] == 128 and getattr(meta, "fresh_prefill", True):
             # MFMA flash prefill (K/V straight from the qkv projection)
             _native().attn_prefill_mfma(out, q, k, v, meta.seq_start,
                                         meta.max_seqlen, scale)
         else:
-            # fallback: per-row attention over the freshly appended cache
+            # per-row attention over the (freshly appended) paged cache:
+            # chunked-prefill continuation rows or odd head dims
             _native().paged_attn(out, q, k_cache, v_cache,
                                  meta.block_tables, meta.ctx_lens,
                                  meta.row_seq, scale, None, 1)
         return out
-    return torch_ref.prefill_attn(q, k, v, meta.seq_start, scale)
+    if getattr(meta, "fresh_prefill", True):
+        return torch_ref.prefill_attn(q, k, v, meta.seq_start, scale)
+    return torch_ref.prefill_attn_paged(q, k_cache, v_cache,
+                                        meta.block_tables, meta.ctx_lens,
+                                        meta.row_seq, scale)
 
 
 def silu_mul(gate_up: torch.Tensor) -> torch.Tensor:

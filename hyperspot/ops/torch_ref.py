@@ -97,6 +97,33 @@ def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
     return out
 
 
+def prefill_attn_paged(q: torch.Tensor, k_cache: torch.Tensor,
+                       v_cache: torch.Tensor, block_tables: torch.Tensor,
+                       ctx_lens: torch.Tensor, row_seq: torch.Tensor,
+                       scale: float) -> torch.Tensor:
+    """Per-row causal attention over the paged cache (chunked-prefill
+    continuation reference: row t attends slots 0..ctx_lens[t])."""
+    T, H, D = q.shape
+    nb, KV, BS, _ = k_cache.shape
+    group = H // KV
+    out = torch.empty_like(q)
+    for t in range(T):
+        si = int(row_seq[t])
+        c = int(ctx_lens[t])
+        nblk = (c + BS - 1) // BS
+        blocks = block_tables[si, :nblk].long()
+        k = k_cache[blocks].permute(1, 0, 2, 3).reshape(KV, nblk * BS,
+                                                        D)[:, :c].float()
+        v = v_cache[blocks].permute(1, 0, 2, 3).reshape(KV, nblk * BS,
+                                                        D)[:, :c].float()
+        qt = q[t].float().view(KV, group, D)
+        att = torch.einsum("kgd,knd->kgn", qt, k) * scale
+        p = att.softmax(-1)
+        o = torch.einsum("kgn,knd->kgd", p, v)
+        out[t] = o.reshape(H, D).to(q.dtype)
+    return out
+
+
 def prefill_attn(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                  seq_start: torch.Tensor, scale: float) -> torch.Tensor:
     """Varlen causal self-attention for prefill.

@@ -160,3 +160,33 @@ def test_long_prompt_exceeding_step_budget_is_not_starved():
     out = eng.generate([long_prompt, [1, 2, 3]],
                        SamplingParams(temperature=0.0, max_tokens=4))
     assert len(out[0]) == 4 and len(out[1]) == 4
+
+
+def test_chunked_prefill_matches_unchunked():
+    """A 199-token prompt through a 64-token/step budget (4 chunks, the
+    continuation chunks attending over the paged cache) must produce the
+    same greedy tokens as whole-prompt prefill."""
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    prompt = [(i * 37) % 500 + 1 for i in range(199)]
+    outs = []
+    for budget in (512, 64):
+        cfg = EngineConfig(model="tiny-llama", max_num_seqs=4,
+                           max_num_batched_tokens=budget, max_model_len=512,
+                           num_gpu_blocks=64, enforce_eager=True, seed=5)
+        eng = LLMEngine(cfg, device="cpu")
+        outs.append(eng.generate(
+            [prompt], SamplingParams(temperature=0.0, max_tokens=6))[0])
+    assert outs[0] == outs[1], outs
+
+
+def test_chunked_prefill_interleaves_short_prompts():
+    """Short prompts keep flowing while a long prompt chunks through."""
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    cfg = EngineConfig(model="tiny-llama", max_num_seqs=4,
+                       max_num_batched_tokens=48, max_model_len=512,
+                       num_gpu_blocks=96, enforce_eager=True)
+    eng = LLMEngine(cfg, device="cpu")
+    long_p = [(i % 90) + 2 for i in range(150)]
+    out = eng.generate([long_p, [5, 6, 7], [9, 10]],
+                       SamplingParams(temperature=0.0, max_tokens=4))
+    assert all(len(o) == 4 for o in out), out

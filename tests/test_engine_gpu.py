@@ -70,3 +70,24 @@ def test_mixtral_tiny_runs_on_gpu():
     out = eng.generate([[1, 2, 3]], SamplingParams(temperature=0.0,
                                                    max_tokens=4))[0]
     assert len(out) == 4
+
+
+@pytest.mark.gpu
+def test_chunked_prefill_matches_unchunked_gpu():
+    """Continuation chunks route through the paged attention kernel on GPU
+    (fresh_prefill=False); greedy output must match whole-prompt prefill."""
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    prompt = [(i * 131) % 120000 + 2 for i in range(700)]
+    outs = []
+    for budget in (8192, 256):
+        cfg = EngineConfig(model="llama3-8b", max_num_seqs=4,
+                           max_num_batched_tokens=budget,
+                           max_model_len=1024, num_gpu_blocks=256,
+                           enforce_eager=True, seed=11)
+        eng = LLMEngine(cfg, device="cuda:0")
+        outs.append(eng.generate(
+            [prompt], SamplingParams(temperature=0.0, max_tokens=5))[0])
+        del eng
+        import torch
+        torch.cuda.empty_cache()
+    assert outs[0] == outs[1], outs
