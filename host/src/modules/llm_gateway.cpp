@@ -881,6 +881,18 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
       if (it->second->status == "queued") {
         it->second->status = "cancelled";
         it->second->finished_at = now_s();
+      } else if (it->second->status == "running") {
+        // abort propagation: the engine request id IS the job id, so a
+        // running job can be cut short server-side (DESIGN timeouts /
+        // cancellation machinery)
+        for (auto& wk : workers_) {
+          if (!wk->ready) continue;
+          EngineConn c(wk->socket);
+          Json ab = Json::object();
+          ab["type"] = "abort";
+          ab["id"] = it->first;
+          if (c.ok()) c.send_json(ab);
+        }
       }
       w.respond(200, "application/json", job_json(*it->second).dump());
     });

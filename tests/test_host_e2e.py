@@ -1100,3 +1100,24 @@ def test_users_info_sse_events(mt_server):
     ev = q.get(timeout=20)
     assert ev["type"] == "user.created"
     assert ev["user"]["email"] == "sse@acme.io"
+
+
+def test_openapi_document_contract(server):
+    """OpenAPI 3.1 document lists the module routes with auth + schemas."""
+    st, body = _http("GET", BASE.format(server.port) + "/openapi.json")
+    assert st == 200
+    doc = json.loads(body)
+    assert doc["openapi"].startswith("3.1")
+    paths = doc["paths"]
+    for p in ["/llm-gateway/v1/chat/completions", "/v1/chat/completions",
+              "/llm-gateway/v1/embeddings", "/llm-gateway/v1/jobs",
+              "/llm-gateway/v1/jobs/{id}", "/llm-gateway/v1/batches",
+              "/simple-user-settings/v1/settings",
+              "/users-info/v1/users", "/users-info/v1/users/events",
+              "/file-parser/v1/info", "/oagw/v1/upstreams",
+              "/module-orchestrator/v1/instances", "/metrics"]:
+        assert p in paths, p
+    chat = paths["/llm-gateway/v1/chat/completions"]["post"]
+    assert chat["security"], "chat must be authenticated"
+    schema = chat["requestBody"]["content"]["application/json"]["schema"]
+    assert set(schema["required"]) == {"model", "messages"}
