@@ -251,13 +251,20 @@ SecureConn::Page SecureConn::select(
 void SecureConn::insert(
     const std::string& table,
     const std::vector<std::pair<std::string, DbValue>>& cols) {
-  if (scope_.is_deny_all() ||
-      (!scope_.unrestricted && scope_.tenant_ids.size() != 1))
+  // writes need an unambiguous tenant: a single-tenant scope (the tenant
+  // column is FORCED from it), or allow_all WITH an explicit tenant col
+  if (scope_.is_deny_all())
+    throw std::runtime_error("insert under deny-all scope");
+  if (!scope_.unrestricted && scope_.tenant_ids.size() != 1)
     throw std::runtime_error("insert requires a single-tenant scope");
+  bool caller_tenant = false;
   std::string names, marks;
   std::vector<DbValue> binds;
   for (auto& [n, v] : cols) {
-    if (n == tenant_col_) continue;   // forced from scope below
+    if (n == tenant_col_) {
+      if (!scope_.unrestricted) continue;   // forced from scope below
+      caller_tenant = true;
+    }
     if (!names.empty()) { names += ","; marks += ","; }
     names += n;
     marks += "?";
@@ -267,6 +274,9 @@ void SecureConn::insert(
     names += std::string(names.empty() ? "" : ",") + tenant_col_;
     marks += std::string(marks.empty() ? "" : ",") + "?";
     binds.push_back(DbValue::S(scope_.tenant_ids[0]));
+  } else if (!caller_tenant) {
+    throw std::runtime_error(
+        "unrestricted insert must name the tenant column explicitly");
   }
   std::string sql = "INSERT INTO " + table + " (" + names + ") VALUES (" +
                     marks + ")";
