@@ -27,8 +27,23 @@ class ByteTokenizer:
         return ([BOS] if add_bos else []) + ids
 
     def decode(self, ids: List[int]) -> str:
-        data = bytes(i - OFFSET for i in ids if OFFSET <= i < 256 + OFFSET)
-        return data.decode("utf-8", errors="replace")
+        # out-of-byte-range ids (everything a random-init model samples)
+        # render as a visible surrogate so streams are never empty
+        parts: List[str] = []
+        buf = bytearray()
+        for i in ids:
+            if OFFSET <= i < 256 + OFFSET:
+                buf.append(i - OFFSET)
+            else:
+                if buf:
+                    parts.append(bytes(buf).decode("utf-8",
+                                                   errors="replace"))
+                    buf.clear()
+                if i not in (BOS, EOS):
+                    parts.append(f"<{i}>")
+        if buf:
+            parts.append(bytes(buf).decode("utf-8", errors="replace"))
+        return "".join(parts)
 
 
 class StreamDetokenizer:
@@ -40,7 +55,11 @@ class StreamDetokenizer:
 
     def push(self, token_id: int) -> str:
         if not (OFFSET <= token_id < 256 + OFFSET):
-            return ""
+            # visible surrogate for non-byte tokens (random-init serving)
+            pre = self.flush()
+            if token_id in (BOS, EOS):
+                return pre
+            return pre + f"<{token_id}>"
         self.buf += bytes([token_id - OFFSET])
         try:
             out = self.buf.decode("utf-8")
