@@ -216,9 +216,20 @@ void LlmGatewayModule::spawn_one(Worker& wk) {
       setenv("HIP_VISIBLE_DEVICES", dev.c_str(), 1);
       setenv("CUDA_VISIBLE_DEVICES", dev.c_str(), 1);
     }
-    std::vector<std::string> args = {
-        python_, "-m", "hyperspot.serving.worker",
-        "--uds", wk.socket, "--model", model_};
+    // tp>1 workers are an SPMD torchrun group (one rank per GPU, RCCL
+    // over xGMI); rank 0 owns the socket and broadcasts the op log
+    const long tp = worker_cfg_.is_object()
+                        ? worker_cfg_.at("tp").as_int(1) : 1;
+    std::vector<std::string> args;
+    if (tp > 1) {
+      args = {python_, "-m", "torch.distributed.run", "--standalone",
+              "--local-addr", "127.0.0.1", "--nnodes", "1",
+              "--nproc-per-node", std::to_string(tp), "-m",
+              "hyperspot.serving.worker"};
+    } else {
+      args = {python_, "-m", "hyperspot.serving.worker"};
+    }
+    args.insert(args.end(), {"--uds", wk.socket, "--model", model_});
     if (worker_cfg_.is_object()) {
       if (worker_cfg_.contains("max_num_seqs")) {
         args.push_back("--max-num-seqs");

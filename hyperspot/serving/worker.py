@@ -30,13 +30,22 @@ log = logging.getLogger("hyperspot.worker")
 
 
 class WorkerState:
-    def __init__(self, engine, tokenizer):
+    """Engine driver.  At tp>1 the worker runs under torchrun as an SPMD
+    group (one rank per GPU, RCCL over xGMI): rank 0 owns the UDS socket
+    and, before every engine step, broadcasts the op log (submits/aborts)
+    so every rank's continuous-batching scheduler makes IDENTICAL
+    decisions (same config, same seed, same op order ⇒ same batches; the
+    TP all-reduces inside the forward keep ranks in lockstep)."""
+
+    def __init__(self, engine, tokenizer, tp: int = 1):
         self.engine = engine
         self.tokenizer = tokenizer
+        self.tp = tp
         self.lock = threading.Lock()          # guards engine scheduler
         self.new_work = threading.Condition(self.lock)
         self.streams = {}                     # rid -> queue.Queue
         self.started = {}                     # rid -> prompt_len
+        self.pending_ops = []                 # ops since last broadcast
         self.stop = False
 
     def submit(self, rid, prompt_ids, sampling):
@@ -45,6 +54,9 @@ class WorkerState:
             self.streams[rid] = q
             self.started[rid] = len(prompt_ids)
             self.engine.add_request(prompt_ids, sampling, request_id=rid)
+            if self.tp > 1:
+                self.pending_ops.append(
+                    ("add", rid, prompt_ids, sampling.__dict__.copy()))
             self.new_work.notify()
         return q
 
@@ -52,14 +64,24 @@ class WorkerState:
         with self.new_work:
             self.engine.abort_request(rid)
             self.streams.pop(rid, None)
+            if self.tp > 1:
+                self.pending_ops.append(("abort", rid))
+                self.new_work.notify()
 
     def step_loop(self):
+        import torch.distributed as dist
         while not self.stop:
             with self.new_work:
-                while not self.engine.has_work() and not self.stop:
+                while not self.engine.has_work() and not self.stop                         and not self.pending_ops:
                     self.new_work.wait(timeout=0.5)
                 if self.stop:
+                    if self.tp > 1:
+                        dist.broadcast_object_list([("stop",)], src=0)
                     return
+                if self.tp > 1:
+                    ops = self.pending_ops
+                    self.pending_ops = []
+                    dist.broadcast_object_list([("step", ops)], src=0)
                 outputs = self.engine.step()
             for out in outputs:
                 q = self.streams.get(out.request_id)
@@ -67,6 +89,27 @@ class WorkerState:
                     q.put(out)
                     if out.finished:
                         self.streams.pop(out.request_id, None)
+
+
+def follower_loop(engine):
+    """tp>1, rank>0: apply rank 0's op log and step in lockstep."""
+    import torch.distributed as dist
+    from hyperspot.engine import SamplingParams
+    while True:
+        box = [None]
+        dist.broadcast_object_list(box, src=0)
+        msg = box[0]
+        if msg[0] == "stop":
+            return
+        _, ops = msg
+        for op in ops:
+            if op[0] == "add":
+                _, rid, prompt_ids, sp = op
+                engine.add_request(prompt_ids, SamplingParams(**sp),
+                                   request_id=rid)
+            elif op[0] == "abort":
+                engine.abort_request(op[1])
+        engine.step()
 
 
 def handle_conn(conn: socket.socket, state: WorkerState, model_name: str):
@@ -99,9 +142,17 @@ def handle_conn(conn: socket.socket, state: WorkerState, model_name: str):
             elif t == "chat":
                 _run_chat(msg, state, send)
             elif t == "embeddings":
+                if state.tp > 1:
+                    send({"event": "error", "message":
+                          "embeddings with tp>1 not supported yet"})
+                    continue
                 _run_embeddings(msg, state, send)
             elif t == "swap":
                 path = msg.get("checkpoint", "")
+                if state.tp > 1:
+                    send({"event": "error", "message":
+                          "hot-swap with tp>1 not supported yet"})
+                    continue
                 try:
                     with state.lock:
                         secs = state.engine.swap_weights(path)
@@ -234,16 +285,26 @@ def main():
         num_gpu_blocks=args.num_gpu_blocks or None,
         enforce_eager=args.eager or not on_gpu, tp_size=args.tp,
         quant=args.quant, kv_dtype=args.kv_dtype)
+    rank = 0
     if args.tp > 1:
         from hyperspot.parallel.state import initialize_model_parallel
         initialize_model_parallel(tp_size=args.tp)
-    log.info("loading engine model=%s device=%s", args.model, args.device)
+        import torch.distributed as dist
+        rank = dist.get_rank()
+        if on_gpu:
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+            args.device = f"cuda:{os.environ.get('LOCAL_RANK', rank)}"
+    log.info("loading engine model=%s device=%s rank=%d", args.model,
+             args.device, rank)
     eng = LLMEngine(cfg, device=args.device,
                     eos_token_id=ByteTokenizer(cfg.spec().vocab_size).eos_token_id)
     if not cfg.enforce_eager:
         eng.capture_graphs()
+    if args.tp > 1 and rank != 0:
+        follower_loop(eng)      # blocks until rank 0 broadcasts stop
+        return
     tok = ByteTokenizer(cfg.spec().vocab_size)
-    state = WorkerState(eng, tok)
+    state = WorkerState(eng, tok, tp=args.tp)
     stepper = threading.Thread(target=state.step_loop, daemon=True)
     stepper.start()
 
