@@ -40,6 +40,9 @@ void launch_moe_gemm(bf16*, const bf16*, const bf16*, const int*,
                      const int*, int, int, int, int, hipStream_t);
 void launch_moe_combine(bf16*, const bf16*, const float*, const int*, long,
                         int, int, hipStream_t);
+void launch_moe_gemm_fp8(bf16*, const unsigned char*, const float*,
+                         const unsigned char*, const float*, const int*,
+                         const int*, int, int, int, int, hipStream_t);
 void launch_greedy_sample(long*, const bf16*, long, int, hipStream_t);
 void launch_inv_cdf_sample(long*, const float*, const float*, long, int,
                            hipStream_t);
@@ -276,6 +279,26 @@ void moe_combine(torch::Tensor out, torch::Tensor y, torch::Tensor wts,
                      (int)out.size(-1), stream());
 }
 
+
+void moe_gemm_fp8(torch::Tensor out, torch::Tensor xq, torch::Tensor xs,
+                  torch::Tensor wq, torch::Tensor ws,
+                  torch::Tensor sorted_ids, torch::Tensor tile_expert,
+                  long gather_div) {
+  check(out, torch::kBFloat16, "out");
+  check(xq, torch::kFloat8_e4m3fn, "xq");
+  check(xs, torch::kFloat, "xs");
+  check(wq, torch::kFloat8_e4m3fn, "wq");
+  check(ws, torch::kFloat, "ws");
+  check(sorted_ids, torch::kInt, "sorted_ids");
+  check(tile_expert, torch::kInt, "tile_expert");
+  TORCH_CHECK(wq.dim() == 3, "wq must be [E, N, K]");
+  launch_moe_gemm_fp8(bf(out), u8(xq), xs.data_ptr<float>(), u8(wq),
+                      ws.data_ptr<float>(), sorted_ids.data_ptr<int>(),
+                      tile_expert.data_ptr<int>(),
+                      (int)tile_expert.numel(), (int)wq.size(1),
+                      (int)wq.size(2), (int)gather_div, stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "HyperSpot-AMD gfx950 (CDNA4) kernels";
   m.def("rmsnorm", &rmsnorm);
@@ -285,6 +308,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_align", &moe_align);
   m.def("moe_gemm", &moe_gemm);
   m.def("moe_combine", &moe_combine);
+  m.def("moe_gemm_fp8", &moe_gemm_fp8);
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm);
   m.def("rope_kv_append", &rope_kv_append);
   m.def("paged_attn", &paged_attn);

@@ -217,3 +217,114 @@ void launch_moe_combine(bf16* out, const bf16* y, const float* wts,
   moe_combine_kernel<<<dim3((unsigned)T), 256, 0, stream>>>(
       out, y, wts, inv_pos, topk, hidden);
 }
+
+// ----------------------------------------------------- fp8 grouped GEMM
+// Same 128x128 tiling as moe_gemm_kernel, e4m3fn operands:
+// out[p,n] = (sum_k A8[p,k] * B8[e,n,k]) * a_scale[row(p)] * b_scale[e,n]
+// via v_mfma_f32_32x32x16_fp8_fp8 (identical C layout to the bf16 tile;
+// each lane feeds 8 fp8 bytes per operand).  Halved LDS/HBM per tile —
+// the MoE step is expert-weight-stream bound, so fp8 is ~2x weight BW.
+typedef long long i64;
+
+// fp8 rows are 64 B; XOR 16-B slots over 4 positions for the b64 reads
+DEV int swz8(int row, int byte_in_row) {
+  return row * 64 + (byte_in_row ^ ((row & 3) << 4));
+}
+
+__global__ __launch_bounds__(512) void moe_gemm_fp8_kernel(
+    bf16* __restrict__ out, const unsigned char* __restrict__ xq,
+    const float* __restrict__ xs, const unsigned char* __restrict__ wq,
+    const float* __restrict__ ws, const int* __restrict__ sorted_ids,
+    const int* __restrict__ tile_expert, int N, int K, int gather_div) {
+  const int e = tile_expert[blockIdx.x];
+  if (e < 0) return;
+  const int m0 = blockIdx.x * MOE_BM;
+  const int n0 = blockIdx.y * 128;
+
+  __shared__ __attribute__((aligned(16))) unsigned char As[128 * 64];
+  __shared__ __attribute__((aligned(16))) unsigned char Bs[128 * 64];
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  const int lcol = lane & 31;
+  const int lhalf = lane >> 5;
+  const int wave_m = wid & 3;
+  const int wave_n = wid >> 2;
+
+  // loaders: 16 B per thread covers a 128x64B tile in one pass
+  const int ar = tid >> 2;
+  const int ac16 = (tid & 3) * 16;
+  int a_src = -1;
+  if (gather_div > 0) {
+    const int pair = sorted_ids[m0 + ar];
+    if (pair >= 0) a_src = pair / gather_div;
+  } else {
+    a_src = m0 + ar;
+  }
+
+  f32x16 acc[2];
+  #pragma unroll
+  for (int nb = 0; nb < 2; ++nb)
+    #pragma unroll
+    for (int i = 0; i < 16; ++i) acc[nb][i] = 0.f;
+
+  const unsigned char* ap0 = a_src >= 0 ? xq + (long)a_src * K + ac16
+                                        : nullptr;
+  const unsigned char* bp0 = wq + ((long)e * N + n0 + ar) * K + ac16;
+
+  for (int k0 = 0; k0 < K; k0 += 64) {
+    uint4 av = uint4{0, 0, 0, 0};
+    if (ap0) av = *reinterpret_cast<const uint4*>(ap0 + k0);
+    uint4 bv = *reinterpret_cast<const uint4*>(bp0 + k0);
+    *reinterpret_cast<uint4*>(&As[swz8(ar, ac16)]) = av;
+    *reinterpret_cast<uint4*>(&Bs[swz8(ar, ac16)]) = bv;
+    __syncthreads();
+    #pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      i64 af = *reinterpret_cast<const i64*>(
+          &As[swz8(wave_m * 32 + lcol, kk * 16 + lhalf * 8)]);
+      #pragma unroll
+      for (int nb = 0; nb < 2; ++nb) {
+        i64 bf = *reinterpret_cast<const i64*>(
+            &Bs[swz8(wave_n * 64 + nb * 32 + lcol,
+                     kk * 16 + lhalf * 8)]);
+        acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_fp8_fp8(
+            af, bf, acc[nb], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  // epilogue: scale by a_scale[row] * b_scale[e, col], store bf16
+  #pragma unroll
+  for (int nb = 0; nb < 2; ++nb) {
+    const int col = n0 + wave_n * 64 + nb * 32 + lcol;
+    const float bs = ws[(long)e * N + col];
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int m = (r & 3) + 8 * (r >> 2) + 4 * lhalf;
+      const int p = m0 + wave_m * 32 + m;
+      float as = 1.f;
+      if (gather_div > 0) {
+        const int pair = sorted_ids[p];
+        as = pair >= 0 ? xs[pair / gather_div] : 0.f;
+      } else {
+        as = xs[p];
+      }
+      *(unsigned short*)(out + (long)p * N + col) =
+          f2bf(acc[nb][r] * as * bs);
+    }
+  }
+}
+
+void launch_moe_gemm_fp8(bf16* out, const unsigned char* xq,
+                         const float* xs, const unsigned char* wq,
+                         const float* ws, const int* sorted_ids,
+                         const int* tile_expert, int ntiles_max, int N,
+                         int K, int gather_div, hipStream_t stream) {
+  if (N % 128 || K % 64)
+    throw std::runtime_error("moe_gemm_fp8: N%128 or K%64 != 0");
+  dim3 grid((unsigned)ntiles_max, (unsigned)(N / 128));
+  moe_gemm_fp8_kernel<<<grid, 512, 0, stream>>>(
+      out, xq, xs, wq, ws, sorted_ids, tile_expert, N, K, gather_div);
+}

@@ -17,7 +17,8 @@ from torch import nn
 from hyperspot import ops
 from hyperspot.engine.config import ModelSpec
 from hyperspot.models.llama import DecoderLayer, LlamaForCausalLM
-from hyperspot.parallel.layers import _init_weight
+from hyperspot.parallel.layers import (_init_weight, _QUANT_MODE,
+                                       quantize_weight_fp8)
 from hyperspot.parallel.state import get_ep_group, get_ep_rank, get_ep_size
 
 
@@ -43,17 +44,47 @@ class MixtralMoE(nn.Module):
                                   _init_weight(i, h, dtype, tag + 1)], 0))
             w2.append(_init_weight(h, i, dtype, tag + 2))
         # [E_local, 2I, H] and [E_local, H, I]
-        self.w13 = nn.Parameter(torch.stack(w13), requires_grad=False)
-        self.w2 = nn.Parameter(torch.stack(w2), requires_grad=False)
+        from hyperspot.parallel import layers as _L
+        self.quant = _L._QUANT_MODE
+        if self.quant == "fp8":
+            qs13 = [quantize_weight_fp8(w) for w in w13]
+            qs2 = [quantize_weight_fp8(w) for w in w2]
+            self.w13 = nn.Parameter(torch.stack([q for q, _ in qs13]),
+                                    requires_grad=False)
+            self.w13_scale = nn.Parameter(torch.stack([s for _, s in qs13]),
+                                          requires_grad=False)
+            self.w2 = nn.Parameter(torch.stack([q for q, _ in qs2]),
+                                   requires_grad=False)
+            self.w2_scale = nn.Parameter(torch.stack([s for _, s in qs2]),
+                                         requires_grad=False)
+        else:
+            self.w13 = nn.Parameter(torch.stack(w13), requires_grad=False)
+            self.w2 = nn.Parameter(torch.stack(w2), requires_grad=False)
 
     def _expert_ffn(self, x: torch.Tensor, e_local: int) -> torch.Tensor:
+        if self.quant == "fp8":   # CPU/EP fallback: dequantized math
+            w13 = (self.w13[e_local].float()
+                   * self.w13_scale[e_local][:, None]).to(x.dtype)
+            w2 = (self.w2[e_local].float()
+                  * self.w2_scale[e_local][:, None]).to(x.dtype)
+            return ops.silu_mul(x @ w13.t()) @ w2.t()
         gu = x @ self.w13[e_local].t()
         return ops.silu_mul(gu) @ self.w2[e_local].t()
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x) -> torch.Tensor:
+        xq = None
+        if isinstance(x, ops.QTensor):
+            # fp8 path: the fused norm producer already quantized x;
+            # dequantize only the router's [T,E] logits input
+            xq = x
+            x = (xq.data.float() * xq.scale[:, None]).to(torch.bfloat16)
         T, H = x.shape
         weights, ids = ops.torch_ref.moe_route(x, self.router, self.top_k) \
             if not x.is_cuda else self._route_gpu(x)
+        if xq is not None and get_ep_size() == 1 and ops.have_native():
+            return ops.moe_ffn_fp8(xq, self.w13, self.w13_scale, self.w2,
+                                   self.w2_scale, weights.float(),
+                                   ids.to(torch.int32))
         if get_ep_size() == 1:
             return self._local_moe(x, weights.to(x.dtype), ids)
         return self._ep_moe(x, weights.to(x.dtype), ids)

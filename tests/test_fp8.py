@@ -161,3 +161,55 @@ def test_fp8_kv_engine_gpu():
     out = eng.generate([[1, 2, 3, 4, 5, 6, 7, 8]],
                        SamplingParams(temperature=0.0, max_tokens=8))[0]
     assert len(out) == 8
+
+
+def test_fp8_moe_engine_cpu():
+    """Mixtral + fp8 mode on CPU exercises the dequant fallback path."""
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    cfg = EngineConfig(model="tiny-moe", max_num_seqs=2,
+                       max_num_batched_tokens=256, max_model_len=128,
+                       num_gpu_blocks=64, enforce_eager=True, quant="fp8")
+    eng = LLMEngine(cfg, device="cpu")
+    out = eng.generate([[1, 2, 3, 4]],
+                       SamplingParams(temperature=0.0, max_tokens=5))[0]
+    assert len(out) == 5
+
+
+@pytest.mark.gpu
+def test_fp8_moe_grouped_gemm_matches_ref_gpu():
+    """fp8 grouped MFMA GEMM vs the dequantized bf16 reference."""
+    import hyperspot.ops as ops
+    torch.manual_seed(3)
+    dev = "cuda:0"
+    T, H, I, E, K = 64, 256, 256, 4, 2
+    x = (torch.randn(T, H, device=dev) * 0.3).to(torch.bfloat16)
+    w13 = (torch.randn(E, 2 * I, H, device=dev) * 0.05).to(torch.bfloat16)
+    w2 = (torch.randn(E, H, I, device=dev) * 0.05).to(torch.bfloat16)
+    logits = torch.randn(T, E, device=dev)
+    wts, ids = logits.softmax(-1).topk(K, dim=-1)
+    wts = (wts / wts.sum(-1, keepdim=True)).float()
+
+    q13 = [L.quantize_weight_fp8(w13[e]) for e in range(E)]
+    q2 = [L.quantize_weight_fp8(w2[e]) for e in range(E)]
+    w13q = torch.stack([q for q, _ in q13])
+    w13s = torch.stack([s for _, s in q13])
+    w2q = torch.stack([q for q, _ in q2])
+    w2s = torch.stack([s for _, s in q2])
+    xq = ops.quant_fp8(x)
+    out8 = ops.moe_ffn_fp8(xq, w13q, w13s, w2q, w2s, wts,
+                           ids.to(torch.int32))
+    ref = ops.moe_ffn(x, w13, w2, wts, ids.to(torch.int32))
+    rel = (out8.float() - ref.float()).norm() / ref.float().norm()
+    assert rel < 0.08, rel.item()
+
+
+@pytest.mark.gpu
+def test_fp8_moe_engine_gpu():
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    cfg = EngineConfig(model="tiny-moe", max_num_seqs=4,
+                       max_num_batched_tokens=512, max_model_len=256,
+                       num_gpu_blocks=128, enforce_eager=True, quant="fp8")
+    eng = LLMEngine(cfg, device="cuda:0")
+    out = eng.generate([[1, 2, 3, 4, 5]],
+                       SamplingParams(temperature=0.0, max_tokens=6))[0]
+    assert len(out) == 6
