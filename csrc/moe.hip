@@ -226,9 +226,12 @@ void launch_moe_combine(bf16* out, const bf16* y, const float* wts,
 // the MoE step is expert-weight-stream bound, so fp8 is ~2x weight BW.
 typedef long long i64;
 
-// fp8 rows are 64 B; XOR 16-B slots over 4 positions for the b64 reads
+// fp8 tiles stage BK=128 (rows of 128 B): half the barriers of the bf16
+// BK=64 tile at the same LDS footprint — with 64-B rows the kernel was
+// measured BARRIER-bound (86 ms vs 69 ms bf16 Mixtral step).  XOR 16-B
+// slots over 8 positions; b64 fragment reads stay 8-B aligned.
 DEV int swz8(int row, int byte_in_row) {
-  return row * 64 + (byte_in_row ^ ((row & 3) << 4));
+  return row * 128 + (byte_in_row ^ ((row & 7) << 4));
 }
 
 __global__ __launch_bounds__(512) void moe_gemm_fp8_kernel(
@@ -241,8 +244,8 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_kernel(
   const int m0 = blockIdx.x * MOE_BM;
   const int n0 = blockIdx.y * 128;
 
-  __shared__ __attribute__((aligned(16))) unsigned char As[128 * 64];
-  __shared__ __attribute__((aligned(16))) unsigned char Bs[128 * 64];
+  __shared__ __attribute__((aligned(16))) unsigned char As[128 * 128];
+  __shared__ __attribute__((aligned(16))) unsigned char Bs[128 * 128];
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
@@ -252,7 +255,7 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_kernel(
   const int wave_m = wid & 3;
   const int wave_n = wid >> 2;
 
-  // loaders: 16 B per thread covers a 128x64B tile in one pass
+  // loaders: 2x16 B per thread cover a 128x128B tile per K step
   const int ar = tid >> 2;
   const int ac16 = (tid & 3) * 16;
   int a_src = -1;
@@ -273,15 +276,21 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_kernel(
                                         : nullptr;
   const unsigned char* bp0 = wq + ((long)e * N + n0 + ar) * K + ac16;
 
-  for (int k0 = 0; k0 < K; k0 += 64) {
-    uint4 av = uint4{0, 0, 0, 0};
-    if (ap0) av = *reinterpret_cast<const uint4*>(ap0 + k0);
-    uint4 bv = *reinterpret_cast<const uint4*>(bp0 + k0);
-    *reinterpret_cast<uint4*>(&As[swz8(ar, ac16)]) = av;
-    *reinterpret_cast<uint4*>(&Bs[swz8(ar, ac16)]) = bv;
+  for (int k0 = 0; k0 < K; k0 += 128) {
+    uint4 av0 = uint4{0, 0, 0, 0}, av1 = uint4{0, 0, 0, 0};
+    if (ap0) {
+      av0 = *reinterpret_cast<const uint4*>(ap0 + k0);
+      av1 = *reinterpret_cast<const uint4*>(ap0 + k0 + 64);
+    }
+    uint4 bv0 = *reinterpret_cast<const uint4*>(bp0 + k0);
+    uint4 bv1 = *reinterpret_cast<const uint4*>(bp0 + k0 + 64);
+    *reinterpret_cast<uint4*>(&As[swz8(ar, ac16)]) = av0;
+    *reinterpret_cast<uint4*>(&As[swz8(ar, ac16 + 64)]) = av1;
+    *reinterpret_cast<uint4*>(&Bs[swz8(ar, ac16)]) = bv0;
+    *reinterpret_cast<uint4*>(&Bs[swz8(ar, ac16 + 64)]) = bv1;
     __syncthreads();
     #pragma unroll
-    for (int kk = 0; kk < 4; ++kk) {
+    for (int kk = 0; kk < 8; ++kk) {
       i64 af = *reinterpret_cast<const i64*>(
           &As[swz8(wave_m * 32 + lcol, kk * 16 + lhalf * 8)]);
       #pragma unroll
@@ -322,8 +331,8 @@ void launch_moe_gemm_fp8(bf16* out, const unsigned char* xq,
                          const float* ws, const int* sorted_ids,
                          const int* tile_expert, int ntiles_max, int N,
                          int K, int gather_div, hipStream_t stream) {
-  if (N % 128 || K % 64)
-    throw std::runtime_error("moe_gemm_fp8: N%128 or K%64 != 0");
+  if (N % 128 || K % 128)
+    throw std::runtime_error("moe_gemm_fp8: N%128 or K%128 != 0");
   dim3 grid((unsigned)ntiles_max, (unsigned)(N / 128));
   moe_gemm_fp8_kernel<<<grid, 512, 0, stream>>>(
       out, xq, xs, wq, ws, sorted_ids, tile_expert, N, K, gather_div);
