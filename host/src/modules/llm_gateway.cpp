@@ -155,6 +155,8 @@ void LlmGatewayModule::init(ModuleCtx& ctx) {
   budget_tokens_ =
       (uint64_t)ctx.config.path("usage.budget_tokens_per_tenant").as_int(0);
   license_feature_ = ctx.config.at("require_license_feature").as_string("");
+  ttft_timeout_ms_ = ctx.config.path("timeouts.ttft_ms").as_int(0);
+  total_timeout_ms_ = ctx.config.path("timeouts.total_ms").as_int(0);
   // worker fleet: explicit device list, or count (devices 0..count-1)
   std::vector<int> devices;
   if (worker_cfg_.at("devices").is_array())
@@ -375,6 +377,13 @@ void LlmGatewayModule::record_usage(const std::string& tenant,
 Json LlmGatewayModule::run_chat_blocking(const Json& body,
                                          const Json& resolved,
                                          const std::string& rid) {
+  const auto t0 = std::chrono::steady_clock::now();
+  auto left = [&](long limit_ms) -> int {
+    if (!limit_ms) return 120000;
+    auto used = std::chrono::duration_cast<std::chrono::milliseconds>(
+        std::chrono::steady_clock::now() - t0).count();
+    return (int)std::max<long>(1, limit_ms - used);
+  };
   Lease lease{pick_worker()};
   if (!lease.w)
     throw Problem{503, "Service Unavailable", "about:blank",
@@ -400,13 +409,25 @@ Json LlmGatewayModule::run_chat_blocking(const Json& body,
   std::string text;
   Json usage;
   std::string finish = "stop";
+  bool first = true;
   while (true) {
-    auto msg = conn.read_json();
-    if (!msg)
+    // TTFT timer until the first delta, then the total timer
+    // (DESIGN.md:706-741)
+    auto msg = conn.read_json(
+        left(first && ttft_timeout_ms_ ? ttft_timeout_ms_
+                                       : total_timeout_ms_));
+    if (!msg) {
+      Json ab = Json::object();
+      ab["type"] = "abort";
+      ab["id"] = rid;
+      conn.send_json(ab);
       throw Problem{504, "Gateway Timeout", "about:blank",
-                    "engine timed out", "provider_timeout"};
+                    first ? "no first token within the TTFT budget"
+                          : "generation exceeded the total budget",
+                    "provider_timeout"};
+    }
     const std::string ev = msg->at("event").as_string();
-    if (ev == "delta") text += msg->at("text").as_string();
+    if (ev == "delta") { first = false; text += msg->at("text").as_string(); }
     else if (ev == "done") {
       usage = msg->at("usage");
       finish = msg->at("finish_reason").as_string("stop");
@@ -445,6 +466,38 @@ static std::string sse_chunk(const std::string& id, const std::string& model,
   return "data: " + c.dump() + "\n\n";
 }
 
+// fallback chain (DESIGN.md:680-704): on provider_error/timeout try the
+// next model in request.fallback.models; mark fallback_used + model_used.
+Json LlmGatewayModule::run_chat_with_fallback(const SecurityContext& sec,
+                                              const Json& body,
+                                              const std::string& rid) {
+  std::vector<std::string> chain = {body.at("model").as_string()};
+  const Json& fb = body.path("fallback.models");
+  if (fb.is_array())
+    for (auto& m : fb.arr()) chain.push_back(m.as_string());
+  Problem last{502, "Bad Gateway", "about:blank", "no model attempted",
+               "provider_error"};
+  for (size_t i = 0; i < chain.size(); ++i) {
+    Json resolved;
+    try {
+      resolved = resolve_model(sec, chain[i]);
+    } catch (const Problem& p) {
+      last = p;
+      continue;            // unknown model in the chain: try the next
+    }
+    try {
+      Json resp = run_chat_blocking(body, resolved, rid);
+      resp["fallback_used"] = i > 0;
+      return resp;
+    } catch (const Problem& p) {
+      if (p.code != "provider_error" && p.code != "provider_timeout")
+        throw;             // validation/budget errors do not fall back
+      last = p;
+    }
+  }
+  throw last;
+}
+
 void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   m_requests_++;
   Json body;
@@ -474,8 +527,11 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
     return;
   }
 
-  // model resolution via model-registry (DESIGN.md:317-346)
-  Json resolved = resolve_model(sec, model);
+  // model resolution via model-registry (DESIGN.md:317-346); with a
+  // fallback chain, resolution failures are handled per-chain-entry
+  const bool has_fb = body.path("fallback.models").is_array();
+  Json resolved;
+  if (!has_fb) resolved = resolve_model(sec, model);
   check_budget(sec.tenant_id);
 
   // per-tenant admission (serverless-runtime quota machinery)
@@ -497,7 +553,8 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   const std::string rid = "chat-" + std::to_string(req_ctr_.fetch_add(1));
 
   if (!stream) {
-    Json resp = run_chat_blocking(body, resolved, rid);
+    Json resp = has_fb ? run_chat_with_fallback(sec, body, rid)
+                       : run_chat_blocking(body, resolved, rid);
     record_usage(sec.tenant_id, resp.at("usage"));
     resp.erase("finish_reason");
     w.respond(200, "application/json", resp.dump(),
@@ -505,6 +562,24 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
     return;
   }
 
+  if (has_fb) {
+    // streaming + fallback: pick the first RESOLVABLE model up front
+    // (mid-stream failover is not possible once bytes are sent)
+    std::vector<std::string> chain = {model};
+    for (auto& m : body.path("fallback.models").arr())
+      chain.push_back(m.as_string());
+    bool ok = false;
+    for (size_t i = 0; i < chain.size() && !ok; ++i) {
+      try {
+        resolved = resolve_model(sec, chain[i]);
+        ok = true;
+      } catch (const Problem&) {}
+    }
+    if (!ok)
+      throw Problem{404, "Not Found", "about:blank",
+                    "no model in the fallback chain is available",
+                    "model_not_found"};
+  }
   m_streams_++;
   Lease lease{pick_worker()};
   if (!lease.w)

@@ -100,6 +100,8 @@ class LlmGatewayModule : public Module {
   // blocking non-stream chat against the engine; throws Problem
   Json run_chat_blocking(const Json& body, const Json& resolved,
                          const std::string& rid);
+  Json run_chat_with_fallback(const SecurityContext& sec, const Json& body,
+                              const std::string& rid);
 
   // jobs
   std::shared_ptr<Job> submit_job(const SecurityContext& sec, Json body,
@@ -145,6 +147,10 @@ class LlmGatewayModule : public Module {
   std::thread watchdog_;
   std::atomic<bool> stopping_{false};
   std::atomic<uint64_t> m_worker_restarts_{0};
+
+  // timeouts (DESIGN.md:706-741 TTFT + total state machine; 0 = off)
+  long ttft_timeout_ms_ = 0;
+  long total_timeout_ms_ = 0;
 
   // usage tracker + budget (tokens per tenant; 0 = unlimited)
   std::mutex usage_mu_;

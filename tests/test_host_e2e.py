@@ -1121,3 +1121,26 @@ def test_openapi_document_contract(server):
     assert chat["security"], "chat must be authenticated"
     schema = chat["requestBody"]["content"]["application/json"]["schema"]
     assert set(schema["required"]) == {"model", "messages"}
+
+
+def test_fallback_chain(server):
+    """fallback.models: unknown first model falls through to a live one;
+    response marks fallback_used + the model actually used."""
+    url = BASE.format(server.port)
+    st, body = _http("POST", url + "/v1/chat/completions",
+                     {"model": "nonexistent-model",
+                      "fallback": {"models": ["tiny-llama"]},
+                      "messages": [{"role": "user", "content":
+                                    [{"type": "text", "text": "fb"}]}],
+                      "max_tokens": 3})
+    assert st == 200, body
+    d = json.loads(body)
+    assert d["fallback_used"] is True
+    assert d["model_used"].endswith("tiny-llama")
+    # chain with no viable model: the problem surfaces
+    st, body = _http("POST", url + "/v1/chat/completions",
+                     {"model": "nope1",
+                      "fallback": {"models": ["nope2"]},
+                      "messages": [{"role": "user", "content":
+                                    [{"type": "text", "text": "x"}]}]})
+    assert st == 404 and json.loads(body)["code"] == "model_not_found"
