@@ -308,7 +308,10 @@ void LlmGatewayModule::start(ModuleCtx& ctx) {
       }
     });
   }
-  job_thread_ = std::thread([this] { job_loop(); });
+  // small executor pool so batch jobs overlap in the engine's
+  // continuous batch instead of serializing
+  for (int i = 0; i < 4; ++i)
+    job_threads_.emplace_back([this] { job_loop(); });
 }
 
 void LlmGatewayModule::stop(ModuleCtx& ctx) {
@@ -317,7 +320,9 @@ void LlmGatewayModule::stop(ModuleCtx& ctx) {
     stopping_ = true;
   }
   jobs_cv_.notify_all();
-  if (job_thread_.joinable()) job_thread_.join();
+  for (auto& t : job_threads_)
+    if (t.joinable()) t.join();
+  job_threads_.clear();
   if (watchdog_.joinable()) watchdog_.join();
   for (auto& w : workers_) {
     if (w->pid <= 0) continue;

@@ -143,7 +143,7 @@ class LlmGatewayModule : public Module {
   std::deque<std::string> job_queue_;
   std::map<std::string, std::shared_ptr<Job>> jobs_;
   std::map<std::string, Batch> batches_;
-  std::thread job_thread_;
+  std::vector<std::thread> job_threads_;
   std::thread watchdog_;
   std::atomic<bool> stopping_{false};
   std::atomic<uint64_t> m_worker_restarts_{0};

@@ -72,8 +72,8 @@ class _LinearCompute(nn.Module):
             self.weight = nn.Parameter(w, requires_grad=False)
 
     def _mm(self, x) -> torch.Tensor:
-        from hyperspot.ops import QTensor
-        if isinstance(x, QTensor):
+        from hyperspot import ops as _O   # local: avoids import cycle at init
+        if isinstance(x, _O.QTensor):
             # pre-quantized by a fused producer kernel (csrc/quant.hip)
             return torch._scaled_mm(
                 x.data, self.weight.t(), scale_a=x.scale.unsqueeze(1),
