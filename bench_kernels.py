@@ -38,7 +38,9 @@ def bench_attn(iters):
     for (bs, ctx, kvh, group, D) in [(256, 512, 8, 4, 128),
                                      (256, 2048, 8, 4, 128),
                                      (64, 512, 8, 4, 128),
-                                     (256, 512, 1, 8, 128)]:
+                                     (256, 512, 1, 8, 128),
+                                     (64, 8192, 8, 4, 128),
+                                     (16, 32768, 8, 4, 128)]:
         BS = 16
         nblk = (ctx + BS - 1) // BS
         tot_blocks = bs * nblk + 7
