@@ -156,6 +156,9 @@ void LlmGatewayModule::init(ModuleCtx& ctx) {
       (uint64_t)ctx.config.path("usage.budget_tokens_per_tenant").as_int(0);
   license_feature_ = ctx.config.at("require_license_feature").as_string("");
   ttft_timeout_ms_ = ctx.config.path("timeouts.ttft_ms").as_int(0);
+  if (ctx.config.path("hooks.blocklist").is_array())
+    for (auto& wd : ctx.config.path("hooks.blocklist").arr())
+      hook_blocklist_.push_back(wd.as_string());
   total_timeout_ms_ = ctx.config.path("timeouts.total_ms").as_int(0);
   // worker fleet: explicit device list, or count (devices 0..count-1)
   std::vector<int> devices;
@@ -471,6 +474,28 @@ static std::string sse_chunk(const std::string& id, const std::string& model,
   return "data: " + c.dump() + "\n\n";
 }
 
+bool LlmGatewayModule::hook_blocks(const std::string& text) const {
+  for (auto& w : hook_blocklist_)
+    if (!w.empty() && text.find(w) != std::string::npos) return true;
+  return false;
+}
+
+// pre_call hook: scan inbound message text (DESIGN.md:743-766)
+void LlmGatewayModule::hook_pre_call(const Json& body) {
+  if (hook_blocklist_.empty()) return;
+  const Json& msgs = body.at("messages");
+  if (!msgs.is_array()) return;
+  for (auto& m : msgs.arr()) {
+    const Json& content = m.at("content");
+    if (!content.is_array()) continue;
+    for (auto& part : content.arr())
+      if (hook_blocks(part.at("text").as_string()))
+        throw Problem{403, "Forbidden", "about:blank",
+                      "request blocked by content policy",
+                      "request_blocked"};
+  }
+}
+
 // fallback chain (DESIGN.md:680-704): on provider_error/timeout try the
 // next model in request.fallback.models; mark fallback_used + model_used.
 Json LlmGatewayModule::run_chat_with_fallback(const SecurityContext& sec,
@@ -534,6 +559,7 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
 
   // model resolution via model-registry (DESIGN.md:317-346); with a
   // fallback chain, resolution failures are handled per-chain-entry
+  hook_pre_call(body);   // may throw request_blocked (DESIGN.md:743-766)
   const bool has_fb = body.path("fallback.models").is_array();
   Json resolved;
   if (!has_fb) resolved = resolve_model(sec, model);
@@ -560,6 +586,12 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   if (!stream) {
     Json resp = has_fb ? run_chat_with_fallback(sec, body, rid)
                        : run_chat_blocking(body, resolved, rid);
+    // post_response hook: generated content can be blocked too
+    if (!hook_blocklist_.empty() &&
+        hook_blocks(resp.at("content").at(0).at("text").as_string()))
+      throw Problem{403, "Forbidden", "about:blank",
+                    "response blocked by content policy",
+                    "response_blocked"};
     record_usage(sec.tenant_id, resp.at("usage"));
     resp.erase("finish_reason");
     w.respond(200, "application/json", resp.dump(),

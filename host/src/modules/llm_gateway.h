@@ -148,6 +148,12 @@ class LlmGatewayModule : public Module {
   std::atomic<bool> stopping_{false};
   std::atomic<uint64_t> m_worker_restarts_{0};
 
+  // hook plugin (DESIGN.md:743-766): pre_call may block a request,
+  // post_response may block/redact the response (config.hooks.blocklist)
+  std::vector<std::string> hook_blocklist_;
+  void hook_pre_call(const Json& body);          // throws request_blocked
+  bool hook_blocks(const std::string& text) const;
+
   // timeouts (DESIGN.md:706-741 TTFT + total state machine; 0 = off)
   long ttft_timeout_ms_ = 0;
   long total_timeout_ms_ = 0;

@@ -1144,3 +1144,62 @@ def test_fallback_chain(server):
                       "messages": [{"role": "user", "content":
                                     [{"type": "text", "text": "x"}]}]})
     assert st == 404 and json.loads(body)["code"] == "model_not_found"
+
+
+@pytest.fixture(scope="module")
+def hooked_server():
+    import tempfile
+    port = _free_port()
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-hook-")
+    cfg = f"""
+server:
+  home_dir: "/tmp/hs-e2e-hook"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+  llm-gateway:
+    config:
+      model: "tiny-llama"
+      worker_socket: "{sock}"
+      auto_start_worker: true
+      hooks:
+        blocklist: ["forbiddenword"]
+      worker:
+        device: "cpu"
+        eager: true
+        max_num_seqs: 4
+        num_gpu_blocks: 128
+"""
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(cfg)
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        srv.wait_worker()
+        yield srv
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
+
+
+def test_hook_request_blocked(hooked_server):
+    url = BASE.format(hooked_server.port)
+    st, body = _http("POST", url + "/v1/chat/completions",
+                     {"model": "tiny-llama",
+                      "messages": [{"role": "user", "content":
+                                    [{"type": "text",
+                                      "text": "say forbiddenword"}]}],
+                      "max_tokens": 2})
+    assert st == 403 and json.loads(body)["code"] == "request_blocked"
+    # clean request passes
+    st, _ = _http("POST", url + "/v1/chat/completions",
+                  {"model": "tiny-llama",
+                   "messages": [{"role": "user", "content":
+                                 [{"type": "text", "text": "fine"}]}],
+                   "max_tokens": 2})
+    assert st == 200
