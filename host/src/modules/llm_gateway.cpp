@@ -156,6 +156,7 @@ void LlmGatewayModule::init(ModuleCtx& ctx) {
       (uint64_t)ctx.config.path("usage.budget_tokens_per_tenant").as_int(0);
   license_feature_ = ctx.config.at("require_license_feature").as_string("");
   ttft_timeout_ms_ = ctx.config.path("timeouts.ttft_ms").as_int(0);
+  job_ttl_s_ = ctx.config.path("jobs.ttl_s").as_int(3600);
   if (ctx.config.path("hooks.blocklist").is_array())
     for (auto& wd : ctx.config.path("hooks.blocklist").arr())
       hook_blocklist_.push_back(wd.as_string());
@@ -354,8 +355,11 @@ Json LlmGatewayModule::resolve_model(const SecurityContext& sec,
                       : std::nullopt;
   if (!resolved)
     throw Problem{404, "Not Found", "about:blank",
-                  "model '" + model + "' not found or not approved",
-                  "model_not_found"};
+                  "model '" + model + "' not found", "model_not_found"};
+  if (resolved->at("approval").as_string("approved") != "approved")
+    throw Problem{403, "Forbidden", "about:blank",
+                  "model '" + model + "' is not approved for this tenant",
+                  "model_not_approved"};
   return *resolved;
 }
 
@@ -560,6 +564,23 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   // model resolution via model-registry (DESIGN.md:317-346); with a
   // fallback chain, resolution failures are handled per-chain-entry
   hook_pre_call(body);   // may throw request_blocked (DESIGN.md:743-766)
+  // capability gate: the engine is text-only today — non-text content
+  // parts or tool use are capability_not_supported (DESIGN error list)
+  if (body.at("tools").is_array() && body.at("tools").size() > 0)
+    throw Problem{400, "Bad Request", "about:blank",
+                  "tool use is not supported by this engine",
+                  "capability_not_supported"};
+  for (auto& m : body.at("messages").arr()) {
+    const Json& content = m.at("content");
+    if (!content.is_array()) continue;
+    for (auto& part : content.arr()) {
+      const std::string pt = part.at("type").as_string("text");
+      if (pt != "text")
+        throw Problem{400, "Bad Request", "about:blank",
+                      "content part type '" + pt + "' is not supported",
+                      "capability_not_supported"};
+    }
+  }
   const bool has_fb = body.path("fallback.models").is_array();
   Json resolved;
   if (!has_fb) resolved = resolve_model(sec, model);
@@ -983,6 +1004,10 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
       if (it == jobs_.end() || it->second->tenant != sec.tenant_id)
         throw Problem{404, "Not Found", "about:blank", "no such job",
                       "job_not_found"};
+      if (it->second->finished_at > 0 && job_ttl_s_ > 0 &&
+          now_s() - it->second->finished_at > job_ttl_s_)
+        throw Problem{410, "Gone", "about:blank",
+                      "job result expired", "job_expired"};
       w.respond(200, "application/json", job_json(*it->second).dump());
     });
   }

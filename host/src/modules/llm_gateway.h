@@ -156,6 +156,7 @@ class LlmGatewayModule : public Module {
 
   // timeouts (DESIGN.md:706-741 TTFT + total state machine; 0 = off)
   long ttft_timeout_ms_ = 0;
+  long job_ttl_s_ = 3600;
   long total_timeout_ms_ = 0;
 
   // usage tracker + budget (tokens per tenant; 0 = unlimited)

@@ -396,7 +396,9 @@ class LocalModelRegistry : public ModelRegistryClient {
         ? "local::" + canonical : canonical;
     auto it = models_.find(cid);
     if (it == models_.end()) return std::nullopt;
-    return it->second;
+    Json m = it->second;
+    m["approval"] = approval(tenant, cid);
+    return m;
   }
   std::vector<Json> list_tenant_models(const std::string&) override {
     std::vector<Json> out;
@@ -404,7 +406,23 @@ class LocalModelRegistry : public ModelRegistryClient {
     return out;
   }
 
+  // ModelApproval states pending->approved|rejected->revoked
+  // (model-registry PRD.md:225-253); default approved so single-tenant
+  // deployments need no workflow
+  std::string approval(const std::string& tenant, const std::string& cid) {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto it = approvals_.find(tenant + "|" + cid);
+    return it == approvals_.end() ? "approved" : it->second;
+  }
+  void set_approval(const std::string& tenant, const std::string& cid,
+                    const std::string& st) {
+    std::lock_guard<std::mutex> lk(mu_);
+    approvals_[tenant + "|" + cid] = st;
+  }
+
  private:
+  std::mutex mu_;
+  std::map<std::string, std::string> approvals_;
   std::map<std::string, Json> models_;
 };
 
@@ -449,6 +467,36 @@ void ModelRegistryModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
                                           rq.path_params["canonical_id"]);
     if (!m) throw Problem::not_found("model_not_found");
     w.respond(200, "application/json", m->dump());
+  });
+  // approval workflow (PRD.md:225-253): approve | reject | revoke
+  OperationSpec ap;
+  ap.method = "POST";
+  ap.path = "/model-registry/v1/models/{canonical_id}/approval";
+  ap.operation_id = "models_set_approval";
+  ap.summary = "Set this tenant's approval state for a model";
+  ap.authenticated = true;
+  ap.allowed_content_types = {"application/json"};
+  ap.tags = {"model-registry"};
+  rest.register_op(ap, [](HttpRequest& rq, ResponseWriter& w) {
+    Json body;
+    try { body = Json::parse(rq.body); }
+    catch (...) { throw Problem::bad_request("invalid JSON body"); }
+    const std::string st = body.at("status").as_string();
+    if (st != "approved" && st != "pending" && st != "rejected" &&
+        st != "revoked")
+      throw Problem::bad_request(
+          "status must be approved|pending|rejected|revoked");
+    const std::string cid = rq.path_params["canonical_id"];
+    auto sec = sec_of(rq);
+    if (!g_registry->get_tenant_model(sec.tenant_id, cid))
+      throw Problem::not_found("model_not_found");
+    g_registry->set_approval(sec.tenant_id,
+                             cid.find("::") == std::string::npos
+                                 ? "local::" + cid : cid, st);
+    Json out = Json::object();
+    out["canonical_id"] = cid;
+    out["status"] = st;
+    w.respond(200, "application/json", out.dump());
   });
 }
 

@@ -1203,3 +1203,49 @@ def test_hook_request_blocked(hooked_server):
                                  [{"type": "text", "text": "fine"}]}],
                    "max_tokens": 2})
     assert st == 200
+
+
+def test_model_approval_workflow(mt_server):
+    """model_not_approved: revoking a model for ONE tenant blocks that
+    tenant only (PRD approval states, tenant-scoped)."""
+    url = BASE.format(mt_server.port)
+    body = {"model": "tiny-llama",
+            "messages": [{"role": "user", "content":
+                          [{"type": "text", "text": "x"}]}],
+            "max_tokens": 2}
+    st, resp = _http("POST",
+                     url + "/model-registry/v1/models/tiny-llama/approval",
+                     {"status": "revoked"}, token="acme-token")
+    assert st == 200, resp
+    st, resp = _http("POST", url + "/v1/chat/completions", body,
+                     token="acme-token")
+    assert st == 403 and json.loads(resp)["code"] == "model_not_approved"
+    # other tenant unaffected (worker off in mt_server -> 503, not 403)
+    st, resp = _http("POST", url + "/v1/chat/completions", body,
+                     token="root-token")
+    assert st == 503, resp
+    # re-approve
+    st, _ = _http("POST",
+                  url + "/model-registry/v1/models/tiny-llama/approval",
+                  {"status": "approved"}, token="acme-token")
+    st, resp = _http("POST", url + "/v1/chat/completions", body,
+                     token="acme-token")
+    assert st == 503, resp   # back to worker-not-running, not approval
+
+
+def test_capability_not_supported(server):
+    url = BASE.format(server.port)
+    st, body = _http("POST", url + "/v1/chat/completions",
+                     {"model": "tiny-llama",
+                      "tools": [{"name": "t"}],
+                      "messages": [{"role": "user", "content":
+                                    [{"type": "text", "text": "x"}]}]})
+    assert st == 400
+    assert json.loads(body)["code"] == "capability_not_supported"
+    st, body = _http("POST", url + "/v1/chat/completions",
+                     {"model": "tiny-llama",
+                      "messages": [{"role": "user", "content":
+                                    [{"type": "image",
+                                      "url": "x://y"}]}]})
+    assert st == 400
+    assert json.loads(body)["code"] == "capability_not_supported"
