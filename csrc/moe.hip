@@ -85,10 +85,14 @@ __global__ __launch_bounds__(512) void moe_gemm_kernel(
     bf16* __restrict__ out, const bf16* __restrict__ x,
     const bf16* __restrict__ w, const int* __restrict__ sorted_ids,
     const int* __restrict__ tile_expert, int N, int K, int gather_div) {
-  const int e = tile_expert[blockIdx.x];
+  // grid is (n_tiles, m_tiles): with n_tiles % 8 == 0 every m-tile of one
+  // (expert, n0) pair lands on the SAME XCD (dispatch round-robins
+  // blockIdx linearly over the 8 XCDs), so the expert's B-slice is read
+  // from that XCD's L2 instead of HBM for the 2nd..kth m-tile
+  const int e = tile_expert[blockIdx.y];
   if (e < 0) return;
-  const int m0 = blockIdx.x * MOE_BM;
-  const int n0 = blockIdx.y * 128;
+  const int m0 = blockIdx.y * MOE_BM;
+  const int n0 = blockIdx.x * 128;
 
   // single-buffered tiles: a 2-deep LDS ring (64 KiB) was measured 10%
   // SLOWER here — it halves WG residency (guide: explicit dbuf at HIP
@@ -204,7 +208,7 @@ void launch_moe_gemm(bf16* out, const bf16* x, const bf16* w,
                      hipStream_t stream) {
   if (N % 128 || K % 64)
     throw std::runtime_error("moe_gemm: N%128 or K%64 != 0");
-  dim3 grid((unsigned)ntiles_max, (unsigned)(N / 128));
+  dim3 grid((unsigned)(N / 128), (unsigned)ntiles_max);
   moe_gemm_kernel<<<grid, 512, 0, stream>>>(out, x, w, sorted_ids,
                                             tile_expert, N, K, gather_div);
 }
@@ -239,10 +243,10 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_kernel(
     const float* __restrict__ xs, const unsigned char* __restrict__ wq,
     const float* __restrict__ ws, const int* __restrict__ sorted_ids,
     const int* __restrict__ tile_expert, int N, int K, int gather_div) {
-  const int e = tile_expert[blockIdx.x];
+  const int e = tile_expert[blockIdx.y];      // (n, m) grid — see bf16 note
   if (e < 0) return;
-  const int m0 = blockIdx.x * MOE_BM;
-  const int n0 = blockIdx.y * 128;
+  const int m0 = blockIdx.y * MOE_BM;
+  const int n0 = blockIdx.x * 128;
 
   __shared__ __attribute__((aligned(16))) unsigned char As[128 * 128];
   __shared__ __attribute__((aligned(16))) unsigned char Bs[128 * 128];
@@ -333,7 +337,7 @@ void launch_moe_gemm_fp8(bf16* out, const unsigned char* xq,
                          int K, int gather_div, hipStream_t stream) {
   if (N % 128 || K % 128)
     throw std::runtime_error("moe_gemm_fp8: N%128 or K%128 != 0");
-  dim3 grid((unsigned)ntiles_max, (unsigned)(N / 128));
+  dim3 grid((unsigned)(N / 128), (unsigned)ntiles_max);
   moe_gemm_fp8_kernel<<<grid, 512, 0, stream>>>(
       out, xq, xs, wq, ws, sorted_ids, tile_expert, N, K, gather_div);
 }
