@@ -1,6 +1,8 @@
 #include "oagw.h"
 
 #include "../http/client.h"
+
+#include <algorithm>
 #include "../util/log.h"
 #include "system_modules.h"
 
@@ -85,6 +87,57 @@ void OagwModule::proxy(HttpRequest& req, ResponseWriter& w) {
                       "upstream rate limit", "rate_limited"};
     }
   }
+  // route matching (oagw-sdk Route: match_rules{methods, path prefix},
+  // priority desc, enabled — models.rs:258-268): when routes exist for
+  // this upstream, the request must match one; a matched route's
+  // rate_limit applies on top of the upstream's
+  {
+    std::lock_guard<std::mutex> lk(mu_);
+    std::vector<const Json*> rts;
+    for (auto& [rid, r] : routes_) {
+      if (r.at("upstream_id").as_string() != up.at("id").as_string())
+        continue;
+      if (!r.at("enabled").as_bool(true)) continue;
+      rts.push_back(&r);
+    }
+    if (!rts.empty()) {
+      std::sort(rts.begin(), rts.end(), [](const Json* a, const Json* b) {
+        return a->at("priority").as_int(0) > b->at("priority").as_int(0);
+      });
+      const Json* hit = nullptr;
+      for (const Json* r : rts) {
+        const Json& mrs = r->at("match_rules");
+        const Json& methods = mrs.at("methods");
+        bool mok = !methods.is_array() || methods.size() == 0;
+        if (!mok)
+          for (auto& mm : methods.arr())
+            if (mm.as_string() == req.method) mok = true;
+        if (!mok) continue;
+        const std::string pfx = mrs.at("path").as_string("");
+        if (!pfx.empty() &&
+            ("/" + suffix).rfind(pfx, 0) != 0)
+          continue;
+        hit = r;
+        break;
+      }
+      if (!hit)
+        throw Problem{404, "Not Found", "about:blank",
+                      "no route matches " + req.method + " /" + suffix,
+                      "route_not_found"};
+      const Json& rl = hit->at("rate_limit");
+      if (rl.is_object()) {
+        auto& b = limiters_["route:" + hit->at("id").as_string()];
+        if (!b)
+          b = std::make_unique<TokenBucket>(
+              rl.at("sustained").as_number(50),
+              rl.at("burst").as_number(100));
+        if (!b->try_acquire())
+          throw Problem{429, "Too Many Requests", "about:blank",
+                        "route rate limit", "rate_limited"};
+      }
+    }
+  }
+
   const Json& ep = up.path("server.endpoints").arr()[0];
   const std::string host = ep.at("host").as_string();
   const int port = (int)ep.at("port").as_int(80);
