@@ -837,6 +837,17 @@ void LlmGatewayModule::job_loop() {
       std::unique_lock<std::mutex> lk(jobs_mu_);
       jobs_cv_.wait(lk, [&] { return stopping_ || !job_queue_.empty(); });
       if (stopping_) return;
+      // prune expired results so the store stays bounded (job_expired TTL)
+      if (job_ttl_s_ > 0) {
+        const double now = now_s();
+        for (auto it = jobs_.begin(); it != jobs_.end();) {
+          if (it->second->finished_at > 0 &&
+              now - it->second->finished_at > 2 * (double)job_ttl_s_)
+            it = jobs_.erase(it);
+          else
+            ++it;
+        }
+      }
       const std::string id = job_queue_.front();
       job_queue_.pop_front();
       auto it = jobs_.find(id);
