@@ -165,7 +165,13 @@ Json RestRegistry::build_openapi(const std::string& title,
       }
     }
     op["responses"] = responses;
-    if (!s.request_schema.is_null()) {
+    if (!s.odata_filter_fields.empty()) {
+      Json ext = Json::object();
+      Json fields = Json::array();
+      for (auto& f : s.odata_filter_fields) fields.push_back(f);
+      ext["allowedFields"] = fields;
+      op["x-odata-filter"] = ext;
+    }
       Json media = Json::object();
       media["schema"] = s.request_schema;
       Json content = Json::object();

@@ -71,6 +71,7 @@ void UserSettingsModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
     op.path = "/simple-user-settings/v1/settings";
     op.operation_id = "list_settings";
     op.summary = "List settings ($filter, $top, cursor)";
+    op.odata_filter_fields = filterable;
     op.authenticated = true;
     op.tags = {"simple-user-settings"};
     rest.register_op(op, [this, filterable](HttpRequest& rq,

@@ -165,6 +165,7 @@ void UsersInfoModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
     op.path = "/users-info/v1/users";
     op.operation_id = "list_users";
     op.summary = "List users ($filter/$top/cursor, Page envelope)";
+    op.odata_filter_fields = filterable;
     op.authenticated = true;
     op.tags = {"users-info"};
     rest.register_op(op, [this, filterable](HttpRequest& rq,

@@ -1121,6 +1121,8 @@ def test_openapi_document_contract(server):
     assert chat["security"], "chat must be authenticated"
     schema = chat["requestBody"]["content"]["application/json"]["schema"]
     assert set(schema["required"]) == {"model", "messages"}
+    lst = paths["/simple-user-settings/v1/settings"]["get"]
+    assert "key" in lst["x-odata-filter"]["allowedFields"]
 
 
 def test_fallback_chain(server):
