@@ -99,6 +99,9 @@ struct OperationSpec {
   std::map<int, std::string> responses;             // status -> description
   Json response_schema;       // schema of the 200 response
   std::vector<std::string> license_features;
+  // OData vendor extension: filterable fields advertised in the OpenAPI
+  // doc as x-odata-filter.allowedFields (operation_builder.rs:227-239)
+  std::vector<std::string> odata_filter_fields;
   bool sse = false;
 };
 
