@@ -172,6 +172,7 @@ Json RestRegistry::build_openapi(const std::string& title,
       ext["allowedFields"] = fields;
       op["x-odata-filter"] = ext;
     }
+    if (!s.request_schema.is_null()) {
       Json media = Json::object();
       media["schema"] = s.request_schema;
       Json content = Json::object();
