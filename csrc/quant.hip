@@ -126,7 +126,10 @@ __global__ __launch_bounds__(256) void fused_add_rmsnorm_fp8_kernel(
   }
 }
 
-// out <- fp8(silu(x[:, :inter]) * x[:, inter:]) with per-row scale
+// out <- fp8(silu(x[:, :inter]) * x[:, inter:]) with per-row scale.
+// CH=16 covers inter up to 32768 (Llama-70B MLP: 28672) at the price of
+// a bigger register file for the row.
+template <int CH>
 __global__ __launch_bounds__(256) void silu_mul_fp8_kernel(
     unsigned char* __restrict__ out, float* __restrict__ scales,
     const bf16* __restrict__ x, int inter) {
@@ -134,10 +137,10 @@ __global__ __launch_bounds__(256) void silu_mul_fp8_kernel(
   const long row = blockIdx.x;
   const bf16* xr = x + row * 2ll * inter;
 
-  float v[QCHUNKS][8];
+  float v[CH][8];
   float amax = 0.f;
   #pragma unroll
-  for (int c = 0; c < QCHUNKS; ++c) {
+  for (int c = 0; c < CH; ++c) {
     const int base = (c * 256 + threadIdx.x) * 8;
     if (base < inter) {
       bf16x8 g = load_bf16x8(xr + base);
@@ -156,7 +159,7 @@ __global__ __launch_bounds__(256) void silu_mul_fp8_kernel(
   const float inv = 1.f / s;
   if (threadIdx.x == 0) scales[row] = s;
   #pragma unroll
-  for (int c = 0; c < QCHUNKS; ++c) {
+  for (int c = 0; c < CH; ++c) {
     const int base = (c * 256 + threadIdx.x) * 8;
     if (base < inter) {
       float t[8];
@@ -189,8 +192,12 @@ void launch_fused_add_rmsnorm_fp8(unsigned char* out, float* scales,
 
 void launch_silu_mul_fp8(unsigned char* out, float* scales, const bf16* x,
                          long rows, int inter, hipStream_t stream) {
-  if (inter % 8 || inter > QCHUNKS * 256 * 8)
+  if (inter % 8 || inter > 2 * QCHUNKS * 256 * 8)
     throw std::runtime_error("silu_mul_fp8: bad inter");
-  silu_mul_fp8_kernel<<<dim3((unsigned)rows), 256, 0, stream>>>(
-      out, scales, x, inter);
+  if (inter <= QCHUNKS * 256 * 8)
+    silu_mul_fp8_kernel<QCHUNKS><<<dim3((unsigned)rows), 256, 0, stream>>>(
+        out, scales, x, inter);
+  else
+    silu_mul_fp8_kernel<2 * QCHUNKS>
+        <<<dim3((unsigned)rows), 256, 0, stream>>>(out, scales, x, inter);
 }
