@@ -170,6 +170,14 @@ void LlmGatewayModule::init(ModuleCtx& ctx) {
     for (int i = 0; i < (int)worker_cfg_.at("count").as_int(1); ++i)
       devices.push_back(i);
   if (devices.empty()) devices.push_back(0);
+  // a tp>1 worker is itself a torchrun group spanning tp GPUs — it owns
+  // the device selection (LOCAL_RANK); don't also shard by fleet index
+  if (worker_cfg_.is_object() && worker_cfg_.at("tp").as_int(1) > 1 &&
+      devices.size() > 1) {
+    LOG_INFO("llm-gateway",
+             "worker.tp>1: collapsing fleet to one TP group");
+    devices.resize(1);
+  }
   for (size_t i = 0; i < devices.size(); ++i) {
     auto w = std::make_unique<Worker>();
     w->index = (int)i;
