@@ -112,6 +112,7 @@ class EngineConfig:
     dtype: str = "bfloat16"
     quant: Optional[str] = None          # None (bf16) | "fp8" (e4m3fn weights)
     kv_dtype: str = "bfloat16"           # "bfloat16" | "fp8" (e4m3fn cache)
+    enable_prefix_caching: bool = False  # shared-prompt KV block reuse
     seed: int = 0
     # decode hipGraph capture batch buckets (padded up to nearest)
     graph_batch_sizes: tuple = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128,

@@ -66,7 +66,12 @@ class LLMEngine:
             return []
         tokens = self.runner.execute(batch)
         outputs: List[StepOutput] = []
+        bm = self.runner.block_manager
         for req, tok in zip(batch.requests, tokens.tolist()):
+            if batch.mode == "prefill" and bm.enable_prefix_caching:
+                # the chunk's full prompt blocks are now computed and
+                # shareable with later prompts
+                bm.commit_hashes(req.request_id, req.num_computed_tokens)
             if batch.mode == "prefill" and \
                     req.num_computed_tokens < req.num_prompt_tokens:
                 continue     # mid-prompt chunk: no token is sampled yet

@@ -31,7 +31,8 @@ class BlockManager:
     """
 
     def __init__(self, num_blocks: int, block_size: int,
-                 capacity: int = 1024, max_blocks_per_seq: int = 2048):
+                 capacity: int = 1024, max_blocks_per_seq: int = 2048,
+                 enable_prefix_caching: bool = False):
         self.num_blocks = num_blocks
         self.block_size = block_size
         self.capacity = capacity
@@ -43,10 +44,45 @@ class BlockManager:
                                   dtype=np.int32)
         self.ntables_np = np.zeros(capacity, dtype=np.int32)
         self.tokens_np = np.zeros(capacity, dtype=np.int64)
+        # ---- prefix caching (hash-chained full prompt blocks) ----
+        # A block becomes shareable only once its tokens are COMPUTED
+        # (commit_hashes after the prefill chunk) — never while a sibling
+        # in the same step could read it half-written.
+        self.enable_prefix_caching = enable_prefix_caching
+        self.hash_to_block: Dict[int, int] = {}     # committed blocks
+        self.block_hash: Dict[int, int] = {}
+        self.block_ref: Dict[int, int] = {}
+        from collections import OrderedDict
+        self.evictable: "OrderedDict[int, int]" = OrderedDict()  # blk->hash
+        self.pending_hashes: Dict[str, List] = {}   # seq -> [(idx, hash)]
+        self.cache_hits = 0
+        self.cache_queries = 0
+
+    @staticmethod
+    def block_hashes(token_ids: List[int], block_size: int) -> List[int]:
+        """Chained content hashes, one per FULL block of the prompt."""
+        out = []
+        h = 0
+        for i in range(len(token_ids) // block_size):
+            h = hash((h, tuple(token_ids[i * block_size:(i + 1)
+                                         * block_size])))
+            out.append(h)
+        return out
+
+    def _take_block(self) -> int:
+        if self.free_blocks:
+            return self.free_blocks.pop()
+        if self.evictable:               # evict the oldest shareable block
+            blk, bh = self.evictable.popitem(last=False)
+            self.hash_to_block.pop(bh, None)
+            self.block_hash.pop(blk, None)
+            return blk
+        raise RuntimeError("KV pool exhausted")
 
     @property
     def num_free(self) -> int:
-        return len(self.free_blocks)
+        # evictable cached blocks are reclaimable on demand
+        return len(self.free_blocks) + len(self.evictable)
 
     def blocks_needed(self, num_tokens: int) -> int:
         return (num_tokens + self.block_size - 1) // self.block_size
@@ -62,11 +98,75 @@ class BlockManager:
             raise RuntimeError("sequence table capacity exhausted")
         row = self._free_rows.pop()
         self.row_of[seq_id] = row
-        blocks = [self.free_blocks.pop() for _ in range(need)]
+        blocks = [self._take_block() for _ in range(need)]
         self.tables_np[row, :need] = blocks
         self.ntables_np[row] = need
         self.tokens_np[row] = num_tokens
         return blocks
+
+    def allocate_with_prefix(self, seq_id: str, num_tokens: int,
+                             token_ids: List[int]) -> int:
+        """Allocate like allocate(), but reuse committed shared-prefix
+        blocks; returns the number of CACHED tokens (KV already computed,
+        prefill may start there).  Never caches the final block of the
+        prompt (its last token's KV is computed with the first output)."""
+        if not self.enable_prefix_caching:
+            self.allocate(seq_id, num_tokens)
+            return 0
+        hashes = self.block_hashes(token_ids, self.block_size)
+        self.cache_queries += 1
+        reused: List[int] = []
+        for h in hashes[:max(0, len(hashes) - 1)]:
+            blk = self.hash_to_block.get(h)
+            if blk is None:
+                break
+            reused.append(blk)
+        need_total = self.blocks_needed(num_tokens)
+        if need_total - len(reused) > self.num_free:
+            raise RuntimeError("KV pool exhausted")
+        if not self._free_rows:
+            raise RuntimeError("sequence table capacity exhausted")
+        if reused:
+            self.cache_hits += 1
+        row = self._free_rows.pop()
+        self.row_of[seq_id] = row
+        blocks: List[int] = []
+        for blk in reused:
+            self.block_ref[blk] = self.block_ref.get(blk, 0) + 1
+            self.evictable.pop(blk, None)    # pinned while referenced
+            blocks.append(blk)
+        fresh_start = len(reused)
+        for _ in range(need_total - fresh_start):
+            blocks.append(self._take_block())
+        self.tables_np[row, :need_total] = blocks
+        self.ntables_np[row] = need_total
+        self.tokens_np[row] = num_tokens
+        # fresh FULL blocks become shareable once computed
+        self.pending_hashes[seq_id] = [
+            (i, hashes[i]) for i in range(fresh_start,
+                                          max(0, len(hashes) - 1))]
+        return fresh_start * self.block_size
+
+    def commit_hashes(self, seq_id: str, computed_tokens: int) -> None:
+        """Publish hashes of fully-computed prompt blocks (post-chunk)."""
+        pend = self.pending_hashes.get(seq_id)
+        if not pend:
+            return
+        row = self.row_of[seq_id]
+        rest = []
+        for idx, h in pend:
+            if (idx + 1) * self.block_size <= computed_tokens:
+                blk = int(self.tables_np[row, idx])
+                if h not in self.hash_to_block:
+                    self.hash_to_block[h] = blk
+                    self.block_hash[blk] = h
+                    self.block_ref[blk] = self.block_ref.get(blk, 0) + 1
+            else:
+                rest.append((idx, h))
+        if rest:
+            self.pending_hashes[seq_id] = rest
+        else:
+            self.pending_hashes.pop(seq_id, None)
 
     def can_extend(self, seq_id: str, num_tokens: int,
                    watermark: int = 0) -> bool:
@@ -84,7 +184,7 @@ class BlockManager:
             raise RuntimeError("KV pool exhausted")
         for _ in range(need):
             nt = int(self.ntables_np[row])
-            self.tables_np[row, nt] = self.free_blocks.pop()
+            self.tables_np[row, nt] = self._take_block()
             self.ntables_np[row] = nt + 1
         self.tokens_np[row] = total
 
@@ -101,10 +201,8 @@ class BlockManager:
         row = self.row_of[seq_id]
         n = int(self.tokens_np[row])
         if n % self.block_size == 0:
-            if not self.free_blocks:
-                raise RuntimeError("KV pool exhausted")
             nt = int(self.ntables_np[row])
-            self.tables_np[row, nt] = self.free_blocks.pop()
+            self.tables_np[row, nt] = self._take_block()
             self.ntables_np[row] = nt + 1
         self.tokens_np[row] = n + 1
         blk = int(self.tables_np[row, n // self.block_size])
@@ -119,9 +217,20 @@ class BlockManager:
         row = self.row_of.pop(seq_id, None)
         if row is None:
             return
+        self.pending_hashes.pop(seq_id, None)
         nt = int(self.ntables_np[row])
-        self.free_blocks.extend(int(b)
-                                for b in self.tables_np[row, :nt][::-1])
+        for b in self.tables_np[row, :nt][::-1]:
+            blk = int(b)
+            h = self.block_hash.get(blk)
+            if h is not None:
+                rc = self.block_ref.get(blk, 1) - 1
+                if rc > 0:
+                    self.block_ref[blk] = rc
+                else:
+                    self.block_ref.pop(blk, None)
+                    self.evictable[blk] = h     # reusable, evict-on-demand
+            else:
+                self.free_blocks.append(blk)
         self.ntables_np[row] = 0
         self.tokens_np[row] = 0
         self._free_rows.append(row)
@@ -139,10 +248,8 @@ class BlockManager:
         boundary = np.nonzero(n % self.block_size == 0)[0]
         for i in boundary:          # rare: one new block per 16 steps/seq
             row = int(rows[i])
-            if not self.free_blocks:
-                raise RuntimeError("KV pool exhausted")
             nt = int(self.ntables_np[row])
-            self.tables_np[row, nt] = self.free_blocks.pop()
+            self.tables_np[row, nt] = self._take_block()
             self.ntables_np[row] = nt + 1
         blk = self.tables_np[rows, n // self.block_size]
         slots = blk.astype(np.int64) * self.block_size + n % self.block_size

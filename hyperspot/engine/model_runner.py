@@ -88,7 +88,8 @@ class ModelRunner:
         self.block_manager = BlockManager(
             self.num_blocks, config.block_size,
             capacity=max(config.max_num_seqs * 2, 64),
-            max_blocks_per_seq=self.max_blocks_per_seq)
+            max_blocks_per_seq=self.max_blocks_per_seq,
+            enable_prefix_caching=config.enable_prefix_caching)
         self._gen = torch.Generator().manual_seed(config.seed)
         # pinned staging for the decode hot path (vectorized input prep)
         pin = self.device.type == "cuda"

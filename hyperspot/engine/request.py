@@ -39,6 +39,7 @@ class Request:
     num_computed_tokens: int = 0
     chunk_start: int = 0
     chunk_len: int = 0
+    blocks_preallocated: bool = False   # prefix-caching whole-prompt alloc
     state: RequestState = RequestState.WAITING
     finish_reason: Optional[FinishReason] = None
     first_token_time: Optional[float] = None

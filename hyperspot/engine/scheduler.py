@@ -83,10 +83,23 @@ class Scheduler:
             done = req.num_computed_tokens
             if done == 0 and len(self.running) + len(batch.requests)                     >= self.config.max_num_seqs:
                 break
+            if done == 0 and self.bm.enable_prefix_caching:
+                # shared-prefix reuse: whole-prompt allocation up front,
+                # compute starts past the cached prefix (the chunked-
+                # prefill machinery handles a mid-prompt start; the
+                # continuation-attention path reads the reused pages)
+                if not self.bm.can_allocate(total, self.watermark):
+                    break
+                cached = self.bm.allocate_with_prefix(
+                    req.request_id, total, req.prompt_token_ids)
+                req.num_computed_tokens = done = cached
+                req.blocks_preallocated = True
             chunk = min(total - done, budget)
             if chunk <= 0:
                 break
-            if done == 0:
+            if getattr(req, "blocks_preallocated", False):
+                pass      # every prompt block already allocated
+            elif done == 0:
                 if not self.bm.can_allocate(chunk, self.watermark):
                     break
                 self.bm.allocate(req.request_id, chunk)
@@ -126,6 +139,7 @@ class Scheduler:
             victim.prompt_token_ids = victim.prompt_token_ids + victim.output_token_ids
             victim.output_token_ids = []
             victim.num_computed_tokens = 0
+            victim.blocks_preallocated = False
             self.waiting.appendleft(victim)
         batch.requests = list(self.running)
         batch.num_tokens = len(batch.requests)
