@@ -91,3 +91,27 @@ def test_chunked_prefill_matches_unchunked_gpu():
         import torch
         torch.cuda.empty_cache()
     assert outs[0] == outs[1], outs
+
+
+@pytest.mark.gpu
+def test_prefix_caching_matches_uncached_gpu():
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    prompt = [(i * 29) % 120000 + 2 for i in range(120)]
+    sp = SamplingParams(temperature=0.0, max_tokens=4)
+    outs = {}
+    for flag in (False, True):
+        cfg = EngineConfig(model="llama3-8b", max_num_seqs=4,
+                           max_num_batched_tokens=2048, max_model_len=512,
+                           num_gpu_blocks=256, enforce_eager=True, seed=2,
+                           enable_prefix_caching=flag)
+        eng = LLMEngine(cfg, device="cuda:0")
+        a = eng.generate([prompt], sp)[0]
+        b = eng.generate([prompt], sp)[0]   # second run hits the cache
+        assert a == b
+        if flag:
+            assert eng.runner.block_manager.cache_hits == 1
+        outs[flag] = a
+        del eng
+        import torch
+        torch.cuda.empty_cache()
+    assert outs[False] == outs[True]
