@@ -63,7 +63,9 @@ class WorkerState:
     def abort(self, rid):
         with self.new_work:
             self.engine.abort_request(rid)
-            self.streams.pop(rid, None)
+            q = self.streams.pop(rid, None)
+            if q is not None:
+                q.put(None)    # wake the serving thread (abort sentinel)
             if self.tp > 1:
                 self.pending_ops.append(("abort", rid))
                 self.new_work.notify()
@@ -210,6 +212,11 @@ def _run_chat(msg, state: WorkerState, send):
         except queue.Empty:
             state.abort(rid)
             send({"event": "error", "id": rid, "message": "engine stall"})
+            return
+        if out is None:   # aborted from another connection
+            send({"event": "done", "id": rid, "finish_reason": "abort",
+                  "usage": {"input_tokens": len(prompt_ids),
+                            "output_tokens": n_out}})
             return
         n_out += 1
         text = detok.push(out.token_id)

@@ -97,3 +97,48 @@ def test_tp2_worker_matches_tp1():
                 proc.wait()
     # greedy TP=2 must reproduce TP=1 exactly (same seed, same op order)
     assert out[1] == out[2], out
+
+
+def test_abort_unblocks_running_chat():
+    """An abort sent on a SECOND connection must end the first
+    connection's stream promptly (finish_reason=abort), not leave the
+    worker thread blocked."""
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-ab-")
+    proc = _spawn_worker(1, sock, 0)
+    try:
+        _wait_sock(sock, proc)
+        s1 = socketlib.socket(socketlib.AF_UNIX, socketlib.SOCK_STREAM)
+        s1.connect(sock)
+        f1 = s1.makefile("rwb")
+        f1.write((json.dumps({
+            "type": "chat", "id": "long1",
+            "messages": [{"role": "user",
+                          "content": [{"type": "text", "text": "go"}]}],
+            "params": {"temperature": 0.0, "max_tokens": 100000,
+                       "ignore_eos": True}}) + "\n").encode())
+        f1.flush()
+        # let a few tokens stream, then abort from a second connection
+        first = json.loads(next(iter(f1)))
+        assert first["event"] == "delta"
+        s2 = socketlib.socket(socketlib.AF_UNIX, socketlib.SOCK_STREAM)
+        s2.connect(sock)
+        s2.sendall((json.dumps({"type": "abort", "id": "long1"})
+                    + "\n").encode())
+        t0 = time.time()
+        done = None
+        for raw in f1:
+            m = json.loads(raw)
+            if m["event"] == "done":
+                done = m
+                break
+        assert done is not None and done["finish_reason"] == "abort"
+        assert time.time() - t0 < 10, "abort did not unblock promptly"
+        s1.close()
+        s2.close()
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=20)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+            proc.wait()
