@@ -50,7 +50,9 @@ class Scheduler:
             for r in list(q):
                 if r.request_id == request_id:
                     q.remove(r)
-                    if r.state == RequestState.RUNNING:
+                    # a mid-prefill (chunked) request holds pages while
+                    # still in `waiting` — free by ownership, not state
+                    if r.request_id in self.bm.row_of:
                         self.bm.free(r.request_id)
                     r.state = RequestState.FINISHED
 

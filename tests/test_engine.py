@@ -190,3 +190,20 @@ def test_chunked_prefill_interleaves_short_prompts():
     out = eng.generate([long_p, [5, 6, 7], [9, 10]],
                        SamplingParams(temperature=0.0, max_tokens=4))
     assert all(len(o) == 4 for o in out), out
+
+
+def test_abort_mid_chunked_prefill_frees_pages():
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    cfg = EngineConfig(model="tiny-llama", max_num_seqs=4,
+                       max_num_batched_tokens=32, max_model_len=512,
+                       num_gpu_blocks=64, enforce_eager=True)
+    eng = LLMEngine(cfg, device="cpu")
+    bm = eng.runner.block_manager
+    free0 = bm.num_free
+    rid = eng.add_request(list(range(2, 150)),
+                          SamplingParams(temperature=0.0, max_tokens=4))
+    eng.step()                       # first 32-token chunk only
+    assert bm.num_free < free0
+    eng.abort_request(rid)
+    assert bm.num_free == free0      # pages reclaimed, no leak
+    assert not eng.has_work()
