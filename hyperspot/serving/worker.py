@@ -64,6 +64,7 @@ class WorkerState:
         with self.new_work:
             self.engine.abort_request(rid)
             q = self.streams.pop(rid, None)
+            self.started.pop(rid, None)
             if q is not None:
                 q.put(None)    # wake the serving thread (abort sentinel)
             if self.tp > 1:
@@ -91,6 +92,7 @@ class WorkerState:
                     q.put(out)
                     if out.finished:
                         self.streams.pop(out.request_id, None)
+                        self.started.pop(out.request_id, None)
 
 
 def follower_loop(engine):
