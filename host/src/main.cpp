@@ -182,8 +182,28 @@ int main(int argc, char** argv) {
   }
   if (do_list) return list_modules(cfg);
   if (command == "migrate") {
-    // migrations run per-module; the in-memory stores need none
-    std::cout << "migrations: nothing to do\n";
+    // run every db-bearing module's migrations and exit (the reference's
+    // separate cloud deploy step, bootstrap/run.rs:111): init-phase only,
+    // no server, no workers
+    ClientHub hub;
+    bool cancel = false;
+    for (auto m : {std::shared_ptr<Module>(
+                       std::make_shared<AuthzResolverModule>()),
+                   std::shared_ptr<Module>(
+                       std::make_shared<TenantResolverModule>()),
+                   std::shared_ptr<Module>(
+                       std::make_shared<UserSettingsModule>()),
+                   std::shared_ptr<Module>(
+                       std::make_shared<UsersInfoModule>())}) {
+      ModuleCtx mc;
+      mc.full_config = cfg;
+      mc.config = cfg.path("modules." + m->name() + ".config");
+      mc.hub = &hub;
+      mc.home_dir = cfg.path("server.home_dir").as_string("~/.hyperspot");
+      mc.cancel = &cancel;
+      m->init(mc);
+      std::cout << "migrated: " << m->name() << "\n";
+    }
     return 0;
   }
   return run_server(cfg, command == "check");

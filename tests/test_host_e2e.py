@@ -1251,3 +1251,19 @@ def test_capability_not_supported(server):
                                       "url": "x://y"}]}]})
     assert st == 400
     assert json.loads(body)["code"] == "capability_not_supported"
+
+
+def test_migrate_command(tmp_path):
+    """hyperspot-server migrate applies module migrations and exits
+    (the reference's separate deploy step, bootstrap/run.rs:111)."""
+    cfg = tmp_path / "m.yaml"
+    cfg.write_text(f'server:\n  home_dir: "{tmp_path}"\n')
+    r = subprocess.run([str(BIN), "migrate", "--config", str(cfg)],
+                       capture_output=True, text=True, timeout=60)
+    assert r.returncode == 0, r.stderr
+    assert "migrated: users-info" in r.stdout
+    assert (tmp_path / "users-info.db").exists()
+    # idempotent
+    r2 = subprocess.run([str(BIN), "migrate", "--config", str(cfg)],
+                        capture_output=True, text=True, timeout=60)
+    assert r2.returncode == 0
