@@ -120,11 +120,15 @@ class MixtralMoE(nn.Module):
 
     def _ep_moe(self, x, weights, ids):
         """Dispatch tokens to expert owners (all-to-all), run local experts,
-        combine back (all-to-all), weight and reduce."""
+        combine back (all-to-all), weight and reduce.  At quant=fp8 the
+        dispatched activations travel as e4m3fn bytes + per-token f32
+        scales — half the xGMI wire bytes — and feed the fp8 grouped GEMM
+        directly on the owner rank."""
         ep = get_ep_size()
         group = get_ep_group()
         T, H = x.shape
         K = self.top_k
+        fp8_wire = self.quant == "fp8"
         flat_ids = ids.reshape(-1)                         # [T*K]
         owner = flat_ids // self.experts_per_rank          # [T*K]
         order = torch.argsort(owner, stable=True)
@@ -136,18 +140,41 @@ class MixtralMoE(nn.Module):
         dist.all_to_all_single(recv_counts, send_counts, group=group)
         sc = send_counts.tolist()
         rc = recv_counts.tolist()
-        recv_x = torch.empty(sum(rc), H, dtype=x.dtype, device=x.device)
         recv_eids = torch.empty(sum(rc), dtype=send_eids.dtype, device=x.device)
-        dist.all_to_all_single(recv_x, send_x, rc, sc, group=group)
         dist.all_to_all_single(recv_eids, send_eids, rc, sc, group=group)
+        if fp8_wire:
+            from hyperspot.parallel.layers import quant_fp8_rowwise
+            sq, ss = quant_fp8_rowwise(send_x)
+            recv_q = torch.empty(sum(rc), H, dtype=torch.uint8,
+                                 device=x.device)
+            recv_s = torch.empty(sum(rc), dtype=torch.float32,
+                                 device=x.device)
+            # fp8 bytes travel as uint8 (gloo/rccl dtype-agnostic payload)
+            dist.all_to_all_single(recv_q, sq.view(torch.uint8), rc, sc,
+                                   group=group)
+            dist.all_to_all_single(recv_s, ss, rc, sc, group=group)
+            recv_x = None
+        else:
+            recv_x = torch.empty(sum(rc), H, dtype=x.dtype, device=x.device)
+            dist.all_to_all_single(recv_x, send_x, rc, sc, group=group)
         # local grouped FFN
         local = recv_eids - self.local_expert_start
-        if recv_x.is_cuda and ops.have_native():
+        if fp8_wire and x.is_cuda and ops.have_native():
+            xq = ops.QTensor(recv_q.view(torch.float8_e4m3fn), recv_s)
+            ones = torch.ones(recv_q.shape[0], 1, device=x.device,
+                              dtype=torch.float32)
+            y = ops.moe_ffn_fp8(xq, self.w13, self.w13_scale, self.w2,
+                                self.w2_scale, ones,
+                                local.view(-1, 1).to(torch.int32))
+        elif not fp8_wire and x.is_cuda and ops.have_native():
             ones = torch.ones(recv_x.shape[0], 1, device=x.device,
                               dtype=torch.float32)
             y = ops.moe_ffn(recv_x, self.w13, self.w2, ones,
                             local.view(-1, 1).to(torch.int32))
         else:
+            if fp8_wire:    # CPU/EP test path: dequantize the wire bytes
+                recv_x = (recv_q.view(torch.float8_e4m3fn).float()
+                          * recv_s[:, None]).to(x.dtype)
             y = torch.empty_like(recv_x)
             for e in range(self.experts_per_rank):
                 m = local == e

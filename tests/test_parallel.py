@@ -124,3 +124,52 @@ def test_ep2_moe_matches_ep1():
     ref = moe(x)
     for r in (0, 1):
         assert torch.allclose(results[r], ref, atol=1e-4), r
+
+
+def _run_ep_moe_fp8(rank, world, port, results):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from hyperspot.engine.config import get_model_spec
+    from hyperspot.models.mixtral import MixtralMoE
+    from hyperspot.parallel import layers as L
+    from hyperspot.parallel.state import (initialize_model_parallel,
+                                          destroy_model_parallel)
+    initialize_model_parallel(tp_size=1, ep_size=world)
+    L.set_quant_mode("fp8")
+    try:
+        spec = get_model_spec("tiny-moe")
+        moe = MixtralMoE(spec, layer_idx=0, dtype=torch.float32)
+        torch.manual_seed(3)
+        x = torch.randn(6, spec.hidden_size)
+        y = moe(x)
+        results[rank] = y.detach()
+    finally:
+        L.set_quant_mode(None)
+    dist.barrier()
+    destroy_model_parallel()
+    dist.destroy_process_group()
+
+
+def test_ep2_fp8_wire_matches_single_rank():
+    """fp8 EP: e4m3fn activation bytes + scales over the all-to-all must
+    reproduce the single-rank fp8 MoE (same quant model both sides)."""
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_run_ep_moe_fp8, args=(2, 29617, results), nprocs=2, join=True)
+    from hyperspot.engine.config import get_model_spec
+    from hyperspot.models.mixtral import MixtralMoE
+    from hyperspot.parallel import layers as L
+    L.set_quant_mode("fp8")
+    try:
+        spec = get_model_spec("tiny-moe")
+        moe = MixtralMoE(spec, layer_idx=0, dtype=torch.float32)
+        torch.manual_seed(3)
+        x = torch.randn(6, spec.hidden_size)
+        ref = moe(x)
+    finally:
+        L.set_quant_mode(None)
+    for r in (0, 1):
+        assert torch.allclose(results[r], ref, atol=2e-2, rtol=2e-2), \
+            (results[r] - ref).abs().max()
