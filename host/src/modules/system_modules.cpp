@@ -566,9 +566,13 @@ void CredStoreModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
   rest.register_op(get, [](HttpRequest& rq, ResponseWriter& w) {
     auto v = g_creds->get(sec_of(rq).tenant_id, rq.path_params["ref"]);
     if (!v) throw Problem::not_found();   // never reveal existence
+    // write-only store: secrets are consumed by the OAGW credential
+    // injector, never read back over REST (credstore DESIGN: no secret
+    // logging / no secret egress)
     Json out = Json::object();
     out["ref"] = rq.path_params["ref"];
-    out["value"] = *v;
+    out["exists"] = true;
+    out["value_length"] = (long)v->size();
     w.respond(200, "application/json", out.dump());
   });
   OperationSpec del;

@@ -263,7 +263,9 @@ def test_credstore_tenant_scoping(server):
     st, _ = _http("PUT", base, body={"value": "s3cr3t"})
     assert st == 204
     st, body = _http("GET", base)
-    assert st == 200 and json.loads(body)["value"] == "s3cr3t"
+    d = json.loads(body)
+    assert st == 200 and d["exists"] is True
+    assert "s3cr3t" not in body     # write-only: value never read back
     st, _ = _http("DELETE", base)
     assert st == 204
     st, _ = _http("GET", base)
@@ -380,7 +382,8 @@ def test_credstore_isolated_between_tenants(mt_server):
     st, _ = _http("GET", base, token="root-token")
     assert st == 404
     st, body = _http("GET", base, token="acme-token")
-    assert st == 200 and json.loads(body)["value"] == "acme-secret"
+    assert st == 200 and json.loads(body)["exists"] is True
+    assert "acme-secret" not in body
 
 
 def test_cli_check_and_introspection():
