@@ -1270,3 +1270,57 @@ def test_migrate_command(tmp_path):
     r2 = subprocess.run([str(BIN), "migrate", "--config", str(cfg)],
                         capture_output=True, text=True, timeout=60)
     assert r2.returncode == 0
+
+
+@pytest.fixture(scope="module")
+def timeout_server():
+    """llm-gateway with a 1ms total budget: every chat must 504 with
+    provider_timeout (DESIGN.md:706-741 state machine)."""
+    import tempfile
+    port = _free_port()
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-to-")
+    cfg = f"""
+server:
+  home_dir: "/tmp/hs-e2e-to"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+  llm-gateway:
+    config:
+      model: "tiny-llama"
+      worker_socket: "{sock}"
+      auto_start_worker: true
+      timeouts:
+        total_ms: 1
+      worker:
+        device: "cpu"
+        eager: true
+        max_num_seqs: 4
+        num_gpu_blocks: 128
+"""
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(cfg)
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        srv.wait_worker()
+        yield srv
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
+
+
+def test_total_timeout_504(timeout_server):
+    url = BASE.format(timeout_server.port)
+    st, body = _http("POST", url + "/v1/chat/completions",
+                     {"model": "tiny-llama",
+                      "messages": [{"role": "user", "content":
+                                    [{"type": "text", "text": "slow"}]}],
+                      "max_tokens": 2000})
+    assert st == 504, body
+    assert json.loads(body)["code"] == "provider_timeout"
