@@ -344,6 +344,42 @@ void NodesRegistryModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
       throw Problem::not_found("unknown node");
     w.respond(200, "application/json", node_info().dump());
   });
+  // syscap: hardware capabilities incl. the GPU inventory (the reference
+  // collects GpuInfo via NVML; the MI355X-native equivalent enumerates
+  // amdgpu devices from sysfs — no vendor daemon needed)
+  OperationSpec syscap_op;
+  syscap_op.method = "GET";
+  syscap_op.path = "/nodes-registry/v1/nodes/{id}/syscap";
+  syscap_op.operation_id = "nodes_syscap";
+  syscap_op.authenticated = true;
+  syscap_op.tags = {"nodes-registry"};
+  rest.register_op(syscap_op, [this](HttpRequest& rq, ResponseWriter& w) {
+    if (rq.path_params["id"] != "local")
+      throw Problem::not_found("unknown node");
+    Json out = node_info();
+    Json gpus = Json::array();
+    for (int card = 0; card < 16; ++card) {
+      const std::string base =
+          "/sys/class/drm/card" + std::to_string(card) + "/device/";
+      std::ifstream vf(base + "vendor");
+      std::string vendor;
+      if (!(vf >> vendor)) continue;
+      if (vendor != "0x1002") continue;            // AMD
+      Json g = Json::object();
+      g["index"] = (long)gpus.size();
+      std::ifstream df(base + "device");
+      std::string dev;
+      if (df >> dev) g["device_id"] = dev;
+      std::ifstream mf(base + "mem_info_vram_total");
+      long long vram = 0;
+      if (mf >> vram) g["vram_mb"] = (long)(vram / (1024 * 1024));
+      g["vendor"] = "AMD";
+      gpus.push_back(g);
+    }
+    out["gpus"] = gpus;
+    out["gpu_count"] = (long)gpus.size();
+    w.respond(200, "application/json", out.dump());
+  });
 }
 
 // -------------------------------------------------------- model-registry

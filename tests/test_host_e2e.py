@@ -1324,3 +1324,12 @@ def test_total_timeout_504(timeout_server):
                       "max_tokens": 2000})
     assert st == 504, body
     assert json.loads(body)["code"] == "provider_timeout"
+
+
+def test_nodes_syscap(server):
+    st, body = _http("GET",
+                     BASE.format(server.port)
+                     + "/nodes-registry/v1/nodes/local/syscap")
+    assert st == 200, body
+    d = json.loads(body)
+    assert "gpu_count" in d and isinstance(d["gpus"], list)
