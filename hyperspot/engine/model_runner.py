@@ -312,7 +312,21 @@ class ModelRunner:
                              dtype=torch.float32)
         top_k = torch.tensor([r.sampling.top_k for r in requests],
                              dtype=torch.int32)
-        uniform = torch.rand(B, generator=self._gen)
+        if any(r.sampling.seed is not None for r in requests):
+            # per-request reproducible draws (SamplingParams.seed)
+            us = []
+            for r in requests:
+                if r.sampling.seed is not None:
+                    g = getattr(r, "_seed_gen", None)
+                    if g is None:
+                        g = torch.Generator().manual_seed(r.sampling.seed)
+                        r._seed_gen = g
+                    us.append(torch.rand(1, generator=g))
+                else:
+                    us.append(torch.rand(1, generator=self._gen))
+            uniform = torch.cat(us)
+        else:
+            uniform = torch.rand(B, generator=self._gen)
         d = logits.device
         return ops.sample(logits, temps.to(d), top_p.to(d), top_k.to(d),
                           uniform.to(d)).cpu()

@@ -207,3 +207,20 @@ def test_abort_mid_chunked_prefill_frees_pages():
     eng.abort_request(rid)
     assert bm.num_free == free0      # pages reclaimed, no leak
     assert not eng.has_work()
+
+
+def test_per_request_sampling_seed_reproducible():
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+
+    def run(seed):
+        cfg = EngineConfig(model="tiny-llama", max_num_seqs=2,
+                           max_num_batched_tokens=128, max_model_len=64,
+                           num_gpu_blocks=32, enforce_eager=True, seed=0)
+        eng = LLMEngine(cfg, device="cpu")
+        return eng.generate([[3, 4, 5]],
+                            SamplingParams(temperature=1.0, top_k=50,
+                                           max_tokens=8, seed=seed))[0]
+    a1, a2 = run(123), run(123)
+    b = run(321)
+    assert a1 == a2          # same request seed => identical draws
+    assert a1 != b           # different seed diverges (w.h.p.)
