@@ -869,10 +869,14 @@ void LlmGatewayModule::job_loop() {
     try {
       SecurityContext sec;
       sec.tenant_id = job->tenant;
-      Json resolved =
-          resolve_model(sec, job->request.at("model").as_string());
       check_budget(job->tenant);
-      result = run_chat_blocking(job->request, resolved, job->id);
+      if (job->request.path("fallback.models").is_array()) {
+        result = run_chat_with_fallback(sec, job->request, job->id);
+      } else {
+        Json resolved =
+            resolve_model(sec, job->request.at("model").as_string());
+        result = run_chat_blocking(job->request, resolved, job->id);
+      }
     } catch (const Problem& p) {
       err = p.code.empty() ? "provider_error" : p.code;
     } catch (...) {
