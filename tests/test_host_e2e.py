@@ -1333,3 +1333,39 @@ def test_nodes_syscap(server):
     assert st == 200, body
     d = json.loads(body)
     assert "gpu_count" in d and isinstance(d["gpus"], list)
+
+
+def test_dp_fleet_survives_single_worker_crash(dp_server):
+    """Killing one of two workers must not fail traffic: the survivor
+    serves while the watchdog respawns the dead one."""
+    url = BASE.format(dp_server.port)
+    kids = subprocess.run(["pgrep", "-P", str(dp_server.proc.pid)],
+                          capture_output=True, text=True).stdout.split()
+    workers = []
+    for pid in kids:
+        try:
+            with open(f"/proc/{pid}/cmdline") as f:
+                if "hyperspot.serving.worker" in f.read():
+                    workers.append(int(pid))
+        except OSError:
+            pass
+    assert len(workers) == 2, workers
+    os.kill(workers[0], 9)
+    ok = 0
+    for i in range(6):
+        st, _ = _http("POST", url + "/v1/chat/completions",
+                      {"model": "tiny-llama",
+                       "messages": [{"role": "user", "content":
+                                     [{"type": "text", "text": f"f{i}"}]}],
+                       "max_tokens": 2}, timeout=30)
+        if st == 200:
+            ok += 1
+    assert ok == 6, ok
+    # respawn brings the fleet back to 2 live workers
+    deadline = time.time() + 60
+    while time.time() < deadline:
+        st, body = _http("GET", url + "/metrics")
+        if "hyperspot_workers_live 2" in body:
+            break
+        time.sleep(1)
+    assert "hyperspot_workers_live 2" in body
