@@ -210,6 +210,22 @@ bool LlmGatewayModule::worker_ready() {
   return false;
 }
 
+LlmGatewayModule::Worker* LlmGatewayModule::pick_live(
+    std::unique_ptr<EngineConn>& conn) {
+  for (size_t att = 0; att < workers_.size() + 1; ++att) {
+    Worker* w = pick_worker();
+    if (!w) return nullptr;
+    conn = std::make_unique<EngineConn>(w->socket);
+    if (conn->ok()) return w;
+    LOG_ERROR("llm-gateway", "worker %d connect failed; demoting",
+              w->index);
+    w->ready = false;          // stale: process died since the last probe
+    w->in_flight--;
+    conn.reset();
+  }
+  return nullptr;
+}
+
 LlmGatewayModule::Worker* LlmGatewayModule::pick_worker() {
   Worker* best = nullptr;
   for (auto& w : workers_) {
@@ -404,14 +420,6 @@ Json LlmGatewayModule::run_chat_blocking(const Json& body,
         std::chrono::steady_clock::now() - t0).count();
     return (int)std::max<long>(1, limit_ms - used);
   };
-  Lease lease{pick_worker()};
-  if (!lease.w)
-    throw Problem{503, "Service Unavailable", "about:blank",
-                  "no engine worker ready", "provider_error"};
-  EngineConn conn(lease.w->socket);
-  if (!conn.ok())
-    throw Problem{503, "Service Unavailable", "about:blank",
-                  "engine connection failed", "provider_error"};
   Json wreq = Json::object();
   wreq["type"] = "chat";
   wreq["id"] = rid;
@@ -423,52 +431,76 @@ Json LlmGatewayModule::run_chat_blocking(const Json& body,
                         "seed"})
     if (body.contains(f)) params[f] = body.at(f);
   wreq["params"] = params;
-  if (!conn.send_json(wreq))
-    throw Problem{502, "Bad Gateway", "about:blank", "engine write failed",
-                  "provider_error"};
-  std::string text;
-  Json usage;
-  std::string finish = "stop";
-  bool first = true;
-  while (true) {
-    // TTFT timer until the first delta, then the total timer
-    // (DESIGN.md:706-741)
-    auto msg = conn.read_json(
-        left(first && ttft_timeout_ms_ ? ttft_timeout_ms_
-                                       : total_timeout_ms_));
-    if (!msg) {
-      Json ab = Json::object();
-      ab["type"] = "abort";
-      ab["id"] = rid;
-      conn.send_json(ab);
-      throw Problem{504, "Gateway Timeout", "about:blank",
-                    first ? "no first token within the TTFT budget"
-                          : "generation exceeded the total budget",
-                    "provider_timeout"};
+
+  // a connect can land in a DYING worker's listen backlog (SIGKILL
+  // teardown window) and then EOF — before any token arrived the request
+  // is trivially retryable on another worker
+  for (int attempt = 0; attempt < 3; ++attempt) {
+    std::unique_ptr<EngineConn> connp;
+    Lease lease{pick_live(connp)};
+    if (!lease.w)
+      throw Problem{503, "Service Unavailable", "about:blank",
+                    "no engine worker ready", "provider_error"};
+    EngineConn& conn = *connp;
+    if (!conn.send_json(wreq)) {
+      lease.w->ready = false;
+      continue;
     }
-    const std::string ev = msg->at("event").as_string();
-    if (ev == "delta") { first = false; text += msg->at("text").as_string(); }
-    else if (ev == "done") {
-      usage = msg->at("usage");
-      finish = msg->at("finish_reason").as_string("stop");
-      break;
-    } else if (ev == "error") {
-      throw Problem{502, "Bad Gateway", "about:blank",
-                    msg->at("message").as_string(), "provider_error"};
+    std::string text;
+    Json usage;
+    std::string finish = "stop";
+    bool first = true;
+    bool retry = false;
+    while (true) {
+      // TTFT timer until the first delta, then the total timer
+      // (DESIGN.md:706-741)
+      auto msg = conn.read_json(
+          left(first && ttft_timeout_ms_ ? ttft_timeout_ms_
+                                         : total_timeout_ms_));
+      if (!msg) {
+        if (first && !ttft_timeout_ms_ && !total_timeout_ms_ &&
+            attempt < 2) {
+          // EOF with no deadline configured: dying-worker race — retry
+          lease.w->ready = false;
+          retry = true;
+          break;
+        }
+        Json ab = Json::object();
+        ab["type"] = "abort";
+        ab["id"] = rid;
+        conn.send_json(ab);
+        throw Problem{504, "Gateway Timeout", "about:blank",
+                      first ? "no first token within the TTFT budget"
+                            : "generation exceeded the total budget",
+                      "provider_timeout"};
+      }
+      const std::string ev = msg->at("event").as_string();
+      if (ev == "delta") { first = false; text += msg->at("text").as_string(); }
+      else if (ev == "done") {
+        usage = msg->at("usage");
+        finish = msg->at("finish_reason").as_string("stop");
+        break;
+      } else if (ev == "error") {
+        throw Problem{502, "Bad Gateway", "about:blank",
+                      msg->at("message").as_string(), "provider_error"};
+      }
     }
+    if (retry) continue;
+    Json part = Json::object();
+    part["type"] = "text";
+    part["text"] = text;
+    Json content = Json::array();
+    content.push_back(part);
+    Json resp = Json::object();
+    resp["content"] = content;
+    resp["usage"] = usage;
+    resp["model_used"] = resolved.at("canonical_id").as_string();
+    resp["fallback_used"] = false;
+    resp["finish_reason"] = finish;
+    return resp;
   }
-  Json part = Json::object();
-  part["type"] = "text";
-  part["text"] = text;
-  Json content = Json::array();
-  content.push_back(part);
-  Json resp = Json::object();
-  resp["content"] = content;
-  resp["usage"] = usage;
-  resp["model_used"] = resolved.at("canonical_id").as_string();
-  resp["fallback_used"] = false;
-  resp["finish_reason"] = finish;
-  return resp;
+  throw Problem{503, "Service Unavailable", "about:blank",
+                "engine workers unavailable", "provider_error"};
 }
 
 // Build a stream_chunk.v1-shaped SSE event
@@ -647,14 +679,6 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
                     "model_not_found"};
   }
   m_streams_++;
-  Lease lease{pick_worker()};
-  if (!lease.w)
-    throw Problem{503, "Service Unavailable", "about:blank",
-                  "no engine worker ready", "provider_error"};
-  EngineConn conn(lease.w->socket);
-  if (!conn.ok())
-    throw Problem{503, "Service Unavailable", "about:blank",
-                  "engine connection failed", "provider_error"};
   Json wreq = Json::object();
   wreq["type"] = "chat";
   wreq["id"] = rid;
@@ -666,9 +690,39 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
                         "seed"})
     if (body.contains(f)) params[f] = body.at(f);
   wreq["params"] = params;
-  if (!conn.send_json(wreq))
-    throw Problem{502, "Bad Gateway", "about:blank", "engine write failed",
-                  "provider_error"};
+
+  // dying-worker race (see run_chat_blocking): obtain the FIRST engine
+  // event before committing the SSE response, so a pre-token EOF can
+  // retry on another worker
+  std::unique_ptr<EngineConn> connp;
+  Lease lease;
+  std::optional<Json> first_msg;
+  for (int attempt = 0; attempt < 3 && !first_msg; ++attempt) {
+    if (lease.w) { lease.w->in_flight--; lease.w = nullptr; }
+    connp.reset();
+    Worker* w2 = pick_live(connp);
+    lease.w = w2;
+    if (!lease.w)
+      throw Problem{503, "Service Unavailable", "about:blank",
+                    "no engine worker ready", "provider_error"};
+    if (!connp->send_json(wreq)) {
+      lease.w->ready = false;
+      continue;
+    }
+    first_msg = connp->read_json(
+        ttft_timeout_ms_ ? (int)ttft_timeout_ms_ : 120000);
+    if (!first_msg) {
+      if (ttft_timeout_ms_)
+        throw Problem{504, "Gateway Timeout", "about:blank",
+                      "no first token within the TTFT budget",
+                      "provider_timeout"};
+      lease.w->ready = false;   // retry
+    }
+  }
+  if (!first_msg)
+    throw Problem{503, "Service Unavailable", "about:blank",
+                  "engine workers unavailable", "provider_error"};
+  EngineConn& conn = *connp;
 
   // SSE stream per DESIGN.md:289-311: role chunk, delta chunks, final
   // finish_reason+usage chunk, then data: [DONE]
@@ -681,8 +735,16 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   role_delta["role"] = "assistant";
   w.write_chunk(sse_chunk(rid, canonical, role_delta));
   bool client_gone = false;
+  bool pending_first = true;
   while (true) {
-    auto msg = conn.read_json();
+    std::optional<Json> msg;
+    if (pending_first) {
+      msg = first_msg;
+      pending_first = false;
+    } else {
+      msg = conn.read_json(
+          total_timeout_ms_ ? (int)total_timeout_ms_ : 120000);
+    }
     if (!msg) {
       w.write_chunk("data: {\"error\":\"provider_timeout\"}\n\n");
       break;
@@ -755,14 +817,12 @@ void LlmGatewayModule::embeddings_handler(HttpRequest& req,
     throw Problem{503, "Service Unavailable", "about:blank",
                   "inference engine is not ready", "provider_error"};
 
-  Lease lease{pick_worker()};
+  std::unique_ptr<EngineConn> connp;
+  Lease lease{pick_live(connp)};
   if (!lease.w)
     throw Problem{503, "Service Unavailable", "about:blank",
                   "no engine worker ready", "provider_error"};
-  EngineConn conn(lease.w->socket);
-  if (!conn.ok())
-    throw Problem{503, "Service Unavailable", "about:blank",
-                  "engine connection failed", "provider_error"};
+  EngineConn& conn = *connp;
   Json wreq = Json::object();
   wreq["type"] = "embeddings";
   wreq["input"] = input;
@@ -1379,13 +1439,14 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
                         "\"validation_error\"}");
           continue;
         }
-        Lease lease{pick_worker()};
+        std::unique_ptr<EngineConn> connp;
+        Lease lease{pick_live(connp)};
         if (!lease.w) {
           ws->send_text("{\"type\":\"error\",\"code\":"
                         "\"provider_error\"}");
           continue;
         }
-        EngineConn conn(lease.w->socket);
+        EngineConn& conn = *connp;
         const std::string rid =
             "rt-" + std::to_string(req_ctr_.fetch_add(1));
         Json wreq = Json::object();

@@ -133,6 +133,9 @@ class LlmGatewayModule : public Module {
   };
   std::vector<std::unique_ptr<Worker>> workers_;
   Worker* pick_worker();                    // least-loaded ready worker
+  // pick + connect with failover: a dead-but-marked-ready worker is
+  // demoted (watchdog respawns it) and the next one is tried
+  Worker* pick_live(std::unique_ptr<EngineConn>& conn);
   bool probe_worker(Worker& wk);
   void spawn_one(Worker& wk);
   std::atomic<uint64_t> req_ctr_{0};
