@@ -74,3 +74,16 @@ def test_concurrent_shared_prefix_same_batch():
     plain = _engine()
     ref = plain.generate([a, b], sp)
     assert outs == ref
+
+
+def test_preemption_with_caching_recovers():
+    """Preempted victims re-admit through the caching path and may reuse
+    their own committed prefix; outputs must match a roomy engine."""
+    sp = SamplingParams(temperature=0.0, max_tokens=24)
+    prompts = [[(s * 41 + i) % 350 + 3 for i in range(40)]
+               for s in range(3)]
+    roomy = _engine(enable_prefix_caching=True, num_gpu_blocks=256)
+    ref = roomy.generate(prompts, sp)
+    tight = _engine(enable_prefix_caching=True, num_gpu_blocks=20)
+    out = tight.generate(prompts, sp)
+    assert out == ref
