@@ -297,6 +297,11 @@ Json load_app_config(const std::string& yaml_path,
       std::string part = key.substr(
           s, sep == std::string::npos ? std::string::npos : sep - s);
       std::transform(part.begin(), part.end(), part.begin(), ::tolower);
+      // dashed keys (module names): _DASH_ spells "-" since env names
+      // cannot carry dashes (APP__MODULES__LLM_DASH_GATEWAY__...)
+      size_t dp;
+      while ((dp = part.find("_dash_")) != std::string::npos)
+        part = part.substr(0, dp) + "-" + part.substr(dp + 6);
       if (sep == std::string::npos) {
         // scalar inference like YAML
         Json v = yaml_parse(part + ": " + val).at(part);

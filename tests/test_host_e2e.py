@@ -1387,3 +1387,35 @@ def test_dp_fleet_survives_single_worker_crash(dp_server):
             break
         time.sleep(1)
     assert "hyperspot_workers_live 2" in body
+
+
+def test_config_layering_precedence(tmp_path):
+    """defaults -> YAML (with ${VAR} expansion) -> APP__ env -> CLI --set
+    (reference bootstrap/config/mod.rs:270-283)."""
+    cfg = tmp_path / "c.yaml"
+    cfg.write_text(
+        "server:\n"
+        "  home_dir: \"${HS_TEST_HOME}\"\n"
+        "modules:\n"
+        "  api-gateway:\n"
+        "    config:\n"
+        "      bind_addr: \"127.0.0.1:7001\"\n"
+        "      enable_docs: true\n")
+    env = dict(os.environ, HS_TEST_HOME="/tmp/layered-home",
+               APP__MODULES__LLM_DASH_GATEWAY__CONFIG__MODEL="env-model")
+    r = subprocess.run(
+        [str(BIN), "run", "--config", str(cfg), "--print-config",
+         "--set", "modules.api-gateway.config.bind_addr",
+         '"127.0.0.1:7002"'],
+        capture_output=True, text=True, env=env, timeout=60)
+    assert r.returncode == 0, r.stderr
+    doc = json.loads(r.stdout[r.stdout.index("{"):])
+    # ${VAR} expanded from the environment
+    assert doc["server"]["home_dir"] == "/tmp/layered-home"
+    # YAML survives where not overridden
+    assert doc["modules"]["api-gateway"]["config"]["enable_docs"] is True
+    # APP__ env overrode the default model
+    assert doc["modules"]["llm-gateway"]["config"]["model"] == "env-model"
+    # CLI --set wins over YAML
+    assert doc["modules"]["api-gateway"]["config"]["bind_addr"] \
+        == "127.0.0.1:7002"
