@@ -317,6 +317,12 @@ def main():
     stepper = threading.Thread(target=state.step_loop, daemon=True)
     stepper.start()
 
+    # graceful drain: SIGTERM (gateway stop ladder) unwinds through the
+    # finally block instead of killing mid-write
+    import signal
+    import sys as _sys
+    signal.signal(signal.SIGTERM, lambda *_: _sys.exit(0))
+
     try:
         os.unlink(args.uds)
     except FileNotFoundError:
