@@ -58,7 +58,6 @@ class ServerProc:
         import threading
         _build_binary()
         env = dict(os.environ, **(extra_env or {}))
-        env["APP__MODULES__API_DASH_GATEWAY"] = ""  # unused; kept simple
         self.port = port
         self.proc = subprocess.Popen(
             [str(BIN), "run", "--config", str(config_path),
