@@ -360,6 +360,14 @@ std::optional<std::string> WsConn::recv_text(int timeout_ms) {
         } else len = SIZE_MAX;
       }
       if (len != SIZE_MAX) {
+        // RFC 6455: clients MUST mask; also cap the declared payload
+        // before computing `need` so an attacker-controlled 64-bit
+        // length can neither wrap size_t nor balloon buf_
+        static const size_t kMaxFrame = 1 << 20;  // 1 MiB
+        if (!masked || len > kMaxFrame) {
+          send_close();
+          return std::nullopt;
+        }
         const size_t need = off + (masked ? 4 : 0) + len;
         if (buf_.size() >= need) {
           std::string payload = buf_.substr(off + (masked ? 4 : 0), len);

@@ -1,5 +1,8 @@
 #include "file_parser.h"
 
+#include <limits.h>
+#include <stdlib.h>
+
 #include <fstream>
 #include <sstream>
 
@@ -310,13 +313,31 @@ void FileParserModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
         throw Problem::bad_request("'path' is required");
       if (path.find("..") != std::string::npos)
         throw Problem::forbidden("path traversal rejected");
-      if (!allowed_roots_.empty()) {
-        bool ok = false;
-        for (auto& r : allowed_roots_)
-          if (path.rfind(r, 0) == 0) ok = true;
-        if (!ok) throw Problem::forbidden("path outside allowed roots");
+      // default-deny: parse-local is only usable when the operator has
+      // configured allowed_roots; both the root and the request path are
+      // canonicalised (symlinks resolved) and the containment check must
+      // end at a path-separator boundary ('/data' must not admit
+      // '/database-secrets')
+      if (allowed_roots_.empty())
+        throw Problem::forbidden(
+            "parse-local disabled: no allowed_roots configured");
+      char canon_buf[PATH_MAX];
+      if (realpath(path.c_str(), canon_buf) == nullptr)
+        throw Problem::not_found("cannot open " + path);
+      const std::string canon(canon_buf);
+      bool ok = false;
+      for (auto& r : allowed_roots_) {
+        char root_buf[PATH_MAX];
+        if (realpath(r.c_str(), root_buf) == nullptr) continue;
+        std::string root(root_buf);
+        if (canon == root ||
+            (canon.size() > root.size() &&
+             canon.compare(0, root.size(), root) == 0 &&
+             canon[root.size()] == '/'))
+          ok = true;
       }
-      std::ifstream f(path, std::ios::binary);
+      if (!ok) throw Problem::forbidden("path outside allowed roots");
+      std::ifstream f(canon, std::ios::binary);
       if (!f) throw Problem::not_found("cannot open " + path);
       std::stringstream ss;
       ss << f.rdbuf();

@@ -712,10 +712,18 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
     first_msg = connp->read_json(
         ttft_timeout_ms_ ? (int)ttft_timeout_ms_ : 120000);
     if (!first_msg) {
-      if (ttft_timeout_ms_)
+      if (ttft_timeout_ms_) {
+        // stop the generation server-side before reporting 504 — the
+        // blocking path does the same; otherwise the engine keeps
+        // producing until the 600 s stall guard notices
+        Json ab = Json::object();
+        ab["type"] = "abort";
+        ab["id"] = rid;
+        connp->send_json(ab);
         throw Problem{504, "Gateway Timeout", "about:blank",
                       "no first token within the TTFT budget",
                       "provider_timeout"};
+      }
       lease.w->ready = false;   // retry
     }
   }
@@ -746,6 +754,10 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
           total_timeout_ms_ ? (int)total_timeout_ms_ : 120000);
     }
     if (!msg) {
+      Json ab = Json::object();     // stop the server-side generation
+      ab["type"] = "abort";
+      ab["id"] = rid;
+      conn.send_json(ab);
       w.write_chunk("data: {\"error\":\"provider_timeout\"}\n\n");
       break;
     }

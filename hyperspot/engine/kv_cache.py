@@ -49,8 +49,8 @@ class BlockManager:
         # (commit_hashes after the prefill chunk) — never while a sibling
         # in the same step could read it half-written.
         self.enable_prefix_caching = enable_prefix_caching
-        self.hash_to_block: Dict[int, int] = {}     # committed blocks
-        self.block_hash: Dict[int, int] = {}
+        self.hash_to_block: Dict[bytes, int] = {}   # committed blocks
+        self.block_hash: Dict[int, bytes] = {}
         self.block_ref: Dict[int, int] = {}
         from collections import OrderedDict
         self.evictable: "OrderedDict[int, int]" = OrderedDict()  # blk->hash
@@ -59,13 +59,21 @@ class BlockManager:
         self.cache_queries = 0
 
     @staticmethod
-    def block_hashes(token_ids: List[int], block_size: int) -> List[int]:
-        """Chained content hashes, one per FULL block of the prompt."""
+    def block_hashes(token_ids: List[int], block_size: int) -> List[bytes]:
+        """Chained content hashes, one per FULL block of the prompt.
+
+        blake2b-128 over the chained token bytes: a collision would
+        silently serve another prompt's KV to the colliding request, so
+        a cryptographic hash (not Python's 64-bit hash()) keys the
+        shared-prefix pool."""
+        import hashlib
         out = []
-        h = 0
+        h = b""
         for i in range(len(token_ids) // block_size):
-            h = hash((h, tuple(token_ids[i * block_size:(i + 1)
-                                         * block_size])))
+            blk = token_ids[i * block_size:(i + 1) * block_size]
+            m = hashlib.blake2b(h, digest_size=16)
+            m.update(np.asarray(blk, dtype=np.int64).tobytes())
+            h = m.digest()
             out.append(h)
         return out
 
@@ -122,7 +130,13 @@ class BlockManager:
                 break
             reused.append(blk)
         need_total = self.blocks_needed(num_tokens)
-        if need_total - len(reused) > self.num_free:
+        # capacity check must not count the request's own reused blocks
+        # as free: reused blocks sitting in `evictable` are about to be
+        # pinned, so they leave the reclaimable pool exactly when the
+        # fresh blocks are taken — otherwise _take_block could raise
+        # mid-allocation after ref-counts were already mutated
+        reused_in_evictable = sum(1 for b in reused if b in self.evictable)
+        if (need_total - len(reused)) + reused_in_evictable > self.num_free:
             raise RuntimeError("KV pool exhausted")
         if not self._free_rows:
             raise RuntimeError("sequence table capacity exhausted")

@@ -138,6 +138,9 @@ modules:
       enable_docs: true
       cors_enabled: true
       auth_disabled: true
+  file-parser:
+    config:
+      allowed_roots: ["{tempfile.gettempdir()}"]
   llm-gateway:
     config:
       model: "tiny-llama"
@@ -659,6 +662,20 @@ def test_file_parser(server):
     st, _ = _http("POST", url + "/file-parser/v1/parse-local",
                   {"path": "/etc/../etc/passwd"})
     assert st == 403
+    # outside allowed_roots rejected (even without '..')
+    st, _ = _http("POST", url + "/file-parser/v1/parse-local",
+                  {"path": "/etc/hostname"})
+    assert st == 403
+    # a symlink inside the root pointing outside it is rejected
+    # (containment is checked on the canonicalised path)
+    link = tempfile.mktemp(suffix=".txt")
+    os.symlink("/etc/hostname", link)
+    try:
+        st, _ = _http("POST", url + "/file-parser/v1/parse-local",
+                      {"path": link})
+        assert st == 403
+    finally:
+        os.unlink(link)
     # multipart upload csv -> markdown table
     boundary = "XbOuNdArYx"
     mp = (f"--{boundary}\r\ncontent-disposition: form-data; "
