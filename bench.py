@@ -1,17 +1,31 @@
 #!/usr/bin/env python3
 """Flagship serving benchmark — BASELINE.json north-star metric.
 
-Measures steady-state decode throughput (tokens/s, whole-job aggregate) of
-the llm-gateway engine on Llama-3-8B bf16, synthetic prompts, random-init
-weights, plus p50 TTFT (prefill latency of one prompt).
+Default mode measures the metric AS STATED: **tokens/sec + p50 TTFT via
+llm-gateway REST** (`POST /v1/chat/completions`, SSE).  It builds/starts
+the real `hyperspot-server` host binary, which spawns the engine worker
+fleet (one worker per GPU, or one TP group with ``--tp N``), then drives
+concurrent streaming chat completions from the client side.
 
-Default (driver contract): one engine replica per GPU rank, TP=1 per
-replica (weak scaling — per-GPU work fixed as N grows; the idiomatic
-serving scale-out).  ``--tp N`` instead runs ONE engine tensor-parallel
-over all N ranks (strong scaling; used for the 70B TP=8 config).
+Timed region (driver contract): once all streams are admitted and in
+steady-state decode, each worker performs W untimed warmup engine steps,
+a `torch.cuda.synchronize()` (+ TP barrier), then EXACTLY K timed steps,
+then sync/barrier again — while the tokens keep streaming out through
+UDS → gateway → SSE to this client.  The reported `value` is the number
+of delta events *delivered over REST to the client* inside the worker's
+[t0, t1] wall-clock window (one SSE delta per generated token), i.e. the
+full path the metric names: engine + detokenize + UDS hop + gateway SSE
+serialization.  `engine_tokens_per_s` (engine-side count over the same
+window) is reported alongside so the gateway overhead is quantified.
+`ms_per_step` is MAX over workers of elapsed/K.
+
+``--engine-direct`` keeps the round-1 in-process engine loop (used for
+rocprof profiling, kernel work, and as the overhead cross-check).
 
 Launch (driver): python -m torch.distributed.run --nnodes=1
   --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+Ranks >0 only wait on a gloo barrier: the GPUs are owned by the worker
+processes the host spawns (REST serving is process-per-GPU by design).
 """
 
 from __future__ import annotations
@@ -19,45 +33,426 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import selectors
+import socket
 import statistics
+import subprocess
+import sys
+import threading
 import time
 
-import torch
-import torch.distributed as dist
+REPO = os.path.dirname(os.path.abspath(__file__))
 
 
-def log(rank, msg):
-    if rank == 0:
-        print(f"[bench] {msg}", flush=True)
+def log(msg):
+    print(f"[bench] {msg}", flush=True)
 
 
-def main():
-    ap = argparse.ArgumentParser()
-    ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=64)
-    ap.add_argument("--warmup", type=int, default=8)
-    ap.add_argument("--model", default="llama3-8b")
-    ap.add_argument("--tp", type=int, default=1,
-                    help=">1: one tensor-parallel engine over all ranks")
-    ap.add_argument("--batch", type=int, default=2048)
-    ap.add_argument("--prompt-len", type=int, default=512)
-    ap.add_argument("--eager", action="store_true")
-    ap.add_argument("--quant", default=None, choices=[None, "fp8"],
-                    help="fp8: e4m3fn weights + per-token act scales "
-                         "(separate config line, NOT the bf16 headline)")
-    ap.add_argument("--kv-dtype", default="bfloat16",
-                    choices=["bfloat16", "fp8"],
-                    help="paged KV cache dtype (fp8 halves attention HBM)")
-    ap.add_argument("--device", default=None)
-    args = ap.parse_args()
+# --------------------------------------------------------------- helpers
 
-    rank = int(os.environ.get("RANK", "0"))
-    world = int(os.environ.get("WORLD_SIZE", "1"))
-    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
-    on_gpu = torch.cuda.is_available() and args.device != "cpu"
-    if on_gpu:
-        torch.cuda.set_device(local_rank)
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
 
+
+def _raise_nofile(target=65535):
+    import resource
+    soft, hard = resource.getrlimit(resource.RLIMIT_NOFILE)
+    want = min(target, hard) if hard > 0 else target
+    if soft < want:
+        try:
+            resource.setrlimit(resource.RLIMIT_NOFILE, (want, hard))
+        except (ValueError, OSError):
+            pass
+
+
+def ensure_host_binary():
+    bin_path = os.path.join(REPO, "host", "build", "hyperspot-server")
+    if not os.path.exists(bin_path):
+        log("building host binary (make -C host)")
+        subprocess.run(["make", "-C", os.path.join(REPO, "host"),
+                        f"-j{os.cpu_count() or 8}"], check=True,
+                       stdout=subprocess.DEVNULL)
+    return bin_path
+
+
+def _http_json(method, url, body=None, timeout=30):
+    import urllib.request
+    data = json.dumps(body).encode() if body is not None else None
+    req = urllib.request.Request(url, data=data, method=method)
+    if data:
+        req.add_header("content-type", "application/json")
+    with urllib.request.urlopen(req, timeout=timeout) as r:
+        return r.status, json.loads(r.read())
+
+
+def _uds_request(path, obj, timeout=900):
+    """One JSON round-trip on a worker's control socket."""
+    s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    s.settimeout(timeout)
+    s.connect(path)
+    s.sendall((json.dumps(obj) + "\n").encode())
+    buf = b""
+    while b"\n" not in buf:
+        chunk = s.recv(65536)
+        if not chunk:
+            break
+        buf += chunk
+    s.close()
+    return json.loads(buf.split(b"\n")[0]) if buf else None
+
+
+# ------------------------------------------------------------ SSE client
+
+class LoadClient(threading.Thread):
+    """Single-threaded selector loop driving N concurrent SSE chat
+    streams.  Counts delivered delta events (one per token — the worker
+    emits a delta for every generated token) with wall-clock timestamps
+    so token delivery can be sliced to the workers' timed window."""
+
+    def __init__(self, port, body_bytes, n_streams):
+        super().__init__(daemon=True)
+        self.port = port
+        self.n = n_streams
+        req = (b"POST /v1/chat/completions HTTP/1.1\r\n"
+               b"host: 127.0.0.1\r\ncontent-type: application/json\r\n"
+               b"content-length: " + str(len(body_bytes)).encode() +
+               b"\r\n\r\n" + body_bytes)
+        self.request = req
+        self.records = []          # (epoch_t, n_delta_events)
+        self.streams_live = 0
+        self.streams_started = 0   # received >= 1 content delta
+        self.total_events = 0
+        self.errors = 0
+        self._stop = threading.Event()
+
+    def stop(self):
+        self._stop.set()
+
+    def run(self):
+        sel = selectors.DefaultSelector()
+        conns = {}                  # fd -> state dict
+        to_open = self.n
+        backoff_until = 0.0
+        while not self._stop.is_set():
+            # open in modest batches so the listen backlog (512) holds
+            now = time.time()
+            opened = 0
+            while to_open > 0 and opened < 64 and now >= backoff_until:
+                s = socket.socket()
+                s.setblocking(False)
+                try:
+                    s.connect(("127.0.0.1", self.port))
+                except BlockingIOError:
+                    pass
+                except OSError:
+                    s.close()
+                    backoff_until = now + 0.2
+                    break
+                st = {"sock": s, "phase": "send", "off": 0, "seen": False}
+                conns[s.fileno()] = st
+                sel.register(s, selectors.EVENT_WRITE, st)
+                to_open -= 1
+                opened += 1
+            for key, _ev in sel.select(timeout=0.05):
+                st = key.data
+                s = st["sock"]
+                if st["phase"] == "send":
+                    try:
+                        sent = s.send(self.request[st["off"]:])
+                    except (BlockingIOError, InterruptedError):
+                        continue
+                    except OSError:
+                        sel.unregister(s)
+                        s.close()
+                        conns.pop(key.fd, None)
+                        self.errors += 1
+                        to_open += 1
+                        continue
+                    st["off"] += sent
+                    if st["off"] >= len(self.request):
+                        st["phase"] = "read"
+                        sel.modify(s, selectors.EVENT_READ, st)
+                        self.streams_live += 1
+                    continue
+                try:
+                    data = s.recv(262144)
+                except (BlockingIOError, InterruptedError):
+                    continue
+                except OSError:
+                    data = b""
+                if not data:
+                    sel.unregister(s)
+                    s.close()
+                    conns.pop(key.fd, None)
+                    self.streams_live -= 1
+                    self.errors += 1
+                    continue
+                nev = data.count(b'"content"')
+                if nev:
+                    if not st["seen"]:
+                        st["seen"] = True
+                        self.streams_started += 1
+                    self.total_events += nev
+                    self.records.append((time.time(), nev))
+        for st in conns.values():
+            try:
+                st["sock"].close()
+            except OSError:
+                pass
+
+    def tokens_in_window(self, t0, t1):
+        return sum(n for (t, n) in self.records if t0 <= t <= t1)
+
+
+def measure_ttft(port, model, prompt_text, iters=10):
+    """Client-side TTFT: POST a streaming chat completion, time until
+    the first delta event that carries content."""
+    vals = []
+    body = json.dumps({
+        "model": model, "stream": True,
+        "messages": [{"role": "user",
+                      "content": [{"type": "text", "text": prompt_text}]}],
+        "max_tokens": 4, "temperature": 0.0,
+    }).encode()
+    req = (b"POST /v1/chat/completions HTTP/1.1\r\n"
+           b"host: 127.0.0.1\r\ncontent-type: application/json\r\n"
+           b"content-length: " + str(len(body)).encode() + b"\r\n\r\n" +
+           body)
+    for _ in range(iters):
+        s = socket.create_connection(("127.0.0.1", port), timeout=120)
+        t0 = time.monotonic()
+        s.sendall(req)
+        buf = b""
+        ttft = None
+        while True:
+            chunk = s.recv(65536)
+            if not chunk:
+                break
+            buf += chunk
+            if ttft is None and b'"content"' in buf:
+                ttft = (time.monotonic() - t0) * 1000
+            if b"[DONE]" in buf:
+                break
+        s.close()
+        if ttft is not None:
+            vals.append(ttft)
+    return statistics.median(vals) if vals else None
+
+
+# --------------------------------------------------------------- REST mode
+
+def write_config(args, port, sock_prefix, n_workers, on_gpu):
+    worker_lines = [
+        f"        max_num_seqs: {args.batch}",
+        # decode headroom: streams keep generating from admission until
+        # the client closes them (ramp + alignment + the timed window)
+        f"        max_model_len: {args.prompt_len + args.steps + args.warmup + 4096}",
+    ]
+    if args.tp > 1:
+        worker_lines.append(f"        tp: {args.tp}")
+    else:
+        worker_lines.append(f"        count: {n_workers}")
+    if args.eager or not on_gpu:
+        worker_lines.append("        eager: true")
+    if not on_gpu:
+        worker_lines.append('        device: "cpu"')
+        worker_lines.append("        num_gpu_blocks: 2048")
+    if args.quant:
+        worker_lines.append(f'        quant: "{args.quant}"')
+    if args.kv_dtype != "bfloat16":
+        worker_lines.append(f'        kv_dtype: "{args.kv_dtype}"')
+    cfg = f"""
+server:
+  home_dir: "/tmp/hs-bench"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+      defaults:
+        rate_limit:
+          rps: 100000000
+          burst: 100000000
+          in_flight: 1000000
+  serverless-runtime:
+    config:
+      limits:
+        max_concurrent_per_tenant: 1000000
+        rps_per_tenant: 100000000
+        burst_per_tenant: 100000000
+  llm-gateway:
+    config:
+      model: "{args.model}"
+      worker_socket: "{sock_prefix}"
+      auto_start_worker: true
+      python: "{sys.executable}"
+      worker:
+{chr(10).join(worker_lines)}
+"""
+    path = f"/tmp/hs-bench-{os.getpid()}.yaml"
+    with open(path, "w") as f:
+        f.write(cfg)
+    return path
+
+
+def wait_workers(port, n_workers, timeout=1200):
+    url = f"http://127.0.0.1:{port}/llm-gateway/v1/status"
+    t0 = time.time()
+    last = None
+    while time.time() - t0 < timeout:
+        try:
+            st, j = _http_json("GET", url, timeout=10)
+            last = j
+            ws = j.get("workers", [])
+            if len(ws) >= n_workers and all(
+                    w.get("ready") and "engine" in w for w in ws):
+                return j
+        except Exception:
+            pass
+        time.sleep(1.0)
+    raise TimeoutError(f"workers not ready after {timeout}s: {last}")
+
+
+def wait_saturated(port, per_worker, timeout=3600):
+    """All streams admitted and through prefill on every worker."""
+    url = f"http://127.0.0.1:{port}/llm-gateway/v1/status"
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        try:
+            _, j = _http_json("GET", url, timeout=10)
+            ws = j.get("workers", [])
+            if ws and all(
+                    w.get("engine", {}).get("num_running", 0)
+                    >= int(per_worker * 0.98)
+                    and w.get("engine", {}).get("num_waiting", 1) == 0
+                    for w in ws):
+                return j
+        except Exception:
+            pass
+        time.sleep(1.0)
+    raise TimeoutError("streams never reached steady-state decode")
+
+
+def rest_bench_rank0(args, n_gpus, on_gpu):
+    _raise_nofile()
+    binary = ensure_host_binary()
+    port = _free_port()
+    sock_prefix = f"/tmp/hs-bench-{os.getpid()}.sock"
+    n_workers = 1 if args.tp > 1 else n_gpus
+    cfg_path = write_config(args, port, sock_prefix, n_workers, on_gpu)
+    env = dict(os.environ)
+    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    logf = open(f"/tmp/hs-bench-{os.getpid()}.log", "wb")
+    host = subprocess.Popen([binary, "run", "--config", cfg_path],
+                            env=env, stdout=logf, stderr=logf, cwd=REPO)
+    sockets = ([sock_prefix] if n_workers == 1 else
+               [f"{sock_prefix}.{i}" for i in range(n_workers)])
+    try:
+        t0 = time.time()
+        wait_workers(port, n_workers)
+        log(f"host + {n_workers} worker(s) ready in {time.time()-t0:.0f}s "
+            f"(model={args.model}, tp={args.tp})")
+
+        # deterministic ASCII prompt of exactly prompt_len tokens
+        # (ByteTokenizer: BOS + 1 token/byte)
+        import random
+        rng = random.Random(7)
+        prompt_text = "".join(
+            chr(rng.randint(0x21, 0x7e)) for _ in range(args.prompt_len - 1))
+
+        ttft_p50 = measure_ttft(port, args.model, prompt_text,
+                                iters=args.ttft_iters)
+        log(f"REST TTFT p50 {ttft_p50:.1f} ms (prompt {args.prompt_len})")
+
+        # ---- sustained load: batch streams per worker ----
+        total_streams = args.batch * n_workers if args.tp == 1 else args.batch
+        body = json.dumps({
+            "model": args.model, "stream": True, "ignore_eos": True,
+            "max_tokens": 1000000000, "temperature": 0.0,
+            "messages": [{"role": "user",
+                          "content": [{"type": "text",
+                                       "text": prompt_text}]}],
+        }).encode()
+        client = LoadClient(port, body, total_streams)
+        client.start()
+        t0 = time.time()
+        wait_saturated(port, args.batch)
+        while (client.streams_started < total_streams * 0.99
+               and time.time() - t0 < 900):
+            time.sleep(0.5)
+        if client.streams_started < total_streams * 0.95:
+            raise RuntimeError(
+                f"only {client.streams_started}/{total_streams} streams "
+                f"came up ({client.errors} errors)")
+        log(f"{client.streams_started}/{total_streams} streams decoding "
+            f"({time.time()-t0:.0f}s ramp, {client.errors} conn errors)")
+
+        # ---- timed window: workers step W+K with sync brackets ----
+        start_at = time.time() + 1.0
+        results = []
+        threads = []
+        def run_bench(sp):
+            results.append(_uds_request(sp, {
+                "type": "bench", "warmup": args.warmup,
+                "steps": args.steps, "start_at": start_at}))
+        for sp in sockets:
+            th = threading.Thread(target=run_bench, args=(sp,))
+            th.start()
+            threads.append(th)
+        for th in threads:
+            th.join(timeout=1800)
+        ok = [r for r in results if r and r.get("event") == "bench_done"]
+        if len(ok) != len(sockets):
+            raise RuntimeError(f"bench op failed on some workers: {results}")
+        time.sleep(1.0)                     # let in-flight SSE drain
+        client.stop()
+
+        w_t0 = max(r["t0"] for r in ok)
+        w_t1 = min(r["t1"] for r in ok)
+        window = w_t1 - w_t0
+        delivered = client.tokens_in_window(w_t0, w_t1)
+        client_rate = delivered / window if window > 0 else 0.0
+        engine_rate = sum(r["produced"] / r["elapsed"] for r in ok)
+        ms_per_step = max(r["elapsed"] for r in ok) / args.steps * 1000
+        overhead = (1 - client_rate / engine_rate) * 100 if engine_rate else 0
+        return {
+            "value": round(client_rate, 2),
+            "ms_per_step": round(ms_per_step, 3),
+            "ttft_ms_p50": round(ttft_p50, 2) if ttft_p50 else None,
+            "engine_tokens_per_s": round(engine_rate, 2),
+            "gateway_overhead_pct": round(overhead, 2),
+            "global_batch": total_streams,
+            "window_s": round(window, 3),
+        }
+    finally:
+        host.send_signal(15)
+        try:
+            host.wait(timeout=30)
+        except subprocess.TimeoutExpired:
+            host.kill()
+        logf.close()
+        for p in [cfg_path]:
+            try:
+                os.unlink(p)
+            except OSError:
+                pass
+
+
+# --------------------------------------------------- engine-direct mode
+
+def engine_direct(args, rank, world, on_gpu):
+    """Round-1 in-process engine loop (profiling / kernel work).
+    Reports the engine-only number; the REST path is the headline."""
+    import torch
+    import torch.distributed as dist
     from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
     from hyperspot.parallel.state import initialize_model_parallel
 
@@ -78,18 +473,21 @@ def main():
         seed=1234 + (0 if tp > 1 else rank))
     t0 = time.monotonic()
     eng = LLMEngine(cfg, device=args.device)
-    log(rank, f"engine init {time.monotonic() - t0:.1f}s "
-              f"(model={args.model}, blocks={eng.runner.num_blocks})")
+    if rank == 0:
+        log(f"engine init {time.monotonic() - t0:.1f}s "
+            f"(model={args.model}, blocks={eng.runner.num_blocks})")
 
-    # ---- TTFT: single-request prefill latency (p50 of 5) ----
-    g = torch.Generator().manual_seed(7)
+    import torch as _t
+    g = _t.Generator().manual_seed(7)
     vocab = cfg.spec().vocab_size
+
     def mk_prompt(n):
-        return torch.randint(0, vocab, (n,), generator=g).tolist()
+        return _t.randint(0, vocab, (n,), generator=g).tolist()
+
     ttfts = []
     for _ in range(5):
-        rid = eng.add_request(mk_prompt(args.prompt_len),
-                              SamplingParams(temperature=0.0, max_tokens=1))
+        eng.add_request(mk_prompt(args.prompt_len),
+                        SamplingParams(temperature=0.0, max_tokens=1))
         t = time.monotonic()
         while eng.has_work():
             eng.step()
@@ -97,29 +495,22 @@ def main():
             torch.cuda.synchronize()
         ttfts.append((time.monotonic() - t) * 1000)
     ttft_p50 = statistics.median(ttfts)
-    log(rank, f"ttft p50 {ttft_p50:.1f} ms")
 
-    # ---- throughput: saturate with args.batch concurrent sequences ----
     sp = SamplingParams(temperature=0.0, max_tokens=10 ** 9)
-    for i in range(args.batch):
+    for _ in range(args.batch):
         eng.add_request(mk_prompt(args.prompt_len), sp)
-    t = time.monotonic()
-    while eng.num_waiting > 0:          # run all prefills
+    while eng.num_waiting > 0:
         eng.step()
     if on_gpu:
         torch.cuda.synchronize()
-    log(rank, f"prefill of {args.batch} x {args.prompt_len} done "
-              f"({time.monotonic() - t:.1f}s); capturing graphs")
     if not cfg.enforce_eager:
         eng.capture_graphs()
-
     for _ in range(args.warmup):
         eng.step()
     if on_gpu:
         torch.cuda.synchronize()
     if world > 1:
         dist.barrier()
-
     t0 = time.monotonic()
     produced = 0
     for _ in range(args.steps):
@@ -129,48 +520,112 @@ def main():
     if world > 1:
         dist.barrier()
     elapsed = time.monotonic() - t0
-
-    # max over ranks
     if world > 1:
-        e = torch.tensor([elapsed])
+        e = _t.tensor([elapsed])
         dist.all_reduce(e, op=dist.ReduceOp.MAX)
         elapsed = float(e[0])
-        p = torch.tensor([produced], dtype=torch.long)
+        p = _t.tensor([produced], dtype=_t.long)
         dist.all_reduce(p)
         produced = int(p[0])
-    else:
-        produced = produced
+    if world > 1:
+        dist.destroy_process_group()
+    return {
+        "value": round(produced / elapsed, 2),
+        "ms_per_step": round(elapsed / args.steps * 1000, 3),
+        "ttft_ms_p50": round(ttft_p50, 2),
+        "global_batch": args.batch * (1 if tp > 1 else world),
+    }
 
+
+# ------------------------------------------------------------------ main
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=64)
+    ap.add_argument("--warmup", type=int, default=8)
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--tp", type=int, default=1,
+                    help=">1: one tensor-parallel engine over all ranks")
+    ap.add_argument("--batch", type=int, default=2048,
+                    help="concurrent streams per engine replica")
+    ap.add_argument("--prompt-len", type=int, default=512)
+    ap.add_argument("--ttft-iters", type=int, default=10)
+    ap.add_argument("--eager", action="store_true")
+    ap.add_argument("--engine-direct", action="store_true",
+                    help="round-1 in-process engine loop (profiling)")
+    ap.add_argument("--quant", default=None, choices=[None, "fp8"],
+                    help="fp8 weights (separate evidence line, NOT the "
+                         "bf16 headline)")
+    ap.add_argument("--kv-dtype", default="bfloat16",
+                    choices=["bfloat16", "fp8"])
+    ap.add_argument("--device", default=None)
+    args = ap.parse_args()
+
+    import torch
+    import torch.distributed as dist
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    on_gpu = torch.cuda.is_available() and args.device != "cpu"
     n_gpus = world if world > 1 else args.gpus
-    value = produced / elapsed
-    ms_per_step = elapsed / args.steps * 1000
-    if rank == 0:
+
+    if not on_gpu and args.model == "llama3-8b" and not args.engine_direct:
+        # CPU dev-run: the 8B model at fp32 on CPU is not a dev loop
+        args.model = "tiny-llama"
+        args.batch = min(args.batch, 8)
+        args.prompt_len = min(args.prompt_len, 64)
+
+    if args.engine_direct:
+        res = engine_direct(args, rank, world, on_gpu)
+        metric = "tokens/sec (engine-direct decode, profiling mode)"
+        parallelism = f"tp{args.tp}" if args.tp > 1 else f"dp{n_gpus}"
+    else:
+        if world > 1:
+            from datetime import timedelta
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            dist.init_process_group("gloo",
+                                    timeout=timedelta(seconds=7200))
+        if rank == 0:
+            res = rest_bench_rank0(args, n_gpus, on_gpu)
+        else:
+            res = None
+        if world > 1:
+            dist.barrier()
+            dist.destroy_process_group()
+        metric = ("tokens/sec + p50 TTFT via llm-gateway REST "
+                  "/v1/chat/completions")
+        parallelism = (f"tp{args.tp}" if args.tp > 1
+                       else f"dp{n_gpus}(worker fleet)")
+
+    if rank == 0 and res is not None:
         out = {
-            "metric": "tokens/sec (llm-gateway decode throughput)",
-            "value": round(value, 2),
+            "metric": metric,
+            "value": res["value"],
             "unit": "tokens/s",
             "n_gpus": n_gpus,
             "steps": args.steps,
             "warmup": args.warmup,
-            "ms_per_step": round(ms_per_step, 3),
+            "ms_per_step": res["ms_per_step"],
             "higher_is_better": True,
-            "scaling": "strong" if tp > 1 else "weak",
+            "scaling": "strong" if args.tp > 1 else "weak",
             "vs_baseline": None,
             "dtype": (((args.quant or "bf16")
                        + ("+fp8kv" if args.kv_dtype == "fp8" else ""))
                       if on_gpu else "fp32(cpu-dev-run)"),
             "data": "synthetic",
-            "ttft_ms_p50": round(ttft_p50, 2),
+            "ttft_ms_p50": res.get("ttft_ms_p50"),
             "config": {
                 "model": args.model,
-                "global_batch": args.batch * (1 if tp > 1 else n_gpus),
+                "global_batch": res.get("global_batch"),
                 "seq_len": args.prompt_len,
-                "parallelism": (f"tp{tp}" if tp > 1 else f"dp{n_gpus}"),
+                "parallelism": parallelism,
             },
         }
+        for k in ("engine_tokens_per_s", "gateway_overhead_pct",
+                  "window_s"):
+            if k in res:
+                out[k] = res[k]
         print(json.dumps(out), flush=True)
-    if world > 1:
-        dist.destroy_process_group()
 
 
 if __name__ == "__main__":

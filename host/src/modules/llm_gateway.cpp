@@ -277,6 +277,11 @@ void LlmGatewayModule::spawn_one(Worker& wk) {
         args.push_back(std::to_string(
             worker_cfg_.at("num_gpu_blocks").as_int(0)));
       }
+      if (worker_cfg_.contains("max_model_len")) {
+        args.push_back("--max-model-len");
+        args.push_back(std::to_string(
+            worker_cfg_.at("max_model_len").as_int(8192)));
+      }
       if (worker_cfg_.contains("tp")) {
         args.push_back("--tp");
         args.push_back(std::to_string(worker_cfg_.at("tp").as_int(1)));
@@ -428,7 +433,7 @@ Json LlmGatewayModule::run_chat_blocking(const Json& body,
   Json params = body.at("params");
   if (params.is_null()) params = Json::object();
   for (const char* f : {"temperature", "top_p", "top_k", "max_tokens",
-                        "seed"})
+                        "seed", "ignore_eos"})
     if (body.contains(f)) params[f] = body.at(f);
   wreq["params"] = params;
 
@@ -687,7 +692,7 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   Json params = body.at("params");
   if (params.is_null()) params = Json::object();
   for (const char* f : {"temperature", "top_p", "top_k", "max_tokens",
-                        "seed"})
+                        "seed", "ignore_eos"})
     if (body.contains(f)) params[f] = body.at(f);
   wreq["params"] = params;
 

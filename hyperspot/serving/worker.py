@@ -47,6 +47,8 @@ class WorkerState:
         self.started = {}                     # rid -> prompt_len
         self.pending_ops = []                 # ops since last broadcast
         self.stop = False
+        self.bench_req = None                 # (warmup, steps, start_at)
+        self.bench_result = queue.Queue()
 
     def submit(self, rid, prompt_ids, sampling):
         q = queue.Queue()
@@ -74,25 +76,85 @@ class WorkerState:
     def step_loop(self):
         import torch.distributed as dist
         while not self.stop:
+            if self.bench_req is not None:
+                self._run_bench()
+                continue
             with self.new_work:
-                while not self.engine.has_work() and not self.stop                         and not self.pending_ops:
+                while not self.engine.has_work() and not self.stop                         and not self.pending_ops and self.bench_req is None:
                     self.new_work.wait(timeout=0.5)
                 if self.stop:
                     if self.tp > 1:
                         dist.broadcast_object_list([("stop",)], src=0)
                     return
+                if self.bench_req is not None:
+                    continue
                 if self.tp > 1:
                     ops = self.pending_ops
                     self.pending_ops = []
                     dist.broadcast_object_list([("step", ops)], src=0)
                 outputs = self.engine.step()
-            for out in outputs:
-                q = self.streams.get(out.request_id)
-                if q is not None:
-                    q.put(out)
-                    if out.finished:
-                        self.streams.pop(out.request_id, None)
-                        self.started.pop(out.request_id, None)
+            self._fanout(outputs)
+
+    def _fanout(self, outputs):
+        for out in outputs:
+            q = self.streams.get(out.request_id)
+            if q is not None:
+                q.put(out)
+                if out.finished:
+                    self.streams.pop(out.request_id, None)
+                    self.started.pop(out.request_id, None)
+
+    def _one_step(self):
+        import torch.distributed as dist
+        with self.new_work:
+            if self.tp > 1:
+                ops = self.pending_ops
+                self.pending_ops = []
+                dist.broadcast_object_list([("step", ops)], src=0)
+            outputs = self.engine.step()
+        self._fanout(outputs)
+        return len(outputs)
+
+    def _sync(self):
+        """Device barrier bracketing a timed region (all TP ranks)."""
+        import torch
+        import torch.distributed as dist
+        if self.tp > 1:
+            dist.broadcast_object_list([("bench_sync",)], src=0)
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        if self.tp > 1:
+            dist.barrier()
+
+    def _run_bench(self):
+        """Timed benchmark window driven by the serving bench client:
+        W untimed warmup steps, device sync + (TP) barrier, exactly K
+        timed steps, sync + barrier.  Token fan-out to the streaming
+        connections stays inside the timed region — the engine is being
+        measured *while serving*, not in isolation.  t0/t1 are epoch
+        seconds so the REST client (same box) can count delivered
+        tokens inside the same wall-clock window."""
+        warmup, steps, start_at = self.bench_req
+        self.bench_req = None
+        try:
+            while time.time() < start_at:       # align fleet windows
+                time.sleep(0.001)
+            for _ in range(warmup):
+                self._one_step()
+            self._sync()
+            t0 = time.time()
+            produced = 0
+            for _ in range(steps):
+                produced += self._one_step()
+            self._sync()
+            t1 = time.time()
+            self.bench_result.put({
+                "event": "bench_done", "t0": t0, "t1": t1,
+                "elapsed": t1 - t0, "produced": produced,
+                "steps": steps, "warmup": warmup})
+        except Exception as e:           # report instead of killing loop
+            self.bench_result.put({"event": "error",
+                                   "message": f"bench failed: {e}"})
 
 
 def follower_loop(engine):
@@ -105,6 +167,12 @@ def follower_loop(engine):
         msg = box[0]
         if msg[0] == "stop":
             return
+        if msg[0] == "bench_sync":
+            import torch
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            dist.barrier()
+            continue
         _, ops = msg
         for op in ops:
             if op[0] == "add":
@@ -143,6 +211,20 @@ def handle_conn(conn: socket.socket, state: WorkerState, model_name: str):
                       "kv_blocks_total": state.engine.runner.num_blocks})
             elif t == "abort":
                 state.abort(msg.get("id", ""))
+            elif t == "bench":
+                # operator-plane op: time W warmup + K steps around the
+                # live serving load (bench.py REST-mode contract)
+                w_ = int(msg.get("warmup", 8))
+                k_ = int(msg.get("steps", 64))
+                start_at = float(msg.get("start_at", 0.0))
+                with state.new_work:
+                    state.bench_req = (w_, k_, start_at)
+                    state.new_work.notify()
+                wait_s = max(0.0, start_at - time.time()) + 600
+                try:
+                    send(state.bench_result.get(timeout=wait_s))
+                except queue.Empty:
+                    send({"event": "error", "message": "bench timed out"})
             elif t == "chat":
                 _run_chat(msg, state, send)
             elif t == "embeddings":
@@ -222,7 +304,10 @@ def _run_chat(msg, state: WorkerState, send):
             return
         n_out += 1
         text = detok.push(out.token_id)
-        if text and not out.finished:
+        if not out.finished:
+            # exactly ONE delta event per generated token (text may be ""
+            # while a multi-byte UTF-8 sequence is incomplete) — REST
+            # clients can count delivered tokens by counting deltas
             if not send({"event": "delta", "id": rid, "text": text,
                          "token_id": out.token_id}):
                 state.abort(rid)
