@@ -62,14 +62,32 @@ def quant_fp8_rowwise(x: torch.Tensor):
 class _LinearCompute(nn.Module):
     """Shared weight-storage / GEMM path for the TP linear layers."""
 
-    def _finalize_weight(self, w: torch.Tensor) -> None:
+    def _finalize_weight(self, w: torch.Tensor, shard_dim: int = None,
+                         slices: list = None,
+                         full_shape: tuple = None) -> None:
+        """`slices` describes how this rank's shard maps into the FULL
+        tensor along `shard_dim`: [(full_start, length, shard_start)].
+        Checkpoint load/save (engine/checkpoint.py) uses it to shard a
+        full checkpoint on load and to reassemble the full tensor on
+        save — TP hot-swap without a per-rank checkpoint format."""
         self.quant = _QUANT_MODE
         if self.quant == "fp8":
             q, s = quantize_weight_fp8(w)
             self.weight = nn.Parameter(q, requires_grad=False)
             self.weight_scale = nn.Parameter(s, requires_grad=False)
+            if slices is not None:
+                if shard_dim == 0:
+                    # per-out-channel scales shard with the rows
+                    self.weight_scale._tp_shard_dim = 0
+                    self.weight_scale._tp_slices = slices
+                    self.weight_scale._tp_full_shape = (full_shape[0],)
+                # shard_dim == 1: scale is per full row — replicated
         else:
             self.weight = nn.Parameter(w, requires_grad=False)
+        if slices is not None:
+            self.weight._tp_shard_dim = shard_dim
+            self.weight._tp_slices = slices
+            self.weight._tp_full_shape = full_shape
 
     def _mm(self, x) -> torch.Tensor:
         from hyperspot import ops as _O   # local: avoids import cycle at init
@@ -112,7 +130,10 @@ class ColumnParallelLinear(_LinearCompute):
         self.in_features, self.out_features = in_features, out_features
         shard = out_features // tp
         full = _init_weight(out_features, in_features, dtype, seed_tag)
-        self._finalize_weight(full[r * shard:(r + 1) * shard].contiguous())
+        self._finalize_weight(full[r * shard:(r + 1) * shard].contiguous(),
+                              shard_dim=0,
+                              slices=[(r * shard, shard, 0)],
+                              full_shape=(out_features, in_features))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return self._mm(x)
@@ -134,12 +155,20 @@ class MergedColumnParallelLinear(_LinearCompute):
         tp, r = get_tp_size(), get_tp_rank()
         self.out_sizes = out_sizes
         parts = []
+        slices = []
+        off_full = 0
+        off_shard = 0
         for i, o in enumerate(out_sizes):
             assert o % tp == 0
             full = _init_weight(o, in_features, dtype, seed_tag + i)
             sh = o // tp
             parts.append(full[r * sh:(r + 1) * sh])
-        self._finalize_weight(torch.cat(parts, 0))
+            slices.append((off_full + r * sh, sh, off_shard))
+            off_full += o
+            off_shard += sh
+        self._finalize_weight(torch.cat(parts, 0), shard_dim=0,
+                              slices=slices,
+                              full_shape=(sum(out_sizes), in_features))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return self._mm(x)
@@ -158,9 +187,15 @@ class QKVParallelLinear(_LinearCompute):
         kf = _init_weight(num_kv_heads * head_dim, hidden, dtype, seed_tag + 1)
         vf = _init_weight(num_kv_heads * head_dim, hidden, dtype, seed_tag + 2)
         qs, ks = self.nh * head_dim, self.nkv * head_dim
-        self._finalize_weight(torch.cat([
-            qf[r * qs:(r + 1) * qs], kf[r * ks:(r + 1) * ks],
-            vf[r * ks:(r + 1) * ks]], 0))
+        nq, nk = num_heads * head_dim, num_kv_heads * head_dim
+        self._finalize_weight(
+            torch.cat([qf[r * qs:(r + 1) * qs], kf[r * ks:(r + 1) * ks],
+                       vf[r * ks:(r + 1) * ks]], 0),
+            shard_dim=0,
+            slices=[(r * qs, qs, 0),
+                    (nq + r * ks, ks, qs),
+                    (nq + nk + r * ks, ks, qs + ks)],
+            full_shape=(nq + 2 * nk, hidden))
 
     def forward(self, x: torch.Tensor):
         """Returns strided [T, n, D] views into ONE fused qkv buffer — no
@@ -184,7 +219,10 @@ class RowParallelLinear(_LinearCompute):
         self.in_features, self.out_features = in_features, out_features
         shard = in_features // tp
         full = _init_weight(out_features, in_features, dtype, seed_tag)
-        self._finalize_weight(full[:, r * shard:(r + 1) * shard].contiguous())
+        self._finalize_weight(
+            full[:, r * shard:(r + 1) * shard].contiguous(),
+            shard_dim=1, slices=[(r * shard, shard, 0)],
+            full_shape=(out_features, in_features))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         out = self._mm(x)

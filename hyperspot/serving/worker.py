@@ -49,6 +49,7 @@ class WorkerState:
         self.stop = False
         self.bench_req = None                 # (warmup, steps, start_at)
         self.bench_result = queue.Queue()
+        self.pending_exec = []                # (kind, arg, reply_q)
 
     def submit(self, rid, prompt_ids, sampling):
         q = queue.Queue()
@@ -73,6 +74,33 @@ class WorkerState:
                 self.pending_ops.append(("abort", rid))
                 self.new_work.notify()
 
+    def exec_collective(self, kind, arg, timeout=900):
+        """Run a model-wide operation (hot-swap / checkpoint save /
+        embeddings) at a step boundary.  At tp>1 these involve the whole
+        SPMD group (shard loads, all-gathers, TP forward), so they are
+        serialized through the same broadcast channel as steps — never
+        executed from a connection thread."""
+        if self.tp == 1:
+            with self.lock:
+                return self._exec(kind, arg)
+        q = queue.Queue()
+        with self.new_work:
+            self.pending_exec.append((kind, arg, q))
+            self.new_work.notify()
+        res = q.get(timeout=timeout)
+        if isinstance(res, Exception):
+            raise res
+        return res
+
+    def _exec(self, kind, arg):
+        if kind == "swap":
+            return self.engine.swap_weights(arg)
+        if kind == "save":
+            return self.engine.save_checkpoint(arg)
+        if kind == "embed":
+            return self.engine.embed(arg)
+        raise ValueError(kind)
+
     def step_loop(self):
         import torch.distributed as dist
         while not self.stop:
@@ -80,13 +108,23 @@ class WorkerState:
                 self._run_bench()
                 continue
             with self.new_work:
-                while not self.engine.has_work() and not self.stop                         and not self.pending_ops and self.bench_req is None:
+                while not self.engine.has_work() and not self.stop                         and not self.pending_ops and self.bench_req is None                         and not self.pending_exec:
                     self.new_work.wait(timeout=0.5)
                 if self.stop:
                     if self.tp > 1:
                         dist.broadcast_object_list([("stop",)], src=0)
                     return
                 if self.bench_req is not None:
+                    continue
+                while self.pending_exec:
+                    kind, arg, q = self.pending_exec.pop(0)
+                    dist.broadcast_object_list([("exec", kind, arg)],
+                                               src=0)
+                    try:
+                        q.put(self._exec(kind, arg))
+                    except Exception as e:
+                        q.put(e)
+                if not self.engine.has_work() and not self.pending_ops:
                     continue
                 if self.tp > 1:
                     ops = self.pending_ops
@@ -173,6 +211,19 @@ def follower_loop(engine):
                 torch.cuda.synchronize()
             dist.barrier()
             continue
+        if msg[0] == "exec":
+            # collective op in lockstep with rank 0 (swap/save/embed)
+            _, kind, arg = msg
+            try:
+                if kind == "swap":
+                    engine.swap_weights(arg)
+                elif kind == "save":
+                    engine.save_checkpoint(arg)
+                elif kind == "embed":
+                    engine.embed(arg)
+            except Exception:
+                log.exception("follower exec %s failed", kind)
+            continue
         _, ops = msg
         for op in ops:
             if op[0] == "add":
@@ -228,28 +279,18 @@ def handle_conn(conn: socket.socket, state: WorkerState, model_name: str):
             elif t == "chat":
                 _run_chat(msg, state, send)
             elif t == "embeddings":
-                if state.tp > 1:
-                    send({"event": "error", "message":
-                          "embeddings with tp>1 not supported yet"})
-                    continue
                 _run_embeddings(msg, state, send)
             elif t == "swap":
                 path = msg.get("checkpoint", "")
-                if state.tp > 1:
-                    send({"event": "error", "message":
-                          "hot-swap with tp>1 not supported yet"})
-                    continue
                 try:
-                    with state.lock:
-                        secs = state.engine.swap_weights(path)
+                    secs = state.exec_collective("swap", path)
                     send({"event": "swapped", "seconds": secs,
                           "checkpoint": path})
                 except Exception as e:
                     send({"event": "error", "message": f"swap failed: {e}"})
             elif t == "save_checkpoint":
                 try:
-                    with state.lock:
-                        state.engine.save_checkpoint(msg.get("path", ""))
+                    state.exec_collective("save", msg.get("path", ""))
                     send({"event": "saved", "path": msg.get("path", "")})
                 except Exception as e:
                     send({"event": "error", "message": f"save failed: {e}"})
@@ -334,8 +375,7 @@ def _run_embeddings(msg, state: WorkerState, send):
         return
     prompts = [state.tokenizer.encode(str(t), add_bos=True) for t in inputs]
     try:
-        with state.lock:
-            vecs = state.engine.embed(prompts)
+        vecs = state.exec_collective("embed", prompts)
     except Exception as e:
         send({"event": "error", "message": f"embed failed: {e}"})
         return
