@@ -1,6 +1,7 @@
 #include "db.h"
 
 #include <cstring>
+#include <functional>
 #include <stdexcept>
 
 namespace hs {
@@ -200,31 +201,75 @@ SecureConn::Page SecureConn::select(
     const std::string& table, const std::string& extra_where,
     std::vector<DbValue> binds0, const std::string& order_by, bool desc,
     int limit, const std::optional<std::string>& cursor) {
+  OrderBy ob;
+  if (!order_by.empty()) ob.emplace_back(order_by, desc);
+  return select(table, extra_where, std::move(binds0), ob, limit, cursor);
+}
+
+SecureConn::Page SecureConn::select(
+    const std::string& table, const std::string& extra_where,
+    std::vector<DbValue> binds0, const OrderBy& order_by, int limit,
+    const std::optional<std::string>& cursor) {
   std::vector<DbValue> binds;
   std::string where = scope_sql(binds);
   for (auto& b : binds0) binds.push_back(b);
   if (!extra_where.empty()) where += " AND (" + extra_where + ")";
-  const std::string ord = order_by.empty() ? "rowid" : order_by;
-  const char* cmp = desc ? "<" : ">";
+
+  // deterministic total order: user keys then rowid tie-break (rowid
+  // direction follows the first key so a single-key page reads like v1)
+  OrderBy keys = order_by;
+  const bool rdesc = keys.empty() ? false : keys[0].second;
+  keys.emplace_back("rowid", rdesc);
+
   if (cursor && !cursor->empty()) {
+    // CursorV2: "v2|<len>:<val>...<len>:<rowid>" base64url — one value
+    // per order key + the rowid (modkit-odata CursorV1 semantics:
+    // opaque, keyset, order-bound — lib.rs:353)
     auto raw = cur_decode(*cursor);
-    if (!raw) throw std::runtime_error("bad cursor");
-    // v1|<ord value>|<rowid>
-    size_t p1 = raw->find('|'), p2 = raw->rfind('|');
-    if (p1 == std::string::npos || p2 <= p1 ||
-        raw->substr(0, p1) != "v1")
+    if (!raw || raw->compare(0, 3, "v2|") != 0)
       throw std::runtime_error("bad cursor");
-    const std::string ov = raw->substr(p1 + 1, p2 - p1 - 1);
-    const std::string rid = raw->substr(p2 + 1);
-    where += " AND (" + ord + " " + cmp + " ?";
-    binds.push_back(DbValue::S(ov));
-    where += " OR (" + ord + " = ? AND rowid " + cmp + " ?))";
-    binds.push_back(DbValue::S(ov));
-    binds.push_back(DbValue::I(atoll(rid.c_str())));
+    std::vector<std::string> vals;
+    size_t i = 3;
+    while (i < raw->size()) {
+      size_t colon = raw->find(':', i);
+      if (colon == std::string::npos) throw std::runtime_error("bad cursor");
+      long len = atol(raw->substr(i, colon - i).c_str());
+      if (len < 0 || colon + 1 + (size_t)len > raw->size())
+        throw std::runtime_error("bad cursor");
+      vals.push_back(raw->substr(colon + 1, (size_t)len));
+      i = colon + 1 + (size_t)len;
+    }
+    if (vals.size() != keys.size())
+      throw std::runtime_error("cursor does not match $orderby");
+    // lexicographic keyset predicate with per-key direction:
+    // k0 cmp0 v0 OR (k0 = v0 AND (k1 cmp1 v1 OR (... rowid cmpN vN)))
+    std::function<std::string(size_t)> pred = [&](size_t k) {
+      const auto& [col, d] = keys[k];
+      const char* cmp = d ? "<" : ">";
+      DbValue v = col == "rowid"
+                      ? DbValue::I(atoll(vals[k].c_str()))
+                      : DbValue::S(vals[k]);
+      if (k + 1 == keys.size()) {
+        binds.push_back(v);
+        return col + " " + cmp + " ?";
+      }
+      std::string out = col + " " + cmp + " ?";
+      binds.push_back(v);
+      out += " OR (" + col + " = ? AND (";
+      binds.push_back(v);
+      out += pred(k + 1) + "))";
+      return out;
+    };
+    where += " AND (" + pred(0) + ")";
+  }
+
+  std::string ord_sql;
+  for (auto& [col, d] : keys) {
+    if (!ord_sql.empty()) ord_sql += ", ";
+    ord_sql += col + (d ? " DESC" : "");
   }
   std::string sql = "SELECT rowid AS _rid, * FROM " + table + " WHERE " +
-                    where + " ORDER BY " + ord + (desc ? " DESC" : "") +
-                    ", rowid" + (desc ? " DESC" : "") + " LIMIT " +
+                    where + " ORDER BY " + ord_sql + " LIMIT " +
                     std::to_string(limit + 1);
   std::lock_guard<std::mutex> lk(db_.mu());
   auto rows = db_.query(sql, binds);
@@ -233,15 +278,20 @@ SecureConn::Page SecureConn::select(
   if (more) rows.resize(limit);
   if (more && !rows.empty()) {
     const DbRow& last = rows.back();
-    std::string ov;
-    auto it = last.find(ord);
-    if (it != last.end()) {
-      ov = it->second.is_string() ? it->second.as_string()
-                                  : it->second.dump();
+    std::string payload = "v2|";
+    for (auto& [col, d] : keys) {
+      std::string ov;
+      if (col == "rowid") {
+        ov = std::to_string((long long)last.at("_rid").as_int(0));
+      } else {
+        auto it = last.find(col);
+        if (it != last.end())
+          ov = it->second.is_string() ? it->second.as_string()
+                                      : it->second.dump();
+      }
+      payload += std::to_string(ov.size()) + ":" + ov;
     }
-    long long rid = (long long)last.at("_rid").as_int(0);
-    page.next_cursor =
-        cur_encode("v1|" + ov + "|" + std::to_string(rid));
+    page.next_cursor = cur_encode(payload);
   }
   for (auto& r : rows) r.erase("_rid");
   page.items = std::move(rows);
@@ -333,6 +383,13 @@ static std::string take_token(const std::string& s, size_t& i) {
 std::string compile_odata_filter(const std::string& filter,
                                  const std::vector<std::string>& fields,
                                  std::vector<DbValue>& binds) {
+  // Recursive-descent $filter compiler (modkit-odata Expr AST parity,
+  // /root/reference/libs/modkit-odata/src/lib.rs:23,:70):
+  //   expr    := and_expr ('or' and_expr)*
+  //   and_expr:= unary ('and' unary)*
+  //   unary   := 'not' unary | '(' expr ')' | primary
+  //   primary := contains|startswith|endswith '(' field ',' lit ')'
+  //            | field (eq|ne|gt|ge|lt|le) literal
   auto allowed = [&](const std::string& f) {
     for (auto& x : fields)
       if (x == f) return true;
@@ -352,64 +409,175 @@ std::string compile_odata_filter(const std::string& filter,
       return DbValue::S(v);
     }
     size_t start = i;
-    while (i < s.size() && s[i] != ' ' && s[i] != ')') ++i;
+    while (i < s.size() && s[i] != ' ' && s[i] != ')' && s[i] != ',') ++i;
     std::string v = s.substr(start, i - start);
+    if (v.empty()) throw std::runtime_error("bad literal");
     if (v == "true") return DbValue::I(1);
     if (v == "false") return DbValue::I(0);
     if (v == "null") return DbValue::null();
     return DbValue::R(atof(v.c_str()));
   };
 
-  std::string sql;
-  size_t i = 0;
-  while (i < filter.size()) {
+  // recursion depth bound: adversarial '((((...' must not overflow the
+  // stack (the reference fuzzes exactly this surface, fuzz_odata_filter)
+  int depth = 0;
+  std::function<std::string(size_t&)> parse_or;
+  auto parse_func = [&](size_t& i, const char* name, size_t nlen,
+                        const char* pre, const char* post) -> std::string {
+    i += nlen;
+    std::string f = take_token(filter, i);
+    allowed(f);
     skip_ws(filter, i);
-    if (i >= filter.size()) break;
-    if (!sql.empty()) {
-      std::string conj = take_token(filter, i);
-      if (conj != "and")
-        throw std::runtime_error("only 'and' is supported in $filter");
-      sql += " AND ";
+    if (i >= filter.size() || filter[i] != ',')
+      throw std::runtime_error(std::string("bad ") + name + "()");
+    ++i;
+    DbValue v = parse_literal(filter, i);
+    skip_ws(filter, i);
+    if (i >= filter.size() || filter[i] != ')')
+      throw std::runtime_error(std::string("bad ") + name + "()");
+    ++i;
+    // escape LIKE wildcards in the needle so user data matches literally
+    std::string needle;
+    for (char c : v.text) {
+      if (c == '%' || c == '_' || c == '\\') needle += '\\';
+      needle += c;
     }
+    binds.push_back(DbValue::S(pre + needle + post));
+    return f + " LIKE ? ESCAPE '\\'";
+  };
+  std::function<std::string(size_t&)> parse_unary =
+      [&](size_t& i) -> std::string {
     skip_ws(filter, i);
-    if (filter.compare(i, 9, "contains(") == 0) {
-      i += 9;
-      std::string f = take_token(filter, i);
-      allowed(f);
-      skip_ws(filter, i);
-      if (i >= filter.size() || filter[i] != ',')
-        throw std::runtime_error("bad contains()");
+    if (++depth > 32) throw std::runtime_error("$filter too deeply nested");
+    std::string out;
+    if (i < filter.size() && filter[i] == '(') {
       ++i;
-      DbValue v = parse_literal(filter, i);
+      out = "(" + parse_or(i) + ")";
       skip_ws(filter, i);
       if (i >= filter.size() || filter[i] != ')')
-        throw std::runtime_error("bad contains()");
+        throw std::runtime_error("unbalanced parentheses");
       ++i;
-      sql += f + " LIKE ?";
-      binds.push_back(DbValue::S("%" + v.text + "%"));
-      continue;
-    }
-    std::string f = take_token(filter, i);
-    if (f.empty()) throw std::runtime_error("bad $filter");
-    allowed(f);
-    std::string op = take_token(filter, i);
-    const char* sqlop = nullptr;
-    if (op == "eq") sqlop = "=";
-    else if (op == "ne") sqlop = "!=";
-    else if (op == "gt") sqlop = ">";
-    else if (op == "ge") sqlop = ">=";
-    else if (op == "lt") sqlop = "<";
-    else if (op == "le") sqlop = "<=";
-    else throw std::runtime_error("bad operator: " + op);
-    DbValue v = parse_literal(filter, i);
-    if (v.kind == DbValue::kNull) {
-      sql += f + (op == "eq" ? " IS NULL" : " IS NOT NULL");
+    } else if (filter.compare(i, 4, "not ") == 0 ||
+               filter.compare(i, 4, "not(") == 0) {
+      i += 3;
+      out = "NOT (" + parse_unary(i) + ")";
+    } else if (filter.compare(i, 9, "contains(") == 0) {
+      out = parse_func(i, "contains", 9, "%", "%");
+    } else if (filter.compare(i, 11, "startswith(") == 0) {
+      out = parse_func(i, "startswith", 11, "", "%");
+    } else if (filter.compare(i, 9, "endswith(") == 0) {
+      out = parse_func(i, "endswith", 9, "%", "");
     } else {
-      sql += f + std::string(sqlop) + "?";
-      binds.push_back(v);
+      std::string f = take_token(filter, i);
+      if (f.empty()) throw std::runtime_error("bad $filter");
+      allowed(f);
+      std::string op = take_token(filter, i);
+      const char* sqlop = nullptr;
+      if (op == "eq") sqlop = "=";
+      else if (op == "ne") sqlop = "!=";
+      else if (op == "gt") sqlop = ">";
+      else if (op == "ge") sqlop = ">=";
+      else if (op == "lt") sqlop = "<";
+      else if (op == "le") sqlop = "<=";
+      else throw std::runtime_error("bad operator: " + op);
+      DbValue v = parse_literal(filter, i);
+      if (v.kind == DbValue::kNull) {
+        if (op != "eq" && op != "ne")
+          throw std::runtime_error("null only supports eq/ne");
+        out = f + (op == "eq" ? " IS NULL" : " IS NOT NULL");
+      } else {
+        out = f + std::string(sqlop) + "?";
+        binds.push_back(v);
+      }
     }
+    --depth;
+    return out;
+  };
+  auto parse_and = [&](size_t& i) -> std::string {
+    std::string out = parse_unary(i);
+    while (true) {
+      size_t save = i;
+      skip_ws(filter, i);
+      size_t t0 = i;
+      std::string tok = take_token(filter, i);
+      if (tok == "and") {
+        out += " AND " + parse_unary(i);
+      } else {
+        i = (tok.empty() ? i : t0);
+        if (tok.empty()) i = save;
+        break;
+      }
+    }
+    return out;
+  };
+  parse_or = [&](size_t& i) -> std::string {
+    std::string out = parse_and(i);
+    while (true) {
+      size_t save = i;
+      skip_ws(filter, i);
+      size_t t0 = i;
+      std::string tok = take_token(filter, i);
+      if (tok == "or") {
+        out += " OR " + parse_and(i);
+      } else {
+        i = (tok.empty() ? i : t0);
+        if (tok.empty()) i = save;
+        break;
+      }
+    }
+    return out;
+  };
+
+  size_t i = 0;
+  skip_ws(filter, i);
+  if (i >= filter.size()) return "1=1";
+  std::string sql = parse_or(i);
+  skip_ws(filter, i);
+  if (i < filter.size())
+    throw std::runtime_error("trailing input in $filter at offset " +
+                             std::to_string(i));
+  return sql;
+}
+
+std::vector<std::pair<std::string, bool>> parse_odata_orderby(
+    const std::string& orderby, const std::vector<std::string>& fields) {
+  // "$orderby=a desc,b asc" (OData) and the reference's signed-token
+  // encoding "-a,b" (modkit-odata lib.rs:135) both parse to the same
+  // (field, desc) list; fields are allow-listed like $filter fields.
+  auto allowed = [&](const std::string& f) {
+    for (auto& x : fields)
+      if (x == f) return true;
+    throw std::runtime_error("field not orderable: " + f);
+  };
+  std::vector<std::pair<std::string, bool>> out;
+  size_t i = 0;
+  while (i < orderby.size()) {
+    skip_ws(orderby, i);
+    bool desc = false;
+    if (i < orderby.size() && (orderby[i] == '-' || orderby[i] == '+')) {
+      desc = orderby[i] == '-';
+      ++i;
+    }
+    std::string f = take_token(orderby, i);
+    if (f.empty()) throw std::runtime_error("bad $orderby");
+    allowed(f);
+    skip_ws(orderby, i);
+    size_t save = i;
+    std::string dir = take_token(orderby, i);
+    if (dir == "desc") desc = true;
+    else if (dir == "asc") desc = false;
+    else i = save;
+    out.emplace_back(f, desc);
+    skip_ws(orderby, i);
+    if (i < orderby.size()) {
+      if (orderby[i] != ',')
+        throw std::runtime_error("bad $orderby separator");
+      ++i;
+    }
+    if (out.size() > 8) throw std::runtime_error("$orderby too long");
   }
-  return sql.empty() ? "1=1" : sql;
+  if (out.empty()) throw std::runtime_error("bad $orderby");
+  return out;
 }
 
 }  // namespace hs

@@ -123,12 +123,19 @@ class SecureConn {
     std::optional<std::string> next_cursor;
   };
 
-  // SELECT with optional extra WHERE, deterministic (order_by, rowid)
+  // multi-field signed ordering: (field, desc) list (modkit-odata
+  // ODataOrderBy signed tokens, lib.rs:135)
+  using OrderBy = std::vector<std::pair<std::string, bool>>;
+
+  // SELECT with optional extra WHERE, deterministic (order keys, rowid)
   // keyset pagination and an opaque base64 cursor (modkit-odata CursorV1).
   Page select(const std::string& table, const std::string& extra_where,
               std::vector<DbValue> binds, const std::string& order_by,
               bool desc, int limit,
               const std::optional<std::string>& cursor);
+  Page select(const std::string& table, const std::string& extra_where,
+              std::vector<DbValue> binds, const OrderBy& order_by,
+              int limit, const std::optional<std::string>& cursor);
 
   // INSERT: the tenant column is forced to the scope's single tenant.
   void insert(const std::string& table,
@@ -150,12 +157,18 @@ class SecureConn {
 };
 
 // Tiny OData-style $filter compiler: supports
-//   <field> eq|ne|gt|ge|lt|le <literal>, contains(field,'s'),
-//   and-combinations.  Fields are validated against an allow-list
-//   (modkit-odata x-odata-filter allowedFields).
+//   <field> eq|ne|gt|ge|lt|le <literal>, contains/startswith/endswith,
+//   and/or/not with parenthesised grouping (modkit-odata Expr AST,
+//   /root/reference/libs/modkit-odata/src/lib.rs:23,:70).  Fields are
+//   validated against an allow-list (x-odata-filter allowedFields).
 // Returns SQL + binds; throws std::runtime_error on a bad filter.
 std::string compile_odata_filter(const std::string& filter,
                                  const std::vector<std::string>& fields,
                                  std::vector<DbValue>& binds);
+
+// "$orderby=a desc,b" / signed tokens "-a,b" → (field, desc) list,
+// allow-listed; throws on unknown fields or malformed input.
+std::vector<std::pair<std::string, bool>> parse_odata_orderby(
+    const std::string& orderby, const std::vector<std::string>& fields);
 
 }  // namespace hs

@@ -136,11 +136,146 @@ std::optional<Json> EngineConn::read_json(int timeout_ms) {
     struct pollfd pf{fd_, POLLIN, 0};
     int pr = poll(&pf, 1, timeout_ms);
     if (pr <= 0) return std::nullopt;
-    char tmp[8192];
+    char tmp[65536];
     ssize_t r = recv(fd_, tmp, sizeof tmp, 0);
-    if (r <= 0) return std::nullopt;
+    if (r <= 0) {
+      eof_ = true;
+      return std::nullopt;
+    }
     buf_.append(tmp, size_t(r));
   }
+}
+
+// ------------------------------------------------------------- MuxClient
+
+MuxClient::MuxClient(const std::string& socket_path) {
+  conn_ = std::make_unique<EngineConn>(socket_path);
+  if (!conn_->ok()) { closed_ = true; return; }
+  Json a = Json::object();
+  a["type"] = "attach_mux";
+  if (!conn_->send_json(a)) { closed_ = true; return; }
+  auto r = conn_->read_json(10000);
+  if (!r || r->at("event").as_string() != "mux_attached") {
+    closed_ = true;
+    return;
+  }
+  attached_ = true;
+  reader_ = std::thread([this] { reader_loop(); });
+}
+
+MuxClient::~MuxClient() {
+  stop_ = true;
+  closed_ = true;
+  conn_.reset();                 // closes the fd; reader unblocks
+  if (reader_.joinable()) reader_.join();
+  fail_all();
+}
+
+void MuxClient::fail_all() {
+  std::lock_guard<std::mutex> lk(mu_);
+  for (auto& [rid, s] : sinks_) {
+    std::lock_guard<std::mutex> sl(s->mu);
+    s->dead = true;
+    s->cv.notify_all();
+  }
+  sinks_.clear();
+}
+
+void MuxClient::reader_loop() {
+  while (!stop_) {
+    auto msg = conn_->read_json(1000);
+    if (!msg) {
+      if (stop_ || conn_->eof()) break;
+      continue;                  // poll timeout, keep waiting
+    }
+    const std::string ev = msg->at("event").as_string();
+    if (ev == "batch") {
+      const Json& items = msg->at("items");
+      std::lock_guard<std::mutex> lk(mu_);
+      for (auto& it : items.arr()) {
+        // [rid, token_id, text] delta | [rid, tok, text, fin, in, out]
+        const std::string rid = it.at(0).as_string();
+        auto f = sinks_.find(rid);
+        if (f == sinks_.end()) continue;       // aborted meanwhile
+        auto& s = *f->second;
+        const bool fin = it.size() > 3;
+        {
+          std::lock_guard<std::mutex> sl(s.mu);
+          Json d = Json::object();
+          d["event"] = "delta";
+          d["token_id"] = it.at(1);
+          d["text"] = it.at(2);
+          s.q.push_back(std::move(d));
+          if (fin) {
+            Json u = Json::object();
+            u["input_tokens"] = it.at(4);
+            u["output_tokens"] = it.at(5);
+            Json dn = Json::object();
+            dn["event"] = "done";
+            dn["finish_reason"] = it.at(3);
+            dn["usage"] = u;
+            s.q.push_back(std::move(dn));
+          }
+          s.cv.notify_all();
+        }
+        if (fin) sinks_.erase(f);
+      }
+    } else if (ev == "error") {
+      const std::string rid = msg->at("id").as_string();
+      std::lock_guard<std::mutex> lk(mu_);
+      auto f = sinks_.find(rid);
+      if (f != sinks_.end()) {
+        std::lock_guard<std::mutex> sl(f->second->mu);
+        f->second->q.push_back(*msg);
+        f->second->cv.notify_all();
+        sinks_.erase(f);
+      }
+    }
+  }
+  closed_ = true;
+  fail_all();
+}
+
+std::shared_ptr<MuxSink> MuxClient::submit(const Json& wreq) {
+  const std::string rid = wreq.at("id").as_string();
+  auto sink = std::make_shared<MuxSink>();
+  {
+    std::lock_guard<std::mutex> lk(mu_);
+    if (closed_) return nullptr;
+    sinks_[rid] = sink;
+    if (!conn_->send_json(wreq)) {
+      sinks_.erase(rid);
+      closed_ = true;
+      return nullptr;
+    }
+  }
+  return sink;
+}
+
+void MuxClient::abort(const std::string& rid) {
+  std::lock_guard<std::mutex> lk(mu_);
+  sinks_.erase(rid);
+  if (closed_) return;
+  Json ab = Json::object();
+  ab["type"] = "abort";
+  ab["id"] = rid;
+  conn_->send_json(ab);
+}
+
+void MuxClient::remove(const std::string& rid) {
+  std::lock_guard<std::mutex> lk(mu_);
+  sinks_.erase(rid);
+}
+
+std::optional<Json> MuxClient::next_event(MuxSink& s, int timeout_ms) {
+  std::unique_lock<std::mutex> lk(s.mu);
+  if (!s.cv.wait_for(lk, std::chrono::milliseconds(timeout_ms),
+                     [&] { return !s.q.empty() || s.dead; }))
+    return std::nullopt;                      // timeout
+  if (s.q.empty()) return std::nullopt;       // dead
+  Json out = std::move(s.q.front());
+  s.q.pop_front();
+  return out;
 }
 
 // ---------------------------------------------------------- LlmGateway
@@ -235,6 +370,15 @@ LlmGatewayModule::Worker* LlmGatewayModule::pick_worker() {
   }
   if (best) best->in_flight++;
   return best;
+}
+
+std::shared_ptr<MuxClient> LlmGatewayModule::ensure_mux(Worker& wk) {
+  std::lock_guard<std::mutex> lk(wk.mux_mu);
+  if (wk.mux && wk.mux->ok()) return wk.mux;
+  auto m = std::make_shared<MuxClient>(wk.socket);
+  if (!m->ok()) return nullptr;
+  wk.mux = m;
+  return m;
 }
 
 void LlmGatewayModule::spawn_one(Worker& wk) {
@@ -437,17 +581,21 @@ Json LlmGatewayModule::run_chat_blocking(const Json& body,
     if (body.contains(f)) params[f] = body.at(f);
   wreq["params"] = params;
 
-  // a connect can land in a DYING worker's listen backlog (SIGKILL
-  // teardown window) and then EOF — before any token arrived the request
-  // is trivially retryable on another worker
+  // a submit can race a DYING worker (SIGKILL teardown window): before
+  // any token arrived the request is trivially retryable on another
+  // worker (the sink goes dead, never delivers)
   for (int attempt = 0; attempt < 3; ++attempt) {
-    std::unique_ptr<EngineConn> connp;
-    Lease lease{pick_live(connp)};
+    Lease lease{pick_worker()};
     if (!lease.w)
       throw Problem{503, "Service Unavailable", "about:blank",
                     "no engine worker ready", "provider_error"};
-    EngineConn& conn = *connp;
-    if (!conn.send_json(wreq)) {
+    auto mux = ensure_mux(*lease.w);
+    if (!mux) {
+      lease.w->ready = false;
+      continue;
+    }
+    auto sink = mux->submit(wreq);
+    if (!sink) {
       lease.w->ready = false;
       continue;
     }
@@ -459,21 +607,26 @@ Json LlmGatewayModule::run_chat_blocking(const Json& body,
     while (true) {
       // TTFT timer until the first delta, then the total timer
       // (DESIGN.md:706-741)
-      auto msg = conn.read_json(
-          left(first && ttft_timeout_ms_ ? ttft_timeout_ms_
-                                         : total_timeout_ms_));
+      auto msg = MuxClient::next_event(
+          *sink, left(first && ttft_timeout_ms_ ? ttft_timeout_ms_
+                                                : total_timeout_ms_));
       if (!msg) {
-        if (first && !ttft_timeout_ms_ && !total_timeout_ms_ &&
-            attempt < 2) {
-          // EOF with no deadline configured: dying-worker race — retry
+        bool dead;
+        {
+          std::lock_guard<std::mutex> sl(sink->mu);
+          dead = sink->dead;
+        }
+        if (first && dead && attempt < 2) {
+          // channel lost pre-token: dying-worker race — retry
           lease.w->ready = false;
           retry = true;
           break;
         }
-        Json ab = Json::object();
-        ab["type"] = "abort";
-        ab["id"] = rid;
-        conn.send_json(ab);
+        if (dead)
+          throw Problem{502, "Bad Gateway", "about:blank",
+                        "engine connection lost mid-generation",
+                        "provider_error"};
+        mux->abort(rid);
         throw Problem{504, "Gateway Timeout", "about:blank",
                       first ? "no first token within the TTFT budget"
                             : "generation exceeded the total budget",
@@ -486,6 +639,7 @@ Json LlmGatewayModule::run_chat_blocking(const Json& body,
         finish = msg->at("finish_reason").as_string("stop");
         break;
       } else if (ev == "error") {
+        mux->remove(rid);
         throw Problem{502, "Bad Gateway", "about:blank",
                       msg->at("message").as_string(), "provider_error"};
       }
@@ -697,45 +851,52 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   wreq["params"] = params;
 
   // dying-worker race (see run_chat_blocking): obtain the FIRST engine
-  // event before committing the SSE response, so a pre-token EOF can
-  // retry on another worker
-  std::unique_ptr<EngineConn> connp;
+  // event before committing the SSE response, so a pre-token channel
+  // loss can retry on another worker
+  std::shared_ptr<MuxClient> mux;
+  std::shared_ptr<MuxSink> sink;
   Lease lease;
   std::optional<Json> first_msg;
   for (int attempt = 0; attempt < 3 && !first_msg; ++attempt) {
     if (lease.w) { lease.w->in_flight--; lease.w = nullptr; }
-    connp.reset();
-    Worker* w2 = pick_live(connp);
-    lease.w = w2;
+    lease.w = pick_worker();
     if (!lease.w)
       throw Problem{503, "Service Unavailable", "about:blank",
                     "no engine worker ready", "provider_error"};
-    if (!connp->send_json(wreq)) {
+    mux = ensure_mux(*lease.w);
+    if (!mux) {
       lease.w->ready = false;
       continue;
     }
-    first_msg = connp->read_json(
-        ttft_timeout_ms_ ? (int)ttft_timeout_ms_ : 120000);
+    sink = mux->submit(wreq);
+    if (!sink) {
+      lease.w->ready = false;
+      continue;
+    }
+    first_msg = MuxClient::next_event(
+        *sink, ttft_timeout_ms_ ? (int)ttft_timeout_ms_ : 120000);
     if (!first_msg) {
-      if (ttft_timeout_ms_) {
+      bool dead;
+      {
+        std::lock_guard<std::mutex> sl(sink->mu);
+        dead = sink->dead;
+      }
+      if (!dead && ttft_timeout_ms_) {
         // stop the generation server-side before reporting 504 — the
         // blocking path does the same; otherwise the engine keeps
-        // producing until the 600 s stall guard notices
-        Json ab = Json::object();
-        ab["type"] = "abort";
-        ab["id"] = rid;
-        connp->send_json(ab);
+        // producing until the total-budget guard notices
+        mux->abort(rid);
         throw Problem{504, "Gateway Timeout", "about:blank",
                       "no first token within the TTFT budget",
                       "provider_timeout"};
       }
+      if (!dead) mux->abort(rid);
       lease.w->ready = false;   // retry
     }
   }
   if (!first_msg)
     throw Problem{503, "Service Unavailable", "about:blank",
                   "engine workers unavailable", "provider_error"};
-  EngineConn& conn = *connp;
 
   // SSE stream per DESIGN.md:289-311: role chunk, delta chunks, final
   // finish_reason+usage chunk, then data: [DONE]
@@ -755,14 +916,11 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
       msg = first_msg;
       pending_first = false;
     } else {
-      msg = conn.read_json(
-          total_timeout_ms_ ? (int)total_timeout_ms_ : 120000);
+      msg = MuxClient::next_event(
+          *sink, total_timeout_ms_ ? (int)total_timeout_ms_ : 120000);
     }
     if (!msg) {
-      Json ab = Json::object();     // stop the server-side generation
-      ab["type"] = "abort";
-      ab["id"] = rid;
-      conn.send_json(ab);
+      mux->abort(rid);              // stop the server-side generation
       w.write_chunk("data: {\"error\":\"provider_timeout\"}\n\n");
       break;
     }
@@ -779,10 +937,7 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
       d["content"] = msg->at("text").as_string();
       if (!w.write_chunk(sse_chunk(rid, canonical, d))) {
         client_gone = true;    // abort generation server-side
-        Json ab = Json::object();
-        ab["type"] = "abort";
-        ab["id"] = rid;
-        conn.send_json(ab);
+        mux->abort(rid);
         break;
       }
     } else if (ev == "done") {

@@ -44,13 +44,53 @@ class EngineConn {
   explicit EngineConn(const std::string& socket_path);
   ~EngineConn();
   bool ok() const { return fd_ >= 0; }
+  bool eof() const { return eof_; }
   bool send_json(const Json& j);
-  // reads one newline-terminated JSON message; empty on EOF/error
+  // reads one newline-terminated JSON message; empty on EOF/error OR
+  // timeout (eof() distinguishes the two)
   std::optional<Json> read_json(int timeout_ms = 120000);
 
  private:
   int fd_ = -1;
+  bool eof_ = false;
   std::string buf_;
+};
+
+// Multiplexed engine channel: ONE UDS connection per worker carries all
+// chat streams (worker "attach_mux" mode).  The worker writes one
+// batched line per engine step; the reader thread here demultiplexes it
+// into per-request sinks consumed by the SSE/blocking chat handlers.
+// This keeps the worker's GPU loop at one json-encode + one send per
+// step instead of per-token fan-out to thousands of connections.
+struct MuxSink {
+  std::mutex mu;
+  std::condition_variable cv;
+  std::deque<Json> q;            // delta/done/error events
+  bool dead = false;             // channel lost before completion
+};
+
+class MuxClient {
+ public:
+  explicit MuxClient(const std::string& socket_path);
+  ~MuxClient();
+  bool ok() const { return attached_ && !closed_; }
+  // register the request id and send the chat request; null on failure
+  std::shared_ptr<MuxSink> submit(const Json& wreq);
+  void abort(const std::string& rid);    // stop generation + drop sink
+  void remove(const std::string& rid);   // drop sink (request finished)
+  // wait for the next event on a sink; nullopt on timeout or dead
+  static std::optional<Json> next_event(MuxSink& s, int timeout_ms);
+
+ private:
+  void reader_loop();
+  void fail_all();
+  std::unique_ptr<EngineConn> conn_;
+  std::atomic<bool> attached_{false};
+  std::atomic<bool> closed_{false};
+  std::atomic<bool> stop_{false};
+  std::mutex mu_;                // guards writes + sinks map
+  std::map<std::string, std::shared_ptr<MuxSink>> sinks_;
+  std::thread reader_;
 };
 
 class LlmGatewayModule : public Module {
@@ -126,7 +166,11 @@ class LlmGatewayModule : public Module {
     pid_t pid = -1;
     std::atomic<bool> ready{false};
     std::atomic<int> in_flight{0};
+    std::mutex mux_mu;
+    std::shared_ptr<MuxClient> mux;   // serving channel (lazy, respawn-safe)
   };
+  // the worker's mux channel, (re)connecting if stale; null if down
+  std::shared_ptr<MuxClient> ensure_mux(Worker& wk);
   struct Lease {
     Worker* w = nullptr;
     ~Lease() { if (w) w->in_flight--; }
