@@ -97,10 +97,13 @@ void UserSettingsModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
       std::optional<std::string> cursor;
       auto cit = rq.query.find("cursor");
       if (cit != rq.query.end()) cursor = cit->second;
+      SecureConn::OrderBy ob{{"key", false}};
+      auto oit = rq.query.find("$orderby");
       SecureConn::Page page;
       try {
-        page = conn.select("settings", where, binds, "key", false, top,
-                           cursor);
+        if (oit != rq.query.end())
+          ob = parse_odata_orderby(oit->second, filterable);
+        page = conn.select("settings", where, binds, ob, top, cursor);
       } catch (const std::exception& e) {
         throw Problem{400, "Bad Request", "about:blank", e.what(),
                       "validation_error"};

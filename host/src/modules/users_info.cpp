@@ -189,10 +189,14 @@ void UsersInfoModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
       std::optional<std::string> cursor;
       auto cit = rq.query.find("cursor");
       if (cit != rq.query.end()) cursor = cit->second;
+      // $orderby: multi-field signed ordering over the same allow-list
+      SecureConn::OrderBy ob{{"email", false}};
+      auto oit = rq.query.find("$orderby");
       SecureConn::Page page;
       try {
-        page = conn.select("users", where, binds, "email", false, top,
-                           cursor);
+        if (oit != rq.query.end())
+          ob = parse_odata_orderby(oit->second, filterable);
+        page = conn.select("users", where, binds, ob, top, cursor);
       } catch (const std::exception& e) {
         throw Problem{400, "Bad Request", "about:blank", e.what(),
                       "validation_error"};

@@ -99,6 +99,29 @@ int test_cursor_pagination() {
   try { c.select("r", "", {}, "k", false, 3, std::string("@@bad@@")); }
   catch (...) { threw = true; }
   CHECK(threw);
+
+  // multi-field signed ordering: k DESC then i ASC, cursor-stable
+  SecureConn::OrderBy ob{{"k", true}, {"i", false}};
+  std::vector<long> seen2;
+  cur.reset();
+  rounds = 0;
+  while (rounds++ < 20) {
+    auto page = c.select("r", "", {}, ob, 3, cur);
+    for (auto& row : page.items)
+      seen2.push_back(row.at("i").as_int());
+    if (!page.next_cursor) break;
+    cur = page.next_cursor;
+  }
+  CHECK(seen2.size() == 10);
+  // k DESC puts k9..k5 first (9,8,7,6,5), then the dup block i ASC
+  std::vector<long> want{9, 8, 7, 6, 5, 0, 1, 2, 3, 4};
+  for (int i = 0; i < 10; ++i) CHECK(seen2[i] == want[i]);
+  // a cursor from one ordering cannot be replayed under another
+  auto p1 = c.select("r", "", {}, ob, 3, std::nullopt);
+  threw = false;
+  try { c.select("r", "", {}, "k", false, 3, p1.next_cursor); }
+  catch (...) { threw = true; }
+  CHECK(threw);
   return 0;
 }
 
@@ -116,11 +139,42 @@ int test_filter_compile() {
     compile_odata_filter("secret eq 'x'", {"k"}, binds);  // not allowed
   } catch (...) { threw = true; }
   CHECK(threw);
+  // or / not / grouping (modkit-odata Expr parity, lib.rs:23)
+  binds.clear();
+  sql = compile_odata_filter(
+      "(k eq 'a' or k eq 'b') and not (i lt 2)", {"k", "i"}, binds);
+  CHECK(sql.find(" OR ") != std::string::npos);
+  CHECK(sql.find("NOT (") != std::string::npos);
+  CHECK(sql.find("(k=? OR k=?)") != std::string::npos);
+  CHECK(binds.size() == 3);
+  // startswith/endswith; LIKE wildcards in the needle are escaped
+  binds.clear();
+  sql = compile_odata_filter("startswith(k,'a%_b')", {"k"}, binds);
+  CHECK(sql.find("LIKE ? ESCAPE") != std::string::npos);
+  CHECK(binds[0].text == "a\\%\\_b%");
+  binds.clear();
+  sql = compile_odata_filter("endswith(k,'z')", {"k"}, binds);
+  CHECK(binds[0].text == "%z");
+  // unbalanced parens / trailing garbage / deep nesting all throw
+  for (const char* bad :
+       {"(k eq 'a'", "k eq 'a' k eq 'b'", "k eq 'a')",
+        "((((((((((((((((((((((((((((((((((k eq 'a'"}) {
+    threw = false;
+    try {
+      binds.clear();
+      compile_odata_filter(bad, {"k"}, binds);
+    } catch (...) { threw = true; }
+    CHECK(threw);
+  }
+  // $orderby: "a desc,b" and signed tokens "-a,+b"; allow-listed
+  auto ob = parse_odata_orderby("k desc, i", {"k", "i"});
+  CHECK(ob.size() == 2 && ob[0].first == "k" && ob[0].second &&
+        ob[1].first == "i" && !ob[1].second);
+  auto ob2 = parse_odata_orderby("-k,+i", {"k", "i"});
+  CHECK(ob2 == ob);
   threw = false;
-  try {
-    binds.clear();
-    compile_odata_filter("k eq 'a' or k eq 'b'", {"k"}, binds);  // no OR
-  } catch (...) { threw = true; }
+  try { parse_odata_orderby("secret", {"k"}); }
+  catch (...) { threw = true; }
   CHECK(threw);
   // quoting: an embedded quote stays a literal (no injection)
   binds.clear();
