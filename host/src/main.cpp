@@ -15,6 +15,7 @@
 #include "modkit/modkit.h"
 #include "modules/api_gateway.h"
 #include "modules/llm_gateway.h"
+#include "modules/serverless_runtime.h"
 #include "modules/user_settings.h"
 #include "modules/file_parser.h"
 #include "modules/users_info.h"

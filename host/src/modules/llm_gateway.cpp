@@ -21,50 +21,7 @@ static double now_s() {
       .count();
 }
 
-// ------------------------------------------------------ serverless-runtime
-
 namespace {
-
-class StaticAdmission : public AdmissionClient {
- public:
-  explicit StaticAdmission(const Json& cfg) {
-    max_concurrent_ = (int)cfg.path("limits.max_concurrent_per_tenant")
-                          .as_int(64);
-    rps_ = cfg.path("limits.rps_per_tenant").as_number(100);
-    burst_ = cfg.path("limits.burst_per_tenant").as_number(200);
-  }
-  std::string admit(const std::string& tenant) override {
-    std::lock_guard<std::mutex> lk(mu_);
-    auto& st = tenants_[tenant];
-    auto now = std::chrono::steady_clock::now();
-    if (st.last.time_since_epoch().count() == 0) st.tokens = burst_;
-    double dt = std::chrono::duration<double>(now - st.last).count();
-    st.last = now;
-    st.tokens = std::min(burst_, st.tokens + dt * rps_);
-    if (st.in_flight >= max_concurrent_) return "rate_limited";
-    if (st.tokens < 1.0) return "rate_limited";
-    st.tokens -= 1.0;
-    st.in_flight++;
-    return "";
-  }
-  void release(const std::string& tenant) override {
-    std::lock_guard<std::mutex> lk(mu_);
-    auto it = tenants_.find(tenant);
-    if (it != tenants_.end() && it->second.in_flight > 0)
-      it->second.in_flight--;
-  }
-
- private:
-  struct State {
-    int in_flight = 0;
-    double tokens = 0;
-    std::chrono::steady_clock::time_point last{};
-  };
-  std::mutex mu_;
-  std::map<std::string, State> tenants_;
-  int max_concurrent_;
-  double rps_, burst_;
-};
 
 // base64 for embedding_response encoding_format=base64 (little-endian f32,
 // the OpenAI-compatible convention the schema references)
@@ -86,11 +43,6 @@ std::string b64_encode(const unsigned char* data, size_t n) {
 }
 
 }  // namespace
-
-void ServerlessRuntimeModule::init(ModuleCtx& ctx) {
-  ctx.hub->register_client<AdmissionClient>(
-      "serverless-runtime", std::make_shared<StaticAdmission>(ctx.config));
-}
 
 // ------------------------------------------------------------- EngineConn
 
@@ -280,8 +232,30 @@ std::optional<Json> MuxClient::next_event(MuxSink& s, int timeout_ms) {
 
 // ---------------------------------------------------------- LlmGateway
 
+namespace {
+class GatewayChatInvoker : public ChatInvoker {
+ public:
+  explicit GatewayChatInvoker(LlmGatewayModule* m) : m_(m) {}
+  Json chat(const SecurityContext& sec, const Json& body) override {
+    return m_->invoke_chat(sec, body);
+  }
+
+ private:
+  LlmGatewayModule* m_;
+};
+}  // namespace
+
+Json LlmGatewayModule::invoke_chat(const SecurityContext& sec,
+                                   const Json& body) {
+  const std::string rid = "sl-" + std::to_string(++req_ctr_);
+  return run_chat_with_fallback(sec, body, rid);
+}
+
 void LlmGatewayModule::init(ModuleCtx& ctx) {
   hub_ = ctx.hub;
+  ctx.hub->register_client<ChatInvoker>(
+      "llm-gateway",
+      std::make_shared<GatewayChatInvoker>(this));
   model_ = ctx.config.at("model").as_string(model_);
   socket_path_ = ctx.config.at("worker_socket").as_string(socket_path_);
   auto_start_ = ctx.config.at("auto_start_worker").as_bool(true);

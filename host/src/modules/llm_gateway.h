@@ -32,10 +32,12 @@ struct AdmissionClient {
   virtual void release(const std::string& tenant) = 0;
 };
 
-class ServerlessRuntimeModule : public Module {
- public:
-  std::string name() const override { return "serverless-runtime"; }
-  void init(ModuleCtx& ctx) override;
+// blocking chat invoker registered by llm-gateway for in-process
+// consumers (serverless-runtime adapter_ref entrypoints — the spec's
+// "entrypoints are model workers" reading)
+struct ChatInvoker {
+  virtual ~ChatInvoker() = default;
+  virtual Json chat(const SecurityContext& sec, const Json& body) = 0;
 };
 
 // Blocking JSON-lines client for one worker request over UDS.
@@ -108,6 +110,8 @@ class LlmGatewayModule : public Module {
 
   const std::string& socket_path() const { return socket_path_; }
   bool worker_ready();                      // any worker ready
+  // in-process blocking chat (ChatInvoker registration target)
+  Json invoke_chat(const SecurityContext& sec, const Json& body);
 
  private:
   // async job state machine (DESIGN.md job/batch schemas; statuses

@@ -1,0 +1,355 @@
+"""serverless-runtime e2e: entrypoint lifecycle, invocation state
+machine, retry w/ backoff, saga compensation, suspend/resume, timeline
+events, durable recovery across host restart, and the LLM adapter.
+
+Contract: reference modules/serverless-runtime/docs/
+ADR_DOMAIN_MODEL_AND_APIS.md (:1030-1087 state machine, :1233 timeline,
+:2599-2636 REST surface) — implemented in
+host/src/modules/serverless_runtime.cpp.
+"""
+
+import json
+import tempfile
+import time
+from pathlib import Path
+
+import pytest
+
+from tests.test_host_e2e import ServerProc, _free_port, _http
+
+BASE = "http://127.0.0.1:{}"
+
+
+def _mk_cfg(port, sock, home):
+    return f"""
+server:
+  home_dir: "{home}"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+  serverless-runtime:
+    config:
+      executors: 2
+  llm-gateway:
+    config:
+      model: "tiny-llama"
+      worker_socket: "{sock}"
+      auto_start_worker: true
+      worker:
+        device: "cpu"
+        eager: true
+        max_num_seqs: 8
+        num_gpu_blocks: 256
+"""
+
+
+@pytest.fixture(scope="module")
+def sl():
+    home = tempfile.mkdtemp(prefix="hs-sl-")
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-sl-")
+    port = _free_port()
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(_mk_cfg(port, sock, home))
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        yield srv, home, cfg_path
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
+
+
+def _url(srv):
+    return BASE.format(srv.port) + "/serverless-runtime/v1"
+
+
+def _mk_active_ep(srv, body):
+    st, resp = _http("POST", _url(srv) + "/entrypoints", body)
+    assert st == 201, resp
+    ep = json.loads(resp)
+    assert ep["status"] == "draft"
+    st, resp = _http("POST",
+                     _url(srv) + f"/entrypoints/{ep['id']}/status",
+                     {"action": "activate"})
+    assert st == 200, resp
+    assert json.loads(resp)["status"] == "active"
+    return ep["id"]
+
+
+def _wait_status(srv, inv_id, statuses, timeout=20):
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        st, resp = _http("GET", _url(srv) + f"/invocations/{inv_id}")
+        assert st == 200, resp
+        j = json.loads(resp)
+        if j["status"] in statuses:
+            return j
+        time.sleep(0.1)
+    raise TimeoutError(f"{inv_id} never reached {statuses}: {j}")
+
+
+def _timeline(srv, inv_id):
+    st, resp = _http("GET", _url(srv) + f"/invocations/{inv_id}/timeline")
+    assert st == 200, resp
+    return [e["event_type"] for e in json.loads(resp)["items"]]
+
+
+FN = {"name": "fn", "entrypoint_type": "function",
+      "implementation": {"adapter":
+                         "gts.x.core.serverless.adapter.builtin.v1~",
+                         "kind": "code",
+                         "code": {"language": "builtin", "source": "echo"}}}
+
+
+def test_entrypoint_lifecycle_and_validation(sl):
+    srv, _, _ = sl
+    # validate endpoint flags issues without saving
+    bad = {"name": "", "entrypoint_type": "nope", "implementation": {}}
+    st, resp = _http("POST", _url(srv) + "/entrypoints/validate", bad)
+    assert st == 200
+    v = json.loads(resp)
+    assert v["valid"] is False and len(v["issues"]) >= 2
+    # create → draft; PUT allowed; activate; PUT now 409; deprecate;
+    # disable; enable
+    st, resp = _http("POST", _url(srv) + "/entrypoints", FN)
+    assert st == 201, resp
+    ep = json.loads(resp)
+    st, _ = _http("PUT", _url(srv) + f"/entrypoints/{ep['id']}",
+                  dict(FN, name="fn2"))
+    assert st == 200
+    st, _ = _http("POST", _url(srv) + f"/entrypoints/{ep['id']}/status",
+                  {"action": "activate"})
+    assert st == 200
+    st, _ = _http("PUT", _url(srv) + f"/entrypoints/{ep['id']}",
+                  dict(FN, name="fn3"))
+    assert st == 409
+    for action, want in (("deprecate", "deprecated"),
+                         ("disable", "disabled"), ("enable", "active")):
+        st, resp = _http("POST",
+                         _url(srv) + f"/entrypoints/{ep['id']}/status",
+                         {"action": action})
+        assert st == 200 and json.loads(resp)["status"] == want, resp
+    # illegal action from current state
+    st, _ = _http("POST", _url(srv) + f"/entrypoints/{ep['id']}/status",
+                  {"action": "activate"})
+    assert st == 409
+    # delete active → archived (soft); draft → hard delete
+    st, _ = _http("DELETE", _url(srv) + f"/entrypoints/{ep['id']}")
+    assert st == 204
+    st, resp = _http("GET", _url(srv) + f"/entrypoints/{ep['id']}")
+    assert st == 200 and json.loads(resp)["status"] == "archived"
+
+
+def test_sync_invocation_echo(sl):
+    srv, _, _ = sl
+    ep = _mk_active_ep(srv, FN)
+    st, resp = _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep, "mode": "sync",
+                      "input": {"x": 42}})
+    assert st == 200, resp
+    j = json.loads(resp)
+    assert j["status"] == "succeeded" and j["result"] == {"x": 42}
+    evs = _timeline(srv, j["id"])
+    assert evs[0] == "started" and evs[-1] == "succeeded"
+
+
+def test_retry_policy_with_backoff(sl):
+    srv, _, _ = sl
+    ep = _mk_active_ep(srv, {
+        "name": "flaky", "entrypoint_type": "function",
+        "retry_policy": {"max_attempts": 3, "backoff_ms": 50},
+        "implementation": {"adapter": "a~", "kind": "code",
+                           "code": {"language": "builtin",
+                                    "source": "fail:2"}}})
+    st, resp = _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep, "input": {"v": 1}})
+    assert st == 202, resp
+    inv = json.loads(resp)
+    j = _wait_status(srv, inv["id"], {"succeeded"})
+    assert j["attempts"] == 3
+    evs = _timeline(srv, inv["id"])
+    assert evs.count("failed") == 2 and evs.count("step_retried") == 2
+    assert evs[-1] == "succeeded"
+
+
+def test_retries_exhausted_dead_letter(sl):
+    srv, _, _ = sl
+    ep = _mk_active_ep(srv, {
+        "name": "doomed", "entrypoint_type": "function",
+        "retry_policy": {"max_attempts": 2, "backoff_ms": 10},
+        "implementation": {"adapter": "a~", "kind": "code",
+                           "code": {"language": "builtin",
+                                    "source": "error:nope"}}})
+    st, resp = _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep, "input": {}})
+    inv = json.loads(resp)
+    j = _wait_status(srv, inv["id"], {"dead_lettered"})
+    assert "nope" in j["error"]
+    evs = _timeline(srv, inv["id"])
+    assert evs[-1] == "dead_lettered"
+    # manual retry only valid from failed — dead_lettered is terminal
+    st, _ = _http("POST",
+                  _url(srv) + f"/invocations/{inv['id']}/control",
+                  {"action": "retry"})
+    assert st == 409
+    # replay creates a NEW invocation
+    st, resp = _http("POST",
+                     _url(srv) + f"/invocations/{inv['id']}/control",
+                     {"action": "replay"})
+    assert st == 202, resp
+    assert json.loads(resp)["id"] != inv["id"]
+
+
+def test_workflow_steps_and_compensation(sl):
+    srv, _, _ = sl
+    ep = _mk_active_ep(srv, {
+        "name": "saga", "entrypoint_type": "workflow",
+        "implementation": {"adapter": "a~", "kind": "workflow_spec",
+                           "workflow": {"steps": [
+                               {"name": "reserve", "op": "echo",
+                                "compensation": "echo"},
+                               {"name": "charge", "op": "upper",
+                                "compensation": "echo"},
+                               {"name": "boom", "op": "error:step3"},
+                           ]}}})
+    st, resp = _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep, "mode": "sync",
+                      "input": {"text": "hi"}})
+    assert st == 200, resp
+    j = json.loads(resp)
+    assert j["status"] == "compensated", j
+    evs = _timeline(srv, j["id"])
+    assert "step_started" in evs and "step_failed" in evs
+    assert "compensation_started" in evs
+    assert evs[-1] == "compensation_completed"
+
+
+def test_workflow_success_timeline(sl):
+    srv, _, _ = sl
+    ep = _mk_active_ep(srv, {
+        "name": "wf-ok", "entrypoint_type": "workflow",
+        "implementation": {"adapter": "a~", "kind": "workflow_spec",
+                           "workflow": {"steps": [
+                               {"name": "s1", "op": "echo"},
+                               {"name": "s2", "op": "upper"},
+                           ]}}})
+    st, resp = _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep, "mode": "sync",
+                      "input": {"text": "abc"}})
+    j = json.loads(resp)
+    assert j["status"] == "succeeded" and j["result"]["text"] == "ABC"
+    evs = _timeline(srv, j["id"])
+    assert evs.count("step_started") == 2
+    assert evs.count("step_completed") == 2
+
+
+def test_suspend_resume_and_cancel(sl):
+    srv, _, _ = sl
+    ep = _mk_active_ep(srv, {
+        "name": "slow", "entrypoint_type": "workflow",
+        "implementation": {"adapter": "a~", "kind": "workflow_spec",
+                           "workflow": {"steps": [
+                               {"name": "s1", "op": "sleep:300"},
+                               {"name": "s2", "op": "sleep:300"},
+                               {"name": "s3", "op": "echo"},
+                           ]}}})
+    st, resp = _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep, "input": {"k": 1}})
+    inv = json.loads(resp)
+    _wait_status(srv, inv["id"], {"running"})
+    st, _ = _http("POST", _url(srv) + f"/invocations/{inv['id']}/control",
+                  {"action": "suspend"})
+    assert st == 200
+    j = _wait_status(srv, inv["id"], {"suspended"})
+    assert "suspended" in _timeline(srv, inv["id"])
+    # resume → continues from the stored step index and completes
+    st, _ = _http("POST", _url(srv) + f"/invocations/{inv['id']}/control",
+                  {"action": "resume"})
+    assert st == 200
+    j = _wait_status(srv, inv["id"], {"succeeded"})
+    assert j["result"] == {"k": 1}
+    evs = _timeline(srv, inv["id"])
+    assert "resumed" in evs
+
+    # cancel mid-run
+    st, resp = _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep, "input": {}})
+    inv2 = json.loads(resp)
+    _wait_status(srv, inv2["id"], {"running"})
+    st, _ = _http("POST",
+                  _url(srv) + f"/invocations/{inv2['id']}/control",
+                  {"action": "cancel"})
+    assert st == 200
+    _wait_status(srv, inv2["id"], {"canceled"})
+
+
+def test_invocation_list_filter_orderby(sl):
+    srv, _, _ = sl
+    st, resp = _http(
+        "GET", _url(srv) + "/invocations?$filter=status%20eq%20"
+        "'succeeded'&$orderby=created_at%20desc,id&$top=5")
+    assert st == 200, resp
+    j = json.loads(resp)
+    assert all(i["status"] == "succeeded" for i in j["items"])
+
+
+def test_llm_adapter_entrypoint(sl):
+    """adapter_ref → in-node LLM engine via the llm-gateway client (the
+    spec's 'entrypoints are model workers' reading)."""
+    srv, _, _ = sl
+    srv.wait_worker()
+    ep = _mk_active_ep(srv, {
+        "name": "chat", "entrypoint_type": "function",
+        "implementation": {"adapter": "gts.x.genai.llm.chat.v1~",
+                           "kind": "adapter_ref"}})
+    st, resp = _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep, "mode": "sync",
+                      "input": {"model": "tiny-llama", "max_tokens": 4,
+                                "temperature": 0.0,
+                                "messages": [{"role": "user", "content":
+                                              [{"type": "text",
+                                                "text": "hi"}]}]}})
+    assert st == 200, resp
+    j = json.loads(resp)
+    assert j["status"] == "succeeded", j
+    assert j["result"]["usage"]["output_tokens"] >= 1
+
+
+def test_durable_recovery_across_restart(sl):
+    """Queued work survives a host restart (PRD.md:44-45 RTO/RPO)."""
+    srv, home, cfg_path = sl
+    ep = _mk_active_ep(srv, {
+        "name": "later", "entrypoint_type": "function",
+        "retry_policy": {"max_attempts": 2, "backoff_ms": 8000},
+        "implementation": {"adapter": "a~", "kind": "code",
+                           "code": {"language": "builtin",
+                                    "source": "fail:1"}}})
+    # first attempt fails; the retry sits in an 8 s backoff window
+    st, resp = _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep, "input": {"z": 9}})
+    inv = json.loads(resp)
+    _wait_status(srv, inv["id"], {"queued"}, timeout=10)
+    srv.stop()
+
+    port2 = _free_port()
+    cfg2 = Path(tempfile.mktemp(suffix=".yaml"))
+    sock2 = tempfile.mktemp(suffix=".sock", prefix="hs-sl2-")
+    cfg2.write_text(_mk_cfg(port2, sock2, home))
+    srv2 = ServerProc(cfg2, port2)
+    try:
+        srv2.wait_ready()
+        # the recovered invocation runs to completion on the new host
+        j = _wait_status(srv2, inv["id"], {"succeeded"}, timeout=30)
+        assert j["result"] == {"z": 9} and j["attempts"] == 2
+        # entrypoints survived too
+        st, resp = _http("GET",
+                         _url(srv2) + f"/entrypoints/{ep}")
+        assert st == 200 and json.loads(resp)["status"] == "active"
+    finally:
+        srv2.stop()
+        cfg2.unlink(missing_ok=True)
