@@ -63,13 +63,11 @@ class WorkerState:
         self.bench_result = queue.Queue()
         self.pending_exec = []                # (kind, arg, reply_q)
         self.mux_of = {}                      # rid -> MuxChannel
-        self.mux_meta = {}                    # rid -> (detok, prompt_len,
-                                              #         n_out)
 
     def submit_mux(self, rid, prompt_ids, sampling, mux, detok):
+        mux.track(rid, detok, len(prompt_ids))
         with self.new_work:
             self.mux_of[rid] = mux
-            self.mux_meta[rid] = [detok, len(prompt_ids), 0]
             self.engine.add_request(prompt_ids, sampling, request_id=rid)
             if self.tp > 1:
                 self.pending_ops.append(
@@ -93,8 +91,9 @@ class WorkerState:
             self.engine.abort_request(rid)
             q = self.streams.pop(rid, None)
             self.started.pop(rid, None)
-            self.mux_of.pop(rid, None)
-            self.mux_meta.pop(rid, None)
+            mux = self.mux_of.pop(rid, None)
+            if mux is not None:
+                mux.forget(rid)
             if q is not None:
                 q.put(None)    # wake the serving thread (abort sentinel)
             if self.tp > 1:
@@ -161,27 +160,21 @@ class WorkerState:
             self._fanout(outputs)
 
     def _fanout(self, outputs):
+        # mux path: hand the RAW outputs to the channel's writer thread;
+        # detokenization + json encoding happen off the GPU loop
         per_mux = None
         for out in outputs:
             rid = out.request_id
             mux = self.mux_of.get(rid)
             if mux is not None:
-                meta = self.mux_meta[rid]
-                text = meta[0].push(out.token_id)
-                meta[2] += 1
                 if per_mux is None:
                     per_mux = {}
-                items = per_mux.setdefault(mux, [])
+                fr = ((out.finish_reason.value if out.finish_reason
+                       else "stop") if out.finished else None)
+                per_mux.setdefault(mux, []).append(
+                    (rid, out.token_id, fr))
                 if out.finished:
-                    text += meta[0].flush()
-                    fr = (out.finish_reason.value if out.finish_reason
-                          else "stop")
-                    items.append([rid, out.token_id, text, fr,
-                                  meta[1], meta[2]])
                     self.mux_of.pop(rid, None)
-                    self.mux_meta.pop(rid, None)
-                else:
-                    items.append([rid, out.token_id, text])
                 continue
             q = self.streams.get(rid)
             if q is not None:
@@ -190,8 +183,8 @@ class WorkerState:
                     self.streams.pop(rid, None)
                     self.started.pop(rid, None)
         if per_mux:
-            for mux, items in per_mux.items():
-                mux.send_batch(items)
+            for mux, raw in per_mux.items():
+                mux.send_outputs(raw)
 
     def _one_step(self):
         import torch.distributed as dist
@@ -247,28 +240,36 @@ class WorkerState:
 
 
 class MuxChannel:
-    """Outbox for one multiplexed gateway connection.  The stepping
-    thread enqueues whole-step batch lines; a writer thread drains them
-    with coalesced sends so a momentarily-slow reader never blocks the
-    GPU loop.  A reader that stops draining for >60 s is treated as dead
-    (the gateway is local; its C++ reader keeps up by construction)."""
+    """Outbox for one multiplexed gateway connection.  The GPU stepping
+    thread enqueues RAW per-step outputs; this channel's writer thread
+    detokenizes, json-encodes and sends them, so the GPU loop pays one
+    list-append per token and nothing else.  A reader that stops
+    draining is treated as dead (the gateway's C++ reader keeps up by
+    construction)."""
 
     MAX_PENDING = 4096          # batch lines (~one per engine step)
 
     def __init__(self, sock: socket.socket):
         self.sock = sock
-        self.pending = []
+        self.pending = []           # bytes | list of raw outputs
         self.cv = threading.Condition()
         self.dead = False
+        self.meta = {}              # rid -> [detok, prompt_len, n_out]
         self.writer = threading.Thread(target=self._write_loop,
                                        daemon=True)
         self.writer.start()
 
-    def send_batch(self, items):
-        self.send_obj({"event": "batch", "items": items})
+    def track(self, rid, detok, prompt_len):
+        with self.cv:
+            self.meta[rid] = [detok, prompt_len, 0]
 
-    def send_obj(self, obj):
-        line = (json.dumps(obj, separators=(",", ":")) + "\n").encode()
+    def forget(self, rid):
+        with self.cv:
+            self.meta.pop(rid, None)
+
+    def send_outputs(self, raw):
+        """Raw (rid, token_id, finish_reason|None) tuples from the GPU
+        loop; the writer thread does the rest."""
         with self.cv:
             if self.dead:
                 return
@@ -276,8 +277,34 @@ class MuxChannel:
                 self.dead = True          # reader stalled: drop channel
                 self.cv.notify()
                 return
+            self.pending.append(raw)
+            self.cv.notify()
+
+    def send_obj(self, obj):
+        line = (json.dumps(obj, separators=(",", ":")) + "\n").encode()
+        with self.cv:
+            if self.dead:
+                return
             self.pending.append(line)
             self.cv.notify()
+
+    def _encode(self, raw):
+        items = []
+        meta = self.meta
+        for rid, tok, fr in raw:
+            m = meta.get(rid)
+            if m is None:
+                continue               # aborted after the step ran
+            text = m[0].push(tok)
+            m[2] += 1
+            if fr is None:
+                items.append([rid, tok, text])
+            else:
+                text += m[0].flush()
+                items.append([rid, tok, text, fr, m[1], m[2]])
+                meta.pop(rid, None)
+        return (json.dumps({"event": "batch", "items": items},
+                           separators=(",", ":")) + "\n").encode()
 
     def _write_loop(self):
         while True:
@@ -286,8 +313,13 @@ class MuxChannel:
                     self.cv.wait(timeout=1.0)
                 if self.dead and not self.pending:
                     return
-                chunk = b"".join(self.pending)
-                self.pending.clear()
+                work = self.pending
+                self.pending = []
+                # encode under the lock: meta is shared with
+                # track/forget, and batches must stay FIFO
+                chunk = b"".join(
+                    w if isinstance(w, bytes) else self._encode(w)
+                    for w in work)
             try:
                 self.sock.sendall(chunk)
             except (BrokenPipeError, OSError):
