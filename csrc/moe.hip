@@ -22,7 +22,7 @@
 //                atomics).
 #include "common.h"
 
-#define MOE_BM 128  // sorted rows per expert tile (the align granularity)
+#define MOE_BM 256  // sorted rows per expert tile (the align granularity)
 
 // ---------------------------------------------------------------- align
 __global__ __launch_bounds__(256) void moe_align_kernel(
@@ -77,94 +77,149 @@ DEV int swz(int row, int byte_in_row) {
   return row * 128 + (byte_in_row ^ ((row & 7) << 4));
 }
 
+// glds helper: each lane supplies a per-lane GLOBAL address; the LDS
+// destination is wave-uniform base + lane*16 (guide §5) — swizzled LDS
+// images are made by pre-swizzling the SOURCE address, LDS stays linear.
+DEV void glds16(const void* gp, void* lp) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned int*)gp,
+      (__attribute__((address_space(3))) unsigned int*)lp, 16, 0, 0);
+}
+
 // out[P, N] (sorted space) = gather(x) @ W[e]^T ; W is [E, N, K] row-major.
-// gather_div > 0: A row p comes from x[sorted_ids[p] / gather_div]
-// gather_div == 0: A row p is x[p] (identity; padded rows are junk and the
-// combine step drops them).
-__global__ __launch_bounds__(512) void moe_gemm_kernel(
+//
+// v2 (guide §5 "canonical CDNA GEMM" + glds table): 256(M) x BN(N) tile
+// per 512-thread block, BK = 64 bf16 (128-B rows), async global->LDS
+// (global_load_lds_dwordx4) into a 2-deep LDS ring, one block/CU.  The
+// 128x128 single-buffered v1 capped at ~620 TF (guide's own number for
+// that shape) = 23.7% MfmaUtil measured; the 256-row tile additionally
+// HALVES the expert-weight re-read (ceil(tokens_e/256) streams of each
+// [N,K] expert matrix per layer — the binding constraint at bs1024,
+// profiles/r01_decode_v3.md).
+//
+// Per k-chunk each wave issues its 1/8 share of the A+B stage as glds
+// (8-row x 128-B pieces, XOR pre-swizzle on the source byte), computes
+// the CURRENT chunk from the other LDS buffer, then __syncthreads()
+// (hipcc inserts the vmcnt(0) drain before the barrier — the guide's
+// recommended 2-buffer recipe).
+template <int BN>
+__global__ __launch_bounds__(512) void moe_gemm_v2_kernel(
     bf16* __restrict__ out, const bf16* __restrict__ x,
     const bf16* __restrict__ w, const int* __restrict__ sorted_ids,
     const int* __restrict__ tile_expert, int N, int K, int gather_div) {
-  // grid is (n_tiles, m_tiles): with n_tiles % 8 == 0 every m-tile of one
-  // (expert, n0) pair lands on the SAME XCD (dispatch round-robins
-  // blockIdx linearly over the 8 XCDs), so the expert's B-slice is read
-  // from that XCD's L2 instead of HBM for the 2nd..kth m-tile
   const int e = tile_expert[blockIdx.y];
   if (e < 0) return;
   const int m0 = blockIdx.y * MOE_BM;
-  const int n0 = blockIdx.x * 128;
+  const int n0 = blockIdx.x * BN;
 
-  // single-buffered tiles: a 2-deep LDS ring (64 KiB) was measured 10%
-  // SLOWER here — it halves WG residency (guide: explicit dbuf at HIP
-  // source is not a lever; occupancy is)
-  __shared__ __attribute__((aligned(16))) unsigned char As[128 * 128];
-  __shared__ __attribute__((aligned(16))) unsigned char Bs[128 * 128];
+  extern __shared__ int lds[];     // dynamic; shared decl w/ moe_align
+  unsigned char* ldsb = reinterpret_cast<unsigned char*>(lds);
+  // layout: A0 | A1 | B0 | B1  (A: 32 KiB each, B: BN*128 each)
+  unsigned char* As[2] = {ldsb, ldsb + 32 * 1024};
+  unsigned char* Bs[2] = {ldsb + 64 * 1024, ldsb + 64 * 1024 + BN * 128};
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int lane = tid & 63;
   const int lcol = lane & 31;
   const int lhalf = lane >> 5;
-  const int wave_m = wid & 3;   // 4 row-blocks of 32
-  const int wave_n = wid >> 2;  // 2 col-blocks of 64
+  const int wave_m = wid & 3;   // 4 row-blocks of 64
+  const int wave_n = wid >> 2;  // 2 col-blocks of BN/2
 
-  // loader role: thread covers 32 B of one row (4 threads per 128-row tile
-  // side, 64-elem K slice = 128 B per row)
-  const int ar = tid >> 2;        // 0..127
-  const int ac32 = (tid & 3) * 32;  // byte offset of this thread's 32 B
-  int a_src = -1;
-  if (gather_div > 0) {
-    const int pair = sorted_ids[m0 + ar];
-    if (pair >= 0) a_src = pair / gather_div;
-  } else {
-    a_src = m0 + ar;
+  // ---- loader geometry: wave wid fills A rows [wid*32, wid*32+32) in 4
+  // glds of 8 rows; lane covers row lrow of its piece at swizzled source
+  // byte gbyte (lane-linear LDS => pre-swizzled global address)
+  const int lrow = lane >> 3;           // 0..7
+  const int gbyte = ((lane & 7) * 16) ^ (lrow << 4);
+  // per-j A source rows (gathered); padded rows read x row 0 (their
+  // outputs are dropped by combine) so the glds always has a valid VA
+  const unsigned char* a_src[4];
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int r = wid * 32 + j * 8 + lrow;
+    long row = 0;
+    if (gather_div > 0) {
+      const int pair = sorted_ids[m0 + r];
+      if (pair >= 0) row = pair / gather_div;
+    } else {
+      row = m0 + r;
+    }
+    a_src[j] = reinterpret_cast<const unsigned char*>(x) +
+               row * (long)K * 2 + gbyte;
+  }
+  // B rows: wave wid covers BN/8 rows in BN/64 glds
+  const int b_glds = BN / 64;
+  const unsigned char* b_src[4];
+  #pragma unroll
+  for (int j = 0; j < b_glds; ++j) {
+    const int r = wid * (BN / 8) + j * 8 + lrow;
+    b_src[j] = reinterpret_cast<const unsigned char*>(w) +
+               ((long)e * N + n0 + r) * (long)K * 2 + gbyte;
   }
 
-  f32x16 acc[2];
+  constexpr int NB = BN / 2 / 32;     // 32-col mfma blocks per wave
+  f32x16 acc[2][NB];
   #pragma unroll
-  for (int nb = 0; nb < 2; ++nb)
+  for (int sm = 0; sm < 2; ++sm)
     #pragma unroll
-    for (int i = 0; i < 16; ++i) acc[nb][i] = 0.f;
+    for (int nb = 0; nb < NB; ++nb)
+      #pragma unroll
+      for (int i = 0; i < 16; ++i) acc[sm][nb][i] = 0.f;
 
-  const bf16* ap0 = a_src >= 0 ? x + (long)a_src * K + ac32 / 2 : nullptr;
-  const bf16* bp0 = w + ((long)e * N + n0 + ar) * K + ac32 / 2;
-  for (int k0 = 0; k0 < K; k0 += 64) {
-    uint4 av0 = uint4{0, 0, 0, 0}, av1 = uint4{0, 0, 0, 0};
-    if (ap0) {
-      av0 = *reinterpret_cast<const uint4*>(ap0 + k0);
-      av1 = *reinterpret_cast<const uint4*>(ap0 + k0 + 8);
-    }
-    uint4 bv0 = *reinterpret_cast<const uint4*>(bp0 + k0);
-    uint4 bv1 = *reinterpret_cast<const uint4*>(bp0 + k0 + 8);
-    *reinterpret_cast<uint4*>(&As[swz(ar, ac32)]) = av0;
-    *reinterpret_cast<uint4*>(&As[swz(ar, ac32 + 16)]) = av1;
-    *reinterpret_cast<uint4*>(&Bs[swz(ar, ac32)]) = bv0;
-    *reinterpret_cast<uint4*>(&Bs[swz(ar, ac32 + 16)]) = bv1;
-    __syncthreads();
+  auto stage = [&](int buf, int k0) {
+    const long cb = (long)k0 * 2;     // chunk byte offset in the row
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+      glds16(a_src[j] + cb, As[buf] + wid * 4096 + j * 1024);
+    #pragma unroll
+    for (int j = 0; j < b_glds; ++j)
+      glds16(b_src[j] + cb, Bs[buf] + wid * (BN * 16) + j * 1024);
+  };
+
+  stage(0, 0);
+  __syncthreads();                    // vmcnt(0) drain inserted by hipcc
+
+  const int nchunks = K >> 6;
+  for (int c = 0; c < nchunks; ++c) {
+    const int cur = c & 1;
+    if (c + 1 < nchunks) stage(cur ^ 1, (c + 1) << 6);
+    const unsigned char* A = As[cur];
+    const unsigned char* B = Bs[cur];
     #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
-      short8 af = *reinterpret_cast<const short8*>(
-          &As[swz(wave_m * 32 + lcol, kk * 32 + lhalf * 16)]);
+      short8 af[2];
       #pragma unroll
-      for (int nb = 0; nb < 2; ++nb) {
+      for (int sm = 0; sm < 2; ++sm)
+        af[sm] = *reinterpret_cast<const short8*>(
+            &A[swz(wave_m * 64 + sm * 32 + lcol,
+                   kk * 32 + lhalf * 16)]);
+      #pragma unroll
+      for (int nb = 0; nb < NB; ++nb) {
         short8 bf = *reinterpret_cast<const short8*>(
-            &Bs[swz(wave_n * 64 + nb * 32 + lcol,
-                    kk * 32 + lhalf * 16)]);
-        acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, acc[nb],
-                                                          0, 0, 0);
+            &B[swz(wave_n * (BN / 2) + nb * 32 + lcol,
+                   kk * 32 + lhalf * 16)]);
+        #pragma unroll
+        for (int sm = 0; sm < 2; ++sm)
+          acc[sm][nb] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af[sm], bf, acc[sm][nb], 0, 0, 0);
       }
     }
-    __syncthreads();
+    __syncthreads();                  // also drains the staged c+1 glds
   }
+
   // ---- epilogue: C row = (reg&3) + 8*(reg>>2) + 4*lhalf, col = lcol ----
   #pragma unroll
-  for (int nb = 0; nb < 2; ++nb)
+  for (int sm = 0; sm < 2; ++sm)
     #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int m = (r & 3) + 8 * (r >> 2) + 4 * lhalf;
-      *(unsigned short*)(out + (long)(m0 + wave_m * 32 + m) * N + n0 +
-                         wave_n * 64 + nb * 32 + lcol) = f2bf(acc[nb][r]);
-    }
+    for (int nb = 0; nb < NB; ++nb)
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int m = (r & 3) + 8 * (r >> 2) + 4 * lhalf;
+        *(unsigned short*)(out +
+                           (long)(m0 + wave_m * 64 + sm * 32 + m) * N +
+                           n0 + wave_n * (BN / 2) + nb * 32 + lcol) =
+            f2bf(acc[sm][nb][r]);
+      }
 }
 
 // ------------------------------------------------------------- combine
@@ -206,11 +261,37 @@ void launch_moe_gemm(bf16* out, const bf16* x, const bf16* w,
                      const int* sorted_ids, const int* tile_expert,
                      int ntiles_max, int N, int K, int gather_div,
                      hipStream_t stream) {
-  if (N % 128 || K % 64)
-    throw std::runtime_error("moe_gemm: N%128 or K%64 != 0");
-  dim3 grid((unsigned)(N / 128), (unsigned)ntiles_max);
-  moe_gemm_kernel<<<grid, 512, 0, stream>>>(out, x, w, sorted_ids,
-                                            tile_expert, N, K, gather_div);
+  if (K % 64)
+    throw std::runtime_error("moe_gemm: K % 64 != 0");
+  // BN=256 when the grid still fills the 256 CUs, else BN=128
+  const bool wide = (N % 256 == 0) &&
+                    (long)(N / 256) * ntiles_max >= 512;
+  static bool attr_set[2] = {false, false};
+  if (wide) {
+    if (N % 256) throw std::runtime_error("moe_gemm: N % 256 != 0");
+    const int lds_bytes = 2 * (32 * 1024 + 256 * 128);
+    if (!attr_set[0]) {
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&moe_gemm_v2_kernel<256>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
+      attr_set[0] = true;
+    }
+    dim3 grid((unsigned)(N / 256), (unsigned)ntiles_max);
+    moe_gemm_v2_kernel<256><<<grid, 512, lds_bytes, stream>>>(
+        out, x, w, sorted_ids, tile_expert, N, K, gather_div);
+  } else {
+    if (N % 128) throw std::runtime_error("moe_gemm: N % 128 != 0");
+    const int lds_bytes = 2 * (32 * 1024 + 128 * 128);
+    if (!attr_set[1]) {
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&moe_gemm_v2_kernel<128>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
+      attr_set[1] = true;
+    }
+    dim3 grid((unsigned)(N / 128), (unsigned)ntiles_max);
+    moe_gemm_v2_kernel<128><<<grid, 512, lds_bytes, stream>>>(
+        out, x, w, sorted_ids, tile_expert, N, K, gather_div);
+  }
 }
 
 void launch_moe_combine(bf16* out, const bf16* y, const float* wts,
@@ -223,33 +304,28 @@ void launch_moe_combine(bf16* out, const bf16* y, const float* wts,
 }
 
 // ----------------------------------------------------- fp8 grouped GEMM
-// Same 128x128 tiling as moe_gemm_kernel, e4m3fn operands:
+// Same 256xBN glds structure as moe_gemm_v2_kernel, e4m3fn operands:
 // out[p,n] = (sum_k A8[p,k] * B8[e,n,k]) * a_scale[row(p)] * b_scale[e,n]
-// via v_mfma_f32_32x32x16_fp8_fp8 (identical C layout to the bf16 tile;
-// each lane feeds 8 fp8 bytes per operand).  Halved LDS/HBM per tile —
-// the MoE step is expert-weight-stream bound, so fp8 is ~2x weight BW.
+// via v_mfma_f32_32x32x16_fp8_fp8.  128-B staged rows = BK 128 fp8
+// elements, so the loader geometry is byte-identical to the bf16 tile;
+// each mfma consumes 8 fp8 bytes per lane (kk 0..7 per chunk).
 typedef long long i64;
 
-// fp8 tiles stage BK=128 (rows of 128 B): half the barriers of the bf16
-// BK=64 tile at the same LDS footprint — with 64-B rows the kernel was
-// measured BARRIER-bound (86 ms vs 69 ms bf16 Mixtral step).  XOR 16-B
-// slots over 8 positions; b64 fragment reads stay 8-B aligned.
-DEV int swz8(int row, int byte_in_row) {
-  return row * 128 + (byte_in_row ^ ((row & 7) << 4));
-}
-
-__global__ __launch_bounds__(512) void moe_gemm_fp8_kernel(
+template <int BN>
+__global__ __launch_bounds__(512) void moe_gemm_fp8_v2_kernel(
     bf16* __restrict__ out, const unsigned char* __restrict__ xq,
     const float* __restrict__ xs, const unsigned char* __restrict__ wq,
     const float* __restrict__ ws, const int* __restrict__ sorted_ids,
     const int* __restrict__ tile_expert, int N, int K, int gather_div) {
-  const int e = tile_expert[blockIdx.y];      // (n, m) grid — see bf16 note
+  const int e = tile_expert[blockIdx.y];
   if (e < 0) return;
   const int m0 = blockIdx.y * MOE_BM;
-  const int n0 = blockIdx.x * 128;
+  const int n0 = blockIdx.x * BN;
 
-  __shared__ __attribute__((aligned(16))) unsigned char As[128 * 128];
-  __shared__ __attribute__((aligned(16))) unsigned char Bs[128 * 128];
+  extern __shared__ int lds[];     // dynamic; shared decl w/ moe_align
+  unsigned char* ldsb = reinterpret_cast<unsigned char*>(lds);
+  unsigned char* As[2] = {ldsb, ldsb + 32 * 1024};
+  unsigned char* Bs[2] = {ldsb + 64 * 1024, ldsb + 64 * 1024 + BN * 128};
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
@@ -259,75 +335,101 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_kernel(
   const int wave_m = wid & 3;
   const int wave_n = wid >> 2;
 
-  // loaders: 2x16 B per thread cover a 128x128B tile per K step
-  const int ar = tid >> 2;
-  const int ac16 = (tid & 3) * 16;
-  int a_src = -1;
-  if (gather_div > 0) {
-    const int pair = sorted_ids[m0 + ar];
-    if (pair >= 0) a_src = pair / gather_div;
-  } else {
-    a_src = m0 + ar;
+  const int lrow = lane >> 3;
+  const int gbyte = ((lane & 7) * 16) ^ (lrow << 4);
+  const unsigned char* a_src[4];
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int r = wid * 32 + j * 8 + lrow;
+    long row = 0;
+    if (gather_div > 0) {
+      const int pair = sorted_ids[m0 + r];
+      if (pair >= 0) row = pair / gather_div;
+    } else {
+      row = m0 + r;
+    }
+    a_src[j] = xq + row * (long)K + gbyte;
+  }
+  const int b_glds = BN / 64;
+  const unsigned char* b_src[4];
+  #pragma unroll
+  for (int j = 0; j < b_glds; ++j) {
+    const int r = wid * (BN / 8) + j * 8 + lrow;
+    b_src[j] = wq + ((long)e * N + n0 + r) * (long)K + gbyte;
   }
 
-  f32x16 acc[2];
+  constexpr int NB = BN / 2 / 32;
+  f32x16 acc[2][NB];
   #pragma unroll
-  for (int nb = 0; nb < 2; ++nb)
+  for (int sm = 0; sm < 2; ++sm)
     #pragma unroll
-    for (int i = 0; i < 16; ++i) acc[nb][i] = 0.f;
+    for (int nb = 0; nb < NB; ++nb)
+      #pragma unroll
+      for (int i = 0; i < 16; ++i) acc[sm][nb][i] = 0.f;
 
-  const unsigned char* ap0 = a_src >= 0 ? xq + (long)a_src * K + ac16
-                                        : nullptr;
-  const unsigned char* bp0 = wq + ((long)e * N + n0 + ar) * K + ac16;
+  auto stage = [&](int buf, int k0) {
+    const long cb = (long)k0;          // 1 B per fp8 element
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+      glds16(a_src[j] + cb, As[buf] + wid * 4096 + j * 1024);
+    #pragma unroll
+    for (int j = 0; j < b_glds; ++j)
+      glds16(b_src[j] + cb, Bs[buf] + wid * (BN * 16) + j * 1024);
+  };
 
-  for (int k0 = 0; k0 < K; k0 += 128) {
-    uint4 av0 = uint4{0, 0, 0, 0}, av1 = uint4{0, 0, 0, 0};
-    if (ap0) {
-      av0 = *reinterpret_cast<const uint4*>(ap0 + k0);
-      av1 = *reinterpret_cast<const uint4*>(ap0 + k0 + 64);
-    }
-    uint4 bv0 = *reinterpret_cast<const uint4*>(bp0 + k0);
-    uint4 bv1 = *reinterpret_cast<const uint4*>(bp0 + k0 + 64);
-    *reinterpret_cast<uint4*>(&As[swz8(ar, ac16)]) = av0;
-    *reinterpret_cast<uint4*>(&As[swz8(ar, ac16 + 64)]) = av1;
-    *reinterpret_cast<uint4*>(&Bs[swz8(ar, ac16)]) = bv0;
-    *reinterpret_cast<uint4*>(&Bs[swz8(ar, ac16 + 64)]) = bv1;
-    __syncthreads();
+  stage(0, 0);
+  __syncthreads();
+
+  const int nchunks = K >> 7;          // 128 fp8 elems per chunk
+  for (int c = 0; c < nchunks; ++c) {
+    const int cur = c & 1;
+    if (c + 1 < nchunks) stage(cur ^ 1, (c + 1) << 7);
+    const unsigned char* A = As[cur];
+    const unsigned char* B = Bs[cur];
     #pragma unroll
     for (int kk = 0; kk < 8; ++kk) {
-      i64 af = *reinterpret_cast<const i64*>(
-          &As[swz8(wave_m * 32 + lcol, kk * 16 + lhalf * 8)]);
+      i64 af[2];
       #pragma unroll
-      for (int nb = 0; nb < 2; ++nb) {
+      for (int sm = 0; sm < 2; ++sm)
+        af[sm] = *reinterpret_cast<const i64*>(
+            &A[swz(wave_m * 64 + sm * 32 + lcol,
+                   kk * 16 + lhalf * 8)]);
+      #pragma unroll
+      for (int nb = 0; nb < NB; ++nb) {
         i64 bf = *reinterpret_cast<const i64*>(
-            &Bs[swz8(wave_n * 64 + nb * 32 + lcol,
-                     kk * 16 + lhalf * 8)]);
-        acc[nb] = __builtin_amdgcn_mfma_f32_32x32x16_fp8_fp8(
-            af, bf, acc[nb], 0, 0, 0);
+            &B[swz(wave_n * (BN / 2) + nb * 32 + lcol,
+                   kk * 16 + lhalf * 8)]);
+        #pragma unroll
+        for (int sm = 0; sm < 2; ++sm)
+          acc[sm][nb] = __builtin_amdgcn_mfma_f32_32x32x16_fp8_fp8(
+              af[sm], bf, acc[sm][nb], 0, 0, 0);
       }
     }
     __syncthreads();
   }
+
   // epilogue: scale by a_scale[row] * b_scale[e, col], store bf16
   #pragma unroll
-  for (int nb = 0; nb < 2; ++nb) {
-    const int col = n0 + wave_n * 64 + nb * 32 + lcol;
-    const float bs = ws[(long)e * N + col];
+  for (int sm = 0; sm < 2; ++sm)
     #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int m = (r & 3) + 8 * (r >> 2) + 4 * lhalf;
-      const int p = m0 + wave_m * 32 + m;
-      float as = 1.f;
-      if (gather_div > 0) {
-        const int pair = sorted_ids[p];
-        as = pair >= 0 ? xs[pair / gather_div] : 0.f;
-      } else {
-        as = xs[p];
+    for (int nb = 0; nb < NB; ++nb) {
+      const int col = n0 + wave_n * (BN / 2) + nb * 32 + lcol;
+      const float bs = ws[(long)e * N + col];
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int m = (r & 3) + 8 * (r >> 2) + 4 * lhalf;
+        const int p = m0 + wave_m * 64 + sm * 32 + m;
+        float as = 1.f;
+        if (gather_div > 0) {
+          const int pair = sorted_ids[p];
+          as = pair >= 0 ? xs[pair / gather_div] : 0.f;
+        } else {
+          as = xs[p];
+        }
+        *(unsigned short*)(out + (long)p * N + col) =
+            f2bf(acc[sm][nb][r] * as * bs);
       }
-      *(unsigned short*)(out + (long)p * N + col) =
-          f2bf(acc[nb][r] * as * bs);
     }
-  }
 }
 
 void launch_moe_gemm_fp8(bf16* out, const unsigned char* xq,
@@ -335,9 +437,33 @@ void launch_moe_gemm_fp8(bf16* out, const unsigned char* xq,
                          const float* ws, const int* sorted_ids,
                          const int* tile_expert, int ntiles_max, int N,
                          int K, int gather_div, hipStream_t stream) {
-  if (N % 128 || K % 128)
-    throw std::runtime_error("moe_gemm_fp8: N%128 or K%128 != 0");
-  dim3 grid((unsigned)(N / 128), (unsigned)ntiles_max);
-  moe_gemm_fp8_kernel<<<grid, 512, 0, stream>>>(
-      out, xq, xs, wq, ws, sorted_ids, tile_expert, N, K, gather_div);
+  if (K % 128)
+    throw std::runtime_error("moe_gemm_fp8: K % 128 != 0");
+  const bool wide = (N % 256 == 0) &&
+                    (long)(N / 256) * ntiles_max >= 512;
+  static bool attr_set[2] = {false, false};
+  if (wide) {
+    const int lds_bytes = 2 * (32 * 1024 + 256 * 128);
+    if (!attr_set[0]) {
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&moe_gemm_fp8_v2_kernel<256>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
+      attr_set[0] = true;
+    }
+    dim3 grid((unsigned)(N / 256), (unsigned)ntiles_max);
+    moe_gemm_fp8_v2_kernel<256><<<grid, 512, lds_bytes, stream>>>(
+        out, xq, xs, wq, ws, sorted_ids, tile_expert, N, K, gather_div);
+  } else {
+    if (N % 128) throw std::runtime_error("moe_gemm_fp8: N % 128 != 0");
+    const int lds_bytes = 2 * (32 * 1024 + 128 * 128);
+    if (!attr_set[1]) {
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&moe_gemm_fp8_v2_kernel<128>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
+      attr_set[1] = true;
+    }
+    dim3 grid((unsigned)(N / 128), (unsigned)ntiles_max);
+    moe_gemm_fp8_v2_kernel<128><<<grid, 512, lds_bytes, stream>>>(
+        out, xq, xs, wq, ws, sorted_ids, tile_expert, N, K, gather_div);
+  }
 }

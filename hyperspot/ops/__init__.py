@@ -203,7 +203,7 @@ def silu_mul_q(gate_up: torch.Tensor) -> QTensor:
     return QTensor(out, scales)
 
 
-MOE_BM = 128  # sorted-pair tile granularity of the MoE kernels (csrc/moe.hip)
+MOE_BM = 256  # sorted-pair tile granularity of the MoE kernels (csrc/moe.hip)
 
 
 def moe_ffn(x: torch.Tensor, w13: torch.Tensor, w2: torch.Tensor,
