@@ -1,12 +1,17 @@
 #include "auth.h"
 
 #include <openssl/bio.h>
+#include <openssl/bn.h>
 #include <openssl/evp.h>
 #include <openssl/hmac.h>
 #include <openssl/pem.h>
+#include <openssl/rsa.h>
 
+#include <chrono>
 #include <ctime>
 #include <vector>
+
+#include "../http/client.h"
 
 namespace hs {
 
@@ -92,6 +97,162 @@ bool const_eq(const std::string& a, const std::string& b) {
 
 }  // namespace
 
+// ------------------------------------------------------------ JwksCache
+
+namespace {
+
+// parse http://host[:port]/path (plain HTTP only — see header note)
+bool parse_http_url(const std::string& url, std::string* host, int* port,
+                    std::string* path) {
+  if (url.rfind("http://", 0) != 0) return false;
+  std::string rest = url.substr(7);
+  size_t slash = rest.find('/');
+  std::string hp = slash == std::string::npos ? rest : rest.substr(0, slash);
+  *path = slash == std::string::npos ? "/" : rest.substr(slash);
+  size_t colon = hp.find(':');
+  if (colon == std::string::npos) {
+    *host = hp;
+    *port = 80;
+  } else {
+    *host = hp.substr(0, colon);
+    *port = atoi(hp.c_str() + colon + 1);
+  }
+  return !host->empty() && *port > 0;
+}
+
+// JWK (kty=RSA, base64url n/e) -> PEM SubjectPublicKeyInfo
+#pragma GCC diagnostic push
+#pragma GCC diagnostic ignored "-Wdeprecated-declarations"
+std::string jwk_rsa_to_pem(const std::string& n_b64,
+                           const std::string& e_b64) {
+  auto n_raw = b64url_decode(n_b64);
+  auto e_raw = b64url_decode(e_b64);
+  if (!n_raw || !e_raw || n_raw->empty() || e_raw->empty()) return "";
+  BIGNUM* n = BN_bin2bn((const unsigned char*)n_raw->data(),
+                        (int)n_raw->size(), nullptr);
+  BIGNUM* e = BN_bin2bn((const unsigned char*)e_raw->data(),
+                        (int)e_raw->size(), nullptr);
+  RSA* rsa = RSA_new();
+  if (!n || !e || !rsa || RSA_set0_key(rsa, n, e, nullptr) != 1) {
+    if (rsa) RSA_free(rsa); else { BN_free(n); BN_free(e); }
+    return "";
+  }
+  EVP_PKEY* pkey = EVP_PKEY_new();
+  EVP_PKEY_assign_RSA(pkey, rsa);      // pkey owns rsa (and n/e)
+  BIO* bio = BIO_new(BIO_s_mem());
+  std::string pem;
+  if (PEM_write_bio_PUBKEY(bio, pkey) == 1) {
+    char* data = nullptr;
+    long len = BIO_get_mem_data(bio, &data);
+    pem.assign(data, (size_t)len);
+  }
+  BIO_free(bio);
+  EVP_PKEY_free(pkey);
+  return pem;
+}
+#pragma GCC diagnostic pop
+
+std::optional<Json> http_get_json(const std::string& url,
+                                  std::string* err) {
+  std::string host, path;
+  int port = 0;
+  if (!parse_http_url(url, &host, &port, &path)) {
+    if (err) *err = "bad url: " + url;
+    return std::nullopt;
+  }
+  auto resp = http_request(host, port, "GET", path,
+                           {{"accept", "application/json"}}, "", 5000);
+  if (!resp || resp->status != 200) {
+    if (err)
+      *err = "fetch failed: " + url + " status " +
+             std::to_string(resp ? resp->status : 0);
+    return std::nullopt;
+  }
+  try {
+    return Json::parse(resp->body);
+  } catch (...) {
+    if (err) *err = "bad JSON from " + url;
+    return std::nullopt;
+  }
+}
+
+double mono_s() {
+  return std::chrono::duration<double>(
+             std::chrono::steady_clock::now().time_since_epoch())
+      .count();
+}
+
+}  // namespace
+
+bool JwksCache::refresh(std::string* err) {
+  // caller holds mu_
+  last_attempt_ = mono_s();
+  if (jwks_uri.empty()) {
+    auto disc = http_get_json(discovery_url, err);
+    if (!disc) return false;
+    jwks_uri = disc->at("jwks_uri").as_string();
+    issuer_ = disc->at("issuer").as_string();
+    if (jwks_uri.empty()) {
+      if (err) *err = "discovery document has no jwks_uri";
+      return false;
+    }
+  }
+  auto jwks = http_get_json(jwks_uri, err);
+  if (!jwks) return false;
+  const Json& keys = jwks->at("keys");
+  if (!keys.is_array()) {
+    if (err) *err = "JWKS has no keys[]";
+    return false;
+  }
+  std::map<std::string, std::string> fresh;
+  for (auto& k : keys.arr()) {
+    if (k.at("kty").as_string() != "RSA") continue;
+    const std::string use = k.at("use").as_string("sig");
+    if (use != "sig") continue;
+    const std::string alg = k.at("alg").as_string("RS256");
+    if (alg != "RS256") continue;
+    const std::string kid = k.at("kid").as_string();
+    std::string pem = jwk_rsa_to_pem(k.at("n").as_string(),
+                                     k.at("e").as_string());
+    if (!kid.empty() && !pem.empty()) fresh[kid] = pem;
+  }
+  keys_ = std::move(fresh);
+  fetched_at_ = mono_s();
+  return true;
+}
+
+std::optional<std::string> JwksCache::key_for(const std::string& kid,
+                                              std::string* err) {
+  std::lock_guard<std::mutex> lk(mu_);
+  const double now = mono_s();
+  auto it = keys_.find(kid);
+  if (it != keys_.end() && now - fetched_at_ < ttl_s) return it->second;
+  // stale hit, or miss (possible key rotation): re-fetch with cool-down
+  if (now - last_attempt_ >= rotate_cooldown_s || fetched_at_ == 0) {
+    std::string ferr;
+    if (!refresh(&ferr)) {
+      // keep serving a stale hit if the IdP is briefly unreachable
+      if (it != keys_.end()) return it->second;
+      if (err) *err = ferr;
+      return std::nullopt;
+    }
+  }
+  it = keys_.find(kid);
+  if (it != keys_.end()) return it->second;
+  if (err) *err = "unknown kid: " + kid;
+  return std::nullopt;
+}
+
+std::string JwksCache::discovered_issuer() {
+  std::lock_guard<std::mutex> lk(mu_);
+  if (issuer_.empty() && !discovery_url.empty() &&
+      mono_s() - last_attempt_ >= rotate_cooldown_s) {
+    std::string err;
+    refresh(&err);
+  }
+  return issuer_;
+}
+
 std::optional<SecurityContext> JwtValidator::validate(
     const std::string& token, std::string* err) const {
   auto fail = [&](const char* m) {
@@ -122,8 +283,19 @@ std::optional<SecurityContext> JwtValidator::validate(
     if (!const_eq(std::string((char*)mac, 32), *sig_raw))
       return fail("bad signature");
   } else if (alg == "RS256") {
-    if (rs256_public_pem.empty()) return fail("RS256 not configured");
-    if (!rs256_verify(rs256_public_pem, signed_part, *sig_raw))
+    std::string pem = rs256_public_pem;
+    const std::string kid = hdr.at("kid").as_string();
+    if (jwks && jwks->configured() && !kid.empty()) {
+      std::string jerr;
+      auto k = jwks->key_for(kid, &jerr);
+      if (!k) {
+        if (err) *err = "jwks: " + jerr;
+        return std::nullopt;
+      }
+      pem = *k;
+    }
+    if (pem.empty()) return fail("RS256 not configured");
+    if (!rs256_verify(pem, signed_part, *sig_raw))
       return fail("bad signature");
   } else {
     return fail("unsupported alg");   // incl. alg=none — always rejected
@@ -133,7 +305,9 @@ std::optional<SecurityContext> JwtValidator::validate(
     return fail("token expired");
   if (pl.contains("nbf") && now + leeway_s < pl.at("nbf").as_int(0))
     return fail("token not yet valid");
-  if (!issuer.empty() && pl.at("iss").as_string() != issuer)
+  std::string want_iss = issuer;
+  if (want_iss.empty() && jwks) want_iss = jwks->discovered_issuer();
+  if (!want_iss.empty() && pl.at("iss").as_string() != want_iss)
     return fail("wrong issuer");
   if (!audience.empty()) {
     const Json& aud = pl.at("aud");

@@ -10,12 +10,47 @@
 // scope/scp -> scopes.
 #pragma once
 
+#include <map>
+#include <memory>
+#include <mutex>
 #include <optional>
 #include <string>
 
 #include "modkit.h"
 
 namespace hs {
+
+// JWKS endpoint fetch + cache with key rotation (reference:
+// libs/modkit-auth/src/providers/jwks.rs).  Keys are resolved by `kid`;
+// an unknown kid triggers ONE re-fetch (rotation) with a cool-down so a
+// flood of bad tokens cannot hammer the IdP.  `discovery_url` points at
+// an OIDC discovery document (/.well-known/openid-configuration,
+// libs/modkit-auth/src/oauth2/discovery.rs) and resolves jwks_uri + the
+// expected issuer.  Plain HTTP (in-cluster IdP / TLS-terminating
+// sidecar — this host's outbound client has no TLS).
+class JwksCache {
+ public:
+  std::string jwks_uri;            // direct JWKS endpoint, or
+  std::string discovery_url;       // resolve jwks_uri + issuer via OIDC
+  int ttl_s = 300;                 // full-refresh interval
+  int rotate_cooldown_s = 2;       // min gap between miss-driven fetches
+
+  bool configured() const {
+    return !jwks_uri.empty() || !discovery_url.empty();
+  }
+  // PEM public key for `kid`; refreshes on stale cache or unknown kid
+  std::optional<std::string> key_for(const std::string& kid,
+                                     std::string* err = nullptr);
+  // issuer from the discovery document ("" until discovered)
+  std::string discovered_issuer();
+
+ private:
+  bool refresh(std::string* err);
+  std::mutex mu_;
+  std::map<std::string, std::string> keys_;   // kid -> PEM
+  std::string issuer_;
+  double fetched_at_ = 0, last_attempt_ = 0;
+};
 
 struct JwtValidator {
   std::string hs256_secret;       // enables HS256 when non-empty
@@ -24,9 +59,11 @@ struct JwtValidator {
   std::string audience;           // checked when non-empty
   std::string tenant_claim = "tid";
   int leeway_s = 30;
+  std::shared_ptr<JwksCache> jwks;  // enables RS256-by-kid when set
 
   bool configured() const {
-    return !hs256_secret.empty() || !rs256_public_pem.empty();
+    return !hs256_secret.empty() || !rs256_public_pem.empty() ||
+           (jwks && jwks->configured());
   }
   // nullopt + err set on any failure (never throws)
   std::optional<SecurityContext> validate(const std::string& token,

@@ -110,6 +110,18 @@ class StaticAuthnResolver : public AuthnResolverClient {
       jwt_.audience = jw.at("audience").as_string("");
       jwt_.tenant_claim = jw.at("tenant_claim").as_string("tid");
       jwt_.leeway_s = (int)jw.at("leeway_s").as_int(30);
+      // real token-validation plane: JWKS endpoint / OIDC discovery
+      // (libs/modkit-auth/src/providers/jwks.rs, oauth2/discovery.rs)
+      const std::string jwks_uri = jw.at("jwks_uri").as_string("");
+      const std::string disc =
+          jw.at("oidc_discovery_url").as_string("");
+      if (!jwks_uri.empty() || !disc.empty()) {
+        auto cache = std::make_shared<JwksCache>();
+        cache->jwks_uri = jwks_uri;
+        cache->discovery_url = disc;
+        cache->ttl_s = (int)jw.at("jwks_ttl_s").as_int(300);
+        jwt_.jwks = cache;
+      }
     }
     // static-authn-plugin shape (plugins/static-authn-plugin/src/config.rs):
     // {tokens: [{token, subject_id, subject_tenant_id, subject_type?,
