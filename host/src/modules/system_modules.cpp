@@ -1,5 +1,7 @@
 #include "system_modules.h"
 
+#include "../modkit/db.h"
+
 #include "../modkit/auth.h"
 
 #include <signal.h>
@@ -552,42 +554,209 @@ void ModelRegistryModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
 
 namespace {
 
-class InMemoryCredStore : public CredStoreClient {
+// sqlite-backed tenant-scoped secret store with sharing modes
+// private|tenant|shared and hierarchical inheritance
+// (reference modules/credstore/docs/DESIGN.md:295-339):
+//   private  — owner(subject)-only, scoped per owner (never conflicts
+//              across owners)
+//   tenant   — tenant-wide (the default)
+//   shared   — visible to DESCENDANT tenants via ancestor resolution;
+//              child tenants may shadow by creating the same reference
+// Inaccessible secrets are ALWAYS 404 (anti-enumeration, DESIGN.md:336);
+// REST GET returns metadata (sharing, owner_tenant_id, is_inherited),
+// never the value — the OAGW credential injector is the only consumer
+// of secret material (documented deviation, stricter than the spec).
+class SqlCredStore : public CredStoreClient {
  public:
+  SqlCredStore(std::unique_ptr<Db> db, ClientHub* hub)
+      : db_(std::move(db)), hub_(hub) {}
+
+  struct Meta {
+    std::string value, sharing, owner_tenant;
+    bool is_inherited = false;
+  };
+
+  std::optional<Meta> resolve(const std::string& tenant,
+                              const std::string& subject,
+                              const std::string& ref) {
+    {
+      std::lock_guard<std::mutex> lk(db_->mu());
+      // 1) caller's own private secret
+      if (!subject.empty()) {
+        auto r = db_->query(
+            "SELECT value, sharing, tenant_id FROM secrets WHERE "
+            "tenant_id=? AND scope_owner=? AND ref=?",
+            {DbValue::S(tenant), DbValue::S(subject), DbValue::S(ref)});
+        if (!r.empty()) return mk(r[0], false);
+      }
+      // 2) tenant-wide secret in the caller's tenant
+      auto r = db_->query(
+          "SELECT value, sharing, tenant_id FROM secrets WHERE "
+          "tenant_id=? AND scope_owner='' AND ref=?",
+          {DbValue::S(tenant), DbValue::S(ref)});
+      if (!r.empty()) return mk(r[0], false);
+    }
+    // 3) hierarchical: nearest ancestor's `shared` secret
+    auto tr = hub_ ? hub_->get<TenantResolverClient>("tenant-resolver")
+                   : nullptr;
+    if (tr) {
+      for (auto& anc : tr->get_ancestors(tenant)) {
+        std::lock_guard<std::mutex> lk(db_->mu());
+        auto r = db_->query(
+            "SELECT value, sharing, tenant_id FROM secrets WHERE "
+            "tenant_id=? AND scope_owner='' AND ref=? AND "
+            "sharing='shared'",
+            {DbValue::S(anc.id), DbValue::S(ref)});
+        if (!r.empty()) return mk(r[0], true);
+      }
+    }
+    return std::nullopt;
+  }
+
+  // create-only (POST): false when the reference already exists in the
+  // same scope (per-owner for private, per-tenant otherwise)
+  bool create(const std::string& tenant, const std::string& subject,
+              const std::string& ref, const std::string& value,
+              const std::string& sharing) {
+    const std::string owner = sharing == "private" ? subject : "";
+    std::lock_guard<std::mutex> lk(db_->mu());
+    auto r = db_->query(
+        "SELECT 1 FROM secrets WHERE tenant_id=? AND scope_owner=? AND "
+        "ref=?",
+        {DbValue::S(tenant), DbValue::S(owner), DbValue::S(ref)});
+    if (!r.empty()) return false;
+    db_->query(
+        "INSERT INTO secrets (tenant_id, scope_owner, ref, value, "
+        "sharing, updated_at) VALUES (?,?,?,?,?,datetime('now'))",
+        {DbValue::S(tenant), DbValue::S(owner), DbValue::S(ref),
+         DbValue::S(value), DbValue::S(sharing)});
+    return true;
+  }
+
+  // upsert (PUT): updates the caller's secret in-scope, creating it
+  // with the given sharing when absent
+  void upsert(const std::string& tenant, const std::string& subject,
+              const std::string& ref, const std::string& value,
+              const std::string& sharing) {
+    const std::string owner = sharing == "private" ? subject : "";
+    std::lock_guard<std::mutex> lk(db_->mu());
+    db_->query(
+        "INSERT INTO secrets (tenant_id, scope_owner, ref, value, "
+        "sharing, updated_at) VALUES (?,?,?,?,?,datetime('now')) "
+        "ON CONFLICT(tenant_id, scope_owner, ref) DO UPDATE SET "
+        "value=excluded.value, sharing=excluded.sharing, "
+        "updated_at=excluded.updated_at",
+        {DbValue::S(tenant), DbValue::S(owner), DbValue::S(ref),
+         DbValue::S(value), DbValue::S(sharing)});
+  }
+
+  // delete own-scope only: an INHERITED secret is not deletable from a
+  // child tenant (404 — the child shadows instead)
+  bool remove(const std::string& tenant, const std::string& subject,
+              const std::string& ref) {
+    std::lock_guard<std::mutex> lk(db_->mu());
+    db_->query(
+        "DELETE FROM secrets WHERE tenant_id=? AND ref=? AND "
+        "(scope_owner='' OR scope_owner=?)",
+        {DbValue::S(tenant), DbValue::S(ref), DbValue::S(subject)});
+    return db_->changes() > 0;
+  }
+
+  // ---- CredStoreClient (OAGW injector: tenant-plane resolution) ----
   std::optional<std::string> get(const std::string& tenant,
                                  const std::string& ref) override {
-    std::lock_guard<std::mutex> lk(mu_);
-    auto it = store_.find(tenant + "\x1f" + ref);
-    if (it == store_.end()) return std::nullopt;
-    return it->second;
+    auto m = resolve(tenant, "", ref);
+    if (!m) return std::nullopt;
+    return m->value;
   }
   void put(const std::string& tenant, const std::string& ref,
            const std::string& value) override {
-    std::lock_guard<std::mutex> lk(mu_);
-    store_[tenant + "\x1f" + ref] = value;
+    upsert(tenant, "", ref, value, "tenant");
   }
   bool del(const std::string& tenant, const std::string& ref) override {
-    std::lock_guard<std::mutex> lk(mu_);
-    return store_.erase(tenant + "\x1f" + ref) > 0;
+    return remove(tenant, "", ref);
   }
 
  private:
-  std::mutex mu_;
-  std::map<std::string, std::string> store_;
+  static Meta mk(const DbRow& r, bool inherited) {
+    Meta m;
+    m.value = r.at("value").as_string();
+    m.sharing = r.at("sharing").as_string();
+    m.owner_tenant = r.at("tenant_id").as_string();
+    m.is_inherited = inherited;
+    return m;
+  }
+  std::unique_ptr<Db> db_;
+  ClientHub* hub_;
 };
 
-std::shared_ptr<InMemoryCredStore> g_creds;
+std::shared_ptr<SqlCredStore> g_creds;
 
 }  // namespace
 
 void CredStoreModule::init(ModuleCtx& ctx) {
-  g_creds = std::make_shared<InMemoryCredStore>();
+  std::string file = ctx.full_config
+                         .path("modules.credstore.database.file")
+                         .as_string("");
+  if (file.empty()) {
+    std::string home = ctx.home_dir;
+    if (!home.empty() && home[0] == '~') {
+      const char* h = getenv("HOME");
+      home = std::string(h ? h : "/tmp") + home.substr(1);
+    }
+    mkdir(home.c_str(), 0755);
+    file = home + "/credstore.db";
+  }
+  auto db = std::make_unique<Db>(file);
+  db->migrate("credstore", {
+      {"0001_secrets",
+       "CREATE TABLE secrets ("
+       "  tenant_id TEXT NOT NULL,"
+       "  scope_owner TEXT NOT NULL DEFAULT '',"   // subject for private
+       "  ref TEXT NOT NULL,"
+       "  value TEXT NOT NULL,"
+       "  sharing TEXT NOT NULL DEFAULT 'tenant',"
+       "  updated_at TEXT NOT NULL,"
+       "  UNIQUE (tenant_id, scope_owner, ref))"},
+  });
+  g_creds = std::make_shared<SqlCredStore>(std::move(db), ctx.hub);
   ctx.hub->register_client<CredStoreClient>("credstore", g_creds);
 }
 
 void CredStoreModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
-  // REST per reference DESIGN.md:283-289; tenant always from SecurityCtx;
-  // inaccessible secrets are ALWAYS 404 (anti-enumeration, DESIGN.md:336).
+  // REST per reference DESIGN.md:283-339; tenant always from SecurityCtx;
+  // inaccessible secrets are ALWAYS 404 (anti-enumeration).
+  auto sharing_of = [](const Json& body) -> std::string {
+    const std::string sh = body.at("sharing").as_string("tenant");
+    if (sh != "private" && sh != "tenant" && sh != "shared")
+      throw Problem::bad_request(
+          "'sharing' must be private|tenant|shared");
+    return sh;
+  };
+
+  OperationSpec post;
+  post.method = "POST";
+  post.path = "/credstore/v1/secrets";
+  post.operation_id = "credstore_create";
+  post.summary = "Create a secret (409 if the reference exists in scope)";
+  post.authenticated = true;
+  post.allowed_content_types = {"application/json"};
+  post.tags = {"credstore"};
+  rest.register_op(post, [sharing_of](HttpRequest& rq, ResponseWriter& w) {
+    Json body;
+    try { body = Json::parse(rq.body); }
+    catch (...) { throw Problem::bad_request("invalid JSON"); }
+    const std::string ref = body.at("reference").as_string();
+    if (ref.empty() || !body.at("value").is_string())
+      throw Problem::bad_request("'reference' and 'value' required");
+    auto sec = sec_of(rq);
+    if (!g_creds->create(sec.tenant_id, sec.subject_id, ref,
+                         body.at("value").as_string(), sharing_of(body)))
+      throw Problem{409, "Conflict", "about:blank",
+                    "secret already exists in this scope", "conflict"};
+    w.respond(201, "application/json", "{\"created\":true}");
+  });
+
   OperationSpec put;
   put.method = "PUT";
   put.path = "/credstore/v1/secrets/{ref}";
@@ -595,16 +764,18 @@ void CredStoreModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
   put.authenticated = true;
   put.allowed_content_types = {"application/json"};
   put.tags = {"credstore"};
-  rest.register_op(put, [](HttpRequest& rq, ResponseWriter& w) {
+  rest.register_op(put, [sharing_of](HttpRequest& rq, ResponseWriter& w) {
     Json body;
     try { body = Json::parse(rq.body); }
     catch (...) { throw Problem::bad_request("invalid JSON"); }
     if (!body.at("value").is_string())
       throw Problem::bad_request("'value' string required");
-    g_creds->put(sec_of(rq).tenant_id, rq.path_params["ref"],
-                 body.at("value").as_string());
+    auto sec = sec_of(rq);
+    g_creds->upsert(sec.tenant_id, sec.subject_id, rq.path_params["ref"],
+                    body.at("value").as_string(), sharing_of(body));
     w.respond(204, "application/json", "");
   });
+
   OperationSpec get;
   get.method = "GET";
   get.path = "/credstore/v1/secrets/{ref}";
@@ -612,17 +783,24 @@ void CredStoreModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
   get.authenticated = true;
   get.tags = {"credstore"};
   rest.register_op(get, [](HttpRequest& rq, ResponseWriter& w) {
-    auto v = g_creds->get(sec_of(rq).tenant_id, rq.path_params["ref"]);
-    if (!v) throw Problem::not_found();   // never reveal existence
-    // write-only store: secrets are consumed by the OAGW credential
-    // injector, never read back over REST (credstore DESIGN: no secret
-    // logging / no secret egress)
+    auto sec = sec_of(rq);
+    auto m = g_creds->resolve(sec.tenant_id, sec.subject_id,
+                              rq.path_params["ref"]);
+    if (!m) throw Problem::not_found();   // never reveal existence
+    // write-only store: values are consumed by the OAGW credential
+    // injector, never read back over REST
+    Json meta = Json::object();
+    meta["owner_tenant_id"] = m->owner_tenant;
+    meta["sharing"] = m->sharing;
+    meta["is_inherited"] = m->is_inherited;
     Json out = Json::object();
     out["ref"] = rq.path_params["ref"];
     out["exists"] = true;
-    out["value_length"] = (long)v->size();
+    out["value_length"] = (long)m->value.size();
+    out["metadata"] = meta;
     w.respond(200, "application/json", out.dump());
   });
+
   OperationSpec del;
   del.method = "DELETE";
   del.path = "/credstore/v1/secrets/{ref}";
@@ -630,7 +808,9 @@ void CredStoreModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
   del.authenticated = true;
   del.tags = {"credstore"};
   rest.register_op(del, [](HttpRequest& rq, ResponseWriter& w) {
-    if (!g_creds->del(sec_of(rq).tenant_id, rq.path_params["ref"]))
+    auto sec = sec_of(rq);
+    if (!g_creds->remove(sec.tenant_id, sec.subject_id,
+                         rq.path_params["ref"]))
       throw Problem::not_found();
     w.respond(204, "application/json", "");
   });

@@ -347,6 +347,9 @@ modules:
         - token: "acme-token"
           subject_id: "acme-user"
           subject_tenant_id: "11111111-1111-1111-1111-111111111111"
+        - token: "acme-token-2"
+          subject_id: "acme-user-2"
+          subject_tenant_id: "11111111-1111-1111-1111-111111111111"
   llm-gateway:
     config:
       model: "tiny-llama"
