@@ -88,35 +88,30 @@ DEV void glds16(const void* gp, void* lp) {
 
 // out[P, N] (sorted space) = gather(x) @ W[e]^T ; W is [E, N, K] row-major.
 //
-// v2 (guide §5 "canonical CDNA GEMM" + glds table): 256(M) x BN(N) tile
-// per 512-thread block, BK = 64 bf16 (128-B rows), async global->LDS
-// (global_load_lds_dwordx4) into a 2-deep LDS ring, one block/CU.  The
-// 128x128 single-buffered v1 capped at ~620 TF (guide's own number for
-// that shape) = 23.7% MfmaUtil measured; the 256-row tile additionally
-// HALVES the expert-weight re-read (ceil(tokens_e/256) streams of each
-// [N,K] expert matrix per layer — the binding constraint at bs1024,
-// profiles/r01_decode_v3.md).
-//
-// Per k-chunk each wave issues its 1/8 share of the A+B stage as glds
-// (8-row x 128-B pieces, XOR pre-swizzle on the source byte), computes
-// the CURRENT chunk from the other LDS buffer, then __syncthreads()
-// (hipcc inserts the vmcnt(0) drain before the barrier — the guide's
-// recommended 2-buffer recipe).
-template <int BN>
-__global__ __launch_bounds__(512) void moe_gemm_v2_kernel(
+// v3 (guide §5 glds table, "3-buf span" row): 256(M) x 128(N) tile per
+// 512-thread block, BK = 64 bf16 (128-B staged rows), async global->LDS
+// (global_load_lds_dwordx4) into a 3-DEEP LDS ring with COUNTED
+// s_waitcnt vmcnt(N) + raw s_barrier, so one whole chunk stays in
+// flight across each barrier.  6 glds per wave per chunk (4 A-pieces +
+// 2 B-pieces of 8 rows x 128 B, source-address XOR swizzle).  The
+// binding constraint is the per-expert weight stream
+// (profiles/r01_decode_v3.md): B traffic = ntiles * N * K * 2 is
+// INVARIANT to the column-tile width, so BN=128 (acc 64 VGPR) buys the
+// LDS headroom for depth 3: 3 x (32 KiB A + 16 KiB B) = 144 KiB.
+// 256-row tiles halve the expert re-read vs the 128-row v1 kernel.
+__global__ __launch_bounds__(512) void moe_gemm_v3_kernel(
     bf16* __restrict__ out, const bf16* __restrict__ x,
     const bf16* __restrict__ w, const int* __restrict__ sorted_ids,
     const int* __restrict__ tile_expert, int N, int K, int gather_div) {
+  constexpr int BN = 128;
   const int e = tile_expert[blockIdx.y];
   if (e < 0) return;
   const int m0 = blockIdx.y * MOE_BM;
   const int n0 = blockIdx.x * BN;
 
-  // dynamic LDS addressed by INTEGER offsets into the extern array so
-  // the compiler keeps the AS3 provenance (a pointer array here decays
-  // to generic AS => fragment reads become flat_load + vmcnt waits in
-  // the MFMA stream — measured 13% slower than the v1 kernel)
-  // layout: A0 | A1 | B0 | B1  (A: 32 KiB each, B: BN*128 each)
+  // dynamic LDS addressed by INTEGER offsets (pointer indirection decays
+  // to generic AS => flat_load + vmcnt stalls in the MFMA stream)
+  // ring layout: A0 A1 A2 | B0 B1 B2
   extern __shared__ int lds[];     // shared decl w/ moe_align
   unsigned char* dyn_lds = reinterpret_cast<unsigned char*>(lds);
 
@@ -126,15 +121,10 @@ __global__ __launch_bounds__(512) void moe_gemm_v2_kernel(
   const int lcol = lane & 31;
   const int lhalf = lane >> 5;
   const int wave_m = wid & 3;   // 4 row-blocks of 64
-  const int wave_n = wid >> 2;  // 2 col-blocks of BN/2
+  const int wave_n = wid >> 2;  // 2 col-blocks of 64
 
-  // ---- loader geometry: wave wid fills A rows [wid*32, wid*32+32) in 4
-  // glds of 8 rows; lane covers row lrow of its piece at swizzled source
-  // byte gbyte (lane-linear LDS => pre-swizzled global address)
   const int lrow = lane >> 3;           // 0..7
   const int gbyte = ((lane & 7) * 16) ^ (lrow << 4);
-  // per-j A source rows (gathered); padded rows read x row 0 (their
-  // outputs are dropped by combine) so the glds always has a valid VA
   const unsigned char* a_src[4];
   #pragma unroll
   for (int j = 0; j < 4; ++j) {
@@ -149,47 +139,37 @@ __global__ __launch_bounds__(512) void moe_gemm_v2_kernel(
     a_src[j] = reinterpret_cast<const unsigned char*>(x) +
                row * (long)K * 2 + gbyte;
   }
-  // B rows: wave wid covers BN/8 rows in BN/64 glds
-  const int b_glds = BN / 64;
-  const unsigned char* b_src[4];
+  const unsigned char* b_src[2];
   #pragma unroll
-  for (int j = 0; j < b_glds; ++j) {
-    const int r = wid * (BN / 8) + j * 8 + lrow;
+  for (int j = 0; j < 2; ++j) {
+    const int r = wid * 16 + j * 8 + lrow;
     b_src[j] = reinterpret_cast<const unsigned char*>(w) +
                ((long)e * N + n0 + r) * (long)K * 2 + gbyte;
   }
 
-  constexpr int NB = BN / 2 / 32;     // 32-col mfma blocks per wave
-  f32x16 acc[2][NB];
+  f32x16 acc[2][2];
   #pragma unroll
   for (int sm = 0; sm < 2; ++sm)
     #pragma unroll
-    for (int nb = 0; nb < NB; ++nb)
+    for (int nb = 0; nb < 2; ++nb)
       #pragma unroll
       for (int i = 0; i < 16; ++i) acc[sm][nb][i] = 0.f;
 
-  auto stage = [&](int buf, int k0) {
-    const long cb = (long)k0 * 2;     // chunk byte offset in the row
+  auto stage = [&](int slot, int k0) {
+    const long cb = (long)k0 * 2;
     #pragma unroll
     for (int j = 0; j < 4; ++j)
       glds16(a_src[j] + cb,
-             &dyn_lds[buf * 32768 + wid * 4096 + j * 1024]);
+             &dyn_lds[slot * 32768 + wid * 4096 + j * 1024]);
     #pragma unroll
-    for (int j = 0; j < b_glds; ++j)
+    for (int j = 0; j < 2; ++j)
       glds16(b_src[j] + cb,
-             &dyn_lds[65536 + buf * (BN * 128) + wid * (BN * 16) +
-                      j * 1024]);
+             &dyn_lds[98304 + slot * 16384 + wid * 2048 + j * 1024]);
   };
 
-  stage(0, 0);
-  __syncthreads();                    // vmcnt(0) drain inserted by hipcc
-
-  const int nchunks = K >> 6;
-  for (int c = 0; c < nchunks; ++c) {
-    const int cur = c & 1;
-    if (c + 1 < nchunks) stage(cur ^ 1, (c + 1) << 6);
-    const int a_base = cur * 32768;
-    const int b_base = 65536 + cur * (BN * 128);
+  auto compute = [&](int slot) {
+    const int a_base = slot * 32768;
+    const int b_base = 98304 + slot * 16384;
     #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
       short8 af[2];
@@ -199,9 +179,9 @@ __global__ __launch_bounds__(512) void moe_gemm_v2_kernel(
             &dyn_lds[a_base + swz(wave_m * 64 + sm * 32 + lcol,
                                   kk * 32 + lhalf * 16)]);
       #pragma unroll
-      for (int nb = 0; nb < NB; ++nb) {
+      for (int nb = 0; nb < 2; ++nb) {
         short8 bf = *reinterpret_cast<const short8*>(
-            &dyn_lds[b_base + swz(wave_n * (BN / 2) + nb * 32 + lcol,
+            &dyn_lds[b_base + swz(wave_n * 64 + nb * 32 + lcol,
                                   kk * 32 + lhalf * 16)]);
         #pragma unroll
         for (int sm = 0; sm < 2; ++sm)
@@ -209,20 +189,44 @@ __global__ __launch_bounds__(512) void moe_gemm_v2_kernel(
               af[sm], bf, acc[sm][nb], 0, 0, 0);
       }
     }
-    __syncthreads();                  // also drains the staged c+1 glds
+  };
+
+  const int nchunks = K >> 6;          // K % 64 == 0 (checked at launch)
+  // 3-stage software pipeline: issue c+2, wait for c (two chunks = 12
+  // glds still in flight), raw-barrier, compute c.  The slot being
+  // overwritten by c+2 was read as chunk c-1 BEFORE the previous
+  // barrier, so every wave is done with it.
+  stage(0, 0);
+  stage(1, 64);
+  int c = 0;
+  for (; c + 2 < nchunks; ++c) {
+    stage((c + 2) % 3, (c + 2) << 6);
+    asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    compute(c % 3);
+  }
+  // drain: one chunk in flight, then none
+  asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  compute(c % 3);
+  ++c;
+  if (c < nchunks) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    compute(c % 3);
   }
 
   // ---- epilogue: C row = (reg&3) + 8*(reg>>2) + 4*lhalf, col = lcol ----
   #pragma unroll
   for (int sm = 0; sm < 2; ++sm)
     #pragma unroll
-    for (int nb = 0; nb < NB; ++nb)
+    for (int nb = 0; nb < 2; ++nb)
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int m = (r & 3) + 8 * (r >> 2) + 4 * lhalf;
         *(unsigned short*)(out +
                            (long)(m0 + wave_m * 64 + sm * 32 + m) * N +
-                           n0 + wave_n * (BN / 2) + nb * 32 + lcol) =
+                           n0 + wave_n * 64 + nb * 32 + lcol) =
             f2bf(acc[sm][nb][r]);
       }
 }
@@ -266,37 +270,19 @@ void launch_moe_gemm(bf16* out, const bf16* x, const bf16* w,
                      const int* sorted_ids, const int* tile_expert,
                      int ntiles_max, int N, int K, int gather_div,
                      hipStream_t stream) {
-  if (K % 64)
-    throw std::runtime_error("moe_gemm: K % 64 != 0");
-  // BN=256 when the grid still fills the 256 CUs, else BN=128
-  const bool wide = (N % 256 == 0) &&
-                    (long)(N / 256) * ntiles_max >= 512;
-  static bool attr_set[2] = {false, false};
-  if (wide) {
-    if (N % 256) throw std::runtime_error("moe_gemm: N % 256 != 0");
-    const int lds_bytes = 2 * (32 * 1024 + 256 * 128);
-    if (!attr_set[0]) {
-      (void)hipFuncSetAttribute(
-          reinterpret_cast<const void*>(&moe_gemm_v2_kernel<256>),
-          hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
-      attr_set[0] = true;
-    }
-    dim3 grid((unsigned)(N / 256), (unsigned)ntiles_max);
-    moe_gemm_v2_kernel<256><<<grid, 512, lds_bytes, stream>>>(
-        out, x, w, sorted_ids, tile_expert, N, K, gather_div);
-  } else {
-    if (N % 128) throw std::runtime_error("moe_gemm: N % 128 != 0");
-    const int lds_bytes = 2 * (32 * 1024 + 128 * 128);
-    if (!attr_set[1]) {
-      (void)hipFuncSetAttribute(
-          reinterpret_cast<const void*>(&moe_gemm_v2_kernel<128>),
-          hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
-      attr_set[1] = true;
-    }
-    dim3 grid((unsigned)(N / 128), (unsigned)ntiles_max);
-    moe_gemm_v2_kernel<128><<<grid, 512, lds_bytes, stream>>>(
-        out, x, w, sorted_ids, tile_expert, N, K, gather_div);
+  if (K % 128 || N % 128)
+    throw std::runtime_error("moe_gemm: K%128 or N%128 != 0");
+  const int lds_bytes = 3 * (32 * 1024 + 16 * 1024);   // 144 KiB ring
+  static bool attr_set = false;
+  if (!attr_set) {
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&moe_gemm_v3_kernel),
+        hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
+    attr_set = true;
   }
+  dim3 grid((unsigned)(N / 128), (unsigned)ntiles_max);
+  moe_gemm_v3_kernel<<<grid, 512, lds_bytes, stream>>>(
+      out, x, w, sorted_ids, tile_expert, N, K, gather_div);
 }
 
 void launch_moe_combine(bf16* out, const bf16* y, const float* wts,
@@ -309,29 +295,23 @@ void launch_moe_combine(bf16* out, const bf16* y, const float* wts,
 }
 
 // ----------------------------------------------------- fp8 grouped GEMM
-// Same 256xBN glds structure as moe_gemm_v2_kernel, e4m3fn operands:
+// Same 256x128 3-deep-ring structure as moe_gemm_v3_kernel, e4m3fn:
 // out[p,n] = (sum_k A8[p,k] * B8[e,n,k]) * a_scale[row(p)] * b_scale[e,n]
 // via v_mfma_f32_32x32x16_fp8_fp8.  128-B staged rows = BK 128 fp8
-// elements, so the loader geometry is byte-identical to the bf16 tile;
-// each mfma consumes 8 fp8 bytes per lane (kk 0..7 per chunk).
+// elements (byte-identical loader geometry; kk 0..7 per chunk).
 typedef long long i64;
 
-template <int BN>
-__global__ __launch_bounds__(512) void moe_gemm_fp8_v2_kernel(
+__global__ __launch_bounds__(512) void moe_gemm_fp8_v3_kernel(
     bf16* __restrict__ out, const unsigned char* __restrict__ xq,
     const float* __restrict__ xs, const unsigned char* __restrict__ wq,
     const float* __restrict__ ws, const int* __restrict__ sorted_ids,
     const int* __restrict__ tile_expert, int N, int K, int gather_div) {
+  constexpr int BN = 128;
   const int e = tile_expert[blockIdx.y];
   if (e < 0) return;
   const int m0 = blockIdx.y * MOE_BM;
   const int n0 = blockIdx.x * BN;
 
-  // dynamic LDS addressed by INTEGER offsets into the extern array so
-  // the compiler keeps the AS3 provenance (a pointer array here decays
-  // to generic AS => fragment reads become flat_load + vmcnt waits in
-  // the MFMA stream — measured 13% slower than the v1 kernel)
-  // layout: A0 | A1 | B0 | B1  (A: 32 KiB each, B: BN*128 each)
   extern __shared__ int lds[];     // shared decl w/ moe_align
   unsigned char* dyn_lds = reinterpret_cast<unsigned char*>(lds);
 
@@ -358,45 +338,36 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_v2_kernel(
     }
     a_src[j] = xq + row * (long)K + gbyte;
   }
-  const int b_glds = BN / 64;
-  const unsigned char* b_src[4];
+  const unsigned char* b_src[2];
   #pragma unroll
-  for (int j = 0; j < b_glds; ++j) {
-    const int r = wid * (BN / 8) + j * 8 + lrow;
+  for (int j = 0; j < 2; ++j) {
+    const int r = wid * 16 + j * 8 + lrow;
     b_src[j] = wq + ((long)e * N + n0 + r) * (long)K + gbyte;
   }
 
-  constexpr int NB = BN / 2 / 32;
-  f32x16 acc[2][NB];
+  f32x16 acc[2][2];
   #pragma unroll
   for (int sm = 0; sm < 2; ++sm)
     #pragma unroll
-    for (int nb = 0; nb < NB; ++nb)
+    for (int nb = 0; nb < 2; ++nb)
       #pragma unroll
       for (int i = 0; i < 16; ++i) acc[sm][nb][i] = 0.f;
 
-  auto stage = [&](int buf, int k0) {
+  auto stage = [&](int slot, int k0) {
     const long cb = (long)k0;          // 1 B per fp8 element
     #pragma unroll
     for (int j = 0; j < 4; ++j)
       glds16(a_src[j] + cb,
-             &dyn_lds[buf * 32768 + wid * 4096 + j * 1024]);
+             &dyn_lds[slot * 32768 + wid * 4096 + j * 1024]);
     #pragma unroll
-    for (int j = 0; j < b_glds; ++j)
+    for (int j = 0; j < 2; ++j)
       glds16(b_src[j] + cb,
-             &dyn_lds[65536 + buf * (BN * 128) + wid * (BN * 16) +
-                      j * 1024]);
+             &dyn_lds[98304 + slot * 16384 + wid * 2048 + j * 1024]);
   };
 
-  stage(0, 0);
-  __syncthreads();
-
-  const int nchunks = K >> 7;          // 128 fp8 elems per chunk
-  for (int c = 0; c < nchunks; ++c) {
-    const int cur = c & 1;
-    if (c + 1 < nchunks) stage(cur ^ 1, (c + 1) << 7);
-    const int a_base = cur * 32768;
-    const int b_base = 65536 + cur * (BN * 128);
+  auto compute = [&](int slot) {
+    const int a_base = slot * 32768;
+    const int b_base = 98304 + slot * 16384;
     #pragma unroll
     for (int kk = 0; kk < 8; ++kk) {
       i64 af[2];
@@ -406,9 +377,9 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_v2_kernel(
             &dyn_lds[a_base + swz(wave_m * 64 + sm * 32 + lcol,
                                   kk * 16 + lhalf * 8)]);
       #pragma unroll
-      for (int nb = 0; nb < NB; ++nb) {
+      for (int nb = 0; nb < 2; ++nb) {
         i64 bf = *reinterpret_cast<const i64*>(
-            &dyn_lds[b_base + swz(wave_n * (BN / 2) + nb * 32 + lcol,
+            &dyn_lds[b_base + swz(wave_n * 64 + nb * 32 + lcol,
                                   kk * 16 + lhalf * 8)]);
         #pragma unroll
         for (int sm = 0; sm < 2; ++sm)
@@ -416,15 +387,34 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_v2_kernel(
               af[sm], bf, acc[sm][nb], 0, 0, 0);
       }
     }
-    __syncthreads();
+  };
+
+  const int nchunks = K >> 7;
+  stage(0, 0);
+  stage(1, 128);
+  int c = 0;
+  for (; c + 2 < nchunks; ++c) {
+    stage((c + 2) % 3, (c + 2) << 7);
+    asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    compute(c % 3);
+  }
+  asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  compute(c % 3);
+  ++c;
+  if (c < nchunks) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    compute(c % 3);
   }
 
   // epilogue: scale by a_scale[row] * b_scale[e, col], store bf16
   #pragma unroll
   for (int sm = 0; sm < 2; ++sm)
     #pragma unroll
-    for (int nb = 0; nb < NB; ++nb) {
-      const int col = n0 + wave_n * (BN / 2) + nb * 32 + lcol;
+    for (int nb = 0; nb < 2; ++nb) {
+      const int col = n0 + wave_n * 64 + nb * 32 + lcol;
       const float bs = ws[(long)e * N + col];
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
@@ -448,33 +438,17 @@ void launch_moe_gemm_fp8(bf16* out, const unsigned char* xq,
                          const float* ws, const int* sorted_ids,
                          const int* tile_expert, int ntiles_max, int N,
                          int K, int gather_div, hipStream_t stream) {
-  if (K % 128)
-    throw std::runtime_error("moe_gemm_fp8: K % 128 != 0");
-  const bool wide = (N % 256 == 0) &&
-                    (long)(N / 256) * ntiles_max >= 512;
-  static bool attr_set[2] = {false, false};
-  if (wide) {
-    const int lds_bytes = 2 * (32 * 1024 + 256 * 128);
-    if (!attr_set[0]) {
-      (void)hipFuncSetAttribute(
-          reinterpret_cast<const void*>(&moe_gemm_fp8_v2_kernel<256>),
-          hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
-      attr_set[0] = true;
-    }
-    dim3 grid((unsigned)(N / 256), (unsigned)ntiles_max);
-    moe_gemm_fp8_v2_kernel<256><<<grid, 512, lds_bytes, stream>>>(
-        out, xq, xs, wq, ws, sorted_ids, tile_expert, N, K, gather_div);
-  } else {
-    if (N % 128) throw std::runtime_error("moe_gemm_fp8: N % 128 != 0");
-    const int lds_bytes = 2 * (32 * 1024 + 128 * 128);
-    if (!attr_set[1]) {
-      (void)hipFuncSetAttribute(
-          reinterpret_cast<const void*>(&moe_gemm_fp8_v2_kernel<128>),
-          hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
-      attr_set[1] = true;
-    }
-    dim3 grid((unsigned)(N / 128), (unsigned)ntiles_max);
-    moe_gemm_fp8_v2_kernel<128><<<grid, 512, lds_bytes, stream>>>(
-        out, xq, xs, wq, ws, sorted_ids, tile_expert, N, K, gather_div);
+  if (K % 128 || K < 256 || N % 128)
+    throw std::runtime_error("moe_gemm_fp8: bad K/N alignment");
+  const int lds_bytes = 3 * (32 * 1024 + 16 * 1024);
+  static bool attr_set = false;
+  if (!attr_set) {
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&moe_gemm_fp8_v3_kernel),
+        hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
+    attr_set = true;
   }
+  dim3 grid((unsigned)(N / 128), (unsigned)ntiles_max);
+  moe_gemm_fp8_v3_kernel<<<grid, 512, lds_bytes, stream>>>(
+      out, xq, xs, wq, ws, sorted_ids, tile_expert, N, K, gather_div);
 }
