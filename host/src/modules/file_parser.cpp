@@ -242,6 +242,7 @@ void FileParserModule::init(ModuleCtx& ctx) {
   backends_.push_back(std::make_unique<HtmlBackend>());
   backends_.push_back(std::make_unique<CsvBackend>());
   backends_.push_back(std::make_unique<JsonBackend>());
+  add_document_backends(backends_);
   const Json& roots = ctx.config.at("allowed_roots");
   if (roots.is_array())
     for (auto& r : roots.arr()) allowed_roots_.push_back(r.as_string());
@@ -288,7 +289,8 @@ void FileParserModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
       out["backends"] = bs;
       Json exts = Json::array();
       for (const char* e : {"txt", "md", "html", "htm", "csv", "tsv",
-                            "json", "yaml"})
+                            "json", "yaml", "docx", "xlsx", "pptx",
+                            "pdf"})
         exts.push_back(e);
       out["extensions"] = exts;
       w.respond(200, "application/json", out.dump());

@@ -18,6 +18,10 @@ struct FileParserBackend {
   virtual std::string parse_markdown(const std::string& bytes) const = 0;
 };
 
+// docx/xlsx/pptx/pdf backends (file_parser_docs.cpp)
+void add_document_backends(
+    std::vector<std::unique_ptr<FileParserBackend>>& backends);
+
 class FileParserModule : public Module {
  public:
   std::string name() const override { return "file-parser"; }
