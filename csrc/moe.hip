@@ -99,15 +99,41 @@ DEV void glds16(const void* gp, void* lp) {
 // INVARIANT to the column-tile width, so BN=128 (acc 64 VGPR) buys the
 // LDS headroom for depth 3: 3 x (32 KiB A + 16 KiB B) = 144 KiB.
 // 256-row tiles halve the expert re-read vs the 128-row v1 kernel.
+// XCD-grouped block decode: consecutive blockIdx round-robin the 8
+// XCDs (guide: workgroup dispatch), so id%8 selects an XCD and the 32
+// ids sharing it form one CU-sized group.  All n-blocks of ONE m-tile
+// are placed on the SAME XCD: the tile's A rows (w2: 7.3 MB — bigger
+// than the 4 MiB per-XCD L2 if spread over all XCDs) are streamed from
+// HBM once per XCD-group instead of once per block.
+struct TileMap {
+  int t, j;      // m-tile index, n-block index (-1 = out of range)
+};
+DEV TileMap xcd_tile_map(int id, int ntiles, int nx) {
+  const int xcd = id & 7;
+  const int s = id >> 3;
+  const int jg = s & 31;             // position in the 32-CU group
+  const int grp = s >> 5;
+  const int tpx = (ntiles + 7) >> 3; // tiles owned per XCD (padded)
+  const int G = (nx + 31) >> 5;      // n-groups per tile
+  const int t = xcd + 8 * (grp % tpx);
+  const int g = grp / tpx;
+  const int j = g * 32 + jg;
+  if (t >= ntiles || g >= G || j >= nx) return {-1, -1};
+  return {t, j};
+}
+
 __global__ __launch_bounds__(512) void moe_gemm_v3_kernel(
     bf16* __restrict__ out, const bf16* __restrict__ x,
     const bf16* __restrict__ w, const int* __restrict__ sorted_ids,
-    const int* __restrict__ tile_expert, int N, int K, int gather_div) {
+    const int* __restrict__ tile_expert, int N, int K, int gather_div,
+    int ntiles) {
   constexpr int BN = 128;
-  const int e = tile_expert[blockIdx.y];
+  const TileMap tm = xcd_tile_map(blockIdx.x, ntiles, N / BN);
+  if (tm.t < 0) return;
+  const int e = tile_expert[tm.t];
   if (e < 0) return;
-  const int m0 = blockIdx.y * MOE_BM;
-  const int n0 = blockIdx.x * BN;
+  const int m0 = tm.t * MOE_BM;
+  const int n0 = tm.j * BN;
 
   // dynamic LDS addressed by INTEGER offsets (pointer indirection decays
   // to generic AS => flat_load + vmcnt stalls in the MFMA stream)
@@ -155,6 +181,20 @@ __global__ __launch_bounds__(512) void moe_gemm_v3_kernel(
       #pragma unroll
       for (int i = 0; i < 16; ++i) acc[sm][nb][i] = 0.f;
 
+  // padding skip: a wave whose 64 output rows are ALL padding keeps its
+  // acc at zero and skips fragment reads + MFMAs (it still issues its
+  // glds share and hits every barrier — the tile stage is cooperative).
+  // ~1/3 of tile rows are padding at bs1024 (256-row align granularity)
+  bool wave_live = true;
+  if (gather_div > 0) {
+    wave_live = false;
+    #pragma unroll
+    for (int r = 0; r < 64; r += 16)        // 4 probes/lane cover 64 rows
+      if (sorted_ids[m0 + wave_m * 64 + r + (lane & 15)] >= 0)
+        wave_live = true;
+    wave_live = __any(wave_live);
+  }
+
   auto stage = [&](int slot, int k0) {
     const long cb = (long)k0 * 2;
     #pragma unroll
@@ -168,6 +208,7 @@ __global__ __launch_bounds__(512) void moe_gemm_v3_kernel(
   };
 
   auto compute = [&](int slot) {
+    if (!wave_live) return;
     const int a_base = slot * 32768;
     const int b_base = 98304 + slot * 16384;
     #pragma unroll
@@ -280,9 +321,12 @@ void launch_moe_gemm(bf16* out, const bf16* x, const bf16* w,
         hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
     attr_set = true;
   }
-  dim3 grid((unsigned)(N / 128), (unsigned)ntiles_max);
-  moe_gemm_v3_kernel<<<grid, 512, lds_bytes, stream>>>(
-      out, x, w, sorted_ids, tile_expert, N, K, gather_div);
+  const int nx = N / 128;
+  const long nblocks =
+      8L * 32 * ((ntiles_max + 7) / 8) * ((nx + 31) / 32);
+  moe_gemm_v3_kernel<<<dim3((unsigned)nblocks), 512, lds_bytes,
+                       stream>>>(out, x, w, sorted_ids, tile_expert, N,
+                                 K, gather_div, ntiles_max);
 }
 
 void launch_moe_combine(bf16* out, const bf16* y, const float* wts,
@@ -305,12 +349,15 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_v3_kernel(
     bf16* __restrict__ out, const unsigned char* __restrict__ xq,
     const float* __restrict__ xs, const unsigned char* __restrict__ wq,
     const float* __restrict__ ws, const int* __restrict__ sorted_ids,
-    const int* __restrict__ tile_expert, int N, int K, int gather_div) {
+    const int* __restrict__ tile_expert, int N, int K, int gather_div,
+    int ntiles) {
   constexpr int BN = 128;
-  const int e = tile_expert[blockIdx.y];
+  const TileMap tm = xcd_tile_map(blockIdx.x, ntiles, N / BN);
+  if (tm.t < 0) return;
+  const int e = tile_expert[tm.t];
   if (e < 0) return;
-  const int m0 = blockIdx.y * MOE_BM;
-  const int n0 = blockIdx.x * BN;
+  const int m0 = tm.t * MOE_BM;
+  const int n0 = tm.j * BN;
 
   extern __shared__ int lds[];     // shared decl w/ moe_align
   unsigned char* dyn_lds = reinterpret_cast<unsigned char*>(lds);
@@ -353,6 +400,20 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_v3_kernel(
       #pragma unroll
       for (int i = 0; i < 16; ++i) acc[sm][nb][i] = 0.f;
 
+  // padding skip: a wave whose 64 output rows are ALL padding keeps its
+  // acc at zero and skips fragment reads + MFMAs (it still issues its
+  // glds share and hits every barrier — the tile stage is cooperative).
+  // ~1/3 of tile rows are padding at bs1024 (256-row align granularity)
+  bool wave_live = true;
+  if (gather_div > 0) {
+    wave_live = false;
+    #pragma unroll
+    for (int r = 0; r < 64; r += 16)        // 4 probes/lane cover 64 rows
+      if (sorted_ids[m0 + wave_m * 64 + r + (lane & 15)] >= 0)
+        wave_live = true;
+    wave_live = __any(wave_live);
+  }
+
   auto stage = [&](int slot, int k0) {
     const long cb = (long)k0;          // 1 B per fp8 element
     #pragma unroll
@@ -366,6 +427,7 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_v3_kernel(
   };
 
   auto compute = [&](int slot) {
+    if (!wave_live) return;
     const int a_base = slot * 32768;
     const int b_base = 98304 + slot * 16384;
     #pragma unroll
@@ -448,7 +510,11 @@ void launch_moe_gemm_fp8(bf16* out, const unsigned char* xq,
         hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
     attr_set = true;
   }
-  dim3 grid((unsigned)(N / 128), (unsigned)ntiles_max);
-  moe_gemm_fp8_v3_kernel<<<grid, 512, lds_bytes, stream>>>(
-      out, xq, xs, wq, ws, sorted_ids, tile_expert, N, K, gather_div);
+  const int nx = N / 128;
+  const long nblocks =
+      8L * 32 * ((ntiles_max + 7) / 8) * ((nx + 31) / 32);
+  moe_gemm_fp8_v3_kernel<<<dim3((unsigned)nblocks), 512, lds_bytes,
+                           stream>>>(out, xq, xs, wq, ws, sorted_ids,
+                                     tile_expert, N, K, gather_div,
+                                     ntiles_max);
 }
