@@ -79,6 +79,7 @@ void ResponseWriter::respond(
     const std::vector<std::pair<std::string, std::string>>& extra) {
   if (started_) return;
   started_ = true;
+  status_ = status;
   std::ostringstream h;
   h << "HTTP/1.1 " << status << " " << status_text(status) << "\r\n"
     << "content-type: " << content_type << "\r\n"
@@ -100,6 +101,7 @@ void ResponseWriter::begin_stream(
     const std::vector<std::pair<std::string, std::string>>& extra) {
   if (started_) return;
   started_ = true;
+  status_ = status;
   streaming_ = true;
   std::ostringstream h;
   h << "HTTP/1.1 " << status << " " << status_text(status) << "\r\n"

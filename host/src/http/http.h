@@ -56,6 +56,7 @@ class ResponseWriter {
   void end_stream();
 
   bool started() const { return started_; }
+  int status() const { return status_; }
   bool keep_alive = true;
   // headers stamped onto every response (request id, traceparent)
   std::vector<std::pair<std::string, std::string>> default_headers;
@@ -65,6 +66,7 @@ class ResponseWriter {
  private:
   bool send_all(const char* p, size_t n);
   int fd_;
+  int status_ = 0;
   bool started_ = false;
   bool streaming_ = false;
 };

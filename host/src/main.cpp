@@ -13,6 +13,7 @@
 #include <iostream>
 
 #include "modkit/modkit.h"
+#include "modkit/telemetry.h"
 #include "modules/api_gateway.h"
 #include "modules/llm_gateway.h"
 #include "modules/serverless_runtime.h"
@@ -30,6 +31,12 @@ static std::atomic<bool> g_cancel{false};
 static void on_signal(int) { g_cancel = true; }
 
 int run_server(const Json& cfg, bool check_only) {
+  // OTLP trace export (telemetry/init.rs equivalent; no-op when
+  // tracing.otlp_endpoint is unset) + boot-time connectivity probe
+  TraceExporter::instance().configure(cfg.path("tracing"));
+  if (TraceExporter::instance().enabled())
+    TraceExporter::instance().probe();
+
   // logging init from config (bootstrap/host/logging.rs equivalent)
   const Json& lg = cfg.path("logging.default");
   Logger::get().configure(
@@ -105,6 +112,7 @@ int run_server(const Json& cfg, bool check_only) {
   cancel_flag = true;
   for (auto it = mods.rbegin(); it != mods.rend(); ++it)
     it->first->stop(it->second);
+  TraceExporter::instance().shutdown();    // flush pending spans
   return 0;
 }
 

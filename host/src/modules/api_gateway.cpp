@@ -1,5 +1,7 @@
 #include "api_gateway.h"
 
+#include "../modkit/telemetry.h"
+
 #include <random>
 
 #include "../util/log.h"
@@ -186,17 +188,37 @@ void ApiGatewayModule::handle(HttpRequest& req, ResponseWriter& w) {
     }
   }
 
-  int status = 0;
+  const uint64_t span_start_ns = (uint64_t)
+      std::chrono::duration_cast<std::chrono::nanoseconds>(
+          std::chrono::system_clock::now().time_since_epoch()).count();
   dispatch(req, w);
   auto dt = std::chrono::duration<double, std::milli>(
       std::chrono::steady_clock::now() - t0).count();
-  // http_request span record (OTel semantic fields, log-exported)
+  // http_request span record (OTel semantic fields): logged + exported
+  // over OTLP/HTTP when tracing.otlp_endpoint is configured
   LOG_DEBUG("http", "%s %s -> done in %.2fms rid=%s trace=%s span=%s",
             req.method.c_str(), req.path.c_str(), dt,
             req.request_id.c_str(),
             req.extensions.path("trace.trace_id").as_string().c_str(),
             req.extensions.path("trace.span_id").as_string().c_str());
-  (void)status;
+  auto& exp = TraceExporter::instance();
+  if (exp.enabled()) {
+    SpanRecord sp;
+    sp.trace_id = req.extensions.path("trace.trace_id").as_string();
+    sp.span_id = req.extensions.path("trace.span_id").as_string();
+    sp.parent_span_id =
+        req.extensions.path("trace.parent_span_id").as_string();
+    sp.name = "http_request";
+    sp.start_ns = span_start_ns;
+    sp.end_ns = span_start_ns + (uint64_t)(dt * 1e6);
+    const int st = w.status();
+    sp.status_code = st >= 500 ? 2 : 1;
+    sp.attrs = {{"http.request.method", req.method},
+                {"url.path", req.path},
+                {"http.response.status_code", std::to_string(st)},
+                {"http.request.id", req.request_id}};
+    exp.record(std::move(sp));
+  }
 }
 
 void ApiGatewayModule::dispatch(HttpRequest& req, ResponseWriter& w) {
