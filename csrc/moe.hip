@@ -108,7 +108,13 @@ DEV void glds16(const void* gp, void* lp) {
 struct TileMap {
   int t, j;      // m-tile index, n-block index (-1 = out of range)
 };
-DEV TileMap xcd_tile_map(int id, int ntiles, int nx) {
+DEV TileMap xcd_tile_map(int id, int ntiles, int nx, int use_xcd) {
+  if (!use_xcd) {
+    const int t = id / nx;
+    const int j = id - t * nx;
+    if (t >= ntiles) return {-1, -1};
+    return {t, j};
+  }
   const int xcd = id & 7;
   const int s = id >> 3;
   const int jg = s & 31;             // position in the 32-CU group
@@ -126,9 +132,9 @@ __global__ __launch_bounds__(512) void moe_gemm_v3_kernel(
     bf16* __restrict__ out, const bf16* __restrict__ x,
     const bf16* __restrict__ w, const int* __restrict__ sorted_ids,
     const int* __restrict__ tile_expert, int N, int K, int gather_div,
-    int ntiles) {
+    int ntiles, int use_xcd) {
   constexpr int BN = 128;
-  const TileMap tm = xcd_tile_map(blockIdx.x, ntiles, N / BN);
+  const TileMap tm = xcd_tile_map(blockIdx.x, ntiles, N / BN, use_xcd);
   if (tm.t < 0) return;
   const int e = tile_expert[tm.t];
   if (e < 0) return;
@@ -232,30 +238,30 @@ __global__ __launch_bounds__(512) void moe_gemm_v3_kernel(
     }
   };
 
-  const int nchunks = K >> 6;          // K % 64 == 0 (checked at launch)
-  // 3-stage software pipeline: issue c+2, wait for c (two chunks = 12
-  // glds still in flight), raw-barrier, compute c.  The slot being
-  // overwritten by c+2 was read as chunk c-1 BEFORE the previous
-  // barrier, so every wave is done with it.
+  const int nchunks = K >> 6;          // K % 128 == 0 (checked at launch)
+  // 3-slot software pipeline, one barrier per chunk:
+  //   [compute d | issue d+2 | wait vmcnt(6) (d+1 landed) | barrier]
+  // The issue of d+2 reuses slot (d-1)%3, whose chunk was computed by
+  // every wave BEFORE the previous barrier — no read/overwrite race —
+  // and one whole chunk (d+2) stays in flight across each barrier
+  // (guide's 3-buf span: counted vmcnt + raw s_barrier, never
+  // __syncthreads while a glds is outstanding).
   stage(0, 0);
   stage(1, 64);
-  int c = 0;
-  for (; c + 2 < nchunks; ++c) {
-    stage((c + 2) % 3, (c + 2) << 6);
-    asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
-    compute(c % 3);
-  }
-  // drain: one chunk in flight, then none
   asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   __builtin_amdgcn_s_barrier();
-  compute(c % 3);
-  ++c;
-  if (c < nchunks) {
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  int d = 0;
+  for (; d + 2 < nchunks; ++d) {
+    compute(d % 3);
+    stage((d + 2) % 3, (d + 2) << 6);
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    compute(c % 3);
   }
+  compute(d % 3);                      // d == nchunks-2
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  ++d;
+  compute(d % 3);                      // d == nchunks-1
 
   // ---- epilogue: C row = (reg&3) + 8*(reg>>2) + 4*lhalf, col = lcol ----
   #pragma unroll
@@ -321,12 +327,17 @@ void launch_moe_gemm(bf16* out, const bf16* x, const bf16* w,
         hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
     attr_set = true;
   }
+  static const int use_xcd = [] {
+    const char* v = getenv("HS_MOE_XCD");
+    return v ? atoi(v) : 1;
+  }();
   const int nx = N / 128;
-  const long nblocks =
-      8L * 32 * ((ntiles_max + 7) / 8) * ((nx + 31) / 32);
+  const long nblocks = use_xcd
+      ? 8L * 32 * ((ntiles_max + 7) / 8) * ((nx + 31) / 32)
+      : (long)nx * ntiles_max;
   moe_gemm_v3_kernel<<<dim3((unsigned)nblocks), 512, lds_bytes,
                        stream>>>(out, x, w, sorted_ids, tile_expert, N,
-                                 K, gather_div, ntiles_max);
+                                 K, gather_div, ntiles_max, use_xcd);
 }
 
 void launch_moe_combine(bf16* out, const bf16* y, const float* wts,
@@ -350,9 +361,9 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_v3_kernel(
     const float* __restrict__ xs, const unsigned char* __restrict__ wq,
     const float* __restrict__ ws, const int* __restrict__ sorted_ids,
     const int* __restrict__ tile_expert, int N, int K, int gather_div,
-    int ntiles) {
+    int ntiles, int use_xcd) {
   constexpr int BN = 128;
-  const TileMap tm = xcd_tile_map(blockIdx.x, ntiles, N / BN);
+  const TileMap tm = xcd_tile_map(blockIdx.x, ntiles, N / BN, use_xcd);
   if (tm.t < 0) return;
   const int e = tile_expert[tm.t];
   if (e < 0) return;
@@ -452,24 +463,23 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_v3_kernel(
   };
 
   const int nchunks = K >> 7;
+  // same race-free 3-slot pipeline as the bf16 kernel
   stage(0, 0);
   stage(1, 128);
-  int c = 0;
-  for (; c + 2 < nchunks; ++c) {
-    stage((c + 2) % 3, (c + 2) << 7);
-    asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
-    compute(c % 3);
-  }
   asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   __builtin_amdgcn_s_barrier();
-  compute(c % 3);
-  ++c;
-  if (c < nchunks) {
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  int d = 0;
+  for (; d + 2 < nchunks; ++d) {
+    compute(d % 3);
+    stage((d + 2) % 3, (d + 2) << 7);
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    compute(c % 3);
   }
+  compute(d % 3);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  ++d;
+  compute(d % 3);
 
   // epilogue: scale by a_scale[row] * b_scale[e, col], store bf16
   #pragma unroll
@@ -510,11 +520,16 @@ void launch_moe_gemm_fp8(bf16* out, const unsigned char* xq,
         hipFuncAttributeMaxDynamicSharedMemorySize, lds_bytes);
     attr_set = true;
   }
+  static const int use_xcd = [] {
+    const char* v = getenv("HS_MOE_XCD");
+    return v ? atoi(v) : 1;
+  }();
   const int nx = N / 128;
-  const long nblocks =
-      8L * 32 * ((ntiles_max + 7) / 8) * ((nx + 31) / 32);
+  const long nblocks = use_xcd
+      ? 8L * 32 * ((ntiles_max + 7) / 8) * ((nx + 31) / 32)
+      : (long)nx * ntiles_max;
   moe_gemm_fp8_v3_kernel<<<dim3((unsigned)nblocks), 512, lds_bytes,
                            stream>>>(out, xq, xs, wq, ws, sorted_ids,
                                      tile_expert, N, K, gather_div,
-                                     ntiles_max);
+                                     ntiles_max, use_xcd);
 }
