@@ -329,7 +329,7 @@ void launch_moe_gemm(bf16* out, const bf16* x, const bf16* w,
   }
   static const int use_xcd = [] {
     const char* v = getenv("HS_MOE_XCD");
-    return v ? atoi(v) : 1;
+    return v ? atoi(v) : 0;
   }();
   const int nx = N / 128;
   const long nblocks = use_xcd
@@ -522,7 +522,7 @@ void launch_moe_gemm_fp8(bf16* out, const unsigned char* xq,
   }
   static const int use_xcd = [] {
     const char* v = getenv("HS_MOE_XCD");
-    return v ? atoi(v) : 1;
+    return v ? atoi(v) : 0;
   }();
   const int nx = N / 128;
   const long nblocks = use_xcd

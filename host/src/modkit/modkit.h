@@ -10,8 +10,10 @@
 #pragma once
 
 #include <functional>
+#include <future>
 #include <map>
 #include <memory>
+#include <mutex>
 #include <optional>
 #include <string>
 #include <vector>
@@ -136,22 +138,100 @@ class RestRegistry {
 // -------------------------------------------------------------- ClientHub
 // Type-erased inter-module DI: modules publish SDK client objects under an
 // interface name; consumers resolve them without direct dependencies.
+// Typed DI hub (reference libs/modkit/src/client_hub.rs:123).  Besides
+// plain interface-keyed clients it supports SCOPED registrations keyed
+// by (interface, ClientScope) where ClientScope is a GTS instance id
+// (client_hub.rs:155-195): get_scoped falls back to the unscoped
+// registration when no scope-specific client exists.
 class ClientHub {
  public:
   template <typename T>
   void register_client(const std::string& iface, std::shared_ptr<T> impl) {
+    std::lock_guard<std::mutex> lk(mu_);
     clients_[iface] = std::static_pointer_cast<void>(impl);
   }
   template <typename T>
+  void register_scoped(const std::string& iface, const std::string& scope,
+                       std::shared_ptr<T> impl) {
+    std::lock_guard<std::mutex> lk(mu_);
+    clients_[iface + "\x1f" + scope] = std::static_pointer_cast<void>(impl);
+  }
+  template <typename T>
   std::shared_ptr<T> get(const std::string& iface) const {
+    std::lock_guard<std::mutex> lk(mu_);
     auto it = clients_.find(iface);
     if (it == clients_.end()) return nullptr;
     return std::static_pointer_cast<T>(it->second);
   }
-  bool has(const std::string& iface) const { return clients_.count(iface); }
+  template <typename T>
+  std::shared_ptr<T> get_scoped(const std::string& iface,
+                                const std::string& scope) const {
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      auto it = clients_.find(iface + "\x1f" + scope);
+      if (it != clients_.end())
+        return std::static_pointer_cast<T>(it->second);
+    }
+    return get<T>(iface);          // fallback: unscoped registration
+  }
+  bool has(const std::string& iface) const {
+    std::lock_guard<std::mutex> lk(mu_);
+    return clients_.count(iface) > 0;
+  }
 
  private:
+  mutable std::mutex mu_;
   std::map<std::string, std::shared_ptr<void>> clients_;
+};
+
+// Single-flight cached plugin-instance resolution (reference
+// libs/modkit/src/plugins/mod.rs:14,:44 GtsPluginSelector): the first
+// caller runs the resolver (a types-registry lookup), concurrent
+// callers wait on the same future, later callers hit the cache until
+// invalidate().
+class GtsPluginSelector {
+ public:
+  using Resolver = std::function<std::string(const std::string&)>;
+  explicit GtsPluginSelector(Resolver r) : resolver_(std::move(r)) {}
+
+  std::string select(const std::string& plugin_type) {
+    std::shared_ptr<std::promise<std::string>> leader;
+    std::shared_future<std::string> fut;
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      auto it = cache_.find(plugin_type);
+      if (it != cache_.end()) {
+        fut = it->second;
+      } else {
+        leader = std::make_shared<std::promise<std::string>>();
+        fut = leader->get_future().share();
+        cache_[plugin_type] = fut;
+      }
+    }
+    if (leader) {
+      try {
+        leader->set_value(resolver_(plugin_type));
+      } catch (...) {
+        {
+          std::lock_guard<std::mutex> lk(mu_);
+          cache_.erase(plugin_type);      // failed lookups are not cached
+        }
+        leader->set_exception(std::current_exception());
+      }
+    }
+    return fut.get();
+  }
+
+  void invalidate(const std::string& plugin_type = "") {
+    std::lock_guard<std::mutex> lk(mu_);
+    if (plugin_type.empty()) cache_.clear();
+    else cache_.erase(plugin_type);
+  }
+
+ private:
+  Resolver resolver_;
+  std::mutex mu_;
+  std::map<std::string, std::shared_future<std::string>> cache_;
 };
 
 // ---------------------------------------------------------------- modules

@@ -207,7 +207,69 @@ void AuthzResolverModule::init(ModuleCtx& ctx) {
 
 // -------------------------------------------------------- types-registry
 
-void TypesRegistryModule::init(ModuleCtx& ctx) {}
+namespace {
+
+class InMemoryTypesRegistry : public TypesRegistryClient {
+ public:
+  std::optional<Json> get(const std::string& gts_id) override {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto it = entities_.find(gts_id);
+    if (it == entities_.end()) return std::nullopt;
+    return it->second;
+  }
+  std::vector<Json> list(const std::string& filter) override {
+    std::vector<Json> out;
+    std::lock_guard<std::mutex> lk(mu_);
+    for (auto& [id, e] : entities_) {
+      if (!filter.empty()) {
+        std::string pre = filter;
+        bool wild = !pre.empty() && pre.back() == '*';
+        if (wild) pre.pop_back();
+        if (wild ? id.rfind(pre, 0) != 0 : id != filter) continue;
+      }
+      out.push_back(e);
+    }
+    return out;
+  }
+  int register_entities(const Json& ents) override {
+    int n = 0;
+    std::lock_guard<std::mutex> lk(mu_);
+    for (const auto& e : ents.arr()) {
+      std::string id = e.at("gts_id").as_string();
+      if (id.empty()) throw Problem::bad_request("entity without gts_id");
+      entities_[id] = e;
+      ++n;
+    }
+    return n;
+  }
+
+ private:
+  std::mutex mu_;
+  std::map<std::string, Json> entities_;   // gts_id -> entity
+};
+
+std::shared_ptr<InMemoryTypesRegistry> g_types;
+
+}  // namespace
+
+void TypesRegistryModule::init(ModuleCtx& ctx) {
+  g_types = std::make_shared<InMemoryTypesRegistry>();
+  ctx.hub->register_client<TypesRegistryClient>("types-registry", g_types);
+  // plugin discovery backbone: a single-flight cached selector resolving
+  // a plugin TYPE id to the first registered instance whose gts_id
+  // extends it (reference GtsPluginSelector over types-registry)
+  auto types = g_types;
+  ctx.hub->register_client<GtsPluginSelector>(
+      "plugin-selector",
+      std::make_shared<GtsPluginSelector>(
+          [types](const std::string& plugin_type) -> std::string {
+            auto hits = types->list(plugin_type + "*");
+            if (hits.empty())
+              throw std::runtime_error("no instance registered for " +
+                                       plugin_type);
+            return hits[0].at("gts_id").as_string();
+          }));
+}
 
 void TypesRegistryModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
   // REST surface per reference: /types-registry/v1/entities{,/{gts_id}}
@@ -218,20 +280,10 @@ void TypesRegistryModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
   list.summary = "List GTS entities (wildcard filter via ?filter=)";
   list.authenticated = true;
   list.tags = {"types-registry"};
-  rest.register_op(list, [this](HttpRequest& rq, ResponseWriter& w) {
+  rest.register_op(list, [](HttpRequest& rq, ResponseWriter& w) {
     std::string filter = rq.query.count("filter") ? rq.query["filter"] : "";
     Json items = Json::array();
-    std::lock_guard<std::mutex> lk(mu_);
-    for (auto& [id, e] : entities_) {
-      if (!filter.empty()) {
-        // '*' suffix wildcard
-        std::string pre = filter;
-        bool wild = !pre.empty() && pre.back() == '*';
-        if (wild) pre.pop_back();
-        if (wild ? id.rfind(pre, 0) != 0 : id != filter) continue;
-      }
-      items.push_back(e);
-    }
+    for (auto& e : g_types->list(filter)) items.push_back(e);
     Json out = Json::object();
     out["items"] = items;
     w.respond(200, "application/json", out.dump());
@@ -245,23 +297,18 @@ void TypesRegistryModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
   reg.authenticated = true;
   reg.allowed_content_types = {"application/json"};
   reg.tags = {"types-registry"};
-  rest.register_op(reg, [this](HttpRequest& rq, ResponseWriter& w) {
+  ClientHub* hub = ctx.hub;
+  rest.register_op(reg, [hub](HttpRequest& rq, ResponseWriter& w) {
     Json body;
     try { body = Json::parse(rq.body); }
     catch (...) { throw Problem::bad_request("invalid JSON body"); }
     const Json& ents = body.at("entities");
     if (!ents.is_array())
       throw Problem::bad_request("'entities' array required");
-    int n = 0;
-    {
-      std::lock_guard<std::mutex> lk(mu_);
-      for (const auto& e : ents.arr()) {
-        std::string id = e.at("gts_id").as_string();
-        if (id.empty()) throw Problem::bad_request("entity without gts_id");
-        entities_[id] = e;
-        ++n;
-      }
-    }
+    const int n = g_types->register_entities(ents);
+    // new instances may shadow cached plugin selections
+    if (auto sel = hub->get<GtsPluginSelector>("plugin-selector"))
+      sel->invalidate();
     Json out = Json::object();
     out["registered"] = n;
     w.respond(201, "application/json", out.dump());
@@ -273,11 +320,10 @@ void TypesRegistryModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
   get.operation_id = "types_registry_get";
   get.authenticated = true;
   get.tags = {"types-registry"};
-  rest.register_op(get, [this](HttpRequest& rq, ResponseWriter& w) {
-    std::lock_guard<std::mutex> lk(mu_);
-    auto it = entities_.find(rq.path_params["gts_id"]);
-    if (it == entities_.end()) throw Problem::not_found("no such entity");
-    w.respond(200, "application/json", it->second.dump());
+  rest.register_op(get, [](HttpRequest& rq, ResponseWriter& w) {
+    auto e = g_types->get(rq.path_params["gts_id"]);
+    if (!e) throw Problem::not_found("no such entity");
+    w.respond(200, "application/json", e->dump());
   });
 }
 

@@ -70,15 +70,20 @@ class AuthzResolverModule : public Module {
 };
 
 // ---- types-registry (GTS entity store; plugin discovery backbone) ----
+// in-process client (reference types-registry-sdk/src/api.rs:22)
+struct TypesRegistryClient {
+  virtual ~TypesRegistryClient() = default;
+  virtual std::optional<Json> get(const std::string& gts_id) = 0;
+  // '*'-suffix wildcard listing
+  virtual std::vector<Json> list(const std::string& filter) = 0;
+  virtual int register_entities(const Json& entities) = 0;
+};
+
 class TypesRegistryModule : public Module {
  public:
   std::string name() const override { return "types-registry"; }
   void init(ModuleCtx& ctx) override;
   void register_rest(ModuleCtx& ctx, RestRegistry& rest) override;
-
- private:
-  std::mutex mu_;
-  std::map<std::string, Json> entities_;   // gts_id -> entity
 };
 
 // ---- nodes-registry (node inventory incl. GPU info) ----
