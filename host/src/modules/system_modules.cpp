@@ -448,7 +448,9 @@ namespace {
 
 class LocalModelRegistry : public ModelRegistryClient {
  public:
-  LocalModelRegistry(const Json& cfg, const std::string& serving_model) {
+  LocalModelRegistry(const Json& cfg, const std::string& serving_model,
+                     std::unique_ptr<Db> db)
+      : db_(std::move(db)) {
     // models from config: modules.model-registry.config.models:
     // [{canonical_id, name, architecture, context_window, managed...}]
     const Json& ms = cfg.at("models");
@@ -486,39 +488,118 @@ class LocalModelRegistry : public ModelRegistryClient {
   std::optional<Json> get_tenant_model(const std::string& tenant,
                                        const std::string& canonical)
       override {
-    // canonical id = {provider_slug}::{provider_model_id}, split on FIRST
-    // "::" (reference PRD.md:11); bare model ids default to "local::"
-    std::string cid = canonical.find("::") == std::string::npos
-        ? "local::" + canonical : canonical;
-    auto it = models_.find(cid);
-    if (it == models_.end()) return std::nullopt;
-    Json m = it->second;
-    m["approval"] = approval(tenant, cid);
+    // resolution (reference PRD.md:11,:298-306): aliases first (tenant
+    // alias shadows a parent/global alias), then canonical id split on
+    // FIRST "::"; bare ids default to "local::".  Tenant-registered
+    // models shadow globals by canonical id (PRD.md:181-188).
+    std::string name = canonical;
+    if (name.find("::") == std::string::npos) {
+      if (auto a = alias_lookup(tenant, name)) name = *a;
+    }
+    std::string cid = name.find("::") == std::string::npos
+        ? "local::" + name : name;
+    std::optional<Json> m = db_model(tenant, cid);
+    if (!m) m = db_model("", cid);
+    if (!m) {
+      auto it = models_.find(cid);
+      if (it == models_.end()) return std::nullopt;
+      m = it->second;
+    }
+    (*m)["approval"] = approval(tenant, cid);
     return m;
   }
-  std::vector<Json> list_tenant_models(const std::string&) override {
+  std::vector<Json> list_tenant_models(const std::string& tenant) override {
     std::vector<Json> out;
-    for (auto& [_, m] : models_) out.push_back(m);
+    std::map<std::string, Json> merged = models_;   // global defaults
+    {
+      std::lock_guard<std::mutex> lk(db_->mu());
+      for (auto& r : db_->query(
+               "SELECT spec FROM models WHERE tenant_id IN ('', ?) "
+               "ORDER BY tenant_id",   // tenant rows overwrite globals
+               {DbValue::S(tenant)})) {
+        try {
+          Json m = Json::parse(r.at("spec").as_string());
+          merged[m.at("canonical_id").as_string()] = m;
+        } catch (...) {}
+      }
+    }
+    for (auto& [_, m] : merged) out.push_back(m);
     return out;
+  }
+
+  // tenant-scoped model registration with shadow-by-canonical-id
+  void put_model(const std::string& tenant, const Json& spec) {
+    std::lock_guard<std::mutex> lk(db_->mu());
+    db_->query(
+        "INSERT INTO models (tenant_id, canonical_id, spec) VALUES "
+        "(?,?,?) ON CONFLICT(tenant_id, canonical_id) DO UPDATE SET "
+        "spec=excluded.spec",
+        {DbValue::S(tenant),
+         DbValue::S(spec.at("canonical_id").as_string()),
+         DbValue::S(spec.dump())});
+  }
+
+  // aliases (PRD.md:298-306): name -> canonical, tenant shadows global
+  void put_alias(const std::string& tenant, const std::string& name,
+                 const std::string& cid) {
+    std::lock_guard<std::mutex> lk(db_->mu());
+    db_->query(
+        "INSERT INTO aliases (tenant_id, name, canonical_id) VALUES "
+        "(?,?,?) ON CONFLICT(tenant_id, name) DO UPDATE SET "
+        "canonical_id=excluded.canonical_id",
+        {DbValue::S(tenant), DbValue::S(name), DbValue::S(cid)});
+  }
+  bool del_alias(const std::string& tenant, const std::string& name) {
+    std::lock_guard<std::mutex> lk(db_->mu());
+    db_->query("DELETE FROM aliases WHERE tenant_id=? AND name=?",
+               {DbValue::S(tenant), DbValue::S(name)});
+    return db_->changes() > 0;
+  }
+  std::optional<std::string> alias_lookup(const std::string& tenant,
+                                          const std::string& name) {
+    std::lock_guard<std::mutex> lk(db_->mu());
+    for (const char* scope : {"t", "g"}) {
+      auto r = db_->query(
+          "SELECT canonical_id FROM aliases WHERE tenant_id=? AND name=?",
+          {DbValue::S(scope[0] == 't' ? tenant : ""), DbValue::S(name)});
+      if (!r.empty()) return r[0].at("canonical_id").as_string();
+    }
+    return std::nullopt;
   }
 
   // ModelApproval states pending->approved|rejected->revoked
   // (model-registry PRD.md:225-253); default approved so single-tenant
-  // deployments need no workflow
+  // deployments need no workflow.  Persisted (survives host restart).
   std::string approval(const std::string& tenant, const std::string& cid) {
-    std::lock_guard<std::mutex> lk(mu_);
-    auto it = approvals_.find(tenant + "|" + cid);
-    return it == approvals_.end() ? "approved" : it->second;
+    std::lock_guard<std::mutex> lk(db_->mu());
+    auto r = db_->query(
+        "SELECT status FROM approvals WHERE tenant_id=? AND "
+        "canonical_id=?",
+        {DbValue::S(tenant), DbValue::S(cid)});
+    return r.empty() ? "approved" : r[0].at("status").as_string();
   }
   void set_approval(const std::string& tenant, const std::string& cid,
                     const std::string& st) {
-    std::lock_guard<std::mutex> lk(mu_);
-    approvals_[tenant + "|" + cid] = st;
+    std::lock_guard<std::mutex> lk(db_->mu());
+    db_->query(
+        "INSERT INTO approvals (tenant_id, canonical_id, status) VALUES "
+        "(?,?,?) ON CONFLICT(tenant_id, canonical_id) DO UPDATE SET "
+        "status=excluded.status",
+        {DbValue::S(tenant), DbValue::S(cid), DbValue::S(st)});
   }
 
  private:
-  std::mutex mu_;
-  std::map<std::string, std::string> approvals_;
+  std::optional<Json> db_model(const std::string& tenant,
+                               const std::string& cid) {
+    std::lock_guard<std::mutex> lk(db_->mu());
+    auto r = db_->query(
+        "SELECT spec FROM models WHERE tenant_id=? AND canonical_id=?",
+        {DbValue::S(tenant), DbValue::S(cid)});
+    if (r.empty()) return std::nullopt;
+    try { return Json::parse(r[0].at("spec").as_string()); }
+    catch (...) { return std::nullopt; }
+  }
+  std::unique_ptr<Db> db_;
   std::map<std::string, Json> models_;
 };
 
@@ -527,10 +608,37 @@ std::shared_ptr<LocalModelRegistry> g_registry;
 }  // namespace
 
 void ModelRegistryModule::init(ModuleCtx& ctx) {
+  std::string file = ctx.full_config
+                         .path("modules.model-registry.database.file")
+                         .as_string("");
+  if (file.empty()) {
+    std::string home = ctx.home_dir;
+    if (!home.empty() && home[0] == '~') {
+      const char* h = getenv("HOME");
+      home = std::string(h ? h : "/tmp") + home.substr(1);
+    }
+    mkdir(home.c_str(), 0755);
+    file = home + "/model-registry.db";
+  }
+  auto db = std::make_unique<Db>(file);
+  db->migrate("model-registry", {
+      {"0001_approvals",
+       "CREATE TABLE approvals (tenant_id TEXT NOT NULL, canonical_id "
+       "TEXT NOT NULL, status TEXT NOT NULL, UNIQUE (tenant_id, "
+       "canonical_id))"},
+      {"0002_aliases",
+       "CREATE TABLE aliases (tenant_id TEXT NOT NULL, name TEXT NOT "
+       "NULL, canonical_id TEXT NOT NULL, UNIQUE (tenant_id, name))"},
+      {"0003_models",
+       "CREATE TABLE models (tenant_id TEXT NOT NULL, canonical_id TEXT "
+       "NOT NULL, spec TEXT NOT NULL, UNIQUE (tenant_id, "
+       "canonical_id))"},
+  });
   g_registry = std::make_shared<LocalModelRegistry>(
       ctx.config,
       ctx.full_config.path("modules.llm-gateway.config.model")
-          .as_string(""));
+          .as_string(""),
+      std::move(db));
   ctx.hub->register_client<ModelRegistryClient>("model-registry",
                                                 g_registry);
 }
@@ -593,6 +701,75 @@ void ModelRegistryModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
     out["canonical_id"] = cid;
     out["status"] = st;
     w.respond(200, "application/json", out.dump());
+  });
+
+  // tenant model registration w/ shadow-by-canonical-id (PRD.md:181-188)
+  OperationSpec post;
+  post.method = "POST";
+  post.path = "/model-registry/v1/models";
+  post.operation_id = "models_register";
+  post.summary = "Register/shadow a model for this tenant";
+  post.authenticated = true;
+  post.allowed_content_types = {"application/json"};
+  post.tags = {"model-registry"};
+  rest.register_op(post, [](HttpRequest& rq, ResponseWriter& w) {
+    Json body;
+    try { body = Json::parse(rq.body); }
+    catch (...) { throw Problem::bad_request("invalid JSON body"); }
+    const std::string cid = body.at("canonical_id").as_string();
+    if (cid.find("::") == std::string::npos)
+      throw Problem::bad_request(
+          "canonical_id must be {provider_slug}::{model_id}");
+    g_registry->put_model(sec_of(rq).tenant_id, body);
+    w.respond(201, "application/json", body.dump());
+  });
+
+  // aliases: name -> canonical_id with tenant->global shadowing
+  // (PRD.md:298-306)
+  OperationSpec aput;
+  aput.method = "PUT";
+  aput.path = "/model-registry/v1/aliases/{name}";
+  aput.operation_id = "alias_put";
+  aput.authenticated = true;
+  aput.allowed_content_types = {"application/json"};
+  aput.tags = {"model-registry"};
+  rest.register_op(aput, [](HttpRequest& rq, ResponseWriter& w) {
+    Json body;
+    try { body = Json::parse(rq.body); }
+    catch (...) { throw Problem::bad_request("invalid JSON body"); }
+    const std::string cid = body.at("canonical_id").as_string();
+    if (cid.empty())
+      throw Problem::bad_request("'canonical_id' required");
+    g_registry->put_alias(sec_of(rq).tenant_id, rq.path_params["name"],
+                          cid);
+    w.respond(204, "application/json", "");
+  });
+  OperationSpec aget;
+  aget.method = "GET";
+  aget.path = "/model-registry/v1/aliases/{name}";
+  aget.operation_id = "alias_get";
+  aget.authenticated = true;
+  aget.tags = {"model-registry"};
+  rest.register_op(aget, [](HttpRequest& rq, ResponseWriter& w) {
+    auto cid = g_registry->alias_lookup(sec_of(rq).tenant_id,
+                                        rq.path_params["name"]);
+    if (!cid) throw Problem::not_found("no such alias");
+    Json out = Json::object();
+    out["name"] = rq.path_params["name"];
+    out["canonical_id"] = *cid;
+    w.respond(200, "application/json", out.dump());
+  });
+  OperationSpec adel;
+  adel.method = "DELETE";
+  adel.path = "/model-registry/v1/aliases/{name}";
+  adel.operation_id = "alias_delete";
+  adel.authenticated = true;
+  adel.tags = {"model-registry"};
+  rest.register_op(adel, [](HttpRequest& rq, ResponseWriter& w) {
+    if (!g_registry->del_alias(sec_of(rq).tenant_id,
+                               rq.path_params["name"]))
+      throw Problem::not_found("no such alias");
+    w.respond(204, "application/json", "");
   });
 }
 
