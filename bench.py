@@ -415,6 +415,11 @@ def rest_bench_rank0(args, n_gpus, on_gpu):
         time.sleep(1.0)                     # let in-flight SSE drain
         client.stop()
 
+        if "ms_engine" in ok[0]:
+            log("worker step breakdown: "
+                f"lock {ok[0]['ms_lock']:.2f} ms, "
+                f"engine {ok[0]['ms_engine']:.2f} ms, "
+                f"fanout {ok[0]['ms_fanout']:.2f} ms")
         w_t0 = max(r["t0"] for r in ok)
         w_t1 = min(r["t1"] for r in ok)
         window = w_t1 - w_t0

@@ -63,6 +63,7 @@ class WorkerState:
         self.bench_result = queue.Queue()
         self.pending_exec = []                # (kind, arg, reply_q)
         self.mux_of = {}                      # rid -> MuxChannel
+        self._bt_lock = self._bt_step = self._bt_fanout = 0.0
 
     def submit_mux(self, rid, prompt_ids, sampling, mux, detok):
         mux.track(rid, detok, len(prompt_ids))
@@ -188,13 +189,20 @@ class WorkerState:
 
     def _one_step(self):
         import torch.distributed as dist
+        t0 = time.perf_counter()
         with self.new_work:
+            t1 = time.perf_counter()
             if self.tp > 1:
                 ops = self.pending_ops
                 self.pending_ops = []
                 dist.broadcast_object_list([("step", ops)], src=0)
             outputs = self.engine.step()
+        t2 = time.perf_counter()
         self._fanout(outputs)
+        t3 = time.perf_counter()
+        self._bt_lock += t1 - t0
+        self._bt_step += t2 - t1
+        self._bt_fanout += t3 - t2
         return len(outputs)
 
     def _sync(self):
@@ -224,6 +232,7 @@ class WorkerState:
             for _ in range(warmup):
                 self._one_step()
             self._sync()
+            self._bt_lock = self._bt_step = self._bt_fanout = 0.0
             t0 = time.time()
             produced = 0
             for _ in range(steps):
@@ -233,7 +242,11 @@ class WorkerState:
             self.bench_result.put({
                 "event": "bench_done", "t0": t0, "t1": t1,
                 "elapsed": t1 - t0, "produced": produced,
-                "steps": steps, "warmup": warmup})
+                "steps": steps, "warmup": warmup,
+                # per-step breakdown (ms): lock wait, engine.step, fanout
+                "ms_lock": self._bt_lock / steps * 1000,
+                "ms_engine": self._bt_step / steps * 1000,
+                "ms_fanout": self._bt_fanout / steps * 1000})
         except Exception as e:           # report instead of killing loop
             self.bench_result.put({"event": "error",
                                    "message": f"bench failed: {e}"})
