@@ -98,6 +98,83 @@ std::optional<Json> EngineConn::read_json(int timeout_ms) {
   }
 }
 
+// ------------------------------------------------------------- ByteDetok
+
+namespace {
+
+// bytes [0, n) that form complete UTF-8 sequences; -1 byte position of
+// the first invalid byte, or n if only an incomplete tail remains
+size_t utf8_complete_prefix(const std::string& b, size_t* bad) {
+  *bad = std::string::npos;
+  size_t i = 0;
+  while (i < b.size()) {
+    unsigned char c = b[i];
+    int need;
+    if (c < 0x80) need = 0;
+    else if ((c & 0xe0) == 0xc0 && c >= 0xc2) need = 1;
+    else if ((c & 0xf0) == 0xe0) need = 2;
+    else if ((c & 0xf8) == 0xf0 && c <= 0xf4) need = 3;
+    else { *bad = i; return i; }
+    if (i + 1 + need > b.size()) return i;     // incomplete tail
+    for (int k = 1; k <= need; ++k)
+      if ((static_cast<unsigned char>(b[i + k]) & 0xc0) != 0x80) {
+        *bad = i;
+        return i;
+      }
+    i += 1 + need;
+  }
+  return i;
+}
+
+// decode with U+FFFD for invalid bytes (python errors="replace")
+std::string utf8_replace(const std::string& b) {
+  std::string out;
+  size_t i = 0;
+  while (i < b.size()) {
+    size_t bad;
+    std::string rest = b.substr(i);
+    size_t ok = utf8_complete_prefix(rest, &bad);
+    out += rest.substr(0, ok);
+    if (ok == rest.size()) break;
+    out += "\xEF\xBF\xBD";        // U+FFFD
+    i += ok + 1;
+  }
+  return out;
+}
+
+}  // namespace
+
+std::string ByteDetok::push(long tok) {
+  if (tok < 4 || tok >= 260) {
+    std::string pre = flush();
+    if (tok == 1 || tok == 2) return pre;       // BOS/EOS silent
+    return pre + "<" + std::to_string(tok) + ">";
+  }
+  buf_.push_back(static_cast<char>(tok - 4));
+  size_t bad;
+  size_t ok = utf8_complete_prefix(buf_, &bad);
+  if (ok == buf_.size()) {                      // fully valid
+    std::string out;
+    out.swap(buf_);
+    return out;
+  }
+  if (ok > 0) {                                 // emit valid prefix
+    std::string out = buf_.substr(0, ok);
+    buf_.erase(0, ok);
+    // python re-checks the remainder next push; if the head byte is
+    // invalid it stays until 4 bytes accumulate — mirror that
+    return out;
+  }
+  if (buf_.size() >= 4) return flush();         // invalid, give up
+  return "";
+}
+
+std::string ByteDetok::flush() {
+  std::string out = utf8_replace(buf_);
+  buf_.clear();
+  return out;
+}
+
 // ------------------------------------------------------------- MuxClient
 
 MuxClient::MuxClient(const std::string& socket_path) {
@@ -145,26 +222,30 @@ void MuxClient::reader_loop() {
       const Json& items = msg->at("items");
       std::lock_guard<std::mutex> lk(mu_);
       for (auto& it : items.arr()) {
-        // [rid, token_id, text] delta | [rid, tok, text, fin, in, out]
+        // [rid, token_id] delta | [rid, tok, fin, in, out] final —
+        // detokenization happens HERE (C++, off the worker's GPU loop)
         const std::string rid = it.at(0).as_string();
         auto f = sinks_.find(rid);
         if (f == sinks_.end()) continue;       // aborted meanwhile
         auto& s = *f->second;
-        const bool fin = it.size() > 3;
+        const bool fin = it.size() > 2;
+        const long tok = it.at(1).as_int(0);
+        std::string text = s.detok.push(tok);
+        if (fin) text += s.detok.flush();
         {
           std::lock_guard<std::mutex> sl(s.mu);
           Json d = Json::object();
           d["event"] = "delta";
-          d["token_id"] = it.at(1);
-          d["text"] = it.at(2);
+          d["token_id"] = (long)tok;
+          d["text"] = text;
           s.q.push_back(std::move(d));
           if (fin) {
             Json u = Json::object();
-            u["input_tokens"] = it.at(4);
-            u["output_tokens"] = it.at(5);
+            u["input_tokens"] = it.at(3);
+            u["output_tokens"] = it.at(4);
             Json dn = Json::object();
             dn["event"] = "done";
-            dn["finish_reason"] = it.at(3);
+            dn["finish_reason"] = it.at(2);
             dn["usage"] = u;
             s.q.push_back(std::move(dn));
           }

@@ -64,11 +64,27 @@ class EngineConn {
 // into per-request sinks consumed by the SSE/blocking chat handlers.
 // This keeps the worker's GPU loop at one json-encode + one send per
 // step instead of per-token fan-out to thousands of connections.
+// Incremental byte-level detokenizer — C++ mirror of the worker's
+// StreamDetokenizer (hyperspot/serving/tokenizer.py): byte tokens are
+// id-4, BOS=1/EOS=2 are silent, other ids render as "<id>"; only
+// complete UTF-8 sequences are emitted, invalid bytes become U+FFFD.
+// Running it gateway-side keeps the worker's batch lines to bare
+// token ids (no per-token Python text work on the GPU host process).
+class ByteDetok {
+ public:
+  std::string push(long tok);
+  std::string flush();
+
+ private:
+  std::string buf_;              // pending (possibly incomplete) UTF-8
+};
+
 struct MuxSink {
   std::mutex mu;
   std::condition_variable cv;
   std::deque<Json> q;            // delta/done/error events
   bool dead = false;             // channel lost before completion
+  ByteDetok detok;               // owned by the mux reader thread
 };
 
 class MuxClient {

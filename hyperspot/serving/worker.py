@@ -63,16 +63,19 @@ class WorkerState:
         self.bench_result = queue.Queue()
         self.pending_exec = []                # (kind, arg, reply_q)
         self.mux_of = {}                      # rid -> MuxChannel
+        self.submit_mu = threading.Lock()
+        self.pending_submits = []             # (rid, ids, sampling, mux)
         self._bt_lock = self._bt_step = self._bt_fanout = 0.0
 
     def submit_mux(self, rid, prompt_ids, sampling, mux, detok):
+        """Enqueue only: the stepping thread drains pending submits in
+        batches at step boundaries.  Taking the engine lock here would
+        convoy behind the in-flight step (one admission per step — a
+        2048-stream ramp took 100+ s on MI355X before this)."""
         mux.track(rid, detok, len(prompt_ids))
+        with self.submit_mu:
+            self.pending_submits.append((rid, prompt_ids, sampling, mux))
         with self.new_work:
-            self.mux_of[rid] = mux
-            self.engine.add_request(prompt_ids, sampling, request_id=rid)
-            if self.tp > 1:
-                self.pending_ops.append(
-                    ("add", rid, prompt_ids, sampling.__dict__.copy()))
             self.new_work.notify()
 
     def submit(self, rid, prompt_ids, sampling):
@@ -88,6 +91,9 @@ class WorkerState:
         return q
 
     def abort(self, rid):
+        with self.submit_mu:
+            self.pending_submits = [t for t in self.pending_submits
+                                    if t[0] != rid]
         with self.new_work:
             self.engine.abort_request(rid)
             q = self.streams.pop(rid, None)
@@ -119,6 +125,20 @@ class WorkerState:
             raise res
         return res
 
+    def _drain_submits_locked(self):
+        """Admit queued mux submissions (caller holds the engine lock)."""
+        if not self.pending_submits:
+            return
+        with self.submit_mu:
+            subs = self.pending_submits
+            self.pending_submits = []
+        for rid, prompt_ids, sampling, mux in subs:
+            self.mux_of[rid] = mux
+            self.engine.add_request(prompt_ids, sampling, request_id=rid)
+            if self.tp > 1:
+                self.pending_ops.append(
+                    ("add", rid, prompt_ids, sampling.__dict__.copy()))
+
     def _exec(self, kind, arg):
         if kind == "swap":
             return self.engine.swap_weights(arg)
@@ -135,7 +155,7 @@ class WorkerState:
                 self._run_bench()
                 continue
             with self.new_work:
-                while not self.engine.has_work() and not self.stop                         and not self.pending_ops and self.bench_req is None                         and not self.pending_exec:
+                while not self.engine.has_work() and not self.stop                         and not self.pending_ops and self.bench_req is None                         and not self.pending_exec and not self.pending_submits:
                     self.new_work.wait(timeout=0.5)
                 if self.stop:
                     if self.tp > 1:
@@ -151,6 +171,7 @@ class WorkerState:
                         q.put(self._exec(kind, arg))
                     except Exception as e:
                         q.put(e)
+                self._drain_submits_locked()
                 if not self.engine.has_work() and not self.pending_ops:
                     continue
                 if self.tp > 1:
@@ -192,6 +213,7 @@ class WorkerState:
         t0 = time.perf_counter()
         with self.new_work:
             t1 = time.perf_counter()
+            self._drain_submits_locked()
             if self.tp > 1:
                 ops = self.pending_ops
                 self.pending_ops = []
@@ -267,18 +289,22 @@ class MuxChannel:
         self.pending = []           # bytes | list of raw outputs
         self.cv = threading.Condition()
         self.dead = False
+        # meta is owned by the WRITER thread; track/forget only enqueue
+        # deltas so they never block behind a batch encode
         self.meta = {}              # rid -> [detok, prompt_len, n_out]
+        self.meta_add = []          # (rid, detok, prompt_len)
+        self.meta_del = []
         self.writer = threading.Thread(target=self._write_loop,
                                        daemon=True)
         self.writer.start()
 
     def track(self, rid, detok, prompt_len):
         with self.cv:
-            self.meta[rid] = [detok, prompt_len, 0]
+            self.meta_add.append((rid, detok, prompt_len))
 
     def forget(self, rid):
         with self.cv:
-            self.meta.pop(rid, None)
+            self.meta_del.append(rid)
 
     def send_outputs(self, raw):
         """Raw (rid, token_id, finish_reason|None) tuples from the GPU
@@ -328,11 +354,19 @@ class MuxChannel:
                     return
                 work = self.pending
                 self.pending = []
-                # encode under the lock: meta is shared with
-                # track/forget, and batches must stay FIFO
-                chunk = b"".join(
-                    w if isinstance(w, bytes) else self._encode(w)
-                    for w in work)
+                adds = self.meta_add
+                self.meta_add = []
+                dels = self.meta_del
+                self.meta_del = []
+            # merge meta deltas, then encode OUTSIDE the lock (the
+            # single writer keeps batches FIFO by construction)
+            for rid, detok, plen in adds:
+                self.meta[rid] = [detok, plen, 0]
+            for rid in dels:
+                self.meta.pop(rid, None)
+            chunk = b"".join(
+                w if isinstance(w, bytes) else self._encode(w)
+                for w in work)
             try:
                 self.sock.sendall(chunk)
             except (BrokenPipeError, OSError):
