@@ -18,13 +18,14 @@ Multiplexed mode (the C++ gateway's serving path): the gateway opens ONE
 connection, sends {"type":"attach_mux"}, then submits every chat stream
 over it.  The stepping thread emits ONE batched line per engine step:
 
-  <- {"event":"batch","items":[[rid,token_id,text], ...               # delta
-                               [rid,token_id,text,finish,in,out],...]}# final
+  <- {"event":"batch","items":[[rid,token_id], ...                  # delta
+                               [rid,token_id,finish,in,out], ...]}  # final
 
-so the per-token Python cost is one list-append + one shared json.dumps
-instead of per-request queue handoffs to thousands of connection threads
-(measured on MI355X: per-request fan-out at batch 2048 tripled the step
-time; the batch line keeps the GPU loop at engine speed).
+Batch items carry bare token ids — the C++ gateway detokenizes
+(ByteDetok mirrors StreamDetokenizer), so the per-token Python cost is
+one tuple-append + a shared json.dumps of ints (measured on MI355X:
+per-request fan-out at batch 2048 tripled the step time; per-token
+Python detok cost another ~3 ms/step of GIL).
 """
 
 from __future__ import annotations
@@ -67,12 +68,12 @@ class WorkerState:
         self.pending_submits = []             # (rid, ids, sampling, mux)
         self._bt_lock = self._bt_step = self._bt_fanout = 0.0
 
-    def submit_mux(self, rid, prompt_ids, sampling, mux, detok):
+    def submit_mux(self, rid, prompt_ids, sampling, mux):
         """Enqueue only: the stepping thread drains pending submits in
         batches at step boundaries.  Taking the engine lock here would
         convoy behind the in-flight step (one admission per step — a
         2048-stream ramp took 100+ s on MI355X before this)."""
-        mux.track(rid, detok, len(prompt_ids))
+        mux.track(rid, len(prompt_ids))
         with self.submit_mu:
             self.pending_submits.append((rid, prompt_ids, sampling, mux))
         with self.new_work:
@@ -298,9 +299,9 @@ class MuxChannel:
                                        daemon=True)
         self.writer.start()
 
-    def track(self, rid, detok, prompt_len):
+    def track(self, rid, prompt_len):
         with self.cv:
-            self.meta_add.append((rid, detok, prompt_len))
+            self.meta_add.append((rid, prompt_len))
 
     def forget(self, rid):
         with self.cv:
@@ -334,13 +335,11 @@ class MuxChannel:
             m = meta.get(rid)
             if m is None:
                 continue               # aborted after the step ran
-            text = m[0].push(tok)
-            m[2] += 1
+            m[1] += 1
             if fr is None:
-                items.append([rid, tok, text])
+                items.append([rid, tok])
             else:
-                text += m[0].flush()
-                items.append([rid, tok, text, fr, m[1], m[2]])
+                items.append([rid, tok, fr, m[0], m[1]])
                 meta.pop(rid, None)
         return (json.dumps({"event": "batch", "items": items},
                            separators=(",", ":")) + "\n").encode()
@@ -360,8 +359,8 @@ class MuxChannel:
                 self.meta_del = []
             # merge meta deltas, then encode OUTSIDE the lock (the
             # single writer keeps batches FIFO by construction)
-            for rid, detok, plen in adds:
-                self.meta[rid] = [detok, plen, 0]
+            for rid, plen in adds:
+                self.meta[rid] = [plen, 0]
             for rid in dels:
                 self.meta.pop(rid, None)
             chunk = b"".join(
@@ -382,9 +381,10 @@ class MuxChannel:
 
 def run_mux_conn(f, conn, state: WorkerState, send):
     """Serve one attached mux connection: read chat/abort commands; the
-    stepping thread writes the batched responses."""
+    stepping thread writes the batched responses (token ids only — the
+    gateway detokenizes)."""
     from hyperspot.engine import SamplingParams
-    from .tokenizer import StreamDetokenizer, render_chat
+    from .tokenizer import render_chat
 
     mux = MuxChannel(conn)
     rids = set()
@@ -418,9 +418,8 @@ def run_mux_conn(f, conn, state: WorkerState, send):
                                    budget),
                     ignore_eos=bool(params.get("ignore_eos", False)),
                 )
-                detok = StreamDetokenizer(state.tokenizer)
                 rids.add(rid)
-                state.submit_mux(rid, prompt_ids, sampling, mux, detok)
+                state.submit_mux(rid, prompt_ids, sampling, mux)
             elif t == "abort":
                 rid = msg.get("id", "")
                 rids.discard(rid)
