@@ -384,14 +384,35 @@ def rest_bench_rank0(args, n_gpus, on_gpu):
         client = LoadClient(port, body, total_streams)
         client.start()
         t0 = time.time()
-        wait_saturated(port, args.batch)
+        last_log = -100.0
+        def ramp_progress():
+            nonlocal last_log
+            if time.time() - last_log < 10:
+                return
+            last_log = time.time()
+            eng = {}
+            try:
+                _, j = _http_json(
+                    "GET",
+                    f"http://127.0.0.1:{port}/llm-gateway/v1/status",
+                    timeout=5)
+                eng = j.get("workers", [{}])[0].get("engine", {})
+            except Exception:
+                pass
+            log(f"ramp t={time.time()-t0:.0f}s "
+                f"started={client.streams_started}/{total_streams} "
+                f"live={client.streams_live} errors={client.errors} "
+                f"running={eng.get('num_running')} "
+                f"waiting={eng.get('num_waiting')}")
         while (client.streams_started < total_streams * 0.99
                and time.time() - t0 < 900):
+            ramp_progress()
             time.sleep(0.5)
         if client.streams_started < total_streams * 0.95:
             raise RuntimeError(
                 f"only {client.streams_started}/{total_streams} streams "
                 f"came up ({client.errors} errors)")
+        wait_saturated(port, args.batch, timeout=300)  # prefills done
         log(f"{client.streams_started}/{total_streams} streams decoding "
             f"({time.time()-t0:.0f}s ramp, {client.errors} conn errors)")
 
