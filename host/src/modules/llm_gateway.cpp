@@ -190,11 +190,14 @@ MuxClient::MuxClient(const std::string& socket_path) {
   }
   attached_ = true;
   reader_ = std::thread([this] { reader_loop(); });
+  flusher_ = std::thread([this] { flusher_loop(); });
 }
 
 MuxClient::~MuxClient() {
   stop_ = true;
   closed_ = true;
+  sub_cv_.notify_all();
+  if (flusher_.joinable()) flusher_.join();
   conn_.reset();                 // closes the fd; reader unblocks
   if (reader_.joinable()) reader_.join();
   fail_all();
@@ -276,18 +279,45 @@ std::shared_ptr<MuxSink> MuxClient::submit(const Json& wreq) {
     std::lock_guard<std::mutex> lk(mu_);
     if (closed_) return nullptr;
     sinks_[rid] = sink;
-    if (!conn_->send_json(wreq)) {
-      sinks_.erase(rid);
+    sub_q_.push_back(wreq);
+  }
+  sub_cv_.notify_one();
+  return sink;
+}
+
+void MuxClient::flusher_loop() {
+  while (!stop_) {
+    std::vector<Json> batch;
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      sub_cv_.wait_for(lk, std::chrono::milliseconds(1),
+                       [&] { return !sub_q_.empty() || stop_; });
+      if (stop_ && sub_q_.empty()) return;
+      batch.swap(sub_q_);
+    }
+    if (batch.empty()) continue;
+    Json reqs = Json::array();
+    for (auto& b : batch) reqs.push_back(std::move(b));
+    Json line = Json::object();
+    line["type"] = "chat_batch";
+    line["reqs"] = std::move(reqs);
+    std::lock_guard<std::mutex> lk(mu_);
+    if (closed_) return;
+    if (!conn_->send_json(line)) {
       closed_ = true;
-      return nullptr;
+      return;
     }
   }
-  return sink;
 }
 
 void MuxClient::abort(const std::string& rid) {
   std::lock_guard<std::mutex> lk(mu_);
   sinks_.erase(rid);
+  for (auto it = sub_q_.begin(); it != sub_q_.end(); ++it)
+    if (it->at("id").as_string() == rid) {
+      sub_q_.erase(it);
+      return;                       // never reached the worker
+    }
   if (closed_) return;
   Json ab = Json::object();
   ab["type"] = "abort";

@@ -379,6 +379,32 @@ class MuxChannel:
             self.cv.notify()
 
 
+def _mux_chat(msg, state: "WorkerState", mux: "MuxChannel", rids: set):
+    from hyperspot.engine import SamplingParams
+    from .tokenizer import render_chat
+
+    rid = msg.get("id") or f"r{time.monotonic_ns()}"
+    params = msg.get("params") or {}
+    prompt_ids = msg.get("prompt_ids")
+    if prompt_ids is None:
+        text = render_chat(msg.get("messages") or [])
+        prompt_ids = state.tokenizer.encode(text, add_bos=True)
+    budget = state.engine.config.max_model_len - len(prompt_ids) - 1
+    if budget <= 0:
+        mux.send_obj({"event": "error", "id": rid,
+                      "message": "prompt exceeds context window"})
+        return
+    sampling = SamplingParams(
+        temperature=float(params.get("temperature", 0.7)),
+        top_p=float(params.get("top_p", 1.0)),
+        top_k=int(params.get("top_k", 0)),
+        max_tokens=min(int(params.get("max_tokens", 256)), budget),
+        ignore_eos=bool(params.get("ignore_eos", False)),
+    )
+    rids.add(rid)
+    state.submit_mux(rid, prompt_ids, sampling, mux)
+
+
 def run_mux_conn(f, conn, state: WorkerState, send):
     """Serve one attached mux connection: read chat/abort commands; the
     stepping thread writes the batched responses (token ids only — the
@@ -396,30 +422,14 @@ def run_mux_conn(f, conn, state: WorkerState, send):
                 mux.send_obj({"event": "error", "message": "bad json"})
                 continue
             t = msg.get("type")
+            if t == "chat_batch":
+                # one GIL slice admits the whole burst (the gateway
+                # micro-batches submissions for exactly this reason)
+                for r in msg.get("reqs") or []:
+                    _mux_chat(r, state, mux, rids)
+                continue
             if t == "chat":
-                rid = msg.get("id") or f"r{time.monotonic_ns()}"
-                params = msg.get("params") or {}
-                prompt_ids = msg.get("prompt_ids")
-                if prompt_ids is None:
-                    text = render_chat(msg.get("messages") or [])
-                    prompt_ids = state.tokenizer.encode(text, add_bos=True)
-                budget = state.engine.config.max_model_len \
-                    - len(prompt_ids) - 1
-                if budget <= 0:
-                    mux.send_obj({"event": "error", "id": rid,
-                                  "message": "prompt exceeds context "
-                                             "window"})
-                    continue
-                sampling = SamplingParams(
-                    temperature=float(params.get("temperature", 0.7)),
-                    top_p=float(params.get("top_p", 1.0)),
-                    top_k=int(params.get("top_k", 0)),
-                    max_tokens=min(int(params.get("max_tokens", 256)),
-                                   budget),
-                    ignore_eos=bool(params.get("ignore_eos", False)),
-                )
-                rids.add(rid)
-                state.submit_mux(rid, prompt_ids, sampling, mux)
+                _mux_chat(msg, state, mux, rids)
             elif t == "abort":
                 rid = msg.get("id", "")
                 rids.discard(rid)
@@ -658,6 +668,10 @@ def main():
 
     logging.basicConfig(level=logging.INFO,
                         format="%(asctime)s %(levelname)s %(name)s %(message)s")
+    # the serving threads (mux reader/writer) share the GIL with the hot
+    # stepping thread; the default 5 ms switch interval starves them
+    import sys as _s
+    _s.setswitchinterval(0.002)
     import torch
     from hyperspot.engine import EngineConfig, LLMEngine
     from hyperspot.serving.tokenizer import ByteTokenizer
