@@ -958,6 +958,7 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
       lease.w->ready = false;
       continue;
     }
+    m_submits_++;
     first_msg = MuxClient::next_event(
         *sink, ttft_timeout_ms_ ? (int)ttft_timeout_ms_ : 120000);
     if (!first_msg) {
@@ -1778,6 +1779,8 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
     Json out = Json::object();
     out["model"] = model_;
     out["worker_ready"] = worker_ready();
+    out["streams_total"] = (long)m_streams_.load();
+    out["submits_total"] = (long)m_submits_.load();
     Json ws_json = Json::array();
     for (auto& wk : workers_) {
       Json wj = Json::object();

@@ -243,6 +243,7 @@ class LlmGatewayModule : public Module {
 
   // metrics
   std::atomic<uint64_t> m_requests_{0}, m_streams_{0}, m_errors_{0};
+  std::atomic<uint64_t> m_submits_{0};
   std::atomic<uint64_t> m_input_tokens_{0}, m_output_tokens_{0};
   std::atomic<uint64_t> m_ttft_us_sum_{0}, m_ttft_count_{0};
 };

@@ -66,6 +66,7 @@ class WorkerState:
         self.mux_of = {}                      # rid -> MuxChannel
         self.submit_mu = threading.Lock()
         self.pending_submits = []             # (rid, ids, sampling, mux)
+        self.mux_channels = []                # live MuxChannel list
         self._bt_lock = self._bt_step = self._bt_fanout = 0.0
 
     def submit_mux(self, rid, prompt_ids, sampling, mux):
@@ -126,6 +127,70 @@ class WorkerState:
             raise res
         return res
 
+    def drain_mux_inputs(self):
+        """Read + process pending mux-channel commands on the STEPPING
+        thread.  A dedicated Python reader thread starves for the GIL
+        behind the hot stepping loop (measured: 256 queued submissions
+        admitted at ~10/s on CPU, ~40/s on MI355X); the stepper already
+        holds the GIL at step boundaries, so it drains the sockets
+        non-blockingly and admits whole bursts at once."""
+        for mux in list(self.mux_channels):
+            data_parts = []
+            dead = False
+            while True:
+                try:
+                    chunk = mux.sock.recv(1 << 20, socket.MSG_DONTWAIT)
+                except (BlockingIOError, InterruptedError):
+                    break
+                except OSError:
+                    dead = True
+                    break
+                if not chunk:
+                    dead = True
+                    break
+                data_parts.append(chunk)
+            if data_parts:
+                mux.rx_buf += b"".join(data_parts)
+                while True:
+                    nl = mux.rx_buf.find(b"\n")
+                    if nl < 0:
+                        break
+                    line = mux.rx_buf[:nl]
+                    mux.rx_buf = mux.rx_buf[nl + 1:]
+                    if line:
+                        self._mux_command(mux, line)
+            if dead:
+                mux.close()
+                self.mux_channels.remove(mux)
+                for rid in list(self.mux_of):
+                    if self.mux_of.get(rid) is mux:
+                        self.abort(rid)
+            mux.rx_event.set()      # waker thread may re-arm select
+
+    def _mux_command(self, mux, line):
+        try:
+            msg = json.loads(line)
+        except json.JSONDecodeError:
+            mux.send_obj({"event": "error", "message": "bad json"})
+            return
+        t = msg.get("type")
+        if t == "chat_batch":
+            for r in msg.get("reqs") or []:
+                _mux_chat(r, self, mux, mux.rids)
+        elif t == "chat":
+            _mux_chat(msg, self, mux, mux.rids)
+        elif t == "abort":
+            rid = msg.get("id", "")
+            mux.rids.discard(rid)
+            self.abort(rid)
+        elif t == "info":
+            mux.send_obj({"event": "info", "ready": True,
+                          "num_running": self.engine.num_running,
+                          "num_waiting": self.engine.num_waiting})
+        else:
+            mux.send_obj({"event": "error",
+                          "message": f"unknown type {t!r}"})
+
     def _drain_submits_locked(self):
         """Admit queued mux submissions (caller holds the engine lock)."""
         if not self.pending_submits:
@@ -155,9 +220,12 @@ class WorkerState:
             if self.bench_req is not None:
                 self._run_bench()
                 continue
+            self.drain_mux_inputs()
             with self.new_work:
                 while not self.engine.has_work() and not self.stop                         and not self.pending_ops and self.bench_req is None                         and not self.pending_exec and not self.pending_submits:
                     self.new_work.wait(timeout=0.5)
+                    if self.mux_channels:
+                        break          # waker notified or timeout: drain
                 if self.stop:
                     if self.tp > 1:
                         dist.broadcast_object_list([("stop",)], src=0)
@@ -211,6 +279,7 @@ class WorkerState:
 
     def _one_step(self):
         import torch.distributed as dist
+        self.drain_mux_inputs()
         t0 = time.perf_counter()
         with self.new_work:
             t1 = time.perf_counter()
@@ -287,6 +356,9 @@ class MuxChannel:
 
     def __init__(self, sock: socket.socket):
         self.sock = sock
+        self.rx_buf = b""           # inbound bytes (drained by stepper)
+        self.rx_event = threading.Event()
+        self.rids = set()
         self.pending = []           # bytes | list of raw outputs
         self.cv = threading.Condition()
         self.dead = False
@@ -406,49 +478,30 @@ def _mux_chat(msg, state: "WorkerState", mux: "MuxChannel", rids: set):
 
 
 def run_mux_conn(f, conn, state: WorkerState, send):
-    """Serve one attached mux connection: read chat/abort commands; the
-    stepping thread writes the batched responses (token ids only — the
-    gateway detokenizes)."""
-    from hyperspot.engine import SamplingParams
-    from .tokenizer import render_chat
+    """Attach a mux channel.  This thread does NOT parse commands (a
+    Python reader starves for the GIL behind the stepping loop); it is
+    a pure WAKER: select() for readability, nudge the stepping thread,
+    wait for it to drain, repeat.  All parsing/admission happens on the
+    stepping thread (WorkerState.drain_mux_inputs)."""
+    import select
 
+    # any buffered bytes the line reader already consumed belong to the
+    # channel (the attach line was read via `f`; switch to raw socket)
     mux = MuxChannel(conn)
-    rids = set()
+    state.mux_channels.append(mux)
     try:
-        for raw in f:
-            try:
-                msg = json.loads(raw)
-            except json.JSONDecodeError:
-                mux.send_obj({"event": "error", "message": "bad json"})
+        while not mux.dead and not state.stop:
+            r, _, _ = select.select([conn], [], [], 0.5)
+            if not r:
                 continue
-            t = msg.get("type")
-            if t == "chat_batch":
-                # one GIL slice admits the whole burst (the gateway
-                # micro-batches submissions for exactly this reason)
-                for r in msg.get("reqs") or []:
-                    _mux_chat(r, state, mux, rids)
-                continue
-            if t == "chat":
-                _mux_chat(msg, state, mux, rids)
-            elif t == "abort":
-                rid = msg.get("id", "")
-                rids.discard(rid)
-                state.abort(rid)
-            elif t == "info":
-                mux.send_obj({
-                    "event": "info", "ready": True,
-                    "num_running": state.engine.num_running,
-                    "num_waiting": state.engine.num_waiting})
-            else:
-                mux.send_obj({"event": "error",
-                              "message": f"unknown type {t!r}"})
-    except (ConnectionResetError, BrokenPipeError, OSError):
+            mux.rx_event.clear()
+            with state.new_work:
+                state.new_work.notify()
+            mux.rx_event.wait(timeout=0.2)
+    except OSError:
         pass
     finally:
         mux.close()
-        for rid in list(rids):    # gateway gone: stop its generations
-            if rid in state.mux_of:
-                state.abort(rid)
 
 
 def follower_loop(engine):
