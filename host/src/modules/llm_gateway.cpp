@@ -3,6 +3,7 @@
 #include <poll.h>
 #include <signal.h>
 #include <sys/socket.h>
+#include <sys/stat.h>
 #include <sys/un.h>
 #include <sys/wait.h>
 #include <unistd.h>
@@ -367,6 +368,35 @@ void LlmGatewayModule::init(ModuleCtx& ctx) {
   ctx.hub->register_client<ChatInvoker>(
       "llm-gateway",
       std::make_shared<GatewayChatInvoker>(this));
+  {
+    std::string file = ctx.full_config
+                           .path("modules.llm-gateway.database.file")
+                           .as_string("");
+    if (file.empty()) {
+      std::string home = ctx.full_config.path("server.home_dir")
+                             .as_string("~/.hyperspot");
+      if (!home.empty() && home[0] == '~') {
+        const char* h = getenv("HOME");
+        home = std::string(h ? h : "/tmp") + home.substr(1);
+      }
+      mkdir(home.c_str(), 0755);
+      file = home + "/llm-gateway.db";
+    }
+    jobs_db_ = std::make_unique<Db>(file);
+    jobs_db_->migrate("llm-gateway", {
+        {"0001_jobs",
+         "CREATE TABLE jobs (tenant_id TEXT NOT NULL, id TEXT NOT NULL "
+         "UNIQUE, batch_id TEXT NOT NULL DEFAULT '', status TEXT NOT "
+         "NULL, request TEXT NOT NULL, result TEXT NOT NULL DEFAULT "
+         "'', error TEXT NOT NULL DEFAULT '', created_at REAL NOT "
+         "NULL, finished_at REAL NOT NULL DEFAULT 0)"},
+        {"0002_batches",
+         "CREATE TABLE batches (tenant_id TEXT NOT NULL, id TEXT NOT "
+         "NULL UNIQUE, job_ids TEXT NOT NULL, created_at REAL NOT "
+         "NULL)"},
+    });
+    load_jobs();
+  }
   model_ = ctx.config.at("model").as_string(model_);
   socket_path_ = ctx.config.at("worker_socket").as_string(socket_path_);
   auto_start_ = ctx.config.at("auto_start_worker").as_bool(true);
@@ -1137,6 +1167,89 @@ Json LlmGatewayModule::job_json(const Job& j) const {
   return o;
 }
 
+void LlmGatewayModule::persist_job(const Job& j) {
+  if (!jobs_db_) return;
+  std::lock_guard<std::mutex> lk(jobs_db_->mu());
+  jobs_db_->query(
+      "INSERT INTO jobs (tenant_id, id, batch_id, status, request, "
+      "result, error, created_at, finished_at) VALUES "
+      "(?,?,?,?,?,?,?,?,?) ON CONFLICT(id) DO UPDATE SET "
+      "status=excluded.status, result=excluded.result, "
+      "error=excluded.error, finished_at=excluded.finished_at",
+      {DbValue::S(j.tenant), DbValue::S(j.id), DbValue::S(j.batch_id),
+       DbValue::S(j.status), DbValue::S(j.request.dump()),
+       DbValue::S(j.result.is_null() ? "" : j.result.dump()),
+       DbValue::S(j.error), DbValue::R(j.created_at),
+       DbValue::R(j.finished_at)});
+}
+
+void LlmGatewayModule::persist_batch(const Batch& b) {
+  if (!jobs_db_) return;
+  Json ids = Json::array();
+  for (auto& id : b.job_ids) ids.push_back(id);
+  std::lock_guard<std::mutex> lk(jobs_db_->mu());
+  jobs_db_->query(
+      "INSERT INTO batches (tenant_id, id, job_ids, created_at) VALUES "
+      "(?,?,?,?) ON CONFLICT(id) DO UPDATE SET job_ids=excluded.job_ids",
+      {DbValue::S(b.tenant), DbValue::S(b.id), DbValue::S(ids.dump()),
+       DbValue::R(b.created_at)});
+}
+
+void LlmGatewayModule::load_jobs() {
+  // restart recovery: finished jobs restore for result fetches (until
+  // TTL pruning), queued/running re-enter the queue (at-least-once)
+  std::vector<DbRow> rows, brows;
+  {
+    std::lock_guard<std::mutex> lk(jobs_db_->mu());
+    rows = jobs_db_->query("SELECT * FROM jobs", {});
+    brows = jobs_db_->query("SELECT * FROM batches", {});
+  }
+  uint64_t max_ctr = 0;
+  std::lock_guard<std::mutex> lk(jobs_mu_);
+  for (auto& r : rows) {
+    auto job = std::make_shared<Job>();
+    job->id = r.at("id").as_string();
+    job->tenant = r.at("tenant_id").as_string();
+    job->batch_id = r.at("batch_id").as_string();
+    job->status = r.at("status").as_string();
+    try { job->request = Json::parse(r.at("request").as_string()); }
+    catch (...) {}
+    const std::string res = r.at("result").as_string();
+    if (!res.empty()) {
+      try { job->result = Json::parse(res); } catch (...) {}
+    }
+    job->error = r.at("error").as_string();
+    job->created_at = r.at("created_at").as_number(0);
+    job->finished_at = r.at("finished_at").as_number(0);
+    if (job->status == "running") job->status = "queued";
+    if (job->status == "queued") job_queue_.push_back(job->id);
+    jobs_[job->id] = job;
+    if (job->id.rfind("job-", 0) == 0)
+      max_ctr = std::max(max_ctr,
+                         (uint64_t)atoll(job->id.c_str() + 4) + 1);
+  }
+  for (auto& r : brows) {
+    Batch b;
+    b.id = r.at("id").as_string();
+    b.tenant = r.at("tenant_id").as_string();
+    b.created_at = r.at("created_at").as_number(0);
+    try {
+      for (auto& x : Json::parse(r.at("job_ids").as_string()).arr())
+        b.job_ids.push_back(x.as_string());
+    } catch (...) {}
+    if (b.id.rfind("batch-", 0) == 0)
+      max_ctr = std::max(max_ctr,
+                         (uint64_t)atoll(b.id.c_str() + 6) + 1);
+    batches_[b.id] = std::move(b);
+  }
+  uint64_t cur = req_ctr_.load();
+  while (max_ctr > cur && !req_ctr_.compare_exchange_weak(cur, max_ctr))
+    ;
+  if (!rows.empty())
+    LOG_INFO("llm-gateway", "restored %zu job(s), %zu batch(es)",
+             rows.size(), brows.size());
+}
+
 std::shared_ptr<LlmGatewayModule::Job> LlmGatewayModule::submit_job(
     const SecurityContext& sec, Json body, const std::string& batch_id) {
   auto job = std::make_shared<Job>();
@@ -1152,6 +1265,7 @@ std::shared_ptr<LlmGatewayModule::Job> LlmGatewayModule::submit_job(
     jobs_[job->id] = job;
     job_queue_.push_back(job->id);
   }
+  persist_job(*job);
   jobs_cv_.notify_one();
   return job;
 }
@@ -1168,10 +1282,16 @@ void LlmGatewayModule::job_loop() {
         const double now = now_s();
         for (auto it = jobs_.begin(); it != jobs_.end();) {
           if (it->second->finished_at > 0 &&
-              now - it->second->finished_at > 2 * (double)job_ttl_s_)
+              now - it->second->finished_at > 2 * (double)job_ttl_s_) {
+            if (jobs_db_) {
+              std::lock_guard<std::mutex> dlk(jobs_db_->mu());
+              jobs_db_->query("DELETE FROM jobs WHERE id=?",
+                              {DbValue::S(it->first)});
+            }
             it = jobs_.erase(it);
-          else
+          } else {
             ++it;
+          }
         }
       }
       const std::string id = job_queue_.front();
@@ -1182,6 +1302,7 @@ void LlmGatewayModule::job_loop() {
       if (job->status != "queued") continue;   // cancelled while queued
       job->status = "running";
     }
+    persist_job(*job);
     Json result;
     std::string err;
     try {
@@ -1212,6 +1333,7 @@ void LlmGatewayModule::job_loop() {
       }
       job->finished_at = now_s();
     }
+    persist_job(*job);
     if (err.empty()) record_usage(job->tenant, job->result.at("usage"));
   }
 }
@@ -1370,6 +1492,8 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
       if (it->second->status == "queued") {
         it->second->status = "cancelled";
         it->second->finished_at = now_s();
+        persist_job(*it->second);
+        it->second->finished_at = now_s();
       } else if (it->second->status == "running") {
         // abort propagation: the engine request id IS the job id, so a
         // running job can be cut short server-side (DESIGN timeouts /
@@ -1420,6 +1544,7 @@ void LlmGatewayModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
       out["status"] = "queued";
       out["num_requests"] = (int64_t)b.job_ids.size();
       out["created_at"] = b.created_at;
+      persist_batch(b);
       {
         std::lock_guard<std::mutex> lk(jobs_mu_);
         batches_[b.id] = std::move(b);

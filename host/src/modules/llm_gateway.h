@@ -19,6 +19,7 @@
 #include <mutex>
 #include <thread>
 
+#include "../modkit/db.h"
 #include "../modkit/modkit.h"
 
 namespace hs {
@@ -213,7 +214,13 @@ class LlmGatewayModule : public Module {
   void spawn_one(Worker& wk);
   std::atomic<uint64_t> req_ctr_{0};
 
-  // jobs/batches (in-memory, tenant-scoped)
+  // jobs/batches: in-memory hot path + write-through sqlite journal so
+  // queued/running jobs survive a host restart (reference: serverless
+  // durable-execution NFR; the spec's async jobs outlive the process)
+  std::unique_ptr<Db> jobs_db_;
+  void persist_job(const Job& j);
+  void persist_batch(const Batch& b);
+  void load_jobs();
   std::mutex jobs_mu_;
   std::condition_variable jobs_cv_;
   std::deque<std::string> job_queue_;
