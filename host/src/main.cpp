@@ -83,15 +83,30 @@ int run_server(const Json& cfg, bool check_only) {
   // phases: init -> post_init -> rest -> start (reference
   // host_runtime.rs:717 run_phases_internal; db/grpc/oop phases live
   // inside the owning modules here)
-  {
-    std::vector<std::string> names;
-    for (auto& m : sorted) names.push_back(m->name());
-    orch->set_modules(names);
-  }
   try {
     for (auto& [m, ctx] : mods) m->init(ctx);
     for (auto& [m, ctx] : mods) m->post_init(ctx);
     for (auto& [m, ctx] : mods) m->register_rest(ctx, gateway->rest());
+    {
+      // ModuleManager view: instance + capabilities + mounted endpoints
+      Json infos = Json::array();
+      for (auto& m : sorted) {
+        Json mi = Json::object();
+        mi["name"] = m->name();
+        Json deps = Json::array();
+        for (auto& d : m->deps()) deps.push_back(d);
+        mi["deps"] = deps;
+        mi["stateful"] = m->is_stateful();
+        Json eps = Json::array();
+        const std::string pre = "/" + m->name() + "/";
+        for (auto& r : gateway->rest().routes())
+          if (r.spec.path.rfind(pre, 0) == 0)
+            eps.push_back(r.spec.method + " " + r.spec.path);
+        mi["endpoints"] = eps;
+        infos.push_back(mi);
+      }
+      orch->set_modules(std::move(infos));
+    }
     if (check_only) {
       std::cout << "config OK; " << mods.size() << " modules, "
                 << gateway->rest().routes().size() << " routes\n";

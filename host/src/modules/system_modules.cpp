@@ -1257,12 +1257,12 @@ void ModuleOrchestratorModule::register_rest(ModuleCtx& ctx,
   list.tags = {"module-orchestrator"};
   rest.register_op(list, [this](HttpRequest& rq, ResponseWriter& w) {
     Json items = Json::array();
-    for (auto& n : module_names_) {
-      Json m = Json::object();
-      m["name"] = n;
-      m["status"] = "running";
-      items.push_back(m);
-    }
+    if (module_infos_.is_array())
+      for (auto& info : module_infos_.arr()) {
+        Json m = info;
+        m["status"] = "running";
+        items.push_back(m);
+      }
     for (auto& c : children_) {
       Json m = Json::object();
       m["name"] = c.name;

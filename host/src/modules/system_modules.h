@@ -166,9 +166,10 @@ class ModuleOrchestratorModule : public Module {
   void start(ModuleCtx& ctx) override;
   void stop(ModuleCtx& ctx) override;
   void register_rest(ModuleCtx& ctx, RestRegistry& rest) override;
-  void set_modules(std::vector<std::string> names) {
-    module_names_ = std::move(names);
-  }
+  // ModuleManager parity (reference runtime/module_manager.rs): the
+  // orchestrator tracks module INSTANCES with capabilities + mounted
+  // endpoints, not just names
+  void set_modules(Json infos) { module_infos_ = std::move(infos); }
 
  private:
   struct OopSpec {
@@ -189,7 +190,7 @@ class ModuleOrchestratorModule : public Module {
     Json meta;
   };
 
-  std::vector<std::string> module_names_;
+  Json module_infos_;   // [{name, deps, stateful, endpoints[]}]
   std::vector<OopSpec> oop_specs_;
   std::vector<Child> children_;
   std::string directory_endpoint_;

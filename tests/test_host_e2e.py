@@ -1476,3 +1476,21 @@ def test_cpp_detok_matches_python_detok(server):
             stream_text += d.get("delta", {}).get("content", "")
     assert stream_text == blocking_text, (stream_text, blocking_text)
     assert len(blocking_text) > 0
+
+
+def test_orchestrator_module_manager_view(server):
+    """/module-orchestrator/v1/modules exposes instances with deps,
+    statefulness and mounted endpoints (ModuleManager parity,
+    reference runtime/module_manager.rs)."""
+    st, body = _http("GET", BASE.format(server.port) +
+                     "/module-orchestrator/v1/modules")
+    assert st == 200, body
+    items = {m["name"]: m for m in json.loads(body)["items"]}
+    assert "llm-gateway" in items
+    lg = items["llm-gateway"]
+    assert lg["stateful"] is True
+    assert "serverless-runtime" in lg["deps"]
+    assert any(e.startswith("POST /llm-gateway/v1/chat/completions")
+               for e in lg["endpoints"])
+    sus = items["simple-user-settings"]
+    assert any("settings" in e for e in sus["endpoints"])
