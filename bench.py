@@ -488,7 +488,11 @@ def engine_direct(args, rank, world, on_gpu):
     tp = args.tp if args.tp > 1 else 1
     if tp > 1:
         assert world == tp, "TP mode needs world == tp"
-        initialize_model_parallel(tp_size=tp)
+        from hyperspot.engine.config import get_model_spec
+        ep = tp if get_model_spec(args.model).is_moe else 1
+        initialize_model_parallel(tp_size=tp, ep_size=ep)
+        if ep > 1:
+            args.eager = True   # EP all-to-all not graph-captured yet
 
     max_len = args.prompt_len + args.warmup + args.steps + 64
     cfg = EngineConfig(

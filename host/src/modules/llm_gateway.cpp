@@ -545,6 +545,10 @@ void LlmGatewayModule::spawn_one(Worker& wk) {
         args.push_back("--tp");
         args.push_back(std::to_string(worker_cfg_.at("tp").as_int(1)));
       }
+      if (worker_cfg_.contains("ep")) {
+        args.push_back("--ep");
+        args.push_back(std::to_string(worker_cfg_.at("ep").as_int(0)));
+      }
       if (worker_cfg_.contains("quant")) {
         args.push_back("--quant");
         args.push_back(worker_cfg_.at("quant").as_string());

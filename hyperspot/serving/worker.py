@@ -714,6 +714,8 @@ def main():
     ap.add_argument("--eager", action="store_true")
     ap.add_argument("--device", default=None)
     ap.add_argument("--tp", type=int, default=1)
+    ap.add_argument("--ep", type=int, default=0,
+                    help="expert-parallel degree; 0 = tp for MoE models")
     ap.add_argument("--quant", default=None, choices=["fp8"])
     ap.add_argument("--kv-dtype", default="bfloat16",
                     choices=["bfloat16", "fp8"])
@@ -737,9 +739,18 @@ def main():
         enforce_eager=args.eager or not on_gpu, tp_size=args.tp,
         quant=args.quant, kv_dtype=args.kv_dtype)
     rank = 0
+    # MoE models shard experts over the same ranks as TP attention
+    # (config 4: Mixtral EP over RCCL all-to-all); EP decode steps do
+    # count exchanges host-side, so graphs are disabled at ep>1
+    ep = args.ep or (args.tp if cfg.spec().is_moe else 1)
+    if ep > 1 and not cfg.enforce_eager:
+        log.info("ep=%d: forcing eager decode (EP all-to-all is not "
+                 "graph-captured yet)", ep)
+        cfg.enforce_eager = True
+    cfg.ep_size = ep
     if args.tp > 1:
         from hyperspot.parallel.state import initialize_model_parallel
-        initialize_model_parallel(tp_size=args.tp)
+        initialize_model_parallel(tp_size=args.tp, ep_size=ep)
         import torch.distributed as dist
         rank = dist.get_rank()
         if on_gpu:

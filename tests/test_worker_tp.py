@@ -230,3 +230,44 @@ def test_tp2_swap_save_embeddings_parity():
             t1, t2 = f1.get_tensor(k), f2.get_tensor(k)
             assert t1.shape == t2.shape, (k, t1.shape, t2.shape)
             assert torch.equal(t1, t2), k
+
+
+def _spawn_moe_worker(tp, sock, port):
+    if tp == 1:
+        cmd = [sys.executable, "-m", "hyperspot.serving.worker"]
+    else:
+        cmd = [sys.executable, "-m", "torch.distributed.run",
+               "--master-addr", "127.0.0.1", "--master-port", str(port),
+               "--nnodes", "1", "--nproc-per-node", str(tp),
+               "-m", "hyperspot.serving.worker", "--tp", str(tp)]
+    cmd += ["--uds", sock, "--model", "tiny-moe", "--device", "cpu",
+            "--max-num-seqs", "4", "--num-gpu-blocks", "128", "--eager"]
+    return subprocess.Popen(cmd, stdout=subprocess.PIPE,
+                            stderr=subprocess.STDOUT)
+
+
+def test_ep2_moe_worker_matches_ep1():
+    """Serving-path EP: a tp=2 MoE worker auto-shards experts over the
+    TP ranks (EP all-to-all dispatch/combine, config 4) and must
+    reproduce the single-rank greedy output exactly."""
+    out = {}
+    for tp in (1, 2):
+        sock = tempfile.mktemp(suffix=".sock", prefix=f"hs-ep{tp}-")
+        s2 = socketlib.socket()
+        s2.bind(("127.0.0.1", 0))
+        port = s2.getsockname()[1]
+        s2.close()
+        proc = _spawn_moe_worker(tp, sock, port)
+        try:
+            _wait_sock(sock, proc)
+            toks, usage = _chat_over_uds(sock, "expert parallel?")
+            assert usage["output_tokens"] == 6
+            out[tp] = toks
+        finally:
+            proc.terminate()
+            try:
+                proc.wait(timeout=20)
+            except subprocess.TimeoutExpired:
+                proc.kill()
+                proc.wait()
+    assert out[1] == out[2], out
