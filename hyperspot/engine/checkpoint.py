@@ -91,8 +91,15 @@ def save_checkpoint(model, path: str, meta: Dict = None) -> None:
     """Write the FULL-layout checkpoint.  Collective under TP (all ranks
     must call); only rank 0 writes the file."""
     from safetensors.torch import save_file
-    from hyperspot.parallel.state import get_tp_group, get_tp_rank, \
-        get_tp_size
+    from hyperspot.parallel.state import get_ep_size, get_tp_group, \
+        get_tp_rank, get_tp_size
+    if get_ep_size() > 1:
+        # expert tensors are sharded per-rank WITHOUT slice metadata (a
+        # rank owns whole experts); gathering them needs an EP-aware
+        # collector — fail loudly rather than write one rank's experts
+        raise NotImplementedError(
+            "checkpoint save at ep>1 (EP expert gather) is not "
+            "implemented; save from an ep=1 deployment")
     tp = get_tp_size()
     state = {}
     for k, v in _params_and_buffers(model).items():
@@ -113,6 +120,10 @@ def load_checkpoint_into(model, path: str) -> float:
     reused (copy_), so captured hipGraphs remain valid.  Full-layout
     checkpoints load into TP shards via the `_tp_slices` metadata."""
     from safetensors import safe_open
+    from hyperspot.parallel.state import get_ep_size
+    if get_ep_size() > 1:
+        raise NotImplementedError(
+            "hot-swap at ep>1 (EP expert scatter) is not implemented")
     t0 = time.monotonic()
     device = next(model.parameters()).device
     params = _params_and_buffers(model)
