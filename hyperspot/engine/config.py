@@ -117,6 +117,9 @@ class EngineConfig:
     # decode hipGraph capture batch buckets (padded up to nearest)
     graph_batch_sizes: tuple = (1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128,
                                 192, 256, 384, 512, 768, 1024, 1536, 2048)
+    # single-request whole-prompt prefill graphs (TTFT fast path):
+    # tokens pad up to the bucket, padded slots write no KV
+    prefill_graph_sizes: tuple = (128, 256, 512, 1024, 2048)
 
     def spec(self) -> ModelSpec:
         return get_model_spec(self.model)

@@ -102,6 +102,8 @@ class ModelRunner:
                                        dtype=torch.int32, pin_memory=pin)
         self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
         self._graph_io: Dict[int, dict] = {}
+        self._pgraphs: Dict[int, torch.cuda.CUDAGraph] = {}
+        self._pgraph_io: Dict[int, dict] = {}
         self._graph_pool = None
 
     # ---------------- input preparation ----------------
@@ -194,12 +196,86 @@ class ModelRunner:
     def execute(self, batch: ScheduledBatch) -> torch.Tensor:
         """Run one step; returns sampled token ids [num_seqs] (cpu)."""
         if batch.mode == "prefill":
-            input_ids, meta = self._prefill_inputs(batch)
-            logits = self.model(input_ids, meta, self.kv_caches)
+            logits = self._prefill_graph_forward(batch)
+            if logits is None:
+                input_ids, meta = self._prefill_inputs(batch)
+                logits = self.model(input_ids, meta, self.kv_caches)
         else:
             input_ids, meta = self._decode_inputs(batch)
             logits = self._decode_forward(input_ids, meta)
         return self._sample(logits, batch.requests)
+
+    # ---- TTFT fast path: single-request whole-prompt prefill graphs ----
+
+    def _prefill_bucket(self, n: int) -> Optional[int]:
+        for s in self.config.prefill_graph_sizes:
+            if s >= n and s <= self.config.max_num_batched_tokens:
+                return s
+        return None
+
+    def _prefill_graph_forward(self, batch: ScheduledBatch):
+        """Replay a padded single-sequence prefill graph (eager prefill
+        costs ~350 kernel launches + the Python layer loop — several ms
+        of the TTFT).  Pad tokens sit AFTER the real prompt, so causal
+        attention keeps real rows exact; padded slots are -1 (no KV
+        write) and the single logit row indexes the real last token."""
+        if (self.device.type != "cuda" or self.config.enforce_eager
+                or len(batch.requests) != 1):
+            return None
+        req = batch.requests[0]
+        n = req.chunk_len or req.num_prompt_tokens
+        if req.chunk_start != 0 or n != req.num_prompt_tokens:
+            return None              # chunked / continued prompt
+        T = self._prefill_bucket(n)
+        if T is None:
+            return None
+        if T not in self._pgraphs:
+            self._capture_prefill(T)
+        import numpy as np
+        bm = self.block_manager
+        row = bm.row_of[req.request_id]
+        p = np.arange(n)
+        slots = bm.tables_np[row, p // bm.block_size].astype(np.int64)             * bm.block_size + p % bm.block_size
+        io = self._pgraph_io[T]
+        d = self.device
+        io["input_ids"][:n] = torch.tensor(req.prompt_token_ids[:n],
+                                           dtype=torch.long, device=d)
+        io["input_ids"][n:] = 0
+        io["positions"][:n] = torch.arange(n, dtype=torch.long, device=d)
+        io["positions"][n:] = 0
+        io["slot_mapping"][:n] = torch.from_numpy(slots).to(d)
+        io["slot_mapping"][n:] = -1
+        io["lidx"][0] = n - 1
+        self._pgraphs[T].replay()
+        return io["logits"]
+
+    def _capture_prefill(self, T: int) -> None:
+        d = self.device
+        log.info("capturing prefill hipGraph for tokens=%d", T)
+        io = {
+            "input_ids": torch.zeros(T, dtype=torch.long, device=d),
+            "positions": torch.zeros(T, dtype=torch.long, device=d),
+            "slot_mapping": torch.full((T,), -1, dtype=torch.long,
+                                       device=d),
+            "lidx": torch.zeros(1, dtype=torch.long, device=d),
+        }
+        meta = ForwardMeta(
+            mode="prefill", positions=io["positions"],
+            slot_mapping=io["slot_mapping"],
+            seq_start=torch.tensor([0, T], dtype=torch.int32, device=d),
+            max_seqlen=T, fresh_prefill=True,
+            logits_indices=io["lidx"])
+        for _ in range(2):
+            self.model(io["input_ids"], meta, self.kv_caches)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g, pool=self._graph_pool):
+            io["logits"] = self.model(io["input_ids"], meta,
+                                      self.kv_caches)
+        if self._graph_pool is None:
+            self._graph_pool = g.pool()
+        self._pgraphs[T] = g
+        self._pgraph_io[T] = io
 
     def _decode_forward(self, input_ids, meta):
         B = input_ids.shape[0]
@@ -262,6 +338,9 @@ class ModelRunner:
         for b in self.config.graph_batch_sizes:
             if b <= self.config.max_num_seqs:
                 self._capture(b)
+        for t in self.config.prefill_graph_sizes:
+            if t <= self.config.max_num_batched_tokens                     and t <= self.config.max_model_len:
+                self._capture_prefill(t)
 
     # ---------------- embeddings ----------------
 

@@ -115,3 +115,26 @@ def test_prefix_caching_matches_uncached_gpu():
         import torch
         torch.cuda.empty_cache()
     assert outs[False] == outs[True]
+
+
+@pytest.mark.gpu
+def test_prefill_graph_matches_eager():
+    """The padded single-request prefill graph (TTFT fast path) must be
+    token-exact vs eager prefill — including a prompt length that is
+    NOT a bucket size (padding correctness) and the KV it leaves behind
+    (greedy continuation must also match)."""
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    prompts = [[7 * i % 311 + 5 for i in range(n)] for n in (100, 512)]
+    outs = {}
+    for eager in (True, False):
+        cfg = EngineConfig(model="llama3-8b", max_num_seqs=4,
+                           max_model_len=1024, num_gpu_blocks=512,
+                           enforce_eager=eager, seed=3)
+        eng = LLMEngine(cfg, device="cuda:0")
+        if not eager:
+            eng.capture_graphs()
+        outs[eager] = eng.generate(
+            prompts, SamplingParams(temperature=0.0, max_tokens=12))
+        del eng
+        torch.cuda.empty_cache()
+    assert outs[True] == outs[False], outs
