@@ -332,14 +332,20 @@ class ModelRunner:
         self._graphs[bucket] = g
         self._graph_io[bucket] = io
 
+    @torch.inference_mode()
     def capture_all_graphs(self) -> None:
+        # inference_mode matches the lazy in-step captures: every graph
+        # io tensor is an inference tensor regardless of capture origin
+        # (mixing modes trips "inplace update to inference tensor")
         if self.device.type != "cuda" or self.config.enforce_eager:
             return
         for b in self.config.graph_batch_sizes:
-            if b <= self.config.max_num_seqs:
+            if b <= self.config.max_num_seqs and b not in self._graphs:
                 self._capture(b)
         for t in self.config.prefill_graph_sizes:
-            if t <= self.config.max_num_batched_tokens                     and t <= self.config.max_model_len:
+            if (t <= self.config.max_num_batched_tokens
+                    and t <= self.config.max_model_len
+                    and t not in self._pgraphs):
                 self._capture_prefill(t)
 
     # ---------------- embeddings ----------------
