@@ -1494,3 +1494,60 @@ def test_orchestrator_module_manager_view(server):
                for e in lg["endpoints"])
     sus = items["simple-user-settings"]
     assert any("settings" in e for e in sus["endpoints"])
+
+
+def test_rest_serving_with_tp2_worker():
+    """Config-3 serving shape end-to-end on CPU: the host spawns a
+    torchrun SPMD worker group (worker.tp=2, gloo here / RCCL on GPUs)
+    and serves REST chat through rank 0's mux socket."""
+    import tempfile
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-tp2e2e-")
+    port = _free_port()
+    cfg = f"""
+server:
+  home_dir: "/tmp/hs-e2e-tp2"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+  llm-gateway:
+    config:
+      model: "tiny-llama"
+      worker_socket: "{sock}"
+      auto_start_worker: true
+      worker:
+        device: "cpu"
+        eager: true
+        tp: 2
+        max_num_seqs: 4
+        num_gpu_blocks: 128
+"""
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(cfg)
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        srv.wait_worker(timeout=180)
+        st, body = _http("POST",
+                         BASE.format(port) + "/v1/chat/completions",
+                         body={"model": "tiny-llama",
+                               "messages": [{"role": "user", "content":
+                                             [{"type": "text",
+                                               "text": "tp2"}]}],
+                               "max_tokens": 6, "temperature": 0.0})
+        assert st == 200, body
+        j = json.loads(body)
+        assert j["usage"]["output_tokens"] == 6
+        # embeddings + hot-swap collective ops work through the SPMD
+        # group too (exec broadcast channel)
+        st, body = _http("POST", BASE.format(port) + "/v1/embeddings",
+                         body={"model": "tiny-llama", "input": "vec me"})
+        assert st == 200, body
+        assert len(json.loads(body)["data"][0]["embedding"]) > 0
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
