@@ -9,8 +9,12 @@
 //   - SecurityContext / AccessScope    (libs/modkit-security/src/context.rs:23)
 #pragma once
 
+#include <atomic>
+#include <chrono>
+#include <condition_variable>
 #include <functional>
 #include <future>
+#include <thread>
 #include <map>
 #include <memory>
 #include <mutex>
@@ -182,6 +186,98 @@ class ClientHub {
  private:
   mutable std::mutex mu_;
   std::map<std::string, std::shared_ptr<void>> clients_;
+};
+
+// Lifecycle state machine for stateful runnables (reference
+// libs/modkit/src/lifecycle.rs:33,:74 WithLifecycle<T: Runnable>):
+// Stopped -> Starting -> Running -> Stopping -> Stopped, with a
+// ReadySignal the entry function fires once it is serving, and a
+// bounded stop (cancel flag -> join with timeout; an overrunning
+// runnable is detached and reported, the thread analog of the
+// reference's stop_timeout kill ladder).
+class WithLifecycle {
+ public:
+  enum class State { Stopped, Starting, Running, Stopping };
+  // entry(cancel, ready): run until `cancel` is set; call ready() once
+  // serving (await_ready semantics)
+  using Runnable = std::function<void(std::atomic<bool>& cancel,
+                                      std::function<void()> ready)>;
+
+  explicit WithLifecycle(Runnable r, int ready_timeout_ms = 30000,
+                         int stop_timeout_ms = 30000)
+      : runnable_(std::move(r)), ready_timeout_ms_(ready_timeout_ms),
+        stop_timeout_ms_(stop_timeout_ms) {}
+  ~WithLifecycle() { stop(); }
+
+  State state() const { return state_.load(); }
+
+  // false if not Stopped, or the runnable missed the ready deadline
+  bool start() {
+    State want = State::Stopped;
+    if (!state_.compare_exchange_strong(want, State::Starting))
+      return false;
+    cancel_ = false;
+    ready_ = false;
+    thread_ = std::thread([this] {
+      runnable_(cancel_, [this] {
+        {
+          std::lock_guard<std::mutex> lk(mu_);
+          ready_ = true;
+        }
+        cv_.notify_all();
+      });
+    });
+    std::unique_lock<std::mutex> lk(mu_);
+    if (!cv_.wait_for(lk, std::chrono::milliseconds(ready_timeout_ms_),
+                      [&] { return ready_; })) {
+      lk.unlock();
+      stop();
+      return false;
+    }
+    state_ = State::Running;
+    return true;
+  }
+
+  // true if the runnable unwound within stop_timeout
+  bool stop() {
+    State s = state_.load();
+    if (s == State::Stopped || s == State::Stopping) return true;
+    state_ = State::Stopping;
+    cancel_ = true;
+    cv_.notify_all();
+    bool clean = true;
+    if (thread_.joinable()) {
+      // bounded join: poll the thread with a deadline
+      auto deadline = std::chrono::steady_clock::now() +
+                      std::chrono::milliseconds(stop_timeout_ms_);
+      std::atomic<bool> joined{false};
+      std::thread waiter([&] {
+        thread_.join();
+        joined = true;
+      });
+      while (!joined &&
+             std::chrono::steady_clock::now() < deadline)
+        std::this_thread::sleep_for(std::chrono::milliseconds(5));
+      if (joined) {
+        waiter.join();
+      } else {
+        clean = false;           // runnable overran its stop budget
+        waiter.detach();
+      }
+    }
+    state_ = State::Stopped;
+    return clean;
+  }
+
+ private:
+  Runnable runnable_;
+  int ready_timeout_ms_, stop_timeout_ms_;
+  std::atomic<State> state_{State::Stopped};
+  std::atomic<bool> cancel_{false};
+  bool ready_ = false;
+  std::mutex mu_;
+  std::condition_variable cv_;
+  std::thread thread_;
 };
 
 // Single-flight cached plugin-instance resolution (reference

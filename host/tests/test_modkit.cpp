@@ -76,7 +76,52 @@ static void test_selector_failure_not_cached() {
   CHECK(sel.select("t") == "ok");   // failure was NOT cached
 }
 
+static void test_with_lifecycle() {
+  using WL = WithLifecycle;
+  std::atomic<int> served{0};
+  WL wl([&](std::atomic<bool>& cancel, std::function<void()> ready) {
+    ready();
+    while (!cancel) {
+      served++;
+      std::this_thread::sleep_for(std::chrono::milliseconds(2));
+    }
+  });
+  CHECK(wl.state() == WL::State::Stopped);
+  CHECK(wl.start());
+  CHECK(wl.state() == WL::State::Running);
+  CHECK(!wl.start());                      // double-start refused
+  std::this_thread::sleep_for(std::chrono::milliseconds(20));
+  CHECK(served > 0);
+  CHECK(wl.stop());                        // clean unwind
+  CHECK(wl.state() == WL::State::Stopped);
+  CHECK(wl.start());                       // restartable
+  CHECK(wl.stop());
+
+  // a runnable that never signals ready fails start()
+  WL late([&](std::atomic<bool>& cancel, std::function<void()>) {
+    while (!cancel)
+      std::this_thread::sleep_for(std::chrono::milliseconds(2));
+  }, /*ready_timeout_ms=*/50);
+  CHECK(!late.start());
+  CHECK(late.state() == WL::State::Stopped);
+
+  // a runnable that ignores cancel overruns its stop budget
+  std::atomic<bool> hang{true};
+  {
+    WL stuck([&](std::atomic<bool>&, std::function<void()> ready) {
+      ready();
+      while (hang)
+        std::this_thread::sleep_for(std::chrono::milliseconds(2));
+    }, 1000, /*stop_timeout_ms=*/50);
+    CHECK(stuck.start());
+    CHECK(!stuck.stop());                  // reported as unclean
+    hang = false;                          // let the detached thread die
+    std::this_thread::sleep_for(std::chrono::milliseconds(20));
+  }
+}
+
 int main() {
+  test_with_lifecycle();
   test_scoped_hub();
   test_plugin_selector_single_flight();
   test_selector_failure_not_cached();
