@@ -216,21 +216,21 @@ class WithLifecycle {
     State want = State::Stopped;
     if (!state_.compare_exchange_strong(want, State::Starting))
       return false;
-    cancel_ = false;
-    ready_ = false;
-    thread_ = std::thread([this] {
-      runnable_(cancel_, [this] {
-        {
-          std::lock_guard<std::mutex> lk(mu_);
-          ready_ = true;
-        }
-        cv_.notify_all();
-      });
+    // shared with the runnable thread: an overrunning runnable is
+    // detached at stop(), so everything it touches must outlive *this
+    auto inner = std::make_shared<Inner>();
+    inner_ = inner;
+    auto fn = runnable_;
+    thread_ = std::thread([inner, fn] {
+      fn(inner->cancel, [inner] { inner->ready = true; });
     });
-    std::unique_lock<std::mutex> lk(mu_);
-    if (!cv_.wait_for(lk, std::chrono::milliseconds(ready_timeout_ms_),
-                      [&] { return ready_; })) {
-      lk.unlock();
+    const auto deadline = std::chrono::steady_clock::now() +
+                          std::chrono::milliseconds(ready_timeout_ms_);
+    while (!inner->ready &&
+           std::chrono::steady_clock::now() < deadline)
+      std::this_thread::sleep_for(std::chrono::milliseconds(1));
+    if (!inner->ready) {
+      state_ = State::Running;   // so stop() proceeds
       stop();
       return false;
     }
@@ -243,22 +243,23 @@ class WithLifecycle {
     State s = state_.load();
     if (s == State::Stopped || s == State::Stopping) return true;
     state_ = State::Stopping;
-    cancel_ = true;
-    cv_.notify_all();
+    auto inner = inner_;
+    if (inner) inner->cancel = true;
     bool clean = true;
     if (thread_.joinable()) {
-      // bounded join: poll the thread with a deadline
+      // bounded join: hand the thread to a waiter so an overrunning
+      // runnable can be detached without dangling into *this
+      auto done = std::make_shared<std::atomic<bool>>(false);
+      std::thread waiter(
+          [t = std::move(thread_), done]() mutable {
+            t.join();
+            *done = true;
+          });
       auto deadline = std::chrono::steady_clock::now() +
                       std::chrono::milliseconds(stop_timeout_ms_);
-      std::atomic<bool> joined{false};
-      std::thread waiter([&] {
-        thread_.join();
-        joined = true;
-      });
-      while (!joined &&
-             std::chrono::steady_clock::now() < deadline)
+      while (!*done && std::chrono::steady_clock::now() < deadline)
         std::this_thread::sleep_for(std::chrono::milliseconds(5));
-      if (joined) {
+      if (*done) {
         waiter.join();
       } else {
         clean = false;           // runnable overran its stop budget
@@ -270,13 +271,14 @@ class WithLifecycle {
   }
 
  private:
+  struct Inner {
+    std::atomic<bool> cancel{false};
+    std::atomic<bool> ready{false};
+  };
   Runnable runnable_;
   int ready_timeout_ms_, stop_timeout_ms_;
   std::atomic<State> state_{State::Stopped};
-  std::atomic<bool> cancel_{false};
-  bool ready_ = false;
-  std::mutex mu_;
-  std::condition_variable cv_;
+  std::shared_ptr<Inner> inner_;
   std::thread thread_;
 };
 
