@@ -794,6 +794,12 @@ def main():
         pass
     finally:
         state.stop = True
+        with state.new_work:
+            state.new_work.notify_all()
+        # let the stepping thread unwind out of torch before the
+        # interpreter finalizes (otherwise libtorch aborts with
+        # "terminate called without an active exception")
+        stepper.join(timeout=10)
         srv.close()
 
 
