@@ -505,6 +505,13 @@ void LlmGatewayModule::spawn_one(Worker& wk) {
       setenv("HIP_VISIBLE_DEVICES", dev.c_str(), 1);
       setenv("CUDA_VISIBLE_DEVICES", dev.c_str(), 1);
     }
+    // worker.environment: extra env for the engine process (A.6
+    // execution.environment envelope) — the RCCL tuning surface for
+    // 8-GPU runs (NCCL_ALGO/NCCL_PROTO/NCCL_MIN_NCHANNELS etc.)
+    if (worker_cfg_.is_object() &&
+        worker_cfg_.at("environment").is_object())
+      for (auto& [k, v] : worker_cfg_.at("environment").obj())
+        setenv(k.c_str(), v.as_string().c_str(), 1);
     // tp>1 workers are an SPMD torchrun group (one rank per GPU, RCCL
     // over xGMI); rank 0 owns the socket and broadcasts the op log
     const long tp = worker_cfg_.is_object()

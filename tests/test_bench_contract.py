@@ -35,3 +35,22 @@ def test_bench_rest_json_line():
                                 "parallelism"}
     # REST-vs-engine overhead is quantified (VERDICT round-1 item 2)
     assert "engine_tokens_per_s" in j and "gateway_overhead_pct" in j
+
+
+def test_bench_rest_two_rank_launch():
+    """The driver's N>1 shape: torchrun two ranks, rank 0 orchestrates
+    the host with a 2-worker DP fleet, barriers synchronize ranks."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29713",
+         os.path.join(ROOT, "bench.py"), "--gpus", "2",
+         "--steps", "4", "--warmup", "1", "--ttft-iters", "2"],
+        cwd=ROOT, capture_output=True, text=True, timeout=420)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    j = json.loads(lines[0])
+    assert j["n_gpus"] == 2
+    assert j["config"]["parallelism"].startswith("dp2")
+    assert j["value"] > 0
