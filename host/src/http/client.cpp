@@ -1,5 +1,8 @@
 #include "client.h"
 
+#include <openssl/err.h>
+#include <openssl/ssl.h>
+
 #include <arpa/inet.h>
 #include <fcntl.h>
 #include <netdb.h>
@@ -10,6 +13,7 @@
 #include <unistd.h>
 
 #include <cstring>
+#include <mutex>
 #include <sstream>
 
 namespace hs {
@@ -47,13 +51,67 @@ int connect_to(const std::string& host, int port, int timeout_ms) {
   return fd;
 }
 
-bool send_all(int fd, const char* p, size_t n) {
-  while (n) {
-    ssize_t w = ::send(fd, p, n, MSG_NOSIGNAL);
-    if (w <= 0) return false;
-    p += w;
-    n -= size_t(w);
+// plain-or-TLS connection IO
+struct Io {
+  int fd = -1;
+  SSL* ssl = nullptr;
+  SSL_CTX* ctx = nullptr;
+
+  ssize_t read(char* p, size_t n) {
+    if (ssl) return (ssize_t)SSL_read(ssl, p, (int)n);
+    return recv(fd, p, n, 0);
   }
+  bool write_all(const char* p, size_t n) {
+    while (n) {
+      ssize_t w = ssl ? (ssize_t)SSL_write(ssl, p, (int)n)
+                      : ::send(fd, p, n, MSG_NOSIGNAL);
+      if (w <= 0) return false;
+      p += w;
+      n -= size_t(w);
+    }
+    return true;
+  }
+  void close_all() {
+    if (ssl) {
+      SSL_shutdown(ssl);
+      SSL_free(ssl);
+      ssl = nullptr;
+    }
+    if (ctx) {
+      SSL_CTX_free(ctx);
+      ctx = nullptr;
+    }
+    if (fd >= 0) {
+      close(fd);
+      fd = -1;
+    }
+  }
+};
+
+bool tls_connect(Io& io, const std::string& host, const TlsOpts& opts) {
+  static std::once_flag once;
+  std::call_once(once, [] {
+    SSL_library_init();
+    SSL_load_error_strings();
+  });
+  io.ctx = SSL_CTX_new(TLS_client_method());
+  if (!io.ctx) return false;
+  if (opts.verify) {
+    SSL_CTX_set_verify(io.ctx, SSL_VERIFY_PEER, nullptr);
+    if (!opts.ca_file.empty()) {
+      if (SSL_CTX_load_verify_locations(io.ctx, opts.ca_file.c_str(),
+                                        nullptr) != 1)
+        return false;
+    } else {
+      SSL_CTX_set_default_verify_paths(io.ctx);
+    }
+  }
+  io.ssl = SSL_new(io.ctx);
+  if (!io.ssl) return false;
+  SSL_set_fd(io.ssl, io.fd);
+  SSL_set_tlsext_host_name(io.ssl, host.c_str());   // SNI
+  if (opts.verify) SSL_set1_host(io.ssl, host.c_str());
+  if (SSL_connect(io.ssl) != 1) return false;
   return true;
 }
 
@@ -65,9 +123,15 @@ std::optional<ClientResponse> http_request(
     const std::map<std::string, std::string>& headers,
     const std::string& body, int connect_timeout_ms,
     const std::function<bool(const char*, size_t)>& on_chunk,
-    const std::function<void(const ClientResponse&)>& on_headers) {
-  int fd = connect_to(host, port, connect_timeout_ms);
-  if (fd < 0) return std::nullopt;
+    const std::function<void(const ClientResponse&)>& on_headers,
+    const TlsOpts* tls) {
+  Io io;
+  io.fd = connect_to(host, port, connect_timeout_ms);
+  if (io.fd < 0) return std::nullopt;
+  if (tls && tls->enable && !tls_connect(io, host, *tls)) {
+    io.close_all();
+    return std::nullopt;
+  }
   std::ostringstream req;
   req << method << " " << target << " HTTP/1.1\r\n"
       << "host: " << host << ":" << port << "\r\n"
@@ -83,9 +147,9 @@ std::optional<ClientResponse> http_request(
     req << "content-length: " << body.size() << "\r\n";
   req << "\r\n";
   std::string head = req.str();
-  if (!send_all(fd, head.data(), head.size()) ||
-      !send_all(fd, body.data(), body.size())) {
-    close(fd);
+  if (!io.write_all(head.data(), head.size()) ||
+      !io.write_all(body.data(), body.size())) {
+    io.close_all();
     return std::nullopt;
   }
   // read response head
@@ -93,10 +157,10 @@ std::optional<ClientResponse> http_request(
   char tmp[16384];
   size_t hdr_end;
   while ((hdr_end = buf.find("\r\n\r\n")) == std::string::npos) {
-    ssize_t r = recv(fd, tmp, sizeof tmp, 0);
-    if (r <= 0) { close(fd); return std::nullopt; }
+    ssize_t r = io.read(tmp, sizeof tmp);
+    if (r <= 0) { io.close_all(); return std::nullopt; }
     buf.append(tmp, size_t(r));
-    if (buf.size() > 256 * 1024) { close(fd); return std::nullopt; }
+    if (buf.size() > 256 * 1024) { io.close_all(); return std::nullopt; }
   }
   ClientResponse resp;
   {
@@ -141,8 +205,8 @@ std::optional<ClientResponse> http_request(
     while (true) {
       size_t nl = acc.find("\r\n");
       while (nl == std::string::npos) {
-        ssize_t r = recv(fd, tmp, sizeof tmp, 0);
-        if (r <= 0) { close(fd); return resp; }
+        ssize_t r = io.read(tmp, sizeof tmp);
+        if (r <= 0) { io.close_all(); return resp; }
         acc.append(tmp, size_t(r));
         nl = acc.find("\r\n");
       }
@@ -150,8 +214,8 @@ std::optional<ClientResponse> http_request(
       if (sz == 0) break;
       size_t need = nl + 2 + size_t(sz) + 2;
       while (acc.size() < need) {
-        ssize_t r = recv(fd, tmp, sizeof tmp, 0);
-        if (r <= 0) { close(fd); return resp; }
+        ssize_t r = io.read(tmp, sizeof tmp);
+        if (r <= 0) { io.close_all(); return resp; }
         acc.append(tmp, size_t(r));
       }
       if (!emit(acc.data() + nl + 2, size_t(sz))) break;
@@ -161,13 +225,13 @@ std::optional<ClientResponse> http_request(
     if (!rest.empty()) emit(rest.data(), rest.size());
     long got = (long)rest.size();
     while (content_len < 0 || got < content_len) {
-      ssize_t r = recv(fd, tmp, sizeof tmp, 0);
+      ssize_t r = io.read(tmp, sizeof tmp);
       if (r <= 0) break;
       got += r;
       if (!emit(tmp, size_t(r))) break;
     }
   }
-  close(fd);
+  io.close_all();
   return resp;
 }
 

@@ -101,18 +101,19 @@ bool const_eq(const std::string& a, const std::string& b) {
 
 namespace {
 
-// parse http://host[:port]/path (plain HTTP only — see header note)
+// parse http(s)://host[:port]/path
 bool parse_http_url(const std::string& url, std::string* host, int* port,
-                    std::string* path) {
-  if (url.rfind("http://", 0) != 0) return false;
-  std::string rest = url.substr(7);
+                    std::string* path, bool* https) {
+  *https = url.rfind("https://", 0) == 0;
+  if (!*https && url.rfind("http://", 0) != 0) return false;
+  std::string rest = url.substr(*https ? 8 : 7);
   size_t slash = rest.find('/');
   std::string hp = slash == std::string::npos ? rest : rest.substr(0, slash);
   *path = slash == std::string::npos ? "/" : rest.substr(slash);
   size_t colon = hp.find(':');
   if (colon == std::string::npos) {
     *host = hp;
-    *port = 80;
+    *port = *https ? 443 : 80;
   } else {
     *host = hp.substr(0, colon);
     *port = atoi(hp.c_str() + colon + 1);
@@ -153,15 +154,21 @@ std::string jwk_rsa_to_pem(const std::string& n_b64,
 #pragma GCC diagnostic pop
 
 std::optional<Json> http_get_json(const std::string& url,
-                                  std::string* err) {
+                                  std::string* err,
+                                  const std::string& ca_file = "") {
   std::string host, path;
   int port = 0;
-  if (!parse_http_url(url, &host, &port, &path)) {
+  bool https = false;
+  if (!parse_http_url(url, &host, &port, &path, &https)) {
     if (err) *err = "bad url: " + url;
     return std::nullopt;
   }
+  TlsOpts tls;
+  tls.enable = https;
+  tls.ca_file = ca_file;
   auto resp = http_request(host, port, "GET", path,
-                           {{"accept", "application/json"}}, "", 5000);
+                           {{"accept", "application/json"}}, "", 5000,
+                           nullptr, nullptr, &tls);
   if (!resp || resp->status != 200) {
     if (err)
       *err = "fetch failed: " + url + " status " +
@@ -188,7 +195,7 @@ bool JwksCache::refresh(std::string* err) {
   // caller holds mu_
   last_attempt_ = mono_s();
   if (jwks_uri.empty()) {
-    auto disc = http_get_json(discovery_url, err);
+    auto disc = http_get_json(discovery_url, err, tls_ca_file);
     if (!disc) return false;
     jwks_uri = disc->at("jwks_uri").as_string();
     issuer_ = disc->at("issuer").as_string();
@@ -197,7 +204,7 @@ bool JwksCache::refresh(std::string* err) {
       return false;
     }
   }
-  auto jwks = http_get_json(jwks_uri, err);
+  auto jwks = http_get_json(jwks_uri, err, tls_ca_file);
   if (!jwks) return false;
   const Json& keys = jwks->at("keys");
   if (!keys.is_array()) {

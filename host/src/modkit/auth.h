@@ -32,6 +32,7 @@ class JwksCache {
  public:
   std::string jwks_uri;            // direct JWKS endpoint, or
   std::string discovery_url;       // resolve jwks_uri + issuer via OIDC
+  std::string tls_ca_file;         // CA bundle for https IdPs
   int ttl_s = 300;                 // full-refresh interval
   int rotate_cooldown_s = 2;       // min gap between miss-driven fetches
 

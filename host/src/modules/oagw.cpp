@@ -140,7 +140,12 @@ void OagwModule::proxy(HttpRequest& req, ResponseWriter& w) {
 
   const Json& ep = up.path("server.endpoints").arr()[0];
   const std::string host = ep.at("host").as_string();
-  const int port = (int)ep.at("port").as_int(80);
+  const std::string scheme = ep.at("scheme").as_string("http");
+  const int port = (int)ep.at("port").as_int(scheme == "https" ? 443 : 80);
+  TlsOpts tls;
+  tls.enable = scheme == "https";
+  tls.ca_file = up.path("tls.ca_file").as_string("");
+  tls.verify = up.path("tls.verify").as_bool(true);
 
   // header policy: drop hop-by-hop + authorization (never forwarded),
   // pass the rest (reference src/infra/proxy/headers.rs allow-list idea)
@@ -193,7 +198,8 @@ void OagwModule::proxy(HttpRequest& req, ResponseWriter& w) {
           upstatus = hr.status;
           auto it = hr.headers.find("content-type");
           if (it != hr.headers.end()) upct = it->second;
-        });
+        },
+        &tls);
     if (started) {
       w.end_stream();
       return;
@@ -208,7 +214,7 @@ void OagwModule::proxy(HttpRequest& req, ResponseWriter& w) {
 
   // forward (buffered)
   auto resp = http_request(host, port, req.method, target, fwd, req.body,
-                           10000);
+                           10000, nullptr, nullptr, &tls);
   if (!resp)
     throw Problem{502, "Bad Gateway", "about:blank",
                   "upstream connect failed", "provider_error"};

@@ -122,6 +122,7 @@ class StaticAuthnResolver : public AuthnResolverClient {
         cache->jwks_uri = jwks_uri;
         cache->discovery_url = disc;
         cache->ttl_s = (int)jw.at("jwks_ttl_s").as_int(300);
+        cache->tls_ca_file = jw.at("tls_ca_file").as_string("");
         jwt_.jwks = cache;
       }
     }

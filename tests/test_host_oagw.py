@@ -227,3 +227,72 @@ def test_route_matching_and_priority(server, upstream):
     st, body = _http("POST", base + "/oagw/v1/proxy/routed/other",
                      body={})
     assert st == 404 and json.loads(body)["code"] == "route_not_found"
+
+
+# ---------------------------------------------------------------------------
+# TLS upstream (https data plane: client.cpp Io/tls_connect; oagw.cpp
+# scheme-aware endpoints + tls.ca_file/tls.verify — reference
+# oagw/src/upstream tls handling)
+
+@pytest.fixture(scope="module")
+def tls_upstream(tmp_path_factory):
+    """MockUpstream behind a self-signed TLS cert for 127.0.0.1."""
+    import ssl
+    import subprocess
+    d = tmp_path_factory.mktemp("hs-oagw-tls")
+    crt, key = d / "srv.crt", d / "srv.key"
+    subprocess.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+         "-keyout", str(key), "-out", str(crt), "-days", "2",
+         "-subj", "/CN=127.0.0.1",
+         "-addext", "subjectAltName=IP:127.0.0.1"],
+        check=True, capture_output=True)
+    srv = http.server.ThreadingHTTPServer(("127.0.0.1", 0), MockUpstream)
+    ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+    ctx.load_cert_chain(str(crt), str(key))
+    srv.socket = ctx.wrap_socket(srv.socket, server_side=True)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    yield srv.server_address[1], str(crt), str(d)
+    srv.shutdown()
+
+
+def test_proxy_https_upstream_with_ca(server, tls_upstream):
+    port, ca, _ = tls_upstream
+    base = BASE.format(server.port)
+    up = {"alias": "tlsup", "server": {"endpoints": [
+        {"scheme": "https", "host": "127.0.0.1", "port": port}]},
+        "tls": {"ca_file": ca},
+        "auth": {"plugin_type": "apikey",
+                 "config": {"header": "x-api-key", "value": "tls-key"}}}
+    st, body = _http("POST", base + "/oagw/v1/upstreams", body=up)
+    assert st == 201, body
+    st, body = _http("POST", base + "/oagw/v1/proxy/tlsup/v1/sec",
+                     body={"q": "tls"})
+    assert st == 200, body
+    j = json.loads(body)
+    assert j["path"] == "/v1/sec" and j["api_key"] == "tls-key"
+
+
+def test_proxy_https_verify_failure_is_502(server, tls_upstream):
+    """Untrusted cert (no CA configured, verify on) must NOT proxy."""
+    port, _, _ = tls_upstream
+    base = BASE.format(server.port)
+    up = {"alias": "tlsbad", "server": {"endpoints": [
+        {"scheme": "https", "host": "127.0.0.1", "port": port}]}}
+    st, body = _http("POST", base + "/oagw/v1/upstreams", body=up)
+    assert st == 201, body
+    st, body = _http("POST", base + "/oagw/v1/proxy/tlsbad/v1/x", body={})
+    assert st == 502 and json.loads(body)["code"] == "provider_error"
+
+
+def test_proxy_https_verify_disabled(server, tls_upstream):
+    port, _, _ = tls_upstream
+    base = BASE.format(server.port)
+    up = {"alias": "tlsnv", "server": {"endpoints": [
+        {"scheme": "https", "host": "127.0.0.1", "port": port}]},
+        "tls": {"verify": False}}
+    st, body = _http("POST", base + "/oagw/v1/upstreams", body=up)
+    assert st == 201, body
+    st, body = _http("GET", base + "/oagw/v1/proxy/tlsnv/ping")
+    assert st == 200, body
