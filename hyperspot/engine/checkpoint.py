@@ -46,8 +46,21 @@ def _shard_from_full(param: torch.Tensor, full: torch.Tensor) -> None:
                          non_blocking=True))
 
 
+def _shard_group_of(param: torch.Tensor):
+    """(group, my_rank, size) the parameter is sharded over: the TP group
+    for the linear layers, the EP group for MoE expert tensors (which
+    carry `_shard_group == "ep"`; mixtral.py block-shards whole experts
+    along dim 0 with the same slice metadata)."""
+    from hyperspot.parallel.state import (get_ep_group, get_ep_rank,
+                                          get_ep_size, get_tp_group,
+                                          get_tp_rank, get_tp_size)
+    if getattr(param, "_shard_group", "tp") == "ep":
+        return get_ep_group(), get_ep_rank(), get_ep_size()
+    return get_tp_group(), get_tp_rank(), get_tp_size()
+
+
 def _full_from_shards(param: torch.Tensor, group) -> torch.Tensor:
-    """All-gather this parameter's shards over the TP group and rebuild
+    """All-gather this parameter's shards over its shard group and rebuild
     the full tensor (every rank computes it; only rank 0 writes)."""
     import torch.distributed as dist
     world = dist.get_world_size(group)
@@ -78,8 +91,7 @@ def _slices_for_rank(param: torch.Tensor, r: int):
     """Derive rank r's slices from this rank's slice pattern.  Within
     each section the full_start is section_base + r*length; section_base
     = full_start - my_rank*length."""
-    from hyperspot.parallel.state import get_tp_rank
-    me = get_tp_rank()
+    _, me, _ = _shard_group_of(param)
     out = []
     for f0, ln, s0 in param._tp_slices:
         base = f0 - me * ln
@@ -90,26 +102,19 @@ def _slices_for_rank(param: torch.Tensor, r: int):
 def save_checkpoint(model, path: str, meta: Dict = None) -> None:
     """Write the FULL-layout checkpoint.  Collective under TP (all ranks
     must call); only rank 0 writes the file."""
+    import torch.distributed as dist
     from safetensors.torch import save_file
-    from hyperspot.parallel.state import get_ep_size, get_tp_group, \
-        get_tp_rank, get_tp_size
-    if get_ep_size() > 1:
-        # expert tensors are sharded per-rank WITHOUT slice metadata (a
-        # rank owns whole experts); gathering them needs an EP-aware
-        # collector — fail loudly rather than write one rank's experts
-        raise NotImplementedError(
-            "checkpoint save at ep>1 (EP expert gather) is not "
-            "implemented; save from an ep=1 deployment")
-    tp = get_tp_size()
     state = {}
     for k, v in _params_and_buffers(model).items():
-        if k.endswith("cos_sin"):
+        if k.endswith("cos_sin") or k.endswith("ep_overflow"):
             continue
-        if tp > 1 and getattr(v, "_tp_slices", None) is not None:
-            state[k] = _full_from_shards(v, get_tp_group())
+        if getattr(v, "_tp_slices", None) is not None:
+            group, _, size = _shard_group_of(v)
+            state[k] = (_full_from_shards(v, group) if size > 1
+                        else v.detach().contiguous().cpu())
         else:
             state[k] = v.detach().contiguous().cpu()
-    if tp > 1 and get_tp_rank() != 0:
+    if dist.is_initialized() and dist.get_rank() != 0:
         return
     os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
     save_file(state, path, metadata={"hyperspot": json.dumps(meta or {})})
@@ -120,17 +125,13 @@ def load_checkpoint_into(model, path: str) -> float:
     reused (copy_), so captured hipGraphs remain valid.  Full-layout
     checkpoints load into TP shards via the `_tp_slices` metadata."""
     from safetensors import safe_open
-    from hyperspot.parallel.state import get_ep_size
-    if get_ep_size() > 1:
-        raise NotImplementedError(
-            "hot-swap at ep>1 (EP expert scatter) is not implemented")
     t0 = time.monotonic()
     device = next(model.parameters()).device
     params = _params_and_buffers(model)
     with safe_open(path, framework="pt", device="cpu") as f:
         keys = set(f.keys())
-        missing = [k for k in params
-                   if k not in keys and not k.endswith("cos_sin")]
+        missing = [k for k in params if k not in keys
+                   and not k.endswith(("cos_sin", "ep_overflow"))]
         if missing:
             raise ValueError(f"checkpoint misses keys: {missing[:5]}...")
         with torch.inference_mode():

@@ -10,6 +10,9 @@ Sharding model (MI355X-first):
 
 from __future__ import annotations
 
+import math
+import os
+
 import torch
 import torch.distributed as dist
 from torch import nn
@@ -35,6 +38,12 @@ class MixtralMoE(nn.Module):
         assert spec.num_experts % ep == 0, (spec.num_experts, ep)
         self.experts_per_rank = spec.num_experts // ep
         self.local_expert_start = r * self.experts_per_rank
+        # fixed-capacity (capture-safe) dispatch knobs; see _ep_moe_static
+        self.capacity_factor = float(os.environ.get("HS_EP_CAPACITY", "2.0"))
+        self.force_static_ep = os.environ.get("HS_EP_STATIC", "") == "1"
+        self.register_buffer("ep_overflow",
+                             torch.zeros((), dtype=torch.long),
+                             persistent=False)
         h, i = spec.hidden_size, spec.intermediate_size
         w13, w2 = [], []
         for e in range(self.local_expert_start,
@@ -60,6 +69,19 @@ class MixtralMoE(nn.Module):
         else:
             self.w13 = nn.Parameter(torch.stack(w13), requires_grad=False)
             self.w2 = nn.Parameter(torch.stack(w2), requires_grad=False)
+        if ep > 1:
+            # expert tensors are block-sharded along dim 0 (whole experts
+            # per rank); the same slice metadata the TP layers carry lets
+            # checkpoint save/load gather/scatter the FULL [E, ...] layout
+            # (engine/checkpoint.py), just over the EP group instead
+            for p in (self.w13, self.w2) + (
+                    (self.w13_scale, self.w2_scale)
+                    if self.quant == "fp8" else ()):
+                p._tp_slices = [(self.local_expert_start,
+                                 self.experts_per_rank, 0)]
+                p._tp_full_shape = (spec.num_experts,) + tuple(p.shape[1:])
+                p._tp_shard_dim = 0
+                p._shard_group = "ep"
 
     def _expert_ffn(self, x: torch.Tensor, e_local: int) -> torch.Tensor:
         if self.quant == "fp8":   # CPU/EP fallback: dequantized math
@@ -123,7 +145,17 @@ class MixtralMoE(nn.Module):
         combine back (all-to-all), weight and reduce.  At quant=fp8 the
         dispatched activations travel as e4m3fn bytes + per-token f32
         scales — half the xGMI wire bytes — and feed the fp8 grouped GEMM
-        directly on the owner rank."""
+        directly on the owner rank.
+
+        Two dispatch modes:
+          - dynamic (exact): per-rank counts exchanged first, variable-split
+            all-to-alls.  Needs a host sync (`.tolist()`) so it cannot be
+            captured in a hipGraph — used for eager/prefill forwards.
+          - static (capture-safe): fixed-capacity buckets, see
+            _ep_moe_static — used under graph capture (decode buckets)."""
+        if self.force_static_ep or (
+                x.is_cuda and torch.cuda.is_current_stream_capturing()):
+            return self._ep_moe_static(x, weights, ids)
         ep = get_ep_size()
         group = get_ep_group()
         T, H = x.shape
@@ -186,6 +218,95 @@ class MixtralMoE(nn.Module):
         out = torch.zeros_like(x)
         token_idx = torch.arange(T, device=x.device).repeat_interleave(K)[order]
         out.index_add_(0, token_idx, back * weights.reshape(-1)[order, None])
+        return out
+
+    def _ep_capacity(self, T: int) -> int:
+        """Tokens-per-destination-rank bucket size: mean load x factor."""
+        ep = get_ep_size()
+        c = math.ceil(T * self.top_k / ep * self.capacity_factor)
+        return max(8, min(T * self.top_k, c))
+
+    def _ep_moe_static(self, x, weights, ids):
+        """Capture-safe EP dispatch: fixed-capacity buckets + EQUAL-split
+        all-to-alls.
+
+        Every tensor shape here is a function of (T, ep, capacity) only —
+        no `.tolist()`/`.item()` host syncs — so hipGraph capture records
+        the RCCL all-to-alls exactly like TP's decode all-reduce, and
+        ep>1 decode no longer forces eager mode.  The trade is the
+        standard fixed-capacity MoE one: a rank sending more than
+        `capacity` tokens to one owner drops the excess top-k assignments
+        (counted in `ep_overflow`; remaining experts still serve the
+        token).  Capacity = mean load x HS_EP_CAPACITY (default 2.0).
+        Eager forwards (prefill) keep the exact dynamic dispatch above.
+        """
+        ep = get_ep_size()
+        group = get_ep_group()
+        T, H = x.shape
+        K = self.top_k
+        C = self._ep_capacity(T)
+        dev = x.device
+        fp8_wire = self.quant == "fp8"
+        flat_ids = ids.reshape(-1).to(torch.long)          # [T*K]
+        owner = flat_ids // self.experts_per_rank
+        order = torch.argsort(owner, stable=True)
+        sorted_owner = owner[order]
+        counts = torch.bincount(owner, minlength=ep)
+        offsets = torch.cumsum(counts, 0) - counts         # exclusive
+        slot = torch.arange(T * K, device=dev) - offsets[sorted_owner]
+        keep = slot < C
+        trash = torch.full_like(slot, ep * C)              # one spare row
+        dest = torch.where(keep, sorted_owner * C + slot, trash)
+        src_rows = torch.arange(T, device=dev).repeat_interleave(K)[order]
+        self.ep_overflow += (~keep).sum()
+        # scatter tokens into their destination buckets (padding rows stay 0)
+        send_x = x.new_zeros(ep * C + 1, H)
+        send_x.index_copy_(0, dest, x[src_rows])
+        send_eid = torch.full((ep * C + 1,), -1, dtype=torch.long, device=dev)
+        send_eid.index_copy_(0, dest, flat_ids[order])
+        recv_eid = torch.empty(ep * C, dtype=torch.long, device=dev)
+        dist.all_to_all_single(recv_eid, send_eid[:ep * C], group=group)
+        if fp8_wire:
+            from hyperspot.parallel.layers import quant_fp8_rowwise
+            sq, ss = quant_fp8_rowwise(send_x[:ep * C])
+            recv_q = torch.empty(ep * C, H, dtype=torch.uint8, device=dev)
+            recv_s = torch.empty(ep * C, dtype=torch.float32, device=dev)
+            dist.all_to_all_single(recv_q, sq.view(torch.uint8), group=group)
+            dist.all_to_all_single(recv_s, ss, group=group)
+            recv_x = None
+        else:
+            recv_x = torch.empty(ep * C, H, dtype=x.dtype, device=dev)
+            dist.all_to_all_single(recv_x, send_x[:ep * C], group=group)
+        # local grouped FFN over all ep*C rows; padding rows (eid -1) are
+        # zero inputs routed to local expert 0 and discarded on combine
+        local = (recv_eid - self.local_expert_start).clamp(
+            0, self.experts_per_rank - 1)
+        if fp8_wire and x.is_cuda and ops.have_native():
+            xq = ops.QTensor(recv_q.view(torch.float8_e4m3fn), recv_s)
+            ones = torch.ones(ep * C, 1, device=dev, dtype=torch.float32)
+            y = ops.moe_ffn_fp8(xq, self.w13, self.w13_scale, self.w2,
+                                self.w2_scale, ones,
+                                local.view(-1, 1).to(torch.int32))
+        elif not fp8_wire and x.is_cuda and ops.have_native():
+            ones = torch.ones(ep * C, 1, device=dev, dtype=torch.float32)
+            y = ops.moe_ffn(recv_x, self.w13, self.w2, ones,
+                            local.view(-1, 1).to(torch.int32))
+        else:
+            if fp8_wire:    # CPU/EP test path: dequantize the wire bytes
+                recv_x = (recv_q.view(torch.float8_e4m3fn).float()
+                          * recv_s[:, None]).to(x.dtype)
+            y = torch.empty_like(recv_x)
+            for e in range(self.experts_per_rank):
+                m = local == e
+                if bool(m.any()):
+                    y[m] = self._expert_ffn(recv_x[m], e)
+        back = torch.empty_like(y)
+        dist.all_to_all_single(back, y, group=group)
+        # gather each kept assignment's result back; dropped ones get w=0
+        gidx = torch.where(keep, dest, torch.zeros_like(dest))
+        w = weights.reshape(-1)[order] * keep.to(weights.dtype)
+        out = torch.zeros_like(x)
+        out.index_add_(0, src_rows, back[gidx] * w[:, None])
         return out
 
 

@@ -740,13 +740,12 @@ def main():
         quant=args.quant, kv_dtype=args.kv_dtype)
     rank = 0
     # MoE models shard experts over the same ranks as TP attention
-    # (config 4: Mixtral EP over RCCL all-to-all); EP decode steps do
-    # count exchanges host-side, so graphs are disabled at ep>1
+    # (config 4: Mixtral EP over RCCL all-to-all).  Decode graphs stay ON
+    # at ep>1: under capture the MoE switches to the fixed-capacity
+    # equal-split dispatch (mixtral._ep_moe_static), so the all-to-alls
+    # are recorded like TP's all-reduce; eager/prefill forwards keep the
+    # exact variable-split dispatch.
     ep = args.ep or (args.tp if cfg.spec().is_moe else 1)
-    if ep > 1 and not cfg.enforce_eager:
-        log.info("ep=%d: forcing eager decode (EP all-to-all is not "
-                 "graph-captured yet)", ep)
-        cfg.enforce_eager = True
     cfg.ep_size = ep
     if args.tp > 1:
         from hyperspot.parallel.state import initialize_model_parallel
