@@ -6,6 +6,8 @@
 #include <algorithm>
 #include <chrono>
 #include <cmath>
+#include <ctime>
+#include <sstream>
 
 #include "../util/log.h"
 #include "llm_gateway.h"       // AdmissionClient, ChatInvoker
@@ -135,7 +137,238 @@ Json validate_entrypoint(const Json& body) {
   const Json& rp = body.at("retry_policy");
   if (rp.is_object() && rp.at("max_attempts").as_int(1) < 1)
     issue("retry_policy.max_attempts must be >= 1");
+  const Json& io = body.at("io_schema");
+  if (!io.is_null() && !io.is_object())
+    issue("io_schema must be an object (params/returns/errors)");
+  if (io.is_object())
+    for (const char* k : {"params", "returns"}) {
+      const Json& s = io.at(k);
+      if (!s.is_null() && !s.is_object())
+        issue(std::string("io_schema.") + k +
+              " must be a JSON Schema object, gts $ref, or null (void)");
+    }
   return issues;
+}
+
+// ---- schedule expressions (ADR:2038-2164: cron | ISO-8601 interval,
+// evaluated in UTC — `timezone` other than UTC is rejected at create) ----
+
+// parse one cron field ("*", "a", "a-b", "*/n", "a-b/n", lists) into an
+// allow-set over [lo, hi]
+bool cron_field(const std::string& f, int lo, int hi,
+                std::vector<bool>& out) {
+  out.assign((size_t)hi + 1, false);
+  size_t pos = 0;
+  while (pos <= f.size()) {
+    size_t comma = f.find(',', pos);
+    std::string part = f.substr(
+        pos, comma == std::string::npos ? std::string::npos : comma - pos);
+    if (part.empty()) return false;
+    int step = 1;
+    size_t slash = part.find('/');
+    bool stepped = slash != std::string::npos;
+    if (stepped) {
+      step = atoi(part.c_str() + slash + 1);
+      part = part.substr(0, slash);
+      if (step < 1) return false;
+    }
+    int a = lo, b = hi;
+    if (part != "*") {
+      size_t dash = part.find('-');
+      if (dash != std::string::npos) {
+        a = atoi(part.substr(0, dash).c_str());
+        b = atoi(part.substr(dash + 1).c_str());
+      } else {
+        a = atoi(part.c_str());
+        b = stepped ? hi : a;   // "5/10" = every 10 starting at 5
+      }
+    }
+    if (a < lo || b > hi || a > b) return false;
+    for (int v = a; v <= b; v += step) out[(size_t)v] = true;
+    if (comma == std::string::npos) break;
+    pos = comma + 1;
+  }
+  return true;
+}
+
+// next fire strictly after `after`, or 0 on parse failure / no match
+// within 366 days.  5 fields: min hour day-of-month month day-of-week.
+time_t cron_next(const std::string& expr, time_t after) {
+  std::istringstream ss(expr);
+  std::string f[5], extra;
+  for (int i = 0; i < 5; ++i)
+    if (!(ss >> f[i])) return 0;
+  if (ss >> extra) return 0;
+  std::vector<bool> mi, ho, dom, mon, dow;
+  if (!cron_field(f[0], 0, 59, mi) || !cron_field(f[1], 0, 23, ho) ||
+      !cron_field(f[2], 1, 31, dom) || !cron_field(f[3], 1, 12, mon) ||
+      !cron_field(f[4], 0, 6, dow))
+    return 0;
+  const bool dom_any = f[2] == "*", dow_any = f[4] == "*";
+  time_t t = after - after % 60 + 60;
+  for (int i = 0; i < 366 * 24 * 60; ++i, t += 60) {
+    tm g{};
+    gmtime_r(&t, &g);
+    if (!mon[(size_t)g.tm_mon + 1] || !mi[(size_t)g.tm_min] ||
+        !ho[(size_t)g.tm_hour])
+      continue;
+    const bool dmatch = dom[(size_t)g.tm_mday];
+    const bool wmatch = dow[(size_t)g.tm_wday];
+    // vixie-cron rule: both day fields restricted -> either may match
+    if (dom_any && dow_any ? true
+        : dom_any          ? wmatch
+        : dow_any          ? dmatch
+                           : (dmatch || wmatch))
+      return t;
+  }
+  return 0;
+}
+
+// ISO-8601 duration (PT1H, P1D, PT90S, P1DT12H...) or plain seconds
+double interval_seconds(const std::string& s) {
+  if (s.empty()) return 0;
+  if (s[0] != 'P') {
+    char* end = nullptr;
+    double v = strtod(s.c_str(), &end);
+    return end && *end == '\0' ? v : 0;
+  }
+  double total = 0, num = 0;
+  bool in_time = false, have = false;
+  for (size_t i = 1; i < s.size(); ++i) {
+    char c = s[i];
+    if (c == 'T') {
+      in_time = true;
+      continue;
+    }
+    if (isdigit((unsigned char)c) || c == '.') {
+      size_t j = i;
+      while (j < s.size() &&
+             (isdigit((unsigned char)s[j]) || s[j] == '.'))
+        j++;
+      num = atof(s.substr(i, j - i).c_str());
+      have = true;
+      i = j - 1;
+      continue;
+    }
+    double mult = 0;
+    if (c == 'W' && !in_time) mult = 604800;
+    else if (c == 'D' && !in_time) mult = 86400;
+    else if (c == 'H' && in_time) mult = 3600;
+    else if (c == 'M') mult = in_time ? 60 : 2592000;
+    else if (c == 'S' && in_time) mult = 1;
+    else return 0;
+    if (!have) return 0;
+    total += num * mult;
+    have = false;
+  }
+  return total;
+}
+
+time_t parse_iso(const std::string& s) {
+  tm g{};
+  if (!strptime(s.c_str(), "%Y-%m-%dT%H:%M:%S", &g)) return 0;
+  return timegm(&g);
+}
+
+std::string iso_of(time_t t) {
+  char buf[32];
+  tm g{};
+  gmtime_r(&t, &g);
+  strftime(buf, sizeof buf, "%Y-%m-%dT%H:%M:%SZ", &g);
+  return buf;
+}
+
+// next occurrence strictly after `after` per the expression, 0 = invalid
+time_t sched_next(const std::string& kind, const std::string& value,
+                  time_t after) {
+  if (kind == "cron") return cron_next(value, after);
+  if (kind == "interval") {
+    double iv = interval_seconds(value);
+    return iv >= 1 ? after + (time_t)iv : 0;
+  }
+  return 0;
+}
+
+// ---- io_schema validation (ADR:131-185; PRD BR-032/BR-037: inputs are
+// validated before invocation).  Minimal JSON-Schema subset: type,
+// required, properties, additionalProperties:false, items, enum, const,
+// minimum/maximum, minLength/maxLength.  `$ref` (gts:// types) passes —
+// resolution against the types registry is out of scope here. ----
+std::string schema_err(const Json& sch, const Json& v,
+                       const std::string& path) {
+  if (!sch.is_object()) return "";
+  if (sch.contains("$ref")) return "";
+  const Json& typ = sch.at("type");
+  auto type_ok = [&](const std::string& t) {
+    if (t == "object") return v.is_object();
+    if (t == "array") return v.is_array();
+    if (t == "string") return v.is_string();
+    if (t == "number") return v.is_number();
+    if (t == "integer")
+      return v.is_number() && v.as_number() == (double)(long long)v.as_number();
+    if (t == "boolean") return v.is_bool();
+    if (t == "null") return v.is_null();
+    return true;
+  };
+  if (typ.is_string() && !type_ok(typ.as_string()))
+    return path + ": expected type " + typ.as_string();
+  if (typ.is_array()) {
+    bool any = false;
+    for (auto& t : typ.arr())
+      if (type_ok(t.as_string())) any = true;
+    if (!any) return path + ": type not in allowed set";
+  }
+  if (sch.at("enum").is_array()) {
+    bool any = false;
+    for (auto& e : sch.at("enum").arr())
+      if (e.dump() == v.dump()) any = true;
+    if (!any) return path + ": value not in enum";
+  }
+  if (sch.contains("const") && sch.at("const").dump() != v.dump())
+    return path + ": value != const";
+  if (v.is_object()) {
+    const Json& req = sch.at("required");
+    if (req.is_array())
+      for (auto& r : req.arr())
+        if (!v.contains(r.as_string()))
+          return path + "." + r.as_string() + ": required";
+    const Json& props = sch.at("properties");
+    if (props.is_object()) {
+      for (auto& [k, psch] : props.obj())
+        if (v.contains(k)) {
+          std::string e = schema_err(psch, v.at(k), path + "." + k);
+          if (!e.empty()) return e;
+        }
+      const Json& ap = sch.at("additionalProperties");
+      if (ap.is_bool() && !ap.as_bool())
+        for (auto& [k, _] : v.obj())
+          if (!props.contains(k))
+            return path + "." + k + ": additional property not allowed";
+    }
+  }
+  if (v.is_array() && sch.at("items").is_object())
+    for (size_t i = 0; i < v.size(); ++i) {
+      std::string e = schema_err(sch.at("items"), v.at(i),
+                                 path + "[" + std::to_string(i) + "]");
+      if (!e.empty()) return e;
+    }
+  if (v.is_string()) {
+    if (sch.contains("minLength") &&
+        v.as_string().size() < (size_t)sch.at("minLength").as_int())
+      return path + ": shorter than minLength";
+    if (sch.contains("maxLength") &&
+        v.as_string().size() > (size_t)sch.at("maxLength").as_int())
+      return path + ": longer than maxLength";
+  }
+  if (v.is_number()) {
+    if (sch.contains("minimum") &&
+        v.as_number() < sch.at("minimum").as_number())
+      return path + ": below minimum";
+    if (sch.contains("maximum") &&
+        v.as_number() > sch.at("maximum").as_number())
+      return path + ": above maximum";
+  }
+  return "";
 }
 
 SecurityContext sec_of(HttpRequest& rq) {
@@ -172,6 +405,32 @@ Json row_invocation(const DbRow& r) {
   }
   const std::string err = r.at("error").as_string();
   if (!err.empty()) j["error"] = err;
+  return j;
+}
+
+Json row_schedule(const DbRow& r) {
+  Json j = Json::object();
+  j["schedule_id"] = r.at("id");
+  j["tenant_id"] = r.at("tenant_id");
+  j["entrypoint_id"] = r.at("entrypoint_id");
+  j["name"] = r.at("name");
+  j["timezone"] = r.at("timezone");
+  Json expr = Json::object();
+  expr["kind"] = r.at("expr_kind");
+  expr["value"] = r.at("expr_value");
+  j["expression"] = expr;
+  try {
+    j["input_overrides"] = Json::parse(r.at("input_overrides").as_string());
+  } catch (...) {}
+  j["missed_policy"] = r.at("missed_policy");
+  j["status"] = r.at("status");
+  for (const char* k : {"next_run_at", "last_run_at"}) {
+    const std::string v = r.at(k).as_string();
+    if (!v.empty()) j[k] = v;
+    else j[k] = Json();
+  }
+  j["created_at"] = r.at("created_at");
+  j["updated_at"] = r.at("updated_at");
   return j;
 }
 
@@ -233,6 +492,25 @@ void ServerlessRuntimeModule::init(ModuleCtx& ctx) {
        "  step_name TEXT NOT NULL DEFAULT '',"
        "  duration_ms INTEGER NOT NULL DEFAULT -1,"
        "  detail TEXT NOT NULL DEFAULT '')"},
+      {"0004_schedules",
+       "CREATE TABLE schedules ("
+       "  tenant_id TEXT NOT NULL,"
+       "  id TEXT NOT NULL UNIQUE,"
+       "  entrypoint_id TEXT NOT NULL,"
+       "  name TEXT NOT NULL,"
+       "  timezone TEXT NOT NULL DEFAULT 'UTC',"
+       "  expr_kind TEXT NOT NULL,"
+       "  expr_value TEXT NOT NULL,"
+       "  input_overrides TEXT NOT NULL DEFAULT '{}',"
+       "  missed_policy TEXT NOT NULL DEFAULT 'skip',"
+       "  status TEXT NOT NULL DEFAULT 'active',"
+       "  next_run_at TEXT NOT NULL DEFAULT '',"
+       "  last_run_at TEXT NOT NULL DEFAULT '',"
+       "  created_at TEXT NOT NULL,"
+       "  updated_at TEXT NOT NULL)"},
+      {"0005_invocation_schedule",
+       "ALTER TABLE invocations ADD COLUMN schedule_id TEXT NOT NULL "
+       "DEFAULT ''"},
   });
 }
 
@@ -556,6 +834,111 @@ void ServerlessRuntimeModule::executor_loop() {
   }
 }
 
+std::string ServerlessRuntimeModule::create_invocation(
+    const std::string& tenant, const std::string& ep_id, const Json& input,
+    const std::string& mode, const std::string& schedule_id,
+    const std::string& preset_status) {
+  SecureConn conn(*db_, AccessScope::for_tenant(tenant));
+  const std::string id =
+      "inv-" + std::to_string(++ctr_) + "-" +
+      std::to_string((long long)(now_s() * 1000) % 100000);
+  conn.insert("invocations",
+              {{"id", DbValue::S(id)},
+               {"entrypoint_id", DbValue::S(ep_id)},
+               {"status", DbValue::S(preset_status.empty() ? "queued"
+                                                           : preset_status)},
+               {"mode", DbValue::S(mode)},
+               {"input", DbValue::S(input.dump())},
+               {"schedule_id", DbValue::S(schedule_id)},
+               {"created_at", DbValue::S(now_iso())},
+               {"updated_at", DbValue::S(now_iso())}});
+  if (preset_status.empty()) enqueue(tenant, id);
+  return id;
+}
+
+void ServerlessRuntimeModule::schedule_tick() {
+  const time_t now = (time_t)now_s();
+  std::vector<DbRow> due;
+  {
+    std::lock_guard<std::mutex> dblk(db_->mu());
+    due = db_->query(
+        "SELECT * FROM schedules WHERE status='active' AND "
+        "next_run_at != '' AND next_run_at <= ?",
+        {DbValue::S(iso_of(now))});
+  }
+  for (auto& s : due) {
+    const std::string tenant = s.at("tenant_id").as_string();
+    const std::string sid = s.at("id").as_string();
+    const std::string kind = s.at("expr_kind").as_string();
+    const std::string value = s.at("expr_value").as_string();
+    const std::string policy = s.at("missed_policy").as_string();
+    time_t next = parse_iso(s.at("next_run_at").as_string());
+    if (next == 0) continue;
+    // occurrences to execute per the missed-schedule policy (BR-022):
+    //   backfill — one per missed slot; catch_up — exactly one;
+    //   skip — one only if the slot is fresh (within a grace window)
+    int fires = 0;
+    if (policy == "backfill") {
+      time_t t = next;
+      while (t != 0 && t <= now && fires < 100) {
+        fires++;
+        t = sched_next(kind, value, t);
+      }
+      next = t;
+    } else {
+      fires = (policy == "catch_up" || now - next <= 5) ? 1 : 0;
+      next = sched_next(kind, value, now);
+    }
+    SecureConn conn(*db_, AccessScope::for_tenant(tenant));
+    // CAS on next_run_at so two ticks / hosts cannot double-fire
+    int won = conn.update(
+        "schedules",
+        {{"next_run_at", DbValue::S(next ? iso_of(next) : "")},
+         {"last_run_at", DbValue::S(iso_of(now))},
+         {"updated_at", DbValue::S(now_iso())}},
+        "id=? AND next_run_at=?",
+        {DbValue::S(sid), DbValue::S(s.at("next_run_at").as_string())});
+    if (won != 1) continue;
+    if (fires == 0) continue;
+    // resolve the entrypoint; merge its input defaults with overrides
+    auto ep = conn.select("entrypoints", "id=?",
+                          {DbValue::S(s.at("entrypoint_id").as_string())},
+                          "id", false, 1, std::nullopt);
+    if (ep.items.empty() ||
+        ep.items[0].at("status").as_string() != "active")
+      continue;               // paused implicitly while not active
+    Json spec;
+    try { spec = Json::parse(ep.items[0].at("spec").as_string()); }
+    catch (...) {}
+    Json input = spec.at("input_defaults").is_object()
+                     ? spec.at("input_defaults")
+                     : Json::object();
+    Json ov;
+    try { ov = Json::parse(s.at("input_overrides").as_string()); }
+    catch (...) {}
+    if (ov.is_object())
+      for (auto& [k, v] : ov.obj()) input[k] = v;
+    const Json& params = spec.path("io_schema.params");
+    std::string verr =
+        params.is_object() ? schema_err(params, input, "input") : "";
+    for (int i = 0; i < fires; ++i) {
+      if (!verr.empty()) {
+        // recorded (visible in schedule history), never executed
+        std::string id = create_invocation(
+            tenant, s.at("entrypoint_id").as_string(), input, "async",
+            sid, "failed");
+        SecureConn c2(*db_, AccessScope::for_tenant(tenant));
+        c2.update("invocations",
+                  {{"error", DbValue::S("input validation: " + verr)}},
+                  "id=?", {DbValue::S(id)});
+      } else {
+        create_invocation(tenant, s.at("entrypoint_id").as_string(),
+                          input, "async", sid);
+      }
+    }
+  }
+}
+
 void ServerlessRuntimeModule::timer_loop() {
   while (!stopping_) {
     std::vector<Timer> due;
@@ -588,6 +971,17 @@ void ServerlessRuntimeModule::timer_loop() {
             timeline(t.tenant, t.id, "dead_lettered", "dead_lettered");
           done_cv_.notify_all();
         }
+      }
+    }
+    // scan schedules every ~500 ms (PRD "scheduled execution start skew
+    // p95 <= 5 s" NFR — a sub-second scan keeps skew well inside that)
+    const double now = now_s();
+    if (now - last_sched_scan_ >= 0.5) {
+      last_sched_scan_ = now;
+      try {
+        schedule_tick();
+      } catch (const std::exception& e) {
+        LOG_ERROR("serverless", "schedule tick failed: %s", e.what());
       }
     }
   }
@@ -890,6 +1284,20 @@ void ServerlessRuntimeModule::register_rest(ModuleCtx& ctx,
         if (ep.items[0].at("status").as_string() != "active")
           throw Problem{409, "Conflict", "about:blank",
                         "entrypoint is not active", "conflict"};
+        // io_schema.params gate (BR-032: inputs validated BEFORE start)
+        Json vinput = body.at("input");
+        if (vinput.is_null()) vinput = Json::object();
+        Json espec;
+        try {
+          espec = Json::parse(ep.items[0].at("spec").as_string());
+        } catch (...) {}
+        const Json& params = espec.path("io_schema.params");
+        if (params.is_object()) {
+          std::string verr = schema_err(params, vinput, "input");
+          if (!verr.empty())
+            throw Problem{400, "Bad Request", "about:blank",
+                          "input validation: " + verr, "invalid_input"};
+        }
         if (body.at("dry_run").as_bool(false)) {
           Json out = Json::object();
           out["valid"] = true;
@@ -1109,6 +1517,214 @@ void ServerlessRuntimeModule::register_rest(ModuleCtx& ctx,
           if (!det.empty()) e["detail"] = det;
           items.push_back(e);
         }
+        Json out = Json::object();
+        out["items"] = items;
+        w.respond(200, "application/json", out.dump());
+      });
+
+  // ---- schedules (ADR:2038-2164 + Schedule API table ADR:2884-2893;
+  // ':pause'/':resume' actions mounted as subresource segments per this
+  // host's route grammar).  Expressions are evaluated in UTC. ----
+  auto load_schedule = [this](SecureConn& conn, const std::string& id) {
+    auto page = conn.select("schedules", "id=?", {DbValue::S(id)}, "id",
+                            false, 1, std::nullopt);
+    if (page.items.empty())
+      throw Problem::not_found("schedule not found");
+    return page.items[0];
+  };
+
+  auto parse_schedule = [parse_body](HttpRequest& rq) {
+    Json b = parse_body(rq);
+    if (b.at("name").as_string().empty())
+      throw Problem{400, "Bad Request", "about:blank",
+                    "'name' is required", "validation_error"};
+    if (b.at("entrypoint_id").as_string().empty())
+      throw Problem{400, "Bad Request", "about:blank",
+                    "'entrypoint_id' is required", "validation_error"};
+    const std::string tz = b.at("timezone").as_string("UTC");
+    if (tz != "UTC")
+      throw Problem{400, "Bad Request", "about:blank",
+                    "only timezone 'UTC' is supported",
+                    "validation_error"};
+    const std::string kind = b.path("expression.kind").as_string();
+    const std::string value = b.path("expression.value").as_string();
+    if (sched_next(kind, value, (time_t)now_s()) == 0)
+      throw Problem{400, "Bad Request", "about:blank",
+                    "expression must be a valid cron (5 fields) or "
+                    "ISO-8601 interval >= 1s",
+                    "validation_error"};
+    const std::string mp = b.at("missed_policy").as_string("skip");
+    if (mp != "skip" && mp != "catch_up" && mp != "backfill")
+      throw Problem{400, "Bad Request", "about:blank",
+                    "missed_policy must be skip|catch_up|backfill",
+                    "validation_error"};
+    return b;
+  };
+
+  reg("POST", "/serverless-runtime/v1/schedules", "create_schedule",
+      [this, parse_schedule, load_schedule](HttpRequest& rq,
+                                            ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        Json b = parse_schedule(rq);
+        SecureConn conn(*db_, scope_for(sec, "create",
+                                        "serverless-runtime:schedules"));
+        auto ep = conn.select(
+            "entrypoints", "id=?",
+            {DbValue::S(b.at("entrypoint_id").as_string())}, "id", false,
+            1, std::nullopt);
+        if (ep.items.empty())
+          throw Problem::not_found("entrypoint not found");
+        const std::string id =
+            "sch-" + std::to_string(++ctr_) + "-" +
+            std::to_string((long long)(now_s() * 1000) % 100000);
+        const std::string kind = b.path("expression.kind").as_string();
+        const std::string value = b.path("expression.value").as_string();
+        time_t next = sched_next(kind, value, (time_t)now_s());
+        Json ov = b.at("input_overrides");
+        if (!ov.is_object()) ov = Json::object();
+        conn.insert(
+            "schedules",
+            {{"id", DbValue::S(id)},
+             {"entrypoint_id",
+              DbValue::S(b.at("entrypoint_id").as_string())},
+             {"name", DbValue::S(b.at("name").as_string())},
+             {"timezone", DbValue::S("UTC")},
+             {"expr_kind", DbValue::S(kind)},
+             {"expr_value", DbValue::S(value)},
+             {"input_overrides", DbValue::S(ov.dump())},
+             {"missed_policy",
+              DbValue::S(b.at("missed_policy").as_string("skip"))},
+             {"status", DbValue::S("active")},
+             {"next_run_at", DbValue::S(iso_of(next))},
+             {"created_at", DbValue::S(now_iso())},
+             {"updated_at", DbValue::S(now_iso())}});
+        w.respond(201, "application/json",
+                  row_schedule(load_schedule(conn, id)).dump());
+      });
+
+  reg("GET", "/serverless-runtime/v1/schedules", "list_schedules",
+      [this](HttpRequest& rq, ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        SecureConn conn(*db_, scope_for(sec, "read",
+                                        "serverless-runtime:schedules"));
+        auto page = conn.select("schedules", "", {}, "id", false, 1000,
+                                std::nullopt);
+        Json items = Json::array();
+        for (auto& r : page.items) items.push_back(row_schedule(r));
+        Json out = Json::object();
+        out["items"] = items;
+        w.respond(200, "application/json", out.dump());
+      });
+
+  reg("GET", "/serverless-runtime/v1/schedules/{id}", "get_schedule",
+      [this, load_schedule](HttpRequest& rq, ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        SecureConn conn(*db_, scope_for(sec, "read",
+                                        "serverless-runtime:schedules"));
+        w.respond(200, "application/json",
+                  row_schedule(load_schedule(conn,
+                                             rq.path_params.at("id")))
+                      .dump());
+      });
+
+  reg("PUT", "/serverless-runtime/v1/schedules/{id}", "update_schedule",
+      [this, parse_schedule, load_schedule](HttpRequest& rq,
+                                            ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        Json b = parse_schedule(rq);
+        SecureConn conn(*db_, scope_for(sec, "update",
+                                        "serverless-runtime:schedules"));
+        const std::string id = rq.path_params.at("id");
+        const std::string kind = b.path("expression.kind").as_string();
+        const std::string value = b.path("expression.value").as_string();
+        Json ov = b.at("input_overrides");
+        if (!ov.is_object()) ov = Json::object();
+        int n = conn.update(
+            "schedules",
+            {{"name", DbValue::S(b.at("name").as_string())},
+             {"expr_kind", DbValue::S(kind)},
+             {"expr_value", DbValue::S(value)},
+             {"input_overrides", DbValue::S(ov.dump())},
+             {"missed_policy",
+              DbValue::S(b.at("missed_policy").as_string("skip"))},
+             {"next_run_at",
+              DbValue::S(iso_of(sched_next(kind, value,
+                                           (time_t)now_s())))},
+             {"updated_at", DbValue::S(now_iso())}},
+            "id=?", {DbValue::S(id)});
+        if (n == 0) throw Problem::not_found("schedule not found");
+        w.respond(200, "application/json",
+                  row_schedule(load_schedule(conn, id)).dump());
+      });
+
+  auto sched_status = [this, load_schedule](HttpRequest& rq,
+                                            ResponseWriter& w,
+                                            const std::string& from_csv,
+                                            const std::string& to) {
+    auto sec = sec_of(rq);
+    SecureConn conn(*db_, scope_for(sec, "update",
+                                    "serverless-runtime:schedules"));
+    const std::string id = rq.path_params.at("id");
+    std::vector<std::pair<std::string, DbValue>> vals = {
+        {"status", DbValue::S(to)},
+        {"updated_at", DbValue::S(now_iso())}};
+    if (to == "active") {
+      // recompute from now: a paused schedule never backfills its gap
+      auto row = load_schedule(conn, id);
+      time_t next = sched_next(row.at("expr_kind").as_string(),
+                               row.at("expr_value").as_string(),
+                               (time_t)now_s());
+      vals.emplace_back("next_run_at", DbValue::S(iso_of(next)));
+    }
+    int n = conn.update("schedules", vals,
+                        "id=? AND status IN (" + from_csv + ")",
+                        {DbValue::S(id)});
+    if (n == 0) {
+      load_schedule(conn, id);   // 404 if absent
+      throw Problem{409, "Conflict", "about:blank",
+                    "illegal schedule state for this action", "conflict"};
+    }
+    w.respond(200, "application/json",
+              row_schedule(load_schedule(conn, id)).dump());
+  };
+
+  reg("POST", "/serverless-runtime/v1/schedules/{id}/pause",
+      "pause_schedule",
+      [sched_status](HttpRequest& rq, ResponseWriter& w) {
+        sched_status(rq, w, "'active'", "paused");
+      });
+
+  reg("POST", "/serverless-runtime/v1/schedules/{id}/resume",
+      "resume_schedule",
+      [sched_status](HttpRequest& rq, ResponseWriter& w) {
+        sched_status(rq, w, "'paused'", "active");
+      });
+
+  reg("DELETE", "/serverless-runtime/v1/schedules/{id}", "delete_schedule",
+      [this](HttpRequest& rq, ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        SecureConn conn(*db_, scope_for(sec, "delete",
+                                        "serverless-runtime:schedules"));
+        int n = conn.remove("schedules", "id=?",
+                            {DbValue::S(rq.path_params.at("id"))});
+        if (n == 0) throw Problem::not_found("schedule not found");
+        w.respond(204, "application/json", "");
+      });
+
+  reg("GET", "/serverless-runtime/v1/schedules/{id}/history",
+      "schedule_history",
+      [this, load_schedule](HttpRequest& rq, ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        SecureConn conn(*db_, scope_for(sec, "read",
+                                        "serverless-runtime:schedules"));
+        const std::string id = rq.path_params.at("id");
+        load_schedule(conn, id);   // 404 if absent
+        auto page = conn.select("invocations", "schedule_id=?",
+                                {DbValue::S(id)},
+                                SecureConn::OrderBy{{"created_at", true}},
+                                200, std::nullopt);
+        Json items = Json::array();
+        for (auto& r : page.items) items.push_back(row_invocation(r));
         Json out = Json::object();
         out["items"] = items;
         w.respond(200, "application/json", out.dump());

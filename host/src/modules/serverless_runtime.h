@@ -45,6 +45,13 @@ class ServerlessRuntimeModule : public Module {
   // ---- execution plane ----
   void executor_loop();
   void timer_loop();
+  // fire due schedules (cron/interval triggers, PRD BR-007/BR-022)
+  void schedule_tick();
+  std::string create_invocation(const std::string& tenant,
+                                const std::string& ep_id, const Json& input,
+                                const std::string& mode,
+                                const std::string& schedule_id,
+                                const std::string& preset_status = "");
   void run_invocation(const std::string& tenant, const std::string& id);
   // one op of a builtin/step implementation; throws on failure
   Json run_op(const std::string& op, const Json& input,
@@ -80,6 +87,7 @@ class ServerlessRuntimeModule : public Module {
   std::atomic<bool> stopping_{false};
   int n_executors_ = 4;
   std::atomic<uint64_t> ctr_{0};
+  double last_sched_scan_ = 0;   // timer-thread only
   // sync-mode waiters notified on terminal status
   std::condition_variable done_cv_;
 };

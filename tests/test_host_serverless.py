@@ -320,6 +320,165 @@ def test_llm_adapter_entrypoint(sl):
     assert j["result"]["usage"]["output_tokens"] >= 1
 
 
+# ---------------------------------------------------------------------------
+# io_schema input validation (ADR:131-185, PRD BR-032/BR-037) and
+# schedules (ADR:2038-2164 + Schedule API)
+
+def test_io_schema_input_validation(sl):
+    srv, _, _ = sl
+    ep = dict(FN)
+    ep = json.loads(json.dumps(FN))
+    ep["name"] = "fn-schema"
+    ep["io_schema"] = {"params": {
+        "type": "object",
+        "required": ["text"],
+        "properties": {"text": {"type": "string", "minLength": 2},
+                       "n": {"type": "integer", "minimum": 1,
+                             "maximum": 10}},
+        "additionalProperties": False,
+    }}
+    ep_id = _mk_active_ep(srv, ep)
+
+    def invoke(inp):
+        return _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep_id, "mode": "sync",
+                      "input": inp})
+
+    # missing required
+    st, resp = invoke({})
+    assert st == 400 and json.loads(resp)["code"] == "invalid_input", resp
+    assert "required" in json.loads(resp)["detail"]
+    # wrong type
+    st, resp = invoke({"text": 7})
+    assert st == 400 and "expected type string" in json.loads(resp)["detail"]
+    # minLength violation
+    st, resp = invoke({"text": "x"})
+    assert st == 400 and "minLength" in json.loads(resp)["detail"]
+    # non-integer / out-of-range number
+    st, resp = invoke({"text": "ok", "n": 2.5})
+    assert st == 400
+    st, resp = invoke({"text": "ok", "n": 99})
+    assert st == 400 and "maximum" in json.loads(resp)["detail"]
+    # additionalProperties: false
+    st, resp = invoke({"text": "ok", "bogus": 1})
+    assert st == 400 and "additional property" in json.loads(resp)["detail"]
+    # valid input passes and runs
+    st, resp = invoke({"text": "ok", "n": 3})
+    assert st == 200, resp
+    assert json.loads(resp)["status"] == "succeeded"
+    # dry_run validates too
+    st, resp = _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep_id, "dry_run": True,
+                      "input": {}})
+    assert st == 400
+
+    # entrypoint validation rejects malformed io_schema
+    bad = json.loads(json.dumps(FN))
+    bad["name"] = "bad-io"
+    bad["io_schema"] = {"params": "not-a-schema"}
+    st, resp = _http("POST", _url(srv) + "/entrypoints", bad)
+    assert st == 400 and "io_schema.params" in json.loads(resp)["detail"]
+
+
+def test_schedule_validation_and_crud(sl):
+    srv, _, _ = sl
+    ep_id = _mk_active_ep(srv, {**json.loads(json.dumps(FN)),
+                                "name": "fn-sched-crud"})
+    # bad expressions rejected
+    for expr in ({"kind": "cron", "value": "* * *"},
+                 {"kind": "cron", "value": "61 * * * *"},
+                 {"kind": "interval", "value": "PT0.2S"},
+                 {"kind": "interval", "value": "bogus"}):
+        st, resp = _http("POST", _url(srv) + "/schedules",
+                         {"name": "s", "entrypoint_id": ep_id,
+                          "expression": expr})
+        assert st == 400, (expr, resp)
+    # non-UTC timezone rejected (documented deviation: UTC-only)
+    st, resp = _http("POST", _url(srv) + "/schedules",
+                     {"name": "s", "entrypoint_id": ep_id,
+                      "timezone": "America/New_York",
+                      "expression": {"kind": "cron",
+                                     "value": "0 12 * * *"}})
+    assert st == 400
+    # valid cron: created with a sane next_run_at within the next minute+
+    st, resp = _http("POST", _url(srv) + "/schedules",
+                     {"name": "hourly", "entrypoint_id": ep_id,
+                      "expression": {"kind": "cron",
+                                     "value": "*/5 * * * *"},
+                      "missed_policy": "catch_up"})
+    assert st == 201, resp
+    sch = json.loads(resp)
+    assert sch["status"] == "active" and sch["next_run_at"]
+    assert sch["missed_policy"] == "catch_up"
+    sid = sch["schedule_id"]
+    st, resp = _http("GET", _url(srv) + "/schedules")
+    assert st == 200
+    assert any(s["schedule_id"] == sid
+               for s in json.loads(resp)["items"])
+    # update swaps the expression and recomputes next_run_at
+    st, resp = _http("PUT", _url(srv) + f"/schedules/{sid}",
+                     {"name": "hourly2", "entrypoint_id": ep_id,
+                      "expression": {"kind": "interval", "value": "PT1H"}})
+    assert st == 200 and json.loads(resp)["name"] == "hourly2"
+    st, _ = _http("DELETE", _url(srv) + f"/schedules/{sid}")
+    assert st == 204
+    st, _ = _http("GET", _url(srv) + f"/schedules/{sid}")
+    assert st == 404
+
+
+def test_schedule_fires_and_pause_resume(sl):
+    srv, _, _ = sl
+    ep_id = _mk_active_ep(srv, {**json.loads(json.dumps(FN)),
+                                "name": "fn-sched-fire"})
+    st, resp = _http("POST", _url(srv) + "/schedules",
+                     {"name": "ticker", "entrypoint_id": ep_id,
+                      "expression": {"kind": "interval", "value": "PT1S"},
+                      "input_overrides": {"from": "schedule"}})
+    assert st == 201, resp
+    sid = json.loads(resp)["schedule_id"]
+
+    def history():
+        st, resp = _http("GET", _url(srv) + f"/schedules/{sid}/history")
+        assert st == 200, resp
+        return json.loads(resp)["items"]
+
+    t0 = time.time()
+    while time.time() - t0 < 15 and len(history()) < 2:
+        time.sleep(0.3)
+    items = history()
+    assert len(items) >= 2, items
+    # scheduled invocations carry the merged input and actually ran
+    inv = _wait_status(srv, items[0]["id"], ("succeeded",))
+    assert inv["input"] == {"from": "schedule"}
+    # pause stops firing
+    st, resp = _http("POST", _url(srv) + f"/schedules/{sid}/pause", {})
+    assert st == 200 and json.loads(resp)["status"] == "paused"
+    time.sleep(1.2)
+    n_paused = len(history())
+    time.sleep(2.5)
+    assert len(history()) == n_paused      # no new fires while paused
+    # double-pause is a conflict
+    st, _ = _http("POST", _url(srv) + f"/schedules/{sid}/pause", {})
+    assert st == 409
+    # resume fires again from NOW (skip policy: the pause gap is not
+    # backfilled)
+    st, resp = _http("POST", _url(srv) + f"/schedules/{sid}/resume", {})
+    assert st == 200 and json.loads(resp)["status"] == "active"
+    t0 = time.time()
+    while time.time() - t0 < 10 and len(history()) <= n_paused:
+        time.sleep(0.3)
+    assert len(history()) > n_paused
+    _http("DELETE", _url(srv) + f"/schedules/{sid}")
+
+
+def test_schedule_rejects_unknown_entrypoint(sl):
+    srv, _, _ = sl
+    st, _ = _http("POST", _url(srv) + "/schedules",
+                  {"name": "s", "entrypoint_id": "ep-nope",
+                   "expression": {"kind": "interval", "value": "PT1M"}})
+    assert st == 404
+
+
 def test_durable_recovery_across_restart(sl):
     """Queued work survives a host restart (PRD.md:44-45 RTO/RPO)."""
     srv, home, cfg_path = sl
