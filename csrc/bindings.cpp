@@ -36,13 +36,13 @@ void launch_silu_mul_fp8(unsigned char*, float*, const bf16*, long, int,
                          hipStream_t);
 void launch_moe_align(int*, int*, int*, const int*, int, int, int,
                       hipStream_t);
-void launch_moe_gemm(bf16*, const bf16*, const bf16*, const int*,
-                     const int*, int, int, int, int, hipStream_t);
+void launch_moe_gemm(bf16*, float*, const bf16*, const bf16*, const int*,
+                     const int*, int, int, int, int, int, hipStream_t);
 void launch_moe_combine(bf16*, const bf16*, const float*, const int*, long,
                         int, int, hipStream_t);
-void launch_moe_gemm_fp8(bf16*, const unsigned char*, const float*,
+void launch_moe_gemm_fp8(bf16*, float*, const unsigned char*, const float*,
                          const unsigned char*, const float*, const int*,
-                         const int*, int, int, int, int, hipStream_t);
+                         const int*, int, int, int, int, int, hipStream_t);
 void launch_greedy_sample(long*, const bf16*, long, int, hipStream_t);
 void launch_inv_cdf_sample(long*, const float*, const float*, long, int,
                            hipStream_t);
@@ -255,17 +255,27 @@ void moe_align(torch::Tensor sorted_ids, torch::Tensor tile_expert,
 
 void moe_gemm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
               torch::Tensor sorted_ids, torch::Tensor tile_expert,
-              long gather_div) {
+              long gather_div, long splitk,
+              c10::optional<torch::Tensor> ws) {
   check(out, torch::kBFloat16, "out");
   check(x, torch::kBFloat16, "x");
   check(w, torch::kBFloat16, "w");
   check(sorted_ids, torch::kInt, "sorted_ids");
   check(tile_expert, torch::kInt, "tile_expert");
   TORCH_CHECK(w.dim() == 3, "w must be [E, N, K]");
-  launch_moe_gemm(bf(out), cbf(x), cbf(w), sorted_ids.data_ptr<int>(),
+  float* wsp = nullptr;
+  if (splitk > 1) {
+    TORCH_CHECK(ws.has_value(), "moe_gemm: splitk>1 needs a workspace");
+    check(*ws, torch::kFloat, "ws");
+    TORCH_CHECK(ws->numel() >= splitk * out.numel(),
+                "moe_gemm: workspace too small");
+    wsp = ws->data_ptr<float>();
+  }
+  launch_moe_gemm(bf(out), wsp, cbf(x), cbf(w),
+                  sorted_ids.data_ptr<int>(),
                   tile_expert.data_ptr<int>(), (int)tile_expert.numel(),
                   (int)w.size(1), (int)w.size(2), (int)gather_div,
-                  stream());
+                  (int)splitk, stream());
 }
 
 void moe_combine(torch::Tensor out, torch::Tensor y, torch::Tensor wts,
@@ -283,7 +293,8 @@ void moe_combine(torch::Tensor out, torch::Tensor y, torch::Tensor wts,
 void moe_gemm_fp8(torch::Tensor out, torch::Tensor xq, torch::Tensor xs,
                   torch::Tensor wq, torch::Tensor ws,
                   torch::Tensor sorted_ids, torch::Tensor tile_expert,
-                  long gather_div) {
+                  long gather_div, long splitk,
+                  c10::optional<torch::Tensor> skw) {
   check(out, torch::kBFloat16, "out");
   check(xq, torch::kFloat8_e4m3fn, "xq");
   check(xs, torch::kFloat, "xs");
@@ -292,11 +303,21 @@ void moe_gemm_fp8(torch::Tensor out, torch::Tensor xq, torch::Tensor xs,
   check(sorted_ids, torch::kInt, "sorted_ids");
   check(tile_expert, torch::kInt, "tile_expert");
   TORCH_CHECK(wq.dim() == 3, "wq must be [E, N, K]");
-  launch_moe_gemm_fp8(bf(out), u8(xq), xs.data_ptr<float>(), u8(wq),
+  float* skp = nullptr;
+  if (splitk > 1) {
+    TORCH_CHECK(skw.has_value(),
+                "moe_gemm_fp8: splitk>1 needs a workspace");
+    check(*skw, torch::kFloat, "skw");
+    TORCH_CHECK(skw->numel() >= splitk * out.numel(),
+                "moe_gemm_fp8: workspace too small");
+    skp = skw->data_ptr<float>();
+  }
+  launch_moe_gemm_fp8(bf(out), skp, u8(xq), xs.data_ptr<float>(), u8(wq),
                       ws.data_ptr<float>(), sorted_ids.data_ptr<int>(),
                       tile_expert.data_ptr<int>(),
                       (int)tile_expert.numel(), (int)wq.size(1),
-                      (int)wq.size(2), (int)gather_div, stream());
+                      (int)wq.size(2), (int)gather_div, (int)splitk,
+                      stream());
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

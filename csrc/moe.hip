@@ -128,13 +128,23 @@ DEV TileMap xcd_tile_map(int id, int ntiles, int nx, int use_xcd) {
   return {t, j};
 }
 
+// Split-K (ws != nullptr, splitk > 1): the w2 GEMM has only
+// ntiles * N/128 ~ 288 blocks at Mixtral decode shapes — barely one
+// round on 256 CUs, so the ragged second round idles ~7/8 of the chip.
+// Each of `splitk` segments owns nchunks/splitk k-chunks and writes an
+// fp32 partial tile to ws[s, P, N] (disjoint — no atomics, so no
+// cross-XCD L2 coherence traffic); moe_sk_reduce sums the segments.
 __global__ __launch_bounds__(512) void moe_gemm_v3_kernel(
-    bf16* __restrict__ out, const bf16* __restrict__ x,
+    bf16* __restrict__ out, float* __restrict__ ws,
+    const bf16* __restrict__ x,
     const bf16* __restrict__ w, const int* __restrict__ sorted_ids,
     const int* __restrict__ tile_expert, int N, int K, int gather_div,
-    int ntiles, int use_xcd) {
+    int ntiles, int use_xcd, int splitk) {
   constexpr int BN = 128;
-  const TileMap tm = xcd_tile_map(blockIdx.x, ntiles, N / BN, use_xcd);
+  const int nb_base = (int)(gridDim.x / (unsigned)splitk);
+  const int seg = (int)blockIdx.x / nb_base;    // k-segment index
+  const TileMap tm =
+      xcd_tile_map((int)blockIdx.x % nb_base, ntiles, N / BN, use_xcd);
   if (tm.t < 0) return;
   const int e = tile_expert[tm.t];
   if (e < 0) return;
@@ -238,7 +248,8 @@ __global__ __launch_bounds__(512) void moe_gemm_v3_kernel(
     }
   };
 
-  const int nchunks = K >> 6;          // K % 128 == 0 (checked at launch)
+  const int nchunks = (K >> 6) / splitk;  // chunks in THIS k-segment
+  const int c0 = seg * nchunks;           // launch checks divisibility
   // 3-slot software pipeline, one barrier per chunk:
   //   [compute d | issue d+2 | wait vmcnt(6) (d+1 landed) | barrier]
   // The issue of d+2 reuses slot (d-1)%3, whose chunk was computed by
@@ -246,14 +257,14 @@ __global__ __launch_bounds__(512) void moe_gemm_v3_kernel(
   // and one whole chunk (d+2) stays in flight across each barrier
   // (guide's 3-buf span: counted vmcnt + raw s_barrier, never
   // __syncthreads while a glds is outstanding).
-  stage(0, 0);
-  stage(1, 64);
+  stage(0, c0 << 6);
+  stage(1, (c0 + 1) << 6);
   asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   __builtin_amdgcn_s_barrier();
   int d = 0;
   for (; d + 2 < nchunks; ++d) {
     compute(d % 3);
-    stage((d + 2) % 3, (d + 2) << 6);
+    stage((d + 2) % 3, (c0 + d + 2) << 6);
     asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
     __builtin_amdgcn_s_barrier();
   }
@@ -264,6 +275,7 @@ __global__ __launch_bounds__(512) void moe_gemm_v3_kernel(
   compute(d % 3);                      // d == nchunks-1
 
   // ---- epilogue: C row = (reg&3) + 8*(reg>>2) + 4*lhalf, col = lcol ----
+  const long P = (long)ntiles * MOE_BM;
   #pragma unroll
   for (int sm = 0; sm < 2; ++sm)
     #pragma unroll
@@ -271,11 +283,40 @@ __global__ __launch_bounds__(512) void moe_gemm_v3_kernel(
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int m = (r & 3) + 8 * (r >> 2) + 4 * lhalf;
-        *(unsigned short*)(out +
-                           (long)(m0 + wave_m * 64 + sm * 32 + m) * N +
-                           n0 + wave_n * 64 + nb * 32 + lcol) =
-            f2bf(acc[sm][nb][r]);
+        const long p = m0 + wave_m * 64 + sm * 32 + m;
+        const long col = n0 + wave_n * 64 + nb * 32 + lcol;
+        if (splitk == 1)
+          *(unsigned short*)(out + p * N + col) = f2bf(acc[sm][nb][r]);
+        else
+          ws[((long)seg * P + p) * N + col] = acc[sm][nb][r];
       }
+}
+
+// sum split-K fp32 partials into bf16 (rows of unused tiles carry
+// garbage in ws AND out alike; the combine gather never reads them)
+__global__ __launch_bounds__(256) void moe_sk_reduce_kernel(
+    bf16* __restrict__ out, const float* __restrict__ ws, long PN,
+    int splitk) {
+  const long i0 = ((long)blockIdx.x * 256 + threadIdx.x) * 4;
+  if (i0 + 3 >= PN) {
+    for (long i = i0; i < PN; ++i) {
+      float a = 0.f;
+      for (int s = 0; s < splitk; ++s) a += ws[s * PN + i];
+      *(unsigned short*)(out + i) = f2bf(a);
+    }
+    return;
+  }
+  float a[4] = {0.f, 0.f, 0.f, 0.f};
+  for (int s = 0; s < splitk; ++s) {
+    const float4 v = *reinterpret_cast<const float4*>(&ws[s * PN + i0]);
+    a[0] += v.x; a[1] += v.y; a[2] += v.z; a[3] += v.w;
+  }
+  unsigned short o[4];
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) o[j] = f2bf(a[j]);
+  *reinterpret_cast<ushort4*>(
+      reinterpret_cast<unsigned short*>(out) + i0) =
+      *reinterpret_cast<ushort4*>(o);
 }
 
 // ------------------------------------------------------------- combine
@@ -313,12 +354,17 @@ void launch_moe_align(int* sorted_ids, int* tile_expert, int* inv_pos,
       ntiles_max);
 }
 
-void launch_moe_gemm(bf16* out, const bf16* x, const bf16* w,
+// splitk segments (>1 needs ws = fp32 [splitk, ntiles*MOE_BM, N]);
+// nchunks must divide evenly — the python op picks splitk accordingly
+void launch_moe_gemm(bf16* out, float* ws, const bf16* x, const bf16* w,
                      const int* sorted_ids, const int* tile_expert,
                      int ntiles_max, int N, int K, int gather_div,
-                     hipStream_t stream) {
+                     int splitk, hipStream_t stream) {
   if (K % 128 || N % 128)
     throw std::runtime_error("moe_gemm: K%128 or N%128 != 0");
+  if (splitk < 1 || (K >> 6) % splitk || ((K >> 6) / splitk) < 3 ||
+      (splitk > 1 && !ws))
+    throw std::runtime_error("moe_gemm: bad splitk");
   const int lds_bytes = 3 * (32 * 1024 + 16 * 1024);   // 144 KiB ring
   static bool attr_set = false;
   if (!attr_set) {
@@ -335,9 +381,15 @@ void launch_moe_gemm(bf16* out, const bf16* x, const bf16* w,
   const long nblocks = use_xcd
       ? 8L * 32 * ((ntiles_max + 7) / 8) * ((nx + 31) / 32)
       : (long)nx * ntiles_max;
-  moe_gemm_v3_kernel<<<dim3((unsigned)nblocks), 512, lds_bytes,
-                       stream>>>(out, x, w, sorted_ids, tile_expert, N,
-                                 K, gather_div, ntiles_max, use_xcd);
+  moe_gemm_v3_kernel<<<dim3((unsigned)(nblocks * splitk)), 512,
+                       lds_bytes, stream>>>(
+      out, ws, x, w, sorted_ids, tile_expert, N, K, gather_div,
+      ntiles_max, use_xcd, splitk);
+  if (splitk > 1) {
+    const long PN = (long)ntiles_max * MOE_BM * N;
+    moe_sk_reduce_kernel<<<dim3((unsigned)((PN / 4 + 255) / 256)), 256,
+                           0, stream>>>(out, ws, PN, splitk);
+  }
 }
 
 void launch_moe_combine(bf16* out, const bf16* y, const float* wts,
@@ -357,13 +409,17 @@ void launch_moe_combine(bf16* out, const bf16* y, const float* wts,
 typedef long long i64;
 
 __global__ __launch_bounds__(512) void moe_gemm_fp8_v3_kernel(
-    bf16* __restrict__ out, const unsigned char* __restrict__ xq,
+    bf16* __restrict__ out, float* __restrict__ skw,
+    const unsigned char* __restrict__ xq,
     const float* __restrict__ xs, const unsigned char* __restrict__ wq,
     const float* __restrict__ ws, const int* __restrict__ sorted_ids,
     const int* __restrict__ tile_expert, int N, int K, int gather_div,
-    int ntiles, int use_xcd) {
+    int ntiles, int use_xcd, int splitk) {
   constexpr int BN = 128;
-  const TileMap tm = xcd_tile_map(blockIdx.x, ntiles, N / BN, use_xcd);
+  const int nb_base = (int)(gridDim.x / (unsigned)splitk);
+  const int seg = (int)blockIdx.x / nb_base;
+  const TileMap tm =
+      xcd_tile_map((int)blockIdx.x % nb_base, ntiles, N / BN, use_xcd);
   if (tm.t < 0) return;
   const int e = tile_expert[tm.t];
   if (e < 0) return;
@@ -462,16 +518,17 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_v3_kernel(
     }
   };
 
-  const int nchunks = K >> 7;
+  const int nchunks = (K >> 7) / splitk;
+  const int c0 = seg * nchunks;
   // same race-free 3-slot pipeline as the bf16 kernel
-  stage(0, 0);
-  stage(1, 128);
+  stage(0, c0 << 7);
+  stage(1, (c0 + 1) << 7);
   asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   __builtin_amdgcn_s_barrier();
   int d = 0;
   for (; d + 2 < nchunks; ++d) {
     compute(d % 3);
-    stage((d + 2) % 3, (d + 2) << 7);
+    stage((d + 2) % 3, (c0 + d + 2) << 7);
     asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
     __builtin_amdgcn_s_barrier();
   }
@@ -482,6 +539,8 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_v3_kernel(
   compute(d % 3);
 
   // epilogue: scale by a_scale[row] * b_scale[e, col], store bf16
+  // (split-K partials are stored ALREADY scaled, so the reduce is a sum)
+  const long P = (long)ntiles * MOE_BM;
   #pragma unroll
   for (int sm = 0; sm < 2; ++sm)
     #pragma unroll
@@ -499,19 +558,26 @@ __global__ __launch_bounds__(512) void moe_gemm_fp8_v3_kernel(
         } else {
           as = xs[p];
         }
-        *(unsigned short*)(out + (long)p * N + col) =
-            f2bf(acc[sm][nb][r] * as * bs);
+        const float v = acc[sm][nb][r] * as * bs;
+        if (splitk == 1)
+          *(unsigned short*)(out + (long)p * N + col) = f2bf(v);
+        else
+          skw[((long)seg * P + p) * N + col] = v;
       }
     }
 }
 
-void launch_moe_gemm_fp8(bf16* out, const unsigned char* xq,
+void launch_moe_gemm_fp8(bf16* out, float* skw, const unsigned char* xq,
                          const float* xs, const unsigned char* wq,
                          const float* ws, const int* sorted_ids,
                          const int* tile_expert, int ntiles_max, int N,
-                         int K, int gather_div, hipStream_t stream) {
+                         int K, int gather_div, int splitk,
+                         hipStream_t stream) {
   if (K % 128 || K < 256 || N % 128)
     throw std::runtime_error("moe_gemm_fp8: bad K/N alignment");
+  if (splitk < 1 || (K >> 7) % splitk || ((K >> 7) / splitk) < 3 ||
+      (splitk > 1 && !skw))
+    throw std::runtime_error("moe_gemm_fp8: bad splitk");
   const int lds_bytes = 3 * (32 * 1024 + 16 * 1024);
   static bool attr_set = false;
   if (!attr_set) {
@@ -528,8 +594,13 @@ void launch_moe_gemm_fp8(bf16* out, const unsigned char* xq,
   const long nblocks = use_xcd
       ? 8L * 32 * ((ntiles_max + 7) / 8) * ((nx + 31) / 32)
       : (long)nx * ntiles_max;
-  moe_gemm_fp8_v3_kernel<<<dim3((unsigned)nblocks), 512, lds_bytes,
-                           stream>>>(out, xq, xs, wq, ws, sorted_ids,
-                                     tile_expert, N, K, gather_div,
-                                     ntiles_max, use_xcd);
+  moe_gemm_fp8_v3_kernel<<<dim3((unsigned)(nblocks * splitk)), 512,
+                           lds_bytes, stream>>>(
+      out, skw, xq, xs, wq, ws, sorted_ids, tile_expert, N, K,
+      gather_div, ntiles_max, use_xcd, splitk);
+  if (splitk > 1) {
+    const long PN = (long)ntiles_max * MOE_BM * N;
+    moe_sk_reduce_kernel<<<dim3((unsigned)((PN / 4 + 255) / 256)), 256,
+                           0, stream>>>(out, skw, PN, splitk);
+  }
 }

@@ -206,6 +206,27 @@ def silu_mul_q(gate_up: torch.Tensor) -> QTensor:
 MOE_BM = 256  # sorted-pair tile granularity of the MoE kernels (csrc/moe.hip)
 
 
+def _moe_splitk(nblocks: int, nchunks: int, out_numel: int, dev):
+    """Split-K degree + fp32 workspace for a grouped GEMM launch.
+
+    The w2 GEMM at Mixtral decode shapes has ~288 blocks for 256 CUs —
+    its ragged second wave idles most of the chip; splitting K into up
+    to 8 segments (each must keep >= 3 k-chunks for the LDS ring and
+    divide the chunk count) multiplies the block count back above ~4
+    per CU.  Shapes-only decision: hipGraph-capture-safe.
+    HS_MOE_SPLITK forces a degree (0 = auto)."""
+    forced = int(os.environ.get("HS_MOE_SPLITK", "0"))
+    s = 1
+    while (s < 8 and nchunks % (s * 2) == 0 and nchunks // (s * 2) >= 3
+           and (nblocks * s < 1024 if not forced else s * 2 <= forced)):
+        s *= 2
+    if forced == 1:
+        s = 1
+    if s == 1:
+        return 1, None
+    return s, torch.empty(s * out_numel, dtype=torch.float32, device=dev)
+
+
 def moe_ffn(x: torch.Tensor, w13: torch.Tensor, w2: torch.Tensor,
             weights: torch.Tensor, ids: torch.Tensor) -> torch.Tensor:
     """Grouped expert SwiGLU FFN (GPU, hipGraph-capture-safe).
@@ -231,10 +252,13 @@ def moe_ffn(x: torch.Tensor, w13: torch.Tensor, w2: torch.Tensor,
     n.moe_align(sorted_ids, tile_expert, inv_pos, flat, E)
     I2 = w13.shape[1]
     h1 = torch.empty(P, I2, dtype=torch.bfloat16, device=dev)
-    n.moe_gemm(h1, x, w13, sorted_ids, tile_expert, K)   # gather pair//K
+    s1, ws1 = _moe_splitk(ntiles * (I2 // 128), H // 64, P * I2, dev)
+    n.moe_gemm(h1, x, w13, sorted_ids, tile_expert, K, s1, ws1)
     a = silu_mul(h1)
     y = torch.empty(P, H, dtype=torch.bfloat16, device=dev)
-    n.moe_gemm(y, a, w2, sorted_ids, tile_expert, 0)     # identity rows
+    I = w2.shape[2]
+    s2, ws2 = _moe_splitk(ntiles * (H // 128), I // 64, P * H, dev)
+    n.moe_gemm(y, a, w2, sorted_ids, tile_expert, 0, s2, ws2)
     out = torch.empty(T, H, dtype=torch.bfloat16, device=dev)
     n.moe_combine(out, y, weights.float().contiguous(), inv_pos, K)
     return out
@@ -263,12 +287,15 @@ def moe_ffn_fp8(xq: "QTensor", w13q: torch.Tensor, w13s: torch.Tensor,
     n.moe_align(sorted_ids, tile_expert, inv_pos, flat, E)
     I2 = w13q.shape[1]
     h1 = torch.empty(P, I2, dtype=torch.bfloat16, device=dev)
+    s1, ws1 = _moe_splitk(ntiles * (I2 // 128), H // 128, P * I2, dev)
     n.moe_gemm_fp8(h1, xq.data, xq.scale, w13q, w13s, sorted_ids,
-                   tile_expert, K)
+                   tile_expert, K, s1, ws1)
     a = silu_mul_q(h1)
     y = torch.empty(P, H, dtype=torch.bfloat16, device=dev)
+    I = w2q.shape[2]
+    s2, ws2 = _moe_splitk(ntiles * (H // 128), I // 128, P * H, dev)
     n.moe_gemm_fp8(y, a.data, a.scale, w2q, w2s, sorted_ids, tile_expert,
-                   0)
+                   0, s2, ws2)
     out = torch.empty(T, H, dtype=torch.bfloat16, device=dev)
     n.moe_combine(out, y, weights.float().contiguous(), inv_pos, K)
     return out
