@@ -255,3 +255,70 @@ def test_moe_grouped_ffn_matches_ref():
     err = (out.float() - ref).abs().max().item()
     scale = ref.abs().max().item()
     assert err < 0.02 * scale + 0.02, (err, scale)
+
+
+@pytest.mark.gpu
+def test_moe_splitk_matches_single_segment():
+    """Split-K grouped GEMM (fp32 partials + reduce) == single-segment
+    kernel.  I=768 gives 12 k-chunks -> auto degree 4 at these tiny
+    block counts; HS_MOE_SPLITK pins each side."""
+    import os
+    import hyperspot.ops as ops
+    torch.manual_seed(1)
+    dev = "cuda:0"
+    T, H, I, E, K = 160, 768, 768, 4, 2
+    x = (torch.randn(T, H, device=dev) * 0.3).to(torch.bfloat16)
+    w13 = (torch.randn(E, 2 * I, H, device=dev) * 0.04).to(torch.bfloat16)
+    w2 = (torch.randn(E, H, I, device=dev) * 0.04).to(torch.bfloat16)
+    logits = torch.randn(T, E, device=dev)
+    w, ids = logits.softmax(-1).topk(K, dim=-1)
+    w = (w / w.sum(-1, keepdim=True)).float()
+    ids = ids.to(torch.int32)
+    try:
+        os.environ["HS_MOE_SPLITK"] = "1"
+        base = ops.moe_ffn(x, w13, w2, w, ids)
+        os.environ["HS_MOE_SPLITK"] = "4"
+        sk = ops.moe_ffn(x, w13, w2, w, ids)
+    finally:
+        del os.environ["HS_MOE_SPLITK"]
+    # identical data path except fp32-partial accumulation order
+    err = (sk.float() - base.float()).abs().max().item()
+    assert err < 0.02 * base.float().abs().max().item() + 0.02, err
+    auto = ops.moe_ffn(x, w13, w2, w, ids)   # auto picks 4 here
+    err2 = (auto.float() - sk.float()).abs().max().item()
+    assert err2 == 0.0, err2
+
+
+@pytest.mark.gpu
+def test_moe_splitk_fp8_matches_single_segment():
+    import os
+    import hyperspot.ops as ops
+    from hyperspot.parallel.layers import quantize_weight_fp8, \
+        quant_fp8_rowwise
+    torch.manual_seed(2)
+    dev = "cuda:0"
+    T, H, I, E, K = 128, 768, 768, 4, 2
+    x = (torch.randn(T, H, device=dev) * 0.3).to(torch.bfloat16)
+    w13 = [torch.randn(2 * I, H, device=dev) * 0.04 for _ in range(E)]
+    w2 = [torch.randn(H, I, device=dev) * 0.04 for _ in range(E)]
+    q13 = [quantize_weight_fp8(m) for m in w13]
+    q2 = [quantize_weight_fp8(m) for m in w2]
+    w13q = torch.stack([q for q, _ in q13])
+    w13s = torch.stack([s for _, s in q13])
+    w2q = torch.stack([q for q, _ in q2])
+    w2s = torch.stack([s for _, s in q2])
+    logits = torch.randn(T, E, device=dev)
+    w, ids = logits.softmax(-1).topk(K, dim=-1)
+    w = (w / w.sum(-1, keepdim=True)).float()
+    ids = ids.to(torch.int32)
+    xq_d, xq_s = quant_fp8_rowwise(x)
+    xq = ops.QTensor(xq_d, xq_s)
+    try:
+        os.environ["HS_MOE_SPLITK"] = "1"
+        base = ops.moe_ffn_fp8(xq, w13q, w13s, w2q, w2s, w, ids)
+        os.environ["HS_MOE_SPLITK"] = "4"
+        sk = ops.moe_ffn_fp8(xq, w13q, w13s, w2q, w2s, w, ids)
+    finally:
+        del os.environ["HS_MOE_SPLITK"]
+    err = (sk.float() - base.float()).abs().max().item()
+    assert err < 0.02 * base.float().abs().max().item() + 0.02, err
