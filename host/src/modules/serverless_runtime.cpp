@@ -123,10 +123,27 @@ Json validate_entrypoint(const Json& body) {
     if (!steps.is_array() || steps.size() == 0)
       issue("workflow_spec needs workflow.steps[]");
     else
-      for (auto& st : steps.arr())
+      for (auto& st : steps.arr()) {
         if (st.at("name").as_string().empty() ||
             st.at("op").as_string().empty())
           issue("every step needs name + op");
+        const Json& wh = st.at("when");
+        if (!wh.is_null()) {
+          if (!wh.is_object() || wh.at("field").as_string().empty())
+            issue("step.when needs {field, op, value}");
+          else {
+            const std::string wop = wh.at("op").as_string("exists");
+            if (wop != "eq" && wop != "ne" && wop != "exists" &&
+                wop != "gt" && wop != "lt")
+              issue("step.when.op must be eq|ne|exists|gt|lt");
+          }
+        }
+        if (!st.at("args").is_null() && !st.at("args").is_object())
+          issue("step.args must be an object");
+        if (!st.at("output_to").is_null() &&
+            !st.at("output_to").is_string())
+          issue("step.output_to must be a string");
+      }
     if (et != "workflow") issue("workflow_spec requires entrypoint_type "
                                 "workflow");
   } else if (kind == "adapter_ref") {
@@ -369,6 +386,22 @@ std::string schema_err(const Json& sch, const Json& v,
       return path + ": above maximum";
   }
   return "";
+}
+
+// step `when` predicate: {field, op: eq|ne|exists|gt|lt, value}
+bool when_matches(const Json& cond, const Json& state) {
+  const std::string field = cond.at("field").as_string();
+  const std::string op = cond.at("op").as_string("exists");
+  const bool has = state.contains(field);
+  if (op == "exists") return has;
+  if (!has) return false;
+  const Json& v = state.at(field);
+  const Json& want = cond.at("value");
+  if (op == "eq") return v.dump() == want.dump();
+  if (op == "ne") return v.dump() != want.dump();
+  if (op == "gt") return v.as_number() > want.as_number();
+  if (op == "lt") return v.as_number() < want.as_number();
+  return false;
 }
 
 SecurityContext sec_of(HttpRequest& rq) {
@@ -771,10 +804,34 @@ void ServerlessRuntimeModule::run_invocation(const std::string& tenant,
         }
         const Json& st = steps.at(si);
         const std::string sname = st.at("name").as_string();
+        // conditional step: `when` predicates on the current pipeline
+        // state ({field, op: eq|ne|exists|gt|lt, value})
+        if (st.at("when").is_object() &&
+            !when_matches(st.at("when"), input)) {
+          timeline(tenant, id, "step_skipped", "running", sname);
+          completed.push_back(si);
+          conn.update("invocations",
+                      {{"step_index", DbValue::I((long long)si + 1)}},
+                      "id=?", {DbValue::S(id)});
+          continue;
+        }
         timeline(tenant, id, "step_started", "running", sname);
         double t0 = now_s();
         try {
-          input = run_op(st.at("op").as_string(), input, tenant, attempts);
+          // per-step args overlay the op input; `output_to` nests the
+          // result under a key instead of replacing the pipeline state
+          Json op_in = input;
+          if (st.at("args").is_object())
+            for (auto& [k, v] : st.at("args").obj()) op_in[k] = v;
+          Json op_out = run_op(st.at("op").as_string(), op_in, tenant,
+                               attempts);
+          const std::string out_to = st.at("output_to").as_string();
+          if (out_to.empty()) {
+            input = op_out;
+          } else {
+            if (!input.is_object()) input = Json::object();
+            input[out_to] = op_out;
+          }
         } catch (const std::exception& e) {
           timeline(tenant, id, "step_failed", "running", sname,
                    (long long)((now_s() - t0) * 1000), e.what());

@@ -380,6 +380,65 @@ def test_io_schema_input_validation(sl):
     assert st == 400 and "io_schema.params" in json.loads(resp)["detail"]
 
 
+def test_workflow_when_args_output_to(sl):
+    """Workflow DSL beyond a linear op chain: conditional steps (`when`),
+    per-step arg overlays (`args`), and result routing (`output_to`)."""
+    srv, _, _ = sl
+    wf = {
+        "name": "wf-dsl", "entrypoint_type": "workflow",
+        "implementation": {
+            "adapter": "gts.x.core.serverless.adapter.builtin.v1~",
+            "kind": "workflow_spec",
+            "workflow": {"steps": [
+                # runs: uppercases text
+                {"name": "up", "op": "upper"},
+                # skipped: `mode` field is absent
+                {"name": "boom", "op": "error",
+                 "when": {"field": "mode", "op": "exists"}},
+                # runs (text == "HI"), result nested under `echoed`
+                {"name": "tag", "op": "echo",
+                 "when": {"field": "text", "op": "eq", "value": "HI"},
+                 "args": {"stamp": True}, "output_to": "echoed"},
+            ]}}}
+    ep_id = _mk_active_ep(srv, wf)
+    st, resp = _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep_id, "mode": "sync",
+                      "input": {"text": "hi"}})
+    assert st == 200, resp
+    j = json.loads(resp)
+    assert j["status"] == "succeeded", j
+    # pipeline state: upper ran, error skipped, echo nested w/ args
+    assert j["result"]["text"] == "HI"
+    assert j["result"]["echoed"] == {"text": "HI", "stamp": True}
+    tl = _timeline(srv, j["id"])
+    assert "step_skipped" in tl and tl.count("step_completed") == 2
+
+    # when.op gating on numbers: gt branch taken, lt skipped
+    wf2 = {
+        "name": "wf-branch", "entrypoint_type": "workflow",
+        "implementation": {
+            "adapter": "a~", "kind": "workflow_spec",
+            "workflow": {"steps": [
+                {"name": "big", "op": "echo", "output_to": "big",
+                 "when": {"field": "n", "op": "gt", "value": 5}},
+                {"name": "small", "op": "error",
+                 "when": {"field": "n", "op": "lt", "value": 5}},
+            ]}}}
+    ep2 = _mk_active_ep(srv, wf2)
+    st, resp = _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep2, "mode": "sync",
+                      "input": {"n": 9}})
+    j = json.loads(resp)
+    assert j["status"] == "succeeded" and "big" in j["result"]
+
+    # validation rejects malformed when/args
+    bad = json.loads(json.dumps(wf))
+    bad["name"] = "wf-bad"
+    bad["implementation"]["workflow"]["steps"][1]["when"] = {"op": "eq"}
+    st, resp = _http("POST", _url(srv) + "/entrypoints", bad)
+    assert st == 400 and "when" in json.loads(resp)["detail"]
+
+
 def test_schedule_validation_and_crud(sl):
     srv, _, _ = sl
     ep_id = _mk_active_ep(srv, {**json.loads(json.dumps(FN)),
