@@ -6,6 +6,9 @@
 #include <fstream>
 #include <sstream>
 
+#include "../http/client.h"
+#include "system_modules.h"
+
 namespace hs {
 
 namespace {
@@ -238,6 +241,7 @@ std::optional<std::pair<std::string, std::string>> parse_multipart(
 }  // namespace
 
 void FileParserModule::init(ModuleCtx& ctx) {
+  hub_ = ctx.hub;
   backends_.push_back(std::make_unique<PlainBackend>());
   backends_.push_back(std::make_unique<HtmlBackend>());
   backends_.push_back(std::make_unique<CsvBackend>());
@@ -246,6 +250,18 @@ void FileParserModule::init(ModuleCtx& ctx) {
   const Json& roots = ctx.config.at("allowed_roots");
   if (roots.is_array())
     for (auto& r : roots.arr()) allowed_roots_.push_back(r.as_string());
+  // remote_backends: [{module: "<oop module name>", extensions: [..]}]
+  // (reference pattern: file-parser as the plugin HOST — a parser can
+  // live out-of-process and be reached via the directory, like the
+  // calculator/calculator-gateway OoP pair)
+  const Json& rbs = ctx.config.at("remote_backends");
+  if (rbs.is_array())
+    for (auto& rb : rbs.arr()) {
+      const std::string mod = rb.at("module").as_string();
+      if (rb.at("extensions").is_array())
+        for (auto& e : rb.at("extensions").arr())
+          remote_ext_[e.as_string()] = mod;
+    }
 }
 
 const FileParserBackend* FileParserModule::backend_for(
@@ -259,6 +275,36 @@ void FileParserModule::parse_and_respond(const std::string& filename,
                                          const std::string& bytes,
                                          bool markdown, ResponseWriter& w) {
   const std::string ext = ext_of(filename);
+  // out-of-process backend: resolve the child via the hub
+  // DirectoryClient and forward the raw bytes (binary-safe body;
+  // filename/format travel as headers)
+  auto rit = remote_ext_.find(ext);
+  if (rit != remote_ext_.end()) {
+    auto dir = hub_ ? hub_->get<DirectoryClient>("module-orchestrator")
+                    : nullptr;
+    const std::string ep = dir ? dir->resolve(rit->second) : "";
+    if (ep.empty())
+      throw Problem{503, "Service Unavailable", "about:blank",
+                    "no live instance of parser module '" + rit->second +
+                        "'", "provider_error"};
+    std::string host;
+    int port = 80;
+    const size_t hs = ep.find("://");
+    std::string rest = hs == std::string::npos ? ep : ep.substr(hs + 3);
+    const size_t cp = rest.find(':');
+    host = rest.substr(0, cp == std::string::npos ? rest.size() : cp);
+    if (cp != std::string::npos) port = atoi(rest.c_str() + cp + 1);
+    auto r = http_request(host, port, "POST", "/parse",
+                          {{"x-filename", filename},
+                           {"x-markdown", markdown ? "1" : "0"},
+                           {"content-type", "application/octet-stream"}},
+                          bytes, 5000);
+    if (!r || r->status != 200)
+      throw Problem{502, "Bad Gateway", "about:blank",
+                    "remote parser failed", "provider_error"};
+    w.respond(200, "application/json", r->body);
+    return;
+  }
   const FileParserBackend* be = backend_for(ext);
   if (!be)
     throw Problem{415, "Unsupported Media Type", "about:blank",

@@ -35,6 +35,10 @@ class FileParserModule : public Module {
                          ResponseWriter& w);
   std::vector<std::unique_ptr<FileParserBackend>> backends_;
   std::vector<std::string> allowed_roots_;   // parse-local path allow-list
+  // extension -> OoP module name (config remote_backends[]; requests for
+  // these extensions forward to the directory-resolved child process)
+  std::map<std::string, std::string> remote_ext_;
+  ClientHub* hub_ = nullptr;
 };
 
 }  // namespace hs

@@ -1176,6 +1176,30 @@ void ModuleOrchestratorModule::init(ModuleCtx& ctx) {
       .path("modules.api-gateway.config.bind_addr")
       .as_string("127.0.0.1:8087");
   directory_endpoint_ = "http://" + bind + "/module-orchestrator/v1";
+
+  // hub DirectoryClient so in-process modules can call OoP services
+  // (the module objects outlive the hub: both owned by the host)
+  struct HubDir : DirectoryClient {
+    ModuleOrchestratorModule* mo;
+    std::string resolve(const std::string& n) override {
+      return mo->resolve_endpoint(n);
+    }
+  };
+  auto dir = std::make_shared<HubDir>();
+  dir->mo = this;
+  ctx.hub->register_client<DirectoryClient>("module-orchestrator", dir);
+}
+
+std::string ModuleOrchestratorModule::resolve_endpoint(
+    const std::string& name) {
+  const double now = mono_s();
+  std::lock_guard<std::mutex> lk(inst_mu_);
+  const Instance* best = nullptr;
+  for (auto& [id, in] : instances_)
+    if (in.name == name && (now - in.last_heartbeat) < 15.0 &&
+        (!best || in.last_heartbeat > best->last_heartbeat))
+      best = &in;
+  return best ? best->endpoint : "";
 }
 
 void ModuleOrchestratorModule::start(ModuleCtx& ctx) {

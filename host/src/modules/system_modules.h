@@ -158,6 +158,17 @@ class FileStorageModule : public Module {
 // (MODKIT_DIRECTORY_ENDPOINT), forward child stdout into the host log
 // (backends/log_forwarder.rs), track register/heartbeat liveness
 // (proto/directory/v1 semantics over REST).
+// in-process service-discovery client (reference: module-orchestrator
+// registers a DirectoryClient in the hub, src/module.rs:80-88, so other
+// modules consume OoP services without going through REST)
+class DirectoryClient {
+ public:
+  virtual ~DirectoryClient() = default;
+  // live endpoint (e.g. "http://127.0.0.1:port") for a module name,
+  // "" when no instance has heartbeated recently
+  virtual std::string resolve(const std::string& name) = 0;
+};
+
 class ModuleOrchestratorModule : public Module {
  public:
   std::string name() const override { return "module-orchestrator"; }
@@ -166,6 +177,8 @@ class ModuleOrchestratorModule : public Module {
   void start(ModuleCtx& ctx) override;
   void stop(ModuleCtx& ctx) override;
   void register_rest(ModuleCtx& ctx, RestRegistry& rest) override;
+  // freshest live instance endpoint ("" if none) — DirectoryClient impl
+  std::string resolve_endpoint(const std::string& name);
   // ModuleManager parity (reference runtime/module_manager.rs): the
   // orchestrator tracks module INSTANCES with capabilities + mounted
   // endpoints, not just names

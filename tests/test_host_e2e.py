@@ -1551,3 +1551,103 @@ modules:
     finally:
         srv.stop()
         cfg_path.unlink(missing_ok=True)
+
+
+@pytest.fixture(scope="module")
+def oop_parser_server(tmp_path_factory):
+    """Host with the OoP file-parser child: the in-process file-parser
+    module routes .rst to the directory-resolved child process (second
+    real OoP module; reference calculator/calculator-gateway flow)."""
+    import tempfile
+    port = _free_port()
+    cfg = f"""
+server:
+  home_dir: "/tmp/hs-e2e-oopfp"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+  llm-gateway:
+    config:
+      auto_start_worker: false
+  file-parser:
+    config:
+      remote_backends:
+        - module: "fileparser-oop"
+          extensions: ["rst"]
+        - module: "ghost-oop"
+          extensions: ["adoc"]
+  fileparser-oop:
+    config:
+      extensions: ["rst"]
+      greeting: "cfg-through-env"
+    runtime:
+      type: oop
+      execution:
+        executable_path: "python3"
+        args: ["tools/oop_fileparser_child.py"]
+"""
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(cfg)
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        yield srv
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
+
+
+def test_oop_fileparser_roundtrip(oop_parser_server):
+    import time as _t
+    import urllib.request
+    url = BASE.format(oop_parser_server.port)
+    rst = "Title\n=====\n\nSome ``text`` here.\n\n.. note:: dropped\n"
+    # wait for the child to register (spawn + register is async)
+    last = None
+    for _ in range(60):
+        req = urllib.request.Request(
+            url + "/file-parser/v1/upload/markdown?filename=doc.rst",
+            method="POST", data=rst.encode(),
+            headers={"content-type": "application/octet-stream"})
+        try:
+            with urllib.request.urlopen(req, timeout=10) as r:
+                last = json.loads(r.read())
+                break
+        except urllib.error.HTTPError as e:
+            last = (e.code, e.read().decode())
+            if e.code != 503:
+                break
+        _t.sleep(0.5)
+    assert isinstance(last, dict), last
+    assert last["backend"] == "oop-rst"
+    assert last["content"].startswith("# Title")
+    # child got its config through MODKIT_MODULE_CONFIG
+    assert last["greeting"] == "cfg-through-env"
+    # text (non-markdown) flavor through the same flow
+    req = urllib.request.Request(
+        url + "/file-parser/v1/upload?filename=doc.rst",
+        method="POST", data=rst.encode(),
+        headers={"content-type": "application/octet-stream"})
+    with urllib.request.urlopen(req, timeout=10) as r:
+        j = json.loads(r.read())
+    assert j["format"] == "text" and "``" not in j["content"]
+
+
+def test_oop_fileparser_no_instance_is_503(oop_parser_server):
+    import urllib.request
+    url = BASE.format(oop_parser_server.port)
+    req = urllib.request.Request(
+        url + "/file-parser/v1/upload?filename=x.adoc",
+        method="POST", data=b"== t\n",
+        headers={"content-type": "application/octet-stream"})
+    try:
+        urllib.request.urlopen(req, timeout=10)
+        assert False, "expected 503"
+    except urllib.error.HTTPError as e:
+        assert e.code == 503
+        assert json.loads(e.read())["code"] == "provider_error"
