@@ -362,7 +362,8 @@ void launch_moe_gemm(bf16* out, float* ws, const bf16* x, const bf16* w,
                      int splitk, hipStream_t stream) {
   if (K % 128 || N % 128)
     throw std::runtime_error("moe_gemm: K%128 or N%128 != 0");
-  if (splitk < 1 || (K >> 6) % splitk || ((K >> 6) / splitk) < 3 ||
+  // each k-segment needs >= 2 chunks (the 2-stage pipeline prologue)
+  if (splitk < 1 || (K >> 6) % splitk || ((K >> 6) / splitk) < 2 ||
       (splitk > 1 && !ws))
     throw std::runtime_error("moe_gemm: bad splitk");
   const int lds_bytes = 3 * (32 * 1024 + 16 * 1024);   // 144 KiB ring
@@ -575,7 +576,7 @@ void launch_moe_gemm_fp8(bf16* out, float* skw, const unsigned char* xq,
                          hipStream_t stream) {
   if (K % 128 || K < 256 || N % 128)
     throw std::runtime_error("moe_gemm_fp8: bad K/N alignment");
-  if (splitk < 1 || (K >> 7) % splitk || ((K >> 7) / splitk) < 3 ||
+  if (splitk < 1 || (K >> 7) % splitk || ((K >> 7) / splitk) < 2 ||
       (splitk > 1 && !skw))
     throw std::runtime_error("moe_gemm_fp8: bad splitk");
   const int lds_bytes = 3 * (32 * 1024 + 16 * 1024);
