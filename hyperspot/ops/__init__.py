@@ -217,6 +217,8 @@ def _moe_splitk(nblocks: int, nchunks: int, out_numel: int, dev):
     HS_MOE_SPLITK forces a degree (0 = auto)."""
     forced = int(os.environ.get("HS_MOE_SPLITK", "0"))
     s = 1
+    if forced and nblocks >= 1024:
+        forced = 0          # forcing never touches saturated launches
     while (s < 8 and nchunks % (s * 2) == 0 and nchunks // (s * 2) >= 3
            and (nblocks * s < 1024 if not forced else s * 2 <= forced)):
         s *= 2
