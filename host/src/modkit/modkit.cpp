@@ -188,6 +188,10 @@ Json RestRegistry::build_openapi(const std::string& title,
       bearer["bearerAuth"] = Json::array();
       sec.push_back(bearer);
       op["security"] = sec;
+    } else {
+      // explicit empty security = deliberately public (the auth stance
+      // is always visible in the spec, mirroring the typestate builder)
+      op["security"] = Json::array();
     }
     // path params
     Json parms = Json::array();
