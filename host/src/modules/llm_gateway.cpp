@@ -1245,8 +1245,10 @@ void LlmGatewayModule::load_jobs() {
     b.tenant = r.at("tenant_id").as_string();
     b.created_at = r.at("created_at").as_number(0);
     try {
-      for (auto& x : Json::parse(r.at("job_ids").as_string()).arr())
-        b.job_ids.push_back(x.as_string());
+      // named first: a range-for over parse(...).arr() iterates a dead
+      // temporary (caught by the ASan e2e suite, tests/test_host_asan.py)
+      Json ids = Json::parse(r.at("job_ids").as_string());
+      for (auto& x : ids.arr()) b.job_ids.push_back(x.as_string());
     } catch (...) {}
     if (b.id.rfind("batch-", 0) == 0)
       max_ctr = std::max(max_ctr,

@@ -19,7 +19,8 @@ from pathlib import Path
 import pytest
 
 ROOT = Path(__file__).resolve().parent.parent
-BIN = ROOT / "host" / "build" / "hyperspot-server"
+BIN = Path(os.environ.get("HS_SERVER_BIN",
+                          ROOT / "host" / "build" / "hyperspot-server"))
 
 
 def _free_port():
