@@ -703,8 +703,14 @@ Json LlmGatewayModule::run_chat_blocking(const Json& body,
   Json params = body.at("params");
   if (params.is_null()) params = Json::object();
   for (const char* f : {"temperature", "top_p", "top_k", "max_tokens",
-                        "seed", "ignore_eos"})
+                        "seed", "ignore_eos", "response_format"})
     if (body.contains(f)) params[f] = body.at(f);
+  // reference request schema: `response_schema` asks for structured
+  // output — the engine enforces the JSON grammar via constrained
+  // decoding (engine/guided.py); the schema body itself is advisory
+  if (body.contains("response_schema") &&
+      !params.contains("response_format"))
+    params["response_format"] = "json";
   wreq["params"] = params;
 
   // a submit can race a DYING worker (SIGKILL teardown window): before
@@ -972,8 +978,14 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   Json params = body.at("params");
   if (params.is_null()) params = Json::object();
   for (const char* f : {"temperature", "top_p", "top_k", "max_tokens",
-                        "seed", "ignore_eos"})
+                        "seed", "ignore_eos", "response_format"})
     if (body.contains(f)) params[f] = body.at(f);
+  // reference request schema: `response_schema` asks for structured
+  // output — the engine enforces the JSON grammar via constrained
+  // decoding (engine/guided.py); the schema body itself is advisory
+  if (body.contains("response_schema") &&
+      !params.contains("response_format"))
+    params["response_format"] = "json";
   wreq["params"] = params;
 
   // dying-worker race (see run_chat_blocking): obtain the FIRST engine

@@ -143,6 +143,11 @@ class SamplingParams:
     stop_token_ids: tuple = ()
     ignore_eos: bool = False
     seed: Optional[int] = None
+    # "json": constrained decoding — the sampler masks every token that
+    # would break RFC-8259 validity and only allows EOS once the
+    # top-level value closes (engine/guided.py; reference request schema
+    # field `response_schema` — the engine enforces the JSON grammar)
+    response_format: Optional[str] = None
 
     @property
     def greedy(self) -> bool:

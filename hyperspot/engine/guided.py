@@ -1,0 +1,201 @@
+"""Guided (constrained) decoding: structured JSON outputs.
+
+The reference request schema carries `response_schema`
+(llm-gateway-sdk/schemas/core/request.v1.schema.json) but the gateway is
+pass-through — enforcement is the serving engine's job.  Here the byte
+tokenizer makes constrained decoding exact and cheap: a pushdown
+automaton over RFC 8259 (minus insignificant whitespace) says which
+BYTES may come next, and the sampler masks every other token id, so a
+`response_format: "json"` request can only ever emit a prefix of valid
+JSON and can only stop (EOS) once the top-level value is closed.
+
+Engine integration: model_runner._apply_guided (mask), worker maps
+`response_schema`/`response_format` from the chat request.
+"""
+
+from __future__ import annotations
+
+from typing import List, Set, Tuple
+
+_DIGITS = set(b"0123456789")
+_HEX = set(b"0123456789abcdefABCDEF")
+# string body: any byte except '"', '\', and C0 controls; multi-byte
+# UTF-8 lead/continuation bytes are allowed (0x80-0xF4) — the model can
+# only have produced them through this mask, so sequences stay valid
+# enough for JSON transport (strict UTF-8 sequence tracking is overkill
+# for a grammar mask; json.loads accepts the decoded text)
+_STR_BODY = {b for b in range(0x20, 0x100) if b not in (0x22, 0x5C)}
+_ESCAPABLE = set(b'"\\/bfnrtu')
+_VALUE_START = set(b'"{[-tfn') | _DIGITS
+
+_LITS = {ord("t"): b"rue", ord("f"): b"alse", ord("n"): b"ull"}
+
+
+class JsonByteMachine:
+    """Incremental JSON validity automaton over bytes.
+
+    allowed() -> (set of permissible next bytes, eos_ok).
+    feed(b) advances; ValueError on a byte not currently allowed.
+    """
+
+    def __init__(self) -> None:
+        self.stack: List[str] = []     # 'arr' | 'obj'
+        self.mode = "value"
+        self.lit = b""                 # remaining literal bytes
+        self.in_key = False
+        self.hex_left = 0
+        self.consumed = 0              # tokens fed (sampler bookkeeping)
+
+    # ---- helpers ----
+
+    def _end_bytes(self) -> Set[int]:
+        """Bytes that may follow a completed value in this context."""
+        if not self.stack:
+            return set()
+        return set(b",]") if self.stack[-1] == "arr" else set(b",}")
+
+    def _num_terminable(self) -> bool:
+        return self.mode in ("num_zero", "num_int", "num_frac", "num_exp")
+
+    @property
+    def done(self) -> bool:
+        return not self.stack and (
+            self.mode == "end" or self._num_terminable())
+
+    # ---- interface ----
+
+    def allowed(self) -> Tuple[Set[int], bool]:
+        m = self.mode
+        if m == "value":
+            return set(_VALUE_START), False
+        if m == "value_first":                 # just after '[' — or close
+            return set(_VALUE_START) | {ord("]")}, False
+        if m == "key":
+            return {ord('"')}, False
+        if m == "key_first":                   # just after '{' — or close
+            return {ord('"'), ord("}")}, False
+        if m == "colon":
+            return {ord(":")}, False
+        if m == "string":
+            return set(_STR_BODY) | {0x22, 0x5C}, False
+        if m == "escape":
+            return set(_ESCAPABLE), False
+        if m == "hex":
+            return set(_HEX), False
+        if m == "lit":
+            return {self.lit[0]}, False
+        if m == "num_minus":
+            return set(_DIGITS), False
+        if m == "num_zero":
+            return {ord("."), ord("e"), ord("E")} | self._end_bytes(), \
+                not self.stack
+        if m == "num_int":
+            return set(_DIGITS) | {ord("."), ord("e"), ord("E")} \
+                | self._end_bytes(), not self.stack
+        if m == "num_dot":
+            return set(_DIGITS), False
+        if m == "num_frac":
+            return set(_DIGITS) | {ord("e"), ord("E")} \
+                | self._end_bytes(), not self.stack
+        if m == "num_esign":
+            return set(_DIGITS), False
+        if m == "num_e":
+            return set(_DIGITS) | {ord("+"), ord("-")}, False
+        if m == "num_exp":
+            return set(_DIGITS) | self._end_bytes(), not self.stack
+        if m == "end":
+            return self._end_bytes(), not self.stack
+        raise AssertionError(m)
+
+    def feed(self, b: int) -> None:
+        ok, _ = self.allowed()
+        if b not in ok:
+            raise ValueError(f"byte {b!r} not allowed in mode {self.mode}")
+        m = self.mode
+        if m in ("value", "value_first"):
+            if m == "value_first" and b == ord("]"):
+                self.stack.pop()
+                self.mode = "end"
+            elif b == ord('"'):
+                self.in_key = False
+                self.mode = "string"
+            elif b == ord("{"):
+                self.stack.append("obj")
+                self.mode = "key_first"
+            elif b == ord("["):
+                self.stack.append("arr")
+                self.mode = "value_first"
+            elif b == ord("-"):
+                self.mode = "num_minus"
+            elif b == ord("0"):
+                self.mode = "num_zero"
+            elif b in _DIGITS:
+                self.mode = "num_int"
+            else:                              # t / f / n
+                self.lit = _LITS[b]
+                self.mode = "lit"
+        elif m in ("key", "key_first"):
+            if b == ord("}"):                  # key_first only
+                self.stack.pop()
+                self.mode = "end"
+            else:
+                self.in_key = True
+                self.mode = "string"
+        elif m == "colon":
+            self.mode = "value"
+        elif m == "string":
+            if b == 0x22:
+                self.mode = "colon" if self.in_key else "end"
+            elif b == 0x5C:
+                self.mode = "escape"
+        elif m == "escape":
+            if b == ord("u"):
+                self.hex_left = 4
+                self.mode = "hex"
+            else:
+                self.mode = "string"
+        elif m == "hex":
+            self.hex_left -= 1
+            if self.hex_left == 0:
+                self.mode = "string"
+        elif m == "lit":
+            self.lit = self.lit[1:]
+            if not self.lit:
+                self.mode = "end"
+        elif m == "num_minus":
+            self.mode = "num_zero" if b == ord("0") else "num_int"
+        elif m in ("num_zero", "num_int", "num_frac", "num_exp"):
+            if b == ord("."):
+                self.mode = "num_dot"
+            elif b in (ord("e"), ord("E")) and m != "num_exp":
+                self.mode = "num_e"
+            elif b in _DIGITS and m != "num_zero":
+                pass                           # stay
+            else:                              # a terminator: close number
+                self.mode = "end"
+                self._feed_end(b)
+        elif m == "num_dot":
+            self.mode = "num_frac"
+        elif m == "num_e":
+            self.mode = "num_exp" if b in _DIGITS else "num_esign"
+        elif m == "num_esign":
+            self.mode = "num_exp"
+        elif m == "end":
+            self._feed_end(b)
+        else:
+            raise AssertionError(m)
+
+    def _feed_end(self, b: int) -> None:
+        top = self.stack[-1]
+        if b == ord(","):
+            self.mode = "key" if top == "obj" else "value"
+        else:                                  # ']' or '}'
+            self.stack.pop()
+            self.mode = "end"
+
+    # ---- token-level wrapper (ByteTokenizer: byte b <-> id b+4) ----
+
+    def feed_token(self, token_id: int) -> None:
+        self.consumed += 1
+        if 4 <= token_id < 4 + 256:
+            self.feed(token_id - 4)

@@ -387,8 +387,41 @@ class ModelRunner:
 
     # ---------------- sampling ----------------
 
+    def _apply_guided(self, logits: torch.Tensor,
+                      requests: List[Request]) -> None:
+        """Structured-output mask (SamplingParams.response_format="json"):
+        only grammar-legal next BYTES (byte tokenizer: id = byte + 4)
+        keep their logits; EOS opens up once the JSON value is closed.
+        In-place on the [B, vocab] logits; no-op without guided seqs."""
+        gis = [i for i, r in enumerate(requests)
+               if r.sampling.response_format == "json"]
+        if not gis:
+            return
+        from .guided import JsonByteMachine
+        NB = 4 + 256                       # specials + byte ids
+        neg = float("-inf")
+        small = torch.full((len(gis), NB), neg, dtype=torch.float32)
+        for k, i in enumerate(gis):
+            r = requests[i]
+            m = getattr(r, "_guided", None)
+            if m is None or m.consumed > len(r.output_token_ids):
+                m = JsonByteMachine()
+                r._guided = m
+            for t in r.output_token_ids[m.consumed:]:
+                m.feed_token(t)
+            allow, eos_ok = m.allowed()
+            for b in allow:
+                small[k, b + 4] = 0.0
+            if eos_ok:
+                small[k, 2] = 0.0          # ByteTokenizer EOS
+        d = logits.device
+        gidx = torch.tensor(gis, device=d)
+        logits[gidx, NB:] = neg
+        logits[gidx, :NB] += small.to(device=d, dtype=logits.dtype)
+
     def _sample(self, logits: torch.Tensor, requests: List[Request]) -> torch.Tensor:
         B = logits.shape[0]
+        self._apply_guided(logits, requests)
         temps = torch.tensor([r.sampling.temperature for r in requests],
                              dtype=torch.float32)
         if bool((temps == 0).all()):

@@ -472,6 +472,8 @@ def _mux_chat(msg, state: "WorkerState", mux: "MuxChannel", rids: set):
         top_k=int(params.get("top_k", 0)),
         max_tokens=min(int(params.get("max_tokens", 256)), budget),
         ignore_eos=bool(params.get("ignore_eos", False)),
+        seed=params.get("seed"),
+        response_format=params.get("response_format"),
     )
     rids.add(rid)
     state.submit_mux(rid, prompt_ids, sampling, mux)
@@ -640,6 +642,8 @@ def _run_chat(msg, state: WorkerState, send):
         top_k=int(params.get("top_k", 0)),
         max_tokens=min(int(params.get("max_tokens", 256)), budget),
         ignore_eos=bool(params.get("ignore_eos", False)),
+        seed=params.get("seed"),
+        response_format=params.get("response_format"),
     )
     q = state.submit(rid, prompt_ids, sampling)
     detok = StreamDetokenizer(state.tokenizer)

@@ -1652,3 +1652,29 @@ def test_oop_fileparser_no_instance_is_503(oop_parser_server):
     except urllib.error.HTTPError as e:
         assert e.code == 503
         assert json.loads(e.read())["code"] == "provider_error"
+
+
+def test_chat_structured_output_json(server):
+    """`response_schema`/`response_format` turns on constrained decoding:
+    the engine's grammar mask makes the reply a JSON prefix, valid JSON
+    when it finished with stop (engine/guided.py)."""
+    import json as _json
+    st, body = _http("POST", BASE.format(server.port) +
+                     "/v1/chat/completions",
+                     body={"model": "tiny-llama",
+                           "messages": [{"role": "user", "content":
+                                         [{"type": "text",
+                                           "text": "give me json"}]}],
+                           "response_schema": {"type": "object"},
+                           "max_tokens": 300, "temperature": 1.0,
+                           "seed": 11})
+    assert st == 200, body
+    j = _json.loads(body)
+    text = "".join(p["text"] for p in j["content"]
+                   if p["type"] == "text")
+    from hyperspot.engine.guided import JsonByteMachine
+    m = JsonByteMachine()
+    for b in text.encode():
+        m.feed(b)                       # grammar-legal prefix always
+    if j.get("finish_reason") == "stop":
+        _json.loads(text)
