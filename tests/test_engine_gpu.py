@@ -138,3 +138,29 @@ def test_prefill_graph_matches_eager():
         del eng
         torch.cuda.empty_cache()
     assert outs[True] == outs[False], outs
+
+
+@pytest.mark.gpu
+def test_guided_json_decode_gpu():
+    """Guided mask on DEVICE logits (advanced indexing + -inf fill on
+    the sampler path): grammar holds on hardware exactly as on CPU."""
+    import json
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    from hyperspot.engine.guided import JsonByteMachine
+    eng = LLMEngine(EngineConfig(model="tiny-llama", max_num_seqs=4,
+                                 max_num_batched_tokens=256,
+                                 max_model_len=512, num_gpu_blocks=128,
+                                 seed=0), device="cuda:0")
+    outs = eng.generate(
+        [[1, 10, 11], [1, 12, 13]],
+        SamplingParams(temperature=1.0, max_tokens=300, seed=7,
+                       response_format="json"))
+    for toks in outs:
+        m = JsonByteMachine()
+        body = [t for t in toks if t != 2]
+        for t in body:
+            assert 4 <= t < 260, toks
+            m.feed(t - 4)
+        if toks and toks[-1] == 2:
+            json.loads(bytes(t - 4 for t in body).decode(
+                "utf-8", errors="replace"))
