@@ -711,6 +711,15 @@ Json LlmGatewayModule::run_chat_blocking(const Json& body,
   if (body.contains("response_schema") &&
       !params.contains("response_format"))
     params["response_format"] = "json";
+  if (body.at("tools").is_array() && body.at("tools").size() > 0) {
+    wreq["tools"] = body.at("tools");
+    if (body.contains("tool_choice"))
+      wreq["tool_choice"] = body.at("tool_choice");
+    if (body.at("tool_choice").as_string("") == "required")
+      // forced call: the engine's ToolCallMachine grammar emits the
+      // {"name":..., "arguments":{...}} skeleton byte-exactly
+      params["response_format"] = "tool_call";
+  }
   wreq["params"] = params;
 
   // a submit can race a DYING worker (SIGKILL teardown window): before
@@ -777,11 +786,34 @@ Json LlmGatewayModule::run_chat_blocking(const Json& body,
       }
     }
     if (retry) continue;
-    Json part = Json::object();
-    part["type"] = "text";
-    part["text"] = text;
     Json content = Json::array();
-    content.push_back(part);
+    bool is_tool_call = false;
+    if (body.at("tool_choice").as_string("") == "required" &&
+        body.at("tools").is_array() && body.at("tools").size() > 0) {
+      // guided decoding forced valid JSON; shape it as a tool call when
+      // it parses as {name, arguments} (ToolCall content schema)
+      try {
+        Json tc = Json::parse(text);
+        if (tc.is_object() && !tc.at("name").as_string().empty()) {
+          Json call = Json::object();
+          call["id"] = "call-" + rid;
+          call["name"] = tc.at("name");
+          call["arguments"] = tc.at("arguments");
+          Json part = Json::object();
+          part["type"] = "tool_call";
+          part["tool_call"] = call;
+          content.push_back(part);
+          is_tool_call = true;
+          finish = "tool_calls";
+        }
+      } catch (...) {}
+    }
+    if (!is_tool_call) {
+      Json part = Json::object();
+      part["type"] = "text";
+      part["text"] = text;
+      content.push_back(part);
+    }
     Json resp = Json::object();
     resp["content"] = content;
     resp["usage"] = usage;
@@ -895,18 +927,24 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   // model resolution via model-registry (DESIGN.md:317-346); with a
   // fallback chain, resolution failures are handled per-chain-entry
   hook_pre_call(body);   // may throw request_blocked (DESIGN.md:743-766)
-  // capability gate: the engine is text-only today — non-text content
-  // parts or tool use are capability_not_supported (DESIGN error list)
-  if (body.at("tools").is_array() && body.at("tools").size() > 0)
-    throw Problem{400, "Bad Request", "about:blank",
-                  "tool use is not supported by this engine",
-                  "capability_not_supported"};
+  // tool calling is PASS-THROUGH (reference FR
+  // cpt-cf-llm-gateway-fr-tool-calling-v1 / ADR-0002: the gateway
+  // converts formats, never executes): tool defs + tool_call/tool_result
+  // parts render into the prompt by the worker; `tool_choice:
+  // "required"` additionally turns on grammar-constrained JSON decoding
+  // so the forced call IS valid JSON.  Media parts stay unsupported.
+  if (body.at("tools").is_array())
+    for (auto& t : body.at("tools").arr())
+      if (t.at("name").as_string().empty() &&
+          t.path("function.name").as_string().empty())
+        throw Problem{400, "Bad Request", "about:blank",
+                      "every tool needs a name", "validation_error"};
   for (auto& m : body.at("messages").arr()) {
     const Json& content = m.at("content");
     if (!content.is_array()) continue;
     for (auto& part : content.arr()) {
       const std::string pt = part.at("type").as_string("text");
-      if (pt != "text")
+      if (pt != "text" && pt != "tool_call" && pt != "tool_result")
         throw Problem{400, "Bad Request", "about:blank",
                       "content part type '" + pt + "' is not supported",
                       "capability_not_supported"};
@@ -940,7 +978,8 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
                        : run_chat_blocking(body, resolved, rid);
     // post_response hook: generated content can be blocked too
     if (!hook_blocklist_.empty() &&
-        hook_blocks(resp.at("content").at(0).at("text").as_string()))
+        resp.at("content").at(0).at("type").as_string("") == "text" &&
+        hook_blocks(resp.at("content").at(0).at("text").as_string("")))
       throw Problem{403, "Forbidden", "about:blank",
                     "response blocked by content policy",
                     "response_blocked"};
@@ -986,6 +1025,15 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   if (body.contains("response_schema") &&
       !params.contains("response_format"))
     params["response_format"] = "json";
+  if (body.at("tools").is_array() && body.at("tools").size() > 0) {
+    wreq["tools"] = body.at("tools");
+    if (body.contains("tool_choice"))
+      wreq["tool_choice"] = body.at("tool_choice");
+    if (body.at("tool_choice").as_string("") == "required")
+      // forced call: the engine's ToolCallMachine grammar emits the
+      // {"name":..., "arguments":{...}} skeleton byte-exactly
+      params["response_format"] = "tool_call";
+  }
   wreq["params"] = params;
 
   // dying-worker race (see run_chat_blocking): obtain the FIRST engine

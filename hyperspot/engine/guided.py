@@ -199,3 +199,62 @@ class JsonByteMachine:
         self.consumed += 1
         if 4 <= token_id < 4 + 256:
             self.feed(token_id - 4)
+
+
+class ToolCallMachine:
+    """Constrained decoding for a FORCED tool call (`tool_choice:
+    "required"`): the skeleton  {"name":"<free>","arguments":{<free>}}
+    is emitted byte-for-byte by the mask, with the model free only
+    inside the name string and the arguments object — so the output
+    always parses as the reference ToolCall shape (content/
+    tool_call.v1.schema.json: name + arguments)."""
+
+    _PRE = b'{"name":"'
+    _MID = b',"arguments":{'
+
+    def __init__(self) -> None:
+        self.m = JsonByteMachine()
+        self.queue: List[int] = list(self._PRE)
+        self.phase = 0          # 0 pre, 1 name, 2 mid, 3 args, 4 done
+        self.consumed = 0
+
+    @property
+    def done(self) -> bool:
+        return self.phase == 4
+
+    def allowed(self) -> Tuple[Set[int], bool]:
+        if self.queue:
+            return {self.queue[0]}, False
+        if self.phase == 4:
+            return set(), True
+        allow, _ = self.m.allowed()
+        return allow, False
+
+    def feed(self, b: int) -> None:
+        ok, _ = self.allowed()
+        if b not in ok:
+            raise ValueError(f"byte {b!r} not allowed in phase "
+                             f"{self.phase}")
+        if self.queue:
+            self.queue.pop(0)
+        self.m.feed(b)
+        if self.queue:
+            return
+        if self.phase == 0:
+            self.phase = 1                      # free: tool name string
+        elif self.phase == 1 and self.m.mode == "end":
+            self.phase = 2                      # name closed
+            self.queue = list(self._MID)
+        elif self.phase == 2:
+            self.phase = 3                      # free: arguments object
+        elif self.phase == 3 and len(self.m.stack) == 1 \
+                and self.m.mode == "end":
+            self.queue = [ord("}")]             # close the outer object
+            self.phase = 9                      # transient: closing
+        elif self.phase == 9:
+            self.phase = 4
+
+    def feed_token(self, token_id: int) -> None:
+        self.consumed += 1
+        if 4 <= token_id < 4 + 256:
+            self.feed(token_id - 4)

@@ -394,10 +394,10 @@ class ModelRunner:
         keep their logits; EOS opens up once the JSON value is closed.
         In-place on the [B, vocab] logits; no-op without guided seqs."""
         gis = [i for i, r in enumerate(requests)
-               if r.sampling.response_format == "json"]
+               if r.sampling.response_format in ("json", "tool_call")]
         if not gis:
             return
-        from .guided import JsonByteMachine
+        from .guided import JsonByteMachine, ToolCallMachine
         NB = 4 + 256                       # specials + byte ids
         neg = float("-inf")
         small = torch.full((len(gis), NB), neg, dtype=torch.float32)
@@ -405,7 +405,9 @@ class ModelRunner:
             r = requests[i]
             m = getattr(r, "_guided", None)
             if m is None or m.consumed > len(r.output_token_ids):
-                m = JsonByteMachine()
+                m = (ToolCallMachine()
+                     if r.sampling.response_format == "tool_call"
+                     else JsonByteMachine())
                 r._guided = m
             for t in r.output_token_ids[m.consumed:]:
                 m.feed_token(t)

@@ -82,19 +82,45 @@ class StreamDetokenizer:
         return out
 
 
-def render_chat(messages: list) -> str:
+def render_chat(messages: list, tools: list = None) -> str:
     """Minimal deterministic chat template over the GTS message shape
     (role + content[] parts — reference message.v1 schema: content is
-    ALWAYS an array of parts; a bare string is tolerated for robustness)."""
+    ALWAYS an array of parts; a bare string is tolerated for robustness).
+
+    Tool calling is pass-through (reference ADR-0002): tool definitions
+    render as a JSON block, tool_call / tool_result parts render as
+    tagged JSON inside their message — the model sees the full exchange
+    as text and the gateway shapes any forced call back into ToolCall
+    content."""
+    import json as _json
     out = []
+    if tools:
+        out.append("<|tools|>\n" +
+                   _json.dumps(tools, sort_keys=True) + "\n")
     for m in messages:
         role = m.get("role", "user")
         content = m.get("content", [])
         if isinstance(content, str):
             text = content
         else:
-            text = "".join(p.get("text", "") for p in content
-                           if isinstance(p, dict) and p.get("type", "text") == "text")
+            parts = []
+            for p in content:
+                if not isinstance(p, dict):
+                    continue
+                pt = p.get("type", "text")
+                if pt == "text":
+                    parts.append(p.get("text", ""))
+                elif pt == "tool_call":
+                    parts.append("<tool_call>" +
+                                 _json.dumps(p.get("tool_call", {}),
+                                             sort_keys=True) +
+                                 "</tool_call>")
+                elif pt == "tool_result":
+                    parts.append("<tool_result>" +
+                                 _json.dumps(p.get("tool_result", {}),
+                                             sort_keys=True) +
+                                 "</tool_result>")
+            text = "".join(parts)
         out.append(f"<|{role}|>\n{text}\n")
     out.append("<|assistant|>\n")
     return "".join(out)

@@ -459,7 +459,8 @@ def _mux_chat(msg, state: "WorkerState", mux: "MuxChannel", rids: set):
     params = msg.get("params") or {}
     prompt_ids = msg.get("prompt_ids")
     if prompt_ids is None:
-        text = render_chat(msg.get("messages") or [])
+        text = render_chat(msg.get("messages") or [],
+                           tools=msg.get("tools"))
         prompt_ids = state.tokenizer.encode(text, add_bos=True)
     budget = state.engine.config.max_model_len - len(prompt_ids) - 1
     if budget <= 0:
@@ -628,7 +629,8 @@ def _run_chat(msg, state: WorkerState, send):
     params = msg.get("params") or {}
     prompt_ids = msg.get("prompt_ids")
     if prompt_ids is None:
-        text = render_chat(msg.get("messages") or [])
+        text = render_chat(msg.get("messages") or [],
+                           tools=msg.get("tools"))
         prompt_ids = state.tokenizer.encode(text, add_bos=True)
     max_model_len = state.engine.config.max_model_len
     budget = max_model_len - len(prompt_ids) - 1

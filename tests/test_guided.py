@@ -99,7 +99,7 @@ def test_engine_guided_generation_is_json():
     eng = LLMEngine(EngineConfig(model="tiny-llama", max_num_seqs=4,
                                  max_num_batched_tokens=256,
                                  max_model_len=512, num_gpu_blocks=128,
-                                 seed=0))
+                                 seed=0), eos_token_id=2)
     outs = eng.generate(
         [[1, 10, 11], [1, 12, 13]],
         SamplingParams(temperature=1.0, max_tokens=300, seed=7,
@@ -114,3 +114,50 @@ def test_engine_guided_generation_is_json():
                                                  errors="replace")
         if toks and toks[-1] == 2:       # closed by EOS -> valid JSON
             json.loads(text)
+
+
+def test_tool_call_machine_walks():
+    """Any masked walk yields the exact ToolCall shape."""
+    from hyperspot.engine.guided import ToolCallMachine
+    for seed in range(50):
+        rng = random.Random(seed)
+        m = ToolCallMachine()
+        out = bytearray()
+        for _ in range(300):
+            allow, eos_ok = m.allowed()
+            if eos_ok:
+                break
+            pool = sorted(allow)
+            if len(out) > 40:
+                pref = [b for b in pool if b in b'"]}']
+                if pref and rng.random() < 0.8:
+                    pool = pref
+            b = rng.choice(pool)
+            m.feed(b)
+            out.append(b)
+        assert m.done, (seed, bytes(out))
+        j = json.loads(out.decode("utf-8", errors="replace"))
+        assert set(j) == {"name", "arguments"}
+        assert isinstance(j["name"], str)
+        assert isinstance(j["arguments"], dict)
+
+
+def test_engine_forced_tool_call():
+    """Forced tool call completes to the exact ToolCall shape (random
+    weights wander long names, hence the generous budget; the walk is
+    fully deterministic under the fixed seeds)."""
+    eng = LLMEngine(EngineConfig(model="tiny-llama", max_num_seqs=2,
+                                 max_num_batched_tokens=256,
+                                 max_model_len=1024, num_gpu_blocks=256,
+                                 seed=0), eos_token_id=2)
+    outs = eng.generate(
+        [[1, 10, 11]],
+        SamplingParams(temperature=1.0, max_tokens=1000, seed=3,
+                       response_format="tool_call"))
+    toks = outs[0]
+    assert toks[-1] == 2, toks          # must close the skeleton + EOS
+    text = bytes(t - 4 for t in toks if t != 2).decode(
+        "utf-8", errors="replace")
+    j = json.loads(text)
+    assert set(j) == {"name", "arguments"} and isinstance(
+        j["arguments"], dict)

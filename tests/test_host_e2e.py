@@ -1678,3 +1678,68 @@ def test_chat_structured_output_json(server):
         m.feed(b)                       # grammar-legal prefix always
     if j.get("finish_reason") == "stop":
         _json.loads(text)
+
+
+def test_chat_tool_calling_passthrough(server):
+    """Tool calling per the reference FR: tool defs + tool_call /
+    tool_result parts pass through (rendered into the prompt); a forced
+    call (`tool_choice: required`) is grammar-constrained to the
+    ToolCall shape and comes back as tool_call content."""
+    base = BASE.format(server.port) + "/v1/chat/completions"
+    tools = [{"name": "lookup", "description": "look something up",
+              "parameters": {"type": "object",
+                             "properties": {"q": {"type": "string"}}}}]
+    # forced call: try a few seeds (random weights wander; the walk is
+    # deterministic per seed, so the outcome is stable)
+    got = None
+    for seed in (3, 4, 5, 7, 11, 13):
+        st, body = _http("POST", base, body={
+            "model": "tiny-llama", "tools": tools,
+            "tool_choice": "required", "max_tokens": 900,
+            "temperature": 1.0, "seed": seed,
+            "messages": [{"role": "user", "content":
+                          [{"type": "text", "text": "call the tool"}]}]})
+        assert st == 200, body
+        j = json.loads(body)
+        if j["content"][0]["type"] == "tool_call":
+            got = j
+            break
+    assert got is not None, "no seed completed a forced tool call"
+    call = got["content"][0]["tool_call"]
+    assert call["id"].startswith("call-")
+    assert isinstance(call["name"], str)
+    assert isinstance(call["arguments"], dict)
+
+    # round 2 of the conversation: assistant tool_call + user tool_result
+    # parts are accepted and the model answers with plain text
+    st, body = _http("POST", base, body={
+        "model": "tiny-llama", "tools": tools, "max_tokens": 8,
+        "temperature": 0.0,
+        "messages": [
+            {"role": "user", "content":
+             [{"type": "text", "text": "call the tool"}]},
+            {"role": "assistant", "content":
+             [{"type": "tool_call", "tool_call": call}]},
+            {"role": "tool", "content":
+             [{"type": "tool_result",
+               "tool_result": {"tool_call_id": call["id"],
+                               "content": "42"}}]},
+        ]})
+    assert st == 200, body
+    j = json.loads(body)
+    assert j["content"][0]["type"] == "text"
+
+    # validation: a tool without a name is rejected
+    st, body = _http("POST", base, body={
+        "model": "tiny-llama", "tools": [{"description": "x"}],
+        "messages": [{"role": "user", "content":
+                      [{"type": "text", "text": "hi"}]}]})
+    assert st == 400 and json.loads(body)["code"] == "validation_error"
+
+    # media parts remain unsupported
+    st, body = _http("POST", base, body={
+        "model": "tiny-llama",
+        "messages": [{"role": "user", "content":
+                      [{"type": "image", "image": {}}]}]})
+    assert st == 400
+    assert json.loads(body)["code"] == "capability_not_supported"
