@@ -1279,13 +1279,8 @@ def test_model_approval_workflow(mt_server):
 
 def test_capability_not_supported(server):
     url = BASE.format(server.port)
-    st, body = _http("POST", url + "/v1/chat/completions",
-                     {"model": "tiny-llama",
-                      "tools": [{"name": "t"}],
-                      "messages": [{"role": "user", "content":
-                                    [{"type": "text", "text": "x"}]}]})
-    assert st == 400
-    assert json.loads(body)["code"] == "capability_not_supported"
+    # tools are pass-through now (test_chat_tool_calling_passthrough);
+    # media content parts remain capability_not_supported
     st, body = _http("POST", url + "/v1/chat/completions",
                      {"model": "tiny-llama",
                       "messages": [{"role": "user", "content":
