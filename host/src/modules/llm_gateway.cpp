@@ -708,9 +708,16 @@ Json LlmGatewayModule::run_chat_blocking(const Json& body,
   // reference request schema: `response_schema` asks for structured
   // output — the engine enforces the JSON grammar via constrained
   // decoding (engine/guided.py); the schema body itself is advisory
-  if (body.contains("response_schema") &&
-      !params.contains("response_format"))
+  if (body.at("response_schema").is_object()) {
+    // schema-SHAPED decoding: the engine forces the declared skeleton
+    // (guided.SchemaMachine); non-object schemas fall back to plain
+    // JSON-grammar enforcement
+    params["response_schema"] = body.at("response_schema");
     params["response_format"] = "json";
+  } else if (body.contains("response_schema") &&
+             !params.contains("response_format")) {
+    params["response_format"] = "json";
+  }
   if (body.at("tools").is_array() && body.at("tools").size() > 0) {
     wreq["tools"] = body.at("tools");
     if (body.contains("tool_choice"))
@@ -1022,9 +1029,16 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   // reference request schema: `response_schema` asks for structured
   // output — the engine enforces the JSON grammar via constrained
   // decoding (engine/guided.py); the schema body itself is advisory
-  if (body.contains("response_schema") &&
-      !params.contains("response_format"))
+  if (body.at("response_schema").is_object()) {
+    // schema-SHAPED decoding: the engine forces the declared skeleton
+    // (guided.SchemaMachine); non-object schemas fall back to plain
+    // JSON-grammar enforcement
+    params["response_schema"] = body.at("response_schema");
     params["response_format"] = "json";
+  } else if (body.contains("response_schema") &&
+             !params.contains("response_format")) {
+    params["response_format"] = "json";
+  }
   if (body.at("tools").is_array() && body.at("tools").size() > 0) {
     wreq["tools"] = body.at("tools");
     if (body.contains("tool_choice"))

@@ -145,9 +145,13 @@ class SamplingParams:
     seed: Optional[int] = None
     # "json": constrained decoding — the sampler masks every token that
     # would break RFC-8259 validity and only allows EOS once the
-    # top-level value closes (engine/guided.py; reference request schema
-    # field `response_schema` — the engine enforces the JSON grammar)
+    # top-level value closes; "tool_call" forces the ToolCall skeleton
+    # (engine/guided.py)
     response_format: Optional[str] = None
+    # a JSON Schema dict: SCHEMA-shaped constrained decoding — the
+    # object skeleton (known keys, declared value types) is forced
+    # byte-exactly (guided.SchemaMachine); overrides response_format
+    response_schema: Optional[dict] = None
 
     @property
     def greedy(self) -> bool:

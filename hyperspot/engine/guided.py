@@ -258,3 +258,152 @@ class ToolCallMachine:
         self.consumed += 1
         if 4 <= token_id < 4 + 256:
             self.feed(token_id - 4)
+
+
+class SchemaMachine:
+    """Schema-SHAPED constrained decoding: `response_schema` compiled to
+    a byte script — forced literal segments (the object skeleton: known
+    keys in schema order) interleaved with typed free regions whose
+    START byte is restricted to the declared type; the JSON automaton
+    then keeps the region well-formed and completion is detected by
+    stack depth.  Supported subset: object (properties + required —
+    required keys are emitted, optional ones omitted), string, number,
+    integer, boolean, null, array(items typed at the start), untyped.
+    Unsupported constructs degrade to an untyped free value, so any
+    schema still yields valid JSON.
+    """
+
+    _TYPE_START = {
+        "string": {0x22},
+        "number": set(b"-0123456789"),
+        "integer": set(b"-0123456789"),
+        "boolean": {ord("t"), ord("f")},
+        "null": {ord("n")},
+        "array": {ord("[")},
+        "object": {ord("{")},
+    }
+
+    def __init__(self, schema: dict) -> None:
+        self.m = JsonByteMachine()
+        # script: ("lit", bytes) | ("free", start_byte_set | None)
+        self.script: List[Tuple[str, object]] = []
+        self._compile(schema if isinstance(schema, dict) else {})
+        self.seg = 0          # script position
+        self.lit_pos = 0      # within a literal segment
+        self.free_depth = -1  # stack depth at free-region start (-1: n/a)
+        self.free_started = False
+        self.consumed = 0
+
+    # ---- schema -> script ----
+
+    def _compile(self, sch: dict) -> None:
+        import json as _json
+        typ = sch.get("type")
+        props = sch.get("properties")
+        if typ == "object" and isinstance(props, dict):
+            req = sch.get("required")
+            keys = [k for k in props if not isinstance(req, list)
+                    or k in req] if req is not None else list(props)
+            if not keys:
+                self.script.append(
+                    ("free", (self._TYPE_START["object"], "object")))
+                return
+            self.script.append(("lit", b"{"))
+            for i, k in enumerate(keys):
+                pre = (b"," if i else b"") + \
+                    _json.dumps(k, sort_keys=True).encode() + b":"
+                self.script.append(("lit", pre))
+                self._compile(props[k] if isinstance(props[k], dict)
+                              else {})
+            self.script.append(("lit", b"}"))
+        elif typ in self._TYPE_START:
+            self.script.append(("free", (self._TYPE_START[typ], typ)))
+        else:
+            self.script.append(("free", (None, None)))  # untyped value
+
+    # ---- runtime ----
+
+    @property
+    def done(self) -> bool:
+        return self.seg >= len(self.script)
+
+    def _free_complete(self) -> bool:
+        """Current free region holds a complete value."""
+        if not self.free_started:
+            return False
+        if len(self.m.stack) != self.free_depth:
+            return False
+        return self.m.mode == "end" or self.m._num_terminable()
+
+    def allowed(self) -> Tuple[Set[int], bool]:
+        if self.done:
+            return set(), True
+        kind, arg = self.script[self.seg]
+        if kind == "lit":
+            return {arg[self.lit_pos]}, False
+        # free region
+        start_set, typ = arg
+        inner, _ = self.m.allowed()
+        if typ == "integer" and self.m.mode in (
+                "num_zero", "num_int", "num_minus"):
+            inner -= {ord("."), ord("e"), ord("E")}
+        if not self.free_started:
+            return (inner if start_set is None
+                    else (inner & start_set)), False
+        if self._free_complete():
+            # the script owns structure now: only bytes that EXTEND the
+            # value (open-ended numbers) or the next literal's first
+            # byte are legal — the machine's generic closers are not
+            ext: Set[int] = set()
+            if self.m._num_terminable():
+                ext = inner - self.m._end_bytes()
+            nxt = self.seg + 1
+            if nxt >= len(self.script):
+                return ext, True          # top-level value may stop
+            _, narg = self.script[nxt]
+            return ext | {narg[0]}, False
+        return inner, False
+
+    def feed(self, b: int) -> None:
+        ok, _ = self.allowed()
+        if b not in ok:
+            raise ValueError(f"byte {b!r} not allowed at segment "
+                             f"{self.seg}")
+        if self.done:
+            raise ValueError("schema value already complete")
+        kind, arg = self.script[self.seg]
+        if kind == "lit":
+            self.m.feed(b)
+            self.lit_pos += 1
+            if self.lit_pos >= len(arg):
+                self.seg += 1
+                self.lit_pos = 0
+            return
+        # free region
+        if self._free_complete():
+            nxt = self.seg + 1
+            if nxt < len(self.script):
+                nkind, narg = self.script[nxt]
+                if nkind == "lit" and b == narg[0]:
+                    # terminator byte: close region AND start literal
+                    self.m.feed(b)
+                    self.seg = nxt
+                    self.lit_pos = 1
+                    self.free_started = False
+                    if self.lit_pos >= len(narg):
+                        self.seg += 1
+                        self.lit_pos = 0
+                    return
+        if not self.free_started:
+            self.free_depth = len(self.m.stack)
+            self.free_started = True
+        self.m.feed(b)
+        # a region that completes exactly at a script boundary with no
+        # literal after it (top-level value) finishes via eos_ok
+
+    def feed_token(self, token_id: int) -> None:
+        self.consumed += 1
+        if token_id == 2 and self.done:
+            return
+        if 4 <= token_id < 4 + 256:
+            self.feed(token_id - 4)

@@ -394,10 +394,12 @@ class ModelRunner:
         keep their logits; EOS opens up once the JSON value is closed.
         In-place on the [B, vocab] logits; no-op without guided seqs."""
         gis = [i for i, r in enumerate(requests)
-               if r.sampling.response_format in ("json", "tool_call")]
+               if r.sampling.response_format in ("json", "tool_call")
+               or r.sampling.response_schema is not None]
         if not gis:
             return
-        from .guided import JsonByteMachine, ToolCallMachine
+        from .guided import (JsonByteMachine, SchemaMachine,
+                             ToolCallMachine)
         NB = 4 + 256                       # specials + byte ids
         neg = float("-inf")
         small = torch.full((len(gis), NB), neg, dtype=torch.float32)
@@ -405,9 +407,12 @@ class ModelRunner:
             r = requests[i]
             m = getattr(r, "_guided", None)
             if m is None or m.consumed > len(r.output_token_ids):
-                m = (ToolCallMachine()
-                     if r.sampling.response_format == "tool_call"
-                     else JsonByteMachine())
+                if r.sampling.response_format == "tool_call":
+                    m = ToolCallMachine()
+                elif r.sampling.response_schema is not None:
+                    m = SchemaMachine(r.sampling.response_schema)
+                else:
+                    m = JsonByteMachine()
                 r._guided = m
             for t in r.output_token_ids[m.consumed:]:
                 m.feed_token(t)

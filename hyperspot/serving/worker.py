@@ -475,6 +475,8 @@ def _mux_chat(msg, state: "WorkerState", mux: "MuxChannel", rids: set):
         ignore_eos=bool(params.get("ignore_eos", False)),
         seed=params.get("seed"),
         response_format=params.get("response_format"),
+        response_schema=params.get("response_schema")
+        if isinstance(params.get("response_schema"), dict) else None,
     )
     rids.add(rid)
     state.submit_mux(rid, prompt_ids, sampling, mux)
@@ -646,6 +648,8 @@ def _run_chat(msg, state: WorkerState, send):
         ignore_eos=bool(params.get("ignore_eos", False)),
         seed=params.get("seed"),
         response_format=params.get("response_format"),
+        response_schema=params.get("response_schema")
+        if isinstance(params.get("response_schema"), dict) else None,
     )
     q = state.submit(rid, prompt_ids, sampling)
     detok = StreamDetokenizer(state.tokenizer)

@@ -161,3 +161,63 @@ def test_engine_forced_tool_call():
     j = json.loads(text)
     assert set(j) == {"name", "arguments"} and isinstance(
         j["arguments"], dict)
+
+
+def test_schema_machine_walks():
+    from hyperspot.engine.guided import SchemaMachine
+    schema = {"type": "object", "required": ["city", "pop", "ok", "meta"],
+              "properties": {"city": {"type": "string"},
+                             "pop": {"type": "integer"},
+                             "ok": {"type": "boolean"},
+                             "opt": {"type": "string"},   # omitted
+                             "meta": {"type": "object", "required": ["v"],
+                                      "properties": {
+                                          "v": {"type": "number"}}}}}
+    for seed in range(60):
+        rng = random.Random(seed)
+        m = SchemaMachine(schema)
+        out = bytearray()
+        eos = False
+        for _ in range(800):
+            allow, eos = m.allowed()
+            if eos:
+                break
+            assert allow, (seed, bytes(out))
+            pool = sorted(allow)
+            if len(out) > 30:
+                pref = [b for b in pool if b in b'"]}0123456789tf,']
+                if pref and rng.random() < 0.85:
+                    pool = pref
+            b = rng.choice(pool)
+            m.feed(b)
+            out.append(b)
+        assert eos, (seed, bytes(out))
+        j = json.loads(out.decode("utf-8", errors="replace"))
+        assert set(j) == {"city", "pop", "ok", "meta"}   # opt omitted
+        assert isinstance(j["city"], str)
+        assert isinstance(j["pop"], int) and not isinstance(j["pop"], bool)
+        assert isinstance(j["ok"], bool)
+        assert isinstance(j["meta"]["v"], (int, float))
+
+
+def test_engine_schema_shaped_output():
+    """Skeleton-forced schema: completion is fast and exact for any
+    seed (only the scalar values are free)."""
+    eng = LLMEngine(EngineConfig(model="tiny-llama", max_num_seqs=2,
+                                 max_num_batched_tokens=256,
+                                 max_model_len=512, num_gpu_blocks=128,
+                                 seed=0), eos_token_id=2)
+    schema = {"type": "object", "required": ["n", "ok"],
+              "properties": {"n": {"type": "integer"},
+                             "ok": {"type": "boolean"}}}
+    for seed in (1, 2, 3):
+        outs = eng.generate(
+            [[1, 10, 11]],
+            SamplingParams(temperature=1.0, max_tokens=200, seed=seed,
+                           response_schema=schema))
+        toks = outs[0]
+        assert toks[-1] == 2, (seed, toks)
+        j = json.loads(bytes(t - 4 for t in toks if t != 2).decode(
+            "utf-8", errors="replace"))
+        assert set(j) == {"n", "ok"}
+        assert isinstance(j["n"], int) and isinstance(j["ok"], bool)

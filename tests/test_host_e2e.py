@@ -1738,3 +1738,31 @@ def test_chat_tool_calling_passthrough(server):
                       [{"type": "image", "image": {}}]}]})
     assert st == 400
     assert json.loads(body)["code"] == "capability_not_supported"
+
+
+def test_chat_schema_shaped_output(server):
+    """response_schema with a concrete object schema: the reply IS that
+    shape — keys, order and value types forced by the engine's
+    SchemaMachine, scalar values free."""
+    st, body = _http("POST", BASE.format(server.port) +
+                     "/v1/chat/completions",
+                     body={"model": "tiny-llama",
+                           "messages": [{"role": "user", "content":
+                                         [{"type": "text",
+                                           "text": "report status"}]}],
+                           "response_schema": {
+                               "type": "object",
+                               "required": ["count", "healthy"],
+                               "properties": {
+                                   "count": {"type": "integer"},
+                                   "healthy": {"type": "boolean"}}},
+                           "max_tokens": 200, "temperature": 1.0,
+                           "seed": 5})
+    assert st == 200, body
+    j = json.loads(body)
+    text = "".join(p["text"] for p in j["content"]
+                   if p["type"] == "text")
+    out = json.loads(text)
+    assert set(out) == {"count", "healthy"}
+    assert isinstance(out["count"], int)
+    assert isinstance(out["healthy"], bool)
