@@ -251,3 +251,32 @@ class HyperspotClient:
 
     def health(self) -> Dict[str, Any]:
         return self.request("GET", "/health")
+
+    # ---- structured outputs ----
+
+    def chat_json(self, messages: List[Dict[str, Any]],
+                  schema: Optional[Dict[str, Any]] = None,
+                  model: Optional[str] = None, **params: Any) -> Any:
+        """Constrained decoding: returns the PARSED JSON reply.  With a
+        schema, the engine forces that object shape (keys in schema
+        order, typed values); without one, any valid JSON value."""
+        body_schema = schema if schema is not None else {}
+        r = self.chat(messages, model, response_schema=body_schema,
+                      **params)
+        text = "".join(p.get("text", "") for p in r.get("content", [])
+                       if p.get("type") == "text")
+        return json.loads(text)
+
+    def chat_tool_call(self, messages: List[Dict[str, Any]],
+                       tools: List[Dict[str, Any]],
+                       model: Optional[str] = None,
+                       **params: Any) -> Dict[str, Any]:
+        """Force a tool call; returns the ToolCall dict
+        (id/name/arguments)."""
+        r = self.chat(messages, model, tools=tools,
+                      tool_choice="required", **params)
+        for p in r.get("content", []):
+            if p.get("type") == "tool_call":
+                return p["tool_call"]
+        raise ProblemError(502, {"title": "no tool call produced",
+                                 "code": "provider_error"})

@@ -34,6 +34,8 @@ def test_e2e_suites_under_asan():
          "tests/test_host_e2e.py::test_chat_completion_sync",
          "tests/test_host_e2e.py::test_chat_completion_sse_stream",
          "tests/test_host_e2e.py::test_settings_crud_and_tenant_isolation",
+         "tests/test_host_e2e.py::test_chat_tool_calling_passthrough",
+         "tests/test_host_e2e.py::test_chat_schema_shaped_output",
          ],
         cwd=ROOT, env=env, capture_output=True, text=True, timeout=800)
     assert out.returncode == 0, out.stdout[-3000:] + out.stderr[-2000:]

@@ -92,3 +92,32 @@ def test_sdk_invoke_serverless(mt_server):  # noqa: F811
               {"action": "activate"})
     inv = c.invoke(ep["id"], {"k": 1})
     assert inv["status"] == "succeeded" and inv["result"] == {"k": 1}
+
+
+def test_sdk_chat_json_schema(server):  # noqa: F811
+    c = HyperspotClient(BASE.format(server.port))
+    out = c.chat_json(
+        [{"role": "user", "content": "status"}],
+        schema={"type": "object", "required": ["n", "ok"],
+                "properties": {"n": {"type": "integer"},
+                               "ok": {"type": "boolean"}}},
+        max_tokens=200, temperature=1.0, seed=5)
+    assert set(out) == {"n", "ok"}
+    assert isinstance(out["n"], int) and isinstance(out["ok"], bool)
+
+
+def test_sdk_chat_tool_call(server):  # noqa: F811
+    c = HyperspotClient(BASE.format(server.port))
+    tools = [{"name": "lookup", "parameters": {"type": "object"}}]
+    for seed in (3, 4, 5, 7, 11):
+        try:
+            call = c.chat_tool_call(
+                [{"role": "user", "content": "use the tool"}], tools,
+                max_tokens=900, temperature=1.0, seed=seed)
+            break
+        except ProblemError:
+            continue
+    else:
+        raise AssertionError("no seed produced a tool call")
+    assert isinstance(call["name"], str)
+    assert isinstance(call["arguments"], dict)
