@@ -316,6 +316,14 @@ class SchemaMachine:
                 self._compile(props[k] if isinstance(props[k], dict)
                               else {})
             self.script.append(("lit", b"}"))
+        elif typ == "array":
+            items = sch.get("items")
+            ist = None
+            if isinstance(items, dict) and \
+                    items.get("type") in self._TYPE_START:
+                ist = self._TYPE_START[items["type"]]
+            self.script.append(
+                ("free", (self._TYPE_START["array"], "array", ist)))
         elif typ in self._TYPE_START:
             self.script.append(("free", (self._TYPE_START[typ], typ)))
         else:
@@ -342,11 +350,19 @@ class SchemaMachine:
         if kind == "lit":
             return {arg[self.lit_pos]}, False
         # free region
-        start_set, typ = arg
+        start_set, typ = arg[0], arg[1]
         inner, _ = self.m.allowed()
         if typ == "integer" and self.m.mode in (
                 "num_zero", "num_int", "num_minus"):
             inner -= {ord("."), ord("e"), ord("E")}
+        if typ == "array" and len(arg) > 2 and arg[2] is not None \
+                and self.free_started \
+                and len(self.m.stack) == self.free_depth + 1 \
+                and self.m.mode in ("value", "value_first"):
+            # direct elements of the typed array: restrict their START
+            # to the declared item type (nested values stay free)
+            keep = {ord("]")} if self.m.mode == "value_first" else set()
+            inner = (inner & arg[2]) | (inner & keep)
         if not self.free_started:
             return (inner if start_set is None
                     else (inner & start_set)), False

@@ -221,3 +221,33 @@ def test_engine_schema_shaped_output():
             "utf-8", errors="replace"))
         assert set(j) == {"n", "ok"}
         assert isinstance(j["n"], int) and isinstance(j["ok"], bool)
+
+
+def test_schema_typed_array_items():
+    """Array `items` typing restricts each direct element's START byte
+    (numeric elements here; nested structure remains free)."""
+    from hyperspot.engine.guided import SchemaMachine
+    schema = {"type": "object", "required": ["xs"],
+              "properties": {"xs": {"type": "array",
+                                    "items": {"type": "number"}}}}
+    for seed in range(40):
+        rng = random.Random(seed)
+        m = SchemaMachine(schema)
+        out = bytearray()
+        eos = False
+        for _ in range(400):
+            allow, eos = m.allowed()
+            if eos:
+                break
+            assert allow, (seed, bytes(out))
+            pool = sorted(allow)
+            if len(out) > 20:
+                pref = [b for b in pool if b in b']}0123456789']
+                if pref and rng.random() < 0.85:
+                    pool = pref
+            b = rng.choice(pool)
+            m.feed(b)
+            out.append(b)
+        assert eos, (seed, bytes(out))
+        j = json.loads(out.decode("utf-8", errors="replace"))
+        assert all(isinstance(x, (int, float)) for x in j["xs"])

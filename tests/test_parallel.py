@@ -313,3 +313,44 @@ def test_ep2_checkpoint_scatter_gather(tmp_path):
         for k in ("w13", "w2", "router"):
             assert torch.equal(f.get_tensor(k),
                                dict(ref.named_parameters())[k].data), k
+
+
+def _run_tp_guided(rank, world, port, results):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    import torch.distributed as dist
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from hyperspot.engine import EngineConfig, LLMEngine, SamplingParams
+    from hyperspot.parallel.state import (initialize_model_parallel,
+                                          destroy_model_parallel)
+    initialize_model_parallel(tp_size=world)
+    cfg = EngineConfig(model="tiny-llama", max_num_seqs=4,
+                       max_num_batched_tokens=256, max_model_len=512,
+                       num_gpu_blocks=128, tp_size=world)
+    eng = LLMEngine(cfg, eos_token_id=2)
+    schema = {"type": "object", "required": ["n", "ok"],
+              "properties": {"n": {"type": "integer"},
+                             "ok": {"type": "boolean"}}}
+    out = eng.generate([[1, 2, 3]],
+                       SamplingParams(temperature=1.0, max_tokens=200,
+                                      seed=4, response_schema=schema))
+    results[rank] = out
+    dist.barrier()
+    destroy_model_parallel()
+    dist.destroy_process_group()
+
+
+def test_tp2_guided_schema_identical_and_valid():
+    """Guided masks are pure functions of the (identical) request state,
+    so TP ranks stay in lockstep; the result obeys the schema."""
+    import json
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_run_tp_guided, args=(2, 29627, results), nprocs=2,
+             join=True)
+    assert results[0] == results[1]
+    toks = results[0][0]
+    assert toks[-1] == 2, toks
+    j = json.loads(bytes(t - 4 for t in toks if t != 2).decode(
+        "utf-8", errors="replace"))
+    assert set(j) == {"n", "ok"} and isinstance(j["n"], int)
