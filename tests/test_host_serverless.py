@@ -538,6 +538,64 @@ def test_schedule_rejects_unknown_entrypoint(sl):
     assert st == 404
 
 
+def test_schedule_survives_restart_catch_up(tmp_path):
+    """Schedules are durable rows: after a host restart the ticker
+    resumes them, and a catch_up missed-policy fires ONCE for the gap
+    (BR-022), then continues on cadence."""
+    import tempfile
+    home = tempfile.mkdtemp(prefix="hs-schedres-")
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-sr-")
+    port = _free_port()
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(_mk_cfg(port, sock, home))
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        ep_id = _mk_active_ep(srv, {**json.loads(json.dumps(FN)),
+                                    "name": "fn-sr"})
+        st, resp = _http("POST", _url(srv) + "/schedules",
+                         {"name": "sr", "entrypoint_id": ep_id,
+                          "expression": {"kind": "interval",
+                                         "value": "PT2S"},
+                          "missed_policy": "catch_up"})
+        assert st == 201, resp
+        sid = json.loads(resp)["schedule_id"]
+    finally:
+        srv.stop()
+    time.sleep(5)       # miss >= 2 slots while the host is down
+    port2 = _free_port()
+    sock2 = tempfile.mktemp(suffix=".sock", prefix="hs-sr2-")
+    cfg2 = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg2.write_text(_mk_cfg(port2, sock2, home))
+    srv2 = ServerProc(cfg2, port2)
+    try:
+        srv2.wait_ready()
+        st, resp = _http("GET", _url(srv2) + f"/schedules/{sid}")
+        assert st == 200, resp          # survived the restart
+
+        def history():
+            st, resp = _http("GET",
+                             _url(srv2) + f"/schedules/{sid}/history")
+            assert st == 200, resp
+            return json.loads(resp)["items"]
+
+        # catch_up: exactly ONE invocation for the whole missed gap...
+        t0 = time.time()
+        while time.time() - t0 < 10 and not history():
+            time.sleep(0.2)
+        n_catchup = len(history())
+        assert n_catchup == 1, history()
+        # ...and the cadence resumes afterwards
+        t0 = time.time()
+        while time.time() - t0 < 10 and len(history()) < 2:
+            time.sleep(0.3)
+        assert len(history()) >= 2
+    finally:
+        srv2.stop()
+        cfg2.unlink(missing_ok=True)
+    cfg_path.unlink(missing_ok=True)
+
+
 def test_durable_recovery_across_restart(sl):
     """Queued work survives a host restart (PRD.md:44-45 RTO/RPO)."""
     srv, home, cfg_path = sl
