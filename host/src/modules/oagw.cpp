@@ -61,6 +61,50 @@ Json* OagwModule::find_upstream(const std::string& tenant,
   return nullptr;
 }
 
+static double mono_s_oagw() {
+  return std::chrono::duration<double>(
+             std::chrono::steady_clock::now().time_since_epoch())
+      .count();
+}
+
+bool OagwModule::breaker_admit(const std::string& key, int threshold,
+                               double open_s, bool* probe) {
+  *probe = false;
+  if (threshold <= 0) return true;          // breaker disabled
+  std::lock_guard<std::mutex> lk(mu_);
+  Breaker& b = breakers_[key];
+  if (!b.open) return true;
+  const double now = mono_s_oagw();
+  if (now - b.opened_at < open_s) return false;       // still cooling
+  if (b.probing) return false;              // one trial at a time
+  b.probing = true;                         // half-open: this call probes
+  *probe = true;
+  return true;
+}
+
+void OagwModule::breaker_report(const std::string& key, bool ok) {
+  std::lock_guard<std::mutex> lk(mu_);
+  auto it = breakers_.find(key);
+  if (it == breakers_.end()) {
+    if (!ok) breakers_[key].fails = 1;
+    return;
+  }
+  Breaker& b = it->second;
+  if (ok) {
+    b.fails = 0;
+    b.open = false;
+    b.probing = false;
+    return;
+  }
+  b.probing = false;
+  if (b.open) {                             // failed half-open probe
+    b.opened_at = mono_s_oagw();
+    return;
+  }
+  b.fails++;
+  // threshold check happens at the call site (config-dependent)
+}
+
 void OagwModule::proxy(HttpRequest& req, ResponseWriter& w) {
   SecurityContext sec = sec_of(req);
   std::string alias = req.path_params["alias"];
@@ -178,6 +222,34 @@ void OagwModule::proxy(HttpRequest& req, ResponseWriter& w) {
   if (req.target.find('?') != std::string::npos)
     target += req.target.substr(req.target.find('?'));
 
+  // circuit breaker (ADR-0004 infrastructure layer): consecutive
+  // connect/5xx failures open the endpoint; after `open_ms` ONE
+  // half-open probe decides recovery.  Config per upstream:
+  //   circuit: {failure_threshold: 5, open_ms: 30000}   (0 disables)
+  const int cb_threshold =
+      (int)up.path("circuit.failure_threshold").as_int(5);
+  const double cb_open_s =
+      up.path("circuit.open_ms").as_number(30000) / 1000.0;
+  const std::string cb_key =
+      alias + "|" + host + ":" + std::to_string(port);
+  bool cb_probe = false;
+  if (!breaker_admit(cb_key, cb_threshold, cb_open_s, &cb_probe))
+    throw Problem{503, "Service Unavailable", "about:blank",
+                  "upstream circuit is open (cooling down)",
+                  "circuit_open"};
+  auto cb_result = [&](bool ok) {
+    if (cb_threshold <= 0) return;
+    breaker_report(cb_key, ok);
+    if (!ok && !cb_probe) {
+      std::lock_guard<std::mutex> lk(mu_);
+      Breaker& b = breakers_[cb_key];
+      if (!b.open && b.fails >= cb_threshold) {
+        b.open = true;
+        b.opened_at = mono_s_oagw();
+      }
+    }
+  };
+
   // SSE pass-through (reference oagw: no total timeout so SSE can
   // stream, src/infra/proxy/service.rs:43-49): when the client asks for
   // an event stream, chunks are forwarded as they arrive
@@ -201,12 +273,16 @@ void OagwModule::proxy(HttpRequest& req, ResponseWriter& w) {
         },
         &tls);
     if (started) {
+      cb_result(true);
       w.end_stream();
       return;
     }
-    if (!r)
+    if (!r) {
+      cb_result(false);
       throw Problem{502, "Bad Gateway", "about:blank",
                     "upstream connect failed", "provider_error"};
+    }
+    cb_result(r->status < 500);
     // empty-bodied response: fall through with the buffered result
     w.respond(r->status, upct, "", {{"x-oagw-upstream", alias}});
     return;
@@ -215,9 +291,12 @@ void OagwModule::proxy(HttpRequest& req, ResponseWriter& w) {
   // forward (buffered)
   auto resp = http_request(host, port, req.method, target, fwd, req.body,
                            10000, nullptr, nullptr, &tls);
-  if (!resp)
+  if (!resp) {
+    cb_result(false);
     throw Problem{502, "Bad Gateway", "about:blank",
                   "upstream connect failed", "provider_error"};
+  }
+  cb_result(resp->status < 500);
   std::string ct = resp->headers.count("content-type")
       ? resp->headers["content-type"] : "application/octet-stream";
   std::vector<std::pair<std::string, std::string>> hdrs;

@@ -33,6 +33,22 @@ class OagwModule : public Module {
   std::map<std::string, Json> routes_;      // id -> route
   std::map<std::string, std::unique_ptr<TokenBucket>> limiters_;
   uint64_t next_id_ = 1;
+
+  // per-endpoint circuit breaker (ADR-0004 infrastructure layer:
+  // reactive fast-fail with automatic half-open recovery; the
+  // llm-gateway's worker watchdog/ready-lease is the business layer)
+  struct Breaker {
+    int fails = 0;              // consecutive failures while closed
+    bool open = false;
+    double opened_at = 0;       // monotonic seconds
+    bool probing = false;       // one half-open trial in flight
+  };
+  std::map<std::string, Breaker> breakers_;   // key: alias|host:port
+  // returns false => fast-fail (circuit open, not yet cool);
+  // *probe set when this call is the single half-open trial
+  bool breaker_admit(const std::string& key, int threshold, double open_s,
+                     bool* probe);
+  void breaker_report(const std::string& key, bool ok);
 };
 
 }  // namespace hs

@@ -296,3 +296,75 @@ def test_proxy_https_verify_disabled(server, tls_upstream):
     assert st == 201, body
     st, body = _http("GET", base + "/oagw/v1/proxy/tlsnv/ping")
     assert st == 200, body
+
+
+# ---------------------------------------------------------------------------
+# circuit breaker (llm-gateway ADR-0004 infrastructure layer, implemented
+# at the OAGW data plane: consecutive connect/5xx failures open the
+# endpoint, fast-fail 503 circuit_open while cooling, one half-open probe
+# recovers)
+
+class FlakyUpstream(http.server.BaseHTTPRequestHandler):
+    fail_remaining = 0          # class-level knob
+
+    def _respond(self):
+        cls = type(self)
+        if cls.fail_remaining > 0:
+            cls.fail_remaining -= 1
+            self.send_response(500)
+            self.send_header("content-length", "0")
+            self.end_headers()
+            return
+        out = b'{"ok": true}'
+        self.send_response(200)
+        self.send_header("content-type", "application/json")
+        self.send_header("content-length", str(len(out)))
+        self.end_headers()
+        self.wfile.write(out)
+
+    do_GET = _respond
+    do_POST = _respond
+
+    def log_message(self, *a):
+        pass
+
+
+def test_circuit_breaker_opens_and_recovers(server):
+    import time
+    srv = http.server.ThreadingHTTPServer(("127.0.0.1", 0), FlakyUpstream)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        base = BASE.format(server.port)
+        up = {"alias": "flaky", "server": {"endpoints": [
+            {"scheme": "http", "host": "127.0.0.1",
+             "port": srv.server_address[1]}]},
+            "circuit": {"failure_threshold": 3, "open_ms": 700}}
+        st, body = _http("POST", base + "/oagw/v1/upstreams", body=up)
+        assert st == 201, body
+
+        FlakyUpstream.fail_remaining = 100
+        # three real failures surface as 502/5xx pass-through...
+        for _ in range(3):
+            st, body = _http("GET", base + "/oagw/v1/proxy/flaky/x")
+            assert st == 500, (st, body)
+        # ...then the circuit opens: fast-fail WITHOUT touching upstream
+        before = FlakyUpstream.fail_remaining
+        st, body = _http("GET", base + "/oagw/v1/proxy/flaky/x")
+        assert st == 503 and json.loads(body)["code"] == "circuit_open"
+        assert FlakyUpstream.fail_remaining == before   # not forwarded
+        # a failed half-open probe re-opens
+        time.sleep(0.8)
+        st, _ = _http("GET", base + "/oagw/v1/proxy/flaky/x")
+        assert st == 500                    # the probe hit the upstream
+        st, body = _http("GET", base + "/oagw/v1/proxy/flaky/x")
+        assert st == 503 and json.loads(body)["code"] == "circuit_open"
+        # recovery: upstream healthy again -> probe closes the circuit
+        FlakyUpstream.fail_remaining = 0
+        time.sleep(0.8)
+        st, body = _http("GET", base + "/oagw/v1/proxy/flaky/x")
+        assert st == 200 and json.loads(body)["ok"] is True
+        st, body = _http("GET", base + "/oagw/v1/proxy/flaky/x")
+        assert st == 200                    # closed: normal traffic
+    finally:
+        srv.shutdown()
