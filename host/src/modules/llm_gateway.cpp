@@ -368,6 +368,13 @@ void LlmGatewayModule::init(ModuleCtx& ctx) {
   ctx.hub->register_client<ChatInvoker>(
       "llm-gateway",
       std::make_shared<GatewayChatInvoker>(this));
+  struct HubHealth : ProviderHealthClient {
+    LlmGatewayModule* m;
+    Json provider_health() override { return m->provider_health(); }
+  };
+  auto hh = std::make_shared<HubHealth>();
+  hh->m = this;
+  ctx.hub->register_client<ProviderHealthClient>("llm-gateway", hh);
   {
     std::string file = ctx.full_config
                            .path("modules.llm-gateway.database.file")
@@ -485,6 +492,39 @@ LlmGatewayModule::Worker* LlmGatewayModule::pick_worker() {
   }
   if (best) best->in_flight++;
   return best;
+}
+
+Json LlmGatewayModule::provider_health() {
+  // status rules per reference PRD:289-294: unhealthy at 3+ consecutive
+  // failures, healthy at 2+ consecutive successes, degraded between
+  Json items = Json::array();
+  for (auto& w : workers_) {
+    std::lock_guard<std::mutex> hl(w->health_mu);
+    Json it = Json::object();
+    it["provider_id"] = "local::worker-" + std::to_string(w->index);
+    it["device"] = (long)w->device;
+    std::string status = "degraded";
+    if (w->consec_fail >= 3) status = "unhealthy";
+    else if (w->consec_ok >= 2) status = "healthy";
+    it["status"] = status;
+    Json m = Json::object();
+    if (!w->probe_ms.empty()) {
+      std::vector<double> v(w->probe_ms.begin(), w->probe_ms.end());
+      std::sort(v.begin(), v.end());
+      m["latency_p50_ms"] = v[v.size() / 2];
+      m["latency_p99_ms"] = v[(v.size() * 99) / 100];
+    }
+    m["consecutive_failures"] = (long)w->consec_fail;
+    m["consecutive_successes"] = (long)w->consec_ok;
+    it["metrics"] = m;
+    if (w->last_check > 0) it["last_check"] = w->last_check;
+    if (w->last_success > 0) it["last_success"] = w->last_success;
+    if (!w->last_error.empty()) it["last_error_message"] = w->last_error;
+    items.push_back(it);
+  }
+  Json out = Json::object();
+  out["items"] = items;
+  return out;
 }
 
 std::shared_ptr<MuxClient> LlmGatewayModule::ensure_mux(Worker& wk) {
@@ -606,6 +646,42 @@ void LlmGatewayModule::start(ModuleCtx& ctx) {
             m_worker_restarts_++;
             unlink(w->socket.c_str());
             spawn_one(*w);
+          }
+          // discovery health probe: one info round-trip per cycle
+          // (ProviderHealth metrics — latency ring + consec counters)
+          const auto t0 = std::chrono::steady_clock::now();
+          bool ok = false;
+          std::string err;
+          {
+            EngineConn c(w->socket);
+            if (!c.ok()) {
+              err = "connect failed";
+            } else {
+              Json q = Json::object();
+              q["type"] = "info";
+              if (!c.send_json(q)) {
+                err = "send failed";
+              } else {
+                auto r = c.read_json(3000);
+                if (r && r->at("ready").as_bool()) ok = true;
+                else err = r ? "not ready" : "info timeout";
+              }
+            }
+          }
+          const double ms = std::chrono::duration<double, std::milli>(
+              std::chrono::steady_clock::now() - t0).count();
+          std::lock_guard<std::mutex> hl(w->health_mu);
+          w->last_check = (double)time(nullptr);
+          if (ok) {
+            w->consec_ok++;
+            w->consec_fail = 0;
+            w->last_success = w->last_check;
+            w->probe_ms.push_back(ms);
+            if (w->probe_ms.size() > 64) w->probe_ms.pop_front();
+          } else {
+            w->consec_fail++;
+            w->consec_ok = 0;
+            w->last_error = err;
           }
         }
       }

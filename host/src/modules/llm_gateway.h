@@ -36,6 +36,14 @@ struct AdmissionClient {
 // blocking chat invoker registered by llm-gateway for in-process
 // consumers (serverless-runtime adapter_ref entrypoints — the spec's
 // "entrypoints are model workers" reading)
+// discovery-level provider health (reference model-registry PRD:280-294
+// ProviderHealth): llm-gateway registers this; model-registry serves it
+// at /model-registry/v1/providers/health
+struct ProviderHealthClient {
+  virtual ~ProviderHealthClient() = default;
+  virtual Json provider_health() = 0;
+};
+
 struct ChatInvoker {
   virtual ~ChatInvoker() = default;
   virtual Json chat(const SecurityContext& sec, const Json& body) = 0;
@@ -198,6 +206,14 @@ class LlmGatewayModule : public Module {
     std::atomic<int> in_flight{0};
     std::mutex mux_mu;
     std::shared_ptr<MuxClient> mux;   // serving channel (lazy, respawn-safe)
+    // discovery-level ProviderHealth (reference model-registry PRD:280-294:
+    // can the gateway reach the worker? NOT routing health, which is the
+    // breaker's job).  Fed by the watchdog's periodic info round-trips.
+    std::mutex health_mu;
+    std::deque<double> probe_ms;      // recent probe latencies (<=64)
+    int consec_fail = 0, consec_ok = 0;
+    double last_check = 0, last_success = 0;   // unix seconds
+    std::string last_error;
   };
   // the worker's mux channel, (re)connecting if stale; null if down
   std::shared_ptr<MuxClient> ensure_mux(Worker& wk);
@@ -207,6 +223,7 @@ class LlmGatewayModule : public Module {
   };
   std::vector<std::unique_ptr<Worker>> workers_;
   Worker* pick_worker();                    // least-loaded ready worker
+  Json provider_health();                   // ProviderHealth snapshot
   // pick + connect with failover: a dead-but-marked-ready worker is
   // demoted (watchdog respawns it) and the next one is tried
   Worker* pick_live(std::unique_ptr<EngineConn>& conn);

@@ -1,5 +1,7 @@
 #include "system_modules.h"
 
+#include "llm_gateway.h"
+
 #include "../modkit/db.h"
 
 #include "../modkit/auth.h"
@@ -645,6 +647,26 @@ void ModelRegistryModule::init(ModuleCtx& ctx) {
 }
 
 void ModelRegistryModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
+  {
+    // ProviderHealth (reference PRD:280-294): discovery-level status of
+    // the in-node serving providers (engine workers), fed by the
+    // llm-gateway watchdog probes through the hub
+    OperationSpec op;
+    op.method = "GET";
+    op.path = "/model-registry/v1/providers/health";
+    op.operation_id = "providers_health";
+    op.summary = "Discovery-level provider health";
+    op.authenticated = true;
+    op.tags = {"model-registry"};
+    ClientHub* hub = ctx.hub;
+    rest.register_op(op, [hub](HttpRequest& rq, ResponseWriter& w) {
+      (void)rq;
+      auto ph = hub->get<ProviderHealthClient>("llm-gateway");
+      Json out = ph ? ph->provider_health() : Json::object();
+      if (!out.contains("items")) out["items"] = Json::array();
+      w.respond(200, "application/json", out.dump());
+    });
+  }
   OperationSpec list;
   list.method = "GET";
   list.path = "/model-registry/v1/models";
