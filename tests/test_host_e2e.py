@@ -1766,3 +1766,28 @@ def test_chat_schema_shaped_output(server):
     assert set(out) == {"count", "healthy"}
     assert isinstance(out["count"], int)
     assert isinstance(out["healthy"], bool)
+
+
+def test_provider_health_endpoint(server):
+    """Discovery-level ProviderHealth (model-registry PRD:280-294):
+    watchdog probes feed status + latency percentiles per worker."""
+    import time as _t
+    url = BASE.format(server.port) + "/model-registry/v1/providers/health"
+    item = None
+    t0 = _t.time()
+    while _t.time() - t0 < 15:          # needs ~2 probe cycles (~2s each)
+        st, body = _http("GET", url)
+        assert st == 200, body
+        items = json.loads(body)["items"]
+        if items and items[0]["status"] == "healthy" \
+                and "latency_p50_ms" in items[0]["metrics"]:
+            item = items[0]
+            break
+        _t.sleep(0.5)
+    assert item, "worker never reported healthy"
+    assert item["provider_id"].startswith("local::worker-")
+    m = item["metrics"]
+    assert m["consecutive_successes"] >= 2
+    assert m["latency_p50_ms"] > 0
+    assert m["latency_p99_ms"] >= m["latency_p50_ms"]
+    assert item["last_success"] > 0
