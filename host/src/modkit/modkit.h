@@ -15,6 +15,7 @@
 #include <functional>
 #include <future>
 #include <thread>
+#include <typeinfo>
 #include <map>
 #include <memory>
 #include <mutex>
@@ -149,21 +150,26 @@ class RestRegistry {
 // registration when no scope-specific client exists.
 class ClientHub {
  public:
+  // keys are (interface TYPE, name): one module may publish several
+  // client interfaces under its own name (e.g. llm-gateway registers
+  // ChatInvoker AND ProviderHealthClient) — keying by name alone would
+  // silently type-confuse the unchecked static cast below
   template <typename T>
   void register_client(const std::string& iface, std::shared_ptr<T> impl) {
     std::lock_guard<std::mutex> lk(mu_);
-    clients_[iface] = std::static_pointer_cast<void>(impl);
+    clients_[key<T>(iface)] = std::static_pointer_cast<void>(impl);
   }
   template <typename T>
   void register_scoped(const std::string& iface, const std::string& scope,
                        std::shared_ptr<T> impl) {
     std::lock_guard<std::mutex> lk(mu_);
-    clients_[iface + "\x1f" + scope] = std::static_pointer_cast<void>(impl);
+    clients_[key<T>(iface + "\x1f" + scope)] =
+        std::static_pointer_cast<void>(impl);
   }
   template <typename T>
   std::shared_ptr<T> get(const std::string& iface) const {
     std::lock_guard<std::mutex> lk(mu_);
-    auto it = clients_.find(iface);
+    auto it = clients_.find(key<T>(iface));
     if (it == clients_.end()) return nullptr;
     return std::static_pointer_cast<T>(it->second);
   }
@@ -172,18 +178,23 @@ class ClientHub {
                                 const std::string& scope) const {
     {
       std::lock_guard<std::mutex> lk(mu_);
-      auto it = clients_.find(iface + "\x1f" + scope);
+      auto it = clients_.find(key<T>(iface + "\x1f" + scope));
       if (it != clients_.end())
         return std::static_pointer_cast<T>(it->second);
     }
     return get<T>(iface);          // fallback: unscoped registration
   }
+  template <typename T>
   bool has(const std::string& iface) const {
     std::lock_guard<std::mutex> lk(mu_);
-    return clients_.count(iface) > 0;
+    return clients_.count(key<T>(iface)) > 0;
   }
 
  private:
+  template <typename T>
+  static std::string key(const std::string& iface) {
+    return std::string(typeid(T).name()) + "\x1e" + iface;
+  }
   mutable std::mutex mu_;
   std::map<std::string, std::shared_ptr<void>> clients_;
 };

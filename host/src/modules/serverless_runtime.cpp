@@ -467,6 +467,32 @@ Json row_schedule(const DbRow& r) {
   return j;
 }
 
+Json row_trigger(const DbRow& r) {
+  Json j = Json::object();
+  j["trigger_id"] = r.at("id");
+  j["tenant_id"] = r.at("tenant_id");
+  j["event_type_id"] = r.at("event_type_id");
+  const std::string f = r.at("event_filter_query").as_string();
+  if (!f.empty()) j["event_filter_query"] = f;
+  j["entrypoint_id"] = r.at("entrypoint_id");
+  j["status"] = r.at("status");
+  try { j["dead_letter_queue"] = Json::parse(r.at("dlq").as_string()); }
+  catch (...) {}
+  j["created_at"] = r.at("created_at");
+  j["updated_at"] = r.at("updated_at");
+  return j;
+}
+
+// gts-style event type match: exact id, or a trigger pattern ending in
+// '*' matches by prefix (x-gts-ref wildcard convention)
+bool event_type_matches(const std::string& pattern,
+                        const std::string& event_type) {
+  if (!pattern.empty() && pattern.back() == '*')
+    return event_type.rfind(pattern.substr(0, pattern.size() - 1), 0)
+           == 0;
+  return pattern == event_type;
+}
+
 }  // namespace
 
 void ServerlessRuntimeModule::init(ModuleCtx& ctx) {
@@ -544,6 +570,17 @@ void ServerlessRuntimeModule::init(ModuleCtx& ctx) {
       {"0005_invocation_schedule",
        "ALTER TABLE invocations ADD COLUMN schedule_id TEXT NOT NULL "
        "DEFAULT ''"},
+      {"0006_triggers",
+       "CREATE TABLE triggers ("
+       "  tenant_id TEXT NOT NULL,"
+       "  id TEXT NOT NULL UNIQUE,"
+       "  event_type_id TEXT NOT NULL,"
+       "  event_filter_query TEXT NOT NULL DEFAULT '',"
+       "  entrypoint_id TEXT NOT NULL,"
+       "  status TEXT NOT NULL DEFAULT 'active',"
+       "  dlq TEXT NOT NULL DEFAULT '{}',"
+       "  created_at TEXT NOT NULL,"
+       "  updated_at TEXT NOT NULL)"},
   });
 }
 
@@ -1766,6 +1803,185 @@ void ServerlessRuntimeModule::register_rest(ModuleCtx& ctx,
                             {DbValue::S(rq.path_params.at("id"))});
         if (n == 0) throw Problem::not_found("schedule not found");
         w.respond(204, "application/json", "");
+      });
+
+  // ---- triggers (event-driven mechanism, ADR:2194-2290 + Trigger API
+  // table; the EventBroker is deliberately minimal here: POST /events is
+  // the in-node publish endpoint — the reference leaves broker, filter
+  // syntax and DLQ management TBD.  event_filter_query takes the same
+  // {field, op, value} predicate as workflow `when`; dead_letter_queue
+  // config is stored, delivery failures follow the entrypoint's
+  // retry_policy into the dead_lettered state) ----
+  auto load_trigger = [this](SecureConn& conn, const std::string& id) {
+    auto page = conn.select("triggers", "id=?", {DbValue::S(id)}, "id",
+                            false, 1, std::nullopt);
+    if (page.items.empty())
+      throw Problem::not_found("trigger not found");
+    return page.items[0];
+  };
+
+  auto parse_trigger = [parse_body](HttpRequest& rq) {
+    Json b = parse_body(rq);
+    if (b.at("event_type_id").as_string().empty())
+      throw Problem{400, "Bad Request", "about:blank",
+                    "'event_type_id' is required", "validation_error"};
+    if (b.at("entrypoint_id").as_string().empty())
+      throw Problem{400, "Bad Request", "about:blank",
+                    "'entrypoint_id' is required", "validation_error"};
+    const std::string st = b.at("status").as_string("active");
+    if (st != "active" && st != "paused" && st != "disabled")
+      throw Problem{400, "Bad Request", "about:blank",
+                    "status must be active|paused|disabled",
+                    "validation_error"};
+    const Json& f = b.at("event_filter_query");
+    if (!f.is_null() && !f.is_string())
+      throw Problem{400, "Bad Request", "about:blank",
+                    "event_filter_query must be a string",
+                    "validation_error"};
+    return b;
+  };
+
+  reg("POST", "/serverless-runtime/v1/triggers", "create_trigger",
+      [this, parse_trigger, load_trigger](HttpRequest& rq,
+                                          ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        Json b = parse_trigger(rq);
+        SecureConn conn(*db_, scope_for(sec, "create",
+                                        "serverless-runtime:triggers"));
+        auto ep = conn.select(
+            "entrypoints", "id=?",
+            {DbValue::S(b.at("entrypoint_id").as_string())}, "id",
+            false, 1, std::nullopt);
+        if (ep.items.empty())
+          throw Problem::not_found("entrypoint not found");
+        const std::string id =
+            "trg-" + std::to_string(++ctr_) + "-" +
+            std::to_string((long long)(now_s() * 1000) % 100000);
+        Json dlq = b.at("dead_letter_queue");
+        if (!dlq.is_object()) dlq = Json::object();
+        conn.insert(
+            "triggers",
+            {{"id", DbValue::S(id)},
+             {"event_type_id",
+              DbValue::S(b.at("event_type_id").as_string())},
+             {"event_filter_query",
+              DbValue::S(b.at("event_filter_query").as_string(""))},
+             {"entrypoint_id",
+              DbValue::S(b.at("entrypoint_id").as_string())},
+             {"status", DbValue::S(b.at("status").as_string("active"))},
+             {"dlq", DbValue::S(dlq.dump())},
+             {"created_at", DbValue::S(now_iso())},
+             {"updated_at", DbValue::S(now_iso())}});
+        w.respond(201, "application/json",
+                  row_trigger(load_trigger(conn, id)).dump());
+      });
+
+  reg("GET", "/serverless-runtime/v1/triggers", "list_triggers",
+      [this](HttpRequest& rq, ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        SecureConn conn(*db_, scope_for(sec, "read",
+                                        "serverless-runtime:triggers"));
+        auto page = conn.select("triggers", "", {}, "id", false, 1000,
+                                std::nullopt);
+        Json items = Json::array();
+        for (auto& r : page.items) items.push_back(row_trigger(r));
+        Json out = Json::object();
+        out["items"] = items;
+        w.respond(200, "application/json", out.dump());
+      });
+
+  reg("GET", "/serverless-runtime/v1/triggers/{id}", "get_trigger",
+      [this, load_trigger](HttpRequest& rq, ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        SecureConn conn(*db_, scope_for(sec, "read",
+                                        "serverless-runtime:triggers"));
+        w.respond(200, "application/json",
+                  row_trigger(load_trigger(conn,
+                                           rq.path_params.at("id")))
+                      .dump());
+      });
+
+  reg("PUT", "/serverless-runtime/v1/triggers/{id}", "update_trigger",
+      [this, parse_trigger, load_trigger](HttpRequest& rq,
+                                          ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        Json b = parse_trigger(rq);
+        SecureConn conn(*db_, scope_for(sec, "update",
+                                        "serverless-runtime:triggers"));
+        const std::string id = rq.path_params.at("id");
+        Json dlq = b.at("dead_letter_queue");
+        if (!dlq.is_object()) dlq = Json::object();
+        int n = conn.update(
+            "triggers",
+            {{"event_type_id",
+              DbValue::S(b.at("event_type_id").as_string())},
+             {"event_filter_query",
+              DbValue::S(b.at("event_filter_query").as_string(""))},
+             {"entrypoint_id",
+              DbValue::S(b.at("entrypoint_id").as_string())},
+             {"status", DbValue::S(b.at("status").as_string("active"))},
+             {"dlq", DbValue::S(dlq.dump())},
+             {"updated_at", DbValue::S(now_iso())}},
+            "id=?", {DbValue::S(id)});
+        if (n == 0) throw Problem::not_found("trigger not found");
+        w.respond(200, "application/json",
+                  row_trigger(load_trigger(conn, id)).dump());
+      });
+
+  reg("DELETE", "/serverless-runtime/v1/triggers/{id}", "delete_trigger",
+      [this](HttpRequest& rq, ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        SecureConn conn(*db_, scope_for(sec, "delete",
+                                        "serverless-runtime:triggers"));
+        int n = conn.remove("triggers", "id=?",
+                            {DbValue::S(rq.path_params.at("id"))});
+        if (n == 0) throw Problem::not_found("trigger not found");
+        w.respond(204, "application/json", "");
+      });
+
+  reg("POST", "/serverless-runtime/v1/events", "publish_event",
+      [this, parse_body](HttpRequest& rq, ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        Json b = parse_body(rq);
+        const std::string et = b.at("event_type_id").as_string();
+        if (et.empty())
+          throw Problem{400, "Bad Request", "about:blank",
+                        "'event_type_id' is required",
+                        "validation_error"};
+        Json payload = b.at("payload");
+        if (payload.is_null()) payload = Json::object();
+        SecureConn conn(*db_, scope_for(sec, "invoke",
+                                        "serverless-runtime:events"));
+        auto page = conn.select("triggers", "status='active'", {}, "id",
+                                false, 1000, std::nullopt);
+        Json fired = Json::array();
+        for (auto& t : page.items) {
+          if (!event_type_matches(t.at("event_type_id").as_string(), et))
+            continue;
+          const std::string fq = t.at("event_filter_query").as_string();
+          if (!fq.empty()) {
+            Json cond;
+            try { cond = Json::parse(fq); } catch (...) { continue; }
+            if (cond.is_object() && !when_matches(cond, payload))
+              continue;
+          }
+          // entrypoint must be active (same gate as schedules)
+          auto ep = conn.select(
+              "entrypoints", "id=? AND status='active'",
+              {DbValue::S(t.at("entrypoint_id").as_string())}, "id",
+              false, 1, std::nullopt);
+          if (ep.items.empty()) continue;
+          const std::string id = create_invocation(
+              sec.tenant_id, t.at("entrypoint_id").as_string(), payload,
+              "async", "");
+          Json f = Json::object();
+          f["trigger_id"] = t.at("id");
+          f["invocation_id"] = id;
+          fired.push_back(f);
+        }
+        Json out = Json::object();
+        out["fired"] = fired;
+        w.respond(202, "application/json", out.dump());
       });
 
   reg("GET", "/serverless-runtime/v1/schedules/{id}/history",

@@ -538,6 +538,78 @@ def test_schedule_rejects_unknown_entrypoint(sl):
     assert st == 404
 
 
+def test_event_triggers(sl):
+    """Event-driven mechanism (BR-007 third trigger type): a trigger
+    binds an event type (exact or gts-wildcard) to an entrypoint; the
+    in-node publish endpoint fans events out, filter predicates gate
+    them, paused triggers stay silent."""
+    srv, _, _ = sl
+    ep_id = _mk_active_ep(srv, {**json.loads(json.dumps(FN)),
+                                "name": "fn-trig"})
+    # validation
+    st, _ = _http("POST", _url(srv) + "/triggers",
+                  {"event_type_id": "", "entrypoint_id": ep_id})
+    assert st == 400
+    st, _ = _http("POST", _url(srv) + "/triggers",
+                  {"event_type_id": "gts.x.app.ev.v1~",
+                   "entrypoint_id": "ep-nope"})
+    assert st == 404
+    # exact-type trigger with a payload filter
+    st, resp = _http("POST", _url(srv) + "/triggers",
+                     {"event_type_id": "gts.x.app.order.v1~",
+                      "event_filter_query":
+                          json.dumps({"field": "kind", "op": "eq",
+                                      "value": "big"}),
+                      "entrypoint_id": ep_id})
+    assert st == 201, resp
+    t1 = json.loads(resp)["trigger_id"]
+    # wildcard trigger, no filter
+    st, resp = _http("POST", _url(srv) + "/triggers",
+                     {"event_type_id": "gts.x.app.*",
+                      "entrypoint_id": ep_id})
+    assert st == 201, resp
+    t2 = json.loads(resp)["trigger_id"]
+
+    # matching event: filter passes on t1, wildcard matches on t2
+    st, resp = _http("POST", _url(srv) + "/events",
+                     {"event_type_id": "gts.x.app.order.v1~",
+                      "payload": {"kind": "big", "n": 1}})
+    assert st == 202, resp
+    fired = json.loads(resp)["fired"]
+    assert {f["trigger_id"] for f in fired} == {t1, t2}
+    # the invocations actually run with the event payload as input
+    inv = _wait_status(srv, fired[0]["invocation_id"], ("succeeded",))
+    assert inv["input"] == {"kind": "big", "n": 1}
+
+    # filter rejects: only the wildcard trigger fires
+    st, resp = _http("POST", _url(srv) + "/events",
+                     {"event_type_id": "gts.x.app.order.v1~",
+                      "payload": {"kind": "small"}})
+    fired = json.loads(resp)["fired"]
+    assert {f["trigger_id"] for f in fired} == {t2}
+
+    # unrelated event type: nothing fires
+    st, resp = _http("POST", _url(srv) + "/events",
+                     {"event_type_id": "gts.x.other.ev.v1~",
+                      "payload": {}})
+    assert json.loads(resp)["fired"] == []
+
+    # pause via PUT -> silent; CRUD round-trip
+    st, resp = _http("GET", _url(srv) + f"/triggers/{t2}")
+    assert st == 200
+    body = json.loads(resp)
+    body["status"] = "paused"
+    st, resp = _http("PUT", _url(srv) + f"/triggers/{t2}", body)
+    assert st == 200 and json.loads(resp)["status"] == "paused"
+    st, resp = _http("POST", _url(srv) + "/events",
+                     {"event_type_id": "gts.x.app.ping.v1~",
+                      "payload": {}})
+    assert json.loads(resp)["fired"] == []
+    st, _ = _http("DELETE", _url(srv) + f"/triggers/{t1}")
+    assert st == 204
+    _http("DELETE", _url(srv) + f"/triggers/{t2}")
+
+
 def test_schedule_survives_restart_catch_up(tmp_path):
     """Schedules are durable rows: after a host restart the ticker
     resumes them, and a catch_up missed-policy fires ONCE for the gap
