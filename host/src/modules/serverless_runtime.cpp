@@ -570,6 +570,12 @@ void ServerlessRuntimeModule::init(ModuleCtx& ctx) {
       {"0005_invocation_schedule",
        "ALTER TABLE invocations ADD COLUMN schedule_id TEXT NOT NULL "
        "DEFAULT ''"},
+      {"0007_tenant_policies",
+       "CREATE TABLE tenant_policies ("
+       "  tenant_id TEXT NOT NULL UNIQUE,"
+       "  enabled INTEGER NOT NULL DEFAULT 1,"
+       "  quotas TEXT NOT NULL DEFAULT '{}',"
+       "  updated_at TEXT NOT NULL)"},
       {"0006_triggers",
        "CREATE TABLE triggers ("
        "  tenant_id TEXT NOT NULL,"
@@ -716,6 +722,25 @@ void ServerlessRuntimeModule::run_invocation(const std::string& tenant,
     std::string deny = adm->admit(tenant);
     if (!deny.empty()) {
       enqueue_at(now_s() + 0.2, tenant, id);
+      return;
+    }
+  }
+  // tenant policy quota: cap concurrently-RUNNING executions (this one
+  // just transitioned to running; quota counts others + us)
+  {
+    long long q = tenant_policy(tenant)
+                      .path("quotas.max_concurrent_executions")
+                      .as_int(-1);
+    if (q >= 0 &&
+        count_rows(tenant, "invocations", "status='running'") > q) {
+      // scheduling retreat (not a domain transition): back to queued,
+      // retried shortly — the invocation is never lost
+      conn.update("invocations",
+                  {{"status", DbValue::S("queued")},
+                   {"updated_at", DbValue::S(now_iso())}},
+                  "id=? AND status='running'", {DbValue::S(id)});
+      if (adm) adm->release(tenant);
+      enqueue_at(now_s() + 0.25, tenant, id);
       return;
     }
   }
@@ -1033,6 +1058,35 @@ void ServerlessRuntimeModule::schedule_tick() {
   }
 }
 
+Json ServerlessRuntimeModule::tenant_policy(const std::string& tenant) {
+  Json pol = Json::object();
+  pol["tenant_id"] = tenant;
+  pol["enabled"] = true;
+  pol["quotas"] = Json::object();
+  SecureConn conn(*db_, AccessScope::for_tenant(tenant));
+  auto page = conn.select("tenant_policies", "", {}, "tenant_id", false,
+                          1, std::nullopt);
+  if (!page.items.empty()) {
+    pol["enabled"] = page.items[0].at("enabled").as_int(1) != 0;
+    try {
+      pol["quotas"] = Json::parse(page.items[0].at("quotas").as_string());
+    } catch (...) {}
+    pol["updated_at"] = page.items[0].at("updated_at");
+  }
+  return pol;
+}
+
+long long ServerlessRuntimeModule::count_rows(const std::string& tenant,
+                                              const char* table,
+                                              const char* extra_where) {
+  std::lock_guard<std::mutex> dblk(db_->mu());
+  std::string q = std::string("SELECT COUNT(*) AS n FROM ") + table +
+                  " WHERE tenant_id=?";
+  if (extra_where) q += std::string(" AND ") + extra_where;
+  auto rows = db_->query(q, {DbValue::S(tenant)});
+  return rows.empty() ? 0 : rows[0].at("n").as_int(0);
+}
+
 void ServerlessRuntimeModule::timer_loop() {
   while (!stopping_) {
     std::vector<Timer> due;
@@ -1168,6 +1222,18 @@ void ServerlessRuntimeModule::register_rest(ModuleCtx& ctx,
           throw Problem{400, "Bad Request", "about:blank",
                         issues.at(0).at("message").as_string(),
                         "validation_error"};
+        {
+          Json pol = tenant_policy(sec.tenant_id);
+          if (!pol.at("enabled").as_bool(true))
+            throw Problem{403, "Forbidden", "about:blank",
+                          "serverless runtime disabled for tenant",
+                          "runtime_disabled"};
+          long long q = pol.path("quotas.max_definitions").as_int(-1);
+          if (q >= 0 && count_rows(sec.tenant_id, "entrypoints") >= q)
+            throw Problem{429, "Too Many Requests", "about:blank",
+                          "max_definitions quota reached",
+                          "quota_exceeded"};
+        }
         SecureConn conn(*db_,
                         scope_for(sec, "create",
                                   "serverless-runtime:entrypoints"));
@@ -1369,6 +1435,10 @@ void ServerlessRuntimeModule::register_rest(ModuleCtx& ctx,
         if (ep_id.empty())
           throw Problem{400, "Bad Request", "about:blank",
                         "'entrypoint_id' is required", "validation_error"};
+        if (!tenant_policy(sec.tenant_id).at("enabled").as_bool(true))
+          throw Problem{403, "Forbidden", "about:blank",
+                        "serverless runtime disabled for tenant",
+                        "runtime_disabled"};
         SecureConn conn(*db_, scope_for(sec, "invoke",
                                         "serverless-runtime:invocations"));
         auto ep = conn.select("entrypoints", "id=?", {DbValue::S(ep_id)},
@@ -1660,6 +1730,14 @@ void ServerlessRuntimeModule::register_rest(ModuleCtx& ctx,
                                             ResponseWriter& w) {
         auto sec = sec_of(rq);
         Json b = parse_schedule(rq);
+        {
+          Json pol = tenant_policy(sec.tenant_id);
+          long long q = pol.path("quotas.max_schedules").as_int(-1);
+          if (q >= 0 && count_rows(sec.tenant_id, "schedules") >= q)
+            throw Problem{429, "Too Many Requests", "about:blank",
+                          "max_schedules quota reached",
+                          "quota_exceeded"};
+        }
         SecureConn conn(*db_, scope_for(sec, "create",
                                         "serverless-runtime:schedules"));
         auto ep = conn.select(
@@ -1805,6 +1883,112 @@ void ServerlessRuntimeModule::register_rest(ModuleCtx& ctx,
         w.respond(204, "application/json", "");
       });
 
+  // ---- tenant runtime policy + quota usage (ADR Tenant Runtime
+  // Policy / Quota Usage APIs; quotas gate the create/start paths) ----
+  reg("GET", "/serverless-runtime/v1/tenants/{tenant_id}/runtime-policy",
+      "get_runtime_policy", [this](HttpRequest& rq, ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        const std::string t = rq.path_params.at("tenant_id");
+        scope_for(sec, "read", "serverless-runtime:policy");
+        if (t != sec.tenant_id)
+          throw Problem::forbidden("cross-tenant policy access");
+        w.respond(200, "application/json", tenant_policy(t).dump());
+      });
+
+  reg("PUT", "/serverless-runtime/v1/tenants/{tenant_id}/runtime-policy",
+      "put_runtime_policy",
+      [this, parse_body](HttpRequest& rq, ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        const std::string t = rq.path_params.at("tenant_id");
+        scope_for(sec, "update", "serverless-runtime:policy");
+        if (t != sec.tenant_id)
+          throw Problem::forbidden("cross-tenant policy access");
+        Json b = parse_body(rq);
+        Json q = b.at("quotas");
+        if (!q.is_null() && !q.is_object())
+          throw Problem{400, "Bad Request", "about:blank",
+                        "quotas must be an object", "validation_error"};
+        if (q.is_object())
+          for (const char* k : {"max_concurrent_executions",
+                                "max_definitions", "max_schedules",
+                                "max_triggers"})
+            if (q.contains(k) && q.at(k).as_int(-1) < 0)
+              throw Problem{400, "Bad Request", "about:blank",
+                            std::string(k) + " must be >= 0",
+                            "validation_error"};
+        SecureConn conn(*db_, AccessScope::for_tenant(t));
+        int n = conn.update(
+            "tenant_policies",
+            {{"enabled", DbValue::I(b.at("enabled").as_bool(true) ? 1
+                                                                  : 0)},
+             {"quotas",
+              DbValue::S(q.is_object() ? q.dump() : "{}")},
+             {"updated_at", DbValue::S(now_iso())}},
+            "tenant_id=?", {DbValue::S(t)});
+        if (n == 0)
+          conn.insert("tenant_policies",
+                      {{"enabled",
+                        DbValue::I(b.at("enabled").as_bool(true) ? 1
+                                                                 : 0)},
+                       {"quotas",
+                        DbValue::S(q.is_object() ? q.dump() : "{}")},
+                       {"updated_at", DbValue::S(now_iso())}});
+        w.respond(200, "application/json", tenant_policy(t).dump());
+      });
+
+  reg("GET", "/serverless-runtime/v1/tenants/{tenant_id}/usage",
+      "get_quota_usage", [this](HttpRequest& rq, ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        const std::string t = rq.path_params.at("tenant_id");
+        scope_for(sec, "read", "serverless-runtime:policy");
+        if (t != sec.tenant_id)
+          throw Problem::forbidden("cross-tenant usage access");
+        Json pol = tenant_policy(t);
+        Json cur = Json::object();
+        cur["definitions"] = (long)count_rows(t, "entrypoints");
+        cur["schedules"] = (long)count_rows(t, "schedules");
+        cur["triggers"] = (long)count_rows(t, "triggers");
+        cur["concurrent_executions"] = (long)count_rows(
+            t, "invocations", "status IN ('queued','running')");
+        Json out = Json::object();
+        out["tenant_id"] = t;
+        out["timestamp"] = now_iso();
+        out["current"] = cur;
+        out["quotas"] = pol.at("quotas");
+        w.respond(200, "application/json", out.dump());
+      });
+
+  reg("GET", "/serverless-runtime/v1/tenants/{tenant_id}/usage/history",
+      "get_quota_usage_history",
+      [this](HttpRequest& rq, ResponseWriter& w) {
+        auto sec = sec_of(rq);
+        const std::string t = rq.path_params.at("tenant_id");
+        scope_for(sec, "read", "serverless-runtime:policy");
+        if (t != sec.tenant_id)
+          throw Problem::forbidden("cross-tenant usage access");
+        // executions per UTC hour over the last 24h, computed from the
+        // durable invocation rows (no separate snapshot store)
+        Json items = Json::array();
+        {
+          std::lock_guard<std::mutex> dblk(db_->mu());
+          auto rows = db_->query(
+              "SELECT substr(created_at, 1, 13) AS hour, COUNT(*) AS n "
+              "FROM invocations WHERE tenant_id=? "
+              "GROUP BY hour ORDER BY hour DESC LIMIT 24",
+              {DbValue::S(t)});
+          for (auto& r : rows) {
+            Json it = Json::object();
+            it["hour"] = r.at("hour").as_string() + ":00Z";
+            it["executions"] = (long)r.at("n").as_int(0);
+            items.push_back(it);
+          }
+        }
+        Json out = Json::object();
+        out["tenant_id"] = t;
+        out["items"] = items;
+        w.respond(200, "application/json", out.dump());
+      });
+
   // ---- triggers (event-driven mechanism, ADR:2194-2290 + Trigger API
   // table; the EventBroker is deliberately minimal here: POST /events is
   // the in-node publish endpoint — the reference leaves broker, filter
@@ -1846,6 +2030,14 @@ void ServerlessRuntimeModule::register_rest(ModuleCtx& ctx,
                                           ResponseWriter& w) {
         auto sec = sec_of(rq);
         Json b = parse_trigger(rq);
+        {
+          Json pol = tenant_policy(sec.tenant_id);
+          long long q = pol.path("quotas.max_triggers").as_int(-1);
+          if (q >= 0 && count_rows(sec.tenant_id, "triggers") >= q)
+            throw Problem{429, "Too Many Requests", "about:blank",
+                          "max_triggers quota reached",
+                          "quota_exceeded"};
+        }
         SecureConn conn(*db_, scope_for(sec, "create",
                                         "serverless-runtime:triggers"));
         auto ep = conn.select(

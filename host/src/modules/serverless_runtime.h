@@ -47,6 +47,11 @@ class ServerlessRuntimeModule : public Module {
   void timer_loop();
   // fire due schedules (cron/interval triggers, PRD BR-007/BR-022)
   void schedule_tick();
+  // tenant runtime policy (ADR Tenant Runtime Policy API): stored row
+  // or defaults; quotas gate create/start paths
+  Json tenant_policy(const std::string& tenant);
+  long long count_rows(const std::string& tenant, const char* table,
+                       const char* extra_where = nullptr);
   std::string create_invocation(const std::string& tenant,
                                 const std::string& ep_id, const Json& input,
                                 const std::string& mode,

@@ -610,6 +610,76 @@ def test_event_triggers(sl):
     _http("DELETE", _url(srv) + f"/triggers/{t2}")
 
 
+def test_tenant_runtime_policy_and_quotas(sl):
+    """Tenant Runtime Policy + Quota Usage APIs: quotas gate the create
+    paths (429 quota_exceeded), enabled=false disables invocation (403),
+    usage reports current counts vs quotas."""
+    srv, _, _ = sl
+    tid = "00000000-df51-5b42-9538-d2b56b7ee953"   # default tenant
+    pol_url = _url(srv) + f"/tenants/{tid}/runtime-policy"
+    # defaults
+    st, resp = _http("GET", pol_url)
+    assert st == 200 and json.loads(resp)["enabled"] is True
+    # cross-tenant access refused
+    st, _ = _http("GET", _url(srv) + "/tenants/other-t/runtime-policy")
+    assert st == 403
+
+    ep_id = _mk_active_ep(srv, {**json.loads(json.dumps(FN)),
+                                "name": "fn-quota"})
+    n_defs = None
+    try:
+        st, resp = _http("GET", _url(srv) + f"/tenants/{tid}/usage")
+        assert st == 200, resp
+        usage = json.loads(resp)
+        n_defs = usage["current"]["definitions"]
+        assert n_defs >= 1
+
+        # cap definitions at the current count -> next create is 429
+        st, resp = _http("PUT", pol_url,
+                         {"enabled": True,
+                          "quotas": {"max_definitions": n_defs}})
+        assert st == 200, resp
+        st, resp = _http("POST", _url(srv) + "/entrypoints",
+                         {**json.loads(json.dumps(FN)),
+                          "name": "fn-over-quota"})
+        assert st == 429, resp
+        assert json.loads(resp)["code"] == "quota_exceeded"
+
+        # schedules quota 0 -> create refused
+        st, _ = _http("PUT", pol_url,
+                      {"enabled": True, "quotas": {"max_schedules": 0}})
+        st, resp = _http("POST", _url(srv) + "/schedules",
+                         {"name": "q", "entrypoint_id": ep_id,
+                          "expression": {"kind": "interval",
+                                         "value": "PT1H"}})
+        assert st == 429
+
+        # disable the runtime -> invocations are 403
+        st, _ = _http("PUT", pol_url, {"enabled": False})
+        st, resp = _http("POST", _url(srv) + "/invocations",
+                         {"entrypoint_id": ep_id, "input": {}})
+        assert st == 403 and json.loads(resp)["code"] == "runtime_disabled"
+
+        # validation: negative quota rejected
+        st, _ = _http("PUT", pol_url,
+                      {"quotas": {"max_triggers": -2}})
+        assert st == 400
+
+        # usage history shape
+        st, resp = _http("GET",
+                         _url(srv) + f"/tenants/{tid}/usage/history")
+        assert st == 200
+        items = json.loads(resp)["items"]
+        assert items and "executions" in items[0]
+    finally:
+        # restore permissive policy for the rest of the suite
+        _http("PUT", pol_url, {"enabled": True, "quotas": {}})
+    st, resp = _http("POST", _url(srv) + "/invocations",
+                     {"entrypoint_id": ep_id, "mode": "sync",
+                      "input": {"z": 1}})
+    assert st == 200 and json.loads(resp)["status"] == "succeeded"
+
+
 def test_schedule_survives_restart_catch_up(tmp_path):
     """Schedules are durable rows: after a host restart the ticker
     resumes them, and a catch_up missed-policy fires ONCE for the gap
