@@ -316,6 +316,18 @@ class SchemaMachine:
                 self._compile(props[k] if isinstance(props[k], dict)
                               else {})
             self.script.append(("lit", b"}"))
+        elif isinstance(sch.get("enum"), list) and sch["enum"]:
+            import json as _json
+            alts = []
+            for v in sch["enum"]:
+                try:
+                    alts.append(_json.dumps(v, sort_keys=True).encode())
+                except (TypeError, ValueError):
+                    pass
+            if alts:
+                self.script.append(("choice", tuple(alts)))
+            else:
+                self.script.append(("free", (None, None)))
         elif typ == "array":
             items = sch.get("items")
             ist = None
@@ -349,6 +361,11 @@ class SchemaMachine:
         kind, arg = self.script[self.seg]
         if kind == "lit":
             return {arg[self.lit_pos]}, False
+        if kind == "choice":
+            # one of the enum literals, matched byte-by-byte; alts that
+            # no longer match the consumed prefix are out
+            return {a[self.lit_pos] for a in arg
+                    if len(a) > self.lit_pos}, False
         # free region
         start_set, typ = arg[0], arg[1]
         inner, _ = self.m.allowed()
@@ -394,6 +411,21 @@ class SchemaMachine:
             if self.lit_pos >= len(arg):
                 self.seg += 1
                 self.lit_pos = 0
+            return
+        if kind == "choice":
+            self.m.feed(b)
+            alts = tuple(a for a in arg if len(a) > self.lit_pos
+                         and a[self.lit_pos] == b)
+            self.lit_pos += 1
+            # done when exactly the consumed prefix equals a full alt and
+            # no longer alt still matches beyond it
+            alive = tuple(a for a in alts if len(a) > self.lit_pos)
+            full = any(len(a) == self.lit_pos for a in alts)
+            if full and not alive:
+                self.seg += 1
+                self.lit_pos = 0
+            else:
+                self.script[self.seg] = ("choice", alts)
             return
         # free region
         if self._free_complete():

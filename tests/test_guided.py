@@ -251,3 +251,37 @@ def test_schema_typed_array_items():
         assert eos, (seed, bytes(out))
         j = json.loads(out.decode("utf-8", errors="replace"))
         assert all(isinstance(x, (int, float)) for x in j["xs"])
+
+
+def test_schema_enum_choice():
+    """enum values are enforced byte-exactly (string quotes make prefix
+    alternatives like fast/fastidious both reachable)."""
+    from hyperspot.engine.guided import SchemaMachine
+    schema = {"type": "object", "required": ["mode"],
+              "properties": {"mode":
+                             {"enum": ["fast", "fastidious", "slow"]}}}
+    seen = set()
+    for seed in range(60):
+        rng = random.Random(seed)
+        m = SchemaMachine(schema)
+        out = bytearray()
+        eos = False
+        for _ in range(100):
+            allow, eos = m.allowed()
+            if eos:
+                break
+            assert allow, (seed, bytes(out))
+            b = rng.choice(sorted(allow))
+            m.feed(b)
+            out.append(b)
+        assert eos
+        j = json.loads(out.decode())
+        assert j["mode"] in ("fast", "fastidious", "slow")
+        seen.add(j["mode"])
+    assert seen == {"fast", "fastidious", "slow"}
+    # out-of-enum bytes rejected
+    m = SchemaMachine(schema)
+    for b in b'{"mode":"f':
+        m.feed(b)
+    with pytest.raises(ValueError):
+        m.feed(ord("x"))                # neither fast nor fastidious
