@@ -1187,6 +1187,12 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
   w.write_chunk(sse_chunk(rid, canonical, role_delta));
   bool client_gone = false;
   bool pending_first = true;
+  // forced tool call: accumulate the (grammar-guaranteed JSON) text so
+  // the FINAL chunk can carry the parsed ToolCall alongside the deltas
+  const bool forced_tool =
+      body.at("tool_choice").as_string("") == "required" &&
+      body.at("tools").is_array() && body.at("tools").size() > 0;
+  std::string acc_text;
   while (true) {
     std::optional<Json> msg;
     if (pending_first) {
@@ -1212,6 +1218,7 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
       }
       Json d = Json::object();
       d["content"] = msg->at("text").as_string();
+      if (forced_tool) acc_text += msg->at("text").as_string();
       if (!w.write_chunk(sse_chunk(rid, canonical, d))) {
         client_gone = true;    // abort generation server-side
         mux->abort(rid);
@@ -1219,8 +1226,22 @@ void LlmGatewayModule::chat_handler(HttpRequest& req, ResponseWriter& w) {
       }
     } else if (ev == "done") {
       record_usage(sec.tenant_id, msg->at("usage"));
-      w.write_chunk(sse_chunk(rid, canonical, Json::object(),
-                              msg->at("finish_reason").as_string("stop"),
+      Json final_delta = Json::object();
+      std::string finish = msg->at("finish_reason").as_string("stop");
+      if (forced_tool) {
+        try {
+          Json tc = Json::parse(acc_text);
+          if (tc.is_object() && !tc.at("name").as_string().empty()) {
+            Json call = Json::object();
+            call["id"] = "call-" + rid;
+            call["name"] = tc.at("name");
+            call["arguments"] = tc.at("arguments");
+            final_delta["tool_call"] = call;
+            finish = "tool_calls";
+          }
+        } catch (...) {}
+      }
+      w.write_chunk(sse_chunk(rid, canonical, final_delta, finish,
                               msg->at("usage")));
       break;
     } else if (ev == "error") {

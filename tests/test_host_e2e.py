@@ -1791,3 +1791,39 @@ def test_provider_health_endpoint(server):
     assert m["latency_p50_ms"] > 0
     assert m["latency_p99_ms"] >= m["latency_p50_ms"]
     assert item["last_success"] > 0
+
+
+def test_chat_tool_calling_streamed(server):
+    """Streamed forced tool call: text deltas stream as usual, the final
+    chunk before [DONE] carries the parsed ToolCall + finish_reason
+    tool_calls."""
+    import urllib.request
+    url = BASE.format(server.port) + "/v1/chat/completions"
+    tools = [{"name": "lookup", "parameters": {"type": "object"}}]
+    got = None
+    for seed in (3, 4, 5, 7, 11, 13):
+        req = urllib.request.Request(url, method="POST")
+        req.add_header("content-type", "application/json")
+        data = json.dumps({
+            "model": "tiny-llama", "stream": True, "tools": tools,
+            "tool_choice": "required", "max_tokens": 900,
+            "temperature": 1.0, "seed": seed,
+            "messages": [{"role": "user", "content":
+                          [{"type": "text", "text": "call it"}]}],
+        }).encode()
+        with urllib.request.urlopen(req, data=data, timeout=120) as r:
+            raw = r.read().decode()
+        events = [json.loads(l[6:]) for l in raw.splitlines()
+                  if l.startswith("data: ") and l != "data: [DONE]"]
+        final = events[-1]
+        if final.get("finish_reason") == "tool_calls":
+            got = final
+            break
+    assert got, "no seed completed a streamed tool call"
+    call = got["delta"]["tool_call"]
+    assert isinstance(call["name"], str)
+    assert isinstance(call["arguments"], dict)
+    # streamed deltas re-assemble to the same JSON
+    text = "".join(e["delta"].get("content", "") for e in events
+                   if "delta" in e)
+    assert json.loads(text)["name"] == call["name"]
