@@ -283,6 +283,9 @@ class SchemaMachine:
         "object": {ord("{")},
     }
 
+    MAX_SEGMENTS = 512      # untrusted input: bound the compiled script
+    MAX_DEPTH = 24
+
     def __init__(self, schema: dict) -> None:
         self.m = JsonByteMachine()
         # script: ("lit", bytes) | ("free", start_byte_set | None)
@@ -296,8 +299,13 @@ class SchemaMachine:
 
     # ---- schema -> script ----
 
-    def _compile(self, sch: dict) -> None:
+    def _compile(self, sch: dict, depth: int = 0) -> None:
         import json as _json
+        if depth > self.MAX_DEPTH or len(self.script) > self.MAX_SEGMENTS:
+            # degrade instead of recursing without bound on untrusted
+            # input — the region is still grammar-valid JSON
+            self.script.append(("free", (None, None)))
+            return
         typ = sch.get("type")
         props = sch.get("properties")
         if typ == "object" and isinstance(props, dict):
@@ -314,7 +322,7 @@ class SchemaMachine:
                     _json.dumps(k, sort_keys=True).encode() + b":"
                 self.script.append(("lit", pre))
                 self._compile(props[k] if isinstance(props[k], dict)
-                              else {})
+                              else {}, depth + 1)
             self.script.append(("lit", b"}"))
         elif isinstance(sch.get("enum"), list) and sch["enum"]:
             import json as _json

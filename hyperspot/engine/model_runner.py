@@ -410,7 +410,12 @@ class ModelRunner:
                 if r.sampling.response_format == "tool_call":
                     m = ToolCallMachine()
                 elif r.sampling.response_schema is not None:
-                    m = SchemaMachine(r.sampling.response_schema)
+                    try:
+                        # untrusted schema: any compile failure degrades
+                        # to plain JSON-grammar enforcement
+                        m = SchemaMachine(r.sampling.response_schema)
+                    except Exception:
+                        m = JsonByteMachine()
                 else:
                     m = JsonByteMachine()
                 r._guided = m

@@ -285,3 +285,48 @@ def test_schema_enum_choice():
         m.feed(b)
     with pytest.raises(ValueError):
         m.feed(ord("x"))                # neither fast nor fastidious
+
+
+@settings(max_examples=150, deadline=None)
+@given(st.recursive(
+    st.one_of(
+        st.fixed_dictionaries({"type": st.sampled_from(
+            ["string", "integer", "number", "boolean", "null",
+             "array", "object", "bogus"])}),
+        st.fixed_dictionaries({"enum": st.lists(
+            st.one_of(st.text(max_size=5), st.integers(-99, 99),
+                      st.booleans()), max_size=4)}),
+        st.just({}),
+    ),
+    lambda children: st.fixed_dictionaries({
+        "type": st.just("object"),
+        "properties": st.dictionaries(
+            st.text(min_size=1, max_size=6), children, max_size=4),
+        "required": st.lists(st.text(min_size=1, max_size=6),
+                             max_size=4),
+    }),
+    max_leaves=12))
+def test_schema_machine_survives_arbitrary_schemas(schema):
+    """response_schema is UNTRUSTED tenant input compiled inside the
+    worker: any schema must produce a working machine whose walks stay
+    valid JSON (degrading where unsupported), never a crash."""
+    from hyperspot.engine.guided import SchemaMachine
+    m = SchemaMachine(schema)
+    rng = random.Random(0)
+    out = bytearray()
+    eos = False
+    for _ in range(600):
+        allow, eos = m.allowed()
+        if eos:
+            break
+        assert allow, (schema, bytes(out))
+        pool = sorted(allow)
+        if len(out) > 40:
+            pref = [b for b in pool if b in b'"]}0123456789tf,']
+            if pref and rng.random() < 0.9:
+                pool = pref
+        b = rng.choice(pool)
+        m.feed(b)
+        out.append(b)
+    if eos:
+        json.loads(out.decode("utf-8", errors="replace"))
