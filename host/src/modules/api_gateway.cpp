@@ -1,5 +1,7 @@
 #include "api_gateway.h"
 
+#include "../modkit/json_schema.h"
+
 #include "../modkit/telemetry.h"
 
 #include <random>
@@ -341,6 +343,25 @@ void ApiGatewayModule::dispatch(HttpRequest& req, ResponseWriter& w) {
     }
   }
   req.extensions["security"] = sec.to_json();
+
+  // declared request schema => automatic body validation (the runtime
+  // analog of the reference's typed OperationBuilder request bodies);
+  // non-JSON/empty bodies are left to the handler's own checks
+  if (route->spec.request_schema.is_object() && !req.body.empty()) {
+    Json body;
+    bool parsed = true;
+    try { body = Json::parse(req.body); } catch (...) { parsed = false; }
+    if (parsed) {
+      const std::string err =
+          json_schema_err(route->spec.request_schema, body, "body");
+      if (!err.empty()) {
+        respond_problem(w, {400, "Bad Request", "about:blank",
+                            "request validation: " + err,
+                            "validation_error"}, req.path);
+        return;
+      }
+    }
+  }
 
   // error-mapping boundary around the handler
   try {

@@ -1827,3 +1827,24 @@ def test_chat_tool_calling_streamed(server):
     text = "".join(e["delta"].get("content", "") for e in events
                    if "delta" in e)
     assert json.loads(text)["name"] == call["name"]
+
+
+def test_declared_request_schema_enforced(server):
+    """OperationSpec.request_schema is enforced by the gateway before
+    the handler (typed OperationBuilder analog): ill-typed chat bodies
+    are 400 validation_error with a pointer."""
+    st, body = _http("POST", BASE.format(server.port) +
+                     "/v1/chat/completions",
+                     body={"model": "tiny-llama",
+                           "temperature": "hot",
+                           "messages": [{"role": "user", "content":
+                                         [{"type": "text", "text": "x"}]}]})
+    assert st == 400, body
+    j = json.loads(body)
+    assert j["code"] == "validation_error"
+    assert "temperature" in j["detail"]
+    # missing required field
+    st, body = _http("POST", BASE.format(server.port) +
+                     "/v1/chat/completions", body={"model": "tiny-llama"})
+    assert st == 400
+    assert "required" in json.loads(body)["detail"]
