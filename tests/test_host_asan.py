@@ -36,6 +36,10 @@ def test_e2e_suites_under_asan():
          "tests/test_host_e2e.py::test_settings_crud_and_tenant_isolation",
          "tests/test_host_e2e.py::test_chat_tool_calling_passthrough",
          "tests/test_host_e2e.py::test_chat_schema_shaped_output",
+         "tests/test_host_e2e.py::test_declared_request_schema_enforced",
+         "tests/test_host_serverless.py::test_event_triggers",
+         "tests/test_host_serverless.py::test_tenant_runtime_policy_and_quotas",
+         "tests/test_host_serverless.py::test_workflow_when_args_output_to",
          ],
         cwd=ROOT, env=env, capture_output=True, text=True, timeout=800)
     assert out.returncode == 0, out.stdout[-3000:] + out.stderr[-2000:]
