@@ -569,21 +569,10 @@ def engine_direct(args, rank, world, on_gpu):
 
 # ------------------------------------------------------------------ main
 
-def _raise_fd_limit():
-    """8-GPU DP runs hold batch x N SSE connections (16k at defaults) in
-    this client plus the same count in the spawned host — lift
-    RLIMIT_NOFILE to the hard cap before anything opens sockets."""
-    try:
-        import resource
-        soft, hard = resource.getrlimit(resource.RLIMIT_NOFILE)
-        if soft < hard:
-            resource.setrlimit(resource.RLIMIT_NOFILE, (hard, hard))
-    except Exception:
-        pass
-
-
 def main():
-    _raise_fd_limit()
+    # 8-GPU DP runs hold batch x N SSE connections (16k at defaults) in
+    # this client plus the same count in the spawned host
+    _raise_nofile(1 << 20)
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=64)
