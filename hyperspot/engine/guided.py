@@ -295,6 +295,7 @@ class SchemaMachine:
         self.lit_pos = 0      # within a literal segment
         self.free_depth = -1  # stack depth at free-region start (-1: n/a)
         self.free_started = False
+        self.free_len = 0     # bytes fed into the current free region
         self.consumed = 0
 
     # ---- schema -> script ----
@@ -344,6 +345,14 @@ class SchemaMachine:
                 ist = self._TYPE_START[items["type"]]
             self.script.append(
                 ("free", (self._TYPE_START["array"], "array", ist)))
+        elif typ == "string":
+            lo = sch.get("minLength")
+            hi = sch.get("maxLength")
+            lo = lo if isinstance(lo, int) and lo >= 0 else None
+            hi = hi if isinstance(hi, int) and hi >= 0 else None
+            self.script.append(
+                ("free", (self._TYPE_START["string"], "string",
+                          None, lo, hi)))
         elif typ in self._TYPE_START:
             self.script.append(("free", (self._TYPE_START[typ], typ)))
         else:
@@ -380,6 +389,15 @@ class SchemaMachine:
         if typ == "integer" and self.m.mode in (
                 "num_zero", "num_int", "num_minus"):
             inner -= {ord("."), ord("e"), ord("E")}
+        if typ == "string" and len(arg) > 4 and self.free_started \
+                and self.m.mode == "string":
+            # body length so far excludes the opening quote
+            body = self.free_len - 1
+            lo, hi = arg[3], arg[4]
+            if hi is not None and body >= hi:
+                inner = inner & {0x22}            # must close now
+            elif lo is not None and body < lo:
+                inner = inner - {0x22}            # may not close yet
         if typ == "array" and len(arg) > 2 and arg[2] is not None \
                 and self.free_started \
                 and len(self.m.stack) == self.free_depth + 1 \
@@ -453,7 +471,9 @@ class SchemaMachine:
         if not self.free_started:
             self.free_depth = len(self.m.stack)
             self.free_started = True
+            self.free_len = 0
         self.m.feed(b)
+        self.free_len += 1
         # a region that completes exactly at a script boundary with no
         # literal after it (top-level value) finishes via eos_ok
 

@@ -330,3 +330,33 @@ def test_schema_machine_survives_arbitrary_schemas(schema):
         out.append(b)
     if eos:
         json.loads(out.decode("utf-8", errors="replace"))
+
+
+def test_schema_string_length_bounds():
+    """string minLength/maxLength enforced at the mask (RAW body bytes:
+    escape sequences and multi-byte UTF-8 count as their bytes)."""
+    from hyperspot.engine.guided import SchemaMachine
+    schema = {"type": "object", "required": ["name", "tag"],
+              "properties": {"name": {"type": "string", "maxLength": 6},
+                             "tag": {"type": "string", "minLength": 3,
+                                     "maxLength": 5}}}
+    ascii_body = set(b"abcdefgh")
+    for seed in range(120):
+        rng = random.Random(seed)
+        m = SchemaMachine(schema)
+        out = bytearray()
+        eos = False
+        for _ in range(200):
+            allow, eos = m.allowed()
+            if eos:
+                break
+            assert allow, (seed, bytes(out))
+            # keep string bodies ASCII so decoded length == raw length
+            pool = sorted(allow & ascii_body) or sorted(allow)
+            b = rng.choice(pool)
+            m.feed(b)
+            out.append(b)
+        assert eos, (seed, bytes(out))
+        j = json.loads(out.decode())
+        assert len(j["name"]) <= 6
+        assert 3 <= len(j["tag"]) <= 5
