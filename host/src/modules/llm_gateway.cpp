@@ -864,6 +864,13 @@ Json LlmGatewayModule::run_chat_blocking(const Json& body,
         break;
       } else if (ev == "error") {
         mux->remove(rid);
+        // worker-declared request errors (e.g. prompt exceeds the
+        // model's context window) are the CLIENT's fault, not a
+        // provider failure
+        if (msg->at("code").as_string("") == "validation_error")
+          throw Problem{400, "Bad Request", "about:blank",
+                        msg->at("message").as_string(),
+                        "validation_error"};
         throw Problem{502, "Bad Gateway", "about:blank",
                       msg->at("message").as_string(), "provider_error"};
       }

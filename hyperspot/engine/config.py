@@ -152,6 +152,9 @@ class SamplingParams:
     # object skeleton (known keys, declared value types) is forced
     # byte-exactly (guided.SchemaMachine); overrides response_format
     response_schema: Optional[dict] = None
+    # declared tool names: a forced tool call ("tool_call" format)
+    # narrows the emitted name to one of these (guided.ToolCallMachine)
+    tool_names: tuple = ()
 
     @property
     def greedy(self) -> bool:

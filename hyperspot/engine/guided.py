@@ -212,11 +212,19 @@ class ToolCallMachine:
     _PRE = b'{"name":"'
     _MID = b',"arguments":{'
 
-    def __init__(self) -> None:
+    def __init__(self, tool_names: Tuple[str, ...] = ()) -> None:
+        import json as _json
         self.m = JsonByteMachine()
         self.queue: List[int] = list(self._PRE)
         self.phase = 0          # 0 pre, 1 name, 2 mid, 3 args, 4 done
         self.consumed = 0
+        # with declared tools the NAME is an enum: json-escaped bodies,
+        # narrowed byte-by-byte (quotes can't appear raw in a body, so
+        # close-vs-continue is never ambiguous)
+        self.alts: Tuple[bytes, ...] = tuple(
+            _json.dumps(n)[1:-1].encode() for n in tool_names
+            if isinstance(n, str) and n)
+        self.npos = 0
 
     @property
     def done(self) -> bool:
@@ -227,6 +235,12 @@ class ToolCallMachine:
             return {self.queue[0]}, False
         if self.phase == 4:
             return set(), True
+        if self.phase == 1 and self.alts:
+            out = {a[self.npos] for a in self.alts
+                   if len(a) > self.npos}
+            if any(len(a) == self.npos for a in self.alts):
+                out.add(0x22)                   # close: full name match
+            return out, False
         allow, _ = self.m.allowed()
         return allow, False
 
@@ -237,22 +251,36 @@ class ToolCallMachine:
                              f"{self.phase}")
         if self.queue:
             self.queue.pop(0)
-        self.m.feed(b)
-        if self.queue:
+            self.m.feed(b)
+            if self.queue:
+                return
+            if self.phase == 0:
+                self.phase = 1                  # tool name string
+            elif self.phase == 2:
+                self.phase = 3                  # free: arguments object
+            elif self.phase == 9:
+                self.phase = 4
             return
-        if self.phase == 0:
-            self.phase = 1                      # free: tool name string
-        elif self.phase == 1 and self.m.mode == "end":
-            self.phase = 2                      # name closed
-            self.queue = list(self._MID)
-        elif self.phase == 2:
-            self.phase = 3                      # free: arguments object
-        elif self.phase == 3 and len(self.m.stack) == 1 \
+        self.m.feed(b)
+        if self.phase == 1:
+            if self.alts:
+                if b == 0x22:                   # closed on a full match
+                    self.phase = 2
+                    self.queue = list(self._MID)
+                    return
+                self.alts = tuple(a for a in self.alts
+                                  if len(a) > self.npos
+                                  and a[self.npos] == b)
+                self.npos += 1
+                return
+            if self.m.mode == "end":            # free-form name closed
+                self.phase = 2
+                self.queue = list(self._MID)
+            return
+        if self.phase == 3 and len(self.m.stack) == 1 \
                 and self.m.mode == "end":
             self.queue = [ord("}")]             # close the outer object
-            self.phase = 9                      # transient: closing
-        elif self.phase == 9:
-            self.phase = 4
+            self.phase = 9
 
     def feed_token(self, token_id: int) -> None:
         self.consumed += 1

@@ -63,6 +63,13 @@ class ModelRunner:
     def __init__(self, config: EngineConfig, device: Optional[str] = None):
         self.config = config
         self.spec = config.spec()
+        if config.max_model_len > self.spec.max_position:
+            # the rope table is sized for max_position; admitting longer
+            # prompts crashed the stepping thread with an index OOB
+            log.warning("max_model_len %d clamped to the model's "
+                        "max_position %d", config.max_model_len,
+                        self.spec.max_position)
+            config.max_model_len = self.spec.max_position
         self.device = torch.device(
             device if device is not None
             else ("cuda" if torch.cuda.is_available() else "cpu"))
@@ -408,7 +415,7 @@ class ModelRunner:
             m = getattr(r, "_guided", None)
             if m is None or m.consumed > len(r.output_token_ids):
                 if r.sampling.response_format == "tool_call":
-                    m = ToolCallMachine()
+                    m = ToolCallMachine(tuple(r.sampling.tool_names))
                 elif r.sampling.response_schema is not None:
                     try:
                         # untrusted schema: any compile failure degrades

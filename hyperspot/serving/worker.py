@@ -215,6 +215,19 @@ class WorkerState:
         raise ValueError(kind)
 
     def step_loop(self):
+        try:
+            self._step_loop()
+        except Exception:
+            # a dead stepping thread would HANG every open request with
+            # the process still alive — invisible to the gateway's
+            # waitpid watchdog.  Die loudly instead: the gateway reaps
+            # the process, respawns a clean worker, and in-flight
+            # requests fail fast / retry on the replacement.
+            log.exception("stepping thread crashed; exiting worker for "
+                          "watchdog respawn")
+            os._exit(17)
+
+    def _step_loop(self):
         import torch.distributed as dist
         while not self.stop:
             if self.bench_req is not None:
@@ -465,6 +478,7 @@ def _mux_chat(msg, state: "WorkerState", mux: "MuxChannel", rids: set):
     budget = state.engine.config.max_model_len - len(prompt_ids) - 1
     if budget <= 0:
         mux.send_obj({"event": "error", "id": rid,
+                      "code": "validation_error",
                       "message": "prompt exceeds context window"})
         return
     sampling = SamplingParams(
@@ -477,6 +491,11 @@ def _mux_chat(msg, state: "WorkerState", mux: "MuxChannel", rids: set):
         response_format=params.get("response_format"),
         response_schema=params.get("response_schema")
         if isinstance(params.get("response_schema"), dict) else None,
+        tool_names=tuple(
+            (t.get("name") or t.get("function", {}).get("name"))
+            for t in (msg.get("tools") or [])
+            if isinstance(t, dict)
+            and (t.get("name") or t.get("function", {}).get("name"))),
     )
     rids.add(rid)
     state.submit_mux(rid, prompt_ids, sampling, mux)
@@ -638,6 +657,7 @@ def _run_chat(msg, state: WorkerState, send):
     budget = max_model_len - len(prompt_ids) - 1
     if budget <= 0:
         send({"event": "error", "id": rid,
+              "code": "validation_error",
               "message": "prompt exceeds context window"})
         return
     sampling = SamplingParams(
@@ -650,6 +670,11 @@ def _run_chat(msg, state: WorkerState, send):
         response_format=params.get("response_format"),
         response_schema=params.get("response_schema")
         if isinstance(params.get("response_schema"), dict) else None,
+        tool_names=tuple(
+            (t.get("name") or t.get("function", {}).get("name"))
+            for t in (msg.get("tools") or [])
+            if isinstance(t, dict)
+            and (t.get("name") or t.get("function", {}).get("name"))),
     )
     q = state.submit(rid, prompt_ids, sampling)
     detok = StreamDetokenizer(state.tokenizer)

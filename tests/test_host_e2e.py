@@ -1690,7 +1690,7 @@ def test_chat_tool_calling_passthrough(server):
     for seed in (3, 4, 5, 7, 11, 13):
         st, body = _http("POST", base, body={
             "model": "tiny-llama", "tools": tools,
-            "tool_choice": "required", "max_tokens": 900,
+            "tool_choice": "required", "max_tokens": 150,
             "temperature": 1.0, "seed": seed,
             "messages": [{"role": "user", "content":
                           [{"type": "text", "text": "call the tool"}]}]})
@@ -1700,6 +1700,8 @@ def test_chat_tool_calling_passthrough(server):
             got = j
             break
     assert got is not None, "no seed completed a forced tool call"
+    # the declared tools CONSTRAIN the emitted name (ToolCallMachine enum)
+    assert got["content"][0]["tool_call"]["name"] == "lookup"
     call = got["content"][0]["tool_call"]
     assert call["id"].startswith("call-")
     assert isinstance(call["name"], str)
@@ -1738,6 +1740,21 @@ def test_chat_tool_calling_passthrough(server):
                       [{"type": "image", "image": {}}]}]})
     assert st == 400
     assert json.loads(body)["code"] == "capability_not_supported"
+
+    # a conversation exceeding the model's context window is the
+    # CLIENT's error (400 validation_error), not a provider failure —
+    # and the worker survives it (clamped rope table, no crash)
+    st, body = _http("POST", base, body={
+        "model": "tiny-llama", "max_tokens": 4,
+        "messages": [{"role": "user", "content":
+                      [{"type": "text", "text": "x" * 5000}]}]})
+    assert st == 400, body
+    assert json.loads(body)["code"] == "validation_error"
+    st, body = _http("POST", base, body={
+        "model": "tiny-llama", "max_tokens": 4, "temperature": 0.0,
+        "messages": [{"role": "user", "content":
+                      [{"type": "text", "text": "still alive?"}]}]})
+    assert st == 200, body
 
 
 def test_chat_schema_shaped_output(server):
