@@ -324,6 +324,11 @@ class SchemaMachine:
         self.free_depth = -1  # stack depth at free-region start (-1: n/a)
         self.free_started = False
         self.free_len = 0     # bytes fed into the current free region
+        # arr_items runtime state (one active at a time per machine;
+        # nesting lives in the CHILD machines)
+        self.arr_open = False
+        self.arr_child: object = None
+        self.arr_after_comma = False
         self.consumed = 0
 
     # ---- schema -> script ----
@@ -367,12 +372,16 @@ class SchemaMachine:
                 self.script.append(("free", (None, None)))
         elif typ == "array":
             items = sch.get("items")
-            ist = None
-            if isinstance(items, dict) and \
-                    items.get("type") in self._TYPE_START:
-                ist = self._TYPE_START[items["type"]]
-            self.script.append(
-                ("free", (self._TYPE_START["array"], "array", ist)))
+            if isinstance(items, dict) and items and \
+                    depth < self.MAX_DEPTH:
+                # full per-element enforcement: each element runs a
+                # child SchemaMachine over the item schema (recursion
+                # via composition — arrays of objects of arrays work)
+                self.script.append(("arr_items", items))
+            else:
+                self.script.append(
+                    ("free", (self._TYPE_START["array"], "array",
+                              None)))
         elif typ == "string":
             lo = sch.get("minLength")
             hi = sch.get("maxLength")
@@ -411,6 +420,18 @@ class SchemaMachine:
             # no longer match the consumed prefix are out
             return {a[self.lit_pos] for a in arg
                     if len(a) > self.lit_pos}, False
+        if kind == "arr_items":
+            if not self.arr_open:
+                return {ord("[")}, False
+            if self.arr_child is None:
+                start, _ = SchemaMachine(arg).allowed()
+                if not self.arr_after_comma:
+                    start = set(start) | {ord("]")}
+                return start, False
+            callow, ceos = self.arr_child.allowed()
+            if ceos:
+                return set(callow) | {ord(","), ord("]")}, False
+            return callow, False
         # free region
         start_set, typ = arg[0], arg[1]
         inner, _ = self.m.allowed()
@@ -480,6 +501,35 @@ class SchemaMachine:
                 self.lit_pos = 0
             else:
                 self.script[self.seg] = ("choice", alts)
+            return
+        if kind == "arr_items":
+            if not self.arr_open:                  # the opening '['
+                self.m.feed(b)
+                self.arr_open = True
+                self.arr_after_comma = False
+                return
+            if self.arr_child is None:
+                if b == ord("]"):
+                    self.m.feed(b)
+                    self.arr_open = False
+                    self.seg += 1
+                    return
+                self.arr_child = SchemaMachine(arg)
+                self.arr_child.feed(b)
+                self.m.feed(b)
+                return
+            callow, ceos = self.arr_child.allowed()
+            if ceos and b in (ord(","), ord("]")) and b not in callow:
+                self.m.feed(b)
+                self.arr_child = None
+                if b == ord("]"):
+                    self.arr_open = False
+                    self.seg += 1
+                else:
+                    self.arr_after_comma = True
+                return
+            self.arr_child.feed(b)
+            self.m.feed(b)
             return
         # free region
         if self._free_complete():

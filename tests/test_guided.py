@@ -293,6 +293,10 @@ def test_schema_enum_choice():
         st.fixed_dictionaries({"type": st.sampled_from(
             ["string", "integer", "number", "boolean", "null",
              "array", "object", "bogus"])}),
+        st.fixed_dictionaries({
+            "type": st.just("array"),
+            "items": st.fixed_dictionaries({"type": st.sampled_from(
+                ["string", "integer", "boolean", "object"])})}),
         st.fixed_dictionaries({"enum": st.lists(
             st.one_of(st.text(max_size=5), st.integers(-99, 99),
                       st.booleans()), max_size=4)}),
@@ -360,3 +364,45 @@ def test_schema_string_length_bounds():
         j = json.loads(out.decode())
         assert len(j["name"]) <= 6
         assert 3 <= len(j["tag"]) <= 5
+
+
+def test_schema_array_of_objects_elements_enforced():
+    """Typed array items run a CHILD machine per element: every element
+    carries the full item skeleton (the common extraction shape)."""
+    from hyperspot.engine.guided import SchemaMachine
+    schema = {"type": "object", "required": ["people"],
+              "properties": {"people": {
+                  "type": "array",
+                  "items": {"type": "object",
+                            "required": ["name", "age"],
+                            "properties": {
+                                "name": {"type": "string",
+                                         "maxLength": 5},
+                                "age": {"type": "integer"}}}}}}
+    saw_multi = False
+    for seed in range(80):
+        rng = random.Random(seed)
+        m = SchemaMachine(schema)
+        out = bytearray()
+        eos = False
+        for _ in range(600):
+            allow, eos = m.allowed()
+            if eos:
+                break
+            assert allow, (seed, bytes(out))
+            pool = sorted(allow)
+            if len(out) > 40:
+                pref = [b for b in pool if b in b'"]}0123456789']
+                if pref and rng.random() < 0.9:
+                    pool = pref
+            b = rng.choice(pool)
+            m.feed(b)
+            out.append(b)
+        assert eos, (seed, bytes(out))
+        j = json.loads(out.decode("utf-8", errors="replace"))
+        if len(j["people"]) > 1:
+            saw_multi = True
+        for pers in j["people"]:
+            assert set(pers) == {"name", "age"}
+            assert isinstance(pers["age"], int)
+    assert saw_multi            # commas + repeated skeletons exercised
