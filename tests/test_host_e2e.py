@@ -1865,3 +1865,38 @@ def test_declared_request_schema_enforced(server):
                      "/v1/chat/completions", body={"model": "tiny-llama"})
     assert st == 400
     assert "required" in json.loads(body)["detail"]
+
+
+def test_chat_schema_array_extraction(server):
+    """The common extraction shape through the full stack: array of
+    typed objects — every element carries the schema skeleton."""
+    st, body = _http("POST", BASE.format(server.port) +
+                     "/v1/chat/completions",
+                     body={"model": "tiny-llama",
+                           "messages": [{"role": "user", "content":
+                                         [{"type": "text",
+                                           "text": "list people"}]}],
+                           "response_schema": {
+                               "type": "object",
+                               "required": ["people"],
+                               "properties": {"people": {
+                                   "type": "array",
+                                   "items": {
+                                       "type": "object",
+                                       "required": ["name", "age"],
+                                       "properties": {
+                                           "name": {"type": "string",
+                                                    "maxLength": 8},
+                                           "age": {"type":
+                                                   "integer"}}}}}},
+                           "max_tokens": 400, "temperature": 1.0,
+                           "seed": 9})
+    assert st == 200, body
+    j = json.loads(body)
+    text = "".join(p["text"] for p in j["content"]
+                   if p["type"] == "text")
+    out = json.loads(text)
+    assert isinstance(out["people"], list)
+    for pers in out["people"]:
+        assert set(pers) == {"name", "age"}
+        assert isinstance(pers["age"], int)
