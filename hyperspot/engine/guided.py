@@ -193,6 +193,15 @@ class JsonByteMachine:
             self.stack.pop()
             self.mode = "end"
 
+    def mask_key(self):
+        """Hashable signature of everything allowed() depends on — lets
+        the sampler cache the [260]-wide logit mask per distinct state
+        instead of looping ~230 bytes in Python per sequence per step
+        (at 2048 guided streams that loop would cost milliseconds).
+        Must stay in lockstep with allowed()."""
+        top = self.stack[-1] if self.stack else None
+        return ("J", self.mode, top, not self.stack, bytes(self.lit))
+
     # ---- token-level wrapper (ByteTokenizer: byte b <-> id b+4) ----
 
     def feed_token(self, token_id: int) -> None:

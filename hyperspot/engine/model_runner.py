@@ -107,6 +107,8 @@ class ModelRunner:
         self._st_lens = torch.empty(cap, dtype=torch.int32, pin_memory=pin)
         self._st_bt_flat = torch.empty(cap * self.max_blocks_per_seq,
                                        dtype=torch.int32, pin_memory=pin)
+        # guided-decoding logit-mask rows keyed by machine state
+        self._guided_mask_cache: Dict[tuple, torch.Tensor] = {}
         self._graphs: Dict[int, torch.cuda.CUDAGraph] = {}
         self._graph_io: Dict[int, dict] = {}
         self._pgraphs: Dict[int, torch.cuda.CUDAGraph] = {}
@@ -410,6 +412,7 @@ class ModelRunner:
         NB = 4 + 256                       # specials + byte ids
         neg = float("-inf")
         small = torch.full((len(gis), NB), neg, dtype=torch.float32)
+        cache = self._guided_mask_cache
         for k, i in enumerate(gis):
             r = requests[i]
             m = getattr(r, "_guided", None)
@@ -428,6 +431,21 @@ class ModelRunner:
                 r._guided = m
             for t in r.output_token_ids[m.consumed:]:
                 m.feed_token(t)
+            # cache mask rows by machine state (mask_key): avoids a
+            # ~230-iteration Python loop per guided sequence per step
+            key = m.mask_key() if hasattr(m, "mask_key") else None
+            if key is not None:
+                row = cache.get(key)
+                if row is None:
+                    allow, eos_ok = m.allowed()
+                    row = torch.full((NB,), neg, dtype=torch.float32)
+                    for b in allow:
+                        row[b + 4] = 0.0
+                    if eos_ok:
+                        row[2] = 0.0       # ByteTokenizer EOS
+                    cache[key] = row
+                small[k] = row
+                continue
             allow, eos_ok = m.allowed()
             for b in allow:
                 small[k, b + 4] = 0.0

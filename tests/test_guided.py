@@ -406,3 +406,22 @@ def test_schema_array_of_objects_elements_enforced():
             assert set(pers) == {"name", "age"}
             assert isinstance(pers["age"], int)
     assert saw_multi            # commas + repeated skeletons exercised
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.integers(0, 10**9))
+def test_mask_key_matches_allowed(seed):
+    """The sampler caches masks by mask_key(): two states with equal
+    keys MUST have identical allowed sets (a collision would leak bytes
+    across grammar states)."""
+    from hyperspot.engine.guided import JsonByteMachine
+    seen = {}
+    out, _ = _walk(seed, soft_limit=25, hard_limit=120)
+    m = JsonByteMachine()
+    for b in out:
+        key = m.mask_key()
+        cur = (frozenset(m.allowed()[0]), m.allowed()[1])
+        if key in seen:
+            assert seen[key] == cur, (key, out)
+        seen[key] = cur
+        m.feed(b)
