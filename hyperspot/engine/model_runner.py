@@ -411,7 +411,7 @@ class ModelRunner:
                              ToolCallMachine)
         NB = 4 + 256                       # specials + byte ids
         neg = float("-inf")
-        small = torch.full((len(gis), NB), neg, dtype=torch.float32)
+        rows: List[torch.Tensor] = []
         cache = self._guided_mask_cache
         for k, i in enumerate(gis):
             r = requests[i]
@@ -434,23 +434,18 @@ class ModelRunner:
             # cache mask rows by machine state (mask_key): avoids a
             # ~230-iteration Python loop per guided sequence per step
             key = m.mask_key() if hasattr(m, "mask_key") else None
-            if key is not None:
-                row = cache.get(key)
-                if row is None:
-                    allow, eos_ok = m.allowed()
-                    row = torch.full((NB,), neg, dtype=torch.float32)
-                    for b in allow:
-                        row[b + 4] = 0.0
-                    if eos_ok:
-                        row[2] = 0.0       # ByteTokenizer EOS
+            row = cache.get(key) if key is not None else None
+            if row is None:
+                allow, eos_ok = m.allowed()
+                row = torch.full((NB,), neg, dtype=torch.float32)
+                for b in allow:
+                    row[b + 4] = 0.0
+                if eos_ok:
+                    row[2] = 0.0           # ByteTokenizer EOS
+                if key is not None:
                     cache[key] = row
-                small[k] = row
-                continue
-            allow, eos_ok = m.allowed()
-            for b in allow:
-                small[k, b + 4] = 0.0
-            if eos_ok:
-                small[k, 2] = 0.0          # ByteTokenizer EOS
+            rows.append(row)
+        small = torch.stack(rows)
         d = logits.device
         gidx = torch.tensor(gis, device=d)
         logits[gidx, NB:] = neg
