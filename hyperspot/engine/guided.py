@@ -338,6 +338,7 @@ class SchemaMachine:
         self.arr_open = False
         self.arr_child: object = None
         self.arr_after_comma = False
+        self.bint = None
         self.consumed = 0
 
     # ---- schema -> script ----
@@ -399,6 +400,14 @@ class SchemaMachine:
             self.script.append(
                 ("free", (self._TYPE_START["string"], "string",
                           None, lo, hi)))
+        elif typ == "integer" \
+                and isinstance(sch.get("minimum"), int) \
+                and isinstance(sch.get("maximum"), int) \
+                and 0 <= sch["minimum"] <= sch["maximum"]:
+            # exact digit-wise range enforcement (non-negative bounds;
+            # other range shapes fall through to start-typed integers)
+            self.script.append(
+                ("bint", (sch["minimum"], sch["maximum"])))
         elif typ in self._TYPE_START:
             self.script.append(("free", (self._TYPE_START[typ], typ)))
         else:
@@ -429,6 +438,17 @@ class SchemaMachine:
             # no longer match the consumed prefix are out
             return {a[self.lit_pos] for a in arg
                     if len(a) > self.lit_pos}, False
+        if kind == "bint":
+            if self.bint is None:
+                self.bint = BoundedIntValue(*arg)
+            callow, complete = self.bint.allowed()
+            if complete:
+                nxt = self.seg + 1
+                if nxt >= len(self.script):
+                    return callow, True
+                _, narg = self.script[nxt]
+                return set(callow) | {narg[0]}, False
+            return callow, False
         if kind == "arr_items":
             if not self.arr_open:
                 return {ord("[")}, False
@@ -511,6 +531,24 @@ class SchemaMachine:
             else:
                 self.script[self.seg] = ("choice", alts)
             return
+        if kind == "bint":
+            if self.bint is None:
+                self.bint = BoundedIntValue(*arg)
+            callow, complete = self.bint.allowed()
+            if complete and b not in callow:       # the next literal
+                self.m.feed(b)
+                self.bint = None
+                nxt = self.seg + 1
+                _, narg = self.script[nxt]
+                self.seg = nxt
+                self.lit_pos = 1
+                if self.lit_pos >= len(narg):
+                    self.seg += 1
+                    self.lit_pos = 0
+                return
+            self.bint.feed(b)
+            self.m.feed(b)
+            return
         if kind == "arr_items":
             if not self.arr_open:                  # the opening '['
                 self.m.feed(b)
@@ -570,3 +608,54 @@ class SchemaMachine:
             return
         if 4 <= token_id < 4 + 256:
             self.feed(token_id - 4)
+
+
+class BoundedIntValue:
+    """A single non-negative JSON integer constrained to [lo, hi],
+    enforced digit-by-digit: a digit is allowed only if SOME completion
+    stays in range, and stopping is allowed exactly when the digits so
+    far ARE in range (used by SchemaMachine for integer minimum/maximum
+    when both bounds are >= 0)."""
+
+    def __init__(self, lo: int, hi: int) -> None:
+        assert 0 <= lo <= hi
+        self.lo, self.hi = lo, hi
+        self.cur = 0
+        self.started = False
+
+    def _viable(self, p: int) -> bool:
+        """Can p followed by j >= 0 more digits land in [lo, hi]?"""
+        span = 1           # 10**j
+        while True:
+            low_j = p * span
+            if low_j > self.hi:
+                return False
+            if low_j + span - 1 >= self.lo:
+                return True
+            span *= 10
+
+    @property
+    def complete(self) -> bool:
+        return self.started and self.lo <= self.cur <= self.hi
+
+    def allowed(self) -> Tuple[Set[int], bool]:
+        out: Set[int] = set()
+        if not self.started:
+            if self.lo == 0:
+                out.add(ord("0"))          # "0" is complete, "0x" illegal
+            for d in range(1, 10):
+                if self._viable(d):
+                    out.add(ord("0") + d)
+            return out, False
+        if self.cur != 0:                  # no digits after a lone "0"
+            for d in range(10):
+                if self._viable(self.cur * 10 + d):
+                    out.add(ord("0") + d)
+        return out, self.complete
+
+    def feed(self, b: int) -> None:
+        ok, _ = self.allowed()
+        if b not in ok:
+            raise ValueError(f"digit {b!r} breaks [{self.lo},{self.hi}]")
+        self.started = True
+        self.cur = self.cur * 10 + (b - ord("0"))

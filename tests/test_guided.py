@@ -297,6 +297,9 @@ def test_schema_enum_choice():
             "type": st.just("array"),
             "items": st.fixed_dictionaries({"type": st.sampled_from(
                 ["string", "integer", "boolean", "object"])})}),
+        st.builds(lambda lo, span: {"type": "integer", "minimum": lo,
+                                    "maximum": lo + span},
+                  st.integers(0, 10**6), st.integers(0, 10**6)),
         st.fixed_dictionaries({"enum": st.lists(
             st.one_of(st.text(max_size=5), st.integers(-99, 99),
                       st.booleans()), max_size=4)}),
@@ -425,3 +428,49 @@ def test_mask_key_matches_allowed(seed):
             assert seen[key] == cur, (key, out)
         seen[key] = cur
         m.feed(b)
+
+
+def test_schema_bounded_integer_range():
+    """integer minimum/maximum (non-negative) enforced digit-wise:
+    every completion is in range, boundaries are reachable, and
+    out-of-range digits are masked."""
+    from hyperspot.engine.guided import BoundedIntValue, SchemaMachine
+    schema = {"type": "object", "required": ["a"],
+              "properties": {"a": {"type": "integer", "minimum": 17,
+                                   "maximum": 4203}}}
+    for seed in range(120):
+        rng = random.Random(seed)
+        m = SchemaMachine(schema)
+        out = bytearray()
+        eos = False
+        for _ in range(40):
+            allow, eos = m.allowed()
+            if eos:
+                break
+            b = rng.choice(sorted(allow))
+            m.feed(b)
+            out.append(b)
+        assert eos
+        assert 17 <= json.loads(out.decode())["a"] <= 4203
+    # exact boundaries complete
+    for val in ("17", "4203", "100"):
+        m = SchemaMachine(schema)
+        for b in b'{"a":' + val.encode():
+            m.feed(b)
+        for b in b"}":
+            m.feed(b)
+        assert m.allowed()[1]
+    # digit that can no longer reach the range is masked
+    v = BoundedIntValue(17, 4203)
+    v.feed(ord("4"))
+    v.feed(ord("2"))
+    v.feed(ord("1"))            # 421 is in range; 421x never is
+    allow, complete = v.allowed()
+    assert complete and not allow          # must stop here
+    v2 = BoundedIntValue(17, 4203)
+    with pytest.raises(ValueError):
+        v2.feed(ord("0"))                  # leading zero: 0 < 17
+    v3 = BoundedIntValue(17, 4203)
+    v3.feed(ord("5"))                      # 5 -> 5x..5xxx reachable
+    assert ord("9") in v3.allowed()[0]     # 59 in range
+    assert not v3.allowed()[1]             # 5 itself is below minimum
