@@ -474,8 +474,13 @@ class SchemaMachine:
             lo, hi = arg[3], arg[4]
             if hi is not None and body >= hi:
                 inner = inner & {0x22}            # must close now
-            elif lo is not None and body < lo:
-                inner = inner - {0x22}            # may not close yet
+            else:
+                if hi is not None and body + 6 > hi:
+                    # no room for a worst-case escape (\uXXXX = 6 raw
+                    # bytes) — the byte bound stays hard
+                    inner = inner - {0x5C}
+                if lo is not None and body < lo:
+                    inner = inner - {0x22}        # may not close yet
         if typ == "array" and len(arg) > 2 and arg[2] is not None \
                 and self.free_started \
                 and len(self.m.stack) == self.free_depth + 1 \
