@@ -271,3 +271,62 @@ def test_ep2_moe_worker_matches_ep1():
                 proc.kill()
                 proc.wait()
     assert out[1] == out[2], out
+
+
+def test_tp2_guided_schema_over_worker_protocol():
+    """response_schema travels through the worker op-log broadcast
+    (sampling params pickled to follower ranks): tp=2 produces the same
+    schema-shaped JSON as tp=1 and every token obeys the mask."""
+    import json as _json
+    schema = {"type": "object", "required": ["n", "ok"],
+              "properties": {"n": {"type": "integer", "minimum": 0,
+                                   "maximum": 99},
+                             "ok": {"type": "boolean"}}}
+
+    def chat_guided(sock_path):
+        s = socketlib.socket(socketlib.AF_UNIX, socketlib.SOCK_STREAM)
+        s.connect(sock_path)
+        f = s.makefile("rwb")
+        f.write((_json.dumps({
+            "type": "chat", "id": "g1",
+            "messages": [{"role": "user",
+                          "content": [{"type": "text",
+                                       "text": "status json"}]}],
+            "params": {"temperature": 1.0, "seed": 5,
+                       "max_tokens": 120,
+                       "response_schema": schema}}) + "\n").encode())
+        f.flush()
+        toks = []
+        for raw in f:
+            m = _json.loads(raw)
+            if m["event"] == "delta":
+                toks.append(m["token_id"])
+            elif m["event"] in ("done", "error"):
+                assert m["event"] == "done", m
+                break
+        s.close()
+        return toks
+
+    out = {}
+    for tp in (1, 2):
+        sock = tempfile.mktemp(suffix=".sock", prefix=f"hs-gtp{tp}-")
+        s2 = socketlib.socket()
+        s2.bind(("127.0.0.1", 0))
+        port = s2.getsockname()[1]
+        s2.close()
+        proc = _spawn_worker(tp, sock, port)
+        try:
+            _wait_sock(sock, proc)
+            out[tp] = chat_guided(sock)
+        finally:
+            proc.terminate()
+            try:
+                proc.wait(timeout=20)
+            except subprocess.TimeoutExpired:
+                proc.kill()
+                proc.wait()
+    assert out[1] == out[2], out
+    text = bytes(t - 4 for t in out[1] if 4 <= t < 260).decode(
+        "utf-8", errors="replace")
+    j = _json.loads(text)
+    assert set(j) == {"n", "ok"} and 0 <= j["n"] <= 99
