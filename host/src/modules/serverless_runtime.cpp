@@ -949,6 +949,8 @@ void ServerlessRuntimeModule::schedule_tick() {
         {DbValue::S(sid), DbValue::S(s.at("next_run_at").as_string())});
     if (won != 1) continue;
     if (fires == 0) continue;
+    if (!tenant_policy(tenant).at("enabled").as_bool(true))
+      continue;              // runtime disabled: cadence advances, no fire
     // resolve the entrypoint; merge its input defaults with overrides
     auto ep = conn.select("entrypoints", "id=?",
                           {DbValue::S(s.at("entrypoint_id").as_string())},
@@ -2072,6 +2074,10 @@ void ServerlessRuntimeModule::register_rest(ModuleCtx& ctx,
                         "validation_error"};
         Json payload = b.at("payload");
         if (payload.is_null()) payload = Json::object();
+        if (!tenant_policy(sec.tenant_id).at("enabled").as_bool(true))
+          throw Problem{403, "Forbidden", "about:blank",
+                        "serverless runtime disabled for tenant",
+                        "runtime_disabled"};
         SecureConn conn(*db_, scope_for(sec, "invoke",
                                         "serverless-runtime:events"));
         auto page = conn.select("triggers", "status='active'", {}, "id",

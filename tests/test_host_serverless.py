@@ -654,10 +654,14 @@ def test_tenant_runtime_policy_and_quotas(sl):
                                          "value": "PT1H"}})
         assert st == 429
 
-        # disable the runtime -> invocations are 403
+        # disable the runtime -> invocations AND event publishes are 403
         st, _ = _http("PUT", pol_url, {"enabled": False})
         st, resp = _http("POST", _url(srv) + "/invocations",
                          {"entrypoint_id": ep_id, "input": {}})
+        assert st == 403 and json.loads(resp)["code"] == "runtime_disabled"
+        st, resp = _http("POST", _url(srv) + "/events",
+                         {"event_type_id": "gts.x.app.any.v1~",
+                          "payload": {}})
         assert st == 403 and json.loads(resp)["code"] == "runtime_disabled"
 
         # validation: negative quota rejected
