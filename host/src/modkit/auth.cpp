@@ -1,5 +1,7 @@
 #include "auth.h"
 
+#include <unistd.h>
+
 #include <openssl/bio.h>
 #include <openssl/bn.h>
 #include <openssl/evp.h>
@@ -166,9 +168,17 @@ std::optional<Json> http_get_json(const std::string& url,
   TlsOpts tls;
   tls.enable = https;
   tls.ca_file = ca_file;
-  auto resp = http_request(host, port, "GET", path,
-                           {{"accept", "application/json"}}, "", 5000,
-                           nullptr, nullptr, &tls);
+  // idempotent GET with bounded retry/backoff (the reference's
+  // modkit-http retry layer, src/layers/retry.rs): transient connect
+  // failures and 5xx from the IdP retry twice before surfacing
+  std::optional<ClientResponse> resp;
+  for (int attempt = 0; attempt < 3; ++attempt) {
+    if (attempt) usleep(100000u << (attempt - 1));   // 100ms, 200ms
+    resp = http_request(host, port, "GET", path,
+                        {{"accept", "application/json"}}, "", 5000,
+                        nullptr, nullptr, &tls);
+    if (resp && resp->status < 500) break;
+  }
   if (!resp || resp->status != 200) {
     if (err)
       *err = "fetch failed: " + url + " status " +

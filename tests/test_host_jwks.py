@@ -189,3 +189,69 @@ def test_unknown_kid_rejected(jwks_env):
     tok = k1.sign_jwt("ghost", _claims(idp))
     st, body = _http("GET", BASE.format(srv.port) + AUTH_ROUTE, token=tok)
     assert st == 401, body
+
+
+def test_jwks_fetch_retries_transient_5xx():
+    """The outbound GET retries transient IdP failures (modkit-http
+    retry-layer analog): a JWKS endpoint that 500s twice then recovers
+    still authenticates the first request."""
+    k = RsaKey()
+    state = {"fails": 2}
+
+    class FlakyH(BaseHTTPRequestHandler):
+        def do_GET(self):
+            if state["fails"] > 0:
+                state["fails"] -= 1
+                self.send_response(500)
+                self.send_header("content-length", "0")
+                self.end_headers()
+                return
+            body = json.dumps({"keys": [k.jwk("k1")]}).encode()
+            self.send_response(200)
+            self.send_header("content-type", "application/json")
+            self.send_header("content-length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+        def log_message(self, *a):
+            pass
+
+    idp_srv = HTTPServer(("127.0.0.1", _free_port()), FlakyH)
+    threading.Thread(target=idp_srv.serve_forever, daemon=True).start()
+    idp_port = idp_srv.server_address[1]
+    port = _free_port()
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(f"""
+server:
+  home_dir: "/tmp/hs-e2e-jwks-flaky"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: false
+  authn-resolver:
+    config:
+      jwt:
+        jwks_uri: "http://127.0.0.1:{idp_port}/jwks"
+  llm-gateway:
+    config:
+      model: "tiny-llama"
+      auto_start_worker: false
+""")
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        now = int(time.time())
+        tok = k.sign_jwt("k1", {
+            "sub": "u", "iat": now, "exp": now + 300,
+            "tid": "00000000-df51-5b42-9538-d2b56b7ee953"})
+        st, body = _http("GET", BASE.format(port) + AUTH_ROUTE, token=tok)
+        assert st == 200, body
+        assert state["fails"] == 0       # both 500s were consumed
+    finally:
+        srv.stop()
+        idp_srv.shutdown()
+        cfg_path.unlink(missing_ok=True)
