@@ -475,6 +475,11 @@ def _mux_chat(msg, state: "WorkerState", mux: "MuxChannel", rids: set):
         text = render_chat(msg.get("messages") or [],
                            tools=msg.get("tools"))
         prompt_ids = state.tokenizer.encode(text, add_bos=True)
+    if not prompt_ids:
+        mux.send_obj({"event": "error", "id": rid,
+                      "code": "validation_error",
+                      "message": "empty prompt"})
+        return
     budget = state.engine.config.max_model_len - len(prompt_ids) - 1
     if budget <= 0:
         mux.send_obj({"event": "error", "id": rid,
@@ -654,6 +659,10 @@ def _run_chat(msg, state: WorkerState, send):
                            tools=msg.get("tools"))
         prompt_ids = state.tokenizer.encode(text, add_bos=True)
     max_model_len = state.engine.config.max_model_len
+    if not prompt_ids:
+        send({"event": "error", "id": rid,
+              "code": "validation_error", "message": "empty prompt"})
+        return
     budget = max_model_len - len(prompt_ids) - 1
     if budget <= 0:
         send({"event": "error", "id": rid,
