@@ -742,6 +742,55 @@ def test_schedule_survives_restart_catch_up(tmp_path):
     cfg_path.unlink(missing_ok=True)
 
 
+def test_schedule_backfill_fires_per_missed_slot(tmp_path):
+    """missed_policy=backfill executes ONCE PER missed occurrence after
+    downtime (vs catch_up's single fire)."""
+    import tempfile
+    home = tempfile.mkdtemp(prefix="hs-bf-")
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-bf-")
+    port = _free_port()
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(_mk_cfg(port, sock, home))
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        ep_id = _mk_active_ep(srv, {**json.loads(json.dumps(FN)),
+                                    "name": "fn-bf"})
+        st, resp = _http("POST", _url(srv) + "/schedules",
+                         {"name": "bf", "entrypoint_id": ep_id,
+                          "expression": {"kind": "interval",
+                                         "value": "PT1S"},
+                          "missed_policy": "backfill"})
+        assert st == 201, resp
+        sid = json.loads(resp)["schedule_id"]
+    finally:
+        srv.stop()
+    time.sleep(4.5)     # ~4 missed 1-second slots while down
+    port2 = _free_port()
+    sock2 = tempfile.mktemp(suffix=".sock", prefix="hs-bf2-")
+    cfg2 = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg2.write_text(_mk_cfg(port2, sock2, home))
+    srv2 = ServerProc(cfg2, port2)
+    try:
+        srv2.wait_ready()
+
+        def history():
+            st, resp = _http("GET",
+                             _url(srv2) + f"/schedules/{sid}/history")
+            assert st == 200, resp
+            return json.loads(resp)["items"]
+
+        # the FIRST tick after recovery backfills every missed slot
+        t0 = time.time()
+        while time.time() - t0 < 10 and len(history()) < 3:
+            time.sleep(0.2)
+        assert len(history()) >= 3, history()
+    finally:
+        srv2.stop()
+        cfg2.unlink(missing_ok=True)
+    cfg_path.unlink(missing_ok=True)
+
+
 def test_durable_recovery_across_restart(sl):
     """Queued work survives a host restart (PRD.md:44-45 RTO/RPO)."""
     srv, home, cfg_path = sl
