@@ -356,6 +356,8 @@ Json row_invocation(const DbRow& r) {
   j["entrypoint_id"] = r.at("entrypoint_id");
   j["status"] = r.at("status");
   j["mode"] = r.at("mode");
+  const std::string sid = r.at("schedule_id").as_string();
+  if (!sid.empty()) j["schedule_id"] = sid;
   j["attempts"] = r.at("attempts");
   j["created_at"] = r.at("created_at");
   j["updated_at"] = r.at("updated_at");
@@ -1117,6 +1119,7 @@ void ServerlessRuntimeModule::register_rest(ModuleCtx& ctx,
                                               "entrypoint_type"};
   const std::vector<std::string> inv_fields = {"id", "entrypoint_id",
                                                "status", "mode",
+                                               "schedule_id",
                                                "created_at"};
 
   auto reg = [&](const char* method, const std::string& path,
