@@ -368,3 +368,24 @@ def test_circuit_breaker_opens_and_recovers(server):
         assert st == 200                    # closed: normal traffic
     finally:
         srv.shutdown()
+
+
+def test_circuit_breaker_disabled_by_zero_threshold(server):
+    """`circuit.failure_threshold: 0` disables the breaker: failures
+    keep passing through as 502s, never fast-fail 503."""
+    import socket as socketlib
+    # a port with nothing listening -> connect failures
+    s = socketlib.socket()
+    s.bind(("127.0.0.1", 0))
+    dead_port = s.getsockname()[1]
+    s.close()
+    base = BASE.format(server.port)
+    up = {"alias": "nobreak", "server": {"endpoints": [
+        {"scheme": "http", "host": "127.0.0.1", "port": dead_port}]},
+        "circuit": {"failure_threshold": 0}}
+    st, body = _http("POST", base + "/oagw/v1/upstreams", body=up)
+    assert st == 201, body
+    for _ in range(6):
+        st, body = _http("GET", base + "/oagw/v1/proxy/nobreak/x")
+        assert st == 502, (st, body)
+        assert json.loads(body)["code"] == "provider_error"
