@@ -1900,3 +1900,73 @@ def test_chat_schema_array_extraction(server):
     for pers in out["people"]:
         assert set(pers) == {"name", "age"}
         assert isinstance(pers["age"], int)
+
+
+def test_chat_per_tenant_admission_limit(tmp_path):
+    """Noisy-neighbor NFR: the serverless-runtime admission plane rate-
+    limits chat per tenant — burst exhausted => 429 rate_limited."""
+    import tempfile
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-adm-")
+    port = _free_port()
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(f"""
+server:
+  home_dir: "/tmp/hs-e2e-adm"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+  serverless-runtime:
+    config:
+      limits:
+        rps_per_tenant: 1
+        burst_per_tenant: 2
+        max_concurrent_per_tenant: 64
+  llm-gateway:
+    config:
+      model: "tiny-llama"
+      worker_socket: "{sock}"
+      auto_start_worker: true
+      worker:
+        device: "cpu"
+        eager: true
+        max_num_seqs: 8
+        num_gpu_blocks: 256
+""")
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        srv.wait_worker()
+        codes = []
+        for _ in range(6):
+            st, body = _http("POST", BASE.format(port) +
+                             "/v1/chat/completions",
+                             body={"model": "tiny-llama",
+                                   "max_tokens": 2, "temperature": 0.0,
+                                   "messages": [{"role": "user",
+                                                 "content":
+                                                 [{"type": "text",
+                                                   "text": "x"}]}]})
+            codes.append(st)
+        assert codes.count(200) >= 2          # the burst served
+        assert 429 in codes                   # then rate-limited
+        limited = [c for c in codes if c == 429]
+        assert len(limited) >= 1
+        # after a pause the bucket refills
+        import time as _t
+        _t.sleep(1.5)
+        st, body = _http("POST", BASE.format(port) +
+                         "/v1/chat/completions",
+                         body={"model": "tiny-llama", "max_tokens": 2,
+                               "temperature": 0.0,
+                               "messages": [{"role": "user", "content":
+                                             [{"type": "text",
+                                               "text": "x"}]}]})
+        assert st == 200, body
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
