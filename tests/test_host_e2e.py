@@ -1970,3 +1970,57 @@ modules:
     finally:
         srv.stop()
         cfg_path.unlink(missing_ok=True)
+
+
+def test_chat_tenant_budget_exhausted(tmp_path):
+    """usage.budget_tokens_per_tenant: once consumed tokens cross the
+    budget, further chats are 429 budget_exceeded (DESIGN error list)."""
+    import tempfile
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-bud-")
+    port = _free_port()
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(f"""
+server:
+  home_dir: "/tmp/hs-e2e-bud"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+  llm-gateway:
+    config:
+      model: "tiny-llama"
+      worker_socket: "{sock}"
+      auto_start_worker: true
+      usage:
+        budget_tokens_per_tenant: 10
+      worker:
+        device: "cpu"
+        eager: true
+        max_num_seqs: 8
+        num_gpu_blocks: 256
+""")
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        srv.wait_worker()
+
+        def chat():
+            return _http("POST", BASE.format(port) +
+                         "/v1/chat/completions",
+                         body={"model": "tiny-llama", "max_tokens": 8,
+                               "temperature": 0.0,
+                               "messages": [{"role": "user", "content":
+                                             [{"type": "text",
+                                               "text": "count"}]}]})
+        st, body = chat()
+        assert st == 200, body          # first request consumes >10 toks
+        st, body = chat()
+        assert st == 429, body
+        assert json.loads(body)["code"] == "budget_exceeded"
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
