@@ -2024,3 +2024,117 @@ modules:
     finally:
         srv.stop()
         cfg_path.unlink(missing_ok=True)
+
+
+def test_hook_response_blocked(tmp_path):
+    """Post-response hook: content matching the blocklist yields 403
+    response_blocked (random-init greedy output always contains the
+    '<id>' rendering of out-of-byte tokens, so '<' triggers it
+    deterministically)."""
+    import tempfile
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-rhook-")
+    port = _free_port()
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(f"""
+server:
+  home_dir: "/tmp/hs-e2e-rhook"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+  llm-gateway:
+    config:
+      model: "tiny-llama"
+      worker_socket: "{sock}"
+      auto_start_worker: true
+      hooks:
+        blocklist: ["<"]
+      worker:
+        device: "cpu"
+        eager: true
+        max_num_seqs: 4
+        num_gpu_blocks: 128
+""")
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        srv.wait_worker()
+        st, body = _http("POST", BASE.format(port) +
+                         "/v1/chat/completions",
+                         body={"model": "tiny-llama", "max_tokens": 16,
+                               "temperature": 0.0,
+                               "messages": [{"role": "user", "content":
+                                             [{"type": "text",
+                                               "text": "hi"}]}]})
+        assert st == 403, body
+        assert json.loads(body)["code"] == "response_blocked"
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
+
+
+def test_job_result_ttl_expires(tmp_path):
+    """jobs.ttl_s: finished results past the TTL are 410 job_expired
+    (the last DESIGN error code without coverage)."""
+    import tempfile
+    import time as _t
+    sock = tempfile.mktemp(suffix=".sock", prefix="hs-ttl-")
+    port = _free_port()
+    cfg_path = Path(tempfile.mktemp(suffix=".yaml"))
+    cfg_path.write_text(f"""
+server:
+  home_dir: "/tmp/hs-e2e-ttl"
+logging:
+  default:
+    console_level: warn
+modules:
+  api-gateway:
+    config:
+      bind_addr: "127.0.0.1:{port}"
+      auth_disabled: true
+  llm-gateway:
+    config:
+      model: "tiny-llama"
+      worker_socket: "{sock}"
+      auto_start_worker: true
+      jobs:
+        ttl_s: 1
+      worker:
+        device: "cpu"
+        eager: true
+        max_num_seqs: 4
+        num_gpu_blocks: 128
+""")
+    srv = ServerProc(cfg_path, port)
+    try:
+        srv.wait_ready()
+        srv.wait_worker()
+        st, body = _http("POST", BASE.format(port) +
+                         "/v1/chat/completions",
+                         body={"model": "tiny-llama", "async": True,
+                               "max_tokens": 4, "temperature": 0.0,
+                               "messages": [{"role": "user", "content":
+                                             [{"type": "text",
+                                               "text": "hi"}]}]})
+        assert st == 202, body
+        jid = json.loads(body)["id"]
+        t0 = _t.time()
+        while _t.time() - t0 < 20:
+            st, body = _http("GET", BASE.format(port) +
+                             f"/llm-gateway/v1/jobs/{jid}")
+            if st == 200 and json.loads(body)["status"] == "succeeded":
+                break
+            _t.sleep(0.2)
+        assert st == 200 and json.loads(body)["status"] == "succeeded"
+        _t.sleep(1.5)
+        st, body = _http("GET", BASE.format(port) +
+                         f"/llm-gateway/v1/jobs/{jid}")
+        assert st == 410, body
+        assert json.loads(body)["code"] == "job_expired"
+    finally:
+        srv.stop()
+        cfg_path.unlink(missing_ok=True)
