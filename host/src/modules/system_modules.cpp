@@ -1152,6 +1152,29 @@ void FileStorageModule::register_rest(ModuleCtx& ctx, RestRegistry& rest) {
     w.respond(200, "application/octet-stream", ss.str());
   });
 
+  // metadata query (PRD "Get file metadata": pre-fetch validation of
+  // size/type without transferring the bytes)
+  OperationSpec md;
+  md.method = "GET";
+  md.path = "/file-storage/v1/metadata/{*name}";
+  md.operation_id = "files_metadata";
+  md.summary = "File metadata (size, mtime) without the body";
+  md.authenticated = true;
+  md.tags = {"file-storage"};
+  rest.register_op(md, [](HttpRequest& rq, ResponseWriter& w) {
+    std::string rel = sanitize_rel_path(rq.path_params["name"]);
+    std::string full =
+        g_files->root_for(sec_of(rq).tenant_id) + "/" + rel;
+    struct stat st{};
+    if (stat(full.c_str(), &st) != 0 || !S_ISREG(st.st_mode))
+      throw Problem::not_found();
+    Json meta = Json::object();
+    meta["name"] = rel;
+    meta["size"] = (long)st.st_size;
+    meta["modified_at"] = (double)st.st_mtime;
+    w.respond(200, "application/json", meta.dump());
+  });
+
   OperationSpec del;
   del.method = "DELETE";
   del.path = "/file-storage/v1/files/{*name}";

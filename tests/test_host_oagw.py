@@ -156,6 +156,12 @@ def test_file_storage_roundtrip(server):
     st, body = _http("GET", base +
                      "/file-storage/v1/files/ckpt/llama/model.bin")
     assert st == 200 and json.loads(body) == data
+    # metadata query (no body transfer)
+    st, body = _http("GET", base +
+                     "/file-storage/v1/metadata/ckpt/llama/model.bin")
+    assert st == 200, body
+    md = json.loads(body)
+    assert md["size"] > 900 and md["modified_at"] > 0
     st, _ = _http("DELETE", base +
                   "/file-storage/v1/files/ckpt/llama/model.bin")
     assert st == 204
