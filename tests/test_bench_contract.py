@@ -54,3 +54,21 @@ def test_bench_rest_two_rank_launch():
     assert j["n_gpus"] == 2
     assert j["config"]["parallelism"].startswith("dp2")
     assert j["value"] > 0
+
+
+def test_bench_engine_direct_json_line():
+    """--engine-direct (the rocprof/kernel-work mode) keeps the same
+    one-JSON-line contract."""
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "bench.py"),
+         "--engine-direct", "--model", "tiny-llama", "--batch", "8",
+         "--prompt-len", "32", "--steps", "2", "--warmup", "1",
+         "--ttft-iters", "1"],
+        cwd=ROOT, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    j = json.loads(lines[0])
+    assert j["metric"].startswith("tokens/sec (engine-direct")
+    assert j["steps"] == 2 and j["value"] > 0 and j["ms_per_step"] > 0
+    assert j["config"]["model"] == "tiny-llama"
