@@ -474,3 +474,24 @@ def test_schema_bounded_integer_range():
     v3.feed(ord("5"))                      # 5 -> 5x..5xxx reachable
     assert ord("9") in v3.allowed()[0]     # 59 in range
     assert not v3.allowed()[1]             # 5 itself is below minimum
+
+
+def test_engine_guided_on_moe_model():
+    """Guided decoding is model-family independent: the MoE engine obeys
+    the same mask (sampling sits after the MoE forward)."""
+    eng = LLMEngine(EngineConfig(model="tiny-moe", max_num_seqs=2,
+                                 max_num_batched_tokens=256,
+                                 max_model_len=512, num_gpu_blocks=128,
+                                 seed=0), eos_token_id=2)
+    outs = eng.generate(
+        [[1, 5, 6]],
+        SamplingParams(temperature=1.0, max_tokens=200, seed=2,
+                       response_schema={
+                           "type": "object", "required": ["n"],
+                           "properties": {"n": {"type": "integer",
+                                                "minimum": 0,
+                                                "maximum": 42}}}))
+    toks = outs[0]
+    assert toks[-1] == 2, toks
+    j = json.loads(bytes(t - 4 for t in toks if t != 2).decode())
+    assert set(j) == {"n"} and 0 <= j["n"] <= 42
